@@ -1,0 +1,417 @@
+"""Gateway filter-config schema.
+
+This is the versioned contract between the control plane (controller /
+``aigw translate`` / autoconfig) and the data plane (the per-GPU gateway
+shard), mirroring the reference's internal/filterapi/filterconfig.go:24-220:
+version+UUID gate, global and route-scoped LLM request costs, backends with
+schema+auth+mutations, declared models, and MCP config. It is deliberately
+not tied to Kubernetes (filterconfig.go:6-12) so the data plane can be
+tested standalone.
+
+One deliberate divergence: because this gateway has no Envoy in front of it,
+each backend carries its concrete ``upstream`` address (the reference
+resolves backends to Envoy clusters via xDS instead). Rate-limit rules are
+also first-class here (the reference delegates enforcement to the external
+Envoy ratelimit service + Redis; we enforce in-process with RCCL-synced
+counters — SURVEY.md §5.8).
+"""
+
+from __future__ import annotations
+
+import enum
+from dataclasses import dataclass, field
+from typing import Optional
+
+import yaml
+
+CURRENT_CONFIG_VERSION = "v1"
+
+
+class ConfigError(ValueError):
+    pass
+
+
+class APISchemaName(str, enum.Enum):
+    """Provider API schema names (filterconfig.go VersionedAPISchema)."""
+
+    OPENAI = "OpenAI"
+    AWS_BEDROCK = "AWSBedrock"
+    AWS_ANTHROPIC = "AWSAnthropic"
+    AZURE_OPENAI = "AzureOpenAI"
+    GCP_VERTEX_AI = "GCPVertexAI"
+    GCP_ANTHROPIC = "GCPAnthropic"
+    ANTHROPIC = "Anthropic"
+    COHERE = "Cohere"
+
+
+@dataclass
+class APISchema:
+    name: APISchemaName = APISchemaName.OPENAI
+    version: str = ""  # e.g. Azure api-version, GCP region qualifier
+
+
+class LLMRequestCostType(str, enum.Enum):
+    INPUT_TOKEN = "InputToken"
+    OUTPUT_TOKEN = "OutputToken"
+    TOTAL_TOKEN = "TotalToken"
+    CACHED_INPUT_TOKEN = "CachedInputToken"
+    CEL = "CEL"
+
+
+@dataclass
+class LLMRequestCost:
+    """A named cost extracted per request (filterconfig.go LLMRequestCost)."""
+
+    metadata_key: str
+    type: LLMRequestCostType = LLMRequestCostType.OUTPUT_TOKEN
+    cel: str = ""
+
+
+@dataclass
+class HeaderMatch:
+    """Route-rule header match. value="" + regex="" means presence match."""
+
+    name: str
+    value: str = ""
+    regex: str = ""
+
+
+@dataclass
+class HeaderMutation:
+    set: dict[str, str] = field(default_factory=dict)
+    remove: list[str] = field(default_factory=list)
+
+
+@dataclass
+class BodyMutation:
+    """JSON field set/remove on the request body (bodymutator, sjson paths)."""
+
+    set: dict[str, object] = field(default_factory=dict)  # dotted path -> value
+    remove: list[str] = field(default_factory=list)
+
+
+@dataclass
+class BackendAuth:
+    """Upstream credential injection (filterconfig.go BackendAuth).
+
+    Exactly one of the kinds should be populated.
+    """
+
+    api_key: str = ""  # -> Authorization: Bearer (backendauth/api_key.go)
+    anthropic_api_key: str = ""  # -> x-api-key (anthropicapikey.go)
+    azure_api_key: str = ""  # -> api-key (azureapikey.go)
+    azure_access_token: str = ""  # -> Authorization: Bearer (azure.go)
+    gcp_access_token: str = ""  # -> Authorization: Bearer + URL rewrite (gcp.go)
+    gcp_project: str = ""
+    gcp_region: str = ""
+    aws_access_key_id: str = ""  # -> SigV4 signing (aws.go:86-160)
+    aws_secret_access_key: str = ""
+    aws_session_token: str = ""
+    aws_region: str = ""
+
+    @property
+    def kind(self) -> str:
+        for k in (
+            "api_key",
+            "anthropic_api_key",
+            "azure_api_key",
+            "azure_access_token",
+            "gcp_access_token",
+            "aws_access_key_id",
+        ):
+            if getattr(self, k):
+                return k
+        return ""
+
+
+@dataclass
+class Upstream:
+    """Concrete upstream address for a backend (replaces Envoy clusters)."""
+
+    host: str = "127.0.0.1"
+    port: int = 443
+    tls: bool = False
+    path_prefix: str = ""  # prepended to the translated path
+    hostname: str = ""  # Host header / SigV4 signing host; defaults to host
+
+    @property
+    def authority(self) -> str:
+        h = self.hostname or self.host
+        return f"{h}:{self.port}"
+
+    @property
+    def base_url(self) -> str:
+        scheme = "https" if self.tls else "http"
+        return f"{scheme}://{self.host}:{self.port}{self.path_prefix}"
+
+
+@dataclass
+class Backend:
+    """One upstream provider binding (filterconfig.go Backend)."""
+
+    name: str
+    schema: APISchema = field(default_factory=APISchema)
+    upstream: Upstream = field(default_factory=Upstream)
+    weight: int = 1
+    priority: int = 0  # 0 = primary; >0 = fallback tiers (ai_gateway_route.go:387-397)
+    model_name_override: str = ""
+    auth: Optional[BackendAuth] = None
+    header_mutation: Optional[HeaderMutation] = None
+    body_mutation: Optional[BodyMutation] = None
+    # Per-try timeout seconds (extensionserver/post_translate_modify.go:206-300)
+    timeout_s: float = 60.0
+
+
+@dataclass
+class RateLimitRule:
+    """Token-budget rule enforced by the gateway (reference: QuotaPolicy +
+    Envoy ratelimit service; here enforced in-process, RCCL-synced)."""
+
+    name: str
+    metadata_key: str  # which LLMRequestCost feeds this bucket
+    limit: int  # tokens per window, across ALL shards
+    window_s: float = 60.0
+    key_headers: list[str] = field(default_factory=list)  # bucket per header value
+
+
+@dataclass
+class Route:
+    """One routing rule (filterconfig.go Config.Rules).
+
+    Matched in order of declaration against the model header + any extra
+    header matches; the first match wins (HTTPRoute semantics).
+    """
+
+    name: str
+    backends: list[Backend] = field(default_factory=list)
+    headers: list[HeaderMatch] = field(default_factory=list)
+    model_name_override: str = ""
+    request_costs: list[LLMRequestCost] = field(default_factory=list)
+    header_mutation: Optional[HeaderMutation] = None
+    # Retry budget across priority tiers (numAttemptsPerPriority analogue).
+    retries: int = 1
+
+
+@dataclass
+class Model:
+    """Declared model served from /v1/models (models_processor.go:40-62)."""
+
+    name: str
+    owned_by: str = "aigw"
+    created_at: int = 0
+    hosts: list[str] = field(default_factory=list)  # empty = unscoped
+
+
+@dataclass
+class MCPBackend:
+    name: str
+    upstream: Upstream = field(default_factory=Upstream)
+    path: str = "/mcp"
+    tool_include: list[str] = field(default_factory=list)
+    tool_exclude: list[str] = field(default_factory=list)
+    headers: Optional[HeaderMutation] = None
+    auth: Optional[BackendAuth] = None
+
+
+@dataclass
+class MCPRoute:
+    name: str
+    path: str = "/mcp"
+    backends: list[MCPBackend] = field(default_factory=list)
+
+
+@dataclass
+class MCPConfig:
+    routes: list[MCPRoute] = field(default_factory=list)
+    session_seed: str = "aigw-mcp"
+
+
+@dataclass
+class Config:
+    """Root gateway config (filterconfig.go Config)."""
+
+    version: str = CURRENT_CONFIG_VERSION
+    uuid: str = ""
+    model_name_header_key: str = "x-ai-eg-model"
+    routes: list[Route] = field(default_factory=list)
+    models: list[Model] = field(default_factory=list)
+    llm_request_costs: list[LLMRequestCost] = field(default_factory=list)
+    rate_limits: list[RateLimitRule] = field(default_factory=list)
+    mcp: Optional[MCPConfig] = None
+
+
+# --- YAML (de)serialization -------------------------------------------------
+
+
+def _dc(cls, d: dict, ctx: str):
+    if not isinstance(d, dict):
+        raise ConfigError(f"{ctx}: expected mapping, got {type(d).__name__}")
+    fields = {f: t for f, t in cls.__dataclass_fields__.items()}  # noqa: C416
+    kwargs = {}
+    for key, val in d.items():
+        pykey = _snake(key)
+        if pykey not in fields:
+            raise ConfigError(f"{ctx}: unknown field {key!r}")
+        kwargs[pykey] = val
+    return kwargs
+
+
+def _snake(name: str) -> str:
+    out = []
+    for i, ch in enumerate(name):
+        if ch.isupper():
+            if i and (not name[i - 1].isupper() or (i + 1 < len(name) and name[i + 1].islower())):
+                out.append("_")
+            out.append(ch.lower())
+        else:
+            out.append(ch)
+    return "".join(out)
+
+
+def _parse_schema(d, ctx) -> APISchema:
+    if isinstance(d, str):
+        d = {"name": d}
+    kw = _dc(APISchema, d, ctx)
+    try:
+        kw["name"] = APISchemaName(kw.get("name", "OpenAI"))
+    except ValueError as e:
+        raise ConfigError(f"{ctx}: unknown schema {kw.get('name')!r}") from e
+    return APISchema(**kw)
+
+
+def _parse_cost(d, ctx) -> LLMRequestCost:
+    kw = _dc(LLMRequestCost, d, ctx)
+    if "metadata_key" not in kw:
+        raise ConfigError(f"{ctx}: metadataKey is required")
+    kw["type"] = LLMRequestCostType(kw.get("type", "OutputToken"))
+    c = LLMRequestCost(**kw)
+    if c.type is LLMRequestCostType.CEL:
+        if not c.cel:
+            raise ConfigError(f"{ctx}: CEL cost requires 'cel' expression")
+        from aigw.llmcost import compile_program
+        from aigw.llmcost.cel import CostExpressionError
+
+        try:
+            compile_program(c.cel)  # validate at load time (gateway.go:201)
+        except CostExpressionError as e:
+            raise ConfigError(f"{ctx}: invalid CEL expression: {e}") from e
+    return c
+
+
+def _parse_backend(d, ctx) -> Backend:
+    kw = _dc(Backend, d, ctx)
+    if "name" not in kw:
+        raise ConfigError(f"{ctx}: backend name is required")
+    if "schema" in kw:
+        kw["schema"] = _parse_schema(kw["schema"], f"{ctx}.schema")
+    if "upstream" in kw:
+        kw["upstream"] = Upstream(**_dc(Upstream, kw["upstream"], f"{ctx}.upstream"))
+    if kw.get("auth") is not None:
+        kw["auth"] = BackendAuth(**_dc(BackendAuth, kw["auth"], f"{ctx}.auth"))
+    if kw.get("header_mutation") is not None:
+        kw["header_mutation"] = HeaderMutation(
+            **_dc(HeaderMutation, kw["header_mutation"], f"{ctx}.headerMutation")
+        )
+    if kw.get("body_mutation") is not None:
+        kw["body_mutation"] = BodyMutation(
+            **_dc(BodyMutation, kw["body_mutation"], f"{ctx}.bodyMutation")
+        )
+    return Backend(**kw)
+
+
+def _parse_route(d, ctx) -> Route:
+    kw = _dc(Route, d, ctx)
+    if "name" not in kw:
+        raise ConfigError(f"{ctx}: route name is required")
+    kw["backends"] = [
+        _parse_backend(b, f"{ctx}.backends[{i}]") for i, b in enumerate(kw.get("backends", []))
+    ]
+    kw["headers"] = [
+        HeaderMatch(**_dc(HeaderMatch, h, f"{ctx}.headers[{i}]"))
+        for i, h in enumerate(kw.get("headers", []))
+    ]
+    kw["request_costs"] = [
+        _parse_cost(c, f"{ctx}.requestCosts[{i}]") for i, c in enumerate(kw.get("request_costs", []))
+    ]
+    if kw.get("header_mutation") is not None:
+        kw["header_mutation"] = HeaderMutation(
+            **_dc(HeaderMutation, kw["header_mutation"], f"{ctx}.headerMutation")
+        )
+    return Route(**kw)
+
+
+def load_config(data: object) -> Config:
+    """Build a Config from parsed YAML/JSON, validating every field."""
+
+    if isinstance(data, (str, bytes)):
+        data = yaml.safe_load(data)
+    if not isinstance(data, dict):
+        raise ConfigError("config root must be a mapping")
+    kw = _dc(Config, data, "config")
+    version = kw.get("version", CURRENT_CONFIG_VERSION)
+    if version != CURRENT_CONFIG_VERSION:
+        # Version gate against mixed-version corruption during rolling
+        # upgrades (filterconfig.go:26-31).
+        raise ConfigError(f"unsupported config version {version!r}")
+    kw["routes"] = [_parse_route(r, f"routes[{i}]") for i, r in enumerate(kw.get("routes", []))]
+    kw["models"] = [
+        Model(**_dc(Model, m, f"models[{i}]")) for i, m in enumerate(kw.get("models", []))
+    ]
+    kw["llm_request_costs"] = [
+        _parse_cost(c, f"llmRequestCosts[{i}]")
+        for i, c in enumerate(kw.get("llm_request_costs", []))
+    ]
+    kw["rate_limits"] = [
+        RateLimitRule(**_dc(RateLimitRule, r, f"rateLimits[{i}]"))
+        for i, r in enumerate(kw.get("rate_limits", []))
+    ]
+    if kw.get("mcp") is not None:
+        mkw = _dc(MCPConfig, kw["mcp"], "mcp")
+        routes = []
+        for i, r in enumerate(mkw.get("routes", [])):
+            rkw = _dc(MCPRoute, r, f"mcp.routes[{i}]")
+            backends = []
+            for j, b in enumerate(rkw.get("backends", [])):
+                bkw = _dc(MCPBackend, b, f"mcp.routes[{i}].backends[{j}]")
+                if "upstream" in bkw:
+                    bkw["upstream"] = Upstream(**_dc(Upstream, bkw["upstream"], "upstream"))
+                if bkw.get("headers") is not None:
+                    bkw["headers"] = HeaderMutation(**_dc(HeaderMutation, bkw["headers"], "headers"))
+                if bkw.get("auth") is not None:
+                    bkw["auth"] = BackendAuth(**_dc(BackendAuth, bkw["auth"], "auth"))
+                backends.append(MCPBackend(**bkw))
+            rkw["backends"] = backends
+            routes.append(MCPRoute(**rkw))
+        mkw["routes"] = routes
+        kw["mcp"] = MCPConfig(**mkw)
+    cfg = Config(**kw)
+    _validate(cfg)
+    return cfg
+
+
+def _validate(cfg: Config) -> None:
+    cost_keys = {c.metadata_key for c in cfg.llm_request_costs}
+    for r in cfg.routes:
+        if not r.backends:
+            raise ConfigError(f"route {r.name!r} has no backends")
+        for c in r.request_costs:
+            cost_keys.add(c.metadata_key)
+        names = [b.name for b in r.backends]
+        if len(set(names)) != len(names):
+            raise ConfigError(f"route {r.name!r} has duplicate backend names")
+    for rl in cfg.rate_limits:
+        if rl.metadata_key not in cost_keys and rl.metadata_key not in (
+            "llm_input_token",
+            "llm_output_token",
+            "llm_total_token",
+        ):
+            raise ConfigError(
+                f"rateLimit {rl.name!r} references unknown cost metadata key {rl.metadata_key!r}"
+            )
+        if rl.limit <= 0:
+            raise ConfigError(f"rateLimit {rl.name!r}: limit must be positive")
+
+
+def load_config_file(path: str) -> Config:
+    with open(path, "r", encoding="utf-8") as f:
+        return load_config(yaml.safe_load(f))

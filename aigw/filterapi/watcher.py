@@ -1,0 +1,84 @@
+"""Config file watcher with hot reload.
+
+Mirrors internal/filterapi/watcher.go:48-160: poll the config path on a
+fixed tick (default 5 s, cmd/extproc/mainlib/main.go:357), compare the
+config UUID (falling back to content hash when the UUID is empty), and on
+change compile a fresh RuntimeConfig and hand it to the update callback.
+In-flight requests keep the RuntimeConfig they already resolved; new
+requests see the swapped pointer — reload never drops streams
+(SURVEY.md §5.4).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import hashlib
+import logging
+from typing import Awaitable, Callable, Optional
+
+import yaml
+
+from aigw.filterapi.config import load_config
+from aigw.filterapi.runtime import RuntimeConfig
+
+logger = logging.getLogger("aigw.filterapi.watcher")
+
+
+class ConfigWatcher:
+    def __init__(
+        self,
+        path: str,
+        on_update: Callable[[RuntimeConfig], Optional[Awaitable[None]]],
+        tick_s: float = 5.0,
+    ):
+        self.path = path
+        self.on_update = on_update
+        self.tick_s = tick_s
+        self._last_key: str = ""
+        self._task: Optional[asyncio.Task] = None
+
+    def load_once(self) -> RuntimeConfig:
+        """Synchronous initial load; raises on invalid config."""
+        with open(self.path, "rb") as f:
+            raw = f.read()
+        cfg = load_config(yaml.safe_load(raw))
+        self._last_key = cfg.uuid or hashlib.sha256(raw).hexdigest()
+        return RuntimeConfig(cfg)
+
+    async def start(self) -> None:
+        self._task = asyncio.create_task(self._loop(), name="aigw-config-watcher")
+
+    async def stop(self) -> None:
+        if self._task:
+            self._task.cancel()
+            try:
+                await self._task
+            except asyncio.CancelledError:
+                pass
+            self._task = None
+
+    async def _loop(self) -> None:
+        while True:
+            await asyncio.sleep(self.tick_s)
+            try:
+                await self._check()
+            except asyncio.CancelledError:
+                raise
+            except Exception:  # keep serving on a bad reload (watcher.go:145)
+                logger.exception("config reload failed; keeping previous config")
+
+    async def _check(self) -> None:
+        with open(self.path, "rb") as f:
+            raw = f.read()
+        data = yaml.safe_load(raw)
+        uuid = (data or {}).get("uuid", "") if isinstance(data, dict) else ""
+        key = uuid or hashlib.sha256(raw).hexdigest()
+        if key == self._last_key:
+            return
+        cfg = load_config(data)
+        rc = RuntimeConfig(cfg)
+        self._last_key = key
+        logger.info("config reloaded (uuid=%s)", key[:12])
+        res = self.on_update(rc)
+        if res is not None:
+            await res
