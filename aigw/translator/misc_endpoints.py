@@ -1,0 +1,211 @@
+"""Remaining endpoint translators: Cohere rerank, Anthropic count_tokens
+family, audio transcription/translation multipart handling.
+
+Parity targets: internal/translator/cohere_rerank_v2.go, counttokens_*.go,
+openai_transcription.go / openai_translation.go + multipart_helper.go.
+"""
+
+from __future__ import annotations
+
+import json
+import re
+import secrets
+
+from aigw.filterapi.config import APISchemaName
+from aigw.translator.base import (
+    RequestTranslation,
+    ResponseTranslation,
+    Translator,
+    Usage,
+    jdump,
+    register,
+)
+
+
+@register("/v2/rerank", APISchemaName.COHERE)
+class CohereRerank(Translator):
+    """Cohere /v2/rerank passthrough with model override + billed-unit
+    usage extraction (cohere_rerank_v2.go)."""
+
+    def __init__(self, **kw):
+        self._model = ""
+
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+        if model_override:
+            body["model"] = model_override
+        self._model = body.get("model", "")
+        return RequestTranslation(path="/v2/rerank", body=jdump(body))
+
+    def response_body(self, status, body):
+        try:
+            resp = json.loads(body)
+            bu = ((resp.get("meta") or {}).get("billed_units") or {})
+            usage = Usage(
+                input_tokens=bu.get("input_tokens", 0) or 0,
+                output_tokens=bu.get("output_tokens", 0) or 0,
+                total_tokens=(bu.get("input_tokens", 0) or 0) + (bu.get("output_tokens", 0) or 0),
+            )
+        except ValueError:
+            usage = Usage()
+        return ResponseTranslation(
+            body=body, usage=usage, response_model=self._model, end_of_stream=True
+        )
+
+
+class _CountTokensBase(Translator):
+    """Anthropic /v1/messages/count_tokens family (counttokens_*.go)."""
+
+    def __init__(self, api_version: str = "", gcp_project: str = "", gcp_region: str = ""):
+        self.gcp_project = gcp_project
+        self.gcp_region = gcp_region
+        self._model = ""
+
+    def _path(self, model):
+        return "/v1/messages/count_tokens"
+
+    def _adjust(self, body, model):
+        return body
+
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+        if model_override:
+            body["model"] = model_override
+        self._model = body.get("model", "")
+        return RequestTranslation(path=self._path(self._model), body=jdump(self._adjust(body, self._model)))
+
+    def response_body(self, status, body):
+        try:
+            resp = json.loads(body)
+            tokens = resp.get("input_tokens", 0) or 0
+        except ValueError:
+            tokens = 0
+        return ResponseTranslation(
+            body=body,
+            usage=Usage(input_tokens=tokens, total_tokens=tokens),
+            response_model=self._model,
+            end_of_stream=True,
+        )
+
+
+@register("/anthropic/v1/messages/count_tokens", APISchemaName.ANTHROPIC)
+class AnthropicCountTokens(_CountTokensBase):
+    pass
+
+
+@register("/anthropic/v1/messages/count_tokens", APISchemaName.GCP_ANTHROPIC)
+class GCPAnthropicCountTokens(_CountTokensBase):
+    def _path(self, model):
+        return (
+            f"/v1/projects/{self.gcp_project}/locations/{self.gcp_region}"
+            f"/publishers/anthropic/models/count-tokens:rawPredict"
+        )
+
+    def _adjust(self, body, model):
+        body["anthropic_version"] = "vertex-2023-10-16"
+        return body
+
+
+# --- multipart audio endpoints -------------------------------------------------
+
+
+def _parse_multipart(body: bytes, content_type: str) -> tuple[list[tuple[str, dict, bytes]], str]:
+    """Minimal multipart/form-data parser (endpointspec.go:872-1076).
+    Returns ([(name, part_headers, payload)], boundary)."""
+    m = re.search(r'boundary="?([^";]+)"?', content_type)
+    if not m:
+        raise ValueError("missing multipart boundary")
+    boundary = m.group(1)
+    delim = b"--" + boundary.encode()
+    parts = []
+    segments = body.split(delim)
+    for seg in segments[1:-1]:
+        seg = seg.lstrip(b"\r\n")
+        if not seg or seg.startswith(b"--"):
+            continue
+        header_blob, _, payload = seg.partition(b"\r\n\r\n")
+        payload = payload[:-2] if payload.endswith(b"\r\n") else payload
+        headers: dict[str, str] = {}
+        name = ""
+        for line in header_blob.split(b"\r\n"):
+            k, _, v = line.partition(b":")
+            headers[k.decode().lower().strip()] = v.decode().strip()
+        cd = headers.get("content-disposition", "")
+        nm = re.search(r'name="([^"]*)"', cd)
+        if nm:
+            name = nm.group(1)
+        parts.append((name, headers, payload))
+    return parts, boundary
+
+
+def _encode_multipart(parts: list[tuple[str, dict, bytes]], boundary: str) -> bytes:
+    out = bytearray()
+    for name, headers, payload in parts:
+        out.extend(b"--" + boundary.encode() + b"\r\n")
+        for k, v in headers.items():
+            out.extend(f"{k}: {v}\r\n".encode())
+        out.extend(b"\r\n")
+        out.extend(payload)
+        out.extend(b"\r\n")
+    out.extend(b"--" + boundary.encode() + b"--\r\n")
+    return bytes(out)
+
+
+class _MultipartAudio(Translator):
+    """Audio transcription/translation: multipart body; model override
+    requires re-encoding the form (multipart_helper.go)."""
+
+    PATH = "/v1/audio/transcriptions"
+
+    def __init__(self, **kw):
+        self._model = ""
+        self.content_type = ""
+
+    def request_multipart(self, body: bytes, content_type: str, *, model_override=""):
+        parts, boundary = _parse_multipart(body, content_type)
+        model = ""
+        for i, (name, headers, payload) in enumerate(parts):
+            if name == "model":
+                model = payload.decode("utf-8", "replace")
+                if model_override:
+                    parts[i] = (name, headers, model_override.encode())
+                    model = model_override
+        self._model = model
+        if model_override:
+            new_boundary = boundary
+            body = _encode_multipart(parts, new_boundary)
+        return RequestTranslation(
+            path=self.PATH,
+            body=body,
+            headers={"content-type": f"multipart/form-data; boundary={boundary}"},
+        ), model
+
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+        raise NotImplementedError("multipart endpoints use request_multipart")
+
+    def response_body(self, status, body):
+        try:
+            resp = json.loads(body)
+            u = resp.get("usage") or {}
+            usage = Usage(
+                input_tokens=u.get("input_tokens", 0) or 0,
+                output_tokens=u.get("output_tokens", 0) or 0,
+                total_tokens=u.get("total_tokens", 0) or 0,
+            )
+        except ValueError:
+            usage = Usage()
+        return ResponseTranslation(
+            body=body, usage=usage, response_model=self._model, end_of_stream=True
+        )
+
+
+@register("/v1/audio/transcriptions", APISchemaName.OPENAI)
+class OpenAITranscription(_MultipartAudio):
+    PATH = "/v1/audio/transcriptions"
+
+
+@register("/v1/audio/translations", APISchemaName.OPENAI)
+class OpenAITranslation(_MultipartAudio):
+    PATH = "/v1/audio/translations"
+
+
+def make_boundary() -> str:
+    return "aigw" + secrets.token_hex(16)
