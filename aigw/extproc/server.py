@@ -1,0 +1,616 @@
+"""The gateway data plane: one asyncio shard, optionally pinned to one GPU.
+
+Replaces the reference's Envoy + router-extproc + upstream-extproc sandwich
+(internal/extproc/server.go, processor_impl.go) with a single in-process
+pipeline per request:
+
+    parse body → model extract → route match (model header + rules)
+    → rate-limit pre-check → semantic-cache lookup
+    → per-try: translate request → header/body mutations → auth signing
+      → upstream dispatch → (retriable failure? next backend, re-translated
+        from the ORIGINAL body with original headers restored — A.8)
+    → response translation (unary or streamed re-chunking)
+    → usage extraction → cost programs → rate-limit charge → metrics
+
+Invariants preserved from the reference: client-spoofable
+x-ai-eg-* headers stripped at ingress; model routing happens after body
+parse and before backend selection; auth signs the FINAL mutated body;
+gateway-generated local replies are never re-translated; content-length is
+dropped on streamed responses.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import time
+import uuid
+from typing import Optional
+
+import aiohttp
+from aiohttp import web
+
+from aigw import internalapi
+from aigw.backendauth import build_auth_handler
+from aigw.extproc.router import RETRIABLE_STATUSES, backend_attempts, max_attempts
+from aigw.filterapi.config import APISchemaName, Backend
+from aigw.filterapi.runtime import CompiledRoute, RuntimeConfig
+from aigw.llmcost import CostVars
+from aigw.metrics import GenAIMetrics
+from aigw.metrics.genai import provider_from_schema
+from aigw.mutator import apply_body_mutation, apply_header_mutation
+from aigw.ratelimit import RateLimiter
+from aigw.translator import TranslationError, Usage, get_translator
+
+logger = logging.getLogger("aigw.server")
+
+# endpoint key -> (operation name for metrics)
+JSON_ENDPOINTS = {
+    "/v1/chat/completions": "chat",
+    "/v1/completions": "text_completion",
+    "/v1/embeddings": "embeddings",
+    "/v1/images/generations": "image_generation",
+    "/v1/responses": "responses",
+    "/v1/audio/speech": "speech",
+    "/tokenize": "tokenize",
+    "/anthropic/v1/messages": "chat",
+    "/anthropic/v1/messages/count_tokens": "count_tokens",
+    "/v2/rerank": "rerank",
+}
+MULTIPART_ENDPOINTS = {
+    "/v1/audio/transcriptions": "transcription",
+    "/v1/audio/translations": "translation",
+}
+# endpoints whose requests may stream responses
+STREAMABLE = {"/v1/chat/completions", "/v1/completions", "/v1/responses", "/anthropic/v1/messages"}
+
+
+# Headers never forwarded upstream (proxy hop-by-hop semantics + the
+# model-header internals, which travel via route state instead).
+_HOP_BY_HOP = frozenset(
+    {
+        "host",
+        "content-length",
+        "content-type",
+        "connection",
+        "keep-alive",
+        "proxy-authenticate",
+        "proxy-authorization",
+        "te",
+        "trailer",
+        "transfer-encoding",
+        "upgrade",
+        "expect",
+        "accept-encoding",
+        internalapi.MODEL_NAME_HEADER,
+        internalapi.ORIGINAL_PATH_HEADER,
+    }
+)
+
+
+def _json_error(status: int, message: str, err_type: str = "invalid_request_error") -> web.Response:
+    return web.json_response(
+        {"error": {"message": message, "type": err_type, "code": str(status)}},
+        status=status,
+    )
+
+
+class GatewayServer:
+    def __init__(
+        self,
+        runtime: RuntimeConfig,
+        *,
+        metrics: Optional[GenAIMetrics] = None,
+        limiter: Optional[RateLimiter] = None,
+        gpu_services=None,
+        max_body_bytes: int = 50 * 1024 * 1024,  # reference raises Envoy's buffer to 50MiB
+    ):
+        self.runtime = runtime
+        self.metrics = metrics or GenAIMetrics()
+        self.limiter = limiter or RateLimiter(runtime.rate_limits)
+        self.gpu = gpu_services  # aigw.gpu.services.GPUServices or None
+        self.max_body_bytes = max_body_bytes
+        self._session: Optional[aiohttp.ClientSession] = None
+        self._started_at = time.time()
+
+    # ---- lifecycle -----------------------------------------------------------
+
+    async def start(self) -> None:
+        if self._session is None:
+            connector = aiohttp.TCPConnector(limit=0, ttl_dns_cache=300, keepalive_timeout=75)
+            self._session = aiohttp.ClientSession(
+                connector=connector, auto_decompress=True, skip_auto_headers=("User-Agent",)
+            )
+
+    async def close(self) -> None:
+        if self._session is not None:
+            await self._session.close()
+            self._session = None
+
+    def swap_runtime(self, rc: RuntimeConfig) -> None:
+        """Hot reload: new requests see the new config; in-flight requests
+        keep the runtime they resolved (server.go:81-88 semantics)."""
+        self.runtime = rc
+        self.limiter.rules = {r.name: r for r in rc.rate_limits}
+
+    def make_app(self) -> web.Application:
+        app = web.Application(client_max_size=self.max_body_bytes)
+        for ep in JSON_ENDPOINTS:
+            app.router.add_post(ep, self._make_handler(ep))
+        for ep in MULTIPART_ENDPOINTS:
+            app.router.add_post(ep, self._make_multipart_handler(ep))
+        app.router.add_get("/v1/models", self._handle_models)
+        app.router.add_get("/anthropic/v1/models", self._handle_anthropic_models)
+        app.router.add_get("/health", self._handle_health)
+        app.router.add_get("/metrics", self._handle_metrics)
+        if self.gpu is not None:
+            app.router.add_post("/v1/gateway/tokenize", self._handle_gpu_tokenize)
+        app.on_startup.append(lambda _app: self.start())
+        app.on_cleanup.append(lambda _app: self.close())
+        return app
+
+    # ---- admin ---------------------------------------------------------------
+
+    async def _handle_health(self, request: web.Request) -> web.Response:
+        return web.json_response({"status": "ok", "uptime_s": time.time() - self._started_at})
+
+    async def _handle_metrics(self, request: web.Request) -> web.Response:
+        return web.Response(body=self.metrics.render(), content_type="text/plain")
+
+    async def _handle_models(self, request: web.Request) -> web.Response:
+        rt = self.runtime
+        models = rt.models_for_host(request.headers.get("host", request.host or ""))
+        return web.json_response(
+            {
+                "object": "list",
+                "data": [
+                    {
+                        "id": m.name,
+                        "object": "model",
+                        "created": m.created_at,
+                        "owned_by": m.owned_by,
+                    }
+                    for m in models
+                ],
+            }
+        )
+
+    async def _handle_anthropic_models(self, request: web.Request) -> web.Response:
+        rt = self.runtime
+        models = rt.models_for_host(request.headers.get("host", request.host or ""))
+        return web.json_response(
+            {
+                "data": [
+                    {
+                        "id": m.name,
+                        "type": "model",
+                        "display_name": m.name,
+                        "created_at": m.created_at,
+                    }
+                    for m in models
+                ],
+                "has_more": False,
+            }
+        )
+
+    async def _handle_gpu_tokenize(self, request: web.Request) -> web.Response:
+        body = await request.json()
+        text = body.get("prompt") or body.get("text") or ""
+        ids = await self.gpu.tokenize(text)
+        return web.json_response({"count": len(ids), "tokens": ids})
+
+    # ---- request processing --------------------------------------------------
+
+    def _make_handler(self, endpoint: str):
+        async def handler(request: web.Request) -> web.StreamResponse:
+            return await self._process(request, endpoint)
+
+        return handler
+
+    def _make_multipart_handler(self, endpoint: str):
+        async def handler(request: web.Request) -> web.StreamResponse:
+            return await self._process_multipart(request, endpoint)
+
+        return handler
+
+    @staticmethod
+    def _ingress_headers(request: web.Request) -> dict[str, str]:
+        headers = {k.lower(): v for k, v in request.headers.items()}
+        for h in internalapi.INTERNAL_HEADERS:
+            headers.pop(h, None)  # spoof protection (server.go:439-455)
+        return headers
+
+    def _translator_kwargs(self, backend: Backend) -> dict:
+        kw: dict = {"api_version": backend.schema.version}
+        if backend.auth is not None:
+            kw["gcp_project"] = backend.auth.gcp_project
+            kw["gcp_region"] = backend.auth.gcp_region
+        return kw
+
+    async def _process(self, request: web.Request, endpoint: str) -> web.StreamResponse:
+        start = time.monotonic()
+        rt = self.runtime
+        try:
+            raw = await request.read()
+            body = json.loads(raw) if raw else {}
+            if not isinstance(body, dict):
+                raise ValueError("body must be a JSON object")
+        except ValueError as e:
+            return _json_error(400, f"invalid request body: {e}")
+
+        headers = self._ingress_headers(request)
+        model = str(body.get("model", ""))
+        headers[rt.model_header] = model
+        headers[internalapi.ORIGINAL_PATH_HEADER] = request.path
+        stream = bool(body.get("stream")) and endpoint in STREAMABLE
+
+        route = rt.select_route(headers)
+        if route is None:
+            return _json_error(404, f"no route matched model {model!r}", "model_not_found")
+
+        # usage-based rate-limit pre-admission
+        decision = self.limiter.check(headers)
+        if not decision.allowed:
+            self.metrics.ratelimit_denials.labels(rule=decision.rule).inc()
+            resp = _json_error(429, "token budget exhausted", "rate_limit_exceeded")
+            resp.headers["retry-after"] = str(int(decision.retry_after_s) + 1)
+            return resp
+
+        # GPU-side input token count (admission accounting / local usage).
+        gpu_input_tokens = 0
+        if self.gpu is not None and endpoint in ("/v1/chat/completions", "/anthropic/v1/messages"):
+            gpu_input_tokens = await self.gpu.count_request_tokens(body)
+
+        # semantic response cache (GPU MFMA embed + HBM index)
+        cache_key_vec = None
+        if (
+            self.gpu is not None
+            and self.gpu.cache_enabled
+            and endpoint == "/v1/chat/completions"
+            and not stream
+        ):
+            hit, cache_key_vec = await self.gpu.cache_lookup(body)
+            if hit is not None:
+                self.metrics.cache_events.labels(event="hit").inc()
+                resp = web.Response(body=hit, content_type="application/json")
+                resp.headers["x-aigw-cache"] = "hit"
+                return resp
+            self.metrics.cache_events.labels(event="miss").inc()
+
+        return await self._dispatch(
+            request, endpoint, route, headers, body, stream, start,
+            model=model, gpu_input_tokens=gpu_input_tokens, cache_key_vec=cache_key_vec,
+        )
+
+    async def _dispatch(
+        self,
+        request: web.Request,
+        endpoint: str,
+        route: CompiledRoute,
+        headers: dict[str, str],
+        body: dict,
+        stream: bool,
+        start: float,
+        *,
+        model: str,
+        gpu_input_tokens: int = 0,
+        cache_key_vec=None,
+    ) -> web.StreamResponse:
+        rt = self.runtime
+        assert self._session is not None, "GatewayServer.start() not called"
+        attempts_left = max_attempts(route)
+        if attempts_left == 0:
+            return _json_error(503, "route has no backends", "no_backend")
+        last_error: Optional[str] = None
+        first = True
+
+        for backend in backend_attempts(route):
+            if attempts_left <= 0:
+                break
+            attempts_left -= 1
+            if not first:
+                self.metrics.retries_total.labels(route=route.route.name).inc()
+            first = False
+
+            try:
+                translator = get_translator(
+                    endpoint, backend.schema.name, **self._translator_kwargs(backend)
+                )
+            except TranslationError as e:
+                last_error = str(e)
+                continue
+
+            # Re-translate the ORIGINAL body per try (A.8 retry semantics).
+            body_copy = json.loads(json.dumps(body))
+            override = backend.model_name_override or route.route.model_name_override
+            try:
+                tr = translator.request(
+                    body_copy,
+                    model_override=override,
+                    stream=stream,
+                    force_include_usage=bool(route.costs),
+                )
+            except TranslationError as e:
+                return _json_error(422, str(e))
+
+            out_body = tr.body
+            if backend.body_mutation is not None:
+                doc = json.loads(out_body)
+                apply_body_mutation(doc, backend.body_mutation)
+                out_body = json.dumps(doc, separators=(",", ":")).encode()
+
+            # Headers: fresh copy of ORIGINAL client headers each try
+            # (headermutator restore-on-retry), then route/backend mutations,
+            # then translator-set headers, then auth over the FINAL body.
+            up_headers = {k: v for k, v in headers.items() if k not in _HOP_BY_HOP}
+            up_headers["content-type"] = "application/json"
+            up_headers = apply_header_mutation(up_headers, route.route.header_mutation)
+            up_headers = apply_header_mutation(up_headers, backend.header_mutation)
+            up_headers.update(tr.headers)
+            for h in tr.remove_headers:
+                up_headers.pop(h, None)
+            auth = build_auth_handler(backend)
+            if auth is not None:
+                up_headers = auth(up_headers, out_body, "POST", tr.path)
+            else:
+                # propagate client Authorization for auth-less backends
+                if "authorization" in headers:
+                    up_headers.setdefault("authorization", headers["authorization"])
+
+            url = backend.upstream.base_url + tr.path
+            if backend.upstream.hostname:
+                up_headers["host"] = backend.upstream.hostname
+            try:
+                timeout = aiohttp.ClientTimeout(total=backend.timeout_s)
+                async with self._session.post(
+                    url, data=out_body, headers=up_headers, timeout=timeout
+                ) as upstream:
+                    status = upstream.status
+                    if status >= 400:
+                        err_body = await upstream.read()
+                        if status in RETRIABLE_STATUSES and attempts_left > 0:
+                            last_error = f"upstream {backend.name} returned {status}"
+                            logger.warning("%s; trying next backend", last_error)
+                            continue
+                        out = translator.response_error(status, err_body, dict(upstream.headers))
+                        self._finish_metrics(
+                            endpoint, route, backend, model, "", Usage(), start,
+                            status=status, error_type=f"upstream_{status}",
+                        )
+                        return web.Response(
+                            body=out, status=status, content_type="application/json"
+                        )
+                    if stream:
+                        return await self._stream_response(
+                            request, endpoint, route, backend, translator, upstream,
+                            headers, model, start, gpu_input_tokens,
+                        )
+                    return await self._unary_response(
+                        endpoint, route, backend, translator, upstream, headers,
+                        model, start, gpu_input_tokens, body, cache_key_vec,
+                    )
+            except (aiohttp.ClientError, asyncio.TimeoutError) as e:
+                last_error = f"upstream {backend.name}: {type(e).__name__}: {e}"
+                logger.warning("%s; %d attempts left", last_error, attempts_left)
+                continue
+
+        self._finish_metrics(
+            endpoint, route, None, model, "", Usage(), start, status=503,
+            error_type="no_healthy_upstream",
+        )
+        return _json_error(503, last_error or "no healthy upstream", "upstream_error")
+
+    # ---- response paths ------------------------------------------------------
+
+    def _apply_costs(
+        self, route: CompiledRoute, backend: Backend, headers: dict[str, str],
+        model: str, usage: Usage, gpu_input_tokens: int,
+    ) -> dict[str, int]:
+        if gpu_input_tokens and not usage.input_tokens:
+            # upstream reported no usage; fall back to the gateway's own
+            # GPU-tokenized count
+            usage.input_tokens = gpu_input_tokens
+            usage.total_tokens = max(usage.total_tokens, usage.input_tokens + usage.output_tokens)
+        v = CostVars(
+            model=model,
+            backend=backend.name if backend else "",
+            route_name=route.route.name,
+            input_tokens=usage.input_tokens,
+            cached_input_tokens=usage.cached_input_tokens,
+            cache_creation_input_tokens=usage.cache_creation_input_tokens,
+            output_tokens=usage.output_tokens,
+            total_tokens=usage.total_tokens,
+            reasoning_tokens=usage.reasoning_tokens,
+        )
+        costs: dict[str, int] = {
+            internalapi.META_INPUT_TOKENS: usage.input_tokens,
+            internalapi.META_OUTPUT_TOKENS: usage.output_tokens,
+            internalapi.META_TOTAL_TOKENS: usage.total_tokens,
+        }
+        for cc in route.costs:
+            try:
+                costs[cc.metadata_key] = cc.evaluate(v)
+            except Exception:
+                logger.exception("cost %s failed", cc.metadata_key)
+        self.limiter.charge(headers, costs)
+        return costs
+
+    def _finish_metrics(
+        self, endpoint, route, backend, model, response_model, usage, start,
+        *, status: int, error_type: str = "", ttft: float = -1.0,
+    ) -> None:
+        provider = (
+            provider_from_schema(backend.schema.name.value, backend.name) if backend else ""
+        )
+        labels = self.metrics.labels(
+            operation=JSON_ENDPOINTS.get(endpoint, endpoint),
+            provider=provider,
+            original_model=model,
+            request_model=model,
+            response_model=response_model,
+        )
+        elapsed = time.monotonic() - start
+        self.metrics.record_request(labels, elapsed, error_type)
+        if usage.total_tokens or usage.input_tokens or usage.output_tokens:
+            self.metrics.record_tokens(labels, usage)
+        if ttft >= 0:
+            self.metrics.record_stream_latency(labels, ttft, elapsed, usage.output_tokens)
+        self.metrics.requests_total.labels(
+            endpoint=endpoint, backend=backend.name if backend else "", status=str(status)
+        ).inc()
+
+    async def _unary_response(
+        self, endpoint, route, backend, translator, upstream, headers, model,
+        start, gpu_input_tokens, orig_body, cache_key_vec,
+    ) -> web.Response:
+        data = await upstream.read()  # aiohttp auto-decompresses gzip/br
+        try:
+            rtl = translator.response_body(upstream.status, data)
+        except Exception as e:
+            logger.exception("response translation failed")
+            return _json_error(502, f"response translation failed: {e}", "translation_error")
+        usage = rtl.usage or Usage()
+        self._apply_costs(route, backend, headers, model, usage, gpu_input_tokens)
+        self._finish_metrics(
+            endpoint, route, backend, model, rtl.response_model, usage, start,
+            status=upstream.status,
+        )
+        hdrs = translator.response_headers(upstream.status, dict(upstream.headers))
+        content_type = hdrs.get("content-type") or upstream.headers.get(
+            "content-type", "application/json"
+        )
+        resp = web.Response(body=rtl.body, status=upstream.status)
+        resp.headers["content-type"] = content_type
+        resp.headers["x-request-id"] = headers.get("x-request-id", str(uuid.uuid4()))
+        if cache_key_vec is not None and upstream.status == 200:
+            await self.gpu.cache_insert(cache_key_vec, rtl.body)
+        return resp
+
+    async def _stream_response(
+        self, request, endpoint, route, backend, translator, upstream,
+        headers, model, start, gpu_input_tokens,
+    ) -> web.StreamResponse:
+        hdrs = translator.response_headers(upstream.status, dict(upstream.headers))
+        content_type = hdrs.get("content-type") or upstream.headers.get(
+            "content-type", "text/event-stream"
+        )
+        resp = web.StreamResponse(status=upstream.status)
+        resp.content_type = content_type
+        resp.headers["cache-control"] = "no-cache"
+        # content-length is dropped for streamed bodies (A.8)
+        await resp.prepare(request)
+
+        usage = Usage()
+        response_model = ""
+        ttft = -1.0
+        try:
+            async for chunk in upstream.content.iter_any():
+                if not chunk:
+                    continue
+                rtl = translator.response_chunk(chunk)
+                if rtl.body:
+                    if ttft < 0:
+                        ttft = time.monotonic() - start
+                    await resp.write(rtl.body)
+                if rtl.usage is not None:
+                    usage.merge_max(rtl.usage)
+                if rtl.response_model:
+                    response_model = rtl.response_model
+            tail = translator.response_flush()
+            if tail.body:
+                await resp.write(tail.body)
+            if tail.usage is not None:
+                usage.merge_max(tail.usage)
+        except (ConnectionResetError, asyncio.CancelledError):
+            logger.info("client disconnected mid-stream")
+            raise
+        finally:
+            self._apply_costs(route, backend, headers, model, usage, gpu_input_tokens)
+            self._finish_metrics(
+                endpoint, route, backend, model, response_model, usage, start,
+                status=upstream.status, ttft=ttft if ttft >= 0 else 0.0,
+            )
+        await resp.write_eof()
+        return resp
+
+    # ---- multipart (audio) ---------------------------------------------------
+
+    async def _process_multipart(self, request: web.Request, endpoint: str) -> web.StreamResponse:
+        from aigw.translator.misc_endpoints import _parse_multipart
+
+        start = time.monotonic()
+        rt = self.runtime
+        raw = await request.read()
+        content_type = request.headers.get("content-type", "")
+        try:
+            parts, _ = _parse_multipart(raw, content_type)
+        except ValueError as e:
+            return _json_error(400, str(e))
+        model = ""
+        for name, _h, payload in parts:
+            if name == "model":
+                model = payload.decode("utf-8", "replace")
+        headers = self._ingress_headers(request)
+        headers[rt.model_header] = model
+        route = rt.select_route(headers)
+        if route is None:
+            return _json_error(404, f"no route matched model {model!r}", "model_not_found")
+        decision = self.limiter.check(headers)
+        if not decision.allowed:
+            self.metrics.ratelimit_denials.labels(rule=decision.rule).inc()
+            return _json_error(429, "token budget exhausted", "rate_limit_exceeded")
+
+        assert self._session is not None
+        attempts_left = max_attempts(route)
+        last_error = None
+        for backend in backend_attempts(route):
+            if attempts_left <= 0:
+                break
+            attempts_left -= 1
+            translator = get_translator(
+                endpoint, backend.schema.name, **self._translator_kwargs(backend)
+            )
+            override = backend.model_name_override or route.route.model_name_override
+            tr, model = translator.request_multipart(raw, content_type, model_override=override)
+            up_headers = dict(tr.headers)
+            up_headers = apply_header_mutation(up_headers, route.route.header_mutation)
+            up_headers = apply_header_mutation(up_headers, backend.header_mutation)
+            auth = build_auth_handler(backend)
+            if auth is not None:
+                up_headers = auth(up_headers, tr.body, "POST", tr.path)
+            url = backend.upstream.base_url + tr.path
+            try:
+                timeout = aiohttp.ClientTimeout(total=backend.timeout_s)
+                async with self._session.post(
+                    url, data=tr.body, headers=up_headers, timeout=timeout
+                ) as upstream:
+                    if upstream.status >= 400:
+                        err = await upstream.read()
+                        if upstream.status in RETRIABLE_STATUSES and attempts_left > 0:
+                            continue
+                        return web.Response(
+                            body=translator.response_error(upstream.status, err, {}),
+                            status=upstream.status,
+                            content_type="application/json",
+                        )
+                    return await self._unary_response(
+                        endpoint, route, backend, translator, upstream, headers,
+                        model, start, 0, None, None,
+                    )
+            except (aiohttp.ClientError, asyncio.TimeoutError) as e:
+                last_error = str(e)
+                continue
+        return _json_error(503, last_error or "no healthy upstream", "upstream_error")
+
+
+async def run_server(
+    server: GatewayServer, host: str = "0.0.0.0", port: int = internalapi.DEFAULT_LISTEN_PORT
+) -> web.AppRunner:
+    app = server.make_app()
+    runner = web.AppRunner(app, access_log=None)
+    await runner.setup()
+    site = web.TCPSite(runner, host, port, backlog=4096, reuse_address=True)
+    await site.start()
+    logger.info("aigw listening on http://%s:%d", host, port)
+    return runner

@@ -33,7 +33,7 @@ from aigw.translator.sse import SSEDecoder, SSEEvent
 class _OpenAIPassthrough(Translator):
     PATH = "/v1/chat/completions"
 
-    def __init__(self, path_prefix: str = "", api_version: str = ""):
+    def __init__(self, path_prefix: str = "", api_version: str = "", **_ignored):
         self.path_prefix = path_prefix
         self.api_version = api_version
         self.stream = False
