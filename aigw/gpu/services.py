@@ -1,0 +1,185 @@
+"""Async facade over the GPU ops for the gateway hot path.
+
+Requests never launch kernels individually: a micro-batching collector
+coalesces concurrent requests (up to ``max_batch`` or ``window_ms``) into
+one packed tokenizer launch — the LDS-staged "batched 4k-token chat
+payloads" admission model from BASELINE.json — and, for cache-enabled
+requests, one embedding GEMM + one fused index lookup. GPU work runs on a
+dedicated single-thread executor so kernel syncs never stall the event
+loop.
+"""
+
+from __future__ import annotations
+
+import asyncio
+from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+
+def extract_chat_text(body: dict) -> bytes:
+    """Concatenated message text of a chat request (OpenAI or Anthropic
+    shape) — the tokenizer input for admission accounting."""
+    parts: list[str] = []
+    system = body.get("system")
+    if isinstance(system, str):
+        parts.append(system)
+    for msg in body.get("messages") or []:
+        c = msg.get("content")
+        if isinstance(c, str):
+            parts.append(c)
+        elif isinstance(c, list):
+            for p in c:
+                if isinstance(p, dict):
+                    t = p.get("text") or ""
+                    if t:
+                        parts.append(t)
+    return "\n".join(parts).encode("utf-8", "replace")[:262144] or b" "
+
+
+@dataclass
+class _Pending:
+    text: bytes
+    want_vec: bool
+    future: asyncio.Future = None  # type: ignore[assignment]
+
+
+class GPUServices:
+    def __init__(
+        self,
+        device: str = "cuda",
+        *,
+        n_merges: int = 32768,
+        enable_cache: bool = False,
+        cache_capacity: int = 65536,
+        cache_threshold: float = 0.92,
+        window_ms: float = 1.0,
+        max_batch: int = 128,
+    ):
+        from aigw.ops.semcache import SemanticCache
+        from aigw.ops.tokenizer import GPUTokenizer
+
+        self.device = device
+        self.tokenizer = GPUTokenizer(n_merges=n_merges, device=device)
+        self.cache: Optional[SemanticCache] = (
+            SemanticCache(
+                self.tokenizer.vocab_size,
+                capacity=cache_capacity,
+                threshold=cache_threshold,
+                device=device,
+            )
+            if enable_cache
+            else None
+        )
+        self.window_ms = window_ms
+        self.max_batch = max_batch
+        self._pending: list[_Pending] = []
+        self._flush_handle = None
+        self._executor = ThreadPoolExecutor(max_workers=1, thread_name_prefix="aigw-gpu")
+        self._lock = asyncio.Lock()
+
+    @property
+    def cache_enabled(self) -> bool:
+        return self.cache is not None
+
+    # ---- batching core -------------------------------------------------------
+
+    async def _submit(self, text: bytes, want_vec: bool):
+        loop = asyncio.get_running_loop()
+        item = _Pending(text=text, want_vec=want_vec, future=loop.create_future())
+        self._pending.append(item)
+        if len(self._pending) >= self.max_batch:
+            if self._flush_handle:
+                self._flush_handle.cancel()
+                self._flush_handle = None
+            asyncio.ensure_future(self._flush())
+        elif self._flush_handle is None:
+            self._flush_handle = loop.call_later(
+                self.window_ms / 1000.0, lambda: asyncio.ensure_future(self._flush())
+            )
+        return await item.future
+
+    async def _flush(self):
+        self._flush_handle = None
+        batch, self._pending = self._pending, []
+        if not batch:
+            return
+        loop = asyncio.get_running_loop()
+        try:
+            results = await loop.run_in_executor(self._executor, self._run_batch, batch)
+        except Exception as e:  # pragma: no cover - defensive
+            for it in batch:
+                if not it.future.done():
+                    it.future.set_exception(e)
+            return
+        for it, res in zip(batch, results):
+            if not it.future.done():
+                it.future.set_result(res)
+
+    def _run_batch(self, batch: list[_Pending]):
+        """Executed on the GPU worker thread: one packed tokenizer launch;
+        embedding GEMM only for the items that asked for a vector."""
+        texts = [it.text for it in batch]
+        counts, _, state = self.tokenizer.encode_batch(texts)
+        counts_host = counts.cpu().tolist()
+        vec_idx = [i for i, it in enumerate(batch) if it.want_vec]
+        vecs = {}
+        if vec_idx and self.cache is not None:
+            qvecs = self.cache.embed(state["out_ids"], state["req_off"])
+            for i in vec_idx:
+                vecs[i] = qvecs[i]
+        return [
+            (counts_host[i], vecs.get(i)) for i in range(len(batch))
+        ]
+
+    # ---- public API ----------------------------------------------------------
+
+    async def count_request_tokens(self, body: dict) -> int:
+        count, _ = await self._submit(extract_chat_text(body), want_vec=False)
+        return count
+
+    async def tokenize(self, text) -> list[int]:
+        if isinstance(text, str):
+            text = text.encode("utf-8", "replace")
+        loop = asyncio.get_running_loop()
+
+        def run():
+            _, ids, _ = self.tokenizer.encode_batch([text or b" "], return_ids=True)
+            return ids[0]
+
+        return await loop.run_in_executor(self._executor, run)
+
+    async def cache_lookup(self, body: dict):
+        """Returns (cached_response_bytes | None, query_vec)."""
+        _, vec = await self._submit(extract_chat_text(body), want_vec=True)
+        if vec is None:
+            return None, None
+        loop = asyncio.get_running_loop()
+
+        def run():
+            hits = self.cache.lookup(vec.unsqueeze(0))
+            if hits[0] is None:
+                return None
+            return self.cache.get(hits[0][0])
+
+        return await loop.run_in_executor(self._executor, run), vec
+
+    async def cache_insert(self, vec, response: bytes) -> None:
+        if vec is None or self.cache is None:
+            return
+        loop = asyncio.get_running_loop()
+        await loop.run_in_executor(self._executor, self.cache.insert, vec, response)
+
+    async def assign_replicas(self, stats: torch.Tensor, predicted: torch.Tensor):
+        loop = asyncio.get_running_loop()
+        hip = self.tokenizer.hip
+
+        def run():
+            return hip.kv_score_assign(stats, predicted, 1.0, 0.1, 0.05).cpu().tolist()
+
+        return await loop.run_in_executor(self._executor, run)
+
+    def close(self):
+        self._executor.shutdown(wait=False)

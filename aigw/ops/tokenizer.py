@@ -1,0 +1,73 @@
+"""GPU BPE tokenizer: batch packing + kernel dispatch.
+
+Packs a batch of texts into one byte buffer + request offsets, runs the
+segmenter + wave-per-segment BPE kernels (csrc/aigw_kernels.hip), and
+returns per-request token counts (and optionally ids). The merge table is
+uploaded once and stays HBM/L2-resident.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+import torch
+
+from aigw.ops import load_hip_module
+from aigw.ops.bpe_ref import BPERef, build_hash_table, make_merges
+
+
+class GPUTokenizer:
+    def __init__(
+        self,
+        n_merges: int = 32768,
+        seed: int = 1355,
+        device: str = "cuda",
+        _hip=None,
+    ):
+        self.merges = make_merges(n_merges, seed)
+        self.device = torch.device(device)
+        self.hip = _hip if _hip is not None else load_hip_module()
+        if self.hip is None:
+            raise RuntimeError("aigw_hip extension unavailable")
+        keys, ranks = build_hash_table(self.merges)
+        self.htab_keys = torch.from_numpy(keys).to(self.device)
+        self.htab_ranks = torch.from_numpy(ranks).to(self.device)
+        self.vocab_size = 256 + len(self.merges)
+
+    @staticmethod
+    def pack(texts: list[bytes]) -> tuple[np.ndarray, np.ndarray]:
+        data = b"".join(texts)
+        offs = np.zeros(len(texts), dtype=np.int64)
+        off = 0
+        for i, t in enumerate(texts):
+            offs[i] = off
+            off += len(t)
+        return np.frombuffer(data, dtype=np.uint8).copy(), offs
+
+    def encode_batch(
+        self, texts: list[bytes], return_ids: bool = False
+    ) -> tuple[torch.Tensor, Optional[list[list[int]]], dict]:
+        """Returns (req_counts int32[n_req] on GPU, ids or None, gpu_state).
+
+        gpu_state carries the on-device tensors (out_ids, req_off) so the
+        embedding stage can consume them without a host round-trip.
+        """
+        arr, offs = self.pack(texts)
+        bytes_t = torch.from_numpy(arr).to(self.device, non_blocking=True)
+        off_t = torch.from_numpy(offs).to(self.device, non_blocking=True)
+        out_ids, req_counts, seg_start, seg_req = self.hip.bpe_encode(
+            bytes_t, off_t, self.htab_keys, self.htab_ranks
+        )
+        ids = None
+        if return_ids:
+            ids = []
+            flat = out_ids.cpu().numpy()
+            bounds = list(offs) + [len(arr)]
+            for i in range(len(texts)):
+                seg = flat[bounds[i] : bounds[i + 1]]
+                ids.append([int(x) for x in seg[seg >= 0]])
+        return req_counts, ids, {"out_ids": out_ids, "req_off": off_t, "n_bytes": len(arr)}
+
+    def reference(self) -> BPERef:
+        return BPERef(self.merges)

@@ -1,0 +1,228 @@
+"""Flagship serving benchmark: OpenAI /v1/chat/completions with 4k-token
+bodies through the gateway shard to a mock upstream (BASELINE.json config #1
++ the GPU token-accounting path of config #2 when a GPU is present).
+
+One rank = one gateway shard pinned to one GPU (plus its own in-process
+mock upstream, so upstream capacity scales with shards and the gateway is
+the measured system). A "step" fires ``--batch`` concurrent requests and
+awaits them all. Timing: W untimed warmup steps, then exactly K steps
+bracketed by a distributed barrier + torch.cuda.synchronize on both sides;
+the reported value is the whole-job aggregate request rate over all ranks
+computed from the MAX per-rank elapsed time.
+
+Run directly (defaults N=1) or via torchrun for N>1 (the driver does this).
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import time
+
+import aiohttp
+import torch
+
+from aigw.extproc.server import GatewayServer, run_server
+from aigw.filterapi import RuntimeConfig, load_config
+from aigw.testing.mockupstream import start_mock_upstream
+
+WORDS = (
+    "the quick brown fox jumps over the lazy dog while seventeen engineers "
+    "profile matrix kernels on a liquid cooled accelerator node during the "
+    "long afternoon of benchmark season"
+).split()
+
+
+def build_payload(n_tokens: int) -> dict:
+    """~n_tokens-token chat body (words ≈ tokens for the synthetic BPE)."""
+    words = [WORDS[i % len(WORDS)] for i in range(n_tokens)]
+    text = " ".join(words)
+    third = len(text) // 3
+    return {
+        "model": "bench-llm",
+        "messages": [
+            {"role": "system", "content": "You are a terse assistant. " + text[:third]},
+            {"role": "user", "content": text[third : 2 * third]},
+            {"role": "assistant", "content": text[2 * third :]},
+            {"role": "user", "content": "Summarize in one word."},
+        ],
+        "max_tokens": 16,
+    }
+
+
+def gateway_config(upstream_port: int) -> dict:
+    return {
+        "version": "v1",
+        "uuid": "bench",
+        "llmRequestCosts": [{"metadataKey": "llm_total_token", "type": "TotalToken"}],
+        "routes": [
+            {
+                "name": "bench",
+                "headers": [{"name": "x-ai-eg-model", "value": "bench-llm"}],
+                "backends": [
+                    {
+                        "name": "mock-openai",
+                        "schema": "OpenAI",
+                        "upstream": {"host": "127.0.0.1", "port": upstream_port},
+                        "auth": {"apiKey": "sk-bench"},
+                    }
+                ],
+            }
+        ],
+        "rateLimits": [
+            {
+                "name": "node-token-budget",
+                "metadataKey": "llm_total_token",
+                "limit": 10**12,
+                "windowS": 3600,
+            }
+        ],
+    }
+
+
+async def fire_step(session, url, payload_bytes, batch, latencies):
+    async def one():
+        t0 = time.perf_counter()
+        async with session.post(
+            url, data=payload_bytes, headers={"content-type": "application/json"}
+        ) as r:
+            await r.read()
+            assert r.status == 200, f"status {r.status}"
+        latencies.append((time.perf_counter() - t0) * 1000.0)
+
+    await asyncio.gather(*(one() for _ in range(batch)))
+
+
+async def amain(args, rank, world, local_rank):
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+
+    mock, up_runner, up_port = await start_mock_upstream(port=0)
+    cfg = load_config(gateway_config(up_port))
+
+    gpu_services = None
+    if use_gpu:
+        from aigw.gpu import GPUServices
+
+        gpu_services = GPUServices(
+            device=f"cuda:{local_rank}", n_merges=32768, enable_cache=False,
+            window_ms=0.5, max_batch=256,
+        )
+
+    server = GatewayServer(RuntimeConfig(cfg), gpu_services=gpu_services)
+    gw_runner = await run_server(server, host="127.0.0.1", port=0)
+    gw_port = gw_runner.addresses[0][1]
+
+    payload = json.dumps(build_payload(args.tokens)).encode()
+    connector = aiohttp.TCPConnector(limit=0)
+    session = aiohttp.ClientSession(connector=connector)
+    gw_url = f"http://127.0.0.1:{gw_port}/v1/chat/completions"
+    direct_url = f"http://127.0.0.1:{up_port}/v1/chat/completions"
+
+    # untimed: direct-to-upstream baseline for "added latency"
+    direct_lat: list[float] = []
+    for _ in range(2):
+        await fire_step(session, direct_url, payload, args.batch, direct_lat)
+
+    # warmup
+    warm_lat: list[float] = []
+    for _ in range(args.warmup):
+        await fire_step(session, gw_url, payload, args.batch, warm_lat)
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    barrier_sync()
+    lat: list[float] = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        await fire_step(session, gw_url, payload, args.batch, lat)
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    barrier_sync()
+
+    # MAX elapsed over ranks defines whole-job time
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if torch.distributed.get_backend() == "nccl":
+            t = t.cuda()
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed_max = float(t.cpu().item())
+    else:
+        elapsed_max = elapsed
+
+    total_requests = args.steps * args.batch * world
+    value = total_requests / elapsed_max
+    p50 = statistics.median(lat)
+    p99 = sorted(lat)[int(len(lat) * 0.99) - 1]
+    p50_direct = statistics.median(direct_lat)
+
+    if rank == 0:
+        out = {
+            "metric": "req/sec (whole node) + p50 added latency, OpenAI chat/completions 4k-token body",
+            "value": round(value, 2),
+            "unit": "req/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed_max / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "aigw standalone: OpenAI /v1/chat/completions -> mock upstream (examples/basic)",
+                "global_batch": args.batch * world,
+                "seq_len": args.tokens,
+                "parallelism": f"dp{world}",
+                "p50_ms": round(p50, 3),
+                "p99_ms": round(p99, 3),
+                "p50_direct_ms": round(p50_direct, 3),
+                "p50_added_latency_ms": round(p50 - p50_direct, 3),
+                "gpu_token_accounting": bool(use_gpu),
+            },
+        }
+        print(json.dumps(out))
+
+    await session.close()
+    await gw_runner.cleanup()
+    await up_runner.cleanup()
+    if gpu_services is not None:
+        gpu_services.close()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--batch", type=int, default=256)
+    ap.add_argument("--tokens", type=int, default=4096)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if world > 1:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
+        torch.distributed.init_process_group(backend=backend)
+    try:
+        asyncio.run(amain(args, rank, world, local_rank))
+    finally:
+        if world > 1:
+            torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

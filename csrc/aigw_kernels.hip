@@ -1,0 +1,598 @@
+// aigw MI355X (gfx950 / CDNA4) kernels.
+//
+// GPU tier of the gateway (BASELINE.json north star; no counterpart code in
+// the reference, which does this work CPU-side or not at all —
+// SURVEY.md §2.4):
+//
+//   1. Byte-level BPE tokenizer (segmenter + wave-per-segment merge loop)
+//      — replaces the reference's provider-usage-JSON token accounting
+//      (translator/openai_openai.go:185-223) and /tokenize passthrough
+//      (translator/tokenize.go:24-81) with gateway-local counting.
+//   2. Mean-pool embedding + MFMA bf16 projection GEMM — semantic response
+//      cache (the reference's cache is provider-side passthrough).
+//   3. Fused MFMA similarity + argmax over the HBM-resident cache index.
+//   4. KV-occupancy endpoint scorer — replaces the external EPP service
+//      (extensionserver/inferencepool.go:39-54).
+//
+// CDNA4 specifics used (per the MI355X HIP guide): 64-wide wavefronts
+// (64-bit ballot masks), __builtin_amdgcn_mfma_f32_16x16x32_bf16 with the
+// C/D mapping col=lane&15 / row=(lane>>4)*4+reg, LDS staging, short8
+// vectorized bf16 loads, grid-stride loops sized for 256 CUs.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <cstdint>
+
+#define AIGW_CHECK(cond, msg) TORCH_CHECK(cond, msg)
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float floatx4;
+typedef __attribute__((ext_vector_type(4))) short short4v;
+
+static inline hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+// ---------------------------------------------------------------------------
+// 1. BPE tokenizer
+// ---------------------------------------------------------------------------
+
+// Byte classes for the segmentation rule (must match aigw/ops/bpe_ref.py):
+// segment starts at i iff i==0, byte==' ', or class(b[i])!=class(b[i-1]) and
+// b[i-1]!=' '.
+__device__ __forceinline__ int byte_class(uint8_t b) {
+  if (b == ' ') return 0;
+  if (b == '\t' || b == '\n' || b == '\r' || b == 0x0b || b == 0x0c) return 1;
+  if (b >= '0' && b <= '9') return 2;
+  if ((b >= 'A' && b <= 'Z') || (b >= 'a' && b <= 'z') || b >= 0x80) return 3;
+  return 4;  // punctuation / other
+}
+
+__global__ void seg_flags_kernel(const uint8_t* __restrict__ bytes, int n,
+                                 uint8_t* __restrict__ flags) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint8_t f;
+  if (i == 0) {
+    f = 1;
+  } else {
+    uint8_t b = bytes[i], p = bytes[i - 1];
+    f = (b == ' ') || (byte_class(b) != byte_class(p) && p != ' ');
+  }
+  flags[i] = f;
+}
+
+__global__ void seg_force_starts_kernel(const int64_t* __restrict__ req_off,
+                                        int n_req, uint8_t* __restrict__ flags) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n_req) flags[req_off[i]] = 1;
+}
+
+// per-256-byte-block flag counts (for the ordered compaction scan)
+__global__ void seg_block_count_kernel(const uint8_t* __restrict__ flags, int n,
+                                       int32_t* __restrict__ blk_counts) {
+  __shared__ int cnt;
+  if (threadIdx.x == 0) cnt = 0;
+  __syncthreads();
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t ballot = __ballot(i < n && flags[i]);
+  if ((threadIdx.x & 63) == 0) atomicAdd(&cnt, __popcll(ballot));
+  __syncthreads();
+  if (threadIdx.x == 0) blk_counts[blockIdx.x] = cnt;
+}
+
+// write segment start offsets + owning request id (ordered compaction)
+__global__ void seg_write_kernel(const uint8_t* __restrict__ flags, int n,
+                                 const int32_t* __restrict__ blk_excl,  // exclusive scan
+                                 const int64_t* __restrict__ req_off, int n_req,
+                                 int32_t* __restrict__ seg_start,
+                                 int32_t* __restrict__ seg_req) {
+  __shared__ int wave_base[4];
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  bool f = (i < n) && flags[i];
+  uint64_t ballot = __ballot(f);
+  // block-local exclusive offsets: wave sums staged through LDS
+  if (lane == 0) wave_base[wave] = __popcll(ballot);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int acc = 0;
+    for (int w = 0; w < 4; ++w) {
+      int c = wave_base[w];
+      wave_base[w] = acc;
+      acc += c;
+    }
+  }
+  __syncthreads();
+  if (!f) return;
+  int pos_in_wave = __popcll(ballot & ((1ull << lane) - 1ull));
+  int pos = blk_excl[blockIdx.x] + wave_base[wave] + pos_in_wave;
+  seg_start[pos] = i;
+  // owning request: upper_bound(req_off, i) - 1
+  int lo = 0, hi = n_req;
+  while (lo < hi) {
+    int mid = (lo + hi) >> 1;
+    if (req_off[mid] <= i) lo = mid + 1; else hi = mid;
+  }
+  seg_req[pos] = lo - 1;
+}
+
+// The BPE merge loop: one 64-lane wave per segment. Tokens live in lane
+// registers; the merge-pair table is an open-addressing hash in HBM (hot
+// entries L2-resident). Pair ranks are wave-min-reduced with __shfl_xor;
+// occurrence selection and merging operate on wave-uniform 64-bit masks.
+__global__ void __launch_bounds__(256)
+bpe_encode_kernel(const uint8_t* __restrict__ bytes,
+                  const int32_t* __restrict__ seg_start,
+                  const int32_t* __restrict__ seg_req, int n_segs, int n_bytes,
+                  const long long* __restrict__ htab_keys,
+                  const int32_t* __restrict__ htab_rank, int htab_mask,
+                  int32_t* __restrict__ out_ids,  // n_bytes, -1 = gap
+                  int32_t* __restrict__ req_counts) {
+  int seg = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  if (seg >= n_segs) return;
+  int lane = threadIdx.x & 63;
+  int s = seg_start[seg];
+  int e = (seg + 1 < n_segs) ? seg_start[seg + 1] : n_bytes;
+  int written = 0;
+
+  for (int chunk = s; chunk < e; chunk += 64) {
+    int len = min(e - chunk, 64);
+    int tok = (lane < len) ? (int)bytes[chunk + lane] : -1;
+    uint64_t active = __ballot(tok >= 0);
+
+    for (;;) {
+      // next active lane above mine
+      uint64_t above = (lane < 63) ? (active & (~0ull << (lane + 1))) : 0ull;
+      int nxt = above ? (__ffsll((long long)above) - 1) : -1;
+      int nxttok = __shfl(tok, nxt < 0 ? 0 : nxt);
+      int rank = INT_MAX;
+      if (tok >= 0 && nxt >= 0) {
+        long long key = ((long long)tok << 32) | (unsigned)nxttok;
+        uint64_t h = (uint64_t)key * 0x9E3779B97F4A7C15ull;
+        int idx = (int)(h >> 40) & htab_mask;
+        for (;;) {
+          long long k = htab_keys[idx];
+          if (k == key) { rank = htab_rank[idx]; break; }
+          if (k == -1) break;
+          idx = (idx + 1) & htab_mask;
+        }
+      }
+      // wave min-reduce of rank
+      int minrank = rank;
+      #pragma unroll
+      for (int off = 32; off > 0; off >>= 1)
+        minrank = min(minrank, __shfl_xor(minrank, off));
+      if (minrank == INT_MAX) break;
+
+      uint64_t occ = __ballot(rank == minrank);
+      // greedy leftmost non-overlapping selection (wave-uniform scalar loop)
+      uint64_t m = occ, sel = 0;
+      while (m) {
+        int i = __ffsll((long long)m) - 1;
+        sel |= 1ull << i;
+        int ni = __shfl(nxt, i);
+        m &= ~(1ull << i);
+        if (ni >= 0) m &= ~(1ull << ni);
+      }
+      int newid = 256 + minrank;
+      // previous active lane (my potential merge head)
+      uint64_t below = active & ((lane ? (1ull << lane) : 1ull) - 1ull);
+      int prev = below ? (63 - __clzll((long long)below)) : -1;
+      bool merged_into_prev = (prev >= 0) && ((sel >> prev) & 1);
+      if ((sel >> lane) & 1) tok = newid;
+      if (merged_into_prev) tok = -1;
+      active = __ballot(tok >= 0);
+    }
+
+    // compact write within the segment's byte span
+    int cnt = __popcll(active);
+    if (tok >= 0) {
+      int pos = __popcll(active & ((lane ? (1ull << lane) : 1ull) - 1ull));
+      out_ids[s + written + pos] = tok;
+    }
+    written += cnt;
+  }
+  // gaps stay -1 (buffer pre-filled); per-request token count
+  if (lane == 0 && written > 0) atomicAdd(&req_counts[seg_req[seg]], written);
+}
+
+// ---------------------------------------------------------------------------
+// 2. Embedding: mean-pool token embeddings per request (memory-bound)
+// ---------------------------------------------------------------------------
+
+// one block (384 threads) per request; thread t owns output column t
+__global__ void meanpool_kernel(const int32_t* __restrict__ ids,  // n_bytes, -1 gaps
+                                const int64_t* __restrict__ req_off, int n_req,
+                                int n_bytes, const bf16* __restrict__ emb,
+                                int vocab, int dim,
+                                float* __restrict__ out /* n_req x dim */) {
+  int r = blockIdx.x;
+  if (r >= n_req) return;
+  int col = threadIdx.x;
+  if (col >= dim) return;
+  long long s = req_off[r];
+  long long e = (r + 1 < n_req) ? req_off[r + 1] : n_bytes;
+  float acc = 0.f;
+  int cnt = 0;
+  for (long long i = s; i < e; ++i) {
+    int tok = ids[i];
+    if (tok < 0) continue;
+    acc += __bfloat162float(emb[(long long)tok * dim + col]);
+    ++cnt;
+  }
+  out[(long long)r * dim + col] = cnt ? acc / (float)cnt : 0.f;
+}
+
+// ---------------------------------------------------------------------------
+// MFMA bf16 GEMM (NT): C[M,N] = A[M,K] * B[N,K]^T, fp32 out.
+// 16x16x32 MFMA; per-wave 16x16 C tile; block = 4 waves covering 16x64.
+// Operand layout (verified on HW by tests/test_gpu_kernels.py::test_mfma_probe
+// against torch.matmul with random asymmetric inputs):
+//   A: lane l holds A[l&15][(l>>4)*8 + j], j=0..7  (8 contiguous bf16)
+//   B: lane l holds B^T[l&15][(l>>4)*8 + j]        (8 contiguous bf16 of B^T)
+//   C: lane l, reg r -> C[(l>>4)*4 + r][l&15]
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(256)
+gemm_bf16_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+                    float* __restrict__ C, int M, int N, int K,
+                    const float* __restrict__ bias, int relu) {
+  int m0 = blockIdx.x * 16;
+  int n0 = blockIdx.y * 64 + (threadIdx.x >> 6) * 16;
+  int lane = threadIdx.x & 63;
+  int row = lane & 15;     // A row within tile / C col group
+  int kgrp = lane >> 4;    // 0..3
+  floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+  bool a_ok = (m0 + row) < M;
+  bool b_ok = (n0 + row) < N;
+  for (int k = 0; k < K; k += 32) {
+    short8 a = {0, 0, 0, 0, 0, 0, 0, 0}, b = {0, 0, 0, 0, 0, 0, 0, 0};
+    int kk = k + kgrp * 8;
+    if (a_ok && kk < K)
+      a = *reinterpret_cast<const short8*>(&A[(long long)(m0 + row) * K + kk]);
+    if (b_ok && kk < K)
+      b = *reinterpret_cast<const short8*>(&Bt[(long long)(n0 + row) * K + kk]);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  int c_row = m0 + kgrp * 4;
+  int c_col = n0 + row;
+  if (c_col >= N) return;
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int cr = c_row + r;
+    if (cr < M) {
+      float v = acc[r];
+      if (bias) v += bias[c_col];
+      if (relu && v < 0.f) v = 0.f;
+      C[(long long)cr * N + c_col] = v;
+    }
+  }
+}
+
+// L2-normalize rows, fp32 -> bf16 (one wave per row)
+__global__ void l2norm_rows_kernel(const float* __restrict__ in, bf16* __restrict__ out,
+                                   int rows, int dim) {
+  int r = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  if (r >= rows) return;
+  int lane = threadIdx.x & 63;
+  float ss = 0.f;
+  for (int c = lane; c < dim; c += 64) {
+    float v = in[(long long)r * dim + c];
+    ss += v * v;
+  }
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1) ss += __shfl_xor(ss, off);
+  float inv = rsqrtf(ss + 1e-12f);
+  for (int c = lane; c < dim; c += 64) {
+    out[(long long)r * dim + c] = __float2bfloat16(in[(long long)r * dim + c] * inv);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 3. Fused cache similarity + argmax: scores[i,q] = Index[i,:].Q[q,:]
+//    (rows pre-normalized -> cosine). Each wave: MFMA over a 16-row index
+//    tile x 16 queries, K-loop over dim; per-query max folded via shfl and
+//    packed (orderable-float<<32 | row) into a global 64-bit atomicMax.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ unsigned long long shfl_xor_u64(unsigned long long v,
+                                                           int off) {
+  int lo = __shfl_xor((int)(v & 0xFFFFFFFFull), off);
+  int hi = __shfl_xor((int)(v >> 32), off);
+  return ((unsigned long long)(unsigned)hi << 32) | (unsigned)lo;
+}
+
+__device__ __forceinline__ unsigned long long pack_score(float s, unsigned idx) {
+  unsigned u = __float_as_uint(s);
+  u = (u & 0x80000000u) ? ~u : (u | 0x80000000u);  // orderable float
+  return ((unsigned long long)u << 32) | idx;
+}
+
+__global__ void __launch_bounds__(256)
+cache_topk_kernel(const bf16* __restrict__ index, long long n_rows,
+                  const bf16* __restrict__ q, int n_q, int dim,
+                  unsigned long long* __restrict__ best /* n_q */) {
+  long long i0 = ((long long)blockIdx.x * 4 + (threadIdx.x >> 6)) * 16;
+  if (i0 >= n_rows) return;
+  int lane = threadIdx.x & 63;
+  int row = lane & 15;
+  int kgrp = lane >> 4;
+  bool i_ok = (i0 + row) < n_rows;
+  for (int q0 = 0; q0 < n_q; q0 += 16) {
+    floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+    bool q_ok = (q0 + row) < n_q;
+    for (int k = 0; k < dim; k += 32) {
+      short8 a = {0, 0, 0, 0, 0, 0, 0, 0}, b = {0, 0, 0, 0, 0, 0, 0, 0};
+      int kk = k + kgrp * 8;
+      if (i_ok && kk < dim)
+        a = *reinterpret_cast<const short8*>(&index[(i0 + row) * dim + kk]);
+      if (q_ok && kk < dim)
+        b = *reinterpret_cast<const short8*>(&q[(long long)(q0 + row) * dim + kk]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    // acc reg r holds C[idx_row = kgrp*4+r][query = row]: fold the 4 regs
+    // locally, then across the 4 lane groups sharing this query column
+    // (lanes differing in bits 4..5), then one atomicMax per query.
+    bool q_in = (q0 + row) < n_q;
+    unsigned long long p = 0;
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      long long irow = i0 + kgrp * 4 + r;
+      float s = (irow < n_rows && q_in) ? acc[r] : -1e30f;
+      unsigned long long pk = pack_score(s, (unsigned)(irow & 0xFFFFFFFF));
+      if (pk > p) p = pk;
+    }
+    #pragma unroll
+    for (int off = 16; off < 64; off <<= 1) {
+      unsigned long long o = shfl_xor_u64(p, off);
+      if (o > p) p = o;
+    }
+    if (kgrp == 0 && q_in) atomicMax(&best[q0 + row], p);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 4. KV-occupancy endpoint scorer: greedy sequential assignment of a request
+//    batch to replicas. One wave; lane = replica. score = w_kv*(free KV frac
+//    after assignment) - w_q*queue_depth - w_a*active. Mirrors the EPP
+//    "prefix-cache + queue depth" scoring the reference delegates to an
+//    external endpoint-picker service.
+// ---------------------------------------------------------------------------
+
+__global__ void kv_scorer_kernel(const float* __restrict__ stats,  // R x 4
+                                 int n_rep, const float* __restrict__ pred_tokens,
+                                 int n_req, float w_kv, float w_q, float w_a,
+                                 int32_t* __restrict__ assign) {
+  int lane = threadIdx.x & 63;
+  bool ok = lane < n_rep;
+  float kv_used = ok ? stats[lane * 4 + 0] : 0.f;
+  float kv_total = ok ? fmaxf(stats[lane * 4 + 1], 1.f) : 1.f;
+  float queue = ok ? stats[lane * 4 + 2] : 0.f;
+  float active = ok ? stats[lane * 4 + 3] : 0.f;
+  for (int i = 0; i < n_req; ++i) {
+    float p = pred_tokens[i];
+    float score = ok ? (w_kv * (1.f - (kv_used + p) / kv_total) - w_q * queue - w_a * active)
+                     : -1e30f;
+    if (ok && kv_used + p > kv_total) score -= 1e6f;  // avoid overflowing a replica
+    // wave argmax
+    float best = score;
+    int best_lane = lane;
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      float os = __shfl_xor(best, off);
+      int ol = __shfl_xor(best_lane, off);
+      if (os > best || (os == best && ol < best_lane)) { best = os; best_lane = ol; }
+    }
+    if (lane == 0) assign[i] = best_lane;
+    if (lane == best_lane) {
+      kv_used += p;
+      active += 1.f;
+      queue += 1.f;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+static void check_cuda(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// Returns (seg_start, seg_req, n_segs is seg_start.numel())
+std::vector<at::Tensor> bpe_segment(at::Tensor bytes, at::Tensor req_off) {
+  check_cuda(bytes, "bytes");
+  check_cuda(req_off, "req_off");
+  int n = (int)bytes.numel();
+  TORCH_CHECK(n > 0, "empty batch");
+  auto stream = current_stream();
+  auto u8 = at::TensorOptions().dtype(at::kByte).device(bytes.device());
+  auto i32 = at::TensorOptions().dtype(at::kInt).device(bytes.device());
+  at::Tensor flags = at::empty({n}, u8);
+  int blocks = (n + 255) / 256;
+  hipLaunchKernelGGL(seg_flags_kernel, dim3(blocks), dim3(256), 0, stream,
+                     bytes.data_ptr<uint8_t>(), n, flags.data_ptr<uint8_t>());
+  int n_req = (int)req_off.numel();
+  hipLaunchKernelGGL(seg_force_starts_kernel, dim3((n_req + 255) / 256), dim3(256), 0,
+                     stream, req_off.data_ptr<int64_t>(), n_req,
+                     flags.data_ptr<uint8_t>());
+  at::Tensor blk_counts = at::empty({blocks}, i32);
+  hipLaunchKernelGGL(seg_block_count_kernel, dim3(blocks), dim3(256), 0, stream,
+                     flags.data_ptr<uint8_t>(), n, blk_counts.data_ptr<int32_t>());
+  at::Tensor scan = blk_counts.cumsum(0, at::kInt);
+  at::Tensor blk_excl = at::zeros({blocks}, i32);
+  if (blocks > 1)
+    blk_excl.narrow(0, 1, blocks - 1).copy_(scan.narrow(0, 0, blocks - 1));
+  int n_segs = scan[blocks - 1].item<int>();  // one D2H sync per batch
+  at::Tensor seg_start = at::empty({n_segs}, i32);
+  at::Tensor seg_req = at::empty({n_segs}, i32);
+  hipLaunchKernelGGL(seg_write_kernel, dim3(blocks), dim3(256), 0, stream,
+                     flags.data_ptr<uint8_t>(), n, blk_excl.data_ptr<int32_t>(),
+                     req_off.data_ptr<int64_t>(), n_req,
+                     seg_start.data_ptr<int32_t>(), seg_req.data_ptr<int32_t>());
+  return {seg_start, seg_req};
+}
+
+std::vector<at::Tensor> bpe_encode(at::Tensor bytes, at::Tensor req_off,
+                                   at::Tensor htab_keys, at::Tensor htab_rank) {
+  check_cuda(bytes, "bytes");
+  check_cuda(req_off, "req_off");
+  check_cuda(htab_keys, "htab_keys");
+  check_cuda(htab_rank, "htab_rank");
+  auto segs = bpe_segment(bytes, req_off);
+  at::Tensor seg_start = segs[0], seg_req = segs[1];
+  int n_segs = (int)seg_start.numel();
+  int n = (int)bytes.numel();
+  int n_req = (int)req_off.numel();
+  auto i32 = at::TensorOptions().dtype(at::kInt).device(bytes.device());
+  at::Tensor out_ids = at::full({n}, -1, i32);
+  at::Tensor req_counts = at::zeros({n_req}, i32);
+  int htab_mask = (int)htab_keys.numel() - 1;
+  TORCH_CHECK((htab_keys.numel() & htab_mask) == 0, "htab size must be power of 2");
+  int blocks = (n_segs + 3) / 4;  // 4 waves per 256-thread block
+  auto stream = current_stream();
+  hipLaunchKernelGGL(bpe_encode_kernel, dim3(blocks), dim3(256), 0, stream,
+                     bytes.data_ptr<uint8_t>(), seg_start.data_ptr<int32_t>(),
+                     seg_req.data_ptr<int32_t>(), n_segs, n,
+                     reinterpret_cast<long long*>(htab_keys.data_ptr<int64_t>()),
+                     htab_rank.data_ptr<int32_t>(), htab_mask,
+                     out_ids.data_ptr<int32_t>(), req_counts.data_ptr<int32_t>());
+  return {out_ids, req_counts, seg_start, seg_req};
+}
+
+at::Tensor meanpool(at::Tensor ids, at::Tensor req_off, at::Tensor emb) {
+  check_cuda(ids, "ids");
+  check_cuda(req_off, "req_off");
+  check_cuda(emb, "emb");
+  TORCH_CHECK(emb.scalar_type() == at::kBFloat16, "emb must be bf16");
+  int n_req = (int)req_off.numel();
+  int dim = (int)emb.size(1);
+  TORCH_CHECK(dim <= 1024, "dim too large for one block");
+  auto out = at::empty({n_req, dim},
+                       at::TensorOptions().dtype(at::kFloat).device(ids.device()));
+  hipLaunchKernelGGL(meanpool_kernel, dim3(n_req), dim3(dim), 0, current_stream(),
+                     ids.data_ptr<int32_t>(), req_off.data_ptr<int64_t>(), n_req,
+                     (int)ids.numel(),
+                     reinterpret_cast<bf16*>(emb.data_ptr<at::BFloat16>()),
+                     (int)emb.size(0), dim, out.data_ptr<float>());
+  return out;
+}
+
+at::Tensor gemm_bf16_nt(at::Tensor a, at::Tensor bt, c10::optional<at::Tensor> bias,
+                        bool relu) {
+  check_cuda(a, "a");
+  check_cuda(bt, "bt");
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16 && bt.scalar_type() == at::kBFloat16,
+              "bf16 required");
+  int M = (int)a.size(0), K = (int)a.size(1), N = (int)bt.size(0);
+  TORCH_CHECK(bt.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+  auto c = at::empty({M, N}, at::TensorOptions().dtype(at::kFloat).device(a.device()));
+  const float* bias_ptr = nullptr;
+  if (bias.has_value()) {
+    check_cuda(*bias, "bias");
+    bias_ptr = bias->data_ptr<float>();
+  }
+  dim3 grid((M + 15) / 16, (N + 63) / 64);
+  hipLaunchKernelGGL(gemm_bf16_nt_kernel, grid, dim3(256), 0, current_stream(),
+                     reinterpret_cast<bf16*>(a.data_ptr<at::BFloat16>()),
+                     reinterpret_cast<bf16*>(bt.data_ptr<at::BFloat16>()),
+                     c.data_ptr<float>(), M, N, K, bias_ptr, relu ? 1 : 0);
+  return c;
+}
+
+at::Tensor l2norm_rows(at::Tensor x) {
+  check_cuda(x, "x");
+  TORCH_CHECK(x.scalar_type() == at::kFloat, "fp32 required");
+  int rows = (int)x.size(0), dim = (int)x.size(1);
+  auto out = at::empty({rows, dim},
+                       at::TensorOptions().dtype(at::kBFloat16).device(x.device()));
+  int blocks = (rows + 3) / 4;
+  hipLaunchKernelGGL(l2norm_rows_kernel, dim3(blocks), dim3(256), 0, current_stream(),
+                     x.data_ptr<float>(),
+                     reinterpret_cast<bf16*>(out.data_ptr<at::BFloat16>()), rows, dim);
+  return out;
+}
+
+std::vector<at::Tensor> cache_topk(at::Tensor index, at::Tensor q) {
+  check_cuda(index, "index");
+  check_cuda(q, "q");
+  TORCH_CHECK(index.scalar_type() == at::kBFloat16 && q.scalar_type() == at::kBFloat16,
+              "bf16 required");
+  long long n_rows = index.size(0);
+  int n_q = (int)q.size(0), dim = (int)q.size(1);
+  TORCH_CHECK(index.size(1) == dim, "dim mismatch");
+  auto best = at::zeros({n_q}, at::TensorOptions()
+                                   .dtype(at::kLong)
+                                   .device(q.device()));
+  long long tiles = (n_rows + 63) / 64;
+  hipLaunchKernelGGL(cache_topk_kernel, dim3((unsigned)tiles), dim3(256), 0,
+                     current_stream(),
+                     reinterpret_cast<bf16*>(index.data_ptr<at::BFloat16>()), n_rows,
+                     reinterpret_cast<bf16*>(q.data_ptr<at::BFloat16>()), n_q, dim,
+                     reinterpret_cast<unsigned long long*>(best.data_ptr<int64_t>()));
+  // unpack: score = orderable^-1(hi32), idx = lo32
+  auto hi = best.bitwise_right_shift(32).to(at::kLong);
+  auto idx = best.bitwise_and(0xFFFFFFFFLL).to(at::kInt);
+  return {hi, idx};
+}
+
+at::Tensor kv_score_assign(at::Tensor stats, at::Tensor pred_tokens, double w_kv,
+                           double w_q, double w_a) {
+  check_cuda(stats, "stats");
+  check_cuda(pred_tokens, "pred_tokens");
+  int n_rep = (int)stats.size(0);
+  TORCH_CHECK(n_rep <= 64, "at most 64 replicas");
+  int n_req = (int)pred_tokens.numel();
+  auto assign = at::empty({n_req},
+                          at::TensorOptions().dtype(at::kInt).device(stats.device()));
+  hipLaunchKernelGGL(kv_scorer_kernel, dim3(1), dim3(64), 0, current_stream(),
+                     stats.data_ptr<float>(), n_rep, pred_tokens.data_ptr<float>(),
+                     n_req, (float)w_kv, (float)w_q, (float)w_a,
+                     assign.data_ptr<int32_t>());
+  return assign;
+}
+
+// MFMA layout probe: C = A(16x32) @ B(32x16) as one intrinsic call; used by
+// the HW test to verify the documented fragment layouts against torch.
+__global__ void mfma_probe_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+                                  float* __restrict__ C) {
+  int lane = threadIdx.x & 63;
+  short8 a = *reinterpret_cast<const short8*>(&A[(lane & 15) * 32 + (lane >> 4) * 8]);
+  short8 b = *reinterpret_cast<const short8*>(&Bt[(lane & 15) * 32 + (lane >> 4) * 8]);
+  floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) C[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+}
+
+at::Tensor mfma_probe(at::Tensor a, at::Tensor bt) {
+  check_cuda(a, "a");
+  check_cuda(bt, "bt");
+  auto c = at::empty({16, 16}, at::TensorOptions().dtype(at::kFloat).device(a.device()));
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, current_stream(),
+                     reinterpret_cast<bf16*>(a.data_ptr<at::BFloat16>()),
+                     reinterpret_cast<bf16*>(bt.data_ptr<at::BFloat16>()),
+                     c.data_ptr<float>());
+  return c;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bpe_segment", &bpe_segment, "segment bytes (GPU)");
+  m.def("bpe_encode", &bpe_encode, "BPE encode a packed byte batch (GPU)");
+  m.def("meanpool", &meanpool, "mean-pool token embeddings per request");
+  m.def("gemm_bf16_nt", &gemm_bf16_nt, "C = A @ Bt^T (MFMA bf16)",
+        py::arg("a"), py::arg("bt"), py::arg("bias") = py::none(),
+        py::arg("relu") = false);
+  m.def("l2norm_rows", &l2norm_rows, "row-wise L2 normalize fp32->bf16");
+  m.def("cache_topk", &cache_topk, "fused cosine-sim argmax over cache index");
+  m.def("kv_score_assign", &kv_score_assign, "greedy KV-occupancy assignment");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+}

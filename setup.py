@@ -1,0 +1,30 @@
+"""Build the gfx950 HIP extension in-tree: `python setup.py build_ext --inplace`.
+
+The resulting aigw_hip*.so lands at the repo root so it travels with the
+source snapshot to GPU boxes (no JIT cache dependence).
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+setup(
+    name="aigw",
+    version="0.1.0",
+    packages=["aigw"],
+    ext_modules=[
+        CUDAExtension(
+            name="aigw_hip",
+            sources=["csrc/aigw_kernels.hip"],
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)

@@ -1,0 +1,192 @@
+"""GPU kernel numerics tests: every HIP kernel vs a plain PyTorch/CPU fp32
+reference (run on MI355X via gpurun; auto-skipped off-GPU)."""
+
+import json
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def hip():
+    import aigw_hip
+
+    return aigw_hip
+
+
+def test_mfma_probe_layout(hip):
+    """Asymmetric-input check of the documented 16x16x32 bf16 fragment
+    layout (guide rule G9: symmetric inputs can hide transposed layouts)."""
+    torch.manual_seed(0)
+    a = torch.randn(16, 32, device="cuda").to(torch.bfloat16)
+    b = torch.randn(32, 16, device="cuda").to(torch.bfloat16)
+    bt = b.t().contiguous()
+    c = hip.mfma_probe(a, bt)
+    ref = a.float() @ b.float()
+    torch.testing.assert_close(c, ref, rtol=2e-2, atol=2e-2)
+
+
+def test_gemm_bf16_nt(hip):
+    torch.manual_seed(1)
+    for m, n, k in [(16, 64, 32), (33, 100, 384), (256, 384, 384), (7, 16, 64)]:
+        a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
+        bt = torch.randn(n, k, device="cuda").to(torch.bfloat16)
+        c = hip.gemm_bf16_nt(a, bt, None, False)
+        ref = a.float() @ bt.float().t()
+        torch.testing.assert_close(c, ref, rtol=5e-2, atol=5e-1)
+
+
+def test_gemm_bias_relu(hip):
+    torch.manual_seed(2)
+    a = torch.randn(32, 64, device="cuda").to(torch.bfloat16)
+    bt = torch.randn(48, 64, device="cuda").to(torch.bfloat16)
+    bias = torch.randn(48, device="cuda")
+    c = hip.gemm_bf16_nt(a, bt, bias, True)
+    ref = torch.relu(a.float() @ bt.float().t() + bias)
+    torch.testing.assert_close(c, ref, rtol=5e-2, atol=5e-1)
+
+
+def test_bpe_segments_match_reference(hip):
+    from aigw.ops.bpe_ref import segment_starts
+    from aigw.ops.tokenizer import GPUTokenizer
+
+    texts = [
+        b"The quick brown fox! jumps over 42 lazy dogs.  Multiple   spaces.",
+        b"short",
+        bytes(range(256)),
+    ]
+    arr, offs = GPUTokenizer.pack(texts)
+    bytes_t = torch.from_numpy(arr).cuda()
+    off_t = torch.from_numpy(offs).cuda()
+    seg_start, seg_req = hip.bpe_segment(bytes_t, off_t)
+    ref_starts = segment_starts(bytes(arr.tobytes()), list(offs))
+    assert seg_start.cpu().tolist() == ref_starts
+    # request assignment
+    reqs = seg_req.cpu().tolist()
+    for s, r in zip(ref_starts, reqs):
+        expect = 0
+        for i, o in enumerate(offs):
+            if s >= o:
+                expect = i
+        assert r == expect
+
+
+def test_bpe_encode_matches_reference(hip):
+    from aigw.ops.tokenizer import GPUTokenizer
+
+    tok = GPUTokenizer(n_merges=8192, device="cuda")
+    ref = tok.reference()
+    rng = np.random.default_rng(3)
+    words = [b"the", b"quick", b"brown", b"fox", b"jump", b"lazy", b"dogs", b"42!", b"\xc3\xa9t\xc3\xa9"]
+    texts = []
+    for i in range(17):
+        n = int(rng.integers(1, 400))
+        texts.append(b" ".join(words[int(j)] for j in rng.integers(0, len(words), n)))
+    texts.append(b"x" * 200)  # long single segment (chunked at 64)
+    counts, ids, _ = tok.encode_batch(texts, return_ids=True)
+    expect = ref.encode_batch(texts)
+    for i, (got, want) in enumerate(zip(ids, expect)):
+        assert got == want, f"text {i}: GPU ids differ from CPU reference"
+    assert counts.cpu().tolist() == [len(w) for w in expect]
+
+
+def test_meanpool_matches_torch(hip):
+    from aigw.ops.tokenizer import GPUTokenizer
+
+    tok = GPUTokenizer(n_merges=4096, device="cuda")
+    texts = [b"hello world this is a test", b"another request body"]
+    counts, ids, state = tok.encode_batch(texts, return_ids=True)
+    emb = torch.randn(tok.vocab_size, 384, device="cuda").to(torch.bfloat16)
+    pooled = hip.meanpool(state["out_ids"], state["req_off"], emb)
+    for i, idlist in enumerate(ids):
+        ref = emb[torch.tensor(idlist, device="cuda")].float().mean(dim=0)
+        torch.testing.assert_close(pooled[i], ref, rtol=2e-2, atol=2e-2)
+
+
+def test_l2norm_rows(hip):
+    x = torch.randn(37, 384, device="cuda")
+    out = hip.l2norm_rows(x)
+    ref = torch.nn.functional.normalize(x, dim=1)
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+
+
+def test_cache_topk_matches_torch(hip):
+    torch.manual_seed(4)
+    index = torch.nn.functional.normalize(torch.randn(5000, 384, device="cuda"), dim=1).to(
+        torch.bfloat16
+    )
+    q = torch.nn.functional.normalize(torch.randn(9, 384, device="cuda"), dim=1).to(
+        torch.bfloat16
+    )
+    hi, idx = hip.cache_topk(index, q)
+    scores = index.float() @ q.float().t()  # (N, B)
+    ref_idx = scores.argmax(dim=0)
+    ref_val = scores.max(dim=0).values
+    from aigw.ops.semcache import _unorder
+
+    for b in range(9):
+        got = _unorder(int(hi[b]))
+        assert abs(got - float(ref_val[b])) < 2e-2
+        # argmax can differ on near-ties in bf16; accept either if scores match
+        assert (
+            int(idx[b]) == int(ref_idx[b])
+            or abs(float(scores[int(idx[b]), b]) - float(ref_val[b])) < 2e-2
+        )
+
+
+def test_semantic_cache_end_to_end(hip):
+    from aigw.ops.semcache import SemanticCache
+    from aigw.ops.tokenizer import GPUTokenizer
+
+    tok = GPUTokenizer(n_merges=4096, device="cuda")
+    cache = SemanticCache(tok.vocab_size, capacity=128, threshold=0.99, device="cuda")
+    texts = [b"what is the capital of france", b"how do i cook rice"]
+    _, _, state = tok.encode_batch(texts)
+    vecs = cache.embed(state["out_ids"], state["req_off"])
+    assert cache.lookup(vecs) == [None, None]
+    cache.insert(vecs[0], b"PARIS")
+    hits = cache.lookup(vecs)
+    assert hits[0] is not None and cache.get(hits[0][0]) == b"PARIS"
+    assert hits[1] is None  # different text does not hit at 0.99
+
+
+def test_kv_scorer_matches_greedy_ref(hip):
+    torch.manual_seed(5)
+    stats = torch.tensor(
+        [[100.0, 1000.0, 0.0, 1.0],
+         [900.0, 1000.0, 2.0, 5.0],
+         [0.0, 500.0, 0.0, 0.0],
+         [499.0, 500.0, 1.0, 2.0]],
+        device="cuda",
+    )
+    pred = torch.tensor([50.0, 50.0, 50.0, 400.0, 50.0], device="cuda")
+    w_kv, w_q, w_a = 1.0, 0.1, 0.05
+    assign = hip.kv_score_assign(stats, pred, w_kv, w_q, w_a).cpu().tolist()
+
+    # greedy CPU reference
+    s = stats.cpu().numpy().copy()
+    ref = []
+    for p in pred.cpu().numpy():
+        scores = []
+        for r in range(len(s)):
+            kv_used, kv_total, queue, active = s[r]
+            sc = w_kv * (1 - (kv_used + p) / max(kv_total, 1)) - w_q * queue - w_a * active
+            if kv_used + p > kv_total:
+                sc -= 1e6
+            scores.append(sc)
+        best = int(np.argmax(scores))
+        ref.append(best)
+        s[best][0] += p
+        s[best][3] += 1
+        s[best][2] += 1
+    assert assign == ref
+
+
+def test_native_extension_is_loaded(hip):
+    """Guard against silent eager fallback: the loaded module must be the
+    in-tree .so."""
+    assert "aigw_hip" in hip.__file__ or hip.__file__.endswith(".so")
+    assert json is not None
