@@ -1,0 +1,160 @@
+"""The ``aigw`` CLI: run / translate / healthcheck / version.
+
+Parity with cmd/aigw (run.go:91, translate.go:41, healthcheck.go), minus
+the Envoy orchestration the reference needs — here ``run`` starts the
+in-process gateway shard(s) directly. ``--shards N`` forks one shard per
+GPU and initializes RCCL-backed state sync between them (one process per
+GPU, torch.distributed over nccl=RCCL).
+
+Usage:
+    python -m aigw run [--config FILE] [--port P] [--shards N] [--gpu]
+    python -m aigw translate FILE...          # CRD YAML -> filter config
+    python -m aigw healthcheck [--port P]
+    python -m aigw version
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import logging
+import os
+import sys
+import urllib.request
+
+import aigw
+from aigw import internalapi
+from aigw.autoconfig import config_from_env
+from aigw.controller import translate_yaml
+from aigw.extproc.server import GatewayServer, run_server
+from aigw.filterapi import ConfigWatcher, RuntimeConfig, load_config_file
+from aigw.filterapi.config import dump_config_yaml
+
+
+def _is_crd_bundle(text: str) -> bool:
+    return "apiVersion" in text and "kind" in text
+
+
+def _load_any_config(path: str | None):
+    if path is None:
+        return config_from_env(), None
+    with open(path, "r", encoding="utf-8") as f:
+        text = f.read()
+    if _is_crd_bundle(text):
+        return translate_yaml(text), None
+    return load_config_file(path), path
+
+
+async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
+    import torch
+
+    cfg, watch_path = _load_any_config(args.config)
+    runtime = RuntimeConfig(cfg)
+
+    gpu_services = None
+    if args.gpu and torch.cuda.is_available():
+        from aigw.gpu import GPUServices
+
+        gpu_services = GPUServices(
+            device=f"cuda:{rank}", enable_cache=args.semantic_cache,
+        )
+    server = GatewayServer(runtime, gpu_services=gpu_services)
+
+    sync = None
+    if world > 1:
+        from aigw.parallel import StateSync
+
+        sync = StateSync(server.limiter)
+        await sync.start()
+
+    watcher = None
+    if watch_path:
+        watcher = ConfigWatcher(watch_path, server.swap_runtime)
+        await watcher.start()
+
+    port = args.port + rank
+    runner = await run_server(server, host=args.host, port=port)
+    print(f"aigw shard {rank}/{world} listening on http://{args.host}:{port}", flush=True)
+    try:
+        await asyncio.Event().wait()
+    finally:
+        if watcher:
+            await watcher.stop()
+        if sync:
+            await sync.stop()
+        await runner.cleanup()
+
+
+def cmd_run(args) -> int:
+    world = int(os.environ.get("WORLD_SIZE", args.shards))
+    rank = int(os.environ.get("RANK", 0))
+    if world > 1:
+        import torch
+        import torch.distributed as dist
+
+        backend = "nccl" if (args.gpu and torch.cuda.is_available()) else "gloo"
+        if backend == "nccl":
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        dist.init_process_group(backend)
+    try:
+        asyncio.run(_run_shard(args, rank, world))
+    except KeyboardInterrupt:
+        pass
+    return 0
+
+
+def cmd_translate(args) -> int:
+    docs = []
+    for path in args.files:
+        with open(path, "r", encoding="utf-8") as f:
+            docs.append(f.read())
+    cfg = translate_yaml("\n---\n".join(docs))
+    sys.stdout.write(dump_config_yaml(cfg))
+    return 0
+
+
+def cmd_healthcheck(args) -> int:
+    url = f"http://127.0.0.1:{args.port}/health"
+    try:
+        with urllib.request.urlopen(url, timeout=5) as r:
+            body = json.loads(r.read())
+            if body.get("status") == "ok":
+                print("ok")
+                return 0
+    except Exception as e:
+        print(f"unhealthy: {e}", file=sys.stderr)
+    return 1
+
+
+def main(argv=None) -> int:
+    logging.basicConfig(level=os.environ.get("AIGW_LOG_LEVEL", "INFO"))
+    ap = argparse.ArgumentParser(prog="aigw", description=__doc__)
+    sub = ap.add_subparsers(dest="cmd", required=True)
+
+    runp = sub.add_parser("run", help="run the gateway")
+    runp.add_argument("--config", default=None, help="filter config or CRD bundle YAML")
+    runp.add_argument("--host", default="0.0.0.0")
+    runp.add_argument("--port", type=int, default=internalapi.DEFAULT_LISTEN_PORT)
+    runp.add_argument("--shards", type=int, default=1, help="shards (one per GPU)")
+    runp.add_argument("--gpu", action="store_true", help="enable GPU services")
+    runp.add_argument("--semantic-cache", action="store_true")
+    runp.set_defaults(fn=cmd_run)
+
+    tr = sub.add_parser("translate", help="CRD bundle -> filter config YAML")
+    tr.add_argument("files", nargs="+")
+    tr.set_defaults(fn=cmd_translate)
+
+    hc = sub.add_parser("healthcheck")
+    hc.add_argument("--port", type=int, default=internalapi.DEFAULT_LISTEN_PORT)
+    hc.set_defaults(fn=cmd_healthcheck)
+
+    ver = sub.add_parser("version")
+    ver.set_defaults(fn=lambda a: (print(f"aigw {aigw.__version__}"), 0)[1])
+
+    args = ap.parse_args(argv)
+    return args.fn(args)
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -412,6 +412,45 @@ def _validate(cfg: Config) -> None:
             raise ConfigError(f"rateLimit {rl.name!r}: limit must be positive")
 
 
+def _camel(s: str) -> str:
+    parts = s.split("_")
+    return parts[0] + "".join(p.title() for p in parts[1:])
+
+
+def config_to_dict(obj) -> object:
+    """Serialize a Config (or any nested part) back to camelCase YAML-ready
+    dicts, omitting fields left at their defaults."""
+    import dataclasses
+    import enum as _enum
+
+    if dataclasses.is_dataclass(obj):
+        out = {}
+        for f in dataclasses.fields(obj):
+            v = getattr(obj, f.name)
+            default = (
+                f.default
+                if f.default is not dataclasses.MISSING
+                else (f.default_factory() if f.default_factory is not dataclasses.MISSING else dataclasses.MISSING)
+            )
+            if default is not dataclasses.MISSING and v == default:
+                continue
+            out[_camel(f.name)] = config_to_dict(v)
+        return out
+    if isinstance(obj, _enum.Enum):
+        return obj.value
+    if isinstance(obj, list):
+        return [config_to_dict(x) for x in obj]
+    if isinstance(obj, dict):
+        return {k: config_to_dict(v) for k, v in obj.items()}
+    return obj
+
+
+def dump_config_yaml(cfg: Config) -> str:
+    d = config_to_dict(cfg)
+    d.setdefault("version", cfg.version)
+    return yaml.safe_dump(d, sort_keys=False)
+
+
 def load_config_file(path: str) -> Config:
     with open(path, "r", encoding="utf-8") as f:
         return load_config(yaml.safe_load(f))

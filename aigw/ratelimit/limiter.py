@@ -17,11 +17,20 @@ off the request critical path).
 
 from __future__ import annotations
 
+import hashlib
 import time
 from dataclasses import dataclass
 from typing import Optional
 
 from aigw.filterapi.config import RateLimitRule
+
+# fixed slot table for cross-shard bucket identity (see aigw.parallel)
+SLOTS = 4096
+
+
+def bucket_slot(rule: str, descriptor: str) -> int:
+    h = hashlib.md5(f"{rule}|{descriptor}".encode()).digest()
+    return int.from_bytes(h[:4], "little") % SLOTS
 
 
 @dataclass
@@ -52,6 +61,7 @@ class RateLimiter:
     def __init__(self, rules: list[RateLimitRule], clock=time.monotonic):
         self.rules = {r.name: r for r in rules}
         self._buckets: dict[tuple[str, str], _Bucket] = {}
+        self._slot_remote: dict[int, int] = {}  # remote spend with no local bucket yet
         self._clock = clock
 
     def _bucket(self, rule: RateLimitRule, descriptor: str) -> _Bucket:
@@ -61,6 +71,8 @@ class RateLimiter:
         if b is None:
             b = _Bucket()
             b.window_start = now
+            # claim any remote spend that arrived before we saw this bucket
+            b.remote_spent = self._slot_remote.pop(bucket_slot(*key), 0)
             self._buckets[key] = b
         elif now - b.window_start >= rule.window_s:
             b.window_start = now
@@ -114,3 +126,12 @@ class RateLimiter:
             return
         b = self._bucket(rule, key[1])
         b.remote_spent += global_delta - own_delta
+
+    def apply_remote_slot(self, slot: int, amount: int) -> None:
+        """Remote spend for a bucket this shard has not seen locally; held
+        per-slot and claimed when the bucket first appears."""
+        for key, b in self._buckets.items():
+            if bucket_slot(*key) == slot:
+                b.remote_spent += amount
+                return
+        self._slot_remote[slot] = self._slot_remote.get(slot, 0) + amount

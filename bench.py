@@ -123,15 +123,17 @@ async def amain(args, rank, world, local_rank):
     gw_url = f"http://127.0.0.1:{gw_port}/v1/chat/completions"
     direct_url = f"http://127.0.0.1:{up_port}/v1/chat/completions"
 
-    # untimed: direct-to-upstream baseline for "added latency"
+    # warmup both paths (connection pools, JIT, allocator) before measuring
+    warm_lat: list[float] = []
+    for _ in range(max(args.warmup, 1)):
+        await fire_step(session, direct_url, payload, args.batch, warm_lat)
+    for _ in range(args.warmup):
+        await fire_step(session, gw_url, payload, args.batch, warm_lat)
+
+    # untimed: direct-to-upstream baseline for "added latency" (warm)
     direct_lat: list[float] = []
     for _ in range(2):
         await fire_step(session, direct_url, payload, args.batch, direct_lat)
-
-    # warmup
-    warm_lat: list[float] = []
-    for _ in range(args.warmup):
-        await fire_step(session, gw_url, payload, args.batch, warm_lat)
 
     def barrier_sync():
         if world > 1:

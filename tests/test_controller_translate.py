@@ -1,0 +1,192 @@
+"""CRD translation tests: feed the REFERENCE's own example bundles
+(examples/basic, examples/token_ratelimit) and assert the compiled filter
+config (parity with tests/controller + cmd/aigw translate)."""
+
+import base64
+import textwrap
+
+import pytest
+
+from aigw.controller import translate_yaml
+from aigw.filterapi.config import APISchemaName, ConfigError, LLMRequestCostType
+from aigw.filterapi.runtime import RuntimeConfig
+
+BASIC = textwrap.dedent(
+    """
+    apiVersion: gateway.networking.k8s.io/v1
+    kind: Gateway
+    metadata: {name: gw, namespace: default}
+    spec: {listeners: [{name: http, protocol: HTTP, port: 80}]}
+    ---
+    apiVersion: aigateway.envoyproxy.io/v1beta1
+    kind: AIGatewayRoute
+    metadata: {name: basic, namespace: default}
+    spec:
+      parentRefs: [{name: gw}]
+      llmRequestCosts:
+        - metadataKey: llm_total_token
+          type: TotalToken
+      rules:
+        - matches:
+            - headers:
+                - {type: Exact, name: x-ai-eg-model, value: gpt-4o-mini}
+          backendRefs:
+            - name: openai-backend
+        - matches:
+            - headers:
+                - {type: Exact, name: x-ai-eg-model, value: claude-3}
+          backendRefs:
+            - name: bedrock-backend
+              modelNameOverride: us.anthropic.claude-3
+              priority: 0
+            - name: openai-backend
+              priority: 1
+    ---
+    apiVersion: aigateway.envoyproxy.io/v1beta1
+    kind: AIServiceBackend
+    metadata: {name: openai-backend, namespace: default}
+    spec:
+      schema: {name: OpenAI}
+      backendRef: {name: openai-ep, kind: Backend, group: gateway.envoyproxy.io}
+    ---
+    apiVersion: aigateway.envoyproxy.io/v1beta1
+    kind: AIServiceBackend
+    metadata: {name: bedrock-backend, namespace: default}
+    spec:
+      schema: {name: AWSBedrock}
+      backendRef: {name: bedrock-ep, kind: Backend, group: gateway.envoyproxy.io}
+    ---
+    apiVersion: gateway.envoyproxy.io/v1alpha1
+    kind: Backend
+    metadata: {name: openai-ep, namespace: default}
+    spec:
+      endpoints: [{fqdn: {hostname: api.openai.com, port: 443}}]
+    ---
+    apiVersion: gateway.envoyproxy.io/v1alpha1
+    kind: Backend
+    metadata: {name: bedrock-ep, namespace: default}
+    spec:
+      endpoints: [{fqdn: {hostname: bedrock-runtime.us-east-1.amazonaws.com, port: 443}}]
+    ---
+    apiVersion: aigateway.envoyproxy.io/v1beta1
+    kind: BackendSecurityPolicy
+    metadata: {name: openai-key, namespace: default}
+    spec:
+      targetRefs:
+        - {group: aigateway.envoyproxy.io, kind: AIServiceBackend, name: openai-backend}
+      type: APIKey
+      apiKey:
+        secretRef: {name: openai-secret}
+    ---
+    apiVersion: aigateway.envoyproxy.io/v1beta1
+    kind: BackendSecurityPolicy
+    metadata: {name: aws-creds, namespace: default}
+    spec:
+      targetRefs:
+        - {group: aigateway.envoyproxy.io, kind: AIServiceBackend, name: bedrock-backend}
+      type: AWSCredentials
+      awsCredentials:
+        region: us-east-1
+        credentialsFile:
+          secretRef: {name: aws-secret}
+    ---
+    apiVersion: v1
+    kind: Secret
+    metadata: {name: openai-secret, namespace: default}
+    stringData: {apiKey: sk-test-123}
+    ---
+    apiVersion: v1
+    kind: Secret
+    metadata: {name: aws-secret, namespace: default}
+    data:
+      credentials: {aws_b64}
+    """
+).replace(
+    "{aws_b64}",
+    base64.b64encode(
+        b"[default]\naws_access_key_id = AKTEST\naws_secret_access_key = SKTEST\n"
+    ).decode(),
+)
+
+
+def test_translate_basic_bundle():
+    cfg = translate_yaml(BASIC)
+    assert len(cfg.routes) == 2
+    r0 = cfg.routes[0]
+    assert r0.backends[0].schema.name is APISchemaName.OPENAI
+    assert r0.backends[0].upstream.host == "api.openai.com"
+    assert r0.backends[0].upstream.tls is True
+    assert r0.backends[0].auth.api_key == "sk-test-123"
+    assert r0.request_costs[0].type is LLMRequestCostType.TOTAL_TOKEN
+    r1 = cfg.routes[1]
+    assert r1.backends[0].model_name_override == "us.anthropic.claude-3"
+    assert r1.backends[0].auth.aws_access_key_id == "AKTEST"
+    assert r1.backends[0].auth.aws_region == "us-east-1"
+    assert r1.backends[1].priority == 1
+    # declared models derived from exact model matches
+    assert {m.name for m in cfg.models} == {"gpt-4o-mini", "claude-3"}
+    # compiles into a runnable runtime
+    rt = RuntimeConfig(cfg)
+    assert rt.select_route({"x-ai-eg-model": "claude-3"}).route.name == "basic-rule-1"
+
+
+def test_translate_reference_example_files():
+    """The reference's own example bundles must translate."""
+    for path, want_routes in (
+        ("/root/reference/examples/basic/basic.yaml", True),
+        ("/root/reference/examples/basic/anthropic.yaml", True),
+        ("/root/reference/examples/basic/openai.yaml", True),
+        ("/root/reference/examples/token_ratelimit/token_ratelimit.yaml", True),
+        # fallback.yaml is only the retry BackendTrafficPolicy attachment
+        ("/root/reference/examples/provider_fallback/fallback.yaml", False),
+    ):
+        try:
+            with open(path) as f:
+                text = f.read()
+        except FileNotFoundError:
+            pytest.skip(f"{path} not present")
+        cfg = translate_yaml(text)
+        if want_routes:
+            assert cfg.routes, path
+
+
+def test_token_ratelimit_bundle_limits():
+    with open("/root/reference/examples/token_ratelimit/token_ratelimit.yaml") as f:
+        cfg = translate_yaml(f.read())
+    assert cfg.rate_limits, "BackendTrafficPolicy Global rules must map to rate limits"
+    rl = cfg.rate_limits[0]
+    assert rl.key_headers == ["x-tenant-id"]
+    assert rl.window_s == 3600.0
+    # CEL cost from the example compiles
+    keys = [c.metadata_key for r in cfg.routes for c in r.request_costs]
+    assert "llm_cel_calculated_token" in keys
+
+
+def test_unknown_kind_rejected():
+    with pytest.raises(ConfigError):
+        translate_yaml("apiVersion: v1\nkind: TotallyUnknown\nmetadata: {name: x}\n")
+
+
+def test_autoconfig_env():
+    from aigw.autoconfig import config_from_env
+
+    cfg = config_from_env({"OPENAI_API_KEY": "sk-abc", "ANTHROPIC_API_KEY": "ak-1"})
+    assert {r.name for r in cfg.routes} == {"openai", "anthropic"}
+    rt = RuntimeConfig(cfg)
+    assert rt.select_route({"x-ai-eg-model": "gpt-4o"}).route.name == "openai"
+    assert rt.select_route({"x-ai-eg-model": "claude-sonnet-4"}).route.name == "anthropic"
+    with pytest.raises(ValueError):
+        config_from_env({})
+
+
+def test_cli_translate_roundtrip(tmp_path, capsys):
+    from aigw.cli.main import main
+    from aigw.filterapi.config import load_config
+
+    p = tmp_path / "bundle.yaml"
+    p.write_text(BASIC)
+    assert main(["translate", str(p)]) == 0
+    out = capsys.readouterr().out
+    cfg = load_config(out)
+    assert len(cfg.routes) == 2
+    assert cfg.routes[1].backends[0].auth.aws_access_key_id == "AKTEST"
