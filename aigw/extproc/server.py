@@ -104,12 +104,20 @@ class GatewayServer:
         metrics: Optional[GenAIMetrics] = None,
         limiter: Optional[RateLimiter] = None,
         gpu_services=None,
+        tracer=None,
         max_body_bytes: int = 50 * 1024 * 1024,  # reference raises Envoy's buffer to 50MiB
     ):
         self.runtime = runtime
         self.metrics = metrics or GenAIMetrics()
         self.limiter = limiter or RateLimiter(runtime.rate_limits)
         self.gpu = gpu_services  # aigw.gpu.services.GPUServices or None
+        self.tracer = tracer  # aigw.tracing.Tracer or None
+        if tracer is not None:
+            from aigw.tracing import ChatSpanRecorder
+
+            self.span_recorder = ChatSpanRecorder(tracer)
+        else:
+            self.span_recorder = None
         self.max_body_bytes = max_body_bytes
         self._session: Optional[aiohttp.ClientSession] = None
         self._started_at = time.time()
@@ -278,10 +286,23 @@ class GatewayServer:
                 return resp
             self.metrics.cache_events.labels(event="miss").inc()
 
-        return await self._dispatch(
-            request, endpoint, route, headers, body, stream, start,
-            model=model, gpu_input_tokens=gpu_input_tokens, cache_key_vec=cache_key_vec,
-        )
+        span = None
+        if self.tracer is not None:
+            span = self.tracer.start_span(f"{JSON_ENDPOINTS.get(endpoint, 'request')} {model}",
+                                          headers)
+        try:
+            resp = await self._dispatch(
+                request, endpoint, route, headers, body, stream, start,
+                model=model, gpu_input_tokens=gpu_input_tokens, cache_key_vec=cache_key_vec,
+                span=span,
+            )
+            if span is not None:
+                self.tracer.end_span(span, error=None if resp.status < 500 else f"status {resp.status}")
+            return resp
+        except Exception as e:
+            if span is not None:
+                self.tracer.end_span(span, error=str(e))
+            raise
 
     async def _dispatch(
         self,
@@ -296,6 +317,7 @@ class GatewayServer:
         model: str,
         gpu_input_tokens: int = 0,
         cache_key_vec=None,
+        span=None,
     ) -> web.StreamResponse:
         rt = self.runtime
         assert self._session is not None, "GatewayServer.start() not called"
@@ -361,6 +383,14 @@ class GatewayServer:
             url = backend.upstream.base_url + tr.path
             if backend.upstream.hostname:
                 up_headers["host"] = backend.upstream.hostname
+            if span is not None:
+                # span context joins provider-side traces (A.8 / §5.1)
+                up_headers["traceparent"] = span.traceparent()
+                self.span_recorder.record_request(
+                    span, body,
+                    provider=provider_from_schema(backend.schema.name.value, backend.name),
+                    backend=backend.name,
+                )
             try:
                 timeout = aiohttp.ClientTimeout(total=backend.timeout_s)
                 async with self._session.post(
@@ -384,11 +414,11 @@ class GatewayServer:
                     if stream:
                         return await self._stream_response(
                             request, endpoint, route, backend, translator, upstream,
-                            headers, model, start, gpu_input_tokens,
+                            headers, model, start, gpu_input_tokens, span=span,
                         )
                     return await self._unary_response(
                         endpoint, route, backend, translator, upstream, headers,
-                        model, start, gpu_input_tokens, body, cache_key_vec,
+                        model, start, gpu_input_tokens, body, cache_key_vec, span=span,
                     )
             except (aiohttp.ClientError, asyncio.TimeoutError) as e:
                 last_error = f"upstream {backend.name}: {type(e).__name__}: {e}"
@@ -462,7 +492,7 @@ class GatewayServer:
 
     async def _unary_response(
         self, endpoint, route, backend, translator, upstream, headers, model,
-        start, gpu_input_tokens, orig_body, cache_key_vec,
+        start, gpu_input_tokens, orig_body, cache_key_vec, span=None,
     ) -> web.Response:
         data = await upstream.read()  # aiohttp auto-decompresses gzip/br
         try:
@@ -472,6 +502,8 @@ class GatewayServer:
             return _json_error(502, f"response translation failed: {e}", "translation_error")
         usage = rtl.usage or Usage()
         self._apply_costs(route, backend, headers, model, usage, gpu_input_tokens)
+        if span is not None:
+            self.span_recorder.record_response(span, usage, rtl.response_model)
         self._finish_metrics(
             endpoint, route, backend, model, rtl.response_model, usage, start,
             status=upstream.status,
@@ -489,7 +521,7 @@ class GatewayServer:
 
     async def _stream_response(
         self, request, endpoint, route, backend, translator, upstream,
-        headers, model, start, gpu_input_tokens,
+        headers, model, start, gpu_input_tokens, span=None,
     ) -> web.StreamResponse:
         hdrs = translator.response_headers(upstream.status, dict(upstream.headers))
         content_type = hdrs.get("content-type") or upstream.headers.get(
@@ -527,6 +559,8 @@ class GatewayServer:
             raise
         finally:
             self._apply_costs(route, backend, headers, model, usage, gpu_input_tokens)
+            if span is not None:
+                self.span_recorder.record_response(span, usage, response_model)
             self._finish_metrics(
                 endpoint, route, backend, model, response_model, usage, start,
                 status=upstream.status, ttft=ttft if ttft >= 0 else 0.0,

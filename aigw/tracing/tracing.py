@@ -1,0 +1,204 @@
+"""Request tracing with OpenInference / OTel-GenAI semantic conventions.
+
+Parity with internal/tracing:
+
+- convention selected by ``AI_GATEWAY_TRACING_SEMCONV`` ∈ {``openinference``
+  (default), ``gen_ai``}; an unknown value is a startup ERROR, not a
+  fallback (semconv.go:66-96);
+- ``OTEL_SDK_DISABLED=true`` disables tracing entirely;
+- spans cover the whole gateway request; streamed chunks are recorded as
+  span events; usage lands in convention-specific attributes;
+- W3C ``traceparent`` is injected into UPSTREAM request headers so
+  provider-side traces join (processor_impl.go:314-320);
+- exporters: JSONL file (``AIGW_TRACE_FILE``), console, in-memory (tests).
+  OTLP wire export is intentionally not implemented in this environment
+  (no OTel SDK available offline); the span model keeps OTLP-compatible
+  fields so an exporter can be added without touching call sites.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import secrets
+import time
+from dataclasses import dataclass, field
+from typing import Optional
+
+
+@dataclass
+class Span:
+    name: str
+    trace_id: str
+    span_id: str
+    parent_span_id: str = ""
+    start_ns: int = 0
+    end_ns: int = 0
+    attributes: dict = field(default_factory=dict)
+    events: list = field(default_factory=list)
+    status: str = "OK"
+
+    def set(self, key: str, value) -> None:
+        if value is not None:
+            self.attributes[key] = value
+
+    def add_event(self, name: str, attributes: Optional[dict] = None) -> None:
+        self.events.append({"name": name, "ts_ns": time.time_ns(),
+                            "attributes": attributes or {}})
+
+    def traceparent(self) -> str:
+        return f"00-{self.trace_id}-{self.span_id}-01"
+
+
+class InMemoryExporter:
+    def __init__(self):
+        self.spans: list[Span] = []
+
+    def export(self, span: Span) -> None:
+        self.spans.append(span)
+
+
+class JSONLExporter:
+    def __init__(self, path: str):
+        self.path = path
+        self._f = open(path, "a", encoding="utf-8")
+
+    def export(self, span: Span) -> None:
+        self._f.write(json.dumps({
+            "name": span.name,
+            "traceId": span.trace_id,
+            "spanId": span.span_id,
+            "parentSpanId": span.parent_span_id,
+            "startTimeUnixNano": span.start_ns,
+            "endTimeUnixNano": span.end_ns,
+            "attributes": span.attributes,
+            "events": span.events,
+            "status": span.status,
+        }) + "\n")
+        self._f.flush()
+
+
+class ConsoleExporter:
+    def export(self, span: Span) -> None:
+        dur_ms = (span.end_ns - span.start_ns) / 1e6
+        print(f"[span] {span.name} {dur_ms:.1f}ms {span.attributes}")
+
+
+class Tracer:
+    def __init__(self, semconv: str = "openinference", exporter=None,
+                 header_attributes: Optional[dict[str, str]] = None):
+        if semconv not in ("openinference", "gen_ai"):
+            raise ValueError(
+                f"unknown tracing semantic convention {semconv!r} "
+                "(expected 'openinference' or 'gen_ai')"
+            )
+        self.semconv = semconv
+        self.exporter = exporter or InMemoryExporter()
+        # request-header name -> span attribute name (mainlib flag
+        # -spanRequestHeaderAttributes)
+        self.header_attributes = header_attributes or {}
+
+    def start_span(self, name: str, headers: Optional[dict[str, str]] = None) -> Span:
+        parent_trace = parent_span = ""
+        if headers:
+            tp = headers.get("traceparent", "")
+            parts = tp.split("-")
+            if len(parts) == 4 and len(parts[1]) == 32:
+                parent_trace, parent_span = parts[1], parts[2]
+        span = Span(
+            name=name,
+            trace_id=parent_trace or secrets.token_hex(16),
+            span_id=secrets.token_hex(8),
+            parent_span_id=parent_span,
+            start_ns=time.time_ns(),
+        )
+        if headers:
+            for hdr, attr in self.header_attributes.items():
+                v = headers.get(hdr.lower())
+                if v is not None:
+                    span.set(attr, v)
+        return span
+
+    def end_span(self, span: Span, error: Optional[str] = None) -> None:
+        span.end_ns = time.time_ns()
+        if error:
+            span.status = "ERROR"
+            span.set("error.message", error)
+        self.exporter.export(span)
+
+
+class ChatSpanRecorder:
+    """Fills span attributes for a chat request per the selected
+    convention (internal/tracing/openinference + otelgenai recorders)."""
+
+    def __init__(self, tracer: Tracer, capture_content: bool = False):
+        self.tracer = tracer
+        self.capture_content = capture_content
+
+    def record_request(self, span: Span, body: dict, *, provider: str, backend: str) -> None:
+        model = body.get("model", "")
+        if self.tracer.semconv == "openinference":
+            span.set("openinference.span.kind", "LLM")
+            span.set("llm.model_name", model)
+            span.set("llm.provider", provider)
+            span.set("llm.system", provider)
+            params = {
+                k: body[k]
+                for k in ("temperature", "top_p", "max_tokens", "max_completion_tokens", "stream")
+                if k in body
+            }
+            span.set("llm.invocation_parameters", json.dumps(params) if params else None)
+            if self.capture_content:
+                for i, m in enumerate(body.get("messages") or []):
+                    span.set(f"llm.input_messages.{i}.message.role", m.get("role"))
+                    c = m.get("content")
+                    if isinstance(c, str):
+                        span.set(f"llm.input_messages.{i}.message.content", c)
+        else:
+            span.set("gen_ai.operation.name", "chat")
+            span.set("gen_ai.provider.name", provider)
+            span.set("gen_ai.request.model", model)
+            span.set("gen_ai.request.temperature", body.get("temperature"))
+            span.set("gen_ai.request.max_tokens",
+                     body.get("max_completion_tokens") or body.get("max_tokens"))
+        span.set("aigw.backend", backend)
+
+    def record_chunk(self, span: Span) -> None:
+        span.add_event("gen_ai.chunk")
+
+    def record_response(self, span: Span, usage, response_model: str,
+                        finish_reason: str = "") -> None:
+        if self.tracer.semconv == "openinference":
+            span.set("llm.token_count.prompt", usage.input_tokens)
+            span.set("llm.token_count.completion", usage.output_tokens)
+            span.set("llm.token_count.total", usage.total_tokens)
+        else:
+            span.set("gen_ai.response.model", response_model)
+            span.set("gen_ai.usage.input_tokens", usage.input_tokens)
+            span.set("gen_ai.usage.output_tokens", usage.output_tokens)
+        if finish_reason:
+            span.set("gen_ai.response.finish_reasons", [finish_reason])
+
+
+def tracing_from_env(env: Optional[dict] = None) -> Optional[Tracer]:
+    """NewTracingFromEnv analogue: returns None when disabled."""
+    env = env if env is not None else dict(os.environ)
+    if env.get("OTEL_SDK_DISABLED", "").lower() == "true":
+        return None
+    semconv = env.get("AI_GATEWAY_TRACING_SEMCONV", "openinference")
+    trace_file = env.get("AIGW_TRACE_FILE", "")
+    if trace_file:
+        exporter = JSONLExporter(trace_file)
+    elif env.get("OTEL_TRACES_EXPORTER", "") == "console":
+        exporter = ConsoleExporter()
+    elif env.get("OTEL_EXPORTER_OTLP_ENDPOINT") or env.get("OTEL_TRACES_EXPORTER"):
+        # no offline OTLP wire support; keep spans observable locally
+        exporter = ConsoleExporter()
+    else:
+        return None
+    header_attrs = {}
+    for pair in env.get("AIGW_SPAN_REQUEST_HEADER_ATTRIBUTES", "").split(","):
+        if ":" in pair:
+            h, a = pair.split(":", 1)
+            header_attrs[h.strip()] = a.strip()
+    return Tracer(semconv=semconv, exporter=exporter, header_attributes=header_attrs)
