@@ -154,6 +154,22 @@ class GatewayServer:
         app.router.add_get("/metrics", self._handle_metrics)
         if self.gpu is not None:
             app.router.add_post("/v1/gateway/tokenize", self._handle_gpu_tokenize)
+        mcp_cfg = self.runtime.config.mcp
+        if mcp_cfg is not None and mcp_cfg.routes:
+            from aigw.mcp import MCPProxy
+
+            self._mcp_proxies = []
+            for mr in mcp_cfg.routes:
+                proxy = MCPProxy(mr, mcp_cfg.session_seed)
+                self._mcp_proxies.append(proxy)
+                for method in ("POST", "GET", "DELETE"):
+                    app.router.add_route(method, mr.path, proxy.handle)
+
+            async def _close_mcp(_app):
+                for p in self._mcp_proxies:
+                    await p.close()
+
+            app.on_cleanup.append(_close_mcp)
         app.on_startup.append(lambda _app: self.start())
         app.on_cleanup.append(lambda _app: self.close())
         return app

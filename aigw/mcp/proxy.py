@@ -1,0 +1,304 @@
+"""MCP (Model Context Protocol) gateway: N upstream MCP servers behind one
+streamable-HTTP endpoint.
+
+Parity targets (internal/mcpproxy/):
+- ``initialize`` fans out to every backend, merges capabilities, and seals
+  the per-backend session ids into one stateless encrypted session blob
+  (handlers.go:569, session.go);
+- ``tools/list`` aggregates with ``<backend>__<tool>`` name prefixing and
+  per-backend ToolSelector filtering (include/exclude exact + RE2-style
+  regex, exclude wins — mcpconfig.go:81-98);
+- ``tools/call`` strips the prefix and routes to the owning backend with
+  that backend's own session id (handlers.go:728);
+- ``resources/list``, ``prompts/list`` aggregate; ``resources/read`` routes
+  by trying the owning backends; ``ping`` answers locally;
+- upstream responses in either JSON or SSE framing are folded to the
+  JSON-RPC response with the matching id.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import re
+from typing import Optional
+
+import aiohttp
+from aiohttp import web
+
+from aigw import internalapi
+from aigw.filterapi.config import MCPBackend, MCPRoute
+from aigw.mcp.session import SessionCrypto, SessionError
+from aigw.mutator import apply_header_mutation
+from aigw.translator.sse import SSEDecoder
+
+logger = logging.getLogger("aigw.mcp")
+
+PREFIX_SEP = "__"
+PROTOCOL_VERSION = "2025-06-18"
+
+
+def _rpc_error(id_, code: int, message: str, status: int = 200) -> web.Response:
+    return web.json_response(
+        {"jsonrpc": "2.0", "id": id_, "error": {"code": code, "message": message}},
+        status=status,
+    )
+
+
+class MCPProxy:
+    def __init__(self, route: MCPRoute, session_seed: str, client_factory=None):
+        self.route = route
+        self.crypto = SessionCrypto(session_seed)
+        self._client_factory = client_factory
+        self._session: Optional[aiohttp.ClientSession] = None
+        self._tool_res: dict[str, list] = {}
+        for b in route.backends:
+            self._tool_res[b.name] = [re.compile(p) for p in b.tool_exclude + b.tool_include]
+
+    async def _client(self) -> aiohttp.ClientSession:
+        if self._session is None:
+            if self._client_factory is not None:
+                self._session = self._client_factory()
+            else:
+                self._session = aiohttp.ClientSession(
+                    connector=aiohttp.TCPConnector(limit=0)
+                )
+        return self._session
+
+    async def close(self) -> None:
+        if self._session is not None:
+            await self._session.close()
+            self._session = None
+
+    # ---- upstream I/O --------------------------------------------------------
+
+    async def _call_backend(
+        self, backend: MCPBackend, payload: dict, session_id: str = ""
+    ) -> tuple[Optional[dict], str]:
+        """POST a JSON-RPC message; returns (response message or None for
+        notifications, new session id if the backend issued one)."""
+        client = await self._client()
+        headers = {
+            "content-type": "application/json",
+            "accept": "application/json, text/event-stream",
+        }
+        if session_id:
+            headers[internalapi.MCP_SESSION_ID_HEADER] = session_id
+        if backend.headers is not None:
+            headers = apply_header_mutation(headers, backend.headers)
+        if backend.auth is not None and backend.auth.api_key:
+            headers["authorization"] = f"Bearer {backend.auth.api_key}"
+        url = backend.upstream.base_url + backend.path
+        async with client.post(url, json=payload, headers=headers) as resp:
+            new_session = resp.headers.get(internalapi.MCP_SESSION_ID_HEADER, "")
+            if resp.status == 202:
+                return None, new_session
+            if resp.status >= 400:
+                body = await resp.text()
+                raise RuntimeError(f"backend {backend.name} returned {resp.status}: {body[:200]}")
+            ctype = resp.headers.get("content-type", "")
+            if ctype.startswith("text/event-stream"):
+                dec = SSEDecoder()
+                want_id = payload.get("id")
+                async for chunk in resp.content.iter_any():
+                    for ev in dec.feed(chunk):
+                        if not ev.data:
+                            continue
+                        try:
+                            msg = json.loads(ev.data)
+                        except ValueError:
+                            continue
+                        if msg.get("id") == want_id and ("result" in msg or "error" in msg):
+                            return msg, new_session
+                return None, new_session
+            data = await resp.read()
+            return (json.loads(data) if data else None), new_session
+
+    # ---- tool filtering ------------------------------------------------------
+
+    @staticmethod
+    def _selected(backend: MCPBackend, tool: str) -> bool:
+        """include/exclude exact + regex; exclude wins (mcpconfig.go:81-98)."""
+
+        def matches(pats: list[str]) -> bool:
+            for p in pats:
+                if p == tool:
+                    return True
+                try:
+                    if re.fullmatch(p, tool):
+                        return True
+                except re.error:
+                    continue
+            return False
+
+        if backend.tool_exclude and matches(backend.tool_exclude):
+            return False
+        if backend.tool_include:
+            return matches(backend.tool_include)
+        return True
+
+    # ---- handlers ------------------------------------------------------------
+
+    async def handle(self, request: web.Request) -> web.StreamResponse:
+        if request.method == "GET":
+            # server-initiated stream endpoint: no push support yet
+            return web.Response(status=405, text="SSE server stream not supported")
+        if request.method == "DELETE":
+            return web.Response(status=202)  # stateless sessions: nothing to delete
+        try:
+            payload = json.loads(await request.read())
+        except ValueError:
+            return _rpc_error(None, -32700, "parse error", status=400)
+        if isinstance(payload, list):
+            return _rpc_error(None, -32600, "batch requests not supported", status=400)
+        method = payload.get("method", "")
+        id_ = payload.get("id")
+        token = request.headers.get(internalapi.MCP_SESSION_ID_HEADER, "")
+        sessions: dict[str, str] = {}
+        if token:
+            try:
+                sessions = self.crypto.open(token).get("s", {})
+            except SessionError:
+                return _rpc_error(id_, -32600, "invalid session", status=404)
+        try:
+            if method == "initialize":
+                return await self._initialize(payload)
+            if method == "ping":
+                return web.json_response({"jsonrpc": "2.0", "id": id_, "result": {}})
+            if method.startswith("notifications/"):
+                await self._fan_out_notification(payload, sessions)
+                return web.Response(status=202)
+            if method == "tools/list":
+                return await self._aggregate_list(payload, sessions, "tools", "name")
+            if method == "prompts/list":
+                return await self._aggregate_list(payload, sessions, "prompts", "name")
+            if method == "resources/list":
+                return await self._aggregate_list(payload, sessions, "resources", "name")
+            if method in ("tools/call", "prompts/get"):
+                return await self._routed_call(payload, sessions, key="name")
+            if method == "resources/read":
+                return await self._resources_read(payload, sessions)
+            return _rpc_error(id_, -32601, f"method {method!r} not found")
+        except Exception as e:
+            logger.exception("mcp %s failed", method)
+            return _rpc_error(id_, -32603, str(e))
+
+    async def _initialize(self, payload: dict) -> web.Response:
+        id_ = payload.get("id")
+        results = await asyncio.gather(
+            *(self._call_backend(b, payload) for b in self.route.backends),
+            return_exceptions=True,
+        )
+        sessions: dict[str, str] = {}
+        caps: dict = {}
+        ok = 0
+        for b, res in zip(self.route.backends, results):
+            if isinstance(res, Exception):
+                logger.warning("initialize failed for backend %s: %s", b.name, res)
+                continue
+            msg, new_session = res
+            if new_session:
+                sessions[b.name] = new_session
+            if msg and "result" in msg:
+                ok += 1
+                for k, v in (msg["result"].get("capabilities") or {}).items():
+                    if isinstance(v, dict):
+                        caps.setdefault(k, {}).update(v)
+                    else:
+                        caps[k] = v
+        if ok == 0:
+            return _rpc_error(id_, -32603, "all MCP backends failed to initialize")
+        token = self.crypto.seal({"s": sessions})
+        resp = web.json_response(
+            {
+                "jsonrpc": "2.0",
+                "id": id_,
+                "result": {
+                    "protocolVersion": PROTOCOL_VERSION,
+                    "capabilities": caps,
+                    "serverInfo": {"name": "aigw-mcp-gateway", "version": "0.1.0"},
+                },
+            }
+        )
+        resp.headers[internalapi.MCP_SESSION_ID_HEADER] = token
+        return resp
+
+    async def _fan_out_notification(self, payload: dict, sessions: dict[str, str]) -> None:
+        await asyncio.gather(
+            *(
+                self._call_backend(b, payload, sessions.get(b.name, ""))
+                for b in self.route.backends
+            ),
+            return_exceptions=True,
+        )
+
+    async def _aggregate_list(
+        self, payload: dict, sessions: dict[str, str], list_key: str, name_key: str
+    ) -> web.Response:
+        id_ = payload.get("id")
+        results = await asyncio.gather(
+            *(
+                self._call_backend(b, payload, sessions.get(b.name, ""))
+                for b in self.route.backends
+            ),
+            return_exceptions=True,
+        )
+        merged: list = []
+        for b, res in zip(self.route.backends, results):
+            if isinstance(res, Exception):
+                logger.warning("%s failed for backend %s: %s", list_key, b.name, res)
+                continue
+            msg, _ = res
+            if not msg or "result" not in msg:
+                continue
+            for item in msg["result"].get(list_key) or []:
+                name = item.get(name_key, "")
+                if list_key == "tools" and not self._selected(b, name):
+                    continue
+                item = dict(item)
+                item[name_key] = f"{b.name}{PREFIX_SEP}{name}"
+                merged.append(item)
+        return web.json_response(
+            {"jsonrpc": "2.0", "id": id_, "result": {list_key: merged}}
+        )
+
+    def _backend_for(self, prefixed: str) -> tuple[Optional[MCPBackend], str]:
+        name, sep, rest = prefixed.partition(PREFIX_SEP)
+        if sep:
+            for b in self.route.backends:
+                if b.name == name:
+                    return b, rest
+        return None, prefixed
+
+    async def _routed_call(self, payload: dict, sessions: dict[str, str], key: str) -> web.Response:
+        id_ = payload.get("id")
+        params = payload.get("params") or {}
+        backend, bare = self._backend_for(params.get(key, ""))
+        if backend is None:
+            return _rpc_error(id_, -32602, f"unknown backend prefix in {params.get(key)!r}")
+        if payload.get("method") == "tools/call" and not self._selected(backend, bare):
+            return _rpc_error(id_, -32602, f"tool {bare!r} is not allowed")
+        fwd = dict(payload)
+        fwd["params"] = dict(params)
+        fwd["params"][key] = bare
+        msg, _ = await self._call_backend(backend, fwd, sessions.get(backend.name, ""))
+        if msg is None:
+            return _rpc_error(id_, -32603, "no response from backend")
+        msg["id"] = id_
+        return web.json_response(msg)
+
+    async def _resources_read(self, payload: dict, sessions: dict[str, str]) -> web.Response:
+        id_ = payload.get("id")
+        last_err = "no backends"
+        for b in self.route.backends:
+            try:
+                msg, _ = await self._call_backend(b, payload, sessions.get(b.name, ""))
+                if msg and "result" in msg:
+                    msg["id"] = id_
+                    return web.json_response(msg)
+                if msg and "error" in msg:
+                    last_err = msg["error"].get("message", "error")
+            except Exception as e:
+                last_err = str(e)
+        return _rpc_error(id_, -32002, f"resource not found: {last_err}")
