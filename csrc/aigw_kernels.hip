@@ -313,37 +313,51 @@ __device__ __forceinline__ unsigned long long pack_score(float s, unsigned idx) 
   return ((unsigned long long)u << 32) | idx;
 }
 
+// Hoists this wave's 16-row index tile into registers ONCE (KSTEPS is a
+// template constant so a_frag stays in VGPRs — a runtime bound would spill
+// to scratch and re-read it per query block, rule #20); the query loop then
+// touches only the L2-resident query block. Global atomics are first folded
+// across the block's 4 waves through LDS (4x fewer device-scope atomicMax
+// on the hot n_q addresses).
+template <int KSTEPS>
 __global__ void __launch_bounds__(256)
-cache_topk_kernel(const bf16* __restrict__ index, long long n_rows,
-                  const bf16* __restrict__ q, int n_q, int dim,
-                  unsigned long long* __restrict__ best /* n_q */) {
+cache_topk_kernel_t(const bf16* __restrict__ index, long long n_rows,
+                    const bf16* __restrict__ q, int n_q, int dim,
+                    unsigned long long* __restrict__ best /* n_q */) {
+  __shared__ unsigned long long blk_best[16];
   long long i0 = ((long long)blockIdx.x * 4 + (threadIdx.x >> 6)) * 16;
-  if (i0 >= n_rows) return;
   int lane = threadIdx.x & 63;
   int row = lane & 15;
   int kgrp = lane >> 4;
-  bool i_ok = (i0 + row) < n_rows;
+  bool i_ok = i0 < n_rows && (i0 + row) < n_rows;
+  short8 a_frag[KSTEPS];
+  #pragma unroll
+  for (int ks = 0; ks < KSTEPS; ++ks) {
+    int kk = ks * 32 + kgrp * 8;
+    if (i_ok)
+      a_frag[ks] = *reinterpret_cast<const short8*>(&index[(i0 + row) * dim + kk]);
+    else
+      a_frag[ks] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
   for (int q0 = 0; q0 < n_q; q0 += 16) {
     floatx4 acc = {0.f, 0.f, 0.f, 0.f};
-    bool q_ok = (q0 + row) < n_q;
-    for (int k = 0; k < dim; k += 32) {
-      short8 a = {0, 0, 0, 0, 0, 0, 0, 0}, b = {0, 0, 0, 0, 0, 0, 0, 0};
-      int kk = k + kgrp * 8;
-      if (i_ok && kk < dim)
-        a = *reinterpret_cast<const short8*>(&index[(i0 + row) * dim + kk]);
-      if (q_ok && kk < dim)
+    bool q_in = (q0 + row) < n_q;
+    #pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      short8 b = {0, 0, 0, 0, 0, 0, 0, 0};
+      int kk = ks * 32 + kgrp * 8;
+      if (q_in)
         b = *reinterpret_cast<const short8*>(&q[(long long)(q0 + row) * dim + kk]);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[ks], b, acc, 0, 0, 0);
     }
     // acc reg r holds C[idx_row = kgrp*4+r][query = row]: fold the 4 regs
     // locally, then across the 4 lane groups sharing this query column
-    // (lanes differing in bits 4..5), then one atomicMax per query.
-    bool q_in = (q0 + row) < n_q;
+    // (lanes differing in bits 4..5), then block-wide via LDS.
     unsigned long long p = 0;
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
       long long irow = i0 + kgrp * 4 + r;
-      float s = (irow < n_rows && q_in) ? acc[r] : -1e30f;
+      float s = (i_ok && irow < n_rows && q_in) ? acc[r] : -1e30f;
       unsigned long long pk = pack_score(s, (unsigned)(irow & 0xFFFFFFFF));
       if (pk > p) p = pk;
     }
@@ -352,7 +366,13 @@ cache_topk_kernel(const bf16* __restrict__ index, long long n_rows,
       unsigned long long o = shfl_xor_u64(p, off);
       if (o > p) p = o;
     }
-    if (kgrp == 0 && q_in) atomicMax(&best[q0 + row], p);
+    if (threadIdx.x < 16) blk_best[threadIdx.x] = 0;
+    __syncthreads();
+    if (kgrp == 0 && q_in) atomicMax(&blk_best[row], p);
+    __syncthreads();
+    if (threadIdx.x < 16 && (q0 + threadIdx.x) < n_q && blk_best[threadIdx.x])
+      atomicMax(&best[q0 + threadIdx.x], blk_best[threadIdx.x]);
+    __syncthreads();
   }
 }
 
@@ -529,15 +549,35 @@ std::vector<at::Tensor> cache_topk(at::Tensor index, at::Tensor q) {
   long long n_rows = index.size(0);
   int n_q = (int)q.size(0), dim = (int)q.size(1);
   TORCH_CHECK(index.size(1) == dim, "dim mismatch");
+  TORCH_CHECK(dim % 32 == 0 && dim <= 512, "dim must be a multiple of 32, <= 512");
   auto best = at::zeros({n_q}, at::TensorOptions()
                                    .dtype(at::kLong)
                                    .device(q.device()));
   long long tiles = (n_rows + 63) / 64;
-  hipLaunchKernelGGL(cache_topk_kernel, dim3((unsigned)tiles), dim3(256), 0,
-                     current_stream(),
-                     reinterpret_cast<bf16*>(index.data_ptr<at::BFloat16>()), n_rows,
-                     reinterpret_cast<bf16*>(q.data_ptr<at::BFloat16>()), n_q, dim,
-                     reinterpret_cast<unsigned long long*>(best.data_ptr<int64_t>()));
+  auto* index_p = reinterpret_cast<bf16*>(index.data_ptr<at::BFloat16>());
+  auto* q_p = reinterpret_cast<bf16*>(q.data_ptr<at::BFloat16>());
+  auto* best_p = reinterpret_cast<unsigned long long*>(best.data_ptr<int64_t>());
+  switch (dim >> 5) {
+    case 12:  // dim = 384 (bge-small) — the hot path
+      hipLaunchKernelGGL((cache_topk_kernel_t<12>), dim3((unsigned)tiles), dim3(256), 0,
+                         current_stream(), index_p, n_rows, q_p, n_q, dim, best_p);
+      break;
+    case 8:  // dim = 256
+      hipLaunchKernelGGL((cache_topk_kernel_t<8>), dim3((unsigned)tiles), dim3(256), 0,
+                         current_stream(), index_p, n_rows, q_p, n_q, dim, best_p);
+      break;
+    case 16:  // dim = 512
+      hipLaunchKernelGGL((cache_topk_kernel_t<16>), dim3((unsigned)tiles), dim3(256), 0,
+                         current_stream(), index_p, n_rows, q_p, n_q, dim, best_p);
+      break;
+    case 4:  // dim = 128
+      hipLaunchKernelGGL((cache_topk_kernel_t<4>), dim3((unsigned)tiles), dim3(256), 0,
+                         current_stream(), index_p, n_rows, q_p, n_q, dim, best_p);
+      break;
+    default:
+      TORCH_CHECK(false, "cache_topk: unsupported dim ", dim,
+                  " (supported: 128/256/384/512)");
+  }
   // unpack: score = orderable^-1(hi32), idx = lo32
   auto hi = best.bitwise_right_shift(32).to(at::kLong);
   auto idx = best.bitwise_and(0xFFFFFFFFLL).to(at::kInt);
