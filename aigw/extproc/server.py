@@ -262,6 +262,8 @@ class GatewayServer:
                 raise ValueError("body must be a JSON object")
         except ValueError as e:
             return _json_error(400, f"invalid request body: {e}")
+        if not raw:
+            raw = b"{}"
 
         headers = self._ingress_headers(request)
         model = str(body.get("model", ""))
@@ -310,7 +312,7 @@ class GatewayServer:
             resp = await self._dispatch(
                 request, endpoint, route, headers, body, stream, start,
                 model=model, gpu_input_tokens=gpu_input_tokens, cache_key_vec=cache_key_vec,
-                span=span,
+                span=span, raw=raw,
             )
             if span is not None:
                 self.tracer.end_span(span, error=None if resp.status < 500 else f"status {resp.status}")
@@ -334,6 +336,7 @@ class GatewayServer:
         gpu_input_tokens: int = 0,
         cache_key_vec=None,
         span=None,
+        raw: bytes = b"",
     ) -> web.StreamResponse:
         rt = self.runtime
         assert self._session is not None, "GatewayServer.start() not called"
@@ -341,15 +344,16 @@ class GatewayServer:
         if attempts_left == 0:
             return _json_error(503, "route has no backends", "no_backend")
         last_error: Optional[str] = None
-        first = True
+        attempt = 0
 
         for backend in backend_attempts(route):
             if attempts_left <= 0:
                 break
             attempts_left -= 1
+            first = attempt == 0
+            attempt += 1
             if not first:
                 self.metrics.retries_total.labels(route=route.route.name).inc()
-            first = False
 
             try:
                 translator = get_translator(
@@ -359,8 +363,11 @@ class GatewayServer:
                 last_error = str(e)
                 continue
 
-            # Re-translate the ORIGINAL body per try (A.8 retry semantics).
-            body_copy = json.loads(json.dumps(body))
+            # First attempt works on the live parsed body (translators may
+            # mutate it); every RETRY re-parses the ORIGINAL raw bytes, which
+            # is the A.8 "restore original body per try" semantics without a
+            # defensive deep copy on the hot path.
+            body_copy = body if first else json.loads(raw)
             override = backend.model_name_override or route.route.model_name_override
             try:
                 tr = translator.request(
@@ -368,6 +375,7 @@ class GatewayServer:
                     model_override=override,
                     stream=stream,
                     force_include_usage=bool(route.costs),
+                    raw=raw if first else b"",
                 )
             except TranslationError as e:
                 return _json_error(422, str(e))

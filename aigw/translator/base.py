@@ -95,7 +95,7 @@ class Translator:
     # streaming translators that re-encode to SSE override response
     # content-type via response_headers().
     def request(self, body: dict, *, model_override: str = "", stream: bool = False,
-                force_include_usage: bool = False) -> RequestTranslation:
+                force_include_usage: bool = False, raw: bytes = b"") -> RequestTranslation:
         raise NotImplementedError
 
     def response_headers(self, status: int, headers: dict[str, str]) -> dict[str, str]:

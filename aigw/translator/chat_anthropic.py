@@ -50,7 +50,7 @@ class _AnthropicBackedChat(Translator):
     def _adjust_request(self, areq: dict, model: str) -> dict:
         return areq
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body = dict(body)
             body["model"] = model_override
@@ -199,7 +199,7 @@ class _AnthropicPassthrough(Translator):
     def _adjust_request(self, body, model):
         return body
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body["model"] = model_override
         self._model = body.get("model", "")
@@ -341,7 +341,7 @@ class AnthropicToOpenAIChat(Translator):
         self._machine = OpenAIToAnthropicStream()
         self._model = ""
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body["model"] = model_override
         self._model = body.get("model", "")

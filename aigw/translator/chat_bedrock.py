@@ -244,7 +244,7 @@ class OpenAIToBedrockChat(Translator):
         self._next_tool = 0
         self._done = False
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body["model"] = model_override
         self._model = body.get("model", "")
@@ -374,7 +374,7 @@ class OpenAIToBedrockEmbeddings(Translator):
     def __init__(self, **kw):
         self._model = ""
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body["model"] = model_override
         self._model = body.get("model", "")

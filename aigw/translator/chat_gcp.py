@@ -224,7 +224,7 @@ class OpenAIToGeminiChat(Translator):
         self._finish = None
         self._done = False
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body["model"] = model_override
         self._model = body.get("model", "")
@@ -341,7 +341,7 @@ class OpenAIToGCPEmbeddings(Translator):
         self.gcp_region = gcp_region
         self._model = ""
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body["model"] = model_override
         self._model = body.get("model", "")

@@ -30,7 +30,7 @@ class CohereRerank(Translator):
     def __init__(self, **kw):
         self._model = ""
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body["model"] = model_override
         self._model = body.get("model", "")
@@ -66,7 +66,7 @@ class _CountTokensBase(Translator):
     def _adjust(self, body, model):
         return body
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if model_override:
             body["model"] = model_override
         self._model = body.get("model", "")
@@ -178,7 +178,7 @@ class _MultipartAudio(Translator):
             headers={"content-type": f"multipart/form-data; boundary={boundary}"},
         ), model
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         raise NotImplementedError("multipart endpoints use request_multipart")
 
     def response_body(self, status, body):

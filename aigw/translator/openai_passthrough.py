@@ -46,13 +46,21 @@ class _OpenAIPassthrough(Translator):
     def _path(self, model: str) -> str:
         return self.path_prefix + self.PATH
 
-    def request(self, body, *, model_override="", stream=False, force_include_usage=False):
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
         if not isinstance(body, dict):
             raise TranslationError("request body must be a JSON object")
+        self.stream = stream
+        needs_usage = stream and force_include_usage and not (
+            (body.get("stream_options") or {}).get("include_usage")
+        )
+        if raw and not model_override and not needs_usage:
+            # True passthrough: forward the client's bytes untouched
+            # (openai_openai.go passthrough path — no re-serialization).
+            self._model = body.get("model", "")
+            return RequestTranslation(path=self._path(self._model), body=raw)
         model = override_model(body, model_override)
         self._model = model
-        self.stream = stream
-        if stream and force_include_usage:
+        if needs_usage:
             so = body.get("stream_options") or {}
             so["include_usage"] = True
             body["stream_options"] = so
