@@ -52,6 +52,7 @@ JSON_ENDPOINTS = {
     "/v1/embeddings": "embeddings",
     "/v1/images/generations": "image_generation",
     "/v1/responses": "responses",
+    "/v1/responses/input_tokens": "responses_input_tokens",
     "/v1/audio/speech": "speech",
     "/tokenize": "tokenize",
     "/anthropic/v1/messages": "chat",
@@ -121,6 +122,9 @@ class GatewayServer:
         self.max_body_bytes = max_body_bytes
         self._session: Optional[aiohttp.ClientSession] = None
         self._started_at = time.time()
+        # per-backend live stats for the KV-occupancy endpoint picker:
+        # name -> [active_requests, estimated_active_tokens]
+        self._ep_stats: dict[str, list] = {}
 
     # ---- lifecycle -----------------------------------------------------------
 
@@ -346,7 +350,31 @@ class GatewayServer:
         last_error: Optional[str] = None
         attempt = 0
 
-        for backend in backend_attempts(route):
+        attempts_iter = backend_attempts(route)
+        if (
+            route.route.endpoint_picker
+            and self.gpu is not None
+            and route.tiers
+            and len(route.tiers[0]) > 1
+        ):
+            tier0 = route.tiers[0]
+            stats_rows = []
+            for b in tier0:
+                st = self._ep_stats.setdefault(b.name, [0, 0.0])
+                stats_rows.append([st[1], 100000.0, float(st[0]), float(st[0])])
+            pick = await self.gpu.pick_endpoint(
+                stats_rows, float(gpu_input_tokens or 256)
+            )
+            ordered = [tier0[pick]] + [b for i, b in enumerate(tier0) if i != pick]
+
+            def _gen():
+                yield from ordered
+                for tier in route.tiers[1:]:
+                    yield from tier
+
+            attempts_iter = _gen()
+
+        for backend in attempts_iter:
             if attempts_left <= 0:
                 break
             attempts_left -= 1
@@ -415,6 +443,9 @@ class GatewayServer:
                     provider=provider_from_schema(backend.schema.name.value, backend.name),
                     backend=backend.name,
                 )
+            st = self._ep_stats.setdefault(backend.name, [0, 0.0])
+            st[0] += 1
+            st[1] += gpu_input_tokens
             try:
                 timeout = aiohttp.ClientTimeout(total=backend.timeout_s)
                 async with self._session.post(
@@ -448,6 +479,9 @@ class GatewayServer:
                 last_error = f"upstream {backend.name}: {type(e).__name__}: {e}"
                 logger.warning("%s; %d attempts left", last_error, attempts_left)
                 continue
+            finally:
+                st[0] -= 1
+                st[1] -= gpu_input_tokens
 
         self._finish_metrics(
             endpoint, route, None, model, "", Usage(), start, status=503,

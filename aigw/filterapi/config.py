@@ -190,6 +190,11 @@ class Route:
     header_mutation: Optional[HeaderMutation] = None
     # Retry budget across priority tiers (numAttemptsPerPriority analogue).
     retries: int = 1
+    # InferencePool-style endpoint picking: rank same-priority backends by
+    # the on-GPU KV-occupancy scorer instead of weighted random
+    # (replaces the reference's external EPP service —
+    # extensionserver/inferencepool.go:39-54).
+    endpoint_picker: bool = False
 
 
 @dataclass

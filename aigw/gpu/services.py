@@ -172,6 +172,19 @@ class GPUServices:
         loop = asyncio.get_running_loop()
         await loop.run_in_executor(self._executor, self.cache.insert, vec, response)
 
+    async def pick_endpoint(self, stats_rows: list[list[float]], predicted: float) -> int:
+        """One KV-occupancy scoring call for a single request: returns the
+        index of the replica to try first (InferencePool EPP analogue)."""
+        loop = asyncio.get_running_loop()
+
+        def run():
+            stats = torch.tensor(stats_rows, dtype=torch.float32, device=self.device)
+            pred = torch.tensor([predicted], dtype=torch.float32, device=self.device)
+            hip = self.tokenizer.hip
+            return int(hip.kv_score_assign(stats, pred, 1.0, 0.1, 0.05).cpu()[0])
+
+        return await loop.run_in_executor(self._executor, run)
+
     async def assign_replicas(self, stats: torch.Tensor, predicted: torch.Tensor):
         loop = asyncio.get_running_loop()
         hip = self.tokenizer.hip

@@ -172,6 +172,33 @@ class OpenAIToOpenAIResponses(_OpenAIPassthrough):
         return ResponseTranslation(body=chunk, usage=usage, response_model=model)
 
 
+@register("/v1/responses/input_tokens", APISchemaName.OPENAI)
+class OpenAIResponsesInputTokens(_OpenAIPassthrough):
+    """Responses input-token counting endpoint (endpointspec.go tokenize
+    family; OpenAI + Azure variants)."""
+
+    PATH = "/v1/responses/input_tokens"
+
+    def response_body(self, status, body):
+        try:
+            parsed = json.loads(body)
+            tokens = parsed.get("input_tokens", 0) or 0
+        except ValueError:
+            tokens = 0
+        return ResponseTranslation(
+            body=body,
+            usage=Usage(input_tokens=tokens, total_tokens=tokens),
+            end_of_stream=True,
+        )
+
+
+@register("/v1/responses/input_tokens", APISchemaName.AZURE_OPENAI)
+class AzureResponsesInputTokens(OpenAIResponsesInputTokens):
+    def _path(self, model):
+        api_version = self.api_version or "2025-01-01-preview"
+        return f"{self.path_prefix}/openai/responses/input_tokens?api-version={api_version}"
+
+
 @register("/v1/audio/speech", APISchemaName.OPENAI)
 class OpenAIToOpenAISpeech(_OpenAIPassthrough):
     PATH = "/v1/audio/speech"
