@@ -63,9 +63,11 @@ class SemanticCache:
         if self.size == 0:
             return [None] * queries.shape[0]
         view = self.index[: self.size]
-        hi, idx = self.hip.cache_topk(view, queries)
-        hi = hi.cpu().tolist()
-        idx = idx.cpu().tolist()
+        hi, idx = [], []
+        for q0 in range(0, queries.shape[0], 256):  # kernel cap: 256 q/call
+            h, i = self.hip.cache_topk(view, queries[q0 : q0 + 256].contiguous())
+            hi.extend(h.cpu().tolist())
+            idx.extend(i.cpu().tolist())
         out = []
         for h, i in zip(hi, idx):
             score = _unorder(h)

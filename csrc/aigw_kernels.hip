@@ -319,61 +319,69 @@ __device__ __forceinline__ unsigned long long pack_score(float s, unsigned idx) 
 // touches only the L2-resident query block. Global atomics are first folded
 // across the block's 4 waves through LDS (4x fewer device-scope atomicMax
 // on the hot n_q addresses).
-template <int KSTEPS>
+template <int KSTEPS, int ROWTILES>
 __global__ void __launch_bounds__(256)
 cache_topk_kernel_t(const bf16* __restrict__ index, long long n_rows,
                     const bf16* __restrict__ q, int n_q, int dim,
-                    unsigned long long* __restrict__ best /* n_q */) {
-  __shared__ unsigned long long blk_best[16];
-  long long i0 = ((long long)blockIdx.x * 4 + (threadIdx.x >> 6)) * 16;
+                    unsigned long long* __restrict__ best /* n_q <= 256 */) {
+  __shared__ unsigned long long blk_best[256];
+  if (threadIdx.x < (unsigned)n_q) blk_best[threadIdx.x] = 0;
+  __syncthreads();
+  int wave = threadIdx.x >> 6;
   int lane = threadIdx.x & 63;
   int row = lane & 15;
   int kgrp = lane >> 4;
-  bool i_ok = i0 < n_rows && (i0 + row) < n_rows;
-  short8 a_frag[KSTEPS];
-  #pragma unroll
-  for (int ks = 0; ks < KSTEPS; ++ks) {
-    int kk = ks * 32 + kgrp * 8;
-    if (i_ok)
-      a_frag[ks] = *reinterpret_cast<const short8*>(&index[(i0 + row) * dim + kk]);
-    else
-      a_frag[ks] = short8{0, 0, 0, 0, 0, 0, 0, 0};
-  }
-  for (int q0 = 0; q0 < n_q; q0 += 16) {
-    floatx4 acc = {0.f, 0.f, 0.f, 0.f};
-    bool q_in = (q0 + row) < n_q;
+  // Each wave walks ROWTILES 16-row index tiles, keeping each tile in
+  // VGPRs (KSTEPS is compile-time so a_frag never spills — rule #20) and
+  // folding per-query maxima into ONE shared table per block: two
+  // barriers and n_q global atomics per block instead of per wave-tile.
+  for (int t = 0; t < ROWTILES; ++t) {
+    long long i0 = ((long long)blockIdx.x * 4 * ROWTILES + wave * ROWTILES + t) * 16;
+    bool tile_ok = i0 < n_rows;
+    bool i_ok = tile_ok && (i0 + row) < n_rows;
+    short8 a_frag[KSTEPS];
     #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
-      short8 b = {0, 0, 0, 0, 0, 0, 0, 0};
       int kk = ks * 32 + kgrp * 8;
-      if (q_in)
-        b = *reinterpret_cast<const short8*>(&q[(long long)(q0 + row) * dim + kk]);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[ks], b, acc, 0, 0, 0);
+      if (i_ok)
+        a_frag[ks] = *reinterpret_cast<const short8*>(&index[(i0 + row) * dim + kk]);
+      else
+        a_frag[ks] = short8{0, 0, 0, 0, 0, 0, 0, 0};
     }
-    // acc reg r holds C[idx_row = kgrp*4+r][query = row]: fold the 4 regs
-    // locally, then across the 4 lane groups sharing this query column
-    // (lanes differing in bits 4..5), then block-wide via LDS.
-    unsigned long long p = 0;
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      long long irow = i0 + kgrp * 4 + r;
-      float s = (i_ok && irow < n_rows && q_in) ? acc[r] : -1e30f;
-      unsigned long long pk = pack_score(s, (unsigned)(irow & 0xFFFFFFFF));
-      if (pk > p) p = pk;
+    if (!tile_ok) continue;
+    for (int q0 = 0; q0 < n_q; q0 += 16) {
+      floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+      bool q_in = (q0 + row) < n_q;
+      #pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        short8 b = {0, 0, 0, 0, 0, 0, 0, 0};
+        int kk = ks * 32 + kgrp * 8;
+        if (q_in)
+          b = *reinterpret_cast<const short8*>(&q[(long long)(q0 + row) * dim + kk]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[ks], b, acc, 0, 0, 0);
+      }
+      // acc reg r holds C[idx_row = kgrp*4+r][query = row]: fold the 4
+      // regs locally, then across the 4 lane groups sharing this query
+      // column (lanes differing in bits 4..5), then into the block table.
+      unsigned long long p = 0;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long long irow = i0 + kgrp * 4 + r;
+        float s = (irow < n_rows && q_in) ? acc[r] : -1e30f;
+        unsigned long long pk = pack_score(s, (unsigned)(irow & 0xFFFFFFFF));
+        if (pk > p) p = pk;
+      }
+      #pragma unroll
+      for (int off = 16; off < 64; off <<= 1) {
+        unsigned long long o = shfl_xor_u64(p, off);
+        if (o > p) p = o;
+      }
+      if (kgrp == 0 && q_in) atomicMax(&blk_best[q0 + row], p);
     }
-    #pragma unroll
-    for (int off = 16; off < 64; off <<= 1) {
-      unsigned long long o = shfl_xor_u64(p, off);
-      if (o > p) p = o;
-    }
-    if (threadIdx.x < 16) blk_best[threadIdx.x] = 0;
-    __syncthreads();
-    if (kgrp == 0 && q_in) atomicMax(&blk_best[row], p);
-    __syncthreads();
-    if (threadIdx.x < 16 && (q0 + threadIdx.x) < n_q && blk_best[threadIdx.x])
-      atomicMax(&best[q0 + threadIdx.x], blk_best[threadIdx.x]);
-    __syncthreads();
   }
+  __syncthreads();
+  if (threadIdx.x < (unsigned)n_q && blk_best[threadIdx.x])
+    atomicMax(&best[threadIdx.x], blk_best[threadIdx.x]);
 }
 
 // ---------------------------------------------------------------------------
@@ -553,26 +561,32 @@ std::vector<at::Tensor> cache_topk(at::Tensor index, at::Tensor q) {
   auto best = at::zeros({n_q}, at::TensorOptions()
                                    .dtype(at::kLong)
                                    .device(q.device()));
-  long long tiles = (n_rows + 63) / 64;
+  TORCH_CHECK(n_q <= 256, "cache_topk: at most 256 queries per call");
+  constexpr int ROWTILES = 8;
+  long long blocks = (n_rows + 64 * ROWTILES - 1) / (64 * ROWTILES);
   auto* index_p = reinterpret_cast<bf16*>(index.data_ptr<at::BFloat16>());
   auto* q_p = reinterpret_cast<bf16*>(q.data_ptr<at::BFloat16>());
   auto* best_p = reinterpret_cast<unsigned long long*>(best.data_ptr<int64_t>());
   switch (dim >> 5) {
     case 12:  // dim = 384 (bge-small) — the hot path
-      hipLaunchKernelGGL((cache_topk_kernel_t<12>), dim3((unsigned)tiles), dim3(256), 0,
-                         current_stream(), index_p, n_rows, q_p, n_q, dim, best_p);
+      hipLaunchKernelGGL((cache_topk_kernel_t<12, ROWTILES>), dim3((unsigned)blocks),
+                         dim3(256), 0, current_stream(), index_p, n_rows, q_p, n_q,
+                         dim, best_p);
       break;
     case 8:  // dim = 256
-      hipLaunchKernelGGL((cache_topk_kernel_t<8>), dim3((unsigned)tiles), dim3(256), 0,
-                         current_stream(), index_p, n_rows, q_p, n_q, dim, best_p);
+      hipLaunchKernelGGL((cache_topk_kernel_t<8, ROWTILES>), dim3((unsigned)blocks),
+                         dim3(256), 0, current_stream(), index_p, n_rows, q_p, n_q,
+                         dim, best_p);
       break;
     case 16:  // dim = 512
-      hipLaunchKernelGGL((cache_topk_kernel_t<16>), dim3((unsigned)tiles), dim3(256), 0,
-                         current_stream(), index_p, n_rows, q_p, n_q, dim, best_p);
+      hipLaunchKernelGGL((cache_topk_kernel_t<16, ROWTILES>), dim3((unsigned)blocks),
+                         dim3(256), 0, current_stream(), index_p, n_rows, q_p, n_q,
+                         dim, best_p);
       break;
     case 4:  // dim = 128
-      hipLaunchKernelGGL((cache_topk_kernel_t<4>), dim3((unsigned)tiles), dim3(256), 0,
-                         current_stream(), index_p, n_rows, q_p, n_q, dim, best_p);
+      hipLaunchKernelGGL((cache_topk_kernel_t<4, ROWTILES>), dim3((unsigned)blocks),
+                         dim3(256), 0, current_stream(), index_p, n_rows, q_p, n_q,
+                         dim, best_p);
       break;
     default:
       TORCH_CHECK(false, "cache_topk: unsupported dim ", dim,
