@@ -43,6 +43,11 @@ from aigw.mutator import apply_body_mutation, apply_header_mutation
 from aigw.ratelimit import RateLimiter
 from aigw.translator import TranslationError, Usage, get_translator
 
+try:  # C++ hot-path helpers (csrc/aigw_native.cpp); built in-tree
+    import aigw_native as _native
+except ImportError:  # pragma: no cover - built by setup.py everywhere
+    _native = None
+
 logger = logging.getLogger("aigw.server")
 
 # endpoint key -> (operation name for metrics)
@@ -88,6 +93,12 @@ _HOP_BY_HOP = frozenset(
         internalapi.ORIGINAL_PATH_HEADER,
     }
 )
+
+
+# Schemas whose request translation is a byte-level passthrough for OpenAI
+# clients (body unchanged modulo model override) — eligible for the
+# never-parse fast path.
+_PASSTHROUGH_SCHEMAS = frozenset({APISchemaName.OPENAI, APISchemaName.AZURE_OPENAI})
 
 
 def _json_error(status: int, message: str, err_type: str = "invalid_request_error") -> web.Response:
@@ -259,21 +270,30 @@ class GatewayServer:
     async def _process(self, request: web.Request, endpoint: str) -> web.StreamResponse:
         start = time.monotonic()
         rt = self.runtime
-        try:
-            raw = await request.read()
-            body = json.loads(raw) if raw else {}
-            if not isinstance(body, dict):
-                raise ValueError("body must be a JSON object")
-        except ValueError as e:
-            return _json_error(400, f"invalid request body: {e}")
-        if not raw:
-            raw = b"{}"
+        raw = await request.read() or b"{}"
+        # Fast path: one C++ pass extracts model/stream/chat-text without
+        # building Python objects; full json.loads happens lazily only when
+        # a translator or mutation needs the dict.
+        body: Optional[dict] = None
+        chat_text = b""
+        if _native is not None:
+            ok, model, stream_flag, chat_text = _native.scan_chat_body(raw)
+            if not ok:
+                return _json_error(400, "invalid request body: malformed JSON")
+        if _native is None or body is not None:
+            try:
+                body = json.loads(raw)
+                if not isinstance(body, dict):
+                    raise ValueError("body must be a JSON object")
+            except ValueError as e:
+                return _json_error(400, f"invalid request body: {e}")
+            model = str(body.get("model", ""))
+            stream_flag = bool(body.get("stream"))
 
         headers = self._ingress_headers(request)
-        model = str(body.get("model", ""))
         headers[rt.model_header] = model
         headers[internalapi.ORIGINAL_PATH_HEADER] = request.path
-        stream = bool(body.get("stream")) and endpoint in STREAMABLE
+        stream = stream_flag and endpoint in STREAMABLE
 
         route = rt.select_route(headers)
         if route is None:
@@ -290,7 +310,13 @@ class GatewayServer:
         # GPU-side input token count (admission accounting / local usage).
         gpu_input_tokens = 0
         if self.gpu is not None and endpoint in ("/v1/chat/completions", "/anthropic/v1/messages"):
-            gpu_input_tokens = await self.gpu.count_request_tokens(body)
+            if not chat_text:
+                if body is None:
+                    body = json.loads(raw)
+                from aigw.gpu.services import extract_chat_text
+
+                chat_text = extract_chat_text(body)
+            gpu_input_tokens = await self.gpu.count_text_tokens(chat_text)
 
         # semantic response cache (GPU MFMA embed + HBM index)
         cache_key_vec = None
@@ -300,7 +326,7 @@ class GatewayServer:
             and endpoint == "/v1/chat/completions"
             and not stream
         ):
-            hit, cache_key_vec = await self.gpu.cache_lookup(body)
+            hit, cache_key_vec = await self.gpu.cache_lookup_text(chat_text or b" ")
             if hit is not None:
                 self.metrics.cache_events.labels(event="hit").inc()
                 resp = web.Response(body=hit, content_type="application/json")
@@ -394,9 +420,25 @@ class GatewayServer:
             # First attempt works on the live parsed body (translators may
             # mutate it); every RETRY re-parses the ORIGINAL raw bytes, which
             # is the A.8 "restore original body per try" semantics without a
-            # defensive deep copy on the hot path.
-            body_copy = body if first else json.loads(raw)
+            # defensive deep copy on the hot path. When the C++ scanner
+            # already yielded model/stream and this backend is a byte-level
+            # passthrough, the dict is never materialized at all.
             override = backend.model_name_override or route.route.model_name_override
+            if first and body is None:
+                if (
+                    backend.schema.name in _PASSTHROUGH_SCHEMAS
+                    and backend.body_mutation is None
+                    and not override
+                    and self.tracer is None
+                    and not (stream and route.costs)
+                ):
+                    body_copy = {"model": model, "stream": stream}
+                else:
+                    body = body_copy = json.loads(raw)
+            elif first:
+                body_copy = body
+            else:
+                body_copy = json.loads(raw)
             try:
                 tr = translator.request(
                     body_copy,

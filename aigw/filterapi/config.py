@@ -456,6 +456,18 @@ def dump_config_yaml(cfg: Config) -> str:
     return yaml.safe_dump(d, sort_keys=False)
 
 
+def _expand_env(text: str) -> str:
+    """``${VAR}`` substitution from the environment (the reference's
+    substitution.aigw.run/env annotations, cmd/aigw/run.go:52-55). Unset
+    variables are left verbatim so configs fail loudly downstream."""
+    import os
+    import re as _re
+
+    return _re.sub(
+        r"\$\{(\w+)\}", lambda m: os.environ.get(m.group(1), m.group(0)), text
+    )
+
+
 def load_config_file(path: str) -> Config:
     with open(path, "r", encoding="utf-8") as f:
-        return load_config(yaml.safe_load(f))
+        return load_config(yaml.safe_load(_expand_env(f.read())))

@@ -140,6 +140,26 @@ class GPUServices:
         count, _ = await self._submit(extract_chat_text(body), want_vec=False)
         return count
 
+    async def count_text_tokens(self, text: bytes) -> int:
+        count, _ = await self._submit(text or b" ", want_vec=False)
+        return count
+
+    async def cache_lookup_text(self, text: bytes):
+        """Like cache_lookup but takes pre-extracted text (from the C++
+        scanner) instead of a parsed body."""
+        _, vec = await self._submit(text or b" ", want_vec=True)
+        if vec is None:
+            return None, None
+        loop = asyncio.get_running_loop()
+
+        def run():
+            hits = self.cache.lookup(vec.unsqueeze(0))
+            if hits[0] is None:
+                return None
+            return self.cache.get(hits[0][0])
+
+        return await loop.run_in_executor(self._executor, run), vec
+
     async def tokenize(self, text) -> list[int]:
         if isinstance(text, str):
             text = text.encode("utf-8", "replace")

@@ -10,6 +10,8 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
+import pybind11  # noqa: E402
+from setuptools import Extension  # noqa: E402
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
 
 setup(
@@ -24,7 +26,14 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
             },
-        )
+        ),
+        Extension(
+            name="aigw_native",
+            sources=["csrc/aigw_native.cpp"],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17"],
+            language="c++",
+        ),
     ],
     cmdclass={"build_ext": BuildExtension},
 )

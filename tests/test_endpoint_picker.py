@@ -23,6 +23,9 @@ class FakeGPU:
     async def count_request_tokens(self, body):
         return 100
 
+    async def count_text_tokens(self, text):
+        return 100
+
     async def pick_endpoint(self, stats_rows, predicted):
         self.seen_stats.append((stats_rows, predicted))
         return self.pick
