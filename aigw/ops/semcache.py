@@ -50,6 +50,9 @@ class SemanticCache:
         self.size = 0
         self.head = 0
         self.values: dict[int, bytes] = {}  # slot -> cached response body
+        # inserts not yet all-gathered to the other shards (aigw.parallel)
+        self.pending: list[tuple[torch.Tensor, bytes]] = []
+        self.record_pending = False
 
     def embed(self, out_ids: torch.Tensor, req_off: torch.Tensor) -> torch.Tensor:
         """(B, dim) bf16 L2-normalized query vectors from tokenizer output."""
@@ -83,7 +86,27 @@ class SemanticCache:
         self.values[slot] = response
         self.head = (self.head + 1) % self.capacity
         self.size = min(self.size + 1, self.capacity)
+        if self.record_pending:
+            self.pending.append((query_vec.detach(), response))
         return slot
+
+    def insert_remote(self, query_vec: torch.Tensor, response: bytes) -> int:
+        """Insert a row replicated from another shard (never re-pended)."""
+        slot = self.head
+        self.index[slot] = query_vec.to(self.index.dtype).to(self.device)
+        self.values[slot] = response
+        self.head = (self.head + 1) % self.capacity
+        self.size = min(self.size + 1, self.capacity)
+        return slot
+
+    def drain_pending(self, max_n: int = 32):
+        """Take up to max_n locally inserted rows for the next all-gather
+        tick: returns (vecs [k, dim] on device, [response bytes])."""
+        batch, self.pending = self.pending[:max_n], self.pending[max_n:]
+        if not batch:
+            return torch.zeros(0, self.dim, dtype=torch.bfloat16, device=self.device), []
+        vecs = torch.stack([v for v, _ in batch]).to(torch.bfloat16)
+        return vecs, [r for _, r in batch]
 
 
 def _unorder(u: int) -> float:

@@ -77,6 +77,8 @@ class StateSync:
             val = torch.tensor(list(own.values()), dtype=torch.int64)
             self._buf[idx.to(self.device)] = val.to(self.device)
         dist.all_reduce(self._buf, op=dist.ReduceOp.SUM, group=self.group)
+        if self.cache is not None:
+            self._sync_cache()
         nonzero = torch.nonzero(self._buf, as_tuple=False).flatten()
         if nonzero.numel():
             host = self._buf[nonzero].cpu().tolist()
@@ -94,6 +96,59 @@ class StateSync:
                 else:
                     self.limiter.apply_remote_slot(s, remote)
         self.ticks += 1
+
+    # ---- semantic-cache index replication ------------------------------------
+
+    MAX_CACHE_ROWS_PER_TICK = 32
+
+    def _sync_cache(self) -> None:
+        """All-gather newly inserted cache rows (embedding vector + cached
+        response body) so every shard serves every shard's hits
+        (RCCL all-gather of semantic-cache index updates).
+        Collective sequence is fixed per tick, so shards stay aligned:
+        sizes all-gather, then padded vec + payload all-gathers."""
+        cache = self.cache
+        rank = dist.get_rank(self.group)
+        world = dist.get_world_size(self.group)
+        vecs, values = cache.drain_pending(self.MAX_CACHE_ROWS_PER_TICK)
+        payload = bytearray()
+        for v in values:
+            payload.extend(len(v).to_bytes(4, "little"))
+            payload.extend(v)
+        sizes = torch.tensor([vecs.shape[0], len(payload)], dtype=torch.int64,
+                             device=self.device)
+        all_sizes = [torch.zeros_like(sizes) for _ in range(world)]
+        dist.all_gather(all_sizes, sizes, group=self.group)
+        max_k = max(int(s[0]) for s in all_sizes)
+        max_p = max(int(s[1]) for s in all_sizes)
+        if max_k == 0:
+            return
+        dim = vecs.shape[1] if vecs.numel() else cache.dim
+        vpad = torch.zeros(max_k, dim, dtype=torch.bfloat16, device=self.device)
+        if vecs.shape[0]:
+            vpad[: vecs.shape[0]] = vecs
+        ppad = torch.zeros(max(max_p, 1), dtype=torch.uint8, device=self.device)
+        if payload:
+            ppad[: len(payload)] = torch.frombuffer(bytes(payload), dtype=torch.uint8).to(
+                self.device
+            )
+        all_vecs = [torch.zeros_like(vpad) for _ in range(world)]
+        all_payload = [torch.zeros_like(ppad) for _ in range(world)]
+        dist.all_gather(all_vecs, vpad, group=self.group)
+        dist.all_gather(all_payload, ppad, group=self.group)
+        for r in range(world):
+            k = int(all_sizes[r][0])
+            if r == rank or k == 0:
+                continue
+            blob = bytes(all_payload[r][: int(all_sizes[r][1])].cpu().numpy().tobytes())
+            off = 0
+            rows = all_vecs[r]
+            for i in range(k):
+                ln = int.from_bytes(blob[off : off + 4], "little")
+                off += 4
+                value = blob[off : off + ln]
+                off += ln
+                cache.insert_remote(rows[i], value)
 
     # ---- background loop -----------------------------------------------------
 
