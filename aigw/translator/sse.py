@@ -33,7 +33,7 @@ class SSEEvent:
 
 
 @dataclass
-class SSEDecoder:
+class PySSEDecoder:
     _buf: bytearray = field(default_factory=bytearray)
     _data_lines: list = field(default_factory=list)
     _event: str = ""
@@ -94,6 +94,37 @@ class SSEDecoder:
         self._event = ""
         self._id = ""
         return ev
+
+
+class NativeSSEDecoder:
+    """Same interface as PySSEDecoder, backed by the C++ incremental
+    splitter (csrc/aigw_native.cpp SSEFeed)."""
+
+    __slots__ = ("_feed",)
+
+    def __init__(self):
+        self._feed = _native.SSEFeed()
+
+    def feed(self, chunk: bytes) -> list[SSEEvent]:
+        return [
+            SSEEvent(data=data.decode("utf-8", "replace"), event=event)
+            for event, data in self._feed.feed(chunk)
+        ]
+
+    def flush(self) -> list[SSEEvent]:
+        return [
+            SSEEvent(data=data.decode("utf-8", "replace"), event=event)
+            for event, data in self._feed.flush()
+        ]
+
+
+try:
+    import aigw_native as _native
+
+    SSEDecoder = NativeSSEDecoder
+except ImportError:  # pragma: no cover - extension built in-tree everywhere
+    _native = None
+    SSEDecoder = PySSEDecoder
 
 
 def encode_data(data: str, event: str = "") -> bytes:

@@ -11,7 +11,10 @@ from aigw.translator.eventstream import (
     encode_event,
     encode_message,
 )
-from aigw.translator.sse import SSEDecoder, SSEEvent
+from aigw.translator.sse import PySSEDecoder, SSEDecoder, SSEEvent
+
+# exercise BOTH the native (C++) and pure-Python SSE decoders
+DECODERS = [SSEDecoder, PySSEDecoder] if SSEDecoder is not PySSEDecoder else [PySSEDecoder]
 
 
 STREAM = (
@@ -35,14 +38,16 @@ def _expected_events():
     ]
 
 
-def test_sse_whole():
-    d = SSEDecoder()
+@pytest.mark.parametrize("cls", DECODERS)
+def test_sse_whole(cls):
+    d = cls()
     assert d.feed(STREAM) + d.flush() == _expected_events()
 
 
+@pytest.mark.parametrize("cls", DECODERS)
 @pytest.mark.parametrize("n", [1, 2, 3, 5, 7, 11, 64])
-def test_sse_any_chunking(n):
-    d = SSEDecoder()
+def test_sse_any_chunking(cls, n):
+    d = cls()
     events = []
     for i in range(0, len(STREAM), n):
         events.extend(d.feed(STREAM[i : i + n]))
@@ -50,20 +55,23 @@ def test_sse_any_chunking(n):
     assert events == _expected_events()
 
 
-def test_sse_crlf():
-    d = SSEDecoder()
+@pytest.mark.parametrize("cls", DECODERS)
+def test_sse_crlf(cls):
+    d = cls()
     evs = d.feed(b"data: x\r\n\r\n")
     assert evs == [SSEEvent(data="x")]
 
 
-def test_sse_encode_roundtrip():
+@pytest.mark.parametrize("cls", DECODERS)
+def test_sse_encode_roundtrip(cls):
     ev = SSEEvent(data='{"x":1}\nline2', event="delta")
-    d = SSEDecoder()
+    d = cls()
     assert d.feed(ev.encode()) == [ev]
 
 
-def test_sse_flush_unterminated():
-    d = SSEDecoder()
+@pytest.mark.parametrize("cls", DECODERS)
+def test_sse_flush_unterminated(cls):
+    d = cls()
     assert d.feed(b"data: partial") == []
     assert d.flush() == [SSEEvent(data="partial")]
 
