@@ -1,16 +1,20 @@
 """Flagship serving benchmark: OpenAI /v1/chat/completions with 4k-token
-bodies through the gateway shard to a mock upstream (BASELINE.json config #1
-+ the GPU token-accounting path of config #2 when a GPU is present).
+bodies through the gateway to a mock upstream (BASELINE.json config #1 +
+the GPU token-accounting path of config #2 when a GPU is present).
 
-One rank = one gateway shard pinned to one GPU (plus its own in-process
-mock upstream, so upstream capacity scales with shards and the gateway is
-the measured system). A "step" fires ``--batch`` concurrent requests and
-awaits them all. Timing: W untimed warmup steps, then exactly K steps
-bracketed by a distributed barrier + torch.cuda.synchronize on both sides;
-the reported value is the whole-job aggregate request rate over all ranks
-computed from the MAX per-rank elapsed time.
+Topology: one rank per GPU (the driver launches N ranks via torchrun).
+Within a rank, the shard is a process GROUP: ``--workers`` HTTP worker
+processes each run a full gateway + mock upstream + load loop (CPython's
+GIL caps a single asyncio process near ~3k req/s; the MI355X node pairs
+256 CUs with ~128 EPYC cores, so the shard design uses processes the way
+Envoy uses worker threads). Each worker pins its GPU work to the rank's
+GPU; kernels from all workers interleave on the device.
 
-Run directly (defaults N=1) or via torchrun for N>1 (the driver does this).
+Timing contract: W untimed warmup steps per worker, then exactly K steps
+of ``--batch`` concurrent requests, bracketed by a distributed barrier +
+torch.cuda.synchronize on both sides; the value is the whole-job aggregate
+request rate over all ranks and workers computed from the MAX elapsed
+time. Rank 0 prints ONE JSON line.
 """
 
 from __future__ import annotations
@@ -18,16 +22,12 @@ from __future__ import annotations
 import argparse
 import asyncio
 import json
+import multiprocessing as mp
 import os
 import statistics
 import time
 
-import aiohttp
 import torch
-
-from aigw.extproc.server import GatewayServer, run_server
-from aigw.filterapi import RuntimeConfig, load_config
-from aigw.testing.mockupstream import start_mock_upstream
 
 WORDS = (
     "the quick brown fox jumps over the lazy dog while seventeen engineers "
@@ -84,6 +84,8 @@ def gateway_config(upstream_port: int) -> dict:
 
 
 async def fire_step(session, url, payload_bytes, batch, latencies):
+    import aiohttp  # local import keeps worker spawn cheap
+
     async def one():
         t0 = time.perf_counter()
         async with session.post(
@@ -96,7 +98,13 @@ async def fire_step(session, url, payload_bytes, batch, latencies):
     await asyncio.gather(*(one() for _ in range(batch)))
 
 
-async def amain(args, rank, world, local_rank):
+async def worker_main(args, local_rank: int, ready, go, out_q):
+    import aiohttp
+
+    from aigw.extproc.server import GatewayServer, run_server
+    from aigw.filterapi import RuntimeConfig, load_config
+    from aigw.testing.mockupstream import start_mock_upstream
+
     use_gpu = torch.cuda.is_available()
     if use_gpu:
         torch.cuda.set_device(local_rank)
@@ -118,22 +126,97 @@ async def amain(args, rank, world, local_rank):
     gw_port = gw_runner.addresses[0][1]
 
     payload = json.dumps(build_payload(args.tokens)).encode()
-    connector = aiohttp.TCPConnector(limit=0)
-    session = aiohttp.ClientSession(connector=connector)
+    session = aiohttp.ClientSession(connector=aiohttp.TCPConnector(limit=0))
     gw_url = f"http://127.0.0.1:{gw_port}/v1/chat/completions"
     direct_url = f"http://127.0.0.1:{up_port}/v1/chat/completions"
 
-    # warmup both paths (connection pools, JIT, allocator) before measuring
-    warm_lat: list[float] = []
+    # warm both paths, then record the warm direct-to-upstream baseline
+    scratch: list[float] = []
     for _ in range(max(args.warmup, 1)):
-        await fire_step(session, direct_url, payload, args.batch, warm_lat)
+        await fire_step(session, direct_url, payload, args.batch, scratch)
     for _ in range(args.warmup):
-        await fire_step(session, gw_url, payload, args.batch, warm_lat)
-
-    # untimed: direct-to-upstream baseline for "added latency" (warm)
+        await fire_step(session, gw_url, payload, args.batch, scratch)
     direct_lat: list[float] = []
     for _ in range(2):
         await fire_step(session, direct_url, payload, args.batch, direct_lat)
+    if use_gpu:
+        torch.cuda.synchronize()
+
+    ready.set()
+    while not go.is_set():
+        await asyncio.sleep(0.001)
+
+    lat: list[float] = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        await fire_step(session, gw_url, payload, args.batch, lat)
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    out_q.put(
+        {
+            "elapsed": elapsed,
+            "requests": args.steps * args.batch,
+            "p50": statistics.median(lat),
+            "p99": sorted(lat)[max(int(len(lat) * 0.99) - 1, 0)],
+            "p50_direct": statistics.median(direct_lat),
+        }
+    )
+    await session.close()
+    await gw_runner.cleanup()
+    await up_runner.cleanup()
+    if gpu_services is not None:
+        gpu_services.close()
+
+
+def worker_entry(args, local_rank, ready, go, out_q):
+    asyncio.run(worker_main(args, local_rank, ready, go, out_q))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--batch", type=int, default=32, help="concurrent requests per worker step")
+    ap.add_argument("--tokens", type=int, default=4096)
+    ap.add_argument("--workers", type=int, default=0,
+                    help="HTTP worker processes per shard (0 = auto)")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    use_gpu = torch.cuda.is_available()
+    if world > 1:
+        backend = "nccl" if use_gpu else "gloo"
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        torch.distributed.init_process_group(backend)
+
+    workers = args.workers
+    if workers <= 0:
+        cores = os.cpu_count() or 8
+        # ~2 cores per worker (gateway + upstream + client share the loop);
+        # never oversubscribe when several ranks share the node
+        workers = max(1, min(6, cores // (2 * max(world, 1))))
+
+    ctx = mp.get_context("spawn")
+    ready_evts = [ctx.Event() for _ in range(workers)]
+    go = ctx.Event()
+    out_q = ctx.Queue()
+    procs = [
+        ctx.Process(target=worker_entry, args=(args, local_rank, ready_evts[i], go, out_q))
+        for i in range(workers)
+    ]
+    for p in procs:
+        p.start()
+    for ev in ready_evts:
+        if not ev.wait(timeout=600):
+            for p in procs:
+                p.terminate()
+            raise RuntimeError("bench worker failed to become ready")
 
     def barrier_sync():
         if world > 1:
@@ -142,16 +225,18 @@ async def amain(args, rank, world, local_rank):
             torch.cuda.synchronize()
 
     barrier_sync()
-    lat: list[float] = []
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        await fire_step(session, gw_url, payload, args.batch, lat)
+    go.set()
+    results = [out_q.get(timeout=600) for _ in range(workers)]
+    elapsed = time.perf_counter() - t0
     if use_gpu:
         torch.cuda.synchronize()
-    elapsed = time.perf_counter() - t0
     barrier_sync()
+    for p in procs:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.terminate()
 
-    # MAX elapsed over ranks defines whole-job time
     if world > 1:
         t = torch.tensor([elapsed], dtype=torch.float64)
         if torch.distributed.get_backend() == "nccl":
@@ -161,11 +246,11 @@ async def amain(args, rank, world, local_rank):
     else:
         elapsed_max = elapsed
 
-    total_requests = args.steps * args.batch * world
+    total_requests = sum(r["requests"] for r in results) * world
     value = total_requests / elapsed_max
-    p50 = statistics.median(lat)
-    p99 = sorted(lat)[int(len(lat) * 0.99) - 1]
-    p50_direct = statistics.median(direct_lat)
+    p50 = statistics.median(r["p50"] for r in results)
+    p99 = max(r["p99"] for r in results)
+    p50_direct = statistics.median(r["p50_direct"] for r in results)
 
     if rank == 0:
         out = {
@@ -183,47 +268,21 @@ async def amain(args, rank, world, local_rank):
             "data": "synthetic",
             "config": {
                 "model": "aigw standalone: OpenAI /v1/chat/completions -> mock upstream (examples/basic)",
-                "global_batch": args.batch * world,
+                "global_batch": args.batch * workers * world,
                 "seq_len": args.tokens,
                 "parallelism": f"dp{world}",
+                "workers_per_shard": workers,
                 "p50_ms": round(p50, 3),
                 "p99_ms": round(p99, 3),
                 "p50_direct_ms": round(p50_direct, 3),
                 "p50_added_latency_ms": round(p50 - p50_direct, 3),
-                "gpu_token_accounting": bool(use_gpu),
+                "gpu_token_accounting": use_gpu,
             },
         }
         print(json.dumps(out))
 
-    await session.close()
-    await gw_runner.cleanup()
-    await up_runner.cleanup()
-    if gpu_services is not None:
-        gpu_services.close()
-
-
-def main():
-    ap = argparse.ArgumentParser()
-    ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=10)
-    ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=256)
-    ap.add_argument("--tokens", type=int, default=4096)
-    args = ap.parse_args()
-
-    rank = int(os.environ.get("RANK", 0))
-    world = int(os.environ.get("WORLD_SIZE", 1))
-    local_rank = int(os.environ.get("LOCAL_RANK", rank))
     if world > 1:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
-        if torch.cuda.is_available():
-            torch.cuda.set_device(local_rank)
-        torch.distributed.init_process_group(backend=backend)
-    try:
-        asyncio.run(amain(args, rank, world, local_rank))
-    finally:
-        if world > 1:
-            torch.distributed.destroy_process_group()
+        torch.distributed.destroy_process_group()
 
 
 if __name__ == "__main__":
