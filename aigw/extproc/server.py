@@ -167,6 +167,7 @@ class GatewayServer:
         app.router.add_get("/anthropic/v1/models", self._handle_anthropic_models)
         app.router.add_get("/health", self._handle_health)
         app.router.add_get("/metrics", self._handle_metrics)
+        app.router.add_get("/debug/tasks", self._handle_debug_tasks)
         if self.gpu is not None:
             app.router.add_post("/v1/gateway/tokenize", self._handle_gpu_tokenize)
         mcp_cfg = self.runtime.config.mcp
@@ -196,6 +197,32 @@ class GatewayServer:
 
     async def _handle_metrics(self, request: web.Request) -> web.Response:
         return web.Response(body=self.metrics.render(), content_type="text/plain")
+
+    async def _handle_debug_tasks(self, request: web.Request) -> web.Response:
+        """Live asyncio task dump — the profiling surface the reference gets
+        from its always-on pprof server (internal/pprof/pprof.go:18-50).
+        Loopback-only, disabled with AIGW_DISABLE_DEBUG (DISABLE_PPROF
+        analogue)."""
+        import os as _os
+
+        if _os.environ.get("AIGW_DISABLE_DEBUG") or request.remote not in (
+            "127.0.0.1",
+            "::1",
+        ):
+            return web.Response(status=403)
+        tasks = []
+        for t in asyncio.all_tasks():
+            frame = t.get_stack(limit=1)
+            tasks.append(
+                {
+                    "name": t.get_name(),
+                    "done": t.done(),
+                    "where": f"{frame[0].f_code.co_filename}:{frame[0].f_lineno}"
+                    if frame
+                    else "",
+                }
+            )
+        return web.json_response({"tasks": tasks, "count": len(tasks)})
 
     async def _handle_models(self, request: web.Request) -> web.Response:
         rt = self.runtime
