@@ -198,9 +198,11 @@ def main():
     workers = args.workers
     if workers <= 0:
         cores = os.cpu_count() or 8
-        # ~2 cores per worker (gateway + upstream + client share the loop);
-        # never oversubscribe when several ranks share the node
-        workers = max(1, min(6, cores // (2 * max(world, 1))))
+        # measured knee on the 256-core MI355X box: throughput peaks at
+        # ~12-16 workers/shard (14.1k req/s @16) and collapses by 32
+        # (profiles/r01 bench_wsweep); ~3 cores per worker keeps headroom
+        # when 8 ranks share the node
+        workers = max(1, min(12, cores // (3 * max(world, 1))))
 
     ctx = mp.get_context("spawn")
     ready_evts = [ctx.Event() for _ in range(workers)]
