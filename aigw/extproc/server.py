@@ -95,6 +95,22 @@ _HOP_BY_HOP = frozenset(
 )
 
 
+# Endpoints that must carry a model in the body (endpointspec ParseBody);
+# /tokenize and audio endpoints resolve the model elsewhere.
+_MODEL_REQUIRED = frozenset(
+    {
+        "/v1/chat/completions",
+        "/v1/completions",
+        "/v1/embeddings",
+        "/v1/images/generations",
+        "/v1/responses",
+        "/anthropic/v1/messages",
+        "/anthropic/v1/messages/count_tokens",
+        "/v2/rerank",
+        "/v1/audio/speech",
+    }
+)
+
 # Schemas whose request translation is a byte-level passthrough for OpenAI
 # clients (body unchanged modulo model override) — eligible for the
 # never-parse fast path.
@@ -316,6 +332,15 @@ class GatewayServer:
                 return _json_error(400, f"invalid request body: {e}")
             model = str(body.get("model", ""))
             stream_flag = bool(body.get("stream"))
+
+        if not model and endpoint in _MODEL_REQUIRED:
+            # parse failure -> 400, matching the reference's ParseBody error
+            # (processor_impl.go:248-252), not a 404 route miss
+            return _json_error(400, "missing required field 'model'")
+        if body is not None and endpoint in ("/v1/chat/completions",) and not isinstance(
+            body.get("messages", []), list
+        ):
+            return _json_error(400, "'messages' must be an array")
 
         headers = self._ingress_headers(request)
         headers[rt.model_header] = model
