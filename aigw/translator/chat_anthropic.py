@@ -388,6 +388,77 @@ class AnthropicToOpenAIChat(Translator):
         )
 
 
+@register("/anthropic/v1/messages", APISchemaName.AWS_BEDROCK)
+class AnthropicToBedrockConverse(Translator):
+    """Anthropic client → Bedrock Converse (anthropic_awsbedrock.go):
+    composed from the shared conversions — Anthropic→OpenAI request,
+    OpenAI→Converse request; Converse→OpenAI response→Anthropic response;
+    for streams, the Bedrock event-stream→OpenAI-chunk machine feeds the
+    OpenAI→Anthropic SSE synthesizer."""
+
+    def __init__(self, **kw):
+        from aigw.translator.anthropic_schema import OpenAIToAnthropicStream
+        from aigw.translator.chat_bedrock import OpenAIToBedrockChat
+
+        self._bedrock = OpenAIToBedrockChat()
+        self._machine = OpenAIToAnthropicStream()
+        self._sse = SSEDecoder()
+        self._model = ""
+        self.stream = False
+
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
+        if model_override:
+            body["model"] = model_override
+        self._model = body.get("model", "")
+        self.stream = stream
+        oreq = anthropic_to_openai_request(body)
+        if stream:
+            oreq["stream"] = True
+        return self._bedrock.request(oreq, stream=stream)
+
+    def response_headers(self, status, headers):
+        if self.stream:
+            return {"content-type": "text/event-stream"}
+        return {}
+
+    def response_body(self, status, body):
+        from aigw.translator.chat_bedrock import converse_to_openai_response
+
+        resp = json.loads(body)
+        oresp, _usage = converse_to_openai_response(resp, self._model)
+        aresp, usage = openai_to_anthropic_response(oresp)
+        return ResponseTranslation(
+            body=jdump(aresp), usage=usage, response_model=self._model, end_of_stream=True
+        )
+
+    def response_chunk(self, chunk):
+        from aigw.translator.anthropic_schema import encode_anthropic_events
+
+        # Bedrock event-stream -> OpenAI SSE chunks (existing machine) ...
+        inner = self._bedrock.response_chunk(chunk)
+        events = []
+        done = False
+        for ev in self._sse.feed(inner.body):
+            if not ev.data:
+                continue
+            if ev.data == "[DONE]":
+                events.extend(self._machine.finish())
+                done = True
+                continue
+            try:
+                data = json.loads(ev.data)
+            except ValueError:
+                continue
+            events.extend(self._machine.feed_chunk(data))
+        usage = self._machine.usage if done else None
+        return ResponseTranslation(
+            body=encode_anthropic_events(events),
+            usage=usage,
+            response_model=self._model,
+            end_of_stream=done,
+        )
+
+
 # --- Anthropic request -> OpenAI request (reverse direction) ------------------
 
 

@@ -104,6 +104,101 @@ class GCPAnthropicCountTokens(_CountTokensBase):
         return body
 
 
+@register("/anthropic/v1/messages/count_tokens", APISchemaName.AWS_ANTHROPIC)
+class AWSAnthropicCountTokens(_CountTokensBase):
+    """count_tokens against Bedrock-hosted Anthropic
+    (counttokens_awsanthropic.go: count-tokens invoke path)."""
+
+    def _path(self, model):
+        return f"/model/{model}/count-tokens"
+
+    def _adjust(self, body, model):
+        body.pop("model", None)
+        body["anthropic_version"] = "bedrock-2023-05-31"
+        return body
+
+
+@register("/tokenize", APISchemaName.GCP_VERTEX_AI)
+class GCPVertexTokenize(Translator):
+    """/tokenize → Vertex :countTokens (tokenize_gcp.go)."""
+
+    def __init__(self, api_version: str = "", gcp_project: str = "", gcp_region: str = ""):
+        self.gcp_project = gcp_project
+        self.gcp_region = gcp_region
+        self._model = ""
+
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
+        if model_override:
+            body["model"] = model_override
+        self._model = body.get("model", "")
+        text = body.get("prompt") or body.get("text") or ""
+        greq = {"contents": [{"role": "user", "parts": [{"text": text}]}]}
+        path = (
+            f"/v1/projects/{self.gcp_project}/locations/{self.gcp_region}"
+            f"/publishers/google/models/{self._model}:countTokens"
+        )
+        return RequestTranslation(path=path, body=jdump(greq))
+
+    def response_body(self, status, body):
+        try:
+            resp = json.loads(body)
+            n = resp.get("totalTokens", 0) or 0
+        except ValueError:
+            n = 0
+        return ResponseTranslation(
+            body=jdump({"count": n, "tokens": []}),
+            usage=Usage(input_tokens=n, total_tokens=n),
+            response_model=self._model,
+            end_of_stream=True,
+        )
+
+
+@register("/tokenize", APISchemaName.ANTHROPIC)
+@register("/tokenize", APISchemaName.GCP_ANTHROPIC)
+class AnthropicTokenize(Translator):
+    """/tokenize → Anthropic count_tokens (tokenize path for Anthropic
+    backends, tokenize.go family)."""
+
+    def __init__(self, api_version: str = "", gcp_project: str = "", gcp_region: str = ""):
+        self.gcp_project = gcp_project
+        self.gcp_region = gcp_region
+        self.is_gcp = bool(gcp_project)
+        self._model = ""
+
+    def request(self, body, *, model_override="", stream=False, force_include_usage=False, raw=b""):
+        if model_override:
+            body["model"] = model_override
+        self._model = body.get("model", "")
+        text = body.get("prompt") or body.get("text") or ""
+        areq = {
+            "model": self._model,
+            "messages": [{"role": "user", "content": text}],
+        }
+        if self.is_gcp:
+            areq.pop("model", None)
+            areq["anthropic_version"] = "vertex-2023-10-16"
+            path = (
+                f"/v1/projects/{self.gcp_project}/locations/{self.gcp_region}"
+                f"/publishers/anthropic/models/count-tokens:rawPredict"
+            )
+        else:
+            path = "/v1/messages/count_tokens"
+        return RequestTranslation(path=path, body=jdump(areq))
+
+    def response_body(self, status, body):
+        try:
+            resp = json.loads(body)
+            n = resp.get("input_tokens", 0) or 0
+        except ValueError:
+            n = 0
+        return ResponseTranslation(
+            body=jdump({"count": n, "tokens": []}),
+            usage=Usage(input_tokens=n, total_tokens=n),
+            response_model=self._model,
+            end_of_stream=True,
+        )
+
+
 # --- multipart audio endpoints -------------------------------------------------
 
 
