@@ -501,17 +501,24 @@ cache_topk_kernel_t(const bf16* __restrict__ index, long long n_rows,
           b = *reinterpret_cast<const short8*>(&q[(long long)(q0 + row) * dim + kk]);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[ks], b, acc, 0, 0, 0);
       }
-      // acc reg r holds C[idx_row = kgrp*4+r][query = row]: fold the 4
-      // regs locally, then across the 4 lane groups sharing this query
-      // column (lanes differing in bits 4..5), then into the block table.
-      unsigned long long p = 0;
+      // acc reg r holds C[idx_row = kgrp*4+r][query = row]. PMC showed the
+      // old 4x pack_score + u64 compare chain made this kernel VALU-bound
+      // (9:1 VALU:MFMA); fold the argmax in float domain first and pack
+      // exactly once per lane per q0, then reduce across the 4 lane groups
+      // sharing this query column (lanes differing in bits 4..5).
+      // out-of-range C rows carry 0.0 (zeroed A fragments), which could
+      // out-rank valid negative scores — mask them before the argmax
+      int lim = (int)min((long long)16, n_rows - i0);
+      float best_s = -1e30f;
+      int best_r = 0;
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        long long irow = i0 + kgrp * 4 + r;
-        float s = (irow < n_rows && q_in) ? acc[r] : -1e30f;
-        unsigned long long pk = pack_score(s, (unsigned)(irow & 0xFFFFFFFF));
-        if (pk > p) p = pk;
+        float s = (kgrp * 4 + r < lim) ? acc[r] : -1e30f;
+        if (s > best_s) { best_s = s; best_r = r; }
       }
+      long long irow = i0 + kgrp * 4 + best_r;
+      if (!q_in) best_s = -1e30f;
+      unsigned long long p = pack_score(best_s, (unsigned)(irow & 0xFFFFFFFF));
       #pragma unroll
       for (int off = 16; off < 64; off <<= 1) {
         unsigned long long o = shfl_xor_u64(p, off);
