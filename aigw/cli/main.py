@@ -114,6 +114,30 @@ def cmd_translate(args) -> int:
     return 0
 
 
+def cmd_mcp_stdio(args) -> int:
+    from aigw.mcp.stdio_bridge import serve_stdio_bridge
+
+    command = [c for c in args.command if c != "--"]
+    if not command:
+        print("mcp-stdio: missing server command", file=sys.stderr)
+        return 2
+
+    async def run():
+        _bridge, runner = await serve_stdio_bridge(command, args.host, args.port)
+        print(f"mcp-stdio bridge on http://{args.host}:{args.port}/mcp -> {command}",
+              flush=True)
+        try:
+            await asyncio.Event().wait()
+        finally:
+            await runner.cleanup()
+
+    try:
+        asyncio.run(run())
+    except KeyboardInterrupt:
+        pass
+    return 0
+
+
 def cmd_healthcheck(args) -> int:
     url = f"http://127.0.0.1:{args.port}/health"
     try:
@@ -151,6 +175,14 @@ def main(argv=None) -> int:
 
     ver = sub.add_parser("version")
     ver.set_defaults(fn=lambda a: (print(f"aigw {aigw.__version__}"), 0)[1])
+
+    br = sub.add_parser("mcp-stdio", help="expose a stdio MCP server over HTTP "
+                                          "(cmd/aigw/stdio2http.go analogue)")
+    br.add_argument("--host", default="127.0.0.1")
+    br.add_argument("--port", type=int, default=internalapi.DEFAULT_MCP_PORT)
+    br.add_argument("command", nargs=argparse.REMAINDER,
+                    help="stdio MCP server command (after --)")
+    br.set_defaults(fn=cmd_mcp_stdio)
 
     args = ap.parse_args(argv)
     return args.fn(args)
