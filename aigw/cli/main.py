@@ -153,6 +153,8 @@ def cmd_healthcheck(args) -> int:
 
 def main(argv=None) -> int:
     logging.basicConfig(level=os.environ.get("AIGW_LOG_LEVEL", "INFO"))
+    if os.environ.get("AIGW_ACCESS_LOG", "").lower() in ("1", "true", "json"):
+        logging.getLogger("aigw.access").setLevel(logging.INFO)
     ap = argparse.ArgumentParser(prog="aigw", description=__doc__)
     sub = ap.add_subparsers(dest="cmd", required=True)
 

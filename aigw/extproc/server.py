@@ -49,6 +49,10 @@ except ImportError:  # pragma: no cover - built by setup.py everywhere
     _native = None
 
 logger = logging.getLogger("aigw.server")
+access_logger = logging.getLogger("aigw.access")
+# per-request JSON access lines are opt-in (AIGW_ACCESS_LOG=1 in the CLI)
+if access_logger.level == logging.NOTSET:
+    access_logger.setLevel(logging.WARNING)
 
 # endpoint key -> (operation name for metrics)
 JSON_ENDPOINTS = {
@@ -641,6 +645,29 @@ class GatewayServer:
         self.metrics.requests_total.labels(
             endpoint=endpoint, backend=backend.name if backend else "", status=str(status)
         ).inc()
+        if access_logger.isEnabledFor(logging.INFO):
+            # enriched access log, carrying what the reference pushes into
+            # Envoy access logs via dynamic metadata (§5.5: backend_name,
+            # token counts, token_latency_ttft/itl)
+            access_logger.info(
+                json.dumps(
+                    {
+                        "endpoint": endpoint,
+                        "route": route.route.name if route else "",
+                        "backend": backend.name if backend else "",
+                        "status": status,
+                        "duration_ms": round(elapsed * 1000.0, 2),
+                        "model": model,
+                        "response_model": response_model,
+                        "input_tokens": usage.input_tokens,
+                        "output_tokens": usage.output_tokens,
+                        "total_tokens": usage.total_tokens,
+                        "ttft_ms": round(ttft * 1000.0, 2) if ttft >= 0 else None,
+                        "error_type": error_type or None,
+                    },
+                    separators=(",", ":"),
+                )
+            )
 
     async def _unary_response(
         self, endpoint, route, backend, translator, upstream, headers, model,
