@@ -223,6 +223,12 @@ class MCPRoute:
     name: str
     path: str = "/mcp"
     backends: list[MCPBackend] = field(default_factory=list)
+    # client authorization at the gateway (mcpconfig.go:102-117): requests
+    # must carry this bearer token; failures answer 401 with an OAuth
+    # protected-resource-metadata WWW-Authenticate challenge when
+    # resource_metadata_url is set
+    bearer_token: str = ""
+    resource_metadata_url: str = ""
 
 
 @dataclass

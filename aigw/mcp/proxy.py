@@ -140,7 +140,29 @@ class MCPProxy:
 
     # ---- handlers ------------------------------------------------------------
 
+    def _authorize(self, request: web.Request):
+        """Bearer gate; 401 carries the OAuth protected-resource metadata
+        pointer (authorization.go WWW-Authenticate behavior)."""
+        if not self.route.bearer_token:
+            return None
+        got = request.headers.get("authorization", "")
+        if got == f"Bearer {self.route.bearer_token}":
+            return None
+        resp = web.json_response(
+            {"jsonrpc": "2.0", "id": None,
+             "error": {"code": -32001, "message": "unauthorized"}},
+            status=401,
+        )
+        if self.route.resource_metadata_url:
+            resp.headers["www-authenticate"] = (
+                f'Bearer resource_metadata="{self.route.resource_metadata_url}"'
+            )
+        return resp
+
     async def handle(self, request: web.Request) -> web.StreamResponse:
+        denied = self._authorize(request)
+        if denied is not None:
+            return denied
         if request.method == "GET":
             # server-initiated stream endpoint: no push support yet
             return web.Response(status=405, text="SSE server stream not supported")

@@ -97,6 +97,47 @@ async def _start(app_handler):
     return runner, runner.addresses[0][1]
 
 
+def test_mcp_authorization_gate():
+    async def main():
+        s1 = FakeMCPServer("a", ["t"])
+        r1, p1 = await _start(s1.handle)
+        cfg = load_config(
+            {
+                "version": "v1",
+                "routes": [
+                    {"name": "d", "backends": [
+                        {"name": "d", "upstream": {"host": "127.0.0.1", "port": 9}}]}
+                ],
+                "mcp": {
+                    "routes": [
+                        {"name": "m", "path": "/mcp",
+                         "bearerToken": "tok-123",
+                         "resourceMetadataUrl": "https://gw/.well-known/oauth-protected-resource",
+                         "backends": [
+                             {"name": "a", "upstream": {"host": "127.0.0.1", "port": p1}}]}
+                    ]
+                },
+            }
+        )
+        server = GatewayServer(RuntimeConfig(cfg))
+        gw = await run_server(server, host="127.0.0.1", port=0)
+        port = gw.addresses[0][1]
+        base = f"http://127.0.0.1:{port}/mcp"
+        async with aiohttp.ClientSession() as client:
+            async with client.post(base, json={"jsonrpc": "2.0", "id": 1, "method": "ping"}) as r:
+                assert r.status == 401
+                assert "resource_metadata=" in r.headers["www-authenticate"]
+            async with client.post(
+                base, json={"jsonrpc": "2.0", "id": 1, "method": "ping"},
+                headers={"authorization": "Bearer tok-123"},
+            ) as r:
+                assert r.status == 200
+        await gw.cleanup()
+        await r1.cleanup()
+
+    asyncio.run(main())
+
+
 def test_mcp_gateway_end_to_end():
     async def main():
         s1 = FakeMCPServer("alpha", ["search", "fetch", "secret_tool"])
