@@ -138,7 +138,12 @@ class GatewayServer:
         gpu_services=None,
         tracer=None,
         max_body_bytes: int = 50 * 1024 * 1024,  # reference raises Envoy's buffer to 50MiB
+        endpoint_prefixes: Optional[dict[str, str]] = None,
+        # extra path prefixes per API family (mainlib --endpointPrefixes,
+        # main.go:143-147), e.g. {"openai": "/openai"} additionally mounts
+        # /openai/v1/chat/completions etc.
     ):
+        self.endpoint_prefixes = endpoint_prefixes or {}
         self.runtime = runtime
         self.metrics = metrics or GenAIMetrics()
         self.limiter = limiter or RateLimiter(runtime.rate_limits)
@@ -179,10 +184,21 @@ class GatewayServer:
 
     def make_app(self) -> web.Application:
         app = web.Application(client_max_size=self.max_body_bytes)
+
+        def mount(path: str, handler) -> None:
+            app.router.add_post(path, handler)
+            for family, prefix in self.endpoint_prefixes.items():
+                if family == "anthropic" and path.startswith("/anthropic/"):
+                    app.router.add_post(prefix + path[len("/anthropic") :], handler)
+                elif family == "cohere" and path.startswith("/v2/"):
+                    app.router.add_post(prefix + path, handler)
+                elif family == "openai" and not path.startswith(("/anthropic/", "/v2/")):
+                    app.router.add_post(prefix + path, handler)
+
         for ep in JSON_ENDPOINTS:
-            app.router.add_post(ep, self._make_handler(ep))
+            mount(ep, self._make_handler(ep))
         for ep in MULTIPART_ENDPOINTS:
-            app.router.add_post(ep, self._make_multipart_handler(ep))
+            mount(ep, self._make_multipart_handler(ep))
         app.router.add_get("/v1/models", self._handle_models)
         app.router.add_get("/anthropic/v1/models", self._handle_anthropic_models)
         app.router.add_get("/health", self._handle_health)

@@ -237,3 +237,43 @@ def test_cli_run_end_to_end(tmp_path):
         asyncio.run_coroutine_threadsafe(state["runner"].cleanup(), loop).result(10)
         loop.call_soon_threadsafe(loop.stop)
         t.join(timeout=10)
+
+
+def test_endpoint_prefixes():
+    """--endpointPrefixes parity: alternate mount points per API family."""
+    from aigw.testing.mockupstream import start_mock_upstream
+
+    async def main():
+        mock, runner, port = await start_mock_upstream()
+        cfg = load_config(
+            {
+                "version": "v1",
+                "routes": [
+                    {"name": "r", "backends": [
+                        {"name": "b", "schema": "OpenAI",
+                         "upstream": {"host": "127.0.0.1", "port": port}}]}
+                ],
+            }
+        )
+        server = GatewayServer(
+            RuntimeConfig(cfg),
+            endpoint_prefixes={"openai": "/openai", "anthropic": "/claude"},
+        )
+        gw = await run_server(server, host="127.0.0.1", port=0)
+        gport = gw.addresses[0][1]
+        async with aiohttp.ClientSession() as c:
+            async with c.post(
+                f"http://127.0.0.1:{gport}/openai/v1/chat/completions",
+                json={"model": "m", "messages": [{"role": "user", "content": "q"}]},
+            ) as r:
+                assert r.status == 200
+            # canonical path still works
+            async with c.post(
+                f"http://127.0.0.1:{gport}/v1/chat/completions",
+                json={"model": "m", "messages": []},
+            ) as r:
+                assert r.status == 200
+        await gw.cleanup()
+        await runner.cleanup()
+
+    asyncio.run(main())
