@@ -63,10 +63,13 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
 
     sync = None
     if world > 1:
-        from aigw.parallel import StateSync
+        import torch.distributed as dist
 
-        sync = StateSync(server.limiter)
-        await sync.start()
+        if dist.is_initialized():  # primary process of each shard only
+            from aigw.parallel import StateSync
+
+            sync = StateSync(server.limiter)
+            await sync.start()
 
     watcher = None
     if watch_path:
@@ -74,7 +77,9 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
         await watcher.start()
 
     port = args.port + rank
-    runner = await run_server(server, host=args.host, port=port)
+    runner = await run_server(
+        server, host=args.host, port=port, reuse_port=getattr(args, "workers", 1) > 1
+    )
     print(f"aigw shard {rank}/{world} listening on http://{args.host}:{port}", flush=True)
     try:
         await asyncio.Event().wait()
@@ -89,6 +94,21 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
 def cmd_run(args) -> int:
     world = int(os.environ.get("WORLD_SIZE", args.shards))
     rank = int(os.environ.get("RANK", 0))
+    if args.workers > 1:
+        # worker processes share the listen port via SO_REUSEPORT; each has
+        # its own event loop + GPU services on the shard's GPU. Token-budget
+        # buckets are per worker within a shard (cross-SHARD consistency is
+        # the RCCL-synced path); divide limits by worker count when exact
+        # in-shard budgets matter.
+        import multiprocessing as _mp
+
+        ctx = _mp.get_context("spawn")
+        procs = [
+            ctx.Process(target=_worker_run, args=(args, rank, world, i))
+            for i in range(1, args.workers)
+        ]
+        for p in procs:
+            p.start()
     if world > 1:
         import torch
         import torch.distributed as dist
@@ -102,6 +122,13 @@ def cmd_run(args) -> int:
     except KeyboardInterrupt:
         pass
     return 0
+
+
+def _worker_run(args, rank: int, world: int, worker_idx: int) -> None:
+    try:
+        asyncio.run(_run_shard(args, rank, world))
+    except KeyboardInterrupt:
+        pass
 
 
 def cmd_translate(args) -> int:
@@ -163,6 +190,9 @@ def main(argv=None) -> int:
     runp.add_argument("--host", default="0.0.0.0")
     runp.add_argument("--port", type=int, default=internalapi.DEFAULT_LISTEN_PORT)
     runp.add_argument("--shards", type=int, default=1, help="shards (one per GPU)")
+    runp.add_argument("--workers", type=int, default=1,
+                      help="HTTP worker processes per shard (SO_REUSEPORT; "
+                           "CPython's GIL caps one loop near ~3k req/s)")
     runp.add_argument("--gpu", action="store_true", help="enable GPU services")
     runp.add_argument("--semantic-cache", action="store_true")
     runp.set_defaults(fn=cmd_run)

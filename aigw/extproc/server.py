@@ -834,12 +834,17 @@ class GatewayServer:
 
 
 async def run_server(
-    server: GatewayServer, host: str = "0.0.0.0", port: int = internalapi.DEFAULT_LISTEN_PORT
+    server: GatewayServer,
+    host: str = "0.0.0.0",
+    port: int = internalapi.DEFAULT_LISTEN_PORT,
+    reuse_port: bool = False,
 ) -> web.AppRunner:
     app = server.make_app()
     runner = web.AppRunner(app, access_log=None)
     await runner.setup()
-    site = web.TCPSite(runner, host, port, backlog=4096, reuse_address=True)
+    site = web.TCPSite(
+        runner, host, port, backlog=4096, reuse_address=True, reuse_port=reuse_port
+    )
     await site.start()
     logger.info("aigw listening on http://%s:%d", host, port)
     return runner
