@@ -20,9 +20,12 @@ reconcilers against a fake client.
 from __future__ import annotations
 
 import base64
+import logging
 from typing import Optional
 
 import yaml
+
+logger = logging.getLogger("aigw.controller")
 
 from aigw.filterapi.config import (
     APISchema,
@@ -214,13 +217,17 @@ def _parse_costs(costs: list) -> list[LLMRequestCost]:
     out = []
     for c in costs or []:
         t = c.get("type", "OutputToken")
-        if t == "CacheCreationInputToken":
-            # v1beta1 extra type: fold into CEL so the engine stays small
+        # v1beta1 extra types fold into CEL so the core enum stays small
+        extra = {
+            "CacheCreationInputToken": "cache_creation_input_tokens",
+            "ReasoningToken": "reasoning_tokens",
+        }
+        if t in extra:
             out.append(
                 LLMRequestCost(
                     metadata_key=c["metadataKey"],
                     type=LLMRequestCostType.CEL,
-                    cel="cache_creation_input_tokens",
+                    cel=extra[t],
                 )
             )
             continue
@@ -270,14 +277,25 @@ def translate_crds(docs: list[dict]) -> Config:
                 asb_name = bref.get("name", "")
                 asb = bundle.find("AIServiceBackend", asb_name, ns)
                 if asb is None:
-                    raise ConfigError(f"AIServiceBackend {asb_name!r} not found")
+                    # cross-file reference (bundles are often split across
+                    # files); the controller leaves such rules unresolved
+                    # rather than failing the whole Gateway
+                    logger.warning(
+                        "translate: AIServiceBackend %r not in bundle; skipping ref",
+                        asb_name,
+                    )
+                    continue
                 aspec = asb.get("spec") or {}
                 schema_d = aspec.get("schema") or {}
                 schema = APISchema(
                     name=APISchemaName(schema_d.get("name", "OpenAI")),
                     version=schema_d.get("version", "") or "",
                 )
-                upstream = _resolve_upstream(bundle, ns, aspec.get("backendRef") or {})
+                try:
+                    upstream = _resolve_upstream(bundle, ns, aspec.get("backendRef") or {})
+                except ConfigError as e:
+                    logger.warning("translate: %s; skipping ref %r", e, asb_name)
+                    continue
                 auth = _bsp_for_backend(bundle, ns, asb_name)
                 timeout = 60.0
                 timeouts = rule.get("timeouts") or {}
@@ -356,27 +374,53 @@ def translate_crds(docs: list[dict]) -> Config:
         for bref in spec.get("backendRefs") or []:
             upstream = _resolve_upstream(bundle, ns, bref)
             sel = bref.get("toolSelector") or {}
+            auth = None
+            sp = bref.get("securityPolicy") or {}
+            if sp.get("apiKey"):
+                ref = (sp["apiKey"].get("secretRef") or {})
+                inline = sp["apiKey"].get("inline")
+                key = inline or _secret_value(
+                    bundle, ref.get("name", ""), ref.get("namespace", ns),
+                    ["apiKey", "api-key", "token"],
+                )
+                auth = BackendAuth(api_key=key.strip())
             mcp_backends.append(
                 MCPBackend(
                     name=bref.get("name", ""),
                     upstream=upstream,
                     path=bref.get("path", "/mcp"),
-                    tool_include=sel.get("include") or [],
-                    tool_exclude=sel.get("exclude") or [],
+                    # v1beta1 uses include/exclude + includeRegex/excludeRegex
+                    # (mcp_route.go ToolSelector); our selector treats entries
+                    # as exact-or-regex, so the four lists merge
+                    tool_include=(sel.get("include") or []) + (sel.get("includeRegex") or []),
+                    tool_exclude=(sel.get("exclude") or []) + (sel.get("excludeRegex") or []),
+                    auth=auth,
                 )
             )
+        # route-level OAuth JWT validation (securityPolicy.oauth) is not
+        # implemented; the bearer-token gate + resource-metadata challenge
+        # covers the WWW-Authenticate discovery flow.
         mcp_routes.append(
             MCPRoute(name=md.get("name", "mcp"), path=spec.get("path", "/mcp"),
                      backends=mcp_backends)
         )
 
-    unknown = set(bundle.by_kind) - _IGNORED_KINDS - {
+    handled = {
         "AIGatewayRoute", "AIServiceBackend", "BackendSecurityPolicy", "Backend",
         "BackendTrafficPolicy", "GatewayConfig", "QuotaPolicy", "MCPRoute", "Secret",
         "Gateway",
     }
-    if unknown:
-        raise ConfigError(f"unsupported kinds in bundle: {sorted(unknown)}")
+    for kind, docs_of_kind in bundle.by_kind.items():
+        if kind in handled or kind in _IGNORED_KINDS:
+            continue
+        # Unknown kinds in OUR API group are typos and must fail loudly;
+        # everything else (RBAC, operators, arbitrary k8s objects that ride
+        # along in real bundles) is skipped like the controller would.
+        if any(
+            "aigateway.envoyproxy.io" in (d.get("apiVersion") or "") for d in docs_of_kind
+        ):
+            raise ConfigError(f"unsupported aigateway.envoyproxy.io kind: {kind}")
+        logger.info("translate: skipping %d %s object(s)", len(docs_of_kind), kind)
 
     cfg = Config(
         uuid="translated",

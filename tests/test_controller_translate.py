@@ -164,7 +164,7 @@ def test_token_ratelimit_bundle_limits():
 
 def test_unknown_kind_rejected():
     with pytest.raises(ConfigError):
-        translate_yaml("apiVersion: v1\nkind: TotallyUnknown\nmetadata: {name: x}\n")
+        translate_yaml("apiVersion: aigateway.envoyproxy.io/v1beta1\nkind: TotallyUnknown\nmetadata: {name: x}\n")
 
 
 def test_autoconfig_env():
