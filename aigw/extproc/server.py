@@ -212,7 +212,8 @@ class GatewayServer:
 
             self._mcp_proxies = []
             for mr in mcp_cfg.routes:
-                proxy = MCPProxy(mr, mcp_cfg.session_seed)
+                proxy = MCPProxy(mr, mcp_cfg.session_seed, metrics=self.metrics,
+                                 tracer=self.tracer)
                 self._mcp_proxies.append(proxy)
                 for method in ("POST", "GET", "DELETE"):
                     app.router.add_route(method, mr.path, proxy.handle)

@@ -47,9 +47,12 @@ def _rpc_error(id_, code: int, message: str, status: int = 200) -> web.Response:
 
 
 class MCPProxy:
-    def __init__(self, route: MCPRoute, session_seed: str, client_factory=None):
+    def __init__(self, route: MCPRoute, session_seed: str, client_factory=None,
+                 metrics=None, tracer=None):
         self.route = route
         self.crypto = SessionCrypto(session_seed)
+        self.metrics = metrics  # aigw.metrics.GenAIMetrics or None
+        self.tracer = tracer  # aigw.tracing.Tracer or None
         self._client_factory = client_factory
         self._session: Optional[aiohttp.ClientSession] = None
         self._tool_res: dict[str, list] = {}
@@ -160,6 +163,34 @@ class MCPProxy:
         return resp
 
     async def handle(self, request: web.Request) -> web.StreamResponse:
+        import time as _time
+
+        t0 = _time.monotonic()
+        span = None
+        if self.tracer is not None:
+            span = self.tracer.start_span("mcp", dict(request.headers))
+        resp = None
+        try:
+            resp = await self._handle_inner(request)
+            return resp
+        finally:
+            method = request.get("aigw_mcp_method", "") or request.method
+            if self.metrics is not None:
+                outcome = "ok"
+                if isinstance(resp, web.Response) and resp.status >= 400:
+                    outcome = f"http_{resp.status}"
+                elif resp is None:
+                    outcome = "error"
+                self.metrics.mcp_requests.labels(method=method, outcome=outcome).inc()
+                self.metrics.mcp_duration.labels(method=method).observe(
+                    _time.monotonic() - t0
+                )
+            if span is not None:
+                span.set("mcp.method", method)
+                span.set("mcp.route", self.route.name)
+                self.tracer.end_span(span)
+
+    async def _handle_inner(self, request: web.Request) -> web.StreamResponse:
         denied = self._authorize(request)
         if denied is not None:
             return denied
@@ -175,6 +206,7 @@ class MCPProxy:
         if isinstance(payload, list):
             return _rpc_error(None, -32600, "batch requests not supported", status=400)
         method = payload.get("method", "")
+        request["aigw_mcp_method"] = method
         id_ = payload.get("id")
         token = request.headers.get(internalapi.MCP_SESSION_ID_HEADER, "")
         sessions: dict[str, str] = {}

@@ -90,6 +90,20 @@ class GenAIMetrics:
             ["event"],
             registry=self.registry,
         )
+        # MCP gateway instruments (internal/metrics/mcp_metrics.go)
+        self.mcp_requests = Counter(
+            "aigw_mcp_requests_total",
+            "MCP JSON-RPC requests by method/outcome",
+            ["method", "outcome"],
+            registry=self.registry,
+        )
+        self.mcp_duration = Histogram(
+            "aigw_mcp_request_duration_seconds",
+            "MCP request duration",
+            ["method"],
+            buckets=_SECONDS_BUCKETS,
+            registry=self.registry,
+        )
 
     def labels(self, *, operation: str, provider: str, original_model: str,
                request_model: str, response_model: str) -> dict[str, str]:
