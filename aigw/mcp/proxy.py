@@ -37,6 +37,7 @@ logger = logging.getLogger("aigw.mcp")
 
 PREFIX_SEP = "__"
 PROTOCOL_VERSION = "2025-06-18"
+_METHOD_KEY = web.AppKey("aigw_mcp_method", str) if hasattr(web, "AppKey") else "aigw_mcp_method"
 
 
 def _rpc_error(id_, code: int, message: str, status: int = 200) -> web.Response:
@@ -174,7 +175,7 @@ class MCPProxy:
             resp = await self._handle_inner(request)
             return resp
         finally:
-            method = request.get("aigw_mcp_method", "") or request.method
+            method = request.get(_METHOD_KEY, "") or request.method
             if self.metrics is not None:
                 outcome = "ok"
                 if isinstance(resp, web.Response) and resp.status >= 400:
@@ -206,7 +207,7 @@ class MCPProxy:
         if isinstance(payload, list):
             return _rpc_error(None, -32600, "batch requests not supported", status=400)
         method = payload.get("method", "")
-        request["aigw_mcp_method"] = method
+        request[_METHOD_KEY] = method
         id_ = payload.get("id")
         token = request.headers.get(internalapi.MCP_SESSION_ID_HEADER, "")
         sessions: dict[str, str] = {}
