@@ -37,7 +37,7 @@ logger = logging.getLogger("aigw.mcp")
 
 PREFIX_SEP = "__"
 PROTOCOL_VERSION = "2025-06-18"
-_METHOD_KEY = web.AppKey("aigw_mcp_method", str) if hasattr(web, "AppKey") else "aigw_mcp_method"
+_METHOD_KEY = web.RequestKey("aigw_mcp_method", str) if hasattr(web, "RequestKey") else "aigw_mcp_method"
 
 
 def _rpc_error(id_, code: int, message: str, status: int = 200) -> web.Response:
