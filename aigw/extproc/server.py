@@ -34,6 +34,7 @@ from aiohttp import web
 from aigw import internalapi
 from aigw.backendauth import build_auth_handler
 from aigw.extproc.router import RETRIABLE_STATUSES, backend_attempts, max_attempts
+from aigw.extproc.upstream_client import UpstreamError
 from aigw.filterapi.config import APISchemaName, Backend
 from aigw.filterapi.runtime import CompiledRoute, RuntimeConfig
 from aigw.llmcost import CostVars
@@ -156,7 +157,7 @@ class GatewayServer:
         else:
             self.span_recorder = None
         self.max_body_bytes = max_body_bytes
-        self._session: Optional[aiohttp.ClientSession] = None
+        self._session = None  # lean upstream client (aigw.extproc.upstream_client)
         self._started_at = time.time()
         # per-backend live stats for the KV-occupancy endpoint picker:
         # name -> [active_requests, estimated_active_tokens]
@@ -166,10 +167,9 @@ class GatewayServer:
 
     async def start(self) -> None:
         if self._session is None:
-            connector = aiohttp.TCPConnector(limit=0, ttl_dns_cache=300, keepalive_timeout=75)
-            self._session = aiohttp.ClientSession(
-                connector=connector, auto_decompress=True, skip_auto_headers=("User-Agent",)
-            )
+            from aigw.extproc.upstream_client import LeanClient
+
+            self._session = LeanClient()
 
     async def close(self) -> None:
         if self._session is not None:
@@ -547,7 +547,6 @@ class GatewayServer:
                 if "authorization" in headers:
                     up_headers.setdefault("authorization", headers["authorization"])
 
-            url = backend.upstream.base_url + tr.path
             if backend.upstream.hostname:
                 up_headers["host"] = backend.upstream.hostname
             if span is not None:
@@ -562,35 +561,42 @@ class GatewayServer:
             st[0] += 1
             st[1] += gpu_input_tokens
             try:
-                timeout = aiohttp.ClientTimeout(total=backend.timeout_s)
-                async with self._session.post(
-                    url, data=out_body, headers=up_headers, timeout=timeout
-                ) as upstream:
-                    status = upstream.status
-                    if status >= 400:
-                        err_body = await upstream.read()
-                        if status in RETRIABLE_STATUSES and attempts_left > 0:
-                            last_error = f"upstream {backend.name} returned {status}"
-                            logger.warning("%s; trying next backend", last_error)
-                            continue
-                        out = translator.response_error(status, err_body, dict(upstream.headers))
-                        self._finish_metrics(
-                            endpoint, route, backend, model, "", Usage(), start,
-                            status=status, error_type=f"upstream_{status}",
-                        )
-                        return web.Response(
-                            body=out, status=status, content_type="application/json"
-                        )
-                    if stream:
-                        return await self._stream_response(
-                            request, endpoint, route, backend, translator, upstream,
-                            headers, model, start, gpu_input_tokens, span=span,
-                        )
-                    return await self._unary_response(
-                        endpoint, route, backend, translator, upstream, headers,
-                        model, start, gpu_input_tokens, body, cache_key_vec, span=span,
+                upstream = await self._session.post(
+                    host=backend.upstream.host,
+                    port=backend.upstream.port,
+                    tls=backend.upstream.tls,
+                    path=backend.upstream.path_prefix + tr.path,
+                    headers=up_headers,
+                    body=out_body,
+                    timeout_s=backend.timeout_s,
+                    server_name=backend.upstream.hostname,
+                )
+                status = upstream.status
+                if status >= 400:
+                    err_body = await upstream.read()
+                    upstream.release()
+                    if status in RETRIABLE_STATUSES and attempts_left > 0:
+                        last_error = f"upstream {backend.name} returned {status}"
+                        logger.warning("%s; trying next backend", last_error)
+                        continue
+                    out = translator.response_error(status, err_body, upstream.headers)
+                    self._finish_metrics(
+                        endpoint, route, backend, model, "", Usage(), start,
+                        status=status, error_type=f"upstream_{status}",
                     )
-            except (aiohttp.ClientError, asyncio.TimeoutError) as e:
+                    return web.Response(
+                        body=out, status=status, content_type="application/json"
+                    )
+                if stream:
+                    return await self._stream_response(
+                        request, endpoint, route, backend, translator, upstream,
+                        headers, model, start, gpu_input_tokens, span=span,
+                    )
+                return await self._unary_response(
+                    endpoint, route, backend, translator, upstream, headers,
+                    model, start, gpu_input_tokens, body, cache_key_vec, span=span,
+                )
+            except (UpstreamError, OSError, asyncio.TimeoutError, asyncio.IncompleteReadError) as e:
                 last_error = f"upstream {backend.name}: {type(e).__name__}: {e}"
                 logger.warning("%s; %d attempts left", last_error, attempts_left)
                 continue
@@ -690,7 +696,8 @@ class GatewayServer:
         self, endpoint, route, backend, translator, upstream, headers, model,
         start, gpu_input_tokens, orig_body, cache_key_vec, span=None,
     ) -> web.Response:
-        data = await upstream.read()  # aiohttp auto-decompresses gzip/br
+        data = await upstream.read()  # LeanResponse decompresses gzip/deflate
+        upstream.release()
         try:
             rtl = translator.response_body(upstream.status, data)
         except Exception as e:
@@ -704,7 +711,7 @@ class GatewayServer:
             endpoint, route, backend, model, rtl.response_model, usage, start,
             status=upstream.status,
         )
-        hdrs = translator.response_headers(upstream.status, dict(upstream.headers))
+        hdrs = translator.response_headers(upstream.status, upstream.headers)
         content_type = hdrs.get("content-type") or upstream.headers.get(
             "content-type", "application/json"
         )
@@ -719,7 +726,7 @@ class GatewayServer:
         self, request, endpoint, route, backend, translator, upstream,
         headers, model, start, gpu_input_tokens, span=None,
     ) -> web.StreamResponse:
-        hdrs = translator.response_headers(upstream.status, dict(upstream.headers))
+        hdrs = translator.response_headers(upstream.status, upstream.headers)
         content_type = hdrs.get("content-type") or upstream.headers.get(
             "content-type", "text/event-stream"
         )
@@ -732,8 +739,9 @@ class GatewayServer:
         usage = Usage()
         response_model = ""
         ttft = -1.0
+        aborted = False
         try:
-            async for chunk in upstream.content.iter_any():
+            async for chunk in upstream.iter_chunks():
                 if not chunk:
                     continue
                 rtl = translator.response_chunk(chunk)
@@ -751,9 +759,14 @@ class GatewayServer:
             if tail.usage is not None:
                 usage.merge_max(tail.usage)
         except (ConnectionResetError, asyncio.CancelledError):
+            aborted = True
             logger.info("client disconnected mid-stream")
             raise
         finally:
+            if aborted:
+                upstream.close()
+            else:
+                upstream.release()
             self._apply_costs(route, backend, headers, model, usage, gpu_input_tokens)
             if span is not None:
                 self.span_recorder.record_response(span, usage, response_model)
@@ -809,26 +822,33 @@ class GatewayServer:
             auth = build_auth_handler(backend)
             if auth is not None:
                 up_headers = auth(up_headers, tr.body, "POST", tr.path)
-            url = backend.upstream.base_url + tr.path
             try:
-                timeout = aiohttp.ClientTimeout(total=backend.timeout_s)
-                async with self._session.post(
-                    url, data=tr.body, headers=up_headers, timeout=timeout
-                ) as upstream:
-                    if upstream.status >= 400:
-                        err = await upstream.read()
-                        if upstream.status in RETRIABLE_STATUSES and attempts_left > 0:
-                            continue
-                        return web.Response(
-                            body=translator.response_error(upstream.status, err, {}),
-                            status=upstream.status,
-                            content_type="application/json",
-                        )
-                    return await self._unary_response(
-                        endpoint, route, backend, translator, upstream, headers,
-                        model, start, 0, None, None,
+                upstream = await self._session.post(
+                    host=backend.upstream.host,
+                    port=backend.upstream.port,
+                    tls=backend.upstream.tls,
+                    path=backend.upstream.path_prefix + tr.path,
+                    headers=up_headers,
+                    body=tr.body,
+                    timeout_s=backend.timeout_s,
+                    server_name=backend.upstream.hostname,
+                )
+                if upstream.status >= 400:
+                    err = await upstream.read()
+                    upstream.release()
+                    if upstream.status in RETRIABLE_STATUSES and attempts_left > 0:
+                        continue
+                    return web.Response(
+                        body=translator.response_error(upstream.status, err, {}),
+                        status=upstream.status,
+                        content_type="application/json",
                     )
-            except (aiohttp.ClientError, asyncio.TimeoutError) as e:
+                return await self._unary_response(
+                    endpoint, route, backend, translator, upstream, headers,
+                    model, start, 0, None, None,
+                )
+            except (UpstreamError, OSError, asyncio.TimeoutError,
+                    asyncio.IncompleteReadError) as e:
                 last_error = str(e)
                 continue
         return _json_error(503, last_error or "no healthy upstream", "upstream_error")
