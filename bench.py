@@ -83,23 +83,25 @@ def gateway_config(upstream_port: int) -> dict:
     }
 
 
-async def fire_step(session, url, payload_bytes, batch, latencies):
-    import aiohttp  # local import keeps worker spawn cheap
+async def fire_step(client, port, path, payload_bytes, batch, latencies):
+    """One step: `batch` concurrent POSTs via the lean pooled client."""
 
     async def one():
         t0 = time.perf_counter()
-        async with session.post(
-            url, data=payload_bytes, headers={"content-type": "application/json"}
-        ) as r:
-            await r.read()
-            assert r.status == 200, f"status {r.status}"
+        r = await client.post(
+            host="127.0.0.1", port=port, tls=False, path=path,
+            headers={"content-type": "application/json"},
+            body=payload_bytes, timeout_s=120.0,
+        )
+        await r.read()
+        r.release()
+        assert r.status == 200, f"status {r.status}"
         latencies.append((time.perf_counter() - t0) * 1000.0)
 
     await asyncio.gather(*(one() for _ in range(batch)))
 
 
 async def worker_main(args, local_rank: int, ready, go, out_q):
-    import aiohttp
 
     from aigw.extproc.server import GatewayServer, run_server
     from aigw.filterapi import RuntimeConfig, load_config
@@ -125,20 +127,21 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
     gw_runner = await run_server(server, host="127.0.0.1", port=0)
     gw_port = gw_runner.addresses[0][1]
 
+    from aigw.extproc.upstream_client import LeanClient
+
     payload = json.dumps(build_payload(args.tokens)).encode()
-    session = aiohttp.ClientSession(connector=aiohttp.TCPConnector(limit=0))
-    gw_url = f"http://127.0.0.1:{gw_port}/v1/chat/completions"
-    direct_url = f"http://127.0.0.1:{up_port}/v1/chat/completions"
+    client = LeanClient()
+    path = "/v1/chat/completions"
 
     # warm both paths, then record the warm direct-to-upstream baseline
     scratch: list[float] = []
     for _ in range(max(args.warmup, 1)):
-        await fire_step(session, direct_url, payload, args.batch, scratch)
+        await fire_step(client, up_port, path, payload, args.batch, scratch)
     for _ in range(args.warmup):
-        await fire_step(session, gw_url, payload, args.batch, scratch)
+        await fire_step(client, gw_port, path, payload, args.batch, scratch)
     direct_lat: list[float] = []
     for _ in range(2):
-        await fire_step(session, direct_url, payload, args.batch, direct_lat)
+        await fire_step(client, up_port, path, payload, args.batch, direct_lat)
     if use_gpu:
         torch.cuda.synchronize()
 
@@ -149,7 +152,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
     lat: list[float] = []
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        await fire_step(session, gw_url, payload, args.batch, lat)
+        await fire_step(client, gw_port, path, payload, args.batch, lat)
     if use_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
@@ -163,7 +166,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
             "p50_direct": statistics.median(direct_lat),
         }
     )
-    await session.close()
+    await client.close()
     await gw_runner.cleanup()
     await up_runner.cleanup()
     if gpu_services is not None:
