@@ -105,13 +105,15 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
 
     from aigw.extproc.server import GatewayServer, run_server
     from aigw.filterapi import RuntimeConfig, load_config
-    from aigw.testing.mockupstream import start_mock_upstream
+    from aigw.testing.fastmock import canned_chat_response, start_fast_mock
 
     use_gpu = torch.cuda.is_available()
     if use_gpu:
         torch.cuda.set_device(local_rank)
 
-    mock, up_runner, up_port = await start_mock_upstream(port=0)
+    up_server, up_port = await start_fast_mock(
+        response=canned_chat_response(prompt_tokens=args.tokens)
+    )
     cfg = load_config(gateway_config(up_port))
 
     gpu_services = None
@@ -168,7 +170,8 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
     )
     await client.close()
     await gw_runner.cleanup()
-    await up_runner.cleanup()
+    up_server.close()
+    await up_server.wait_closed()
     if gpu_services is not None:
         gpu_services.close()
 
