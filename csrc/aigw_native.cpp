@@ -279,8 +279,9 @@ class SSEFeed {
       --value_len;
     }
     if (name_len == 4 && std::memcmp(line, "data", 4) == 0) {
-      if (!data_.empty()) data_.push_back('\n');
+      if (has_data_) data_.push_back('\n');  // joins EMPTY data lines too
       data_.append(value, value_len);
+      has_data_ = true;
       has_fields_ = true;
     } else if (name_len == 5 && std::memcmp(line, "event", 5) == 0) {
       event_.assign(value, value_len);
@@ -295,12 +296,14 @@ class SSEFeed {
     data_.clear();
     event_.clear();
     has_fields_ = false;
+    has_data_ = false;
   }
 
   std::string buf_;
   std::string data_;
   std::string event_;
   bool has_fields_ = false;
+  bool has_data_ = false;
 };
 
 PYBIND11_MODULE(aigw_native, m) {
