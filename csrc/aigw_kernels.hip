@@ -415,6 +415,102 @@ gemm_bf16_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
   }
 }
 
+// Large-shape MFMA GEMM: 128x128 tile, BK=32, 4 waves (2x2), LDS staged
+// via async global_load_lds width-16 (the HIP guide's m97 structure:
+// naive direct-load was 71 TF; this structure reaches ~900 TF at 4096^3
+// on the guide's ladder). Both operands NT ([row][k] contiguous) so each
+// lane's fragment is one ds_read_b128. Requires M%128==0, N%128==0,
+// K%32==0 (host dispatches the simple kernel otherwise). blockIdx is
+// XCD-swizzled with the bijective m204 mapping so neighbor tiles share a
+// per-XCD L2.
+__global__ void __launch_bounds__(256)
+gemm_bf16_nt_tiled_kernel(const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+                          float* __restrict__ C, int M, int N, int K,
+                          const float* __restrict__ bias, int relu) {
+  __shared__ short As[128 * 32];
+  __shared__ short Bs[128 * 32];
+  int nwg = (int)(gridDim.x * gridDim.y);
+  int orig = (int)(blockIdx.y * gridDim.x + blockIdx.x);
+  int wg = orig;
+  if (nwg >= 8) {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = orig & 7, seq = orig >> 3;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + seq;
+  }
+  int bx = wg % (int)gridDim.x;  // N tile
+  int by = wg / (int)gridDim.x;  // M tile
+  long long m0 = (long long)by * 128;
+  long long n0 = (long long)bx * 128;
+
+  int t = threadIdx.x;
+  int lane = t & 63;
+  int wave = t >> 6;
+  int wr = wave >> 1, wc = wave & 1;
+  int frow = lane & 15;
+  int kgrp = lane >> 4;
+
+  floatx4 acc[4][4];
+  #pragma unroll
+  for (int m = 0; m < 4; ++m)
+    #pragma unroll
+    for (int n = 0; n < 4; ++n) acc[m][n] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < K; k0 += 32) {
+    // stage A and B tiles: 512 16-byte chunks each; chunk f covers
+    // row = f>>2, ks = f&3 of the [128][32] tile. Each wave's 64 lanes
+    // write one contiguous 1 KiB LDS span (gload_lds dest is uniform
+    // base + lane*16).
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int f = wave * 128 + i * 64 + lane;
+      int row = f >> 2, ks = f & 3;
+      const bf16* ga = &A[(m0 + row) * K + k0 + ks * 8];
+      const bf16* gb = &Bt[(n0 + row) * K + k0 + ks * 8];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)&As[(wave * 128 + i * 64) * 8],
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)&Bs[(wave * 128 + i * 64) * 8],
+          16, 0, 0);
+    }
+    __syncthreads();  // drains vmcnt: staged tiles visible
+
+    short8 a[4], b[4];
+    #pragma unroll
+    for (int m = 0; m < 4; ++m)
+      a[m] = *reinterpret_cast<const short8*>(
+          &As[(wr * 64 + m * 16 + frow) * 32 + kgrp * 8]);
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      b[n] = *reinterpret_cast<const short8*>(
+          &Bs[(wc * 64 + n * 16 + frow) * 32 + kgrp * 8]);
+    #pragma unroll
+    for (int m = 0; m < 4; ++m)
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[m], b[n], acc[m][n], 0, 0, 0);
+    __syncthreads();  // tile fully consumed before restaging
+  }
+
+  #pragma unroll
+  for (int m = 0; m < 4; ++m) {
+    long long crow_base = m0 + wr * 64 + m * 16 + kgrp * 4;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      long long ccol = n0 + wc * 64 + n * 16 + frow;
+      float bv = bias ? bias[ccol] : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[m][n][r] + bv;
+        if (relu && v < 0.f) v = 0.f;
+        C[(crow_base + r) * N + ccol] = v;
+      }
+    }
+  }
+}
+
 // L2-normalize rows, fp32 -> bf16 (one wave per row)
 __global__ void l2norm_rows_kernel(const float* __restrict__ in, bf16* __restrict__ out,
                                    int rows, int dim) {
@@ -704,6 +800,15 @@ at::Tensor gemm_bf16_nt(at::Tensor a, at::Tensor bt, c10::optional<at::Tensor> b
   if (bias.has_value()) {
     check_cuda(*bias, "bias");
     bias_ptr = bias->data_ptr<float>();
+  }
+  if (M % 128 == 0 && N % 128 == 0 && K % 32 == 0) {
+    dim3 grid(N / 128, M / 128);
+    hipLaunchKernelGGL(gemm_bf16_nt_tiled_kernel, grid, dim3(256), 0,
+                       current_stream(),
+                       reinterpret_cast<bf16*>(a.data_ptr<at::BFloat16>()),
+                       reinterpret_cast<bf16*>(bt.data_ptr<at::BFloat16>()),
+                       c.data_ptr<float>(), M, N, K, bias_ptr, relu ? 1 : 0);
+    return c;
   }
   dim3 grid((M + 15) / 16, (N + 63) / 64);
   hipLaunchKernelGGL(gemm_bf16_nt_kernel, grid, dim3(256), 0, current_stream(),

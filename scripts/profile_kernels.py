@@ -72,7 +72,23 @@ def main():
         "req_per_s": round(args.batch / dt, 1),
     }))
 
-    # standalone MFMA GEMM rate at a fatter shape
+    # standalone MFMA GEMM rate: square compute-bound shape (tiled path)
+    m, n, k = 4096, 4096, 4096
+    a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
+    bt = torch.randn(n, k, device="cuda").to(torch.bfloat16)
+
+    def run_gemm_sq():
+        tok.hip.gemm_bf16_nt(a, bt, None, False)
+
+    dt = timeit(run_gemm_sq)
+    print(json.dumps({
+        "kernel": "gemm_bf16_nt_tiled",
+        "shape": [m, n, k],
+        "ms": round(dt * 1e3, 3),
+        "TFLOPs": round(2 * m * n * k / dt / 1e12, 2),
+    }))
+
+    # skinny embed-projection-like shape
     m, n, k = 4096, 4096, 384
     a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
     bt = torch.randn(n, k, device="cuda").to(torch.bfloat16)

@@ -31,7 +31,9 @@ def test_mfma_probe_layout(hip):
 
 def test_gemm_bf16_nt(hip):
     torch.manual_seed(1)
-    for m, n, k in [(16, 64, 32), (33, 100, 384), (256, 384, 384), (7, 16, 64)]:
+    for m, n, k in [(16, 64, 32), (33, 100, 384), (256, 384, 384), (7, 16, 64),
+                    (128, 128, 32), (256, 128, 384), (384, 384, 384),
+                    (512, 256, 256)]:  # last four take the tiled LDS path
         a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
         bt = torch.randn(n, k, device="cuda").to(torch.bfloat16)
         c = hip.gemm_bf16_nt(a, bt, None, False)
