@@ -28,7 +28,6 @@ import time
 import uuid
 from typing import Optional
 
-import aiohttp
 from aiohttp import web
 
 from aigw import internalapi
