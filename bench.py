@@ -107,7 +107,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
     from aigw.filterapi import RuntimeConfig, load_config
     from aigw.testing.fastmock import canned_chat_response, start_fast_mock
 
-    use_gpu = torch.cuda.is_available()
+    use_gpu = torch.cuda.is_available() and not getattr(args, "no_gpu", False)
     if use_gpu:
         torch.cuda.set_device(local_rank)
 
@@ -189,6 +189,8 @@ def main():
     ap.add_argument("--tokens", type=int, default=4096)
     ap.add_argument("--workers", type=int, default=0,
                     help="HTTP worker processes per shard (0 = auto)")
+    ap.add_argument("--no-gpu", action="store_true",
+                    help="disable GPU token accounting (contention diagnosis)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
