@@ -144,6 +144,15 @@ class GPUServices:
         count, _ = await self._submit(text or b" ", want_vec=False)
         return count
 
+    async def count_texts_batch(self, texts: list[bytes]) -> list[int]:
+        """Batch entry used by the shard GPU service: the texts join the
+        SAME micro-batch queue, so batches from many workers coalesce into
+        one kernel launch."""
+        results = await asyncio.gather(
+            *(self._submit(t or b" ", want_vec=False) for t in texts)
+        )
+        return [c for c, _ in results]
+
     async def cache_lookup_text(self, text: bytes):
         """Like cache_lookup but takes pre-extracted text (from the C++
         scanner) instead of a parsed body."""

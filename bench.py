@@ -117,13 +117,13 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
     cfg = load_config(gateway_config(up_port))
 
     gpu_services = None
-    if use_gpu:
-        from aigw.gpu import GPUServices
+    if use_gpu and getattr(args, "gpu_socket", None):
+        # one GPU context per SHARD: workers RPC to the rank-primary's
+        # admission service (per-worker contexts thrash the device —
+        # measured collapse in profiles/r01 diag)
+        from aigw.gpu.service import RemoteGPUClient
 
-        gpu_services = GPUServices(
-            device=f"cuda:{local_rank}", n_merges=32768, enable_cache=False,
-            window_ms=0.5, max_batch=256,
-        )
+        gpu_services = RemoteGPUClient(args.gpu_socket, window_ms=0.5, max_batch=256)
 
     server = GatewayServer(RuntimeConfig(cfg), gpu_services=gpu_services)
     gw_runner = await run_server(server, host="127.0.0.1", port=0)
@@ -180,6 +180,35 @@ def worker_entry(args, local_rank, ready, go, out_q):
     asyncio.run(worker_main(args, local_rank, ready, go, out_q))
 
 
+def _start_gpu_host(socket_path: str, local_rank: int):
+    """Run the shard's GPU admission service on a dedicated thread/loop in
+    the rank-primary process (the only process owning a GPU context)."""
+    import threading
+
+    ready_evt = threading.Event()
+
+    def run():
+        torch.cuda.set_device(local_rank)
+        from aigw.gpu import GPUServices
+        from aigw.gpu.service import GPUServiceHost
+
+        async def amain():
+            gpu = GPUServices(device=f"cuda:{local_rank}", n_merges=32768,
+                              enable_cache=False, window_ms=0.5, max_batch=1024)
+            host = GPUServiceHost(gpu, socket_path)
+            await host.start()
+            ready_evt.set()
+            await asyncio.Event().wait()
+
+        asyncio.run(amain())
+
+    t = threading.Thread(target=run, daemon=True, name="aigw-gpu-host")
+    t.start()
+    if not ready_evt.wait(timeout=600):
+        raise RuntimeError("GPU admission service failed to start")
+    return t
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -211,6 +240,13 @@ def main():
         # (profiles/r01 bench_wsweep); ~3 cores per worker keeps headroom
         # when 8 ranks share the node
         workers = max(1, min(12, cores // (3 * max(world, 1))))
+
+    gpu_socket = None
+    gpu_host_state = None
+    if use_gpu:
+        gpu_socket = f"/tmp/aigw-gpu-{rank}-{os.getpid()}.sock"
+        gpu_host_state = _start_gpu_host(gpu_socket, local_rank)
+    args.gpu_socket = gpu_socket
 
     ctx = mp.get_context("spawn")
     ready_evts = [ctx.Event() for _ in range(workers)]
