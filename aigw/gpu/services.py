@@ -41,7 +41,11 @@ def extract_chat_text(body: dict) -> bytes:
 
 @dataclass
 class _Pending:
-    text: bytes
+    """One queued work item: a BATCH of texts sharing a future (single
+    requests are 1-element batches; the shard GPU service submits whole
+    worker batches so the per-text future overhead disappears)."""
+
+    texts: list
     want_vec: bool
     future: asyncio.Future = None  # type: ignore[assignment]
 
@@ -76,6 +80,7 @@ class GPUServices:
         self.window_ms = window_ms
         self.max_batch = max_batch
         self._pending: list[_Pending] = []
+        self._pending_texts = 0
         self._flush_handle = None
         self._executor = ThreadPoolExecutor(max_workers=1, thread_name_prefix="aigw-gpu")
         self._lock = asyncio.Lock()
@@ -86,11 +91,12 @@ class GPUServices:
 
     # ---- batching core -------------------------------------------------------
 
-    async def _submit(self, text: bytes, want_vec: bool):
+    async def _submit_batch(self, texts: list, want_vec: bool):
         loop = asyncio.get_running_loop()
-        item = _Pending(text=text, want_vec=want_vec, future=loop.create_future())
+        item = _Pending(texts=texts, want_vec=want_vec, future=loop.create_future())
         self._pending.append(item)
-        if len(self._pending) >= self.max_batch:
+        self._pending_texts += len(texts)
+        if self._pending_texts >= self.max_batch:
             if self._flush_handle:
                 self._flush_handle.cancel()
                 self._flush_handle = None
@@ -101,9 +107,14 @@ class GPUServices:
             )
         return await item.future
 
+    async def _submit(self, text: bytes, want_vec: bool):
+        out = await self._submit_batch([text], want_vec)
+        return out[0]
+
     async def _flush(self):
         self._flush_handle = None
         batch, self._pending = self._pending, []
+        self._pending_texts = 0
         if not batch:
             return
         loop = asyncio.get_running_loop()
@@ -119,20 +130,26 @@ class GPUServices:
                 it.future.set_result(res)
 
     def _run_batch(self, batch: list[_Pending]):
-        """Executed on the GPU worker thread: one packed tokenizer launch;
-        embedding GEMM only for the items that asked for a vector."""
-        texts = [it.text for it in batch]
+        """Executed on the GPU worker thread: one packed tokenizer launch
+        over every text of every queued batch; embedding GEMM only when
+        some batch asked for vectors. Returns one result list per item."""
+        texts = []
+        spans = []
+        for it in batch:
+            spans.append((len(texts), len(texts) + len(it.texts)))
+            texts.extend(it.texts)
         counts, _, state = self.tokenizer.encode_batch(texts)
         counts_host = counts.cpu().tolist()
-        vec_idx = [i for i, it in enumerate(batch) if it.want_vec]
-        vecs = {}
-        if vec_idx and self.cache is not None:
+        qvecs = None
+        if any(it.want_vec for it in batch) and self.cache is not None:
             qvecs = self.cache.embed(state["out_ids"], state["req_off"])
-            for i in vec_idx:
-                vecs[i] = qvecs[i]
-        return [
-            (counts_host[i], vecs.get(i)) for i in range(len(batch))
-        ]
+        results = []
+        for it, (lo, hi) in zip(batch, spans):
+            if it.want_vec and qvecs is not None:
+                results.append([(counts_host[i], qvecs[i]) for i in range(lo, hi)])
+            else:
+                results.append([(counts_host[i], None) for i in range(lo, hi)])
+        return results
 
     # ---- public API ----------------------------------------------------------
 
@@ -145,12 +162,9 @@ class GPUServices:
         return count
 
     async def count_texts_batch(self, texts: list[bytes]) -> list[int]:
-        """Batch entry used by the shard GPU service: the texts join the
-        SAME micro-batch queue, so batches from many workers coalesce into
-        one kernel launch."""
-        results = await asyncio.gather(
-            *(self._submit(t or b" ", want_vec=False) for t in texts)
-        )
+        """Batch entry used by the shard GPU service: one future per worker
+        batch; texts from many workers coalesce into one kernel launch."""
+        results = await self._submit_batch([t or b" " for t in texts], want_vec=False)
         return [c for c, _ in results]
 
     async def cache_lookup_text(self, text: bytes):
