@@ -119,11 +119,18 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
     gpu_services = None
     if use_gpu and getattr(args, "gpu_socket", None):
         # one GPU context per SHARD: workers RPC to the rank-primary's
-        # admission service (per-worker contexts thrash the device —
-        # measured collapse in profiles/r01 diag)
+        # admission service (scales past the ~16-worker direct-context
+        # knee; see --gpu-service help)
         from aigw.gpu.service import RemoteGPUClient
 
         gpu_services = RemoteGPUClient(args.gpu_socket, window_ms=0.5, max_batch=256)
+    elif use_gpu:
+        from aigw.gpu import GPUServices
+
+        gpu_services = GPUServices(
+            device=f"cuda:{local_rank}", n_merges=32768, enable_cache=False,
+            window_ms=0.5, max_batch=256,
+        )
 
     server = GatewayServer(RuntimeConfig(cfg), gpu_services=gpu_services)
     gw_runner = await run_server(server, host="127.0.0.1", port=0)
@@ -220,6 +227,13 @@ def main():
                     help="HTTP worker processes per shard (0 = auto)")
     ap.add_argument("--no-gpu", action="store_true",
                     help="disable GPU token accounting (contention diagnosis)")
+    ap.add_argument("--gpu-service", action="store_true",
+                    help="route GPU work through ONE per-shard admission "
+                         "service over IPC instead of per-worker GPU "
+                         "contexts. Wins above ~16 workers/shard (48k req/s "
+                         "@40w measured) at higher p50; direct contexts win "
+                         "at the ~12 workers/shard an 8-rank node affords "
+                         "(38k req/s, 5.8 ms added)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -243,7 +257,7 @@ def main():
 
     gpu_socket = None
     gpu_host_state = None
-    if use_gpu:
+    if use_gpu and args.gpu_service:
         gpu_socket = f"/tmp/aigw-gpu-{rank}-{os.getpid()}.sock"
         gpu_host_state = _start_gpu_host(gpu_socket, local_rank)
     args.gpu_socket = gpu_socket
