@@ -53,12 +53,28 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
     runtime = RuntimeConfig(cfg)
 
     gpu_services = None
+    gpu_host = None
     if args.gpu and torch.cuda.is_available():
-        from aigw.gpu import GPUServices
+        gpu_socket = getattr(args, "_gpu_socket", "")
+        if gpu_socket and getattr(args, "_is_worker", False):
+            # worker process: RPC to the shard primary's GPU admission
+            # service (one GPU context + one shared semantic cache per shard)
+            from aigw.gpu.service import RemoteGPUClient
 
-        gpu_services = GPUServices(
-            device=f"cuda:{rank}", enable_cache=args.semantic_cache,
-        )
+            gpu_services = RemoteGPUClient(
+                gpu_socket, enable_cache=args.semantic_cache
+            )
+        else:
+            from aigw.gpu import GPUServices
+
+            gpu_services = GPUServices(
+                device=f"cuda:{rank}", enable_cache=args.semantic_cache,
+            )
+            if gpu_socket:
+                from aigw.gpu.service import GPUServiceHost
+
+                gpu_host = GPUServiceHost(gpu_services, gpu_socket)
+                await gpu_host.start()
     server = GatewayServer(runtime, gpu_services=gpu_services)
 
     sync = None
@@ -88,6 +104,8 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
             await watcher.stop()
         if sync:
             await sync.stop()
+        if gpu_host is not None:
+            await gpu_host.stop()
         await runner.cleanup()
 
 
@@ -95,13 +113,16 @@ def cmd_run(args) -> int:
     world = int(os.environ.get("WORLD_SIZE", args.shards))
     rank = int(os.environ.get("RANK", 0))
     if args.workers > 1:
-        # worker processes share the listen port via SO_REUSEPORT; each has
-        # its own event loop + GPU services on the shard's GPU. Token-budget
+        # worker processes share the listen port via SO_REUSEPORT; GPU work
+        # funnels through the primary's per-shard admission service (one GPU
+        # context + one shared semantic cache per shard). Token-budget
         # buckets are per worker within a shard (cross-SHARD consistency is
         # the RCCL-synced path); divide limits by worker count when exact
         # in-shard budgets matter.
         import multiprocessing as _mp
 
+        if args.gpu:
+            args._gpu_socket = f"/tmp/aigw-gpu-{rank}-{os.getpid()}.sock"
         ctx = _mp.get_context("spawn")
         procs = [
             ctx.Process(target=_worker_run, args=(args, rank, world, i))
@@ -125,6 +146,10 @@ def cmd_run(args) -> int:
 
 
 def _worker_run(args, rank: int, world: int, worker_idx: int) -> None:
+    args._is_worker = True
+    import time as _t
+
+    _t.sleep(1.0)  # let the primary bind the GPU service socket first
     try:
         asyncio.run(_run_shard(args, rank, world))
     except KeyboardInterrupt:
