@@ -135,6 +135,11 @@ __device__ __forceinline__ uint64_t bpe_merge_lanes(
     int& tok, int myseg, uint64_t active, int lane,
     const long long* __restrict__ htab_keys, const int32_t* __restrict__ htab_rank,
     int htab_mask) {
+  // pair-rank memo: a merge round only changes the pairs adjacent to a
+  // merge site, so most lanes can reuse last round's hash-probe result
+  // (the probe's L2 loads dominate the VALU-bound loop otherwise)
+  long long cached_key = -1;
+  int cached_rank = INT_MAX;
   for (;;) {
     // next active lane above mine
     uint64_t above = (lane < 63) ? (active & (~0ull << (lane + 1))) : 0ull;
@@ -144,13 +149,19 @@ __device__ __forceinline__ uint64_t bpe_merge_lanes(
     int rank = INT_MAX;
     if (tok >= 0 && nxt >= 0 && nxtseg == myseg) {
       long long key = ((long long)tok << 32) | (unsigned)nxttok;
-      uint64_t h = (uint64_t)key * 0x9E3779B97F4A7C15ull;
-      int idx = (int)(h >> 40) & htab_mask;
-      for (;;) {
-        long long k = htab_keys[idx];
-        if (k == key) { rank = htab_rank[idx]; break; }
-        if (k == -1) break;
-        idx = (idx + 1) & htab_mask;
+      if (key == cached_key) {
+        rank = cached_rank;
+      } else {
+        uint64_t h = (uint64_t)key * 0x9E3779B97F4A7C15ull;
+        int idx = (int)(h >> 40) & htab_mask;
+        for (;;) {
+          long long k = htab_keys[idx];
+          if (k == key) { rank = htab_rank[idx]; break; }
+          if (k == -1) break;
+          idx = (idx + 1) & htab_mask;
+        }
+        cached_key = key;
+        cached_rank = rank;
       }
     }
     // wave min-reduce of rank

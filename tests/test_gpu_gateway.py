@@ -111,3 +111,52 @@ def test_gpu_batched_token_counting_concurrent():
         gpu.close()
 
     asyncio.run(main())
+
+
+def test_gpu_admission_service_path():
+    """Production topology on hardware: gateway worker -> unix-socket RPC ->
+    shard GPU service -> kernels (the aigw run --workers path)."""
+    from aigw.extproc.server import GatewayServer, run_server
+    from aigw.filterapi import RuntimeConfig, load_config
+    from aigw.gpu import GPUServices
+    from aigw.gpu.service import GPUServiceHost, RemoteGPUClient
+    from aigw.testing.mockupstream import start_mock_upstream
+
+    async def main():
+        import tempfile
+
+        sock = tempfile.mktemp(suffix=".sock")
+        mock, up_runner, up_port = await start_mock_upstream()
+        gpu = GPUServices(device="cuda", n_merges=8192, enable_cache=True,
+                          cache_threshold=0.95, window_ms=0.5)
+        host = GPUServiceHost(gpu, sock)
+        await host.start()
+        client = RemoteGPUClient(sock, enable_cache=True, window_ms=0.5)
+        server = GatewayServer(RuntimeConfig(load_config(_cfg(up_port))),
+                               gpu_services=client)
+        gw = await run_server(server, host="127.0.0.1", port=0)
+        port = gw.addresses[0][1]
+        payload = {"model": "m",
+                   "messages": [{"role": "user", "content": "service path test " * 30}]}
+        async with aiohttp.ClientSession() as c:
+            async with c.post(f"http://127.0.0.1:{port}/v1/chat/completions",
+                              json=payload) as r:
+                assert r.status == 200
+                body = await r.read()
+            # identical request -> cache hit SERVED FROM THE SHARD SERVICE
+            async with c.post(f"http://127.0.0.1:{port}/v1/chat/completions",
+                              json=payload) as r:
+                assert r.status == 200
+                assert r.headers.get("x-aigw-cache") == "hit"
+                assert await r.read() == body
+            # direct tokenize through the RPC
+            ids = await client.tokenize("hello world")
+            ref = gpu.tokenizer.reference().encode_batch([b"hello world"])[0]
+            assert ids == ref
+        client.close()
+        await host.stop()
+        await gw.cleanup()
+        await up_runner.cleanup()
+        gpu.close()
+
+    asyncio.run(main())
