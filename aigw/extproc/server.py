@@ -121,6 +121,63 @@ _MODEL_REQUIRED = frozenset(
 _PASSTHROUGH_SCHEMAS = frozenset({APISchemaName.OPENAI, APISchemaName.AZURE_OPENAI})
 
 
+class RequestView:
+    """Transport-agnostic request: the processing core never touches the
+    HTTP front's request object, so the same pipeline runs under the
+    aiohttp front (full surface) and the raw lean front (hot paths)."""
+
+    __slots__ = ("method", "path", "host", "remote", "headers", "body", "_stream_factory")
+
+    def __init__(self, method, path, host, remote, headers, body, stream_factory):
+        self.method = method
+        self.path = path
+        self.host = host
+        self.remote = remote
+        self.headers = headers  # lowercase dict, pre-spoof-strip
+        self.body = body
+        self._stream_factory = stream_factory
+
+    async def start_stream(self, status: int, headers: dict[str, str]):
+        """Begin a streamed response; returns a writer with
+        write(bytes)/finish()/result()."""
+        return await self._stream_factory(status, headers)
+
+
+class _AiohttpStreamWriter:
+    __slots__ = ("_resp",)
+
+    def __init__(self, resp: web.StreamResponse):
+        self._resp = resp
+
+    async def write(self, data: bytes) -> None:
+        await self._resp.write(data)
+
+    async def finish(self):
+        await self._resp.write_eof()
+
+    def result(self) -> web.StreamResponse:
+        return self._resp
+
+
+def _aiohttp_view(request: web.Request, body: bytes) -> RequestView:
+    async def stream_factory(status: int, headers: dict[str, str]):
+        resp = web.StreamResponse(status=status)
+        for k, v in headers.items():
+            resp.headers[k] = v
+        await resp.prepare(request)
+        return _AiohttpStreamWriter(resp)
+
+    return RequestView(
+        method=request.method,
+        path=request.path,
+        host=request.headers.get("host", request.host or ""),
+        remote=request.remote,
+        headers={k.lower(): v for k, v in request.headers.items()},
+        body=body,
+        stream_factory=stream_factory,
+    )
+
+
 def _json_error(status: int, message: str, err_type: str = "invalid_request_error") -> web.Response:
     return web.json_response(
         {"error": {"message": message, "type": err_type, "code": str(status)}},
@@ -306,7 +363,8 @@ class GatewayServer:
 
     def _make_handler(self, endpoint: str):
         async def handler(request: web.Request) -> web.StreamResponse:
-            return await self._process(request, endpoint)
+            body = await request.read()
+            return await self._process(_aiohttp_view(request, body), endpoint)
 
         return handler
 
@@ -317,8 +375,8 @@ class GatewayServer:
         return handler
 
     @staticmethod
-    def _ingress_headers(request: web.Request) -> dict[str, str]:
-        headers = {k.lower(): v for k, v in request.headers.items()}
+    def _ingress_headers(view) -> dict[str, str]:
+        headers = dict(view.headers)
         for h in internalapi.INTERNAL_HEADERS:
             headers.pop(h, None)  # spoof protection (server.go:439-455)
         return headers
@@ -330,10 +388,10 @@ class GatewayServer:
             kw["gcp_region"] = backend.auth.gcp_region
         return kw
 
-    async def _process(self, request: web.Request, endpoint: str) -> web.StreamResponse:
+    async def _process(self, request: RequestView, endpoint: str):
         start = time.monotonic()
         rt = self.runtime
-        raw = await request.read() or b"{}"
+        raw = request.body or b"{}"
         # Fast path: one C++ pass extracts model/stream/chat-text without
         # building Python objects; full json.loads happens lazily only when
         # a translator or mutation needs the dict.
@@ -729,11 +787,10 @@ class GatewayServer:
         content_type = hdrs.get("content-type") or upstream.headers.get(
             "content-type", "text/event-stream"
         )
-        resp = web.StreamResponse(status=upstream.status)
-        resp.content_type = content_type
-        resp.headers["cache-control"] = "no-cache"
         # content-length is dropped for streamed bodies (A.8)
-        await resp.prepare(request)
+        writer = await request.start_stream(
+            upstream.status, {"content-type": content_type, "cache-control": "no-cache"}
+        )
 
         usage = Usage()
         response_model = ""
@@ -747,14 +804,14 @@ class GatewayServer:
                 if rtl.body:
                     if ttft < 0:
                         ttft = time.monotonic() - start
-                    await resp.write(rtl.body)
+                    await writer.write(rtl.body)
                 if rtl.usage is not None:
                     usage.merge_max(rtl.usage)
                 if rtl.response_model:
                     response_model = rtl.response_model
             tail = translator.response_flush()
             if tail.body:
-                await resp.write(tail.body)
+                await writer.write(tail.body)
             if tail.usage is not None:
                 usage.merge_max(tail.usage)
         except (ConnectionResetError, asyncio.CancelledError):
@@ -773,8 +830,8 @@ class GatewayServer:
                 endpoint, route, backend, model, response_model, usage, start,
                 status=upstream.status, ttft=ttft if ttft >= 0 else 0.0,
             )
-        await resp.write_eof()
-        return resp
+        await writer.finish()
+        return writer.result()
 
     # ---- multipart (audio) ---------------------------------------------------
 
@@ -793,7 +850,9 @@ class GatewayServer:
         for name, _h, payload in parts:
             if name == "model":
                 model = payload.decode("utf-8", "replace")
-        headers = self._ingress_headers(request)
+        headers = {k.lower(): v for k, v in request.headers.items()}
+        for h in internalapi.INTERNAL_HEADERS:
+            headers.pop(h, None)
         headers[rt.model_header] = model
         route = rt.select_route(headers)
         if route is None:

@@ -103,7 +103,7 @@ async def fire_step(client, port, path, payload_bytes, batch, latencies):
 
 async def worker_main(args, local_rank: int, ready, go, out_q):
 
-    from aigw.extproc.server import GatewayServer, run_server
+    from aigw.extproc.server import GatewayServer
     from aigw.filterapi import RuntimeConfig, load_config
     from aigw.testing.fastmock import canned_chat_response, start_fast_mock
 
@@ -133,8 +133,11 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
         )
 
     server = GatewayServer(RuntimeConfig(cfg), gpu_services=gpu_services)
-    gw_runner = await run_server(server, host="127.0.0.1", port=0)
-    gw_port = gw_runner.addresses[0][1]
+    from aigw.extproc.lean_front import serve_lean
+
+    _, gw_port, gw_cleanup = await serve_lean(
+        server, "127.0.0.1", 0, with_fallback=False
+    )
 
     from aigw.extproc.upstream_client import LeanClient
 
@@ -176,7 +179,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
         }
     )
     await client.close()
-    await gw_runner.cleanup()
+    await gw_cleanup()
     up_server.close()
     await up_server.wait_closed()
     if gpu_services is not None:
