@@ -69,5 +69,20 @@ class GPUTokenizer:
                 ids.append([int(x) for x in seg[seg >= 0]])
         return req_counts, ids, {"out_ids": out_ids, "req_off": off_t, "n_bytes": len(arr)}
 
+    def encode_batch_async(self, texts: list[bytes]):
+        """Sync-free batch encode: returns (req_counts_gpu, out_ids_gpu,
+        req_off_gpu, event). Await the event (cooperatively, via
+        event.query()) before reading the tensors — no host sync happens
+        here, which matters because ROCm host syncs busy-spin a core."""
+        arr, offs = self.pack(texts)
+        bytes_t = torch.from_numpy(arr).to(self.device, non_blocking=True)
+        off_t = torch.from_numpy(offs).to(self.device, non_blocking=True)
+        out_ids, req_counts = self.hip.bpe_count_async(
+            bytes_t, off_t, self.htab_keys, self.htab_ranks
+        )
+        ev = torch.cuda.Event()
+        ev.record()
+        return req_counts, out_ids, off_t, ev
+
     def reference(self) -> BPERef:
         return BPERef(self.merges)

@@ -226,16 +226,13 @@ __device__ __forceinline__ int bpe_encode_one_segment(
 // segment's start is < 32 bytes, and the whole group is packed whenever
 // it ends within 64 bytes of the group start (long tails fall back to the
 // chunked path).
-__global__ void __launch_bounds__(256)
-bpe_encode_grouped_kernel(const uint8_t* __restrict__ bytes,
-                          const int32_t* __restrict__ seg_start,
-                          const int32_t* __restrict__ seg_req, int n_segs,
-                          int n_bytes,
-                          const int32_t* __restrict__ ghead, int n_groups,
-                          const long long* __restrict__ htab_keys,
-                          const int32_t* __restrict__ htab_rank, int htab_mask,
-                          int32_t* __restrict__ out_ids,  // n_bytes, -1 = gap
-                          int32_t* __restrict__ req_counts) {
+__device__ __forceinline__ void bpe_encode_grouped_body(
+    const uint8_t* __restrict__ bytes, const int32_t* __restrict__ seg_start,
+    const int32_t* __restrict__ seg_req, int n_segs, int n_bytes,
+    const int32_t* __restrict__ ghead, int n_groups,
+    const long long* __restrict__ htab_keys,
+    const int32_t* __restrict__ htab_rank, int htab_mask,
+    int32_t* __restrict__ out_ids, int32_t* __restrict__ req_counts) {
   int g = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   if (g >= n_groups) return;
   int lane = threadIdx.x & 63;
@@ -289,6 +286,43 @@ bpe_encode_grouped_kernel(const uint8_t* __restrict__ bytes,
   if (lane == 0 && written > 0) atomicAdd(&req_counts[seg_req[first]], written);
 }
 
+__global__ void __launch_bounds__(256)
+bpe_encode_grouped_kernel(const uint8_t* __restrict__ bytes,
+                          const int32_t* __restrict__ seg_start,
+                          const int32_t* __restrict__ seg_req, int n_segs,
+                          int n_bytes,
+                          const int32_t* __restrict__ ghead, int n_groups,
+                          const long long* __restrict__ htab_keys,
+                          const int32_t* __restrict__ htab_rank, int htab_mask,
+                          int32_t* __restrict__ out_ids,
+                          int32_t* __restrict__ req_counts) {
+  bpe_encode_grouped_body(bytes, seg_start, seg_req, n_segs, n_bytes, ghead,
+                          n_groups, htab_keys, htab_rank, htab_mask, out_ids,
+                          req_counts);
+}
+
+// Sync-free variant: segment/group totals live in DEVICE scalars (the scan
+// tails), so the host never calls .item() — the whole pipeline is launched
+// blind with upper-bound grids and each kernel self-bounds. This is what
+// lets the serving path await a hipEvent cooperatively instead of
+// busy-polling a host sync (ROCm host syncs spin a core).
+__global__ void __launch_bounds__(256)
+bpe_encode_grouped_dev_kernel(const uint8_t* __restrict__ bytes,
+                              const int32_t* __restrict__ seg_start,
+                              const int32_t* __restrict__ seg_req,
+                              const int32_t* __restrict__ n_segs_dev,
+                              int n_bytes,
+                              const int32_t* __restrict__ ghead,
+                              const int32_t* __restrict__ n_groups_dev,
+                              const long long* __restrict__ htab_keys,
+                              const int32_t* __restrict__ htab_rank, int htab_mask,
+                              int32_t* __restrict__ out_ids,
+                              int32_t* __restrict__ req_counts) {
+  bpe_encode_grouped_body(bytes, seg_start, seg_req, *n_segs_dev, n_bytes, ghead,
+                          *n_groups_dev, htab_keys, htab_rank, htab_mask, out_ids,
+                          req_counts);
+}
+
 // group-head flags over the segment array: a group breaks on request
 // change or a new 32-byte cell (relative to the request start)
 __global__ void group_head_flags_kernel(const int32_t* __restrict__ seg_start,
@@ -298,6 +332,23 @@ __global__ void group_head_flags_kernel(const int32_t* __restrict__ seg_start,
                                         uint8_t* __restrict__ flags) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n_segs) return;
+  uint8_t f;
+  if (i == 0 || seg_req[i] != seg_req[i - 1]) {
+    f = 1;
+  } else {
+    long long off = req_off[seg_req[i]];
+    f = ((seg_start[i] - off) >> 5) != ((seg_start[i - 1] - off) >> 5);
+  }
+  flags[i] = f;
+}
+
+__global__ void group_head_flags_dev_kernel(const int32_t* __restrict__ seg_start,
+                                            const int32_t* __restrict__ seg_req,
+                                            const int32_t* __restrict__ n_segs_dev,
+                                            const int64_t* __restrict__ req_off,
+                                            uint8_t* __restrict__ flags) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= *n_segs_dev) return;
   uint8_t f;
   if (i == 0 || seg_req[i] != seg_req[i - 1]) {
     f = 1;
@@ -774,6 +825,79 @@ std::vector<at::Tensor> bpe_encode(at::Tensor bytes, at::Tensor req_off,
   return {out_ids, req_counts, seg_start, seg_req};
 }
 
+// Fully async tokenize: ZERO host syncs. Segment/group totals stay in
+// device scalars (scan tails) and every kernel self-bounds; intermediate
+// arrays use upper bounds (#segments <= n bytes; #groups <= n/32 + n_req).
+// The caller awaits a hipEvent instead of a blocking sync — on ROCm a
+// blocking host sync busy-spins a core, which was measured to halve the
+// serving throughput with 12 gateway workers per GPU (profiles/r01).
+std::vector<at::Tensor> bpe_count_async(at::Tensor bytes, at::Tensor req_off,
+                                        at::Tensor htab_keys, at::Tensor htab_rank) {
+  check_cuda(bytes, "bytes");
+  check_cuda(req_off, "req_off");
+  int n = (int)bytes.numel();
+  int n_req = (int)req_off.numel();
+  TORCH_CHECK(n > 0, "empty batch");
+  int htab_mask = (int)htab_keys.numel() - 1;
+  auto stream = current_stream();
+  auto u8 = at::TensorOptions().dtype(at::kByte).device(bytes.device());
+  auto i32 = at::TensorOptions().dtype(at::kInt).device(bytes.device());
+  int blocks = (n + 255) / 256;
+
+  at::Tensor flags = at::empty({n}, u8);
+  hipLaunchKernelGGL(seg_flags_kernel, dim3(blocks), dim3(256), 0, stream,
+                     bytes.data_ptr<uint8_t>(), n, flags.data_ptr<uint8_t>());
+  hipLaunchKernelGGL(seg_force_starts_kernel, dim3((n_req + 255) / 256), dim3(256), 0,
+                     stream, req_off.data_ptr<int64_t>(), n_req,
+                     flags.data_ptr<uint8_t>());
+  at::Tensor blk_counts = at::empty({blocks}, i32);
+  hipLaunchKernelGGL(seg_block_count_kernel, dim3(blocks), dim3(256), 0, stream,
+                     flags.data_ptr<uint8_t>(), n, blk_counts.data_ptr<int32_t>());
+  at::Tensor scan = blk_counts.cumsum(0, at::kInt);
+  at::Tensor blk_excl = at::zeros({blocks}, i32);
+  if (blocks > 1)
+    blk_excl.narrow(0, 1, blocks - 1).copy_(scan.narrow(0, 0, blocks - 1));
+  const int32_t* n_segs_dev = scan.data_ptr<int32_t>() + (blocks - 1);
+
+  at::Tensor seg_start = at::empty({n}, i32);
+  at::Tensor seg_req = at::empty({n}, i32);
+  hipLaunchKernelGGL(seg_write_kernel, dim3(blocks), dim3(256), 0, stream,
+                     flags.data_ptr<uint8_t>(), n, blk_excl.data_ptr<int32_t>(),
+                     req_off.data_ptr<int64_t>(), n_req,
+                     seg_start.data_ptr<int32_t>(), seg_req.data_ptr<int32_t>());
+
+  at::Tensor gflags = at::zeros({n}, u8);  // beyond n_segs stays 0
+  hipLaunchKernelGGL(group_head_flags_dev_kernel, dim3(blocks), dim3(256), 0, stream,
+                     seg_start.data_ptr<int32_t>(), seg_req.data_ptr<int32_t>(),
+                     n_segs_dev, req_off.data_ptr<int64_t>(),
+                     gflags.data_ptr<uint8_t>());
+  at::Tensor gblk = at::empty({blocks}, i32);
+  hipLaunchKernelGGL(seg_block_count_kernel, dim3(blocks), dim3(256), 0, stream,
+                     gflags.data_ptr<uint8_t>(), n, gblk.data_ptr<int32_t>());
+  at::Tensor gscan = gblk.cumsum(0, at::kInt);
+  at::Tensor gexcl = at::zeros({blocks}, i32);
+  if (blocks > 1)
+    gexcl.narrow(0, 1, blocks - 1).copy_(gscan.narrow(0, 0, blocks - 1));
+  const int32_t* n_groups_dev = gscan.data_ptr<int32_t>() + (blocks - 1);
+  at::Tensor ghead = at::empty({n}, i32);
+  hipLaunchKernelGGL(flag_compact_write_kernel, dim3(blocks), dim3(256), 0, stream,
+                     gflags.data_ptr<uint8_t>(), n, gexcl.data_ptr<int32_t>(),
+                     ghead.data_ptr<int32_t>());
+
+  at::Tensor out_ids = at::full({n}, -1, i32);
+  at::Tensor req_counts = at::zeros({n_req}, i32);
+  long long group_bound = (long long)n / 32 + n_req + 1;
+  int blocks2 = (int)((group_bound + 3) / 4);
+  hipLaunchKernelGGL(bpe_encode_grouped_dev_kernel, dim3(blocks2), dim3(256), 0,
+                     stream, bytes.data_ptr<uint8_t>(), seg_start.data_ptr<int32_t>(),
+                     seg_req.data_ptr<int32_t>(), n_segs_dev, n,
+                     ghead.data_ptr<int32_t>(), n_groups_dev,
+                     reinterpret_cast<long long*>(htab_keys.data_ptr<int64_t>()),
+                     htab_rank.data_ptr<int32_t>(), htab_mask,
+                     out_ids.data_ptr<int32_t>(), req_counts.data_ptr<int32_t>());
+  return {out_ids, req_counts};
+}
+
 at::Tensor meanpool(at::Tensor ids, at::Tensor req_off, at::Tensor emb) {
   check_cuda(ids, "ids");
   check_cuda(req_off, "req_off");
@@ -934,6 +1058,8 @@ at::Tensor mfma_probe(at::Tensor a, at::Tensor bt) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bpe_segment", &bpe_segment, "segment bytes (GPU)");
   m.def("bpe_encode", &bpe_encode, "BPE encode a packed byte batch (GPU)");
+  m.def("bpe_count_async", &bpe_count_async,
+        "sync-free BPE encode: returns (out_ids, req_counts) without host syncs");
   m.def("meanpool", &meanpool, "mean-pool token embeddings per request");
   m.def("gemm_bf16_nt", &gemm_bf16_nt, "C = A @ Bt^T (MFMA bf16)",
         py::arg("a"), py::arg("bt"), py::arg("bias") = py::none(),

@@ -95,6 +95,26 @@ def test_bpe_encode_matches_reference(hip):
     assert counts.cpu().tolist() == [len(w) for w in expect]
 
 
+def test_bpe_count_async_matches_sync(hip):
+    import torch
+
+    from aigw.ops.tokenizer import GPUTokenizer
+
+    tok = GPUTokenizer(n_merges=8192, device="cuda")
+    texts = [b"the quick brown fox " * 50, b"hello", b"42! punctuation, here."]
+    counts_sync, ids_sync, _ = tok.encode_batch(texts, return_ids=True)
+    counts_gpu, out_ids, off_t, ev = tok.encode_batch_async(texts)
+    while not ev.query():
+        pass
+    assert counts_gpu.cpu().tolist() == counts_sync.cpu().tolist()
+    # ids identical too (same kernels, device-bounded grids)
+    flat = out_ids.cpu().numpy()
+    offs = off_t.cpu().tolist() + [sum(len(t) for t in texts)]
+    for i in range(len(texts)):
+        seg = flat[offs[i] : offs[i + 1]]
+        assert [int(x) for x in seg[seg >= 0]] == ids_sync[i]
+
+
 def test_meanpool_matches_torch(hip):
     from aigw.ops.tokenizer import GPUTokenizer
 
