@@ -79,6 +79,10 @@ class GPUServices:
         )
         self.window_ms = window_ms
         self.max_batch = max_batch
+        import os as _os
+
+        # A/B toggle for the sync-free count path (profiles/r01 study)
+        self.async_counts = _os.environ.get("AIGW_GPU_SYNC_PATH", "") != "1"
         self._pending: list[_Pending] = []
         self._pending_texts = 0
         self._flush_handle = None
@@ -118,7 +122,7 @@ class GPUServices:
         if not batch:
             return
         loop = asyncio.get_running_loop()
-        if not any(it.want_vec for it in batch):
+        if self.async_counts and not any(it.want_vec for it in batch):
             # count-only fast path: launch without ANY host sync and await
             # the completion event cooperatively — a blocking torch sync on
             # ROCm busy-spins a core per worker, which halved 12-worker

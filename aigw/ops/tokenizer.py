@@ -34,6 +34,9 @@ class GPUTokenizer:
         self.htab_keys = torch.from_numpy(keys).to(self.device)
         self.htab_ranks = torch.from_numpy(ranks).to(self.device)
         self.vocab_size = 256 + len(self.merges)
+        # pinned staging buffers for truly-async H2D (pageable copies block)
+        self._pin_bytes = None
+        self._pin_offs = None
 
     @staticmethod
     def pack(texts: list[bytes]) -> tuple[np.ndarray, np.ndarray]:
@@ -75,8 +78,17 @@ class GPUTokenizer:
         event.query()) before reading the tensors — no host sync happens
         here, which matters because ROCm host syncs busy-spin a core."""
         arr, offs = self.pack(texts)
-        bytes_t = torch.from_numpy(arr).to(self.device, non_blocking=True)
-        off_t = torch.from_numpy(offs).to(self.device, non_blocking=True)
+        n = len(arr)
+        if self._pin_bytes is None or self._pin_bytes.numel() < n:
+            self._pin_bytes = torch.empty(max(n, 1 << 20), dtype=torch.uint8,
+                                          pin_memory=True)
+        if self._pin_offs is None or self._pin_offs.numel() < len(offs):
+            self._pin_offs = torch.empty(max(len(offs), 4096), dtype=torch.int64,
+                                         pin_memory=True)
+        self._pin_bytes.numpy()[:n] = arr
+        self._pin_offs.numpy()[: len(offs)] = offs
+        bytes_t = self._pin_bytes[:n].to(self.device, non_blocking=True)
+        off_t = self._pin_offs[: len(offs)].to(self.device, non_blocking=True)
         out_ids, req_counts = self.hip.bpe_count_async(
             bytes_t, off_t, self.htab_keys, self.htab_ranks
         )
