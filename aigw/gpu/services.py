@@ -81,8 +81,12 @@ class GPUServices:
         self.max_batch = max_batch
         import os as _os
 
-        # A/B toggle for the sync-free count path (profiles/r01 study)
-        self.async_counts = _os.environ.get("AIGW_GPU_SYNC_PATH", "") != "1"
+        # Measured NEGATIVE (profiles/r01 ab.jsonl): the sync-free
+        # event-polling count path is ~20% slower at 12 workers — executor
+        # threads may freely spin on this many-core node, while 0.2 ms
+        # event polling adds per-batch latency. Kept behind an opt-in for
+        # core-constrained deployments.
+        self.async_counts = _os.environ.get("AIGW_GPU_ASYNC_PATH", "") == "1"
         self._pending: list[_Pending] = []
         self._pending_texts = 0
         self._flush_handle = None
