@@ -615,6 +615,13 @@ class GatewayServer:
                     backend=backend.name,
                 )
             st = self._ep_stats.setdefault(backend.name, [0, 0.0])
+            if backend.max_concurrency and st[0] >= backend.max_concurrency:
+                # circuit open: saturated backend = failed attempt
+                last_error = f"backend {backend.name} saturated ({st[0]} in flight)"
+                self.metrics.requests_total.labels(
+                    endpoint=endpoint, backend=backend.name, status="circuit_open"
+                ).inc()
+                continue
             st[0] += 1
             st[1] += gpu_input_tokens
             try:

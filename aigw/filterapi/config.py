@@ -160,6 +160,10 @@ class Backend:
     body_mutation: Optional[BodyMutation] = None
     # Per-try timeout seconds (extensionserver/post_translate_modify.go:206-300)
     timeout_s: float = 60.0
+    # Circuit breaking: max in-flight requests to this backend (parity with
+    # the Envoy cluster circuit breakers the reference relies on; 0 = off).
+    # A saturated backend counts as a failed attempt and fallback proceeds.
+    max_concurrency: int = 0
 
 
 @dataclass

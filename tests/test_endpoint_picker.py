@@ -126,3 +126,54 @@ def test_picker_fallback_on_picked_failure():
         await up_runner.cleanup()
 
     asyncio.run(main())
+
+
+def test_circuit_breaker_spills_to_fallback():
+    """Backend.maxConcurrency opens the circuit and fallback takes over
+    (Envoy cluster circuit-breaker parity)."""
+    from aigw.testing.mockupstream import start_mock_upstream
+
+    async def main():
+        mock, up_runner, up_port = await start_mock_upstream()
+        mock.record = True
+        up = {"host": "127.0.0.1", "port": up_port}
+        cfg = load_config(
+            {
+                "version": "v1",
+                "routes": [
+                    {
+                        "name": "cb",
+                        "retries": 2,
+                        "backends": [
+                            {"name": "tiny", "schema": "OpenAI", "upstream": up,
+                             "maxConcurrency": 1,
+                             "headerMutation": {"set": {"x-replica": "tiny",
+                                                        "x-mock-delay-ms": "300"}}},
+                            {"name": "spill", "schema": "OpenAI", "upstream": up,
+                             "priority": 1,
+                             "headerMutation": {"set": {"x-replica": "spill"}}},
+                        ],
+                    }
+                ],
+            }
+        )
+        server = GatewayServer(RuntimeConfig(cfg))
+        gw = await run_server(server, host="127.0.0.1", port=0)
+        port = gw.addresses[0][1]
+
+        async def one(c):
+            async with c.post(
+                f"http://127.0.0.1:{port}/v1/chat/completions",
+                json={"model": "m", "messages": [{"role": "user", "content": "q"}]},
+            ) as r:
+                assert r.status == 200
+
+        async with aiohttp.ClientSession() as c:
+            await asyncio.gather(*(one(c) for _ in range(4)))
+        replicas = [r["headers"].get("x-replica") for r in mock.requests]
+        assert replicas.count("tiny") >= 1
+        assert replicas.count("spill") >= 1  # overflow spilled to fallback
+        await gw.cleanup()
+        await up_runner.cleanup()
+
+    asyncio.run(main())
