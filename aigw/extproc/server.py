@@ -448,20 +448,34 @@ class GatewayServer:
                 chat_text = extract_chat_text(body)
             gpu_input_tokens = await self.gpu.count_text_tokens(chat_text)
 
-        # semantic response cache (GPU MFMA embed + HBM index)
+        # semantic response cache (GPU MFMA embed + HBM index). Streamed
+        # requests are cached too: the translated SSE transcript is stored
+        # on miss and replayed chunk-wise on hit. Stream/unary entries are
+        # disjoint (key prefix) so a cached transcript never answers a
+        # unary request and vice versa.
         cache_key_vec = None
         if (
             self.gpu is not None
             and self.gpu.cache_enabled
             and endpoint == "/v1/chat/completions"
-            and not stream
         ):
-            hit, cache_key_vec = await self.gpu.cache_lookup_text(chat_text or b" ")
+            cache_key = (b"s:" if stream else b"u:") + (chat_text or b" ")
+            hit, cache_key_vec = await self.gpu.cache_lookup_text(cache_key)
             if hit is not None:
                 self.metrics.cache_events.labels(event="hit").inc()
-                resp = web.Response(body=hit, content_type="application/json")
-                resp.headers["x-aigw-cache"] = "hit"
-                return resp
+                if not stream:
+                    resp = web.Response(body=hit, content_type="application/json")
+                    resp.headers["x-aigw-cache"] = "hit"
+                    return resp
+                writer = await request.start_stream(
+                    200,
+                    {"content-type": "text/event-stream",
+                     "cache-control": "no-cache", "x-aigw-cache": "hit"},
+                )
+                for off in range(0, len(hit), 16384):
+                    await writer.write(hit[off : off + 16384])
+                await writer.finish()
+                return writer.result()
             self.metrics.cache_events.labels(event="miss").inc()
 
         span = None
@@ -655,6 +669,7 @@ class GatewayServer:
                     return await self._stream_response(
                         request, endpoint, route, backend, translator, upstream,
                         headers, model, start, gpu_input_tokens, span=span,
+                        cache_key_vec=cache_key_vec,
                     )
                 return await self._unary_response(
                     endpoint, route, backend, translator, upstream, headers,
@@ -788,8 +803,9 @@ class GatewayServer:
 
     async def _stream_response(
         self, request, endpoint, route, backend, translator, upstream,
-        headers, model, start, gpu_input_tokens, span=None,
+        headers, model, start, gpu_input_tokens, span=None, cache_key_vec=None,
     ) -> web.StreamResponse:
+        transcript = bytearray() if cache_key_vec is not None else None
         hdrs = translator.response_headers(upstream.status, upstream.headers)
         content_type = hdrs.get("content-type") or upstream.headers.get(
             "content-type", "text/event-stream"
@@ -812,6 +828,8 @@ class GatewayServer:
                     if ttft < 0:
                         ttft = time.monotonic() - start
                     await writer.write(rtl.body)
+                    if transcript is not None and len(transcript) < (2 << 20):
+                        transcript.extend(rtl.body)
                 if rtl.usage is not None:
                     usage.merge_max(rtl.usage)
                 if rtl.response_model:
@@ -819,8 +837,16 @@ class GatewayServer:
             tail = translator.response_flush()
             if tail.body:
                 await writer.write(tail.body)
+                if transcript is not None:
+                    transcript.extend(tail.body)
             if tail.usage is not None:
                 usage.merge_max(tail.usage)
+            if (
+                transcript is not None
+                and upstream.status == 200
+                and len(transcript) < (2 << 20)
+            ):
+                await self.gpu.cache_insert(cache_key_vec, bytes(transcript))
         except (ConnectionResetError, asyncio.CancelledError):
             aborted = True
             logger.info("client disconnected mid-stream")

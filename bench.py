@@ -129,7 +129,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
 
         gpu_services = GPUServices(
             device=f"cuda:{local_rank}", n_merges=32768, enable_cache=False,
-            window_ms=0.5, max_batch=256,
+            window_ms=getattr(args, "gpu_window", 0.5), max_batch=256,
         )
 
     server = GatewayServer(RuntimeConfig(cfg), gpu_services=gpu_services)
@@ -230,6 +230,8 @@ def main():
                     help="HTTP worker processes per shard (0 = auto)")
     ap.add_argument("--no-gpu", action="store_true",
                     help="disable GPU token accounting (contention diagnosis)")
+    ap.add_argument("--gpu-window", type=float, default=0.5,
+                    help="GPU micro-batch window per worker, ms")
     ap.add_argument("--gpu-service", action="store_true",
                     help="route GPU work through ONE per-shard admission "
                          "service over IPC instead of per-worker GPU "

@@ -160,3 +160,49 @@ def test_gpu_admission_service_path():
         gpu.close()
 
     asyncio.run(main())
+
+
+def test_gpu_streaming_cache_replay():
+    """Streamed requests hit the semantic cache too: the translated SSE
+    transcript is stored on miss and replayed on hit."""
+    from aigw.extproc.server import GatewayServer, run_server
+    from aigw.filterapi import RuntimeConfig, load_config
+    from aigw.gpu import GPUServices
+    from aigw.testing.mockupstream import start_mock_upstream
+
+    async def main():
+        mock, up_runner, up_port = await start_mock_upstream()
+        gpu = GPUServices(device="cuda", n_merges=8192, enable_cache=True,
+                          cache_threshold=0.95, window_ms=0.5)
+        server = GatewayServer(RuntimeConfig(load_config(_cfg(up_port))),
+                               gpu_services=gpu)
+        gw = await run_server(server, host="127.0.0.1", port=0)
+        port = gw.addresses[0][1]
+        payload = {"model": "m", "stream": True,
+                   "messages": [{"role": "user", "content": "stream cache test " * 25}]}
+        async with aiohttp.ClientSession() as c:
+            async with c.post(f"http://127.0.0.1:{port}/v1/chat/completions",
+                              json=payload,
+                              headers={"x-mock-response-tokens": "6"}) as r:
+                assert r.status == 200
+                first = await r.read()
+                assert first.endswith(b"data: [DONE]\n\n")
+            n_up = len(mock.requests)
+            async with c.post(f"http://127.0.0.1:{port}/v1/chat/completions",
+                              json=payload) as r:
+                assert r.status == 200
+                assert r.headers.get("x-aigw-cache") == "hit"
+                assert await r.read() == first
+            assert len(mock.requests) == n_up
+            # the UNARY form of the same prompt must NOT hit the stream entry
+            unary = dict(payload)
+            unary.pop("stream")
+            async with c.post(f"http://127.0.0.1:{port}/v1/chat/completions",
+                              json=unary) as r:
+                assert r.status == 200
+                assert "x-aigw-cache" not in r.headers
+        await gw.cleanup()
+        await up_runner.cleanup()
+        gpu.close()
+
+    asyncio.run(main())
