@@ -450,21 +450,23 @@ class GatewayServer:
 
         # semantic response cache (GPU MFMA embed + HBM index). Streamed
         # requests are cached too: the translated SSE transcript is stored
-        # on miss and replayed chunk-wise on hit. Stream/unary entries are
-        # disjoint (key prefix) so a cached transcript never answers a
-        # unary request and vice versa.
+        # on miss and replayed chunk-wise on hit. Mode lives in a value tag
+        # (a key prefix cannot separate stream/unary in embedding space —
+        # mean-pooled vectors of near-identical texts coincide), so a
+        # cached transcript never answers a unary request or vice versa.
         cache_key_vec = None
+        cache_tag = b"S" if stream else b"U"
         if (
             self.gpu is not None
             and self.gpu.cache_enabled
             and endpoint == "/v1/chat/completions"
         ):
-            cache_key = (b"s:" if stream else b"u:") + (chat_text or b" ")
-            hit, cache_key_vec = await self.gpu.cache_lookup_text(cache_key)
-            if hit is not None:
+            hit, cache_key_vec = await self.gpu.cache_lookup_text(chat_text or b" ")
+            if hit is not None and hit[:1] == cache_tag:
                 self.metrics.cache_events.labels(event="hit").inc()
+                body_bytes = hit[1:]
                 if not stream:
-                    resp = web.Response(body=hit, content_type="application/json")
+                    resp = web.Response(body=body_bytes, content_type="application/json")
                     resp.headers["x-aigw-cache"] = "hit"
                     return resp
                 writer = await request.start_stream(
@@ -472,8 +474,8 @@ class GatewayServer:
                     {"content-type": "text/event-stream",
                      "cache-control": "no-cache", "x-aigw-cache": "hit"},
                 )
-                for off in range(0, len(hit), 16384):
-                    await writer.write(hit[off : off + 16384])
+                for off in range(0, len(body_bytes), 16384):
+                    await writer.write(body_bytes[off : off + 16384])
                 await writer.finish()
                 return writer.result()
             self.metrics.cache_events.labels(event="miss").inc()
@@ -798,7 +800,7 @@ class GatewayServer:
         resp.headers["content-type"] = content_type
         resp.headers["x-request-id"] = headers.get("x-request-id", str(uuid.uuid4()))
         if cache_key_vec is not None and upstream.status == 200:
-            await self.gpu.cache_insert(cache_key_vec, rtl.body)
+            await self.gpu.cache_insert(cache_key_vec, b"U" + rtl.body)
         return resp
 
     async def _stream_response(
@@ -846,7 +848,7 @@ class GatewayServer:
                 and upstream.status == 200
                 and len(transcript) < (2 << 20)
             ):
-                await self.gpu.cache_insert(cache_key_vec, bytes(transcript))
+                await self.gpu.cache_insert(cache_key_vec, b"S" + bytes(transcript))
         except (ConnectionResetError, asyncio.CancelledError):
             aborted = True
             logger.info("client disconnected mid-stream")
