@@ -89,9 +89,12 @@ class GPUServiceHost:
             if op == "count_batch":
                 out["counts"] = await self.gpu.count_texts_batch(msg["texts"])
             elif op == "lookup_batch":
+                # ONE batched submit for the whole RPC (a sequential
+                # per-text await would serialize a micro-batch window per
+                # text)
+                pairs = await self.gpu.lookup_texts_batch(msg["texts"])
                 hits, handles = [], []
-                for text in msg["texts"]:
-                    hit, vec = await self.gpu.cache_lookup_text(text)
+                for hit, vec in pairs:
                     hits.append(hit)
                     if vec is not None and hit is None:
                         h = next(self._handles)

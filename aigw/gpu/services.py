@@ -117,7 +117,7 @@ class GPUServices:
 
     async def _submit(self, text: bytes, want_vec: bool):
         out = await self._submit_batch([text], want_vec)
-        return out[0]
+        return out[0]  # (count, vec, cached_value)
 
     async def _flush(self):
         self._flush_handle = None
@@ -147,7 +147,7 @@ class GPUServices:
             for it, (lo, hi) in zip(batch, spans):
                 if not it.future.done():
                     it.future.set_result(
-                        [(counts_host[i], None) for i in range(lo, hi)]
+                        [(counts_host[i], None, None) for i in range(lo, hi)]
                     )
             return
         try:
@@ -173,8 +173,10 @@ class GPUServices:
 
     def _run_batch(self, batch: list[_Pending]):
         """Executed on the GPU worker thread: one packed tokenizer launch
-        over every text of every queued batch; embedding GEMM only when
-        some batch asked for vectors. Returns one result list per item."""
+        over every text of every queued batch; for vector-wanting items,
+        ONE batched embedding GEMM and ONE fused index lookup (a lookup per
+        request would mean a single-query kernel launch each). Returns one
+        result list per item of (count, vec, cached_value_or_None)."""
         texts = []
         spans = []
         for it in batch:
@@ -183,47 +185,49 @@ class GPUServices:
         counts, _, state = self.tokenizer.encode_batch(texts)
         counts_host = counts.cpu().tolist()
         qvecs = None
+        hits = None
         if any(it.want_vec for it in batch) and self.cache is not None:
             qvecs = self.cache.embed(state["out_ids"], state["req_off"])
+            found = self.cache.lookup(qvecs)  # one fused kernel, all queries
+            hits = [self.cache.get(h[0]) if h is not None else None for h in found]
         results = []
         for it, (lo, hi) in zip(batch, spans):
             if it.want_vec and qvecs is not None:
-                results.append([(counts_host[i], qvecs[i]) for i in range(lo, hi)])
+                results.append(
+                    [(counts_host[i], qvecs[i], hits[i]) for i in range(lo, hi)]
+                )
             else:
-                results.append([(counts_host[i], None) for i in range(lo, hi)])
+                results.append([(counts_host[i], None, None) for i in range(lo, hi)])
         return results
 
     # ---- public API ----------------------------------------------------------
 
     async def count_request_tokens(self, body: dict) -> int:
-        count, _ = await self._submit(extract_chat_text(body), want_vec=False)
+        count, _, _ = await self._submit(extract_chat_text(body), want_vec=False)
         return count
 
     async def count_text_tokens(self, text: bytes) -> int:
-        count, _ = await self._submit(text or b" ", want_vec=False)
+        count, _, _ = await self._submit(text or b" ", want_vec=False)
         return count
 
     async def count_texts_batch(self, texts: list[bytes]) -> list[int]:
         """Batch entry used by the shard GPU service: one future per worker
         batch; texts from many workers coalesce into one kernel launch."""
         results = await self._submit_batch([t or b" " for t in texts], want_vec=False)
-        return [c for c, _ in results]
+        return [r[0] for r in results]
 
     async def cache_lookup_text(self, text: bytes):
         """Like cache_lookup but takes pre-extracted text (from the C++
-        scanner) instead of a parsed body."""
-        _, vec = await self._submit(text or b" ", want_vec=True)
-        if vec is None:
-            return None, None
-        loop = asyncio.get_running_loop()
+        scanner) instead of a parsed body. Lookup is part of the batched
+        flush — no per-request kernel launches."""
+        _, vec, hit = await self._submit(text or b" ", want_vec=True)
+        return hit, vec
 
-        def run():
-            hits = self.cache.lookup(vec.unsqueeze(0))
-            if hits[0] is None:
-                return None
-            return self.cache.get(hits[0][0])
-
-        return await loop.run_in_executor(self._executor, run), vec
+    async def lookup_texts_batch(self, texts: list[bytes]):
+        """Batch entry for the shard GPU service: one future per worker
+        batch; returns [(hit_or_None, vec)] aligned with texts."""
+        results = await self._submit_batch([t or b" " for t in texts], want_vec=True)
+        return [(hit, vec) for _c, vec, hit in results]
 
     async def tokenize(self, text) -> list[int]:
         if isinstance(text, str):
@@ -238,18 +242,8 @@ class GPUServices:
 
     async def cache_lookup(self, body: dict):
         """Returns (cached_response_bytes | None, query_vec)."""
-        _, vec = await self._submit(extract_chat_text(body), want_vec=True)
-        if vec is None:
-            return None, None
-        loop = asyncio.get_running_loop()
-
-        def run():
-            hits = self.cache.lookup(vec.unsqueeze(0))
-            if hits[0] is None:
-                return None
-            return self.cache.get(hits[0][0])
-
-        return await loop.run_in_executor(self._executor, run), vec
+        _, vec, hit = await self._submit(extract_chat_text(body), want_vec=True)
+        return hit, vec
 
     async def cache_insert(self, vec, response: bytes) -> None:
         if vec is None or self.cache is None:

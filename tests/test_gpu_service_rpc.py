@@ -22,6 +22,9 @@ class FakeGPU:
             return self.cache[text], None
         return None, ("vec", text)
 
+    async def lookup_texts_batch(self, texts):
+        return [await self.cache_lookup_text(t) for t in texts]
+
     async def cache_insert(self, vec, response):
         _, text = vec
         self.cache[text] = response
