@@ -83,22 +83,29 @@ def gateway_config(upstream_port: int) -> dict:
     }
 
 
-async def fire_step(client, port, path, payload_bytes, batch, latencies):
-    """One step: `batch` concurrent POSTs via the lean pooled client."""
+async def fire_step(client, port, path, payloads, batch, latencies, counter=None):
+    """One step: `batch` concurrent POSTs via the lean pooled client.
+    ``payloads`` is one bytes object or a pool cycled via ``counter``
+    (the semantic-cache mode reuses a payload pool to produce hits)."""
 
-    async def one():
+    pool = payloads if isinstance(payloads, list) else [payloads]
+
+    async def one(i):
+        body = pool[(counter[0] + i) % len(pool)] if counter else pool[0]
         t0 = time.perf_counter()
         r = await client.post(
             host="127.0.0.1", port=port, tls=False, path=path,
             headers={"content-type": "application/json"},
-            body=payload_bytes, timeout_s=120.0,
+            body=body, timeout_s=120.0,
         )
         await r.read()
         r.release()
         assert r.status == 200, f"status {r.status}"
         latencies.append((time.perf_counter() - t0) * 1000.0)
 
-    await asyncio.gather(*(one() for _ in range(batch)))
+    await asyncio.gather(*(one(i) for i in range(batch)))
+    if counter:
+        counter[0] += batch
 
 
 async def worker_main(args, local_rank: int, ready, go, out_q):
@@ -116,6 +123,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
     )
     cfg = load_config(gateway_config(up_port))
 
+    cache_mode = getattr(args, "cache_payloads", 0) > 0
     gpu_services = None
     if use_gpu and getattr(args, "gpu_socket", None):
         # one GPU context per SHARD: workers RPC to the rank-primary's
@@ -128,8 +136,9 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
         from aigw.gpu import GPUServices
 
         gpu_services = GPUServices(
-            device=f"cuda:{local_rank}", n_merges=32768, enable_cache=False,
-            window_ms=getattr(args, "gpu_window", 0.5), max_batch=256,
+            device=f"cuda:{local_rank}", n_merges=32768, enable_cache=cache_mode,
+            cache_threshold=0.98, window_ms=getattr(args, "gpu_window", 0.5),
+            max_batch=256,
         )
 
     server = GatewayServer(RuntimeConfig(cfg), gpu_services=gpu_services)
@@ -141,19 +150,28 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
 
     from aigw.extproc.upstream_client import LeanClient
 
-    payload = json.dumps(build_payload(args.tokens)).encode()
+    if cache_mode:
+        # distinct payload pool -> steady-state hit ratio 1 - D/total
+        payloads = []
+        for i in range(args.cache_payloads):
+            pl = build_payload(args.tokens)
+            pl["messages"][1]["content"] = f"variant {i}: " + pl["messages"][1]["content"]
+            payloads.append(json.dumps(pl).encode())
+    else:
+        payloads = [json.dumps(build_payload(args.tokens)).encode()]
     client = LeanClient()
     path = "/v1/chat/completions"
+    counter = [0]
 
     # warm both paths, then record the warm direct-to-upstream baseline
     scratch: list[float] = []
     for _ in range(max(args.warmup, 1)):
-        await fire_step(client, up_port, path, payload, args.batch, scratch)
+        await fire_step(client, up_port, path, payloads, args.batch, scratch)
     for _ in range(args.warmup):
-        await fire_step(client, gw_port, path, payload, args.batch, scratch)
+        await fire_step(client, gw_port, path, payloads, args.batch, scratch, counter)
     direct_lat: list[float] = []
     for _ in range(2):
-        await fire_step(client, up_port, path, payload, args.batch, direct_lat)
+        await fire_step(client, up_port, path, payloads, args.batch, direct_lat)
     if use_gpu:
         torch.cuda.synchronize()
 
@@ -164,7 +182,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
     lat: list[float] = []
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        await fire_step(client, gw_port, path, payload, args.batch, lat)
+        await fire_step(client, gw_port, path, payloads, args.batch, lat, counter)
     if use_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
@@ -230,6 +248,10 @@ def main():
                     help="HTTP worker processes per shard (0 = auto)")
     ap.add_argument("--no-gpu", action="store_true",
                     help="disable GPU token accounting (contention diagnosis)")
+    ap.add_argument("--cache-payloads", type=int, default=0,
+                    help="semantic-cache mode: cycle this many distinct "
+                         "payloads per worker (0 = cache off); steady-state "
+                         "hit ratio approaches 1")
     ap.add_argument("--gpu-window", type=float, default=0.5,
                     help="GPU micro-batch window per worker, ms")
     ap.add_argument("--gpu-service", action="store_true",
@@ -342,6 +364,7 @@ def main():
                 "p50_direct_ms": round(p50_direct, 3),
                 "p50_added_latency_ms": round(p50 - p50_direct, 3),
                 "gpu_token_accounting": use_gpu,
+                "semantic_cache_payload_pool": args.cache_payloads or None,
             },
         }
         print(json.dumps(out))
