@@ -20,9 +20,45 @@ from __future__ import annotations
 
 from typing import Callable, Optional
 
+import os
+
 from aigw import internalapi
 from aigw.backendauth.sigv4 import sign_sigv4
 from aigw.filterapi.config import Backend, BackendAuth
+
+
+class _FileCredential:
+    """mtime-cached credential file reader — picks up rotated credentials
+    without a restart (the BSP rotator analogue)."""
+
+    __slots__ = ("path", "_mtime", "_value")
+
+    def __init__(self, path: str):
+        self.path = path
+        self._mtime = -1.0
+        self._value = ""
+
+    def read(self) -> str:
+        try:
+            mtime = os.stat(self.path).st_mtime
+        except OSError:
+            return self._value
+        if mtime != self._mtime:
+            with open(self.path, "r", encoding="utf-8") as f:
+                self._value = f.read().strip()
+            self._mtime = mtime
+        return self._value
+
+
+def _parse_aws_ini(text: str) -> dict[str, str]:
+    out: dict[str, str] = {}
+    for line in text.splitlines():
+        line = line.strip()
+        if not line or line.startswith(("#", "[", ";")):
+            continue
+        k, _, v = line.partition("=")
+        out[k.strip().lower()] = v.strip()
+    return out
 
 # An AuthHandler mutates (headers, body) right before dispatch and returns
 # the new headers. Body is already final (mutation ordering: A.8).
@@ -52,6 +88,16 @@ def build_auth_handler(backend: Backend) -> Optional[AuthHandler]:
         return None
     kind = auth.kind
     if kind == "api_key":
+        if auth.api_key_file:
+            cred = _FileCredential(auth.api_key_file)
+
+            def handler(headers, body, method, path, _cred=cred):
+                ov = _pop_overrides(headers)
+                key = ov.get(internalapi.API_KEY_OVERRIDE_HEADER, _cred.read())
+                headers["authorization"] = f"Bearer {key}"
+                return headers
+
+            return handler
         return _bearer_handler(auth.api_key)
     if kind == "anthropic_api_key":
         return _anthropic_handler(auth)
@@ -100,12 +146,23 @@ def _azure_key_handler(auth: BackendAuth) -> AuthHandler:
 def _aws_handler(backend: Backend, auth: BackendAuth) -> AuthHandler:
     host = backend.upstream.hostname or backend.upstream.host
     region = auth.aws_region or "us-east-1"
+    cred = _FileCredential(auth.aws_credentials_file) if auth.aws_credentials_file else None
 
     def handler(headers, body, method, path):
         ov = _pop_overrides(headers)
-        access = ov.get(internalapi.AWS_ACCESS_KEY_OVERRIDE_HEADER, auth.aws_access_key_id)
-        secret = ov.get(internalapi.AWS_SECRET_KEY_OVERRIDE_HEADER, auth.aws_secret_access_key)
-        token = ov.get(internalapi.AWS_SESSION_TOKEN_OVERRIDE_HEADER, auth.aws_session_token)
+        file_creds = _parse_aws_ini(cred.read()) if cred is not None else {}
+        access = ov.get(
+            internalapi.AWS_ACCESS_KEY_OVERRIDE_HEADER,
+            file_creds.get("aws_access_key_id", auth.aws_access_key_id),
+        )
+        secret = ov.get(
+            internalapi.AWS_SECRET_KEY_OVERRIDE_HEADER,
+            file_creds.get("aws_secret_access_key", auth.aws_secret_access_key),
+        )
+        token = ov.get(
+            internalapi.AWS_SESSION_TOKEN_OVERRIDE_HEADER,
+            file_creds.get("aws_session_token", auth.aws_session_token),
+        )
         headers.pop("authorization", None)
         return sign_sigv4(
             method,

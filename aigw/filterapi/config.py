@@ -108,9 +108,20 @@ class BackendAuth:
     aws_secret_access_key: str = ""
     aws_session_token: str = ""
     aws_region: str = ""
+    # File-based credentials, re-read on change (mtime-cached): the hot-
+    # rotation analogue of the reference's BSP rotators, which refresh
+    # OIDC/exchanged credentials into mounted Secrets
+    # (internal/controller/rotators/). The file content replaces api_key /
+    # the AWS credential trio (INI credentials-file format for AWS).
+    api_key_file: str = ""
+    aws_credentials_file: str = ""
 
     @property
     def kind(self) -> str:
+        if self.api_key_file:
+            return "api_key"
+        if self.aws_credentials_file:
+            return "aws_access_key_id"
         for k in (
             "api_key",
             "anthropic_api_key",

@@ -108,3 +108,31 @@ def test_body_mutation_escaped_dot():
     doc = {}
     apply_body_mutation(doc, BodyMutation(set={r"metadata\.key.sub": 1}))
     assert doc == {"metadata.key": {"sub": 1}}
+
+
+def test_file_based_credentials_rotate(tmp_path):
+    """File-backed credentials are re-read on change (BSP rotator analogue)."""
+    key_file = tmp_path / "key"
+    key_file.write_text("sk-old\n")
+    b = Backend(name="o", auth=BackendAuth(api_key_file=str(key_file)))
+    h = build_auth_handler(b)
+    assert h({}, b"", "POST", "/x")["authorization"] == "Bearer sk-old"
+    import os
+    import time
+
+    key_file.write_text("sk-new\n")
+    os.utime(key_file, (time.time() + 5, time.time() + 5))
+    assert h({}, b"", "POST", "/x")["authorization"] == "Bearer sk-new"
+
+
+def test_aws_credentials_file(tmp_path):
+    cf = tmp_path / "credentials"
+    cf.write_text("[default]\naws_access_key_id = AKFILE\naws_secret_access_key = SKFILE\n")
+    b = Backend(
+        name="aws",
+        upstream=Upstream(host="bedrock.example", port=443),
+        auth=BackendAuth(aws_credentials_file=str(cf), aws_region="eu-west-1"),
+    )
+    h = build_auth_handler(b)({}, b"{}", "POST", "/model/m/converse")
+    assert "Credential=AKFILE/" in h["authorization"]
+    assert "/eu-west-1/bedrock/aws4_request" in h["authorization"]
