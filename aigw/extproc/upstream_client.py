@@ -26,7 +26,7 @@ from __future__ import annotations
 import asyncio
 import ssl as ssl_mod
 import zlib
-from typing import AsyncIterator, Optional
+from typing import AsyncIterator
 
 _MAX_HEADER_BYTES = 64 * 1024
 
