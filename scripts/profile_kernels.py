@@ -121,6 +121,22 @@ def main():
         "TFLOPs": round(2 * args.index_rows * args.batch * 384 / dt / 1e12, 2),
     }))
 
+    index8 = index.to(torch.float8_e4m3fn)
+    q8 = q.to(torch.float8_e4m3fn)
+
+    def run_topk8():
+        tok.hip.cache_topk(index8, q8)
+
+    dt = timeit(run_topk8)
+    print(json.dumps({
+        "kernel": "cache_topk_fp8",
+        "rows": args.index_rows,
+        "queries": args.batch,
+        "ms": round(dt * 1e3, 3),
+        "TB_per_s_index_read": round(args.index_rows * 384 / dt / 1e12, 3),
+        "TFLOPs": round(2 * args.index_rows * args.batch * 384 / dt / 1e12, 2),
+    }))
+
     stats = torch.rand(8, 4, device="cuda")
     stats[:, 1] = 1000
     pred = torch.rand(args.batch, device="cuda") * 100
