@@ -1046,7 +1046,7 @@ std::vector<at::Tensor> cache_topk(at::Tensor index, at::Tensor q) {
                                    .dtype(at::kLong)
                                    .device(q.device()));
   TORCH_CHECK(n_q <= 256, "cache_topk: at most 256 queries per call");
-  constexpr int ROWTILES = 8;
+  constexpr int ROWTILES = 16;
   long long blocks = (n_rows + 64 * ROWTILES - 1) / (64 * ROWTILES);
   auto* best_p = reinterpret_cast<unsigned long long*>(best.data_ptr<int64_t>());
   if (fp8) {
