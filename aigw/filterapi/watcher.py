@@ -7,6 +7,12 @@ change compile a fresh RuntimeConfig and hand it to the update callback.
 In-flight requests keep the RuntimeConfig they already resolved; new
 requests see the swapped pointer — reload never drops streams
 (SURVEY.md §5.4).
+
+If ``path`` is a directory containing an ``index.yaml``, it is treated as
+a sharded config bundle (aigw/filterapi/bundle.py, the analogue of
+StartConfigBundleWatcher in config_bundle.go): the watcher gates on the
+index UUID and only swaps after the sha256 over the reassembled parts
+verifies, so partially-synced part files are never loaded.
 """
 
 from __future__ import annotations
@@ -39,6 +45,13 @@ class ConfigWatcher:
 
     def load_once(self) -> RuntimeConfig:
         """Synchronous initial load; raises on invalid config."""
+        from aigw.filterapi import bundle
+
+        if bundle.is_bundle_dir(self.path):
+            raw, index = bundle.read_bundle_payload(self.path)
+            cfg = load_config(yaml.safe_load(raw))
+            self._last_key = index.uuid or hashlib.sha256(raw).hexdigest()
+            return RuntimeConfig(cfg)
         with open(self.path, "rb") as f:
             raw = f.read()
         cfg = load_config(yaml.safe_load(raw))
@@ -68,11 +81,23 @@ class ConfigWatcher:
                 logger.exception("config reload failed; keeping previous config")
 
     async def _check(self) -> None:
-        with open(self.path, "rb") as f:
-            raw = f.read()
-        data = yaml.safe_load(raw)
-        uuid = (data or {}).get("uuid", "") if isinstance(data, dict) else ""
-        key = uuid or hashlib.sha256(raw).hexdigest()
+        from aigw.filterapi import bundle
+
+        if bundle.is_bundle_dir(self.path):
+            # cheap gate: only the small index file is read per tick
+            with open(f"{self.path}/{bundle.INDEX_FILE}", "rb") as f:
+                index = bundle.parse_index(f.read())
+            key = index.uuid or index.checksum
+            if key == self._last_key:
+                return
+            raw, _ = bundle.read_bundle_payload(self.path)  # checksum-verified
+            data = yaml.safe_load(raw)
+        else:
+            with open(self.path, "rb") as f:
+                raw = f.read()
+            data = yaml.safe_load(raw)
+            uuid = (data or {}).get("uuid", "") if isinstance(data, dict) else ""
+            key = uuid or hashlib.sha256(raw).hexdigest()
         if key == self._last_key:
             return
         cfg = load_config(data)
