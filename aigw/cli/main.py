@@ -39,6 +39,12 @@ def _is_crd_bundle(text: str) -> bool:
 def _load_any_config(path: str | None):
     if path is None:
         return config_from_env(), None
+    import os as _os
+
+    if _os.path.isdir(path):  # sharded config bundle dir (filterapi.bundle)
+        from aigw.filterapi import bundle as _bundle
+
+        return _bundle.load_bundle(path), path
     with open(path, "r", encoding="utf-8") as f:
         text = f.read()
     if _is_crd_bundle(text):
@@ -211,7 +217,7 @@ def main(argv=None) -> int:
     sub = ap.add_subparsers(dest="cmd", required=True)
 
     runp = sub.add_parser("run", help="run the gateway")
-    runp.add_argument("--config", default=None, help="filter config or CRD bundle YAML")
+    runp.add_argument("--config", default=None, help="filter config YAML, CRD bundle YAML, or sharded bundle dir")
     runp.add_argument("--host", default="0.0.0.0")
     runp.add_argument("--port", type=int, default=internalapi.DEFAULT_LISTEN_PORT)
     runp.add_argument("--shards", type=int, default=1, help="shards (one per GPU)")
