@@ -10,10 +10,9 @@ Parity with internal/tracing:
   span events; usage lands in convention-specific attributes;
 - W3C ``traceparent`` is injected into UPSTREAM request headers so
   provider-side traces join (processor_impl.go:314-320);
-- exporters: JSONL file (``AIGW_TRACE_FILE``), console, in-memory (tests).
-  OTLP wire export is intentionally not implemented in this environment
-  (no OTel SDK available offline); the span model keeps OTLP-compatible
-  fields so an exporter can be added without touching call sites.
+- exporters: OTLP/HTTP JSON (``OTEL_EXPORTER_OTLP_ENDPOINT``, batched on a
+  background thread — aigw/tracing/otlp.py, no SDK dependency), JSONL file
+  (``AIGW_TRACE_FILE``), console, in-memory (tests).
 """
 
 from __future__ import annotations
@@ -191,8 +190,21 @@ def tracing_from_env(env: Optional[dict] = None) -> Optional[Tracer]:
         exporter = JSONLExporter(trace_file)
     elif env.get("OTEL_TRACES_EXPORTER", "") == "console":
         exporter = ConsoleExporter()
-    elif env.get("OTEL_EXPORTER_OTLP_ENDPOINT") or env.get("OTEL_TRACES_EXPORTER"):
-        # no offline OTLP wire support; keep spans observable locally
+    elif env.get("OTEL_EXPORTER_OTLP_ENDPOINT") or env.get(
+        "OTEL_EXPORTER_OTLP_TRACES_ENDPOINT"
+    ):
+        from aigw.tracing.otlp import OTLPHTTPExporter, parse_otlp_headers
+
+        endpoint = env.get("OTEL_EXPORTER_OTLP_TRACES_ENDPOINT") or env[
+            "OTEL_EXPORTER_OTLP_ENDPOINT"
+        ]
+        exporter = OTLPHTTPExporter(
+            endpoint,
+            headers=parse_otlp_headers(env.get("OTEL_EXPORTER_OTLP_HEADERS", "")),
+            service_name=env.get("OTEL_SERVICE_NAME", "ai-gateway"),
+            flush_interval_s=float(env.get("OTEL_BSP_SCHEDULE_DELAY", "5000")) / 1000.0,
+        )
+    elif env.get("OTEL_TRACES_EXPORTER"):
         exporter = ConsoleExporter()
     else:
         return None
