@@ -20,191 +20,11 @@
 #include <string>
 #include <vector>
 
+#include "native_core.h"
+
 namespace py = pybind11;
 
-namespace {
-
-struct Scan {
-  const char* p;
-  const char* end;
-  std::string model;
-  int stream = 0;
-  std::string text;
-  bool ok = true;
-
-  void ws() {
-    while (p < end && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
-  }
-
-  // parse a JSON string; when collect != nullptr, append the unescaped value.
-  // The common no-escape span is scanned with a tight byte loop (auto-
-  // vectorized) and appended in bulk — per-char push_back made the scanner
-  // slower than CPython's json for 16 KiB bodies.
-  bool parse_string(std::string* collect) {
-    if (p >= end || *p != '"') return fail();
-    ++p;
-    while (p < end) {
-      const char* seg = p;
-      while (p < end && *p != '"' && *p != '\\') ++p;
-      if (collect && p != seg) collect->append(seg, p - seg);
-      if (p >= end) return fail();
-      unsigned char c = *p;
-      if (c == '"') {
-        ++p;
-        return true;
-      }
-      if (c == '\\') {
-        if (p + 1 >= end) return fail();
-        char e = p[1];
-        p += 2;
-        if (!collect) continue;
-        switch (e) {
-          case '"': collect->push_back('"'); break;
-          case '\\': collect->push_back('\\'); break;
-          case '/': collect->push_back('/'); break;
-          case 'b': collect->push_back('\b'); break;
-          case 'f': collect->push_back('\f'); break;
-          case 'n': collect->push_back('\n'); break;
-          case 'r': collect->push_back('\r'); break;
-          case 't': collect->push_back('\t'); break;
-          case 'u': {
-            if (p + 4 > end) return fail();
-            unsigned v = 0;
-            for (int i = 0; i < 4; ++i) {
-              char h = p[i];
-              v <<= 4;
-              if (h >= '0' && h <= '9') v |= h - '0';
-              else if (h >= 'a' && h <= 'f') v |= h - 'a' + 10;
-              else if (h >= 'A' && h <= 'F') v |= h - 'A' + 10;
-              else return fail();
-            }
-            p += 4;
-            // surrogate pair
-            if (v >= 0xD800 && v <= 0xDBFF && p + 6 <= end && p[0] == '\\' &&
-                p[1] == 'u') {
-              unsigned lo = 0;
-              bool okp = true;
-              for (int i = 0; i < 4; ++i) {
-                char h = p[2 + i];
-                lo <<= 4;
-                if (h >= '0' && h <= '9') lo |= h - '0';
-                else if (h >= 'a' && h <= 'f') lo |= h - 'a' + 10;
-                else if (h >= 'A' && h <= 'F') lo |= h - 'A' + 10;
-                else { okp = false; break; }
-              }
-              if (okp && lo >= 0xDC00 && lo <= 0xDFFF) {
-                v = 0x10000 + ((v - 0xD800) << 10) + (lo - 0xDC00);
-                p += 6;
-              }
-            }
-            // UTF-8 encode
-            if (v < 0x80) collect->push_back((char)v);
-            else if (v < 0x800) {
-              collect->push_back((char)(0xC0 | (v >> 6)));
-              collect->push_back((char)(0x80 | (v & 0x3F)));
-            } else if (v < 0x10000) {
-              collect->push_back((char)(0xE0 | (v >> 12)));
-              collect->push_back((char)(0x80 | ((v >> 6) & 0x3F)));
-              collect->push_back((char)(0x80 | (v & 0x3F)));
-            } else {
-              collect->push_back((char)(0xF0 | (v >> 18)));
-              collect->push_back((char)(0x80 | ((v >> 12) & 0x3F)));
-              collect->push_back((char)(0x80 | ((v >> 6) & 0x3F)));
-              collect->push_back((char)(0x80 | (v & 0x3F)));
-            }
-            break;
-          }
-          default:
-            return fail();
-        }
-        continue;
-      }
-    }
-    return fail();
-  }
-
-  bool fail() {
-    ok = false;
-    return false;
-  }
-
-  bool parse_value(int depth, const std::string& key, bool at_root) {
-    if (depth > 64 || !ok) return fail();
-    ws();
-    if (p >= end) return fail();
-    char c = *p;
-    if (c == '{') {
-      ++p;
-      ws();
-      if (p < end && *p == '}') { ++p; return true; }
-      while (p < end) {
-        std::string k;
-        ws();
-        if (!parse_string(&k)) return false;
-        ws();
-        if (p >= end || *p != ':') return fail();
-        ++p;
-        if (!parse_value(depth + 1, k, false)) return false;
-        ws();
-        if (p < end && *p == ',') { ++p; continue; }
-        if (p < end && *p == '}') { ++p; return true; }
-        return fail();
-      }
-      return fail();
-    }
-    if (c == '[') {
-      ++p;
-      ws();
-      if (p < end && *p == ']') { ++p; return true; }
-      while (p < end) {
-        if (!parse_value(depth + 1, key, false)) return false;
-        ws();
-        if (p < end && *p == ',') { ++p; continue; }
-        if (p < end && *p == ']') { ++p; return true; }
-        return fail();
-      }
-      return fail();
-    }
-    if (c == '"') {
-      bool is_model = depth == 1 && key == "model";
-      bool is_text = key == "content" || key == "text" ||
-                     (depth == 1 && key == "system");
-      if (is_model) return parse_string(&model);
-      if (is_text) {
-        bool r = parse_string(&text);
-        text.push_back('\n');
-        return r;
-      }
-      return parse_string(nullptr);
-    }
-    // literals / numbers
-    if (c == 't') {
-      if (end - p < 4 || std::memcmp(p, "true", 4) != 0) return fail();
-      if (depth == 1 && key == "stream") stream = 1;
-      p += 4;
-      return true;
-    }
-    if (c == 'f') {
-      if (end - p < 5 || std::memcmp(p, "false", 5) != 0) return fail();
-      p += 5;
-      return true;
-    }
-    if (c == 'n') {
-      if (end - p < 4 || std::memcmp(p, "null", 4) != 0) return fail();
-      p += 4;
-      return true;
-    }
-    // number
-    const char* s = p;
-    while (p < end && (*p == '-' || *p == '+' || *p == '.' || *p == 'e' ||
-                       *p == 'E' || (*p >= '0' && *p <= '9')))
-      ++p;
-    if (p == s) return fail();
-    return true;
-  }
-};
-
-}  // namespace
+using aigw_core::Scan;
 
 // Returns (ok, model, stream, text_bytes)
 static py::tuple scan_chat_body(py::buffer buf) {
@@ -227,7 +47,7 @@ static py::tuple scan_chat_body(py::buffer buf) {
 static bool contains_usage(py::buffer buf) {
   py::buffer_info info = buf.request();
   const char* p = static_cast<const char*>(info.ptr);
-  return memmem(p, info.size, "\"usage\"", 7) != nullptr;
+  return aigw_core::contains_usage_raw(p, info.size);
 }
 
 class SSEFeed {
@@ -236,74 +56,24 @@ class SSEFeed {
   // event in the buffered stream; partial tails stay buffered.
   py::list feed(py::buffer chunk) {
     py::buffer_info info = chunk.request();
-    buf_.append(static_cast<const char*>(info.ptr), info.size);
     py::list out;
-    size_t start = 0;
-    for (;;) {
-      size_t nl = buf_.find('\n', start);
-      if (nl == std::string::npos) break;
-      size_t len = nl - start;
-      if (len && buf_[start + len - 1] == '\r') --len;
-      feed_line(buf_.data() + start, len, out);
-      start = nl + 1;
-    }
-    buf_.erase(0, start);
+    core_.feed(static_cast<const char*>(info.ptr), info.size,
+               [&out](const std::string& ev, const std::string& data) {
+                 out.append(py::make_tuple(py::str(ev), py::bytes(data)));
+               });
     return out;
   }
 
   py::list flush() {
     py::list out;
-    if (!buf_.empty()) {
-      size_t len = buf_.size();
-      if (len && buf_[len - 1] == '\r') --len;
-      feed_line(buf_.data(), len, out);
-      buf_.clear();
-    }
-    if (!data_.empty() || !event_.empty()) dispatch(out);
+    core_.flush([&out](const std::string& ev, const std::string& data) {
+      out.append(py::make_tuple(py::str(ev), py::bytes(data)));
+    });
     return out;
   }
 
  private:
-  void feed_line(const char* line, size_t n, py::list& out) {
-    if (n == 0) {
-      if (!data_.empty() || !event_.empty() || has_fields_) dispatch(out);
-      return;
-    }
-    if (line[0] == ':') return;  // comment
-    const char* colon = static_cast<const char*>(memchr(line, ':', n));
-    size_t name_len = colon ? (size_t)(colon - line) : n;
-    const char* value = colon ? colon + 1 : line + n;
-    size_t value_len = colon ? n - name_len - 1 : 0;
-    if (value_len && *value == ' ') {
-      ++value;
-      --value_len;
-    }
-    if (name_len == 4 && std::memcmp(line, "data", 4) == 0) {
-      if (has_data_) data_.push_back('\n');  // joins EMPTY data lines too
-      data_.append(value, value_len);
-      has_data_ = true;
-      has_fields_ = true;
-    } else if (name_len == 5 && std::memcmp(line, "event", 5) == 0) {
-      event_.assign(value, value_len);
-      has_fields_ = true;
-    } else if (name_len == 2 && std::memcmp(line, "id", 2) == 0) {
-      has_fields_ = true;
-    }
-  }
-
-  void dispatch(py::list& out) {
-    out.append(py::make_tuple(py::str(event_), py::bytes(data_)));
-    data_.clear();
-    event_.clear();
-    has_fields_ = false;
-    has_data_ = false;
-  }
-
-  std::string buf_;
-  std::string data_;
-  std::string event_;
-  bool has_fields_ = false;
-  bool has_data_ = false;
+  aigw_core::SSECore core_{};
 };
 
 PYBIND11_MODULE(aigw_native, m) {
