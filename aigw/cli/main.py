@@ -103,8 +103,21 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
         server, host=args.host, port=port, reuse_port=getattr(args, "workers", 1) > 1
     )
     print(f"aigw shard {rank}/{world} listening on http://{args.host}:{port}", flush=True)
+    stop = asyncio.Event()
+    loop = asyncio.get_running_loop()
+    import signal as _signal
+
+    for sig in (_signal.SIGTERM, _signal.SIGINT):
+        try:
+            loop.add_signal_handler(sig, stop.set)
+        except (NotImplementedError, RuntimeError):
+            pass
     try:
-        await asyncio.Event().wait()
+        await stop.wait()
+        # graceful drain: fail /health, 503 new work, finish in-flight
+        left = await server.drain(float(os.environ.get("AIGW_DRAIN_TIMEOUT", "30")))
+        if left:
+            print(f"aigw shard {rank}: drain timeout with {left} in flight", flush=True)
     finally:
         if watcher:
             await watcher.stop()

@@ -181,6 +181,24 @@ class LeanFront(asyncio.Protocol):
     async def _handle(self, req):
         method, path, headers, body = req
         server = self.server
+        if server.draining and path not in ("/health", "/metrics"):
+            self._write_simple(
+                503, {"content-type": "application/json"},
+                b'{"error":{"message":"shutting down","type":"unavailable"}}',
+                keep_alive=False,
+            )
+            return
+        server.inflight += 1
+        try:
+            await self._handle_inner(req)
+        finally:
+            server.inflight -= 1
+            if server.inflight == 0 and server.draining:
+                server._idle_event.set()
+
+    async def _handle_inner(self, req):
+        method, path, headers, body = req
+        server = self.server
         if method == "POST" and path in JSON_ENDPOINTS:
             view = RequestView(
                 method=method,

@@ -215,6 +215,13 @@ class GatewayServer:
         self.max_body_bytes = max_body_bytes
         self._session = None  # lean upstream client (aigw.extproc.upstream_client)
         self._started_at = time.time()
+        # graceful drain (the Envoy drain-sequence analogue): when
+        # draining, /health fails (LBs stop routing here), new requests
+        # get 503 + connection:close, and drain() waits for in-flight
+        # work to finish before the process exits
+        self.inflight = 0
+        self.draining = False
+        self._idle_event = asyncio.Event()
         # per-backend live stats for the KV-occupancy endpoint picker:
         # name -> [active_requests, estimated_active_tokens]
         self._ep_stats: dict[str, list] = {}
@@ -238,8 +245,40 @@ class GatewayServer:
         self.runtime = rc
         self.limiter.rules = {r.name: r for r in rc.rate_limits}
 
+    async def drain(self, timeout_s: float = 30.0) -> int:
+        """Stop taking new requests and wait (bounded) for in-flight ones.
+        Returns the number still in flight at the deadline (0 = clean)."""
+        self.draining = True
+        if self.inflight == 0:
+            return 0
+        self._idle_event.clear()
+        try:
+            await asyncio.wait_for(self._idle_event.wait(), timeout=timeout_s)
+        except asyncio.TimeoutError:
+            pass
+        return self.inflight
+
+    @web.middleware
+    async def _drain_middleware(self, request, handler):
+        if self.draining and request.path not in ("/health", "/metrics"):
+            return web.Response(
+                status=503,
+                text='{"error": {"message": "shutting down", "type": "unavailable"}}',
+                content_type="application/json",
+                headers={"connection": "close"},
+            )
+        self.inflight += 1
+        try:
+            return await handler(request)
+        finally:
+            self.inflight -= 1
+            if self.inflight == 0 and self.draining:
+                self._idle_event.set()
+
     def make_app(self) -> web.Application:
-        app = web.Application(client_max_size=self.max_body_bytes)
+        app = web.Application(
+            client_max_size=self.max_body_bytes, middlewares=[self._drain_middleware]
+        )
 
         def mount(path: str, handler) -> None:
             app.router.add_post(path, handler)
@@ -286,6 +325,10 @@ class GatewayServer:
     # ---- admin ---------------------------------------------------------------
 
     async def _handle_health(self, request: web.Request) -> web.Response:
+        if self.draining:  # LBs/k8s stop routing here during the drain window
+            return web.json_response(
+                {"status": "draining", "inflight": self.inflight}, status=503
+            )
         return web.json_response({"status": "ok", "uptime_s": time.time() - self._started_at})
 
     async def _handle_metrics(self, request: web.Request) -> web.Response:
