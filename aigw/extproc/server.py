@@ -506,7 +506,7 @@ class GatewayServer:
         ):
             hit, cache_key_vec = await self.gpu.cache_lookup_text(chat_text or b" ")
             if hit is not None and hit[:1] == cache_tag:
-                self.metrics.cache_events.labels(event="hit").inc()
+                self.metrics._child(self.metrics.cache_events, ("hit",)).inc()
                 body_bytes = hit[1:]
                 if not stream:
                     resp = web.Response(body=body_bytes, content_type="application/json")
@@ -521,7 +521,7 @@ class GatewayServer:
                     await writer.write(body_bytes[off : off + 16384])
                 await writer.finish()
                 return writer.result()
-            self.metrics.cache_events.labels(event="miss").inc()
+            self.metrics._child(self.metrics.cache_events, ("miss",)).inc()
 
         span = None
         if self.tracer is not None:
@@ -677,8 +677,9 @@ class GatewayServer:
             if backend.max_concurrency and st[0] >= backend.max_concurrency:
                 # circuit open: saturated backend = failed attempt
                 last_error = f"backend {backend.name} saturated ({st[0]} in flight)"
-                self.metrics.requests_total.labels(
-                    endpoint=endpoint, backend=backend.name, status="circuit_open"
+                self.metrics._child(
+                    self.metrics.requests_total,
+                    (endpoint, backend.name, "circuit_open"),
                 ).inc()
                 continue
             st[0] += 1
@@ -789,8 +790,9 @@ class GatewayServer:
             self.metrics.record_tokens(labels, usage)
         if ttft >= 0:
             self.metrics.record_stream_latency(labels, ttft, elapsed, usage.output_tokens)
-        self.metrics.requests_total.labels(
-            endpoint=endpoint, backend=backend.name if backend else "", status=str(status)
+        self.metrics._child(
+            self.metrics.requests_total,
+            (endpoint, backend.name if backend else "", str(status)),
         ).inc()
         if access_logger.isEnabledFor(logging.INFO):
             # enriched access log, carrying what the reference pushes into

@@ -38,6 +38,7 @@ _LABELS = [
 class GenAIMetrics:
     def __init__(self, registry: CollectorRegistry | None = None):
         self.registry = registry or CollectorRegistry()
+        self._children: dict = {}
         self.token_usage = Histogram(
             "gen_ai_client_token_usage",
             "OTel gen_ai.client.token.usage: tokens used per request",
@@ -106,16 +107,21 @@ class GenAIMetrics:
         )
 
     def labels(self, *, operation: str, provider: str, original_model: str,
-               request_model: str, response_model: str) -> dict[str, str]:
-        return {
-            "gen_ai_operation_name": operation,
-            "gen_ai_provider_name": provider,
-            "gen_ai_original_model": original_model,
-            "gen_ai_request_model": request_model,
-            "gen_ai_response_model": response_model or request_model,
-        }
+               request_model: str, response_model: str) -> tuple:
+        # ordered to match _LABELS; consumed positionally via the child
+        # cache below (prometheus_client's labels() re-validates and locks
+        # on every call — ~7% of a worker's loop time uncached)
+        return (operation, provider, original_model, request_model,
+                response_model or request_model)
 
-    def record_tokens(self, labels: dict[str, str], usage) -> None:
+    def _child(self, metric, values: tuple):
+        key = (id(metric), values)
+        c = self._children.get(key)
+        if c is None:
+            c = self._children[key] = metric.labels(*values)
+        return c
+
+    def record_tokens(self, labels: tuple, usage) -> None:
         for token_type, value in (
             ("input", usage.input_tokens),
             ("output", usage.output_tokens),
@@ -125,17 +131,17 @@ class GenAIMetrics:
             ("reasoning", usage.reasoning_tokens),
         ):
             if value:
-                self.token_usage.labels(**labels, gen_ai_token_type=token_type).observe(value)
+                self._child(self.token_usage, labels + (token_type,)).observe(value)
 
-    def record_request(self, labels: dict[str, str], seconds: float, error_type: str = "") -> None:
-        self.request_duration.labels(**labels, error_type=error_type or "").observe(seconds)
+    def record_request(self, labels: tuple, seconds: float, error_type: str = "") -> None:
+        self._child(self.request_duration, labels + (error_type or "",)).observe(seconds)
 
     def record_stream_latency(
-        self, labels: dict[str, str], ttft_s: float, elapsed_s: float, output_tokens: int
+        self, labels: tuple, ttft_s: float, elapsed_s: float, output_tokens: int
     ) -> None:
-        self.ttft.labels(**labels).observe(ttft_s)
+        self._child(self.ttft, labels).observe(ttft_s)
         if output_tokens > 1:
-            self.itl.labels(**labels).observe((elapsed_s - ttft_s) / (output_tokens - 1))
+            self._child(self.itl, labels).observe((elapsed_s - ttft_s) / (output_tokens - 1))
 
     def render(self) -> bytes:
         return generate_latest(self.registry)
