@@ -55,6 +55,10 @@ def _load_any_config(path: str | None):
 async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
     import torch
 
+    from aigw.utils import tune_gc
+
+    tune_gc()
+
     cfg, watch_path = _load_any_config(args.config)
     runtime = RuntimeConfig(cfg)
 

@@ -205,6 +205,9 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
 
 
 def worker_entry(args, local_rank, ready, go, out_q):
+    from aigw.utils import tune_gc
+
+    tune_gc()
     asyncio.run(worker_main(args, local_rank, ready, go, out_q))
 
 
