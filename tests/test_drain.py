@@ -91,3 +91,49 @@ def test_drain_immediate_when_idle():
         assert server.draining
 
     asyncio.run(run())
+
+
+@pytest.mark.timeout(60)
+def test_drain_rejects_on_lean_front():
+    from aigw.extproc.lean_front import serve_lean
+    from aigw.extproc.upstream_client import LeanClient
+
+    async def run():
+        up_runner, up_port = await _slow_upstream()
+        cfg = load_config(yaml.safe_load(CFG % up_port))
+        server = GatewayServer(RuntimeConfig(cfg))
+        await server.start()
+        _, gw_port, gw_cleanup = await serve_lean(
+            server, "127.0.0.1", 0, with_fallback=False
+        )
+        body = json.dumps(
+            {"model": "m", "messages": [{"role": "user", "content": "q"}]}
+        ).encode()
+        client = LeanClient()
+
+        async def post():
+            return await client.post(
+                host="127.0.0.1", port=gw_port, tls=False,
+                path="/v1/chat/completions",
+                headers={"content-type": "application/json"}, body=body,
+            )
+
+        task = asyncio.create_task(post())
+        await asyncio.sleep(0.1)
+        drain = asyncio.create_task(server.drain(timeout_s=10))
+        await asyncio.sleep(0.05)
+        r = await post()
+        assert r.status == 503
+        assert json.loads(await r.read())["error"]["type"] == "unavailable"
+        r.close()
+        resp = await task  # in-flight request still completes
+        assert resp.status == 200
+        assert json.loads(await resp.read())["choices"][0]["message"]["content"] == "hi"
+        resp.release()
+        assert await drain == 0
+        await gw_cleanup()
+        await up_runner.cleanup()
+        await client.close()
+        await server.close()
+
+    asyncio.run(run())
