@@ -134,7 +134,7 @@ class RemoteGPUClient:
     shard's GPU service. Exposes the same interface GatewayServer uses."""
 
     def __init__(self, socket_path: str, *, enable_cache: bool = False,
-                 window_ms: float = 1.0, max_batch: int = 128):
+                 window_ms: float = 0.1, max_batch: int = 128):
         self.socket_path = socket_path
         self.cache_enabled = enable_cache
         self.window_ms = window_ms

@@ -59,7 +59,7 @@ class GPUServices:
         enable_cache: bool = False,
         cache_capacity: int = 65536,
         cache_threshold: float = 0.92,
-        window_ms: float = 1.0,
+        window_ms: float = 0.1,
         max_batch: int = 128,
     ):
         from aigw.ops.semcache import SemanticCache

@@ -131,13 +131,13 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
         # knee; see --gpu-service help)
         from aigw.gpu.service import RemoteGPUClient
 
-        gpu_services = RemoteGPUClient(args.gpu_socket, window_ms=0.5, max_batch=256)
+        gpu_services = RemoteGPUClient(args.gpu_socket, window_ms=args.gpu_window, max_batch=256)
     elif use_gpu:
         from aigw.gpu import GPUServices
 
         gpu_services = GPUServices(
             device=f"cuda:{local_rank}", n_merges=32768, enable_cache=cache_mode,
-            cache_threshold=0.98, window_ms=getattr(args, "gpu_window", 0.5),
+            cache_threshold=0.98, window_ms=getattr(args, "gpu_window", 0.1),
             max_batch=256,
         )
 
@@ -211,7 +211,7 @@ def worker_entry(args, local_rank, ready, go, out_q):
     asyncio.run(worker_main(args, local_rank, ready, go, out_q))
 
 
-def _start_gpu_host(socket_path: str, local_rank: int):
+def _start_gpu_host(socket_path: str, local_rank: int, window_ms: float = 0.1):
     """Run the shard's GPU admission service on a dedicated thread/loop in
     the rank-primary process (the only process owning a GPU context)."""
     import threading
@@ -225,7 +225,7 @@ def _start_gpu_host(socket_path: str, local_rank: int):
 
         async def amain():
             gpu = GPUServices(device=f"cuda:{local_rank}", n_merges=32768,
-                              enable_cache=False, window_ms=0.5, max_batch=1024)
+                              enable_cache=False, window_ms=window_ms, max_batch=1024)
             host = GPUServiceHost(gpu, socket_path)
             await host.start()
             ready_evt.set()
@@ -255,7 +255,7 @@ def main():
                     help="semantic-cache mode: cycle this many distinct "
                          "payloads per worker (0 = cache off); steady-state "
                          "hit ratio approaches 1")
-    ap.add_argument("--gpu-window", type=float, default=0.5,
+    ap.add_argument("--gpu-window", type=float, default=0.1,
                     help="GPU micro-batch window per worker, ms")
     ap.add_argument("--gpu-service", action="store_true",
                     help="route GPU work through ONE per-shard admission "
@@ -289,7 +289,7 @@ def main():
     gpu_host_state = None
     if use_gpu and args.gpu_service:
         gpu_socket = f"/tmp/aigw-gpu-{rank}-{os.getpid()}.sock"
-        gpu_host_state = _start_gpu_host(gpu_socket, local_rank)
+        gpu_host_state = _start_gpu_host(gpu_socket, local_rank, args.gpu_window)
     args.gpu_socket = gpu_socket
 
     ctx = mp.get_context("spawn")
