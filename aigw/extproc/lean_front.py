@@ -82,14 +82,20 @@ _STREAMED = object()  # sentinel: response already written to the transport
 
 class LeanFront(asyncio.Protocol):
     def __init__(self, server: GatewayServer, fallback_port: Optional[int],
-                 fallback_client: Optional[LeanClient]):
+                 fallback_client: Optional[LeanClient],
+                 idle_timeout_s: float = 60.0):
         self.server = server
         self.fallback_port = fallback_port
         self.fallback = fallback_client
+        self.idle_timeout_s = idle_timeout_s
         self._buf = bytearray()
         self._transport = None
         self._chain: Optional[asyncio.Future] = None
         self._peer = ""
+        # slowloris guard: armed only while an INCOMPLETE request sits in
+        # the buffer; fires -> if no new bytes arrived since, close
+        self._idle_handle = None
+        self._buf_len_at_arm = -1
 
     # ---- protocol ------------------------------------------------------------
 
@@ -104,6 +110,9 @@ class LeanFront(asyncio.Protocol):
             sock.setsockopt(_s.IPPROTO_TCP, _s.TCP_NODELAY, 1)
 
     def connection_lost(self, exc):
+        if self._idle_handle is not None:
+            self._idle_handle.cancel()
+            self._idle_handle = None
         if self._chain is not None and not self._chain.done():
             self._chain.cancel()
 
@@ -112,9 +121,32 @@ class LeanFront(asyncio.Protocol):
         while True:
             req = self._try_parse()
             if req is None:
+                if self._buf and self._idle_handle is None:
+                    self._arm_idle_guard()
                 return
             prev = self._chain
             self._chain = asyncio.ensure_future(self._handle_ordered(prev, req))
+
+    def _arm_idle_guard(self):
+        self._buf_len_at_arm = len(self._buf)
+        self._idle_handle = asyncio.get_running_loop().call_later(
+            self.idle_timeout_s, self._idle_check
+        )
+
+    def _idle_check(self):
+        self._idle_handle = None
+        if not self._buf:
+            return
+        if len(self._buf) == self._buf_len_at_arm:
+            # partial request made zero progress for a full window
+            if self._transport is not None and not self._transport.is_closing():
+                self._transport.write(
+                    b"HTTP/1.1 408 Request Timeout\r\ncontent-length: 0\r\n"
+                    b"connection: close\r\n\r\n"
+                )
+                self._transport.close()
+            return
+        self._arm_idle_guard()
 
     def _try_parse(self):
         end = self._buf.find(b"\r\n\r\n")
@@ -133,7 +165,17 @@ class LeanFront(asyncio.Protocol):
         for line in lines[1:]:
             k, _, v = line.partition(b":")
             headers[k.decode("latin1").strip().lower()] = v.decode("latin1").strip()
-        clen = int(headers.get("content-length", "0") or 0)
+        try:
+            clen = int(headers.get("content-length", "0") or 0)
+        except ValueError:
+            clen = -1
+        if clen < 0 or clen > 64 * 1024 * 1024:
+            self._transport.write(
+                b"HTTP/1.1 400 Bad Request\r\ncontent-length: 0\r\n"
+                b"connection: close\r\n\r\n"
+            )
+            self._transport.close()
+            return None
         total = end + 4 + clen
         if len(self._buf) < total:
             return None
@@ -269,7 +311,8 @@ class LeanFront(asyncio.Protocol):
 
 
 async def serve_lean(server: GatewayServer, host: str, port: int,
-                     *, with_fallback: bool = True, reuse_port: bool = False):
+                     *, with_fallback: bool = True, reuse_port: bool = False,
+                     idle_timeout_s: float = 60.0):
     """Start the lean front (and, when with_fallback, a loopback aiohttp
     app for the cold paths). Returns (asyncio.Server, actual_port, cleanup)."""
     fallback_port = None
@@ -288,7 +331,8 @@ async def serve_lean(server: GatewayServer, host: str, port: int,
 
     loop = asyncio.get_running_loop()
     srv = await loop.create_server(
-        lambda: LeanFront(server, fallback_port, fallback_client),
+        lambda: LeanFront(server, fallback_port, fallback_client,
+                          idle_timeout_s=idle_timeout_s),
         host, port, reuse_port=reuse_port, backlog=4096,
     )
     actual = srv.sockets[0].getsockname()[1]

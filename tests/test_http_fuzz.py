@@ -1,0 +1,173 @@
+"""Robustness fuzz of the gateway HTTP surface (CPU-only): malformed
+JSON bodies and broken HTTP framing must produce clean 4xx responses or
+connection closes — never unhandled exceptions, hangs, or 5xx (the
+reference inherits this hardening from Envoy's codec; the lean front
+implements framing itself, so it gets fuzzed here)."""
+
+import asyncio
+import json
+
+import pytest
+import yaml
+
+from aigw.extproc.lean_front import serve_lean
+from aigw.extproc.server import GatewayServer
+from aigw.extproc.upstream_client import LeanClient
+from aigw.filterapi.config import load_config
+from aigw.filterapi.runtime import RuntimeConfig
+from aigw.testing.fastmock import start_fast_mock
+
+
+def _rng(seed=0x5EED):
+    state = seed or 1
+
+    def nxt(n):
+        nonlocal state
+        state ^= (state << 13) & 0xFFFFFFFFFFFFFFFF
+        state ^= state >> 7
+        state ^= (state << 17) & 0xFFFFFFFFFFFFFFFF
+        return state % n
+
+    return nxt
+
+
+async def _gateway(idle_timeout_s: float = 2.0):
+    up_srv, up_port = await start_fast_mock("127.0.0.1", 0)
+    cfg = load_config(yaml.safe_load(f"""
+routes:
+  - name: r
+    backends:
+      - name: b
+        schema: OpenAI
+        upstream: {{host: 127.0.0.1, port: {up_port}}}
+"""))
+    server = GatewayServer(RuntimeConfig(cfg))
+    await server.start()
+    _, gw_port, gw_cleanup = await serve_lean(
+        server, "127.0.0.1", 0, with_fallback=False, idle_timeout_s=idle_timeout_s
+    )
+
+    async def cleanup():
+        await gw_cleanup()
+        await server.close()
+        up_srv.close()
+
+    return server, gw_port, cleanup
+
+
+@pytest.mark.timeout(120)
+def test_malformed_json_bodies_get_4xx():
+    async def run():
+        _, port, cleanup = await _gateway()
+        client = LeanClient()
+        nxt = _rng()
+        valid = json.dumps({"model": "m", "messages": [{"role": "user", "content": "hi"}]})
+        alphabet = '{}[]",:null true false 0123456789.eE-\\u00'
+        cases = [b"", b"{", b"[1,", b'{"model":}', b'{"model":"m"',
+                 b"\xff\xfe\x00", b'{"model": "m", "messages": "not-a-list"}',
+                 b"[" * 100, json.dumps({"messages": []}).encode()]
+        for _ in range(150):
+            b = bytearray(valid.encode())
+            for _m in range(1 + nxt(4)):
+                b[nxt(len(b))] = ord(alphabet[nxt(len(alphabet))])
+            cases.append(bytes(b))
+        for body in cases:
+            r = await client.post(
+                host="127.0.0.1", port=port, tls=False,
+                path="/v1/chat/completions",
+                headers={"content-type": "application/json"}, body=body,
+            )
+            data = await r.read()
+            # mutations can leave the body valid (-> 200 via mock) or
+            # make it invalid (-> 4xx); a 5xx or protocol error = bug
+            assert r.status < 500, (r.status, body[:120], data[:200])
+            r.release()
+        await client.close()
+        await cleanup()
+
+    asyncio.run(run())
+
+
+@pytest.mark.timeout(120)
+def test_broken_http_framing_never_hangs():
+    async def run():
+        _, port, cleanup = await _gateway()
+        frames = [
+            b"GARBAGE\r\n\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\ncontent-length: -5\r\n\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\ncontent-length: zzz\r\n\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\n" + b"x: y\r\n" * 5000 + b"\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\ncontent-length: 10\r\n\r\nshort",
+            # ^ stalls mid-body: the idle guard must answer 408 and close
+            b"POST /nope HTTP/1.1\r\ncontent-length: 2\r\n\r\n{}",
+            b"GET /health HTTP/1.1\r\n\r\n" * 3,  # pipelined
+            b"\r\n\r\n\r\n",
+        ]
+        for payload in frames:
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            writer.write(payload)
+            try:
+                await writer.drain()
+                # server must either answer or close within the deadline
+                await asyncio.wait_for(reader.read(65536), timeout=6)
+            except (asyncio.TimeoutError,) as e:
+                raise AssertionError(f"hang on frame {payload[:60]!r}") from e
+            except (ConnectionResetError, BrokenPipeError):
+                pass
+            finally:
+                writer.close()
+        # the server must still be healthy afterwards
+        client = LeanClient()
+        r = await client.post(
+            host="127.0.0.1", port=port, tls=False, path="/v1/chat/completions",
+            headers={"content-type": "application/json"},
+            body=json.dumps({"model": "m", "messages": [{"role": "user", "content": "ok"}]}).encode(),
+        )
+        assert r.status == 200
+        await r.read()
+        r.release()
+        await client.close()
+        await cleanup()
+
+    asyncio.run(run())
+
+
+@pytest.mark.timeout(120)
+def test_slow_and_partial_requests():
+    """Byte-at-a-time delivery and mid-body disconnects must not wedge
+    the worker or leak the connection slot."""
+
+    async def run():
+        _, port, cleanup = await _gateway()
+        body = json.dumps({"model": "m", "messages": [{"role": "user", "content": "x"}]}).encode()
+        head = (b"POST /v1/chat/completions HTTP/1.1\r\n"
+                b"content-type: application/json\r\n"
+                b"content-length: %d\r\n\r\n" % len(body))
+        # trickle a full request
+        reader, writer = await asyncio.open_connection("127.0.0.1", port)
+        for i in range(0, len(head + body), 7):
+            writer.write((head + body)[i : i + 7])
+            await writer.drain()
+            await asyncio.sleep(0)
+        resp = await asyncio.wait_for(reader.readline(), timeout=5)
+        assert b"200" in resp
+        writer.close()
+        # disconnect mid-body x20
+        for _ in range(20):
+            _, w = await asyncio.open_connection("127.0.0.1", port)
+            w.write(head + body[: len(body) // 2])
+            await w.drain()
+            w.close()
+        # server still serves
+        client = LeanClient()
+        r = await client.post(
+            host="127.0.0.1", port=port, tls=False, path="/v1/chat/completions",
+            headers={"content-type": "application/json"}, body=body,
+        )
+        assert r.status == 200
+        await r.read()
+        r.release()
+        await client.close()
+        await cleanup()
+
+    asyncio.run(run())
