@@ -95,7 +95,8 @@ class LeanFront(asyncio.Protocol):
         # slowloris guard: armed only while an INCOMPLETE request sits in
         # the buffer; fires -> if no new bytes arrived since, close
         self._idle_handle = None
-        self._buf_len_at_arm = -1
+        self._rx_total = 0          # monotonic; snapshot taken at arm time
+        self._rx_at_arm = -1
 
     # ---- protocol ------------------------------------------------------------
 
@@ -117,6 +118,7 @@ class LeanFront(asyncio.Protocol):
             self._chain.cancel()
 
     def data_received(self, data: bytes):
+        self._rx_total += len(data)
         self._buf.extend(data)
         while True:
             req = self._try_parse()
@@ -128,7 +130,7 @@ class LeanFront(asyncio.Protocol):
             self._chain = asyncio.ensure_future(self._handle_ordered(prev, req))
 
     def _arm_idle_guard(self):
-        self._buf_len_at_arm = len(self._buf)
+        self._rx_at_arm = self._rx_total
         self._idle_handle = asyncio.get_running_loop().call_later(
             self.idle_timeout_s, self._idle_check
         )
@@ -137,7 +139,7 @@ class LeanFront(asyncio.Protocol):
         self._idle_handle = None
         if not self._buf:
             return
-        if len(self._buf) == self._buf_len_at_arm:
+        if self._rx_total == self._rx_at_arm:
             # partial request made zero progress for a full window
             if self._transport is not None and not self._transport.is_closing():
                 self._transport.write(
