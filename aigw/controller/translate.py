@@ -258,6 +258,19 @@ def translate_crds(docs: list[dict]) -> Config:
         ns = md.get("namespace", "default")
         spec = route_doc.get("spec") or {}
         route_costs = _parse_costs(spec.get("llmRequestCosts"))
+        owned_by = spec.get("modelsOwnedBy") or "aigw"
+        created_at = 0
+        if spec.get("modelsCreatedAt"):
+            import datetime as _dt
+
+            try:
+                created_at = int(
+                    _dt.datetime.fromisoformat(
+                        str(spec["modelsCreatedAt"]).replace("Z", "+00:00")
+                    ).timestamp()
+                )
+            except ValueError:
+                created_at = 0
         for i, rule in enumerate(spec.get("rules") or []):
             headers: list[HeaderMatch] = []
             for match in rule.get("matches") or []:
@@ -271,7 +284,9 @@ def translate_crds(docs: list[dict]) -> Config:
                         v = h.get("value", "")
                         if v and v not in seen_models:
                             seen_models.add(v)
-                            models.append(Model(name=v))
+                            models.append(
+                                Model(name=v, owned_by=owned_by, created_at=created_at)
+                            )
             backends: list[Backend] = []
             for bref in rule.get("backendRefs") or []:
                 asb_name = bref.get("name", "")

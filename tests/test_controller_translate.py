@@ -190,3 +190,34 @@ def test_cli_translate_roundtrip(tmp_path, capsys):
     cfg = load_config(out)
     assert len(cfg.routes) == 2
     assert cfg.routes[1].backends[0].auth.aws_access_key_id == "AKTEST"
+
+
+def test_models_owned_by_and_created_at():
+    from aigw.controller import translate_yaml
+
+    cfg = translate_yaml(
+        """
+apiVersion: aigateway.envoyproxy.io/v1beta1
+kind: AIGatewayRoute
+metadata: {name: r, namespace: default}
+spec:
+  modelsOwnedBy: my-org
+  modelsCreatedAt: "2024-05-01T00:00:00Z"
+  rules:
+    - matches:
+        - headers:
+            - {type: Exact, name: x-ai-eg-model, value: gpt-4o}
+      backendRefs: [{name: b}]
+---
+apiVersion: aigateway.envoyproxy.io/v1beta1
+kind: AIServiceBackend
+metadata: {name: b, namespace: default}
+spec:
+  schema: {name: OpenAI}
+  backendRef: {name: up, port: 8080}
+"""
+    )
+    (m,) = cfg.models
+    assert m.name == "gpt-4o"
+    assert m.owned_by == "my-org"
+    assert m.created_at == 1714521600
