@@ -258,20 +258,22 @@ def translate_crds(docs: list[dict]) -> Config:
         ns = md.get("namespace", "default")
         spec = route_doc.get("spec") or {}
         route_costs = _parse_costs(spec.get("llmRequestCosts"))
-        owned_by = spec.get("modelsOwnedBy") or "aigw"
-        created_at = 0
-        if spec.get("modelsCreatedAt"):
-            import datetime as _dt
-
-            try:
-                created_at = int(
-                    _dt.datetime.fromisoformat(
-                        str(spec["modelsCreatedAt"]).replace("Z", "+00:00")
-                    ).timestamp()
-                )
-            except ValueError:
-                created_at = 0
         for i, rule in enumerate(spec.get("rules") or []):
+            # per-rule /v1/models attribution (ai_gateway_route.go
+            # ModelsOwnedBy/ModelsCreatedAt — rule-level fields)
+            owned_by = rule.get("modelsOwnedBy") or "aigw"
+            created_at = 0
+            if rule.get("modelsCreatedAt"):
+                import datetime as _dt
+
+                try:
+                    created_at = int(
+                        _dt.datetime.fromisoformat(
+                            str(rule["modelsCreatedAt"]).replace("Z", "+00:00")
+                        ).timestamp()
+                    )
+                except ValueError:
+                    created_at = 0
             headers: list[HeaderMatch] = []
             for match in rule.get("matches") or []:
                 for h in match.get("headers") or []:

@@ -201,10 +201,10 @@ apiVersion: aigateway.envoyproxy.io/v1beta1
 kind: AIGatewayRoute
 metadata: {name: r, namespace: default}
 spec:
-  modelsOwnedBy: my-org
-  modelsCreatedAt: "2024-05-01T00:00:00Z"
   rules:
-    - matches:
+    - modelsOwnedBy: my-org
+      modelsCreatedAt: "2024-05-01T00:00:00Z"
+      matches:
         - headers:
             - {type: Exact, name: x-ai-eg-model, value: gpt-4o}
       backendRefs: [{name: b}]
