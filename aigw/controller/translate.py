@@ -318,6 +318,9 @@ def translate_crds(docs: list[dict]) -> Config:
                 timeouts = rule.get("timeouts") or {}
                 if timeouts.get("request"):
                     timeout = _parse_duration(timeouts["request"])
+                idle = 0.0
+                if rule.get("streamIdleTimeout"):
+                    idle = _parse_duration(rule["streamIdleTimeout"])
                 backends.append(
                     Backend(
                         name=asb_name,
@@ -328,6 +331,7 @@ def translate_crds(docs: list[dict]) -> Config:
                         model_name_override=bref.get("modelNameOverride", ""),
                         auth=auth,
                         timeout_s=timeout,
+                        stream_idle_timeout_s=idle,
                     )
                 )
             if backends:

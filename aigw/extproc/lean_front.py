@@ -73,6 +73,11 @@ class _LeanStreamWriter:
         if not self.transport.is_closing():
             self.transport.write(b"0\r\n\r\n")
 
+    def abort(self) -> None:
+        """Close without the terminal chunk: the client sees truncated
+        chunked framing (mid-stream idle cut)."""
+        self.transport.close()
+
     def result(self):
         return _STREAMED
 

@@ -155,6 +155,10 @@ class _AiohttpStreamWriter:
     async def finish(self):
         await self._resp.write_eof()
 
+    def abort(self) -> None:
+        """Terminate without proper EOF framing (mid-stream cut)."""
+        self._resp.force_close()
+
     def result(self) -> web.StreamResponse:
         return self._resp
 
@@ -176,6 +180,13 @@ def _aiohttp_view(request: web.Request, body: bytes) -> RequestView:
         body=body,
         stream_factory=stream_factory,
     )
+
+
+async def _prepend(first, aiter):
+    if first is not None:
+        yield first
+    async for item in aiter:
+        yield item
 
 
 def _json_error(status: int, message: str, err_type: str = "invalid_request_error") -> web.Response:
@@ -822,7 +833,10 @@ class GatewayServer:
         self, endpoint, route, backend, translator, upstream, headers, model,
         start, gpu_input_tokens, orig_body, cache_key_vec, span=None,
     ) -> web.Response:
-        data = await upstream.read()  # LeanResponse decompresses gzip/deflate
+        # idle-bounded read: a stalled upstream raises UpstreamError, which
+        # _dispatch treats as a failed try (fallback proceeds — nothing has
+        # been written to the client yet)
+        data = await upstream.read(idle_timeout=backend.stream_idle_timeout_s)
         upstream.release()
         try:
             rtl = translator.response_body(upstream.status, data)
@@ -853,6 +867,18 @@ class GatewayServer:
         headers, model, start, gpu_input_tokens, span=None, cache_key_vec=None,
     ) -> web.StreamResponse:
         transcript = bytearray() if cache_key_vec is not None else None
+        idle = backend.stream_idle_timeout_s
+        chunks = upstream.iter_chunks(idle_timeout=idle)
+        first_chunk = None
+        if idle > 0:
+            # wait for the first body byte BEFORE committing a response to
+            # the client: an idle upstream here resets the try and fallback
+            # proceeds (per_try_idle_timeout pre-first-byte semantics) —
+            # UpstreamError propagates to _dispatch's retry loop
+            try:
+                first_chunk = await chunks.__anext__()
+            except StopAsyncIteration:
+                first_chunk = None
         hdrs = translator.response_headers(upstream.status, upstream.headers)
         content_type = hdrs.get("content-type") or upstream.headers.get(
             "content-type", "text/event-stream"
@@ -866,8 +892,9 @@ class GatewayServer:
         response_model = ""
         ttft = -1.0
         aborted = False
+        cut = False
         try:
-            async for chunk in upstream.iter_chunks():
+            async for chunk in _prepend(first_chunk, chunks):
                 if not chunk:
                     continue
                 rtl = translator.response_chunk(chunk)
@@ -894,12 +921,18 @@ class GatewayServer:
                 and len(transcript) < (2 << 20)
             ):
                 await self.gpu.cache_insert(cache_key_vec, b"S" + bytes(transcript))
+        except UpstreamError as e:
+            # mid-stream idle: the response has started, so fallback is no
+            # longer possible — cut the stream (the client sees truncated
+            # chunked framing; Envoy's analogue is a stream reset / 504)
+            cut = True
+            logger.warning("cutting stream to %s: %s", backend.name, e)
         except (ConnectionResetError, asyncio.CancelledError):
             aborted = True
             logger.info("client disconnected mid-stream")
             raise
         finally:
-            if aborted:
+            if aborted or cut:
                 upstream.close()
             else:
                 upstream.release()
@@ -910,6 +943,11 @@ class GatewayServer:
                 endpoint, route, backend, model, response_model, usage, start,
                 status=upstream.status, ttft=ttft if ttft >= 0 else 0.0,
             )
+        if cut:
+            abort = getattr(writer, "abort", None)
+            if abort is not None:
+                abort()
+            return writer.result()
         await writer.finish()
         return writer.result()
 

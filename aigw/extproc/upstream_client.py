@@ -96,9 +96,24 @@ class LeanResponse:
             self._conn.reusable = False
         return data
 
-    async def iter_chunks(self) -> AsyncIterator[bytes]:
+    async def iter_chunks(self, idle_timeout: float = 0) -> AsyncIterator[bytes]:
+        """Yield decompressed body chunks. idle_timeout > 0 bounds the gap
+        between consecutive raw chunks (per-try idle timeout — the
+        reference's route.retry_policy.per_try_idle_timeout); on expiry
+        the connection is dropped and UpstreamError raised."""
         while True:
-            raw = await self._read_raw_chunk()
+            if idle_timeout > 0:
+                try:
+                    raw = await asyncio.wait_for(
+                        self._read_raw_chunk(), timeout=idle_timeout
+                    )
+                except asyncio.TimeoutError:
+                    self._conn.reusable = False
+                    raise UpstreamError(
+                        f"no upstream bytes within {idle_timeout}s (idle timeout)"
+                    )
+            else:
+                raw = await self._read_raw_chunk()
             if not raw and self._eof:
                 tail = self._decomp.flush() if self._decomp is not None else b""
                 if tail:
@@ -108,9 +123,9 @@ class LeanResponse:
             if out:
                 yield out
 
-    async def read(self) -> bytes:
+    async def read(self, idle_timeout: float = 0) -> bytes:
         parts = []
-        async for c in self.iter_chunks():
+        async for c in self.iter_chunks(idle_timeout):
             parts.append(c)
         return b"".join(parts)
 

@@ -171,6 +171,11 @@ class Backend:
     body_mutation: Optional[BodyMutation] = None
     # Per-try timeout seconds (extensionserver/post_translate_modify.go:206-300)
     timeout_s: float = 60.0
+    # Per-try idle timeout: max gap without upstream bytes. Before the
+    # first response byte it resets the try and fallback proceeds; after
+    # bytes have flowed the stream is cut (ai_gateway_route.go
+    # StreamIdleTimeout -> per_try_idle_timeout). 0 = off.
+    stream_idle_timeout_s: float = 0.0
     # Circuit breaking: max in-flight requests to this backend (parity with
     # the Envoy cluster circuit breakers the reference relies on; 0 = off).
     # A saturated backend counts as a failed attempt and fallback proceeds.
