@@ -418,12 +418,31 @@ def translate_crds(docs: list[dict]) -> Config:
                     auth=auth,
                 )
             )
-        # route-level OAuth JWT validation (securityPolicy.oauth) is not
-        # implemented; the bearer-token gate + resource-metadata challenge
-        # covers the WWW-Authenticate discovery flow.
+        # route-level OAuth JWT validation (securityPolicy.oauth ->
+        # MCPOAuth): inline/local JWKS only; remote JWKS needs egress, so
+        # it is reported loudly rather than silently leaving the gate open
+        oauth = None
+        rsp = spec.get("securityPolicy") or {}
+        oa = rsp.get("oauth")
+        if oa:
+            jwks = oa.get("jwks") or {}
+            local = jwks.get("localJWKS") or {}
+            inline = local.get("inline", "")
+            if not inline and jwks.get("remoteJWKS"):
+                raise ConfigError(
+                    f"MCPRoute {md.get('name')!r}: remoteJWKS requires network "
+                    "egress; provide jwks.localJWKS.inline instead"
+                )
+            from aigw.filterapi.config import MCPOAuth
+
+            oauth = MCPOAuth(
+                issuer=oa.get("issuer", ""),
+                audiences=oa.get("audiences") or [],
+                jwks=inline,
+            )
         mcp_routes.append(
             MCPRoute(name=md.get("name", "mcp"), path=spec.get("path", "/mcp"),
-                     backends=mcp_backends)
+                     backends=mcp_backends, oauth=oauth)
         )
 
     handled = {

@@ -239,10 +239,23 @@ class MCPBackend:
 
 
 @dataclass
+class MCPOAuth:
+    """JWT validation of client access tokens (mcp_route.go securityPolicy
+    oauth: issuer/audiences/jwks). JWKS is inline JSON or a file path —
+    remote JWKS discovery needs egress and is not available here."""
+
+    issuer: str = ""
+    audiences: list[str] = field(default_factory=list)
+    jwks: str = ""       # inline JWKS JSON (localJWKS analogue)
+    jwks_file: str = ""
+
+
+@dataclass
 class MCPRoute:
     name: str
     path: str = "/mcp"
     backends: list[MCPBackend] = field(default_factory=list)
+    oauth: Optional[MCPOAuth] = None
     # client authorization at the gateway (mcpconfig.go:102-117): requests
     # must carry this bearer token; failures answer 401 with an OAuth
     # protected-resource-metadata WWW-Authenticate challenge when
@@ -412,6 +425,10 @@ def load_config(data: object) -> Config:
                     bkw["auth"] = BackendAuth(**_dc(BackendAuth, bkw["auth"], "auth"))
                 backends.append(MCPBackend(**bkw))
             rkw["backends"] = backends
+            if rkw.get("oauth") is not None:
+                rkw["oauth"] = MCPOAuth(
+                    **_dc(MCPOAuth, rkw["oauth"], f"mcp.routes[{i}].oauth")
+                )
             routes.append(MCPRoute(**rkw))
         mkw["routes"] = routes
         kw["mcp"] = MCPConfig(**mkw)

@@ -56,6 +56,16 @@ class MCPProxy:
         self.tracer = tracer  # aigw.tracing.Tracer or None
         self._client_factory = client_factory
         self._session: Optional[aiohttp.ClientSession] = None
+        self._jwt = None
+        if getattr(route, "oauth", None) is not None:
+            from aigw.mcp.jwt_auth import JWTValidator
+
+            # raises at startup on unusable JWKS (config error, not a
+            # silent open gate)
+            self._jwt = JWTValidator.from_config(
+                route.oauth.issuer, route.oauth.audiences,
+                jwks_json=route.oauth.jwks, jwks_file=route.oauth.jwks_file,
+            )
         self._tool_res: dict[str, list] = {}
         for b in route.backends:
             self._tool_res[b.name] = [re.compile(p) for p in b.tool_exclude + b.tool_include]
@@ -145,13 +155,27 @@ class MCPProxy:
     # ---- handlers ------------------------------------------------------------
 
     def _authorize(self, request: web.Request):
-        """Bearer gate; 401 carries the OAuth protected-resource metadata
-        pointer (authorization.go WWW-Authenticate behavior)."""
+        """Bearer / JWT gate; 401 carries the OAuth protected-resource
+        metadata pointer (authorization.go WWW-Authenticate behavior)."""
+        if self._jwt is not None:
+            got = request.headers.get("authorization", "")
+            if got.startswith("Bearer "):
+                from aigw.mcp.jwt_auth import JWTError
+
+                try:
+                    self._jwt.validate(got[7:])
+                    return None
+                except JWTError:
+                    pass
+            return self._deny(request)
         if not self.route.bearer_token:
             return None
         got = request.headers.get("authorization", "")
         if got == f"Bearer {self.route.bearer_token}":
             return None
+        return self._deny(request)
+
+    def _deny(self, request: web.Request):
         resp = web.json_response(
             {"jsonrpc": "2.0", "id": None,
              "error": {"code": -32001, "message": "unauthorized"}},
