@@ -153,11 +153,51 @@ def test_issuer_and_audience_checks():
     v.validate(_token(_claims(aud=["other", "mcp-api"])))
 
 
-def test_jwks_parsing_skips_non_rsa():
+def test_jwks_parsing_skips_unsupported():
     doc = json.loads(_jwks())
-    doc["keys"].append({"kty": "EC", "crv": "P-256", "x": "AA", "y": "AA"})
+    doc["keys"].append({"kty": "EC", "crv": "P-384", "x": "AA", "y": "AA"})
+    doc["keys"].append({"kty": "oct", "k": "AA"})
     keys = parse_jwks(doc)
     assert len(keys) == 1 and keys[0].kid == "k1"
+
+
+def test_es256_sign_verify_roundtrip():
+    from aigw.mcp import jwt_auth as ja
+
+    rng = random.Random(7)
+    d = rng.randrange(1, ja._N)
+    pub = ja._ec_mul(d, (ja._GX, ja._GY))
+
+    def es_sign(signing_input: bytes) -> bytes:
+        e = int.from_bytes(hashlib.sha256(signing_input).digest(), "big") % ja._N
+        while True:
+            k = rng.randrange(1, ja._N)
+            pt = ja._ec_mul(k, (ja._GX, ja._GY))
+            r = pt[0] % ja._N
+            if r == 0:
+                continue
+            s = pow(k, -1, ja._N) * (e + r * d) % ja._N
+            if s:
+                return r.to_bytes(32, "big") + s.to_bytes(32, "big")
+
+    jwks = json.dumps({"keys": [{
+        "kty": "EC", "crv": "P-256", "use": "sig", "kid": "ec1",
+        "x": _b64url(pub[0].to_bytes(32, "big")),
+        "y": _b64url(pub[1].to_bytes(32, "big")),
+    }]})
+    v = JWTValidator.from_config("https://auth.example", ["mcp-api"], jwks_json=jwks)
+    header = _b64url(json.dumps({"alg": "ES256", "typ": "JWT", "kid": "ec1"}).encode())
+    payload = _b64url(json.dumps(_claims()).encode())
+    sig = es_sign(f"{header}.{payload}".encode())
+    assert v.validate(f"{header}.{payload}.{_b64url(sig)}")["sub"] == "u1"
+    # tampered signature rejected
+    bad = bytes([sig[5] ^ 1]) + sig[1:] if False else sig[:-1] + bytes([sig[-1] ^ 1])
+    with pytest.raises(JWTError, match="signature"):
+        v.validate(f"{header}.{payload}.{_b64url(bad)}")
+    # RS256 token cannot pass against an EC-only JWKS
+    rs_tok = _token(_claims())
+    with pytest.raises(JWTError, match="signature"):
+        v.validate(rs_tok)
 
 
 def test_mcp_proxy_gate_end_to_end():
