@@ -250,3 +250,4 @@ def test_semantic_cache_fp8_end_to_end(hip):
     hits = cache.lookup(vecs)
     assert hits[0] is not None and cache.get(hits[0][0]) == b"MADRID"
     assert hits[1] is None
+
