@@ -20,6 +20,7 @@
 // vectorized bf16 loads, grid-stride loops sized for 256 CUs.
 
 #include <hip/hip_runtime.h>
+#include <algorithm>
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -755,6 +756,112 @@ cache_topk_kernel_t(const bf16* __restrict__ index, long long n_rows,
     atomicMax(&best[threadIdx.x], blk_best[threadIdx.x]);
 }
 
+// LDS-staged variant: PMC on the register-tile kernel above shows a
+// 27:1 SQ_WAIT:SQ_BUSY ratio — the serial L2 B-load -> MFMA dependency
+// chain in its query loop leaves waves stalled on ~300-cycle L2 hits.
+// Staging the whole (<=128-query) block in LDS once per CTA turns those
+// into ~30-cycle ds_reads the scheduler hides behind MFMAs; the index
+// still streams HBM once per 128-query pass (the host chunks 256-query
+// calls into two passes). Row stride padded by 8 halves so the 16 query
+// rows of a q-tile land on distinct banks (KSTEPS*32+8 halves = 4 mod 64
+// dwords -> banks 4*row mod 64, conflict-free for 16 rows).
+template <int KSTEPS, int ROWTILES>
+__global__ void __launch_bounds__(256)
+cache_topk_lds_kernel_t(const bf16* __restrict__ index, long long n_rows,
+                        const bf16* __restrict__ q, int n_q /* <= 128 */, int dim,
+                        unsigned long long* __restrict__ best) {
+  constexpr int DIMP = KSTEPS * 32 + 8;
+  extern __shared__ unsigned char smem[];
+  bf16* qs = reinterpret_cast<bf16*>(smem);
+  unsigned long long* blk_best =
+      reinterpret_cast<unsigned long long*>(smem + 128 * DIMP * sizeof(bf16));
+  if (threadIdx.x < 128) blk_best[threadIdx.x] = 0;
+  int chunks_per_row = dim / 8;
+  for (int idx = threadIdx.x; idx < n_q * chunks_per_row; idx += 256) {
+    int r = idx / chunks_per_row, c = (idx - r * chunks_per_row) * 8;
+    *reinterpret_cast<short8*>(&qs[r * DIMP + c]) =
+        *reinterpret_cast<const short8*>(&q[(long long)r * dim + c]);
+  }
+  __syncthreads();
+  int wave = threadIdx.x >> 6;
+  int lane = threadIdx.x & 63;
+  int row = lane & 15;
+  int kgrp = lane >> 4;
+  // A-tile software pipeline: at 1 CTA/CU (the 100 KB query stage) a
+  // wave that loads a row tile and only then computes exposes the full
+  // HBM latency every tile. Ping-pong buffers (compile-time indexed —
+  // rule #20) let tile t+1's 12 dwordx4 loads fly while tile t's 96
+  // MFMAs issue.
+  auto load_tile = [&](short8 (&frag)[KSTEPS], int t) {
+    long long i0 = ((long long)blockIdx.x * 4 * ROWTILES + wave * ROWTILES + t) * 16;
+    bool i_ok = i0 < n_rows && (i0 + row) < n_rows;
+    #pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      int kk = ks * 32 + kgrp * 8;
+      if (i_ok)
+        frag[ks] = *reinterpret_cast<const short8*>(&index[(i0 + row) * dim + kk]);
+      else
+        frag[ks] = short8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  };
+  // ks-outer with one accumulator per query tile: at 1 CTA/CU the q-inner
+  // form serializes on each MFMA's ~5-cycle result latency (12-deep acc
+  // dependency chain per q-tile, nothing else in flight); QT independent
+  // chains issued back-to-back hide it completely.
+  constexpr int QT = 8;  // 128 queries / 16
+  auto compute_tile = [&](short8 (&frag)[KSTEPS], int t) {
+    long long i0 = ((long long)blockIdx.x * 4 * ROWTILES + wave * ROWTILES + t) * 16;
+    if (i0 >= n_rows) return;
+    floatx4 acc[QT];
+    #pragma unroll
+    for (int qt = 0; qt < QT; ++qt) acc[qt] = floatx4{0.f, 0.f, 0.f, 0.f};
+    int qrow = min(row, n_q - 1);  // LDS reads always in-bounds
+    #pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      int kk = ks * 32 + kgrp * 8;
+      #pragma unroll
+      for (int qt = 0; qt < QT; ++qt) {
+        short8 b = *reinterpret_cast<const short8*>(&qs[(qt * 16 + qrow) * DIMP + kk]);
+        acc[qt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(frag[ks], b, acc[qt], 0, 0, 0);
+      }
+    }
+    int lim = (int)min((long long)16, n_rows - i0);
+    #pragma unroll
+    for (int qt = 0; qt < QT; ++qt) {
+      int q0 = qt * 16;
+      bool q_in = (q0 + row) < n_q;
+      float best_s = -1e30f;
+      int best_r = 0;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = (kgrp * 4 + r < lim) ? acc[qt][r] : -1e30f;
+        if (s > best_s) { best_s = s; best_r = r; }
+      }
+      long long irow = i0 + kgrp * 4 + best_r;
+      if (!q_in) best_s = -1e30f;
+      unsigned long long p = pack_score(best_s, (unsigned)(irow & 0xFFFFFFFF));
+      #pragma unroll
+      for (int off = 16; off < 64; off <<= 1) {
+        unsigned long long o = shfl_xor_u64(p, off);
+        if (o > p) p = o;
+      }
+      if (kgrp == 0 && q_in) atomicMax(&blk_best[q0 + row], p);
+    }
+  };
+  short8 frag_a[KSTEPS], frag_b[KSTEPS];
+  load_tile(frag_a, 0);
+  static_assert(ROWTILES % 2 == 0, "pipeline assumes even ROWTILES");
+  for (int t = 0; t < ROWTILES; t += 2) {
+    load_tile(frag_b, t + 1);
+    compute_tile(frag_a, t);
+    if (t + 2 < ROWTILES) load_tile(frag_a, t + 2);
+    compute_tile(frag_b, t + 1);
+  }
+  __syncthreads();
+  if (threadIdx.x < (unsigned)n_q && blk_best[threadIdx.x])
+    atomicMax(&best[threadIdx.x], blk_best[threadIdx.x]);
+}
+
 // ---------------------------------------------------------------------------
 // 4. KV-occupancy endpoint scorer: greedy sequential assignment of a request
 //    batch to replicas. One wave; lane = replica. score = w_kv*(free KV frac
@@ -1082,30 +1189,47 @@ std::vector<at::Tensor> cache_topk(at::Tensor index, at::Tensor q) {
   }
   auto* index_p = reinterpret_cast<bf16*>(index.data_ptr<at::BFloat16>());
   auto* q_p = reinterpret_cast<bf16*>(q.data_ptr<at::BFloat16>());
-  switch (dim >> 5) {
-    case 12:  // dim = 384 (bge-small) — the hot path
-      hipLaunchKernelGGL((cache_topk_kernel_t<12, ROWTILES>), dim3((unsigned)blocks),
-                         dim3(256), 0, current_stream(), index_p, n_rows, q_p, n_q,
-                         dim, best_p);
-      break;
-    case 8:  // dim = 256
-      hipLaunchKernelGGL((cache_topk_kernel_t<8, ROWTILES>), dim3((unsigned)blocks),
-                         dim3(256), 0, current_stream(), index_p, n_rows, q_p, n_q,
-                         dim, best_p);
-      break;
-    case 16:  // dim = 512
-      hipLaunchKernelGGL((cache_topk_kernel_t<16, ROWTILES>), dim3((unsigned)blocks),
-                         dim3(256), 0, current_stream(), index_p, n_rows, q_p, n_q,
-                         dim, best_p);
-      break;
-    case 4:  // dim = 128
-      hipLaunchKernelGGL((cache_topk_kernel_t<4, ROWTILES>), dim3((unsigned)blocks),
-                         dim3(256), 0, current_stream(), index_p, n_rows, q_p, n_q,
-                         dim, best_p);
-      break;
-    default:
-      TORCH_CHECK(false, "cache_topk: unsupported dim ", dim,
-                  " (supported: 128/256/384/512)");
+  // bf16 path: LDS-staged kernel, <=128 queries per pass (the query block
+  // lives in LDS; see cache_topk_lds_kernel_t). The index streams once
+  // per pass — two passes for 256 queries still beat the register-tile
+  // kernel's latency stalls by a wide margin.
+  auto launch_lds = [&](auto kernel, int kq, const bf16* qp, unsigned long long* bp,
+                        size_t lds_bytes) {
+    (void)hipFuncSetAttribute(reinterpret_cast<const void*>(kernel),
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              160 * 1024);
+    hipLaunchKernelGGL(kernel, dim3((unsigned)blocks), dim3(256), lds_bytes,
+                       current_stream(), index_p, n_rows, qp, kq, dim, bp);
+  };
+  for (int q0 = 0; q0 < n_q; q0 += 128) {
+    int kq = std::min(128, n_q - q0);
+    const bf16* qp = q_p + (long long)q0 * dim;
+    unsigned long long* bp = best_p + q0;
+    switch (dim >> 5) {
+      case 12: {  // dim = 384 (bge-small) — the hot path
+        constexpr size_t L = 128 * (12 * 32 + 8) * sizeof(bf16) + 128 * 8;
+        launch_lds(&cache_topk_lds_kernel_t<12, ROWTILES>, kq, qp, bp, L);
+        break;
+      }
+      case 8: {  // dim = 256
+        constexpr size_t L = 128 * (8 * 32 + 8) * sizeof(bf16) + 128 * 8;
+        launch_lds(&cache_topk_lds_kernel_t<8, ROWTILES>, kq, qp, bp, L);
+        break;
+      }
+      case 16: {  // dim = 512
+        constexpr size_t L = 128 * (16 * 32 + 8) * sizeof(bf16) + 128 * 8;
+        launch_lds(&cache_topk_lds_kernel_t<16, ROWTILES>, kq, qp, bp, L);
+        break;
+      }
+      case 4: {  // dim = 128
+        constexpr size_t L = 128 * (4 * 32 + 8) * sizeof(bf16) + 128 * 8;
+        launch_lds(&cache_topk_lds_kernel_t<4, ROWTILES>, kq, qp, bp, L);
+        break;
+      }
+      default:
+        TORCH_CHECK(false, "cache_topk: unsupported dim ", dim,
+                    " (supported: 128/256/384/512)");
+    }
   }
   // unpack: score = orderable^-1(hi32), idx = lo32
   auto hi = best.bitwise_right_shift(32).to(at::kLong);
