@@ -613,132 +613,72 @@ __device__ __forceinline__ unsigned long long pack_score(float s, unsigned idx) 
   return ((unsigned long long)u << 32) | idx;
 }
 
-// fp8 (OCP e4m3) variant of the fused cache lookup: same structure as the
-// bf16 kernel but 1-byte elements — half the index HBM/L2 traffic and 2x
-// the rows per GB of the 288 GB budget. Same MFMA family at K=32 per step
-// (operands are 8 packed fp8 = one i64 per lane; C/D layout is
-// dtype-independent on gfx950 per the HIP guide).
+// fp8 LDS-staged variant — same structure as cache_topk_lds_kernel_t
+// below (query block in LDS, A-tile ping-pong, ks-outer interleave) with
+// 1-byte elements: the 128-query stage is only ~50 KB, so 3 CTAs/CU fit
+// and occupancy recovers on top of the latency fixes. Row stride padded
+// to KSTEPS*32+16 bytes (100 dwords = 36 mod 64 -> 16 distinct banks).
 template <int KSTEPS, int ROWTILES>
 __global__ void __launch_bounds__(256)
-cache_topk_fp8_kernel_t(const uint8_t* __restrict__ index, long long n_rows,
-                        const uint8_t* __restrict__ q, int n_q, int dim,
-                        unsigned long long* __restrict__ best /* n_q <= 256 */) {
-  __shared__ unsigned long long blk_best[256];
-  if (threadIdx.x < (unsigned)n_q) blk_best[threadIdx.x] = 0;
+cache_topk_fp8_lds_kernel_t(const uint8_t* __restrict__ index, long long n_rows,
+                            const uint8_t* __restrict__ q, int n_q /* <= 128 */,
+                            int dim, unsigned long long* __restrict__ best) {
+  constexpr int DIMP = KSTEPS * 32 + 16;
+  extern __shared__ unsigned char smem[];
+  uint8_t* qs = smem;
+  unsigned long long* blk_best =
+      reinterpret_cast<unsigned long long*>(smem + 128 * DIMP);
+  if (threadIdx.x < 128) blk_best[threadIdx.x] = 0;
+  int chunks_per_row = dim / 8;
+  for (int idx = threadIdx.x; idx < n_q * chunks_per_row; idx += 256) {
+    int r = idx / chunks_per_row, c = (idx - r * chunks_per_row) * 8;
+    *reinterpret_cast<long*>(&qs[r * DIMP + c]) =
+        *reinterpret_cast<const long*>(&q[(long long)r * dim + c]);
+  }
   __syncthreads();
   int wave = threadIdx.x >> 6;
   int lane = threadIdx.x & 63;
   int row = lane & 15;
   int kgrp = lane >> 4;
-  for (int t = 0; t < ROWTILES; ++t) {
+  auto load_tile = [&](long (&frag)[KSTEPS], int t) {
     long long i0 = ((long long)blockIdx.x * 4 * ROWTILES + wave * ROWTILES + t) * 16;
-    bool tile_ok = i0 < n_rows;
-    bool i_ok = tile_ok && (i0 + row) < n_rows;
-    long a_frag[KSTEPS];
+    bool i_ok = i0 < n_rows && (i0 + row) < n_rows;
     #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
       int kk = ks * 32 + kgrp * 8;
-      a_frag[ks] = i_ok
+      frag[ks] = i_ok
           ? *reinterpret_cast<const long*>(&index[(i0 + row) * dim + kk])
           : 0L;
     }
-    if (!tile_ok) continue;
-    for (int q0 = 0; q0 < n_q; q0 += 16) {
-      floatx4 acc = {0.f, 0.f, 0.f, 0.f};
-      bool q_in = (q0 + row) < n_q;
-      #pragma unroll
-      for (int ks = 0; ks < KSTEPS; ++ks) {
-        long b = q_in ? *reinterpret_cast<const long*>(
-                            &q[(long long)(q0 + row) * dim + ks * 32 + kgrp * 8])
-                      : 0L;
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a_frag[ks], b, acc, 0, 0, 0);
-      }
-      int lim = (int)min((long long)16, n_rows - i0);
-      float best_s = -1e30f;
-      int best_r = 0;
-      #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float s = (kgrp * 4 + r < lim) ? acc[r] : -1e30f;
-        if (s > best_s) { best_s = s; best_r = r; }
-      }
-      long long irow = i0 + kgrp * 4 + best_r;
-      if (!q_in) best_s = -1e30f;
-      unsigned long long p = pack_score(best_s, (unsigned)(irow & 0xFFFFFFFF));
-      #pragma unroll
-      for (int off = 16; off < 64; off <<= 1) {
-        unsigned long long o = shfl_xor_u64(p, off);
-        if (o > p) p = o;
-      }
-      if (kgrp == 0 && q_in) atomicMax(&blk_best[q0 + row], p);
-    }
-  }
-  __syncthreads();
-  if (threadIdx.x < (unsigned)n_q && blk_best[threadIdx.x])
-    atomicMax(&best[threadIdx.x], blk_best[threadIdx.x]);
-}
-
-
-// Hoists this wave's 16-row index tile into registers ONCE (KSTEPS is a
-// template constant so a_frag stays in VGPRs — a runtime bound would spill
-// to scratch and re-read it per query block, rule #20); the query loop then
-// touches only the L2-resident query block. Global atomics are first folded
-// across the block's 4 waves through LDS (4x fewer device-scope atomicMax
-// on the hot n_q addresses).
-template <int KSTEPS, int ROWTILES>
-__global__ void __launch_bounds__(256)
-cache_topk_kernel_t(const bf16* __restrict__ index, long long n_rows,
-                    const bf16* __restrict__ q, int n_q, int dim,
-                    unsigned long long* __restrict__ best /* n_q <= 256 */) {
-  __shared__ unsigned long long blk_best[256];
-  if (threadIdx.x < (unsigned)n_q) blk_best[threadIdx.x] = 0;
-  __syncthreads();
-  int wave = threadIdx.x >> 6;
-  int lane = threadIdx.x & 63;
-  int row = lane & 15;
-  int kgrp = lane >> 4;
-  // Each wave walks ROWTILES 16-row index tiles, keeping each tile in
-  // VGPRs (KSTEPS is compile-time so a_frag never spills — rule #20) and
-  // folding per-query maxima into ONE shared table per block: two
-  // barriers and n_q global atomics per block instead of per wave-tile.
-  for (int t = 0; t < ROWTILES; ++t) {
+  };
+  constexpr int QT = 8;
+  auto compute_tile = [&](long (&frag)[KSTEPS], int t) {
     long long i0 = ((long long)blockIdx.x * 4 * ROWTILES + wave * ROWTILES + t) * 16;
-    bool tile_ok = i0 < n_rows;
-    bool i_ok = tile_ok && (i0 + row) < n_rows;
-    short8 a_frag[KSTEPS];
+    if (i0 >= n_rows) return;
+    floatx4 acc[QT];
+    #pragma unroll
+    for (int qt = 0; qt < QT; ++qt) acc[qt] = floatx4{0.f, 0.f, 0.f, 0.f};
+    int qrow = min(row, n_q - 1);
     #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
       int kk = ks * 32 + kgrp * 8;
-      if (i_ok)
-        a_frag[ks] = *reinterpret_cast<const short8*>(&index[(i0 + row) * dim + kk]);
-      else
-        a_frag[ks] = short8{0, 0, 0, 0, 0, 0, 0, 0};
-    }
-    if (!tile_ok) continue;
-    for (int q0 = 0; q0 < n_q; q0 += 16) {
-      floatx4 acc = {0.f, 0.f, 0.f, 0.f};
-      bool q_in = (q0 + row) < n_q;
       #pragma unroll
-      for (int ks = 0; ks < KSTEPS; ++ks) {
-        short8 b = {0, 0, 0, 0, 0, 0, 0, 0};
-        int kk = ks * 32 + kgrp * 8;
-        if (q_in)
-          b = *reinterpret_cast<const short8*>(&q[(long long)(q0 + row) * dim + kk]);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[ks], b, acc, 0, 0, 0);
+      for (int qt = 0; qt < QT; ++qt) {
+        long b = *reinterpret_cast<const long*>(&qs[(qt * 16 + qrow) * DIMP + kk]);
+        acc[qt] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(frag[ks], b, acc[qt], 0, 0, 0);
       }
-      // acc reg r holds C[idx_row = kgrp*4+r][query = row]. PMC showed the
-      // old 4x pack_score + u64 compare chain made this kernel VALU-bound
-      // (9:1 VALU:MFMA); fold the argmax in float domain first and pack
-      // exactly once per lane per q0, then reduce across the 4 lane groups
-      // sharing this query column (lanes differing in bits 4..5).
-      // out-of-range C rows carry 0.0 (zeroed A fragments), which could
-      // out-rank valid negative scores — mask them before the argmax
-      int lim = (int)min((long long)16, n_rows - i0);
+    }
+    int lim = (int)min((long long)16, n_rows - i0);
+    #pragma unroll
+    for (int qt = 0; qt < QT; ++qt) {
+      int q0 = qt * 16;
+      bool q_in = (q0 + row) < n_q;
       float best_s = -1e30f;
       int best_r = 0;
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float s = (kgrp * 4 + r < lim) ? acc[r] : -1e30f;
-        if (s > best_s) { best_s = s; best_r = r; }
+        float sc = (kgrp * 4 + r < lim) ? acc[qt][r] : -1e30f;
+        if (sc > best_s) { best_s = sc; best_r = r; }
       }
       long long irow = i0 + kgrp * 4 + best_r;
       if (!q_in) best_s = -1e30f;
@@ -750,6 +690,15 @@ cache_topk_kernel_t(const bf16* __restrict__ index, long long n_rows,
       }
       if (kgrp == 0 && q_in) atomicMax(&blk_best[q0 + row], p);
     }
+  };
+  long frag_a[KSTEPS], frag_b[KSTEPS];
+  load_tile(frag_a, 0);
+  static_assert(ROWTILES % 2 == 0, "pipeline assumes even ROWTILES");
+  for (int t = 0; t < ROWTILES; t += 2) {
+    load_tile(frag_b, t + 1);
+    compute_tile(frag_a, t);
+    if (t + 2 < ROWTILES) load_tile(frag_a, t + 2);
+    compute_tile(frag_b, t + 1);
   }
   __syncthreads();
   if (threadIdx.x < (unsigned)n_q && blk_best[threadIdx.x])
@@ -1158,30 +1107,43 @@ std::vector<at::Tensor> cache_topk(at::Tensor index, at::Tensor q) {
   auto* best_p = reinterpret_cast<unsigned long long*>(best.data_ptr<int64_t>());
   if (fp8) {
     auto* ip = reinterpret_cast<const uint8_t*>(index.data_ptr());
-    auto* qp = reinterpret_cast<const uint8_t*>(q.data_ptr());
-    switch (dim >> 5) {
-      case 12:
-        hipLaunchKernelGGL((cache_topk_fp8_kernel_t<12, ROWTILES>),
-                           dim3((unsigned)blocks), dim3(256), 0, current_stream(),
-                           ip, n_rows, qp, n_q, dim, best_p);
-        break;
-      case 8:
-        hipLaunchKernelGGL((cache_topk_fp8_kernel_t<8, ROWTILES>),
-                           dim3((unsigned)blocks), dim3(256), 0, current_stream(),
-                           ip, n_rows, qp, n_q, dim, best_p);
-        break;
-      case 16:
-        hipLaunchKernelGGL((cache_topk_fp8_kernel_t<16, ROWTILES>),
-                           dim3((unsigned)blocks), dim3(256), 0, current_stream(),
-                           ip, n_rows, qp, n_q, dim, best_p);
-        break;
-      case 4:
-        hipLaunchKernelGGL((cache_topk_fp8_kernel_t<4, ROWTILES>),
-                           dim3((unsigned)blocks), dim3(256), 0, current_stream(),
-                           ip, n_rows, qp, n_q, dim, best_p);
-        break;
-      default:
-        TORCH_CHECK(false, "cache_topk fp8: unsupported dim ", dim);
+    auto* q_base = reinterpret_cast<const uint8_t*>(q.data_ptr());
+    auto launch_fp8 = [&](auto kernel, int kq, const uint8_t* qp,
+                          unsigned long long* bp, size_t lds_bytes) {
+      (void)hipFuncSetAttribute(reinterpret_cast<const void*>(kernel),
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                160 * 1024);
+      hipLaunchKernelGGL(kernel, dim3((unsigned)blocks), dim3(256), lds_bytes,
+                         current_stream(), ip, n_rows, qp, kq, dim, bp);
+    };
+    for (int q0 = 0; q0 < n_q; q0 += 128) {
+      int kq = std::min(128, n_q - q0);
+      const uint8_t* qp = q_base + (long long)q0 * dim;
+      unsigned long long* bp = best_p + q0;
+      switch (dim >> 5) {
+        case 12: {
+          constexpr size_t L = 128 * (12 * 32 + 16) + 128 * 8;
+          launch_fp8(&cache_topk_fp8_lds_kernel_t<12, ROWTILES>, kq, qp, bp, L);
+          break;
+        }
+        case 8: {
+          constexpr size_t L = 128 * (8 * 32 + 16) + 128 * 8;
+          launch_fp8(&cache_topk_fp8_lds_kernel_t<8, ROWTILES>, kq, qp, bp, L);
+          break;
+        }
+        case 16: {
+          constexpr size_t L = 128 * (16 * 32 + 16) + 128 * 8;
+          launch_fp8(&cache_topk_fp8_lds_kernel_t<16, ROWTILES>, kq, qp, bp, L);
+          break;
+        }
+        case 4: {
+          constexpr size_t L = 128 * (4 * 32 + 16) + 128 * 8;
+          launch_fp8(&cache_topk_fp8_lds_kernel_t<4, ROWTILES>, kq, qp, bp, L);
+          break;
+        }
+        default:
+          TORCH_CHECK(false, "cache_topk fp8: unsupported dim ", dim);
+      }
     }
     auto hi = best.bitwise_right_shift(32).to(at::kLong);
     auto idx = best.bitwise_and(0xFFFFFFFFLL).to(at::kInt);
