@@ -25,7 +25,6 @@ import asyncio
 import json
 import logging
 import time
-import uuid
 from typing import Optional
 
 from aiohttp import web
@@ -49,6 +48,20 @@ except ImportError:  # pragma: no cover - built by setup.py everywhere
     _native = None
 
 logger = logging.getLogger("aigw.server")
+
+# request-id generation: uuid4() costs ~3-5 us per call (fork-safe RNG +
+# string formatting) and runs once per unary response when the client
+# sent no x-request-id — a random prefix + counter is unique enough for
+# log correlation at a fraction of the cost
+import itertools as _itertools
+import os as _os
+
+_REQ_ID_PREFIX = _os.urandom(6).hex()
+_req_id_counter = _itertools.count()
+
+
+def _next_request_id() -> str:
+    return f"{_REQ_ID_PREFIX}-{next(_req_id_counter):x}"
 access_logger = logging.getLogger("aigw.access")
 # per-request JSON access lines are opt-in (AIGW_ACCESS_LOG=1 in the CLI)
 if access_logger.level == logging.NOTSET:
@@ -857,7 +870,8 @@ class GatewayServer:
         )
         resp = web.Response(body=rtl.body, status=upstream.status)
         resp.headers["content-type"] = content_type
-        resp.headers["x-request-id"] = headers.get("x-request-id", str(uuid.uuid4()))
+        rid = headers.get("x-request-id")
+        resp.headers["x-request-id"] = rid if rid is not None else _next_request_id()
         if cache_key_vec is not None and upstream.status == 200:
             await self.gpu.cache_insert(cache_key_vec, b"U" + rtl.body)
         return resp
