@@ -251,3 +251,33 @@ def test_semantic_cache_fp8_end_to_end(hip):
     assert hits[0] is not None and cache.get(hits[0][0]) == b"MADRID"
     assert hits[1] is None
 
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dim", [128, 256, 384, 512])
+@pytest.mark.parametrize("n_q", [1, 17, 128, 129, 255, 256])
+def test_cache_topk_shapes_and_chunking(hip, dim, n_q):
+    """Every supported dim x query-count edge: exercises the host's
+    128-query chunked launches (129 -> 128+1 passes) and the LDS kernels'
+    partial-tile masking for both dtypes."""
+    torch.manual_seed(dim * 1000 + n_q)
+    n_rows = 4100  # not a multiple of the 16x16 tile grid
+    index = torch.nn.functional.normalize(
+        torch.randn(n_rows, dim, device="cuda"), dim=1
+    ).to(torch.bfloat16)
+    q = torch.nn.functional.normalize(
+        torch.randn(n_q, dim, device="cuda"), dim=1
+    ).to(torch.bfloat16)
+    from aigw.ops.semcache import _unorder
+
+    for dtype in (torch.bfloat16, torch.float8_e4m3fn):
+        idx_t = index.to(dtype)
+        q_t = q.to(dtype)
+        hi, idx = hip.cache_topk(idx_t, q_t)
+        scores = idx_t.to(torch.float32) @ q_t.to(torch.float32).t()
+        ref_val = scores.max(dim=0).values
+        tol = 2e-2 if dtype == torch.bfloat16 else 8e-2
+        for b in range(n_q):
+            got = _unorder(int(hi[b]))
+            assert abs(got - float(ref_val[b])) < tol, (dtype, b, got, float(ref_val[b]))
+            assert abs(float(scores[int(idx[b]), b]) - float(ref_val[b])) < tol
