@@ -61,7 +61,6 @@ _IGNORED_KINDS = {
     "EnvoyExtensionPolicy",
     "SecurityPolicy",
     "InferencePool",
-    "BackendTLSPolicy",
     "EnvoyPatchPolicy",
     "GRPCRoute",
     "Job",
@@ -187,6 +186,30 @@ def _bsp_for_backend(bundle: _Bundle, ns: str, asb_name: str) -> Optional[Backen
     return None
 
 
+def _apply_tls_policy(bundle: _Bundle, ns: str, backend_name: str, up: Upstream) -> Upstream:
+    """BackendTLSPolicy targeting this Backend: TLS on, SNI/verify hostname
+    from validation.hostname, and a private trust anchor from
+    caCertificateRefs (ConfigMap ca.crt) when not wellKnownCACertificates."""
+    for pol in bundle.by_kind.get("BackendTLSPolicy", []):
+        spec = pol.get("spec") or {}
+        refs = spec.get("targetRefs") or []
+        if not any(r.get("kind") == "Backend" and r.get("name") == backend_name
+                   for r in refs):
+            continue
+        v = spec.get("validation") or {}
+        up.tls = True
+        if v.get("hostname"):
+            up.hostname = v["hostname"]
+        for cref in v.get("caCertificateRefs") or []:
+            cm = bundle.find(cref.get("kind", "ConfigMap"), cref.get("name", ""), ns)
+            if cm is not None:
+                pem = (cm.get("data") or {}).get("ca.crt", "")
+                if pem:
+                    up.ca_pem = pem
+        break
+    return up
+
+
 def _resolve_upstream(bundle: _Bundle, ns: str, backend_ref: dict) -> Upstream:
     name = backend_ref.get("name", "")
     kind = backend_ref.get("kind", "Backend")
@@ -205,9 +228,11 @@ def _resolve_upstream(bundle: _Bundle, ns: str, backend_ref: dict) -> Upstream:
     if "fqdn" in ep:
         host = ep["fqdn"].get("hostname", "")
         port = ep["fqdn"].get("port", 443)
-        return Upstream(host=host, port=port, tls=port == 443, hostname=host)
+        up = Upstream(host=host, port=port, tls=port == 443, hostname=host)
+        return _apply_tls_policy(bundle, ns, name, up)
     if "ip" in ep:
-        return Upstream(host=ep["ip"].get("address", ""), port=ep["ip"].get("port", 80))
+        up = Upstream(host=ep["ip"].get("address", ""), port=ep["ip"].get("port", 80))
+        return _apply_tls_policy(bundle, ns, name, up)
     if "unix" in ep:
         raise ConfigError("unix-socket backends are not supported")
     raise ConfigError(f"Backend {name!r}: unknown endpoint type")
@@ -448,6 +473,7 @@ def translate_crds(docs: list[dict]) -> Config:
     handled = {
         "AIGatewayRoute", "AIServiceBackend", "BackendSecurityPolicy", "Backend",
         "BackendTrafficPolicy", "GatewayConfig", "QuotaPolicy", "MCPRoute", "Secret",
+        "BackendTLSPolicy",
         "Gateway",
     }
     for kind, docs_of_kind in bundle.by_kind.items():

@@ -718,6 +718,8 @@ class GatewayServer:
                     body=out_body,
                     timeout_s=backend.timeout_s,
                     server_name=backend.upstream.hostname,
+                    ca_file=backend.upstream.ca_file,
+                    ca_pem=backend.upstream.ca_pem,
                 )
                 status = upstream.status
                 if status >= 400:
@@ -1022,6 +1024,8 @@ class GatewayServer:
                     body=tr.body,
                     timeout_s=backend.timeout_s,
                     server_name=backend.upstream.hostname,
+                    ca_file=backend.upstream.ca_file,
+                    ca_pem=backend.upstream.ca_pem,
                 )
                 if upstream.status >= 400:
                     err = await upstream.read()

@@ -150,11 +150,15 @@ class _Connection:
 
 
 class _OriginPool:
-    def __init__(self, host: str, port: int, tls: bool, server_name: str = ""):
+    def __init__(self, host: str, port: int, tls: bool, server_name: str = "",
+                 ca_file: str = "", ca_pem: str = ""):
         self.host = host
         self.port = port
         self.tls = tls
         self.server_name = server_name or host
+        self.ca_file = ca_file
+        self.ca_pem = ca_pem
+        self._ssl_ctx = None
         self._idle: list[_Connection] = []
 
     async def acquire(self) -> _Connection:
@@ -165,7 +169,14 @@ class _OriginPool:
             conn.writer.close()
         ssl_ctx = None
         if self.tls:
-            ssl_ctx = ssl_mod.create_default_context()
+            if self._ssl_ctx is None:
+                # custom trust anchor for private upstreams (BackendTLSPolicy
+                # caCertificateRefs analogue); system store otherwise
+                self._ssl_ctx = ssl_mod.create_default_context(
+                    cafile=self.ca_file or None,
+                    cadata=self.ca_pem or None,
+                )
+            ssl_ctx = self._ssl_ctx
         reader, writer = await asyncio.open_connection(
             self.host, self.port, ssl=ssl_ctx,
             server_hostname=self.server_name if self.tls else None,
@@ -195,11 +206,12 @@ class LeanClient:
     def __init__(self):
         self._pools: dict[tuple[str, int, bool], _OriginPool] = {}
 
-    def _pool(self, host: str, port: int, tls: bool, server_name: str = "") -> _OriginPool:
-        key = (host, port, tls)
+    def _pool(self, host: str, port: int, tls: bool, server_name: str = "",
+              ca_file: str = "", ca_pem: str = "") -> _OriginPool:
+        key = (host, port, tls, ca_file, bool(ca_pem))
         p = self._pools.get(key)
         if p is None:
-            p = _OriginPool(host, port, tls, server_name)
+            p = _OriginPool(host, port, tls, server_name, ca_file, ca_pem)
             self._pools[key] = p
         return p
 
@@ -214,14 +226,18 @@ class LeanClient:
         body: bytes,
         timeout_s: float = 60.0,
         server_name: str = "",
+        ca_file: str = "",
+        ca_pem: str = "",
     ) -> LeanResponse:
         return await asyncio.wait_for(
-            self._post(host, port, tls, path, headers, body, server_name),
+            self._post(host, port, tls, path, headers, body, server_name,
+                       ca_file, ca_pem),
             timeout=timeout_s,
         )
 
-    async def _post(self, host, port, tls, path, headers, body, server_name) -> LeanResponse:
-        pool = self._pool(host, port, tls, server_name)
+    async def _post(self, host, port, tls, path, headers, body, server_name,
+                    ca_file="", ca_pem="") -> LeanResponse:
+        pool = self._pool(host, port, tls, server_name, ca_file, ca_pem)
         conn = await pool.acquire()
         try:
             authority = headers.get("host") or (

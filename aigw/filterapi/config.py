@@ -143,7 +143,12 @@ class Upstream:
     port: int = 443
     tls: bool = False
     path_prefix: str = ""  # prepended to the translated path
-    hostname: str = ""  # Host header / SigV4 signing host; defaults to host
+    hostname: str = ""  # Host header / TLS SNI / SigV4 signing host; defaults to host
+    # Custom trust anchor for private upstreams (the BackendTLSPolicy
+    # caCertificateRefs analogue): PEM file path or inline PEM. Empty =
+    # system trust store. Verification is never disabled.
+    ca_file: str = ""
+    ca_pem: str = ""
 
     @property
     def authority(self) -> str:
