@@ -20,7 +20,6 @@ from __future__ import annotations
 import hashlib
 import time
 from dataclasses import dataclass
-from typing import Optional
 
 from aigw.filterapi.config import RateLimitRule
 

@@ -10,7 +10,6 @@ stream event. Also the Bedrock embeddings variant
 
 from __future__ import annotations
 
-import base64
 import json
 import time
 from typing import Optional

@@ -27,7 +27,7 @@ from aigw.translator.base import (
     register,
     usage_from_openai,
 )
-from aigw.translator.sse import SSEDecoder, SSEEvent
+from aigw.translator.sse import SSEDecoder
 
 
 class _OpenAIPassthrough(Translator):

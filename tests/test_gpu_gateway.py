@@ -2,7 +2,6 @@
 loop — GPU token accounting, semantic cache hit/miss, GPU tokenize API."""
 
 import asyncio
-import json
 
 import aiohttp
 import pytest

@@ -7,7 +7,6 @@ import asyncio
 import json
 
 import aiohttp
-import pytest
 
 from aigw.extproc.lean_front import serve_lean
 from aigw.extproc.server import GatewayServer

@@ -3,7 +3,6 @@ flight simultaneously — guards the shared-state paths (limiter, metrics,
 per-request translators) against cross-request interference."""
 
 import asyncio
-import json
 import random
 
 import aiohttp

@@ -11,7 +11,6 @@ import sys
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-import pytest
 
 from aigw.filterapi.config import APISchemaName
 from aigw.translator import get_translator
