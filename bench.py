@@ -318,9 +318,10 @@ def main():
     t0 = time.perf_counter()
     go.set()
     results = [out_q.get(timeout=600) for _ in range(workers)]
-    elapsed = time.perf_counter() - t0
+    # closing bracket: all GPU work retired before the clock stops
     if use_gpu:
         torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
     barrier_sync()
     for p in procs:
         p.join(timeout=60)
