@@ -328,7 +328,28 @@ def main():
         if p.is_alive():
             p.terminate()
 
+    statesync_tick_us = None
     if world > 1:
+        # demonstrate the RCCL-synced rate-limit counters on the same
+        # communicator the scale bench uses (BASELINE config: "RCCL-synced
+        # rate-limit counters"): a fixed number of fused all-reduce ticks
+        # per rank, timed. Deterministic tick count keeps the collective
+        # sequence identical across ranks (no timer-driven ticks racing
+        # the final all-reduce).
+        from aigw.filterapi import RuntimeConfig as _RC
+        from aigw.filterapi import load_config as _lc
+        from aigw.parallel import StateSync
+        from aigw.ratelimit import RateLimiter
+
+        lim = RateLimiter(_RC(_lc(gateway_config(9))).rate_limits)
+        lim.check({})  # touch the bucket so deltas exist
+        sync = StateSync(lim)
+        sync.tick_sync()  # warm
+        ts0 = time.perf_counter()
+        for _ in range(20):
+            lim.charge({}, {"llm_total_token": 17})
+            sync.tick_sync()
+        statesync_tick_us = (time.perf_counter() - ts0) / 20 * 1e6
         t = torch.tensor([elapsed], dtype=torch.float64)
         if torch.distributed.get_backend() == "nccl":
             t = t.cuda()
@@ -368,6 +389,7 @@ def main():
                 "p50_direct_ms": round(p50_direct, 3),
                 "p50_added_latency_ms": round(p50 - p50_direct, 3),
                 "gpu_token_accounting": use_gpu,
+                "statesync_tick_us": round(statesync_tick_us, 1) if statesync_tick_us else None,
                 "semantic_cache_payload_pool": args.cache_payloads or None,
             },
         }
