@@ -138,8 +138,15 @@ struct Scan {
     return false;
   }
 
-  bool parse_value(int depth, const std::string& key, bool at_root) {
+  bool parse_value(int depth, const std::string& key, bool at_root,
+                   bool in_msgs = false) {
     if (depth > 64 || !ok) return fail();
+    // chat text is only collected inside the top-level "messages" tree
+    // (plus a top-level "system" string) so stray "text"/"content" keys
+    // elsewhere in the body never leak into admission/cache-key text —
+    // keeps this fast path byte-identical with the strict extractor
+    // (aigw/gpu/services.py extract_chat_text)
+    in_msgs = in_msgs || (depth == 1 && key == "messages");
     ws();
     if (p >= end) return fail();
     char c = *p;
@@ -154,7 +161,7 @@ struct Scan {
         ws();
         if (p >= end || *p != ':') return fail();
         ++p;
-        if (!parse_value(depth + 1, k, false)) return false;
+        if (!parse_value(depth + 1, k, false, in_msgs)) return false;
         ws();
         if (p < end && *p == ',') { ++p; continue; }
         if (p < end && *p == '}') { ++p; return true; }
@@ -167,7 +174,7 @@ struct Scan {
       ws();
       if (p < end && *p == ']') { ++p; return true; }
       while (p < end) {
-        if (!parse_value(depth + 1, key, false)) return false;
+        if (!parse_value(depth + 1, key, false, in_msgs)) return false;
         ws();
         if (p < end && *p == ',') { ++p; continue; }
         if (p < end && *p == ']') { ++p; return true; }
@@ -177,7 +184,7 @@ struct Scan {
     }
     if (c == '"') {
       bool is_model = depth == 1 && key == "model";
-      bool is_text = key == "content" || key == "text" ||
+      bool is_text = (in_msgs && (key == "content" || key == "text")) ||
                      (depth == 1 && key == "system");
       if (is_model) return parse_string(&model);
       if (is_text) {

@@ -61,6 +61,12 @@ static void test_scan_extraction() {
   CHECK(r.ok);
   CHECK(r.text == std::string("q\n\"A\xF0\x9D\x84\x9E\n"));
 
+  // text/content OUTSIDE the messages tree is never collected (keeps
+  // the fast path identical to the strict python extractor)
+  r = scan(R"({"model":"m","metadata":{"text":"noise","content":"x"},)"
+           R"("messages":[{"content":"real"}]})");
+  CHECK(r.ok && r.text == "real\n");
+
   // stream:false and absent stream
   CHECK(!scan(R"({"model":"m","stream":false})").stream);
   CHECK(!scan(R"({"model":"m"})").stream);
