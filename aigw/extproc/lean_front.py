@@ -207,7 +207,7 @@ class LeanFront(asyncio.Protocol):
         except Exception:
             logger.exception("lean front request failed")
             self._write_simple(500, {"content-type": "application/json"},
-                               b'{"error":{"message":"internal error"}}')
+                               b'{"type":"error","error":{"type":"internal_error","code":"500","message":"internal error"}}')
 
     def _write_simple(self, status: int, headers: dict, body: bytes,
                       keep_alive: bool = True):
@@ -233,7 +233,7 @@ class LeanFront(asyncio.Protocol):
         if server.draining and path not in ("/health", "/metrics"):
             self._write_simple(
                 503, {"content-type": "application/json"},
-                b'{"error":{"message":"shutting down","type":"unavailable"}}',
+                b'{"type":"error","error":{"type":"unavailable","code":"503","message":"shutting down"}}',
                 keep_alive=False,
             )
             return
@@ -299,7 +299,7 @@ class LeanFront(asyncio.Protocol):
         aiohttp instance serving the full app."""
         if self.fallback_port is None:
             self._write_simple(404, {"content-type": "application/json"},
-                               b'{"error":{"message":"not found"}}')
+                               b'{"type":"error","error":{"type":"not_found","code":"404","message":"not found"}}')
             return
         if method != "POST":
             self._write_simple(405, {"content-type": "application/json"},

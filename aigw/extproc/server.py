@@ -203,8 +203,12 @@ async def _prepend(first, aiter):
 
 
 def _json_error(status: int, message: str, err_type: str = "invalid_request_error") -> web.Response:
+    # gateway-generated ("local reply") error envelope, wire-compatible
+    # with the reference's formatUserFacingErrorJSON
+    # (processor_impl.go:190-196): {"type":"error","error":{type,code,message}}
     return web.json_response(
-        {"error": {"message": message, "type": err_type, "code": str(status)}},
+        {"type": "error",
+         "error": {"type": err_type, "code": str(status), "message": message}},
         status=status,
     )
 
