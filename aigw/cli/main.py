@@ -85,7 +85,9 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
 
                 gpu_host = GPUServiceHost(gpu_services, gpu_socket)
                 await gpu_host.start()
-    server = GatewayServer(runtime, gpu_services=gpu_services)
+    server = GatewayServer(
+        runtime, gpu_services=gpu_services, root_prefix=getattr(args, "root_prefix", "")
+    )
 
     sync = None
     if world > 1:
@@ -238,6 +240,8 @@ def main(argv=None) -> int:
     runp.add_argument("--host", default="0.0.0.0")
     runp.add_argument("--port", type=int, default=internalapi.DEFAULT_LISTEN_PORT)
     runp.add_argument("--shards", type=int, default=1, help="shards (one per GPU)")
+    runp.add_argument("--root-prefix", default="", dest="root_prefix",
+                      help="global path prefix for every endpoint")
     runp.add_argument("--workers", type=int, default=1,
                       help="HTTP worker processes per shard (SO_REUSEPORT; "
                            "CPython's GIL caps one loop near ~3k req/s)")

@@ -248,6 +248,19 @@ class LeanFront(asyncio.Protocol):
     async def _handle_inner(self, req):
         method, path, headers, body = req
         server = self.server
+        root = server.root_prefix
+        if root:
+            if path.startswith(root + "/"):
+                path = path[len(root):]
+            elif path not in ("/health", "/metrics", "/debug/tasks"):
+                # admin endpoints stay unprefixed; everything else must
+                # carry the root prefix (mainlib --rootPrefix semantics)
+                self._write_simple(
+                    404, {"content-type": "application/json"},
+                    b'{"type":"error","error":{"type":"not_found",'
+                    b'"code":"404","message":"not found"}}',
+                )
+                return
         if method == "POST" and path in JSON_ENDPOINTS:
             view = RequestView(
                 method=method,

@@ -224,11 +224,15 @@ class GatewayServer:
         tracer=None,
         max_body_bytes: int = 50 * 1024 * 1024,  # reference raises Envoy's buffer to 50MiB
         endpoint_prefixes: Optional[dict[str, str]] = None,
+        root_prefix: str = "",
+        # global path prefix joined onto every endpoint (mainlib
+        # --rootPrefix; main.go path.Join(flags.rootPrefix, ...))
         # extra path prefixes per API family (mainlib --endpointPrefixes,
         # main.go:143-147), e.g. {"openai": "/openai"} additionally mounts
         # /openai/v1/chat/completions etc.
     ):
         self.endpoint_prefixes = endpoint_prefixes or {}
+        self.root_prefix = root_prefix.rstrip("/")
         self.runtime = runtime
         self.metrics = metrics or GenAIMetrics()
         self.limiter = limiter or RateLimiter(runtime.rate_limits)
@@ -308,22 +312,24 @@ class GatewayServer:
             client_max_size=self.max_body_bytes, middlewares=[self._drain_middleware]
         )
 
+        root = self.root_prefix
+
         def mount(path: str, handler) -> None:
-            app.router.add_post(path, handler)
+            app.router.add_post(root + path, handler)
             for family, prefix in self.endpoint_prefixes.items():
                 if family == "anthropic" and path.startswith("/anthropic/"):
-                    app.router.add_post(prefix + path[len("/anthropic") :], handler)
+                    app.router.add_post(root + prefix + path[len("/anthropic") :], handler)
                 elif family == "cohere" and path.startswith("/v2/"):
-                    app.router.add_post(prefix + path, handler)
+                    app.router.add_post(root + prefix + path, handler)
                 elif family == "openai" and not path.startswith(("/anthropic/", "/v2/")):
-                    app.router.add_post(prefix + path, handler)
+                    app.router.add_post(root + prefix + path, handler)
 
         for ep in JSON_ENDPOINTS:
             mount(ep, self._make_handler(ep))
         for ep in MULTIPART_ENDPOINTS:
             mount(ep, self._make_multipart_handler(ep))
-        app.router.add_get("/v1/models", self._handle_models)
-        app.router.add_get("/anthropic/v1/models", self._handle_anthropic_models)
+        app.router.add_get(root + "/v1/models", self._handle_models)
+        app.router.add_get(root + "/anthropic/v1/models", self._handle_anthropic_models)
         app.router.add_get("/health", self._handle_health)
         app.router.add_get("/metrics", self._handle_metrics)
         app.router.add_get("/debug/tasks", self._handle_debug_tasks)

@@ -277,3 +277,73 @@ def test_endpoint_prefixes():
         await runner.cleanup()
 
     asyncio.run(main())
+
+
+def test_root_prefix_mounts_everything():
+    """mainlib --rootPrefix: every endpoint is served under the global
+    prefix (path.Join(flags.rootPrefix, ...)), on both fronts."""
+    import asyncio
+    import json as _json
+
+    import yaml as _yaml
+
+    from aigw.extproc.lean_front import serve_lean
+    from aigw.extproc.server import GatewayServer, run_server
+    from aigw.extproc.upstream_client import LeanClient
+    from aigw.filterapi.config import load_config
+    from aigw.filterapi.runtime import RuntimeConfig
+    from aigw.testing.fastmock import start_fast_mock
+
+    async def run():
+        up_srv, up_port = await start_fast_mock("127.0.0.1", 0)
+        cfg = load_config(_yaml.safe_load(f"""
+routes:
+  - name: r
+    backends:
+      - name: b
+        schema: OpenAI
+        upstream: {{host: 127.0.0.1, port: {up_port}}}
+models: [{{name: m}}]
+"""))
+        server = GatewayServer(RuntimeConfig(cfg), root_prefix="/ai/v2")
+        await server.start()
+        runner = await run_server(server, host="127.0.0.1", port=0)
+        aio_port = runner.addresses[0][1]
+        _, lean_port, lean_cleanup = await serve_lean(
+            server, "127.0.0.1", 0, with_fallback=False
+        )
+        body = _json.dumps(
+            {"model": "m", "messages": [{"role": "user", "content": "q"}]}
+        ).encode()
+        client = LeanClient()
+        for port in (aio_port, lean_port):
+            r = await client.post(
+                host="127.0.0.1", port=port, tls=False,
+                path="/ai/v2/v1/chat/completions",
+                headers={"content-type": "application/json"}, body=body,
+            )
+            data = await r.read()
+            assert r.status == 200, (port, r.status, data[:200])
+            r.release()
+            # unprefixed path 404s
+            r = await client.post(
+                host="127.0.0.1", port=port, tls=False,
+                path="/v1/chat/completions",
+                headers={"content-type": "application/json"}, body=body,
+            )
+            await r.read()
+            assert r.status == 404, port
+            r.close()
+        import aiohttp
+
+        async with aiohttp.ClientSession() as s:
+            async with s.get(f"http://127.0.0.1:{aio_port}/ai/v2/v1/models") as r:
+                assert r.status == 200
+                assert (await r.json())["data"][0]["id"] == "m"
+        await client.close()
+        await lean_cleanup()
+        await runner.cleanup()
+        await server.close()
+        up_srv.close()
+
+    asyncio.run(run())
