@@ -108,6 +108,19 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
     runner = await run_server(
         server, host=args.host, port=port, reuse_port=getattr(args, "workers", 1) > 1
     )
+    admin_runner = None
+    admin_port = getattr(args, "admin_port", 0)
+    if admin_port:
+        # separate admin server for /health + /metrics (mainlib --adminPort;
+        # keeps scrapers and probes off the data-plane listener)
+        from aiohttp import web as _web
+
+        admin_app = _web.Application()
+        admin_app.router.add_get("/health", server._handle_health)
+        admin_app.router.add_get("/metrics", server._handle_metrics)
+        admin_runner = _web.AppRunner(admin_app, access_log=None)
+        await admin_runner.setup()
+        await _web.TCPSite(admin_runner, "127.0.0.1", admin_port + rank).start()
     print(f"aigw shard {rank}/{world} listening on http://{args.host}:{port}", flush=True)
     stop = asyncio.Event()
     loop = asyncio.get_running_loop()
@@ -131,6 +144,8 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
             await sync.stop()
         if gpu_host is not None:
             await gpu_host.stop()
+        if admin_runner is not None:
+            await admin_runner.cleanup()
         await runner.cleanup()
 
 
@@ -242,6 +257,8 @@ def main(argv=None) -> int:
     runp.add_argument("--shards", type=int, default=1, help="shards (one per GPU)")
     runp.add_argument("--root-prefix", default="", dest="root_prefix",
                       help="global path prefix for every endpoint")
+    runp.add_argument("--admin-port", type=int, default=0, dest="admin_port",
+                      help="separate localhost admin server for /health and /metrics (0 = serve on the data port only)")
     runp.add_argument("--workers", type=int, default=1,
                       help="HTTP worker processes per shard (SO_REUSEPORT; "
                            "CPython's GIL caps one loop near ~3k req/s)")

@@ -342,7 +342,8 @@ class GatewayServer:
             self._mcp_proxies = []
             for mr in mcp_cfg.routes:
                 proxy = MCPProxy(mr, mcp_cfg.session_seed, metrics=self.metrics,
-                                 tracer=self.tracer)
+                                 tracer=self.tracer,
+                                 fallback_seed=mcp_cfg.fallback_session_seed)
                 self._mcp_proxies.append(proxy)
                 for method in ("POST", "GET", "DELETE"):
                     app.router.add_route(method, mr.path, proxy.handle)

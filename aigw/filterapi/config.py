@@ -273,6 +273,9 @@ class MCPRoute:
 class MCPConfig:
     routes: list[MCPRoute] = field(default_factory=list)
     session_seed: str = "aigw-mcp"
+    # previous seed(s) still accepted for DEcryption during seed rotation
+    # (mainlib --mcpFallbackSessionEncryptionSeed)
+    fallback_session_seed: str = ""
 
 
 @dataclass

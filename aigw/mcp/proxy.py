@@ -49,9 +49,11 @@ def _rpc_error(id_, code: int, message: str, status: int = 200) -> web.Response:
 
 class MCPProxy:
     def __init__(self, route: MCPRoute, session_seed: str, client_factory=None,
-                 metrics=None, tracer=None):
+                 metrics=None, tracer=None, fallback_seed: str = ""):
         self.route = route
-        self.crypto = SessionCrypto(session_seed)
+        self.crypto = SessionCrypto(
+            session_seed, fallback_seeds=[fallback_seed] if fallback_seed else None
+        )
         self.metrics = metrics  # aigw.metrics.GenAIMetrics or None
         self.tracer = tracer  # aigw.tracing.Tracer or None
         self._client_factory = client_factory

@@ -234,3 +234,21 @@ def test_mcp_gateway_end_to_end():
         await r2.cleanup()
 
     asyncio.run(main())
+
+
+def test_session_seed_rotation():
+    """mainlib --mcpFallbackSessionEncryptionSeed: sessions minted under
+    the OLD seed keep decrypting after the seed rotates."""
+    from aigw.mcp.session import SessionCrypto
+
+    old = SessionCrypto("seed-v1")
+    token = old.seal({"backends": {"a": "s1"}})
+    rotated = SessionCrypto("seed-v2", fallback_seeds=["seed-v1"])
+    assert rotated.open(token) == {"backends": {"a": "s1"}}
+    # and new sessions use the new seed
+    t2 = rotated.seal({"x": 1})
+    assert SessionCrypto("seed-v2").open(t2) == {"x": 1}
+    import pytest as _pytest
+
+    with _pytest.raises(Exception):
+        SessionCrypto("seed-v3").open(token)
