@@ -68,7 +68,13 @@ def _content_to_anthropic(content) -> list:
     for part in content:
         t = part.get("type")
         if t == "text":
-            blocks.append({"type": "text", "text": part.get("text", "")})
+            block = {"type": "text", "text": part.get("text", "")}
+            # provider-agnostic prompt caching: cache_control rides through
+            # to Anthropic-family backends untouched (the reference's
+            # unified cache_control API, capabilities/prompt-caching.md)
+            if part.get("cache_control"):
+                block["cache_control"] = part["cache_control"]
+            blocks.append(block)
         elif t == "image_url":
             url = (part.get("image_url") or {}).get("url", "")
             if url.startswith("data:"):
@@ -194,8 +200,13 @@ def openai_to_anthropic_request(body: dict) -> dict:
                 "type": "tool",
                 "name": (choice.get("function") or {}).get("name", ""),
             }
+    # unified `thinking` extension field (vendor-specific-fields.md):
+    # passed to Anthropic-family backends verbatim, overriding the
+    # reasoning_effort mapping below
+    if isinstance(body.get("thinking"), dict):
+        out["thinking"] = body["thinking"]
     # reasoning_effort -> thinking budget (gcp/aws anthropic thinking map)
-    effort = body.get("reasoning_effort")
+    effort = None if "thinking" in out else body.get("reasoning_effort")
     if effort:
         budgets = {"minimal": 1024, "low": 1024, "medium": 8192, "high": 24576}
         if effort in budgets:

@@ -57,6 +57,11 @@ def _content_to_converse(content) -> list:
         t = part.get("type")
         if t == "text":
             blocks.append({"text": part.get("text", "")})
+            # unified prompt caching (prompt-caching.md): a cache_control
+            # breakpoint becomes a Converse cachePoint block AFTER the
+            # cached content
+            if part.get("cache_control"):
+                blocks.append({"cachePoint": {"type": "default"}})
         elif t == "image_url":
             url = (part.get("image_url") or {}).get("url", "")
             if not url.startswith("data:"):
@@ -74,6 +79,10 @@ def openai_to_converse_request(body: dict) -> dict:
     """OpenAI ChatCompletionRequest → Bedrock ConverseRequest
     (openai_awsbedrock.go:103-284)."""
     out: dict = {}
+    # unified `thinking` extension field (vendor-specific-fields.md):
+    # Bedrock carries model-native fields in additionalModelRequestFields
+    if isinstance(body.get("thinking"), dict):
+        out["additionalModelRequestFields"] = {"thinking": body["thinking"]}
     system: list[dict] = []
     messages: list[dict] = []
     for msg in body.get("messages", []):

@@ -141,8 +141,21 @@ def openai_to_gemini_request(body: dict) -> dict:
     rf = body.get("response_format")
     if isinstance(rf, dict) and rf.get("type") == "json_object":
         gen["responseMimeType"] = "application/json"
+    # extension fields (vendor-specific-fields.md): unified `thinking` ->
+    # generationConfig.thinkingConfig; `safetySettings` passed through
+    th = body.get("thinking")
+    if isinstance(th, dict):
+        tc = {}
+        if th.get("budget_tokens") is not None:
+            tc["thinkingBudget"] = th["budget_tokens"]
+        if th.get("type") == "enabled":
+            tc.setdefault("includeThoughts", True)
+        if tc:
+            gen["thinkingConfig"] = tc
     if gen:
         out["generationConfig"] = gen
+    if body.get("safetySettings"):
+        out["safetySettings"] = body["safetySettings"]
 
     tools = body.get("tools")
     if tools:

@@ -166,3 +166,51 @@ def test_max_completion_tokens_priority():
     assert openai_to_anthropic_request(req)["max_tokens"] == 99
     assert openai_to_converse_request(req)["inferenceConfig"]["maxTokens"] == 99
     assert openai_to_gemini_request(req)["generationConfig"]["maxOutputTokens"] == 99
+
+
+def test_extension_fields_thinking_and_cache_control():
+    """vendor-specific-fields.md + prompt-caching.md: the unified
+    `thinking` field and `cache_control` breakpoints translate per
+    backend."""
+    import json as _json
+
+    from aigw.filterapi.config import APISchemaName
+    from aigw.translator import get_translator
+
+    base = {
+        "model": "m",
+        "thinking": {"type": "enabled", "budget_tokens": 2048},
+        "messages": [{"role": "user", "content": [
+            {"type": "text", "text": "big context",
+             "cache_control": {"type": "ephemeral"}},
+            {"type": "text", "text": "question"},
+        ]}],
+    }
+    # -> Anthropic: thinking verbatim, cache_control on the block
+    t = get_translator("/v1/chat/completions", APISchemaName.ANTHROPIC)
+    body = _json.loads(t.request(_json.loads(_json.dumps(base))).body)
+    assert body["thinking"] == {"type": "enabled", "budget_tokens": 2048}
+    blocks = body["messages"][0]["content"]
+    assert blocks[0]["cache_control"] == {"type": "ephemeral"}
+    assert "cache_control" not in blocks[1]
+
+    # -> Bedrock Converse: thinking in additionalModelRequestFields,
+    # cache_control becomes a cachePoint block after the cached text
+    t = get_translator("/v1/chat/completions", APISchemaName.AWS_BEDROCK)
+    body = _json.loads(t.request(_json.loads(_json.dumps(base))).body)
+    assert body["additionalModelRequestFields"]["thinking"]["budget_tokens"] == 2048
+    content = body["messages"][0]["content"]
+    assert content[0] == {"text": "big context"}
+    assert content[1] == {"cachePoint": {"type": "default"}}
+    assert content[2] == {"text": "question"}
+
+    # -> Gemini: thinking -> generationConfig.thinkingConfig;
+    # safetySettings passthrough
+    t = get_translator("/v1/chat/completions", APISchemaName.GCP_VERTEX_AI,
+                       gcp_project="p", gcp_region="r")
+    gbase = _json.loads(_json.dumps(base))
+    gbase["safetySettings"] = [{"category": "HARM_CATEGORY_HARASSMENT",
+                                "threshold": "BLOCK_NONE"}]
+    body = _json.loads(t.request(gbase).body)
+    assert body["generationConfig"]["thinkingConfig"]["thinkingBudget"] == 2048
+    assert body["safetySettings"][0]["threshold"] == "BLOCK_NONE"
