@@ -160,7 +160,13 @@ def openai_to_gemini_request(body: dict) -> dict:
     tools = body.get("tools")
     if tools:
         decls = []
+        extra_tools = []
         for t in tools:
+            if t.get("type") == "google_search":
+                # Google Search grounding (vendor-specific-fields.md:
+                # openai.go ToolTypeGoogleSearch) — config passes through
+                extra_tools.append({"googleSearch": t.get("google_search") or {}})
+                continue
             if t.get("type") != "function":
                 continue
             fn = t.get("function") or {}
@@ -171,7 +177,7 @@ def openai_to_gemini_request(body: dict) -> dict:
                     "parameters": fn.get("parameters") or {"type": "object"},
                 }
             )
-        out["tools"] = [{"functionDeclarations": decls}]
+        out["tools"] = ([{"functionDeclarations": decls}] if decls else []) + extra_tools
         choice = body.get("tool_choice")
         mode = None
         allowed = None

@@ -214,3 +214,19 @@ def test_extension_fields_thinking_and_cache_control():
     body = _json.loads(t.request(gbase).body)
     assert body["generationConfig"]["thinkingConfig"]["thinkingBudget"] == 2048
     assert body["safetySettings"][0]["threshold"] == "BLOCK_NONE"
+
+
+def test_gemini_google_search_tool():
+    import json as _json
+
+    from aigw.filterapi.config import APISchemaName
+    from aigw.translator import get_translator
+
+    t = get_translator("/v1/chat/completions", APISchemaName.GCP_VERTEX_AI,
+                       gcp_project="p", gcp_region="r")
+    body = _json.loads(t.request({
+        "model": "gemini", "messages": [{"role": "user", "content": "q"}],
+        "tools": [{"type": "google_search",
+                   "google_search": {"excludeDomains": ["x.com"]}}],
+    }).body)
+    assert body["tools"] == [{"googleSearch": {"excludeDomains": ["x.com"]}}]
