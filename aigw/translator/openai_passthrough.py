@@ -231,6 +231,11 @@ class OpenAIToAzureEmbeddings(_AzurePathMixin, _OpenAIPassthrough):
     AZURE_SUFFIX = "embeddings"
 
 
+@register("/v1/completions", APISchemaName.AZURE_OPENAI)
+class OpenAIToAzureCompletions(_AzurePathMixin, _OpenAIPassthrough):
+    AZURE_SUFFIX = "completions"
+
+
 @register("/v1/responses", APISchemaName.AZURE_OPENAI)
 class OpenAIToAzureResponses(OpenAIToOpenAIResponses):
     def _path(self, model: str) -> str:
