@@ -56,6 +56,10 @@ CASES = [
 ] + [
     ("/v1/embeddings", s, {"model": "base-model", "input": "embed me"})
     for s in ("OpenAI", "AzureOpenAI", "GCPVertexAI", "AWSBedrock")
+] + [
+    ("/v1/completions", s, {"model": "base-model", "prompt": "continue this",
+                            "max_tokens": 8})
+    for s in ("OpenAI", "AzureOpenAI")
 ]
 
 
