@@ -948,10 +948,11 @@ class GatewayServer:
                 and len(transcript) < (2 << 20)
             ):
                 await self.gpu.cache_insert(cache_key_vec, b"S" + bytes(transcript))
-        except UpstreamError as e:
-            # mid-stream idle: the response has started, so fallback is no
-            # longer possible — cut the stream (the client sees truncated
-            # chunked framing; Envoy's analogue is a stream reset / 504)
+        except (UpstreamError, ValueError) as e:
+            # mid-stream failure (idle timeout, or malformed provider bytes
+            # the translator/codec rejects): the response has started, so
+            # fallback is no longer possible — cut the stream (the client
+            # sees truncated framing; Envoy's analogue is a stream reset)
             cut = True
             logger.warning("cutting stream to %s: %s", backend.name, e)
         except (ConnectionResetError, asyncio.CancelledError):

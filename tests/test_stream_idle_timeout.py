@@ -182,3 +182,69 @@ def test_mid_stream_idle_cuts_stream():
 async def _collect(resp, sink):
     async for c in resp.iter_chunks():
         sink.extend(c)
+
+
+@pytest.mark.timeout(60)
+def test_malformed_provider_stream_cuts_not_500():
+    """A provider emitting garbage mid-stream (here: a corrupt Bedrock
+    event-stream frame) truncates the client stream instead of crashing
+    the worker or emitting a second response."""
+
+    async def run():
+        async def chat(request):
+            resp = web.StreamResponse(status=200)
+            resp.headers["content-type"] = "application/vnd.amazon.eventstream"
+            await resp.prepare(request)
+            await resp.write(b"\xde\xad\xbe\xef" * 64)  # not a valid frame
+            await resp.write_eof()
+            return resp
+
+        app = web.Application()
+        app.router.add_post("/model/m/converse-stream", chat)
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.TCPSite(runner, "127.0.0.1", 0)
+        await site.start()
+        up_port = site._server.sockets[0].getsockname()[1]
+
+        cfg = load_config(yaml.safe_load(f"""
+routes:
+  - name: r
+    backends:
+      - name: bedrock
+        schema: AWSBedrock
+        upstream: {{host: 127.0.0.1, port: {up_port}}}
+"""))
+        server = GatewayServer(RuntimeConfig(cfg))
+        gw = await run_server(server, host="127.0.0.1", port=0)
+        port = gw.addresses[0][1]
+        client = LeanClient()
+        body = json.dumps({"model": "m", "stream": True,
+                           "messages": [{"role": "user", "content": "q"}]}).encode()
+        r = await client.post(host="127.0.0.1", port=port, tls=False,
+                              path="/v1/chat/completions",
+                              headers={"content-type": "application/json"}, body=body)
+        assert r.status == 200  # headers were already committed
+        try:
+            await asyncio.wait_for(r.read(), timeout=10)
+        except UpstreamError:
+            pass  # truncated framing is the expected signal
+        r.close()
+        # the worker still serves (no crash): same request round-trips
+        # again, truncating the same way
+        h = await client.post(host="127.0.0.1", port=port, tls=False,
+                              path="/v1/chat/completions",
+                              headers={"content-type": "application/json"},
+                              body=body)
+        assert h.status == 200
+        try:
+            await asyncio.wait_for(h.read(), timeout=10)
+        except UpstreamError:
+            pass
+        h.close()
+        await client.close()
+        await gw.cleanup()
+        await server.close()
+        await runner.cleanup()
+
+    asyncio.run(run())
