@@ -174,3 +174,30 @@ def test_cost_arithmetic_matches_python(a, b, mul):
     v = CostVars(input_tokens=a, output_tokens=b)
     assert CostProgram("input_tokens + output_tokens * {}".format(mul)).evaluate(v) == a + b * mul
     assert CostProgram("max(input_tokens, output_tokens)").evaluate(v) == max(a, b)
+
+
+@given(
+    data=st.binary(min_size=0, max_size=400),
+    schema=st.sampled_from(["Anthropic", "AWSBedrock", "GCPVertexAI", "OpenAI"]),
+)
+@settings(max_examples=200, deadline=None)
+def test_response_chunk_never_raises_unhandled(data, schema):
+    """Streaming translators fed arbitrary provider bytes may reject them
+    (ValueError/TranslationError — the gateway cuts the stream) but must
+    never raise anything else or corrupt their internal state."""
+    from aigw.filterapi.config import APISchemaName
+    from aigw.translator import TranslationError, get_translator
+
+    t = get_translator("/v1/chat/completions", APISchemaName(schema),
+                       gcp_project="p", gcp_region="r")
+    t.request({"model": "m", "messages": [{"role": "user", "content": "q"}],
+               "stream": True}, stream=True)
+    for chunk in (data[:137], data[137:]):
+        try:
+            t.response_chunk(chunk)
+        except (ValueError, TranslationError):
+            pass  # handled by the gateway as a stream cut
+    try:
+        t.response_flush()
+    except (ValueError, TranslationError):
+        pass
