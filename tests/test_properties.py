@@ -188,9 +188,12 @@ def test_response_chunk_never_raises_unhandled(data, schema):
     from aigw.filterapi.config import APISchemaName
     from aigw.translator import TranslationError, get_translator
 
-    t = get_translator("/v1/chat/completions", APISchemaName(schema),
+    endpoint = ("/anthropic/v1/messages" if schema == "OpenAI"
+                else "/v1/chat/completions")  # covers both SSE machines
+    t = get_translator(endpoint, APISchemaName(schema),
                        gcp_project="p", gcp_region="r")
-    t.request({"model": "m", "messages": [{"role": "user", "content": "q"}],
+    t.request({"model": "m", "max_tokens": 8,
+               "messages": [{"role": "user", "content": "q"}],
                "stream": True}, stream=True)
     for chunk in (data[:137], data[137:]):
         try:
