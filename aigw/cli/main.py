@@ -48,7 +48,11 @@ def _load_any_config(path: str | None):
     with open(path, "r", encoding="utf-8") as f:
         text = f.read()
     if _is_crd_bundle(text):
-        return translate_yaml(text), None
+        from aigw.filterapi.config import _expand_env
+
+        # ${ENV} expansion applies at load time on the run path (the
+        # offline `aigw translate` keeps placeholders for GitOps)
+        return translate_yaml(_expand_env(text)), None
     return load_config_file(path), path
 
 

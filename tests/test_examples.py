@@ -43,3 +43,17 @@ def test_full_example_surface(monkeypatch):
     assert cfg.mcp.routes[0].backends[0].tool_exclude == ["delete_.*"]
     # the CEL cost compiles and references reasoning_tokens
     assert any(c.type.value == "CEL" for c in cfg.llm_request_costs)
+
+
+def test_crd_bundle_example_runs_with_env_expansion(monkeypatch):
+    monkeypatch.setenv("OPENAI_API_KEY", "sk-live-123")
+    from aigw.cli.main import _load_any_config
+
+    cfg, watch = _load_any_config(
+        os.path.join(os.path.dirname(__file__), "..", "examples", "crd_bundle.yaml")
+    )
+    assert watch is None  # CRD bundles don't hot-reload (one-shot translate)
+    b = cfg.routes[0].backends[0]
+    assert b.auth.api_key == "sk-live-123"
+    assert b.stream_idle_timeout_s == 15.0
+    assert cfg.models[0].owned_by == "my-org"
