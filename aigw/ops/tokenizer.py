@@ -37,6 +37,7 @@ class GPUTokenizer:
         # pinned staging buffers for truly-async H2D (pageable copies block)
         self._pin_bytes = None
         self._pin_offs = None
+        self._h2d_ev = None  # guards pinned-buffer reuse vs in-flight DMA
 
     @staticmethod
     def pack(texts: list[bytes]) -> tuple[np.ndarray, np.ndarray]:
@@ -57,8 +58,10 @@ class GPUTokenizer:
         embedding stage can consume them without a host round-trip.
         """
         arr, offs = self.pack(texts)
-        bytes_t = torch.from_numpy(arr).to(self.device, non_blocking=True)
-        off_t = torch.from_numpy(offs).to(self.device, non_blocking=True)
+        # stage through pinned buffers: pageable H2D copies BLOCK on ROCm
+        # (the driver bounces them through an internal staging buffer),
+        # which taxed every admission batch on the serving hot path
+        bytes_t, off_t = self._stage(arr, offs)
         out_ids, req_counts, seg_start, seg_req = self.hip.bpe_encode(
             bytes_t, off_t, self.htab_keys, self.htab_ranks
         )
@@ -72,13 +75,15 @@ class GPUTokenizer:
                 ids.append([int(x) for x in seg[seg >= 0]])
         return req_counts, ids, {"out_ids": out_ids, "req_off": off_t, "n_bytes": len(arr)}
 
-    def encode_batch_async(self, texts: list[bytes]):
-        """Sync-free batch encode: returns (req_counts_gpu, out_ids_gpu,
-        req_off_gpu, event). Await the event (cooperatively, via
-        event.query()) before reading the tensors — no host sync happens
-        here, which matters because ROCm host syncs busy-spin a core."""
-        arr, offs = self.pack(texts)
+    def _stage(self, arr, offs):
+        """Copy packed host arrays into reusable pinned buffers and issue
+        truly-async H2D copies; returns the device tensors."""
         n = len(arr)
+        if self._h2d_ev is not None:
+            # the pinned buffers may still be the source of an in-flight
+            # copy from the PREVIOUS batch; wait for that DMA (usually
+            # already complete) before overwriting them
+            self._h2d_ev.synchronize()
         if self._pin_bytes is None or self._pin_bytes.numel() < n:
             self._pin_bytes = torch.empty(max(n, 1 << 20), dtype=torch.uint8,
                                           pin_memory=True)
@@ -89,6 +94,17 @@ class GPUTokenizer:
         self._pin_offs.numpy()[: len(offs)] = offs
         bytes_t = self._pin_bytes[:n].to(self.device, non_blocking=True)
         off_t = self._pin_offs[: len(offs)].to(self.device, non_blocking=True)
+        self._h2d_ev = torch.cuda.Event()
+        self._h2d_ev.record()
+        return bytes_t, off_t
+
+    def encode_batch_async(self, texts: list[bytes]):
+        """Sync-free batch encode: returns (req_counts_gpu, out_ids_gpu,
+        req_off_gpu, event). Await the event (cooperatively, via
+        event.query()) before reading the tensors — no host sync happens
+        here, which matters because ROCm host syncs busy-spin a core."""
+        arr, offs = self.pack(texts)
+        bytes_t, off_t = self._stage(arr, offs)
         out_ids, req_counts = self.hip.bpe_count_async(
             bytes_t, off_t, self.htab_keys, self.htab_ranks
         )
