@@ -138,6 +138,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
         gpu_services = GPUServices(
             device=f"cuda:{local_rank}", n_merges=32768, enable_cache=cache_mode,
             cache_threshold=0.98, window_ms=getattr(args, "gpu_window", 0.1),
+            cache_index_dtype="fp8" if getattr(args, "cache_fp8", False) else "bf16",
             max_batch=256,
         )
 
@@ -251,6 +252,8 @@ def main():
                     help="HTTP worker processes per shard (0 = auto)")
     ap.add_argument("--no-gpu", action="store_true",
                     help="disable GPU token accounting (contention diagnosis)")
+    ap.add_argument("--cache-fp8", action="store_true",
+                    help="fp8 (e4m3) semantic-cache index instead of bf16")
     ap.add_argument("--cache-payloads", type=int, default=0,
                     help="semantic-cache mode: cycle this many distinct "
                          "payloads per worker (0 = cache off); steady-state "
