@@ -885,7 +885,13 @@ class GatewayServer:
         resp.headers["content-type"] = content_type
         rid = headers.get("x-request-id")
         resp.headers["x-request-id"] = rid if rid is not None else _next_request_id()
-        if cache_key_vec is not None and upstream.status == 200:
+        if (
+            cache_key_vec is not None
+            and upstream.status == 200
+            and len(rtl.body) < (2 << 20)
+            # same 2 MiB cap as streamed transcripts: 64k slots x unbounded
+            # bodies would otherwise grow host memory without limit
+        ):
             await self.gpu.cache_insert(cache_key_vec, b"U" + rtl.body)
         return resp
 

@@ -205,3 +205,69 @@ def test_gpu_streaming_cache_replay():
         gpu.close()
 
     asyncio.run(main())
+
+
+@pytest.mark.gpu
+def test_oversized_unary_response_not_cached():
+    """Responses over the 2 MiB cap are served but never inserted into
+    the semantic cache (bounded host/HBM value storage)."""
+    import asyncio
+
+    import yaml as _yaml
+    from aiohttp import web
+
+    from aigw.extproc.server import GatewayServer, run_server
+    from aigw.extproc.upstream_client import LeanClient
+    from aigw.filterapi.config import load_config
+    from aigw.filterapi.runtime import RuntimeConfig
+    from aigw.gpu import GPUServices
+
+    async def run():
+        big = "x" * (3 << 20)
+
+        async def chat(request):
+            return web.json_response(
+                {"id": "b", "object": "chat.completion", "model": "m",
+                 "choices": [{"index": 0, "message": {"role": "assistant",
+                                                      "content": big},
+                              "finish_reason": "stop"}],
+                 "usage": {"prompt_tokens": 1, "completion_tokens": 1,
+                           "total_tokens": 2}})
+
+        app = web.Application()
+        app.router.add_post("/v1/chat/completions", chat)
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.TCPSite(runner, "127.0.0.1", 0)
+        await site.start()
+        up_port = site._server.sockets[0].getsockname()[1]
+        cfg = load_config(_yaml.safe_load(f"""
+routes:
+  - name: r
+    backends:
+      - name: b
+        schema: OpenAI
+        upstream: {{host: 127.0.0.1, port: {up_port}}}
+"""))
+        gpu = GPUServices(device="cuda:0", enable_cache=True)
+        server = GatewayServer(RuntimeConfig(cfg), gpu_services=gpu)
+        gw = await run_server(server, host="127.0.0.1", port=0)
+        port = gw.addresses[0][1]
+        client = LeanClient()
+        body = json.dumps({"model": "m",
+                           "messages": [{"role": "user", "content": "q"}]}).encode()
+        r = await client.post(host="127.0.0.1", port=port, tls=False,
+                              path="/v1/chat/completions",
+                              headers={"content-type": "application/json"},
+                              body=body)
+        data = await r.read()
+        assert r.status == 200 and len(data) > (3 << 20)
+        r.release()
+        assert gpu.cache.size == 0  # oversized body was not inserted
+        await client.close()
+        await gw.cleanup()
+        await server.close()
+        await runner.cleanup()
+        gpu.close()
+
+    asyncio.run(run())
