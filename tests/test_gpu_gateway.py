@@ -223,6 +223,8 @@ def test_oversized_unary_response_not_cached():
     from aigw.gpu import GPUServices
 
     async def run():
+        import json
+
         big = "x" * (3 << 20)
 
         async def chat(request):
