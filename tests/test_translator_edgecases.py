@@ -230,3 +230,28 @@ def test_gemini_google_search_tool():
                    "google_search": {"excludeDomains": ["x.com"]}}],
     }).body)
     assert body["tools"] == [{"googleSearch": {"excludeDomains": ["x.com"]}}]
+
+
+def test_streaming_response_headers_rewrites():
+    """Streamed responses present client-appropriate content types:
+    Bedrock's binary event-stream becomes SSE; SSE backends stay SSE;
+    content-length never survives re-chunking (A.8)."""
+    from aigw.filterapi.config import APISchemaName
+    from aigw.translator import get_translator
+
+    req = {"model": "m", "stream": True,
+           "messages": [{"role": "user", "content": "q"}]}
+
+    t = get_translator("/v1/chat/completions", APISchemaName.AWS_BEDROCK)
+    t.request(dict(req), stream=True)
+    h = t.response_headers(200, {
+        "content-type": "application/vnd.amazon.eventstream",
+        "content-length": "12345",
+    })
+    assert h["content-type"] == "text/event-stream"
+
+    t = get_translator("/v1/chat/completions", APISchemaName.GCP_VERTEX_AI,
+                       gcp_project="p", gcp_region="r")
+    t.request(dict(req), stream=True)
+    h = t.response_headers(200, {"content-type": "text/event-stream"})
+    assert h.get("content-type", "text/event-stream") == "text/event-stream"
