@@ -110,9 +110,20 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
         await watcher.start()
 
     port = args.port + rank
-    runner = await run_server(
-        server, host=args.host, port=port, reuse_port=getattr(args, "workers", 1) > 1
-    )
+    reuse = getattr(args, "workers", 1) > 1
+    if getattr(args, "front", "lean") == "aiohttp":
+        runner = await run_server(server, host=args.host, port=port, reuse_port=reuse)
+        front_cleanup = runner.cleanup
+    else:
+        # default: the raw-asyncio lean front the benchmarks measure; the
+        # loopback aiohttp fallback keeps the cold paths (multipart, MCP)
+        # on the same port
+        from aigw.extproc.lean_front import serve_lean
+
+        _, _, lean_cleanup = await serve_lean(
+            server, args.host, port, with_fallback=True, reuse_port=reuse
+        )
+        front_cleanup = lean_cleanup
     admin_runner = None
     admin_port = getattr(args, "admin_port", 0)
     if admin_port:
@@ -151,7 +162,7 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
             await gpu_host.stop()
         if admin_runner is not None:
             await admin_runner.cleanup()
-        await runner.cleanup()
+        await front_cleanup()
 
 
 def cmd_run(args) -> int:
@@ -262,6 +273,8 @@ def main(argv=None) -> int:
     runp.add_argument("--shards", type=int, default=1, help="shards (one per GPU)")
     runp.add_argument("--root-prefix", default="", dest="root_prefix",
                       help="global path prefix for every endpoint")
+    runp.add_argument("--front", choices=["lean", "aiohttp"], default="lean",
+                      help="HTTP front: lean raw-asyncio (benchmarked default; aiohttp loopback serves multipart/MCP) or pure aiohttp")
     runp.add_argument("--admin-port", type=int, default=0, dest="admin_port",
                       help="separate localhost admin server for /health and /metrics (0 = serve on the data port only)")
     runp.add_argument("--workers", type=int, default=1,
