@@ -136,3 +136,47 @@ def test_aws_credentials_file(tmp_path):
     h = build_auth_handler(b)({}, b"{}", "POST", "/model/m/converse")
     assert "Credential=AKFILE/" in h["authorization"]
     assert "/eu-west-1/bedrock/aws4_request" in h["authorization"]
+
+
+def test_sigv4_cross_validated_independent_implementation():
+    """The production signer vs a from-first-principles SigV4 written
+    here (RFC-style 4 steps) — catches shared-bug blindness that a
+    self-consistency known-answer cannot."""
+    import hashlib
+    import hmac as hmac_mod
+    from datetime import datetime, timezone
+
+    method, path, body = "POST", "/model/claude/converse", b'{"messages":[]}'
+    host, region, service = "bedrock-runtime.us-east-1.amazonaws.com", "us-east-1", "bedrock"
+    ak, sk = "AKIDEXAMPLE", "wJalrXUtnFEMI/K7MDENG+bPxRfiCYEXAMPLEKEY"
+    now = datetime(2015, 8, 30, 12, 36, 0, tzinfo=timezone.utc)
+
+    got = sign_sigv4(method, path, {"content-type": "application/json"}, body,
+                     host=host, region=region, service=service,
+                     access_key_id=ak, secret_access_key=sk, now=now)
+
+    # independent computation
+    amz_date, datestamp = "20150830T123600Z", "20150830"
+    payload_hash = hashlib.sha256(body).hexdigest()
+    signed_names = sorted(
+        got["authorization"].split("SignedHeaders=")[1].split(",")[0].split(";")
+    )
+    all_headers = {"host": host, "x-amz-date": amz_date,
+                   "x-amz-content-sha256": payload_hash,
+                   "content-type": "application/json"}
+    canonical_headers = "".join(f"{n}:{all_headers[n]}\n" for n in signed_names)
+    canonical = "\n".join([method, path, "", canonical_headers,
+                           ";".join(signed_names), payload_hash])
+    scope = f"{datestamp}/{region}/{service}/aws4_request"
+    sts = "\n".join(["AWS4-HMAC-SHA256", amz_date, scope,
+                     hashlib.sha256(canonical.encode()).hexdigest()])
+
+    def hm(key, msg):
+        return hmac_mod.new(key, msg.encode(), hashlib.sha256).digest()
+
+    k = hm(hm(hm(hm(b"AWS4" + sk.encode(), datestamp), region), service),
+           "aws4_request")
+    expected_sig = hmac_mod.new(k, sts.encode(), hashlib.sha256).hexdigest()
+    assert got["authorization"].endswith(f"Signature={expected_sig}"), (
+        got["authorization"], expected_sig)
+    assert got["x-amz-date"] == amz_date
