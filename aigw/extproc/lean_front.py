@@ -155,6 +155,13 @@ class LeanFront(asyncio.Protocol):
             return
         self._arm_idle_guard()
 
+    def _reject(self, status_line: bytes) -> None:
+        self._transport.write(
+            b"HTTP/1.1 " + status_line + b"\r\ncontent-length: 0\r\n"
+            b"connection: close\r\n\r\n"
+        )
+        self._transport.close()
+
     def _try_parse(self):
         end = self._buf.find(b"\r\n\r\n")
         if end < 0:
@@ -169,19 +176,34 @@ class LeanFront(asyncio.Protocol):
             self._transport.close()
             return None
         headers: dict[str, str] = {}
+        clen_values: list[str] = []
+        te_present = False
         for line in lines[1:]:
             k, _, v = line.partition(b":")
-            headers[k.decode("latin1").strip().lower()] = v.decode("latin1").strip()
+            key = k.decode("latin1").strip().lower()
+            val = v.decode("latin1").strip()
+            if key == "content-length":
+                clen_values.append(val)
+            elif key == "transfer-encoding":
+                te_present = True
+            headers[key] = val
+        if te_present:
+            # This front frames requests by Content-Length only. Accepting a
+            # chunked body would reinterpret its bytes as pipelined requests
+            # (request smuggling through any intermediary that forwards TE),
+            # so chunked uploads are refused outright rather than mis-framed.
+            self._reject(b"501 Not Implemented")
+            return None
+        if len(set(clen_values)) > 1:
+            # conflicting Content-Length values: classic desync vector
+            self._reject(b"400 Bad Request")
+            return None
         try:
-            clen = int(headers.get("content-length", "0") or 0)
+            clen = int(clen_values[0]) if clen_values else 0
         except ValueError:
             clen = -1
         if clen < 0 or clen > 64 * 1024 * 1024:
-            self._transport.write(
-                b"HTTP/1.1 400 Bad Request\r\ncontent-length: 0\r\n"
-                b"connection: close\r\n\r\n"
-            )
-            self._transport.close()
+            self._reject(b"400 Bad Request")
             return None
         total = end + 4 + clen
         if len(self._buf) < total:

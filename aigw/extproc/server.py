@@ -734,7 +734,13 @@ class GatewayServer:
                 )
                 status = upstream.status
                 if status >= 400:
-                    err_body = await upstream.read()
+                    try:
+                        err_body = await upstream.read()
+                    except BaseException:
+                        # failed mid-error-body read: drop the connection
+                        # instead of leaking it into the retry loop
+                        upstream.close()
+                        raise
                     upstream.release()
                     if status in RETRIABLE_STATUSES and attempts_left > 0:
                         last_error = f"upstream {backend.name} returned {status}"
@@ -1046,7 +1052,11 @@ class GatewayServer:
                     ca_pem=backend.upstream.ca_pem,
                 )
                 if upstream.status >= 400:
-                    err = await upstream.read()
+                    try:
+                        err = await upstream.read()
+                    except BaseException:
+                        upstream.close()
+                        raise
                     upstream.release()
                     if upstream.status in RETRIABLE_STATUSES and attempts_left > 0:
                         continue

@@ -48,6 +48,14 @@ class LeanResponse:
         self._remaining = int(clen) if clen is not None else None
         self._chunked = headers.get("transfer-encoding", "").lower() == "chunked"
         self._eof = False
+        if status in (204, 304) or 100 <= status < 200:
+            # statuses defined to carry no body: EOF immediately, whatever
+            # the framing headers say — otherwise a keep-alive 204 with no
+            # Content-Length would block as close-delimited until timeout
+            # and poison the connection
+            self._eof = True
+            self._chunked = False
+            self._remaining = 0
 
     def _maybe_decompress(self, data: bytes) -> bytes:
         if self._decomp is None or not data:

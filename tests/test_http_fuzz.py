@@ -133,6 +133,66 @@ def test_broken_http_framing_never_hangs():
 
 
 @pytest.mark.timeout(120)
+def test_transfer_encoding_and_duplicate_content_length_rejected():
+    """Request-desync hardening: this front frames by Content-Length only,
+    so any Transfer-Encoding must be refused (501) and conflicting
+    Content-Length values must be refused (400) — the body bytes of a
+    chunked request must NEVER be parsed as pipelined follow-up requests."""
+
+    async def run():
+        _, port, cleanup = await _gateway()
+
+        async def send(payload: bytes) -> bytes:
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            writer.write(payload)
+            try:
+                await writer.drain()
+                return await asyncio.wait_for(reader.read(65536), timeout=6)
+            finally:
+                writer.close()
+
+        # chunked request whose body smuggles a pipelined GET /health —
+        # must be rejected wholesale, not answered twice
+        smuggle = (
+            b"POST /v1/chat/completions HTTP/1.1\r\n"
+            b"transfer-encoding: chunked\r\n\r\n"
+            b"2\r\n{}\r\n0\r\n\r\n"
+            b"GET /health HTTP/1.1\r\n\r\n"
+        )
+        data = await send(smuggle)
+        assert data.startswith(b"HTTP/1.1 501"), data[:80]
+        assert data.count(b"HTTP/1.1") == 1, "smuggled request was answered"
+
+        for te in (b"chunked", b"identity", b"gzip, chunked", b"ChUnKeD"):
+            data = await send(
+                b"POST /v1/chat/completions HTTP/1.1\r\n"
+                b"Transfer-Encoding: " + te + b"\r\ncontent-length: 2\r\n\r\n{}"
+            )
+            assert data.startswith(b"HTTP/1.1 501"), (te, data[:80])
+
+        # conflicting Content-Length values
+        data = await send(
+            b"POST /v1/chat/completions HTTP/1.1\r\n"
+            b"content-length: 2\r\ncontent-length: 5\r\n\r\n{}123"
+        )
+        assert data.startswith(b"HTTP/1.1 400"), data[:80]
+
+        # duplicate but AGREEING Content-Length is tolerated (last-win is
+        # safe when the values match)
+        body = json.dumps({"model": "m",
+                           "messages": [{"role": "user", "content": "ok"}]}).encode()
+        data = await send(
+            b"POST /v1/chat/completions HTTP/1.1\r\n"
+            b"content-length: %d\r\ncontent-length: %d\r\n"
+            b"content-type: application/json\r\n\r\n" % (len(body), len(body)) + body
+        )
+        assert data.startswith(b"HTTP/1.1 200"), data[:80]
+        await cleanup()
+
+    asyncio.run(run())
+
+
+@pytest.mark.timeout(120)
 def test_slow_and_partial_requests():
     """Byte-at-a-time delivery and mid-body disconnects must not wedge
     the worker or leak the connection slot."""

@@ -187,3 +187,33 @@ def test_request_wire_format():
         await srv.stop()
 
     run(main())
+
+
+def test_bodyless_204_304_keepalive_not_close_delimited():
+    """204/304 carry no body by definition: with neither Content-Length nor
+    chunked framing they must yield EOF immediately (not block as
+    close-delimited) and keep the connection reusable."""
+
+    async def main():
+        ok = b"HTTP/1.1 200 OK\r\ncontent-length: 2\r\n\r\nok"
+        for status in (204, 304):
+            resp = b"HTTP/1.1 %d X\r\ndate: now\r\n\r\n" % status
+            srv = ScriptedServer([(resp, False), (ok, False)])
+            port = await srv.start()
+            c = LeanClient()
+            r = await c.post(host="127.0.0.1", port=port, tls=False, path="/x",
+                             headers={}, body=b"{}")
+            assert r.status == status
+            body = await asyncio.wait_for(r.read(), timeout=2)
+            assert body == b""
+            r.release()
+            # connection stays alive and framed: the next request reuses it
+            r2 = await c.post(host="127.0.0.1", port=port, tls=False, path="/x",
+                              headers={}, body=b"{}")
+            assert r2.status == 200 and await r2.read() == b"ok"
+            r2.release()
+            assert srv.connections == 1
+            await c.close()
+            await srv.stop()
+
+    run(main())
