@@ -11,9 +11,13 @@ as internal/backendauth/auth.go:18-68 does:
                      by the GCP translators, which own the vendor path)
 - AWS credentials -> SigV4 over the final mutated body (aws.go:86-160)
 
-Per-request credential override from client headers is honored before the
-configured credentials (credential_override.go), and the override headers
-are always stripped before the request goes upstream.
+Per-request credential override is honored ONLY when the backend config
+sets ``credentialOverride`` (credential_override.go): the configured source
+header supplies the credential; when it is absent, ``fallbackToConfigured``
+selects between the static credential and a 401 (CredentialMissingError).
+Override headers are consumed by the handler and the server additionally
+strips every configured override header name at egress, so they never
+reach an upstream.
 """
 
 from __future__ import annotations
@@ -24,7 +28,18 @@ import os
 
 from aigw import internalapi
 from aigw.backendauth.sigv4 import sign_sigv4
-from aigw.filterapi.config import Backend, BackendAuth
+from aigw.filterapi.config import Backend, BackendAuth, CredentialOverride
+
+
+class CredentialMissingError(Exception):
+    """Per-request credential source configured, absent, and
+    fallbackToConfigured=false — the caller answers 401
+    (reference: ErrCredentialMissing, credential_override.go:22-26)."""
+
+
+class IncompleteAWSCredentialError(CredentialMissingError):
+    """The source carried some but not all SigV4 inputs
+    (reference: ErrIncompleteAWSCredential, credential_override.go:153-156)."""
 
 
 class _FileCredential:
@@ -65,21 +80,45 @@ def _parse_aws_ini(text: str) -> dict[str, str]:
 AuthHandler = Callable[[dict[str, str], bytes, str, str], dict[str, str]]
 # signature: (headers, body, method, path) -> headers
 
-_OVERRIDE_HEADERS = (
-    internalapi.API_KEY_OVERRIDE_HEADER,
-    internalapi.AWS_ACCESS_KEY_OVERRIDE_HEADER,
-    internalapi.AWS_SECRET_KEY_OVERRIDE_HEADER,
-    internalapi.AWS_SESSION_TOKEN_OVERRIDE_HEADER,
-)
+
+def override_header_names(auth: BackendAuth) -> tuple[str, ...]:
+    """Header names this backend's credentialOverride reads (and the server
+    must strip at egress). Empty when overrides are not configured."""
+    co = auth.credential_override
+    if co is None:
+        return ()
+    if auth.kind == "aws_access_key_id":
+        prefix = co.header_name or internalapi.AWS_CREDENTIAL_OVERRIDE_HEADER_PREFIX
+        return internalapi.aws_credential_override_header_names(prefix)
+    return (co.header_name,) if co.header_name else ()
 
 
-def _pop_overrides(headers: dict[str, str]) -> dict[str, str]:
-    ov = {}
-    for h in _OVERRIDE_HEADERS:
-        v = headers.pop(h, None)
-        if v is not None:
-            ov[h] = v
-    return ov
+def _resolve_override(co: CredentialOverride, headers: dict[str, str]) -> str:
+    """Pop and return the per-request credential; '' when absent."""
+    if not co.header_name:
+        return ""
+    return (headers.pop(co.header_name.lower(), "") or "").strip()
+
+
+def _with_override(auth: BackendAuth, apply_static: AuthHandler,
+                   apply_cred: Callable[[dict[str, str], str], dict[str, str]]) -> AuthHandler:
+    """Wrap a static handler with per-request credential sourcing
+    (credentialOverrideHandler.Do semantics, credential_override.go:98-118)."""
+    co = auth.credential_override
+    if co is None:
+        return apply_static
+
+    def handler(headers, body, method, path):
+        cred = _resolve_override(co, headers)
+        if not cred:
+            if not co.fallback_to_configured:
+                raise CredentialMissingError(
+                    f"missing per-request credential header {co.header_name!r}"
+                )
+            return apply_static(headers, body, method, path)
+        return apply_cred(headers, cred)
+
+    return handler
 
 
 def build_auth_handler(backend: Backend) -> Optional[AuthHandler]:
@@ -87,26 +126,42 @@ def build_auth_handler(backend: Backend) -> Optional[AuthHandler]:
     if auth is None:
         return None
     kind = auth.kind
+
+    def bearer_apply(headers, cred):
+        headers["authorization"] = f"Bearer {cred}"
+        return headers
+
     if kind == "api_key":
         if auth.api_key_file:
             cred = _FileCredential(auth.api_key_file)
 
-            def handler(headers, body, method, path, _cred=cred):
-                ov = _pop_overrides(headers)
-                key = ov.get(internalapi.API_KEY_OVERRIDE_HEADER, _cred.read())
-                headers["authorization"] = f"Bearer {key}"
+            def static(headers, body, method, path, _cred=cred):
+                headers["authorization"] = f"Bearer {_cred.read()}"
                 return headers
 
-            return handler
-        return _bearer_handler(auth.api_key)
+            return _with_override(auth, static, bearer_apply)
+        return _with_override(auth, _bearer_handler(auth.api_key), bearer_apply)
     if kind == "anthropic_api_key":
-        return _anthropic_handler(auth)
+
+        def anthropic_apply(headers, cred):
+            headers["x-api-key"] = cred
+            headers.setdefault("anthropic-version", "2023-06-01")
+            headers.pop("authorization", None)
+            return headers
+
+        return _with_override(auth, _anthropic_handler(auth), anthropic_apply)
     if kind == "azure_api_key":
-        return _azure_key_handler(auth)
+
+        def azure_apply(headers, cred):
+            headers["api-key"] = cred
+            headers.pop("authorization", None)
+            return headers
+
+        return _with_override(auth, _azure_key_handler(auth), azure_apply)
     if kind == "azure_access_token":
-        return _bearer_handler(auth.azure_access_token)
+        return _with_override(auth, _bearer_handler(auth.azure_access_token), bearer_apply)
     if kind == "gcp_access_token":
-        return _bearer_handler(auth.gcp_access_token)
+        return _with_override(auth, _bearer_handler(auth.gcp_access_token), bearer_apply)
     if kind == "aws_access_key_id":
         return _aws_handler(backend, auth)
     return None
@@ -114,9 +169,7 @@ def build_auth_handler(backend: Backend) -> Optional[AuthHandler]:
 
 def _bearer_handler(token: str) -> AuthHandler:
     def handler(headers, body, method, path):
-        ov = _pop_overrides(headers)
-        key = ov.get(internalapi.API_KEY_OVERRIDE_HEADER, token)
-        headers["authorization"] = f"Bearer {key}"
+        headers["authorization"] = f"Bearer {token}"
         return headers
 
     return handler
@@ -124,8 +177,7 @@ def _bearer_handler(token: str) -> AuthHandler:
 
 def _anthropic_handler(auth: BackendAuth) -> AuthHandler:
     def handler(headers, body, method, path):
-        ov = _pop_overrides(headers)
-        headers["x-api-key"] = ov.get(internalapi.API_KEY_OVERRIDE_HEADER, auth.anthropic_api_key)
+        headers["x-api-key"] = auth.anthropic_api_key
         headers.setdefault("anthropic-version", "2023-06-01")
         headers.pop("authorization", None)
         return headers
@@ -135,8 +187,7 @@ def _anthropic_handler(auth: BackendAuth) -> AuthHandler:
 
 def _azure_key_handler(auth: BackendAuth) -> AuthHandler:
     def handler(headers, body, method, path):
-        ov = _pop_overrides(headers)
-        headers["api-key"] = ov.get(internalapi.API_KEY_OVERRIDE_HEADER, auth.azure_api_key)
+        headers["api-key"] = auth.azure_api_key
         headers.pop("authorization", None)
         return headers
 
@@ -147,22 +198,31 @@ def _aws_handler(backend: Backend, auth: BackendAuth) -> AuthHandler:
     host = backend.upstream.hostname or backend.upstream.host
     region = auth.aws_region or "us-east-1"
     cred = _FileCredential(auth.aws_credentials_file) if auth.aws_credentials_file else None
+    co = auth.credential_override
+    if co is not None:
+        prefix = co.header_name or internalapi.AWS_CREDENTIAL_OVERRIDE_HEADER_PREFIX
+        h_access, h_secret, h_session = internalapi.aws_credential_override_header_names(prefix)
 
     def handler(headers, body, method, path):
-        ov = _pop_overrides(headers)
-        file_creds = _parse_aws_ini(cred.read()) if cred is not None else {}
-        access = ov.get(
-            internalapi.AWS_ACCESS_KEY_OVERRIDE_HEADER,
-            file_creds.get("aws_access_key_id", auth.aws_access_key_id),
-        )
-        secret = ov.get(
-            internalapi.AWS_SECRET_KEY_OVERRIDE_HEADER,
-            file_creds.get("aws_secret_access_key", auth.aws_secret_access_key),
-        )
-        token = ov.get(
-            internalapi.AWS_SESSION_TOKEN_OVERRIDE_HEADER,
-            file_creds.get("aws_session_token", auth.aws_session_token),
-        )
+        access = secret = token = ""
+        if co is not None:
+            access = (headers.pop(h_access, "") or "").strip()
+            secret = (headers.pop(h_secret, "") or "").strip()
+            token = (headers.pop(h_session, "") or "").strip()
+            if bool(access) != bool(secret):
+                raise IncompleteAWSCredentialError(
+                    "incomplete per-request AWS credential: access key ID and "
+                    "secret access key must both be present"
+                )
+            if not access and not co.fallback_to_configured:
+                raise CredentialMissingError(
+                    f"missing per-request AWS credential headers {h_access!r}/{h_secret!r}"
+                )
+        if not access:
+            file_creds = _parse_aws_ini(cred.read()) if cred is not None else {}
+            access = file_creds.get("aws_access_key_id", auth.aws_access_key_id)
+            secret = file_creds.get("aws_secret_access_key", auth.aws_secret_access_key)
+            token = file_creds.get("aws_session_token", auth.aws_session_token)
         headers.pop("authorization", None)
         return sign_sigv4(
             method,

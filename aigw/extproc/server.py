@@ -31,6 +31,7 @@ from aiohttp import web
 
 from aigw import internalapi
 from aigw.backendauth import build_auth_handler
+from aigw.backendauth.auth import CredentialMissingError
 from aigw.extproc.router import RETRIABLE_STATUSES, backend_attempts, max_attempts
 from aigw.extproc.upstream_client import UpstreamError
 from aigw.filterapi.config import APISchemaName, Backend
@@ -692,11 +693,21 @@ class GatewayServer:
                 up_headers.pop(h, None)
             auth = build_auth_handler(backend)
             if auth is not None:
-                up_headers = auth(up_headers, out_body, "POST", tr.path)
+                try:
+                    up_headers = auth(up_headers, out_body, "POST", tr.path)
+                except CredentialMissingError as e:
+                    # per-request credential source configured but absent
+                    # (or incomplete) with no fallback -> 401 local reply
+                    # (credential_override.go ErrCredentialMissing)
+                    return _json_error(401, str(e), "authentication_error")
             else:
                 # propagate client Authorization for auth-less backends
                 if "authorization" in headers:
                     up_headers.setdefault("authorization", headers["authorization"])
+            for h in rt.override_strip_headers:
+                # overrides not consumed by THIS backend's handler never
+                # travel upstream (reserved override names + all configured)
+                up_headers.pop(h, None)
 
             if backend.upstream.hostname:
                 up_headers["host"] = backend.upstream.hostname
@@ -1037,7 +1048,12 @@ class GatewayServer:
             up_headers = apply_header_mutation(up_headers, backend.header_mutation)
             auth = build_auth_handler(backend)
             if auth is not None:
-                up_headers = auth(up_headers, tr.body, "POST", tr.path)
+                try:
+                    up_headers = auth(up_headers, tr.body, "POST", tr.path)
+                except CredentialMissingError as e:
+                    return _json_error(401, str(e), "authentication_error")
+            for h in rt.override_strip_headers:
+                up_headers.pop(h, None)
             try:
                 upstream = await self._session.post(
                     host=backend.upstream.host,

@@ -91,6 +91,24 @@ class BodyMutation:
 
 
 @dataclass
+class CredentialOverride:
+    """Per-request credential sourcing for a backend (filterconfig.go
+    CredentialOverride :221-240). Override headers are only honored when
+    this is configured; otherwise client-sent override headers are
+    stripped at egress and ignored.
+
+    header_name: request header carrying the credential. For AWS it is a
+    PREFIX, not a full name — the three SigV4 header names derive from it
+    (internalapi.aws_credential_override_header_names).
+    fallback_to_configured: when the source header is absent, True falls
+    back to the static credential; False makes the gateway answer 401.
+    """
+
+    header_name: str = ""
+    fallback_to_configured: bool = False
+
+
+@dataclass
 class BackendAuth:
     """Upstream credential injection (filterconfig.go BackendAuth).
 
@@ -115,6 +133,8 @@ class BackendAuth:
     # the AWS credential trio (INI credentials-file format for AWS).
     api_key_file: str = ""
     aws_credentials_file: str = ""
+    # Per-request credential sourcing; None = overrides disabled.
+    credential_override: Optional["CredentialOverride"] = None
 
     @property
     def kind(self) -> str:
@@ -350,6 +370,16 @@ def _parse_cost(d, ctx) -> LLMRequestCost:
     return c
 
 
+def _parse_backend_auth(d, ctx) -> BackendAuth:
+    kw = _dc(BackendAuth, d, ctx)
+    if kw.get("credential_override") is not None:
+        kw["credential_override"] = CredentialOverride(
+            **_dc(CredentialOverride, kw["credential_override"],
+                  f"{ctx}.credentialOverride")
+        )
+    return BackendAuth(**kw)
+
+
 def _parse_backend(d, ctx) -> Backend:
     kw = _dc(Backend, d, ctx)
     if "name" not in kw:
@@ -359,7 +389,7 @@ def _parse_backend(d, ctx) -> Backend:
     if "upstream" in kw:
         kw["upstream"] = Upstream(**_dc(Upstream, kw["upstream"], f"{ctx}.upstream"))
     if kw.get("auth") is not None:
-        kw["auth"] = BackendAuth(**_dc(BackendAuth, kw["auth"], f"{ctx}.auth"))
+        kw["auth"] = _parse_backend_auth(kw["auth"], f"{ctx}.auth")
     if kw.get("header_mutation") is not None:
         kw["header_mutation"] = HeaderMutation(
             **_dc(HeaderMutation, kw["header_mutation"], f"{ctx}.headerMutation")
@@ -430,7 +460,7 @@ def load_config(data: object) -> Config:
                 if bkw.get("headers") is not None:
                     bkw["headers"] = HeaderMutation(**_dc(HeaderMutation, bkw["headers"], "headers"))
                 if bkw.get("auth") is not None:
-                    bkw["auth"] = BackendAuth(**_dc(BackendAuth, bkw["auth"], "auth"))
+                    bkw["auth"] = _parse_backend_auth(bkw["auth"], "auth")
                 backends.append(MCPBackend(**bkw))
             rkw["backends"] = backends
             if rkw.get("oauth") is not None:

@@ -119,6 +119,21 @@ class RuntimeConfig:
             else:
                 self.unscoped_models.append(m)
         self.rate_limits = list(cfg.rate_limits)
+        # Every credential-override header name configured anywhere, plus
+        # the reserved default AWS trio: the server strips these at egress
+        # so a client-sent override can never leak to an upstream that did
+        # not opt into per-request credentials (credential_override.go; the
+        # reference's controller builds the same list as
+        # CredentialOverride.InputHeadersToRemove).
+        from aigw import internalapi
+        from aigw.backendauth.auth import override_header_names
+
+        strip: set[str] = set(internalapi.aws_credential_override_header_names())
+        for r in cfg.routes:
+            for b in r.backends:
+                if b.auth is not None:
+                    strip.update(h.lower() for h in override_header_names(b.auth))
+        self.override_strip_headers = frozenset(strip)
 
     def select_route(self, headers: dict[str, str]) -> Optional[CompiledRoute]:
         """First-match route selection over lowercase header dict."""

@@ -51,12 +51,25 @@ INTERNAL_HEADERS = (
     MCP_BACKEND_HEADER,
 )
 
-# Per-request credential-override headers (reference: internalapi.go AWS
-# credential-override header names).
-AWS_ACCESS_KEY_OVERRIDE_HEADER = "x-ai-eg-aws-access-key-id"
-AWS_SECRET_KEY_OVERRIDE_HEADER = "x-ai-eg-aws-secret-access-key"
-AWS_SESSION_TOKEN_OVERRIDE_HEADER = "x-ai-eg-aws-session-token"
-API_KEY_OVERRIDE_HEADER = "x-ai-eg-api-key"
+# Per-request credential override (reference: internalapi.go:77-92).
+# Override headers are only honored when the backend config sets
+# credentialOverride (filterapi CredentialOverride); the header NAME is
+# config-supplied. For AWS the configured value is a PREFIX and the three
+# SigV4 header names derive from it, default prefix below.
+AWS_CREDENTIAL_OVERRIDE_HEADER_PREFIX = "x-aigw-aws-"
+
+
+def aws_credential_override_header_names(
+    prefix: str = AWS_CREDENTIAL_OVERRIDE_HEADER_PREFIX,
+) -> tuple[str, str, str]:
+    """(access-key-id, secret-access-key, session-token) header names for a
+    configured prefix (reference: AWSCredentialOverrideHeaderNames,
+    internalapi.go:86-92)."""
+    return (
+        prefix + "access-key-id",
+        prefix + "secret-access-key",
+        prefix + "session-token",
+    )
 
 # Default ports (reference: mainlib/main.go:104-121, internalapi.go:48-52).
 DEFAULT_LISTEN_PORT = 1975

@@ -38,7 +38,7 @@ def redact_json_tree(node, *, _key: str = ""):
 def redact_headers(headers: dict[str, str]) -> dict[str, str]:
     """Credential headers are ALWAYS redacted in logs (server.go:474-516)."""
     sensitive = {"authorization", "x-api-key", "api-key", "proxy-authorization",
-                 "cookie", "set-cookie", "x-ai-eg-api-key"}
+                 "cookie", "set-cookie", "x-ai-eg-api-key", "x-aigw-aws-access-key-id", "x-aigw-aws-secret-access-key", "x-aigw-aws-session-token"}
     return {
         k: (redact_string(v) if k.lower() in sensitive else v)
         for k, v in headers.items()

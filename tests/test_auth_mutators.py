@@ -3,10 +3,20 @@ internal/headermutator, internal/bodymutator)."""
 
 from datetime import datetime, timezone
 
+import pytest
+
 from aigw import internalapi
 from aigw.backendauth import build_auth_handler
+from aigw.backendauth.auth import CredentialMissingError, IncompleteAWSCredentialError
 from aigw.backendauth.sigv4 import sign_sigv4
-from aigw.filterapi.config import Backend, BackendAuth, BodyMutation, HeaderMutation, Upstream
+from aigw.filterapi.config import (
+    Backend,
+    BackendAuth,
+    BodyMutation,
+    CredentialOverride,
+    HeaderMutation,
+    Upstream,
+)
 from aigw.mutator import apply_body_mutation, apply_header_mutation
 
 
@@ -30,12 +40,59 @@ def test_azure_key():
     assert h["api-key"] == "zk"
 
 
-def test_credential_override_header():
+def test_credential_override_gated_on_config():
+    """Override headers are ignored unless the backend opts in via
+    credentialOverride (credential_override.go semantics)."""
+    # not configured: client-sent override header is ignored
     b = Backend(name="o", auth=BackendAuth(api_key="sk-config"))
-    headers = {internalapi.API_KEY_OVERRIDE_HEADER: "sk-override"}
-    h = build_auth_handler(b)(headers, b"", "POST", "/v1/chat/completions")
+    h = build_auth_handler(b)({"x-client-key": "sk-override"}, b"", "POST", "/v1/x")
+    assert h["authorization"] == "Bearer sk-config"
+
+    # configured: the named header supplies the credential and is consumed
+    b = Backend(name="o", auth=BackendAuth(
+        api_key="sk-config",
+        credential_override=CredentialOverride(header_name="x-client-key",
+                                               fallback_to_configured=True)))
+    h = build_auth_handler(b)({"x-client-key": "sk-override"}, b"", "POST", "/v1/x")
     assert h["authorization"] == "Bearer sk-override"
-    assert internalapi.API_KEY_OVERRIDE_HEADER not in h
+    assert "x-client-key" not in h
+
+    # configured + absent + fallback: static credential
+    h = build_auth_handler(b)({}, b"", "POST", "/v1/x")
+    assert h["authorization"] == "Bearer sk-config"
+
+    # configured + absent + NO fallback: 401 (CredentialMissingError)
+    b = Backend(name="o", auth=BackendAuth(
+        api_key="sk-config",
+        credential_override=CredentialOverride(header_name="x-client-key")))
+    with pytest.raises(CredentialMissingError):
+        build_auth_handler(b)({}, b"", "POST", "/v1/x")
+
+
+def test_aws_credential_override_headers():
+    """AWS overrides use the x-aigw-aws- prefix trio; incomplete pairs are
+    rejected (ErrIncompleteAWSCredential)."""
+    b = Backend(name="aws", auth=BackendAuth(
+        aws_access_key_id="AKIDSTATIC", aws_secret_access_key="s3cr3t",
+        aws_region="us-east-1",
+        credential_override=CredentialOverride(fallback_to_configured=True)))
+    names = internalapi.aws_credential_override_header_names()
+    assert names == ("x-aigw-aws-access-key-id",
+                     "x-aigw-aws-secret-access-key",
+                     "x-aigw-aws-session-token")
+    hdrs = {names[0]: "AKIDOVER", names[1]: "oversecret"}
+    h = build_auth_handler(b)(hdrs, b"{}", "POST", "/model/m/converse")
+    assert "AKIDOVER" in h["authorization"]
+    for n in names:
+        assert n not in h
+
+    # access key without secret -> incomplete -> 401-class error
+    with pytest.raises(IncompleteAWSCredentialError):
+        build_auth_handler(b)({names[0]: "AKIDONLY"}, b"{}", "POST", "/x")
+
+    # absent + fallback: static credential signs
+    h = build_auth_handler(b)({}, b"{}", "POST", "/model/m/converse")
+    assert "AKIDSTATIC" in h["authorization"]
 
 
 # SigV4 known-answer test: vector computed with the canonical algorithm

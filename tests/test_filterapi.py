@@ -153,3 +153,38 @@ def test_regex_header_match():
     rc = RuntimeConfig(cfg)
     assert rc.select_route({"x-ai-eg-model": "gpt-4o"}) is not None
     assert rc.select_route({"x-ai-eg-model": "claude"}) is None
+
+
+def test_credential_override_config_and_strip_set():
+    """credentialOverride parses from YAML config and every configured
+    override header name (plus the reserved default AWS trio) lands in the
+    runtime egress strip set."""
+    from aigw.filterapi.config import load_config
+    from aigw.filterapi.runtime import RuntimeConfig
+
+    cfg = load_config({
+        "routes": [{
+            "name": "r",
+            "backends": [
+                {"name": "a", "schema": "OpenAI",
+                 "upstream": {"host": "h", "port": 80},
+                 "auth": {"apiKey": "sk",
+                          "credentialOverride": {"headerName": "X-Client-Key",
+                                                 "fallbackToConfigured": True}}},
+                {"name": "b", "schema": "AWSBedrock",
+                 "upstream": {"host": "h2", "port": 80},
+                 "auth": {"awsAccessKeyId": "AK", "awsSecretAccessKey": "s",
+                          "credentialOverride": {"headerName": "x-corp-aws-"}}},
+            ],
+        }],
+    })
+    a = cfg.routes[0].backends[0].auth
+    assert a.credential_override.header_name == "X-Client-Key"
+    assert a.credential_override.fallback_to_configured is True
+    rt = RuntimeConfig(cfg)
+    assert "x-client-key" in rt.override_strip_headers
+    for h in ("x-corp-aws-access-key-id", "x-corp-aws-secret-access-key",
+              "x-corp-aws-session-token"):
+        assert h in rt.override_strip_headers
+    # reserved defaults are always stripped, configured or not
+    assert "x-aigw-aws-access-key-id" in rt.override_strip_headers
