@@ -203,6 +203,37 @@ async def _prepend(first, aiter):
         yield item
 
 
+_SAMPLING_KEYS = (
+    "temperature", "top_p", "top_k", "n", "max_tokens", "max_completion_tokens",
+    "stop", "presence_penalty", "frequency_penalty", "logit_bias", "seed",
+    "response_format", "tools", "tool_choice", "logprobs", "top_logprobs",
+    "reasoning_effort",
+)
+
+
+def _cache_fingerprint(model: str, route_name: str, body: dict,
+                       client_headers: dict[str, str]) -> bytes:
+    """Scope tag for semantic-cache values: two near-identical prompts may
+    embed to the same key vector, so everything that must NOT be shared —
+    model, route, generation parameters, and the client credential — is
+    folded into a 16-byte digest compared on the VALUE at hit time."""
+    import hashlib
+
+    ident = (
+        client_headers.get("authorization")
+        or client_headers.get("x-api-key")
+        or client_headers.get("api-key")
+        or ""
+    )
+    params = {k: body[k] for k in _SAMPLING_KEYS if k in body}
+    blob = json.dumps(
+        [model, route_name, params,
+         hashlib.sha256(ident.encode("utf-8", "replace")).hexdigest()],
+        sort_keys=True, separators=(",", ":"), default=str,
+    )
+    return hashlib.sha256(blob.encode()).digest()[:16]
+
+
 def _json_error(status: int, message: str, err_type: str = "invalid_request_error") -> web.Response:
     # gateway-generated ("local reply") error envelope, wire-compatible
     # with the reference's formatUserFacingErrorJSON
@@ -529,21 +560,28 @@ class GatewayServer:
 
         # semantic response cache (GPU MFMA embed + HBM index). Streamed
         # requests are cached too: the translated SSE transcript is stored
-        # on miss and replayed chunk-wise on hit. Mode lives in a value tag
-        # (a key prefix cannot separate stream/unary in embedding space —
-        # mean-pooled vectors of near-identical texts coincide), so a
-        # cached transcript never answers a unary request or vice versa.
+        # on miss and replayed chunk-wise on hit. Scope lives in a value
+        # tag (a key prefix cannot separate scopes in embedding space —
+        # mean-pooled vectors of near-identical texts coincide): one mode
+        # byte (stream/unary) plus a fingerprint of (model, route,
+        # sampling params, client credential), so a cached answer never
+        # crosses models, generation settings, or tenants.
         cache_key_vec = None
-        cache_tag = b"S" if stream else b"U"
+        cache_tag = b""
         if (
             self.gpu is not None
             and self.gpu.cache_enabled
             and endpoint == "/v1/chat/completions"
         ):
+            if body is None:
+                body = json.loads(raw)
+            cache_tag = (b"S" if stream else b"U") + _cache_fingerprint(
+                model, route.route.name, body, request.headers
+            )
             hit, cache_key_vec = await self.gpu.cache_lookup_text(chat_text or b" ")
-            if hit is not None and hit[:1] == cache_tag:
+            if hit is not None and hit[: len(cache_tag)] == cache_tag:
                 self.metrics._child(self.metrics.cache_events, ("hit",)).inc()
-                body_bytes = hit[1:]
+                body_bytes = hit[len(cache_tag):]
                 if not stream:
                     resp = web.Response(body=body_bytes, content_type="application/json")
                     resp.headers["x-aigw-cache"] = "hit"
@@ -567,6 +605,7 @@ class GatewayServer:
             resp = await self._dispatch(
                 request, endpoint, route, headers, body, stream, start,
                 model=model, gpu_input_tokens=gpu_input_tokens, cache_key_vec=cache_key_vec,
+                cache_tag=cache_tag,
                 span=span, raw=raw,
             )
             if span is not None:
@@ -590,6 +629,7 @@ class GatewayServer:
         model: str,
         gpu_input_tokens: int = 0,
         cache_key_vec=None,
+        cache_tag: bytes = b"",
         span=None,
         raw: bytes = b"",
     ) -> web.StreamResponse:
@@ -769,11 +809,12 @@ class GatewayServer:
                     return await self._stream_response(
                         request, endpoint, route, backend, translator, upstream,
                         headers, model, start, gpu_input_tokens, span=span,
-                        cache_key_vec=cache_key_vec,
+                        cache_key_vec=cache_key_vec, cache_tag=cache_tag,
                     )
                 return await self._unary_response(
                     endpoint, route, backend, translator, upstream, headers,
-                    model, start, gpu_input_tokens, body, cache_key_vec, span=span,
+                    model, start, gpu_input_tokens, body, cache_key_vec,
+                    cache_tag=cache_tag, span=span,
                 )
             except (UpstreamError, OSError, asyncio.TimeoutError, asyncio.IncompleteReadError) as e:
                 last_error = f"upstream {backend.name}: {type(e).__name__}: {e}"
@@ -874,7 +915,7 @@ class GatewayServer:
 
     async def _unary_response(
         self, endpoint, route, backend, translator, upstream, headers, model,
-        start, gpu_input_tokens, orig_body, cache_key_vec, span=None,
+        start, gpu_input_tokens, orig_body, cache_key_vec, cache_tag=b"", span=None,
     ) -> web.Response:
         # idle-bounded read: a stalled upstream raises UpstreamError, which
         # _dispatch treats as a failed try (fallback proceeds — nothing has
@@ -909,12 +950,13 @@ class GatewayServer:
             # same 2 MiB cap as streamed transcripts: 64k slots x unbounded
             # bodies would otherwise grow host memory without limit
         ):
-            await self.gpu.cache_insert(cache_key_vec, b"U" + rtl.body)
+            await self.gpu.cache_insert(cache_key_vec, cache_tag + rtl.body)
         return resp
 
     async def _stream_response(
         self, request, endpoint, route, backend, translator, upstream,
         headers, model, start, gpu_input_tokens, span=None, cache_key_vec=None,
+        cache_tag=b"",
     ) -> web.StreamResponse:
         transcript = bytearray() if cache_key_vec is not None else None
         idle = backend.stream_idle_timeout_s
@@ -970,7 +1012,7 @@ class GatewayServer:
                 and upstream.status == 200
                 and len(transcript) < (2 << 20)
             ):
-                await self.gpu.cache_insert(cache_key_vec, b"S" + bytes(transcript))
+                await self.gpu.cache_insert(cache_key_vec, cache_tag + bytes(transcript))
         except (UpstreamError, ValueError) as e:
             # mid-stream failure (idle timeout, or malformed provider bytes
             # the translator/codec rejects): the response has started, so
