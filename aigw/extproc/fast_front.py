@@ -1,0 +1,194 @@
+"""Native fast front: the C++ data-plane server (csrc/fastpath.cpp)
+composed from a RuntimeConfig.
+
+The fast front serves the OpenAI passthrough hot loop (frame → scan →
+route → rate-limit → auth → upstream → usage-tap relay) entirely in
+native threads; everything it cannot serve natively — provider
+translation, multipart audio, MCP, admin endpoints — rides a loopback
+aiohttp instance running the full Python app, exactly like the lean
+front's fallback. CPython is out of the per-request path.
+
+Eligibility is decided per route at composition time (a matched but
+ineligible route is relayed to the fallback per request):
+- backend schema is OpenAI (byte-level passthrough; other providers need
+  the Python translators);
+- backend auth is None or a static/file API key (SigV4 and vendor auth
+  flows stay in Python);
+- plain HTTP upstream (TLS upstreams stay in Python);
+- no endpoint picker, no body mutations, no tracing.
+Whole-config blockers (any present -> FastFrontUnsupported, callers fall
+back to the Python fronts): non-model header matches, keyed or CEL rate
+rules (the native limiter is fixed-window per rule, like the Python
+limiter's un-keyed buckets).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Optional
+
+from aiohttp import web
+
+from aigw.filterapi.config import APISchemaName, LLMRequestCostType
+from aigw.filterapi.runtime import RuntimeConfig
+
+try:
+    import aigw_fast as _fast
+except ImportError:  # pragma: no cover - built by setup.py everywhere
+    _fast = None
+
+logger = logging.getLogger("aigw.fast_front")
+
+
+class FastFrontUnsupported(RuntimeError):
+    """The config needs features only the Python fronts provide."""
+
+
+def _backend_eligible(b) -> tuple[bool, Optional[dict]]:
+    if b.schema.name is not APISchemaName.OPENAI:
+        return False, None
+    if b.upstream.tls or b.upstream.path_prefix:
+        return False, None
+    if b.body_mutation is not None or b.header_mutation is not None:
+        return False, None
+    if b.max_concurrency:
+        return False, None
+    auth = b.auth
+    bearer = ""
+    api_key_file = ""
+    if auth is not None:
+        if auth.credential_override is not None:
+            return False, None
+        kind = auth.kind
+        if kind == "api_key":
+            bearer = auth.api_key
+            api_key_file = auth.api_key_file
+        elif kind == "":
+            pass
+        else:
+            return False, None
+    return True, {
+        "name": b.name,
+        "host": b.upstream.host,
+        "port": b.upstream.port,
+        "bearer": bearer,
+        "api_key_file": api_key_file,
+        "model_override": b.model_name_override or "",
+        "weight": float(b.weight),
+        "priority": int(b.priority),
+        "timeout_s": float(b.timeout_s),
+    }
+
+
+def build_fast_server(runtime: RuntimeConfig):
+    """Compose a FastServer from a RuntimeConfig; raises
+    FastFrontUnsupported when the config as a whole cannot be served."""
+    if _fast is None:
+        raise FastFrontUnsupported("aigw_fast extension not built")
+    cfg = runtime.config
+    for rule in cfg.rate_limits:
+        if rule.key_headers:
+            raise FastFrontUnsupported(
+                f"rate rule {rule.name!r} uses key headers (python fronts only)"
+            )
+    srv = _fast.FastServer()
+    for cr in runtime.routes:
+        model_match = ""
+        for m in cr.matches:
+            if m.name == runtime.model_header and m.regex is None and m.value:
+                model_match = m.value
+            elif m.regex is not None or m.value or m.name != runtime.model_header:
+                raise FastFrontUnsupported(
+                    f"route {cr.route.name!r} matches on non-model headers"
+                )
+        eligible = not cr.route.endpoint_picker
+        has_costs = bool(cr.costs)
+        for c in cr.costs:
+            if c.type is LLMRequestCostType.CEL:
+                eligible = False  # CEL costs need the Python evaluator
+        backends = []
+        for tier in cr.tiers:
+            for b in tier:
+                ok, desc = _backend_eligible(b)
+                if not ok:
+                    eligible = False
+                    break
+                backends.append(desc)
+            if not eligible:
+                break
+        if cr.route.header_mutation is not None:
+            eligible = False
+        if cr.route.model_name_override:
+            for d in backends:
+                if not d["model_override"]:
+                    d["model_override"] = cr.route.model_name_override
+        srv.add_route(
+            cr.route.name, model_match, int(cr.route.retries), has_costs,
+            eligible, backends if eligible else [],
+        )
+    for rule in cfg.rate_limits:
+        srv.add_rate_rule(rule.name, int(rule.limit), float(rule.window_s),
+                          rule.metadata_key)
+    return srv
+
+
+class FastFront:
+    """Lifecycle wrapper: fallback aiohttp app + native server + a
+    metrics bridge exposing the native counters on the Python /metrics."""
+
+    def __init__(self, server, runtime: RuntimeConfig, *, gpu_socket: str = "",
+                 gpu_window_us: int = 100, gpu_max_batch: int = 256):
+        # `server` is the Python GatewayServer used for cold paths
+        self.py_server = server
+        self.runtime = runtime
+        self.fast = build_fast_server(runtime)
+        self.gpu_socket = gpu_socket
+        self.gpu_window_us = gpu_window_us
+        self.gpu_max_batch = gpu_max_batch
+        self._fallback_runner = None
+        self.port = None
+
+    async def start(self, host: str, port: int) -> int:
+        app = self.py_server.make_app()
+        self._fallback_runner = web.AppRunner(app, access_log=None)
+        await self._fallback_runner.setup()
+        site = web.TCPSite(self._fallback_runner, "127.0.0.1", 0)
+        await site.start()
+        fallback_port = self._fallback_runner.addresses[0][1]
+        self.fast.set_fallback("127.0.0.1", fallback_port)
+        if self.gpu_socket:
+            self.fast.enable_gpu(self.gpu_socket, self.gpu_window_us,
+                                 self.gpu_max_batch)
+        self.port = self.fast.start(host, port)
+        logger.info("fast front listening on %s:%d (fallback :%d)", host,
+                    self.port, fallback_port)
+        return self.port
+
+    def stats(self) -> dict:
+        return self.fast.stats()
+
+    def render_metrics_extra(self) -> str:
+        """Prometheus lines for the native counters, appended to the
+        Python registry's render by the CLI's /metrics."""
+        s = self.stats()
+        lines = []
+        for k in ("requests", "responses_2xx", "responses_4xx", "responses_5xx",
+                  "local_429", "fallback", "retries", "gpu_tokens",
+                  "input_tokens", "output_tokens", "total_tokens",
+                  "bytes_in", "bytes_out"):
+            lines.append(f'aigw_fast_{k}_total {s[k]}')
+        lines.append(f'aigw_fast_active_connections {s["active_connections"]}')
+        hist = s["latency_us_log2"]
+        cum = 0
+        for i, n in enumerate(hist):
+            cum += n
+            lines.append(
+                f'aigw_fast_latency_us_bucket{{le="{2 ** (i + 1)}"}} {cum}')
+        return "\n".join(lines) + "\n"
+
+    async def stop(self) -> None:
+        self.fast.stop()
+        if self._fallback_runner is not None:
+            await self._fallback_runner.cleanup()
+            self._fallback_runner = None
+        await self.py_server.close()

@@ -1,0 +1,92 @@
+// pybind11 bindings for the native fast-path server (csrc/fastpath.cpp).
+// CPython composes the config (from aigw.filterapi.RuntimeConfig) and
+// owns the cold-path fallback app; the server itself runs entirely in
+// native threads with the GIL released.
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "fastpath.h"
+
+namespace py = pybind11;
+using namespace aigw_fast;
+
+PYBIND11_MODULE(aigw_fast, m) {
+  m.doc() = "aigw native data-plane fast path";
+
+  py::class_<FastServer>(m, "FastServer")
+      .def(py::init<>())
+      .def(
+          "add_route",
+          [](FastServer& s, const std::string& name, const std::string& model_match,
+             int retries, bool has_costs, bool eligible, py::list backends) {
+            FastRoute r;
+            r.name = name;
+            r.model_match = model_match;
+            r.retries = retries;
+            r.has_costs = has_costs;
+            r.eligible = eligible;
+            for (auto item : backends) {
+              py::dict d = item.cast<py::dict>();
+              FastBackend b;
+              b.name = d["name"].cast<std::string>();
+              b.host = d["host"].cast<std::string>();
+              b.port = d["port"].cast<uint16_t>();
+              if (d.contains("bearer")) b.bearer = d["bearer"].cast<std::string>();
+              if (d.contains("api_key_file"))
+                b.api_key_file = d["api_key_file"].cast<std::string>();
+              if (d.contains("model_override"))
+                b.model_override = d["model_override"].cast<std::string>();
+              if (d.contains("weight")) b.weight = d["weight"].cast<double>();
+              if (d.contains("priority")) b.priority = d["priority"].cast<int>();
+              if (d.contains("timeout_s")) b.timeout_s = d["timeout_s"].cast<double>();
+              r.backends.push_back(std::move(b));
+            }
+            s.add_route(std::move(r));
+          },
+          py::arg("name"), py::arg("model_match"), py::arg("retries"),
+          py::arg("has_costs"), py::arg("eligible"), py::arg("backends"))
+      .def("add_rate_rule",
+           [](FastServer& s, const std::string& name, int64_t limit,
+              double window_s, const std::string& metadata_key) {
+             RateRule r;
+             r.name = name;
+             r.limit = limit;
+             r.window_s = window_s;
+             r.metadata_key = metadata_key == "llm_input_token"    ? 1
+                              : metadata_key == "llm_output_token" ? 2
+                                                                   : 0;
+             s.add_rate_rule(r);
+           })
+      .def("set_fallback", &FastServer::set_fallback)
+      .def("enable_gpu", &FastServer::enable_gpu, py::arg("socket_path"),
+           py::arg("window_us") = 100, py::arg("max_batch") = 256)
+      .def("start", &FastServer::start, py::arg("host"), py::arg("port"),
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &FastServer::stop, py::call_guard<py::gil_scoped_release>())
+      .def("rl_collect_deltas", &FastServer::rl_collect_deltas)
+      .def("rl_apply_remote", &FastServer::rl_apply_remote)
+      .def("rl_local_spent", &FastServer::rl_local_spent)
+      .def("stats", [](const FastServer& s) {
+        const ServerStats& st = s.stats();
+        py::dict d;
+        d["requests"] = st.requests.load();
+        d["responses_2xx"] = st.responses_2xx.load();
+        d["responses_4xx"] = st.responses_4xx.load();
+        d["responses_5xx"] = st.responses_5xx.load();
+        d["local_429"] = st.local_429.load();
+        d["fallback"] = st.fallback.load();
+        d["retries"] = st.retries.load();
+        d["gpu_tokens"] = st.gpu_tokens.load();
+        d["input_tokens"] = st.input_tokens.load();
+        d["output_tokens"] = st.output_tokens.load();
+        d["total_tokens"] = st.total_tokens.load();
+        d["bytes_in"] = st.bytes_in.load();
+        d["bytes_out"] = st.bytes_out.load();
+        d["active_connections"] = st.active_connections.load();
+        py::list hist;
+        for (int i = 0; i < 32; ++i) hist.append(st.latency_us_log2[i].load());
+        d["latency_us_log2"] = hist;
+        return d;
+      });
+}

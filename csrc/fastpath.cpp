@@ -1,0 +1,1416 @@
+// aigw fast path implementation — see fastpath.h for the design notes.
+//
+// Behavior parity anchors (same invariants as aigw/extproc/server.py and
+// the reference):
+//   - requests are framed by Content-Length only; any Transfer-Encoding
+//     is refused with 501 and conflicting duplicate Content-Length with
+//     400 (request-desync hardening, identical to the Python lean front);
+//   - client-spoofable x-ai-eg-* / credential-override headers never
+//     travel upstream;
+//   - model routing happens after the body scan, before backend pick;
+//   - per-try translation: every retry splices the ORIGINAL body bytes;
+//   - auth (static bearer) is applied over the FINAL body;
+//   - local replies (400/404/429/503) are never re-translated;
+//   - streamed responses relay the upstream framing unchanged and are
+//     tapped for cumulative usage; content-length responses forward
+//     byte-identical bodies.
+
+#include "fastpath.h"
+
+#include <arpa/inet.h>
+#include <errno.h>
+#include <fcntl.h>
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <string.h>
+#include <sys/socket.h>
+#include <sys/time.h>
+#include <sys/types.h>
+#include <sys/un.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <chrono>
+#include <cmath>
+#include <random>
+
+#include "msgpack_mini.h"
+#include "native_core.h"
+
+namespace aigw_fast {
+
+namespace {
+
+int64_t now_ms() {
+  return std::chrono::duration_cast<std::chrono::milliseconds>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+int64_t now_us() {
+  return std::chrono::duration_cast<std::chrono::microseconds>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+void set_timeout(int fd, double seconds) {
+  struct timeval tv;
+  tv.tv_sec = (time_t)seconds;
+  tv.tv_usec = (suseconds_t)((seconds - (double)tv.tv_sec) * 1e6);
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+  setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
+}
+
+void set_nodelay(int fd) {
+  int one = 1;
+  setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+}
+
+// full-buffer send; false on error/timeout
+bool write_all(int fd, const char* p, size_t n) {
+  while (n > 0) {
+    ssize_t w = ::send(fd, p, n, MSG_NOSIGNAL);
+    if (w < 0) {
+      if (errno == EINTR) continue;
+      return false;
+    }
+    p += w;
+    n -= (size_t)w;
+  }
+  return true;
+}
+
+bool write_all(int fd, const std::string& s) { return write_all(fd, s.data(), s.size()); }
+
+// one recv; 0 = clean EOF, -1 = error/timeout
+ssize_t read_some(int fd, char* p, size_t cap) {
+  for (;;) {
+    ssize_t r = ::recv(fd, p, cap, 0);
+    if (r < 0 && errno == EINTR) continue;
+    return r;
+  }
+}
+
+int tcp_connect(const std::string& host, uint16_t port, double timeout_s) {
+  struct sockaddr_in addr;
+  memset(&addr, 0, sizeof(addr));
+  addr.sin_family = AF_INET;
+  addr.sin_port = htons(port);
+  if (inet_pton(AF_INET, host.c_str(), &addr.sin_addr) != 1) {
+    struct addrinfo hints, *res = nullptr;
+    memset(&hints, 0, sizeof(hints));
+    hints.ai_family = AF_INET;
+    hints.ai_socktype = SOCK_STREAM;
+    if (getaddrinfo(host.c_str(), nullptr, &hints, &res) != 0 || !res) return -1;
+    addr.sin_addr = ((struct sockaddr_in*)res->ai_addr)->sin_addr;
+    freeaddrinfo(res);
+  }
+  int fd = ::socket(AF_INET, SOCK_STREAM, 0);
+  if (fd < 0) return -1;
+  set_timeout(fd, timeout_s);
+  if (::connect(fd, (struct sockaddr*)&addr, sizeof(addr)) != 0) {
+    ::close(fd);
+    return -1;
+  }
+  set_nodelay(fd);
+  return fd;
+}
+
+std::string lower(std::string s) {
+  for (auto& c : s) c = (char)tolower((unsigned char)c);
+  return s;
+}
+
+struct Header {
+  std::string name;  // lowercased
+  std::string value;
+};
+
+struct HttpHead {
+  std::string method, path;  // requests
+  int status = 0;            // responses
+  std::vector<Header> headers;
+  size_t head_len = 0;  // bytes consumed incl. CRLFCRLF
+
+  const std::string* get(const char* name) const {
+    for (const auto& h : headers)
+      if (h.name == name) return &h.value;
+    return nullptr;
+  }
+};
+
+constexpr size_t kMaxHead = 64 * 1024;
+constexpr size_t kMaxBody = 64 * 1024 * 1024;
+
+// parse a full head present in buf[0..); returns false on malformed
+bool parse_head(const std::string& buf, size_t head_end, bool is_request, HttpHead* out) {
+  out->head_len = head_end + 4;
+  size_t line_end = buf.find("\r\n");
+  if (line_end == std::string::npos || line_end > head_end) return false;
+  const std::string first = buf.substr(0, line_end);
+  if (is_request) {
+    size_t sp1 = first.find(' ');
+    if (sp1 == std::string::npos) return false;
+    size_t sp2 = first.find(' ', sp1 + 1);
+    if (sp2 == std::string::npos) return false;
+    out->method = first.substr(0, sp1);
+    out->path = first.substr(sp1 + 1, sp2 - sp1 - 1);
+  } else {
+    if (first.compare(0, 7, "HTTP/1.") != 0) return false;
+    size_t sp = first.find(' ');
+    if (sp == std::string::npos) return false;
+    out->status = atoi(first.c_str() + sp + 1);
+  }
+  size_t pos = line_end + 2;
+  while (pos < head_end) {
+    size_t eol = buf.find("\r\n", pos);
+    if (eol == std::string::npos || eol > head_end) eol = head_end;
+    size_t colon = buf.find(':', pos);
+    if (colon != std::string::npos && colon < eol) {
+      Header h;
+      h.name = lower(buf.substr(pos, colon - pos));
+      // trim
+      size_t vs = colon + 1;
+      while (vs < eol && (buf[vs] == ' ' || buf[vs] == '\t')) ++vs;
+      size_t ve = eol;
+      while (ve > vs && (buf[ve - 1] == ' ' || buf[ve - 1] == '\t')) --ve;
+      while (!h.name.empty() && (h.name.back() == ' ' || h.name.back() == '\t'))
+        h.name.pop_back();
+      h.value = buf.substr(vs, ve - vs);
+      out->headers.push_back(std::move(h));
+    }
+    pos = eol + 2;
+  }
+  return true;
+}
+
+// -------- body framing ------------------------------------------------------
+
+enum class Framing { kLength, kChunked, kClose, kNone };
+
+Framing response_framing(const HttpHead& h, int64_t* length) {
+  const std::string* te = h.get("transfer-encoding");
+  if (te && lower(*te).find("chunked") != std::string::npos) return Framing::kChunked;
+  if (h.status == 204 || h.status == 304 || (h.status >= 100 && h.status < 200)) {
+    *length = 0;
+    return Framing::kLength;
+  }
+  const std::string* cl = h.get("content-length");
+  if (cl) {
+    *length = atoll(cl->c_str());
+    return Framing::kLength;
+  }
+  return Framing::kClose;
+}
+
+// incremental chunked-transfer parser: feeds raw bytes, emits payload
+// spans, reports completion. Used both to find the response end for
+// connection reuse and to tap SSE payloads mid-relay.
+class ChunkedParser {
+ public:
+  bool done() const { return state_ == State::kDone; }
+  bool error() const { return state_ == State::kError; }
+
+  // feed raw bytes; calls payload(p, n) for chunk-data spans
+  template <class F>
+  void feed(const char* p, size_t n, F&& payload) {
+    size_t i = 0;
+    while (i < n && state_ != State::kDone && state_ != State::kError) {
+      switch (state_) {
+        case State::kSize: {
+          char c = p[i++];
+          line_.push_back(c);
+          if (line_.size() > 256) { state_ = State::kError; break; }
+          if (c == '\n') {
+            size_t semi = line_.find(';');
+            std::string hex = line_.substr(0, semi);
+            remaining_ = strtoll(hex.c_str(), nullptr, 16);
+            line_.clear();
+            if (remaining_ < 0) state_ = State::kError;
+            else if (remaining_ == 0) state_ = State::kTrailer;
+            else state_ = State::kData;
+          }
+          break;
+        }
+        case State::kData: {
+          size_t take = std::min((size_t)remaining_, n - i);
+          payload(p + i, take);
+          i += take;
+          remaining_ -= (int64_t)take;
+          if (remaining_ == 0) state_ = State::kDataCrlf, crlf_ = 0;
+          break;
+        }
+        case State::kDataCrlf: {
+          char c = p[i++];
+          if (c == '\n') state_ = State::kSize;
+          else if (c != '\r') state_ = State::kError;
+          break;
+        }
+        case State::kTrailer: {
+          char c = p[i++];
+          line_.push_back(c);
+          if (line_.size() > 4096) { state_ = State::kError; break; }
+          if (c == '\n') {
+            if (line_ == "\r\n" || line_ == "\n") state_ = State::kDone;
+            line_.clear();
+          }
+          break;
+        }
+        default:
+          break;
+      }
+    }
+  }
+
+ private:
+  enum class State { kSize, kData, kDataCrlf, kTrailer, kDone, kError };
+  State state_ = State::kSize;
+  std::string line_;
+  int64_t remaining_ = 0;
+  int crlf_ = 0;
+};
+
+// -------- usage extraction --------------------------------------------------
+
+// scan a JSON fragment for the token counters inside its "usage" object
+// (translator/openai_openai.go:185-223 equivalent: cumulative counts, the
+// caller keeps the max). Flat byte scan — the fragment is trusted JSON
+// from the upstream and the keys are unambiguous in OpenAI responses.
+bool extract_usage(const char* p, size_t n, Usage* u) {
+  const char* up = (const char*)memmem(p, n, "\"usage\"", 7);
+  if (!up) return false;
+  size_t off = (size_t)(up - p);
+  auto grab = [&](const char* key, size_t klen) -> int64_t {
+    const char* k = (const char*)memmem(p + off, n - off, key, klen);
+    if (!k) return -1;
+    const char* q = k + klen;
+    const char* e = p + n;
+    while (q < e && (*q == ':' || *q == ' ' || *q == '\t')) ++q;
+    if (q >= e || *q < '0' || *q > '9') return -1;
+    int64_t v = 0;
+    while (q < e && *q >= '0' && *q <= '9') v = v * 10 + (*q++ - '0');
+    return v;
+  };
+  int64_t in = grab("\"prompt_tokens\"", 15);
+  int64_t out = grab("\"completion_tokens\"", 19);
+  int64_t total = grab("\"total_tokens\"", 14);
+  if (in < 0) in = grab("\"input_tokens\"", 14);
+  if (out < 0) out = grab("\"output_tokens\"", 15);
+  bool any = false;
+  if (in > 0) u->input = std::max(u->input, in), any = true;
+  if (out > 0) u->output = std::max(u->output, out), any = true;
+  if (total > 0) u->total = std::max(u->total, total), any = true;
+  if (u->total == 0 && (u->input || u->output)) u->total = u->input + u->output;
+  return any;
+}
+
+// -------- body splices ------------------------------------------------------
+
+// replace the top-level "model" value span with the override
+std::string splice_model(const std::string& body, size_t vs, size_t ve,
+                         const std::string& override_model) {
+  std::string out;
+  out.reserve(body.size() + override_model.size() + 2);
+  out.append(body, 0, vs);
+  out.push_back('"');
+  out += override_model;  // config-supplied, no escaping needed
+  out.push_back('"');
+  out.append(body, ve, body.size() - ve);
+  return out;
+}
+
+// force stream_options.include_usage=true (endpointspec.go:138-154): when
+// the body has no stream_options, insert one before the final '}'.
+bool splice_include_usage(std::string* body) {
+  if (memmem(body->data(), body->size(), "\"stream_options\"", 16)) return false;
+  size_t close = body->rfind('}');
+  if (close == std::string::npos) return false;
+  body->insert(close, ",\"stream_options\":{\"include_usage\":true}");
+  return true;
+}
+
+}  // namespace
+
+// -------- upstream pool -----------------------------------------------------
+
+class UpstreamPool {
+ public:
+  int acquire(const std::string& host, uint16_t port, double timeout_s) {
+    const std::string key = host + ":" + std::to_string(port);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = idle_.find(key);
+      while (it != idle_.end() && !it->second.empty()) {
+        int fd = it->second.back();
+        it->second.pop_back();
+        // liveness probe: a reusable idle conn has no readable bytes;
+        // readable or EOF means the peer closed or sent garbage
+        char tmp;
+        ssize_t r = ::recv(fd, &tmp, 1, MSG_PEEK | MSG_DONTWAIT);
+        if (r == -1 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+          set_timeout(fd, timeout_s);
+          return fd;
+        }
+        ::close(fd);
+      }
+    }
+    return tcp_connect(host, port, timeout_s);
+  }
+
+  void release(const std::string& host, uint16_t port, int fd, bool reusable) {
+    if (!reusable) {
+      ::close(fd);
+      return;
+    }
+    const std::string key = host + ":" + std::to_string(port);
+    std::lock_guard<std::mutex> lk(mu_);
+    auto& v = idle_[key];
+    if (v.size() >= 512) {
+      ::close(fd);
+      return;
+    }
+    v.push_back(fd);
+  }
+
+  void close_all() {
+    std::lock_guard<std::mutex> lk(mu_);
+    for (auto& kv : idle_)
+      for (int fd : kv.second) ::close(fd);
+    idle_.clear();
+  }
+
+ private:
+  std::mutex mu_;
+  std::map<std::string, std::vector<int>> idle_;
+};
+
+// -------- GPU admission client ----------------------------------------------
+
+class GpuAdmissionClient {
+ public:
+  struct Waiter {
+    std::mutex m;
+    std::condition_variable cv;
+    bool done = false;
+    int64_t count = 0;
+  };
+
+  bool start(const std::string& socket_path, int window_us, int max_batch) {
+    window_us_ = window_us;
+    max_batch_ = max_batch;
+    fd_ = ::socket(AF_UNIX, SOCK_STREAM, 0);
+    if (fd_ < 0) return false;
+    struct sockaddr_un addr;
+    memset(&addr, 0, sizeof(addr));
+    addr.sun_family = AF_UNIX;
+    strncpy(addr.sun_path, socket_path.c_str(), sizeof(addr.sun_path) - 1);
+    if (::connect(fd_, (struct sockaddr*)&addr, sizeof(addr)) != 0) {
+      ::close(fd_);
+      fd_ = -1;
+      return false;
+    }
+    batcher_ = std::thread([this] { batch_loop(); });
+    reader_ = std::thread([this] { read_loop(); });
+    return true;
+  }
+
+  void stop() {
+    stopping_ = true;
+    {
+      std::lock_guard<std::mutex> lk(qmu_);
+      qcv_.notify_all();
+    }
+    if (fd_ >= 0) ::shutdown(fd_, SHUT_RDWR);
+    if (batcher_.joinable()) batcher_.join();
+    if (reader_.joinable()) reader_.join();
+    if (fd_ >= 0) ::close(fd_), fd_ = -1;
+    fail_all();
+  }
+
+  ~GpuAdmissionClient() { stop(); }
+
+  // blocks until the batched RPC answers; 0 on service failure (the
+  // request proceeds without GPU accounting, like the Python path when
+  // gpu_services is absent)
+  int64_t count_text(const std::string& text) {
+    if (stopping_ || fd_ < 0) return 0;
+    auto w = std::make_shared<Waiter>();
+    {
+      std::lock_guard<std::mutex> lk(qmu_);
+      q_texts_.push_back(text);
+      q_waiters_.push_back(w);
+      qcv_.notify_one();
+    }
+    std::unique_lock<std::mutex> lk(w->m);
+    w->cv.wait_for(lk, std::chrono::seconds(30), [&] { return w->done; });
+    return w->count;
+  }
+
+ private:
+  void batch_loop() {
+    while (!stopping_) {
+      std::vector<std::string> texts;
+      std::vector<std::shared_ptr<Waiter>> waiters;
+      {
+        std::unique_lock<std::mutex> lk(qmu_);
+        qcv_.wait(lk, [&] { return stopping_ || !q_texts_.empty(); });
+        if (stopping_) return;
+      }
+      // micro-batch window: collect arrivals across ALL connections
+      if (window_us_ > 0)
+        std::this_thread::sleep_for(std::chrono::microseconds(window_us_));
+      {
+        std::lock_guard<std::mutex> lk(qmu_);
+        size_t take = std::min(q_texts_.size(), (size_t)max_batch_);
+        texts.assign(q_texts_.begin(), q_texts_.begin() + take);
+        waiters.assign(q_waiters_.begin(), q_waiters_.begin() + take);
+        q_texts_.erase(q_texts_.begin(), q_texts_.begin() + take);
+        q_waiters_.erase(q_waiters_.begin(), q_waiters_.begin() + take);
+      }
+      if (texts.empty()) continue;
+      uint64_t id = next_id_++;
+      {
+        std::lock_guard<std::mutex> lk(pmu_);
+        pending_[id] = waiters;
+      }
+      MsgpackWriter w;
+      w.map_header(3);
+      w.str("id", 2);
+      w.uint(id);
+      w.str("op", 2);
+      w.str("count_batch", 11);
+      w.str("texts", 5);
+      w.array_header((uint32_t)texts.size());
+      for (const auto& t : texts) w.bin(t.data(), t.size());
+      uint32_t len = (uint32_t)w.out.size();
+      char hdr[4] = {(char)(len & 0xff), (char)((len >> 8) & 0xff),
+                     (char)((len >> 16) & 0xff), (char)((len >> 24) & 0xff)};
+      std::string frame(hdr, 4);
+      frame += w.out;
+      if (!write_all(fd_, frame)) {
+        stopping_ = true;
+        fail_all();
+        return;
+      }
+    }
+  }
+
+  void read_loop() {
+    std::string buf;
+    char tmp[65536];
+    while (!stopping_) {
+      ssize_t r = read_some(fd_, tmp, sizeof(tmp));
+      if (r <= 0) break;
+      buf.append(tmp, (size_t)r);
+      while (buf.size() >= 4) {
+        uint32_t len = (uint8_t)buf[0] | ((uint8_t)buf[1] << 8) |
+                       ((uint8_t)buf[2] << 16) | ((uint8_t)buf[3] << 24);
+        if (buf.size() < 4 + (size_t)len) break;
+        handle_reply(buf.data() + 4, len);
+        buf.erase(0, 4 + (size_t)len);
+      }
+    }
+    stopping_ = true;
+    fail_all();
+  }
+
+  void handle_reply(const char* p, size_t n) {
+    MsgpackReader r(p, n);
+    int pairs = r.map_header();
+    uint64_t id = 0;
+    std::vector<int64_t> counts;
+    for (int i = 0; i < pairs && r.ok(); ++i) {
+      std::string key;
+      if (!r.str(&key)) return;
+      if (key == "id") {
+        int64_t v;
+        if (!r.integer(&v)) return;
+        id = (uint64_t)v;
+      } else if (key == "counts") {
+        int m = r.array_header();
+        for (int j = 0; j < m && r.ok(); ++j) {
+          int64_t v;
+          if (!r.integer(&v)) return;
+          counts.push_back(v);
+        }
+      } else {
+        r.skip();
+      }
+    }
+    std::vector<std::shared_ptr<Waiter>> waiters;
+    {
+      std::lock_guard<std::mutex> lk(pmu_);
+      auto it = pending_.find(id);
+      if (it == pending_.end()) return;
+      waiters = std::move(it->second);
+      pending_.erase(it);
+    }
+    for (size_t i = 0; i < waiters.size(); ++i) {
+      auto& w = waiters[i];
+      std::lock_guard<std::mutex> lk(w->m);
+      w->count = i < counts.size() ? counts[i] : 0;
+      w->done = true;
+      w->cv.notify_all();
+    }
+  }
+
+  void fail_all() {
+    std::vector<std::shared_ptr<Waiter>> all;
+    {
+      std::lock_guard<std::mutex> lk(qmu_);
+      all.insert(all.end(), q_waiters_.begin(), q_waiters_.end());
+      q_waiters_.clear();
+      q_texts_.clear();
+    }
+    {
+      std::lock_guard<std::mutex> lk(pmu_);
+      for (auto& kv : pending_)
+        all.insert(all.end(), kv.second.begin(), kv.second.end());
+      pending_.clear();
+    }
+    for (auto& w : all) {
+      std::lock_guard<std::mutex> lk(w->m);
+      w->done = true;
+      w->cv.notify_all();
+    }
+  }
+
+  int fd_ = -1;
+  int window_us_ = 100;
+  int max_batch_ = 256;
+  std::atomic<bool> stopping_{false};
+  std::atomic<uint64_t> next_id_{1};
+  std::mutex qmu_;
+  std::condition_variable qcv_;
+  std::vector<std::string> q_texts_;
+  std::vector<std::shared_ptr<GpuAdmissionClient::Waiter>> q_waiters_;
+  std::mutex pmu_;
+  std::map<uint64_t, std::vector<std::shared_ptr<Waiter>>> pending_;
+  std::thread batcher_, reader_;
+};
+
+// -------- connection handler ------------------------------------------------
+
+namespace {
+
+// client headers never forwarded upstream (aigw/extproc/server.py
+// _HOP_BY_HOP + internal x-ai-eg-* spoof-strip + override-strip)
+bool banned_upstream_header(const std::string& name) {
+  static const char* kBanned[] = {
+      "host", "content-length", "content-type", "connection", "keep-alive",
+      "proxy-authenticate", "proxy-authorization", "te", "trailer",
+      "transfer-encoding", "upgrade", "expect", "accept-encoding",
+      "authorization"};
+  for (const char* b : kBanned)
+    if (name == b) return true;
+  if (name.compare(0, 8, "x-ai-eg-") == 0) return true;
+  if (name.compare(0, 7, "x-aigw-") == 0) return true;
+  return false;
+}
+
+}  // namespace
+
+class ConnHandler {
+ public:
+  ConnHandler(FastServer* srv, int fd) : srv_(srv), fd_(fd) {}
+
+  void run() {
+    std::string buf;
+    char tmp[65536];
+    for (;;) {
+      // locate a complete head
+      size_t head_end;
+      for (;;) {
+        head_end = buf.find("\r\n\r\n");
+        if (head_end != std::string::npos) break;
+        if (buf.size() > kMaxHead) return;
+        ssize_t r = read_some(fd_, tmp, sizeof(tmp));
+        if (r <= 0) return;
+        srv_->stats_.bytes_in += (uint64_t)r;
+        buf.append(tmp, (size_t)r);
+      }
+      HttpHead req;
+      if (!parse_head(buf, head_end, true, &req)) {
+        simple_reply(400, "invalid_request_error", "malformed request", true);
+        return;
+      }
+      // desync hardening (identical to the Python lean front)
+      int64_t clen = 0;
+      bool te = false, dup_conflict = false;
+      {
+        const std::string* seen = nullptr;
+        for (const auto& h : req.headers) {
+          if (h.name == "transfer-encoding") te = true;
+          if (h.name == "content-length") {
+            if (seen && *seen != h.value) dup_conflict = true;
+            seen = &h.value;
+          }
+        }
+        if (seen) clen = atoll(seen->c_str());
+      }
+      if (te) {
+        simple_reply(501, "invalid_request_error", "transfer-encoding not supported", true);
+        return;
+      }
+      if (dup_conflict || clen < 0 || (size_t)clen > kMaxBody) {
+        simple_reply(400, "invalid_request_error", "bad content-length", true);
+        return;
+      }
+      while (buf.size() < req.head_len + (size_t)clen) {
+        ssize_t r = read_some(fd_, tmp, sizeof(tmp));
+        if (r <= 0) return;
+        srv_->stats_.bytes_in += (uint64_t)r;
+        buf.append(tmp, (size_t)r);
+      }
+      std::string body = buf.substr(req.head_len, (size_t)clen);
+      buf.erase(0, req.head_len + (size_t)clen);
+
+      bool keep = handle_request(req, body);
+      if (!keep) return;
+    }
+  }
+
+ private:
+  bool handle_request(const HttpHead& req, const std::string& body) {
+    srv_->stats_.requests++;
+    if (req.method == "GET" && req.path == "/health") {
+      return raw_reply(200, "application/json",
+                       "{\"status\":\"ok\",\"front\":\"fast\"}");
+    }
+    bool hot = req.method == "POST" &&
+               (req.path == "/v1/chat/completions" || req.path == "/v1/completions" ||
+                req.path == "/v1/embeddings");
+    if (!hot) return fallback(req, body);
+
+    int64_t t0 = now_us();
+    // one-pass native scan: model + stream + chat text + model span
+    aigw_core::Scan sc{body.data(), body.data() + body.size()};
+    sc.base = body.data();
+    bool ok = sc.parse_value(0, "", true) && sc.ok;
+    if (ok) {
+      sc.ws();
+      ok = sc.p == sc.end;
+    }
+    if (!ok) {
+      srv_->stats_.responses_4xx++;
+      return simple_reply(400, "invalid_request_error",
+                          "invalid request body: malformed JSON", false);
+    }
+    if (sc.model.empty()) {
+      srv_->stats_.responses_4xx++;
+      return simple_reply(400, "invalid_request_error",
+                          "missing required field 'model'", false);
+    }
+    const FastRoute* route = nullptr;
+    for (const auto& r : srv_->routes_) {
+      if (r.catch_all || r.model_match == sc.model) {
+        route = &r;
+        break;
+      }
+    }
+    if (!route) {
+      srv_->stats_.responses_4xx++;
+      return simple_reply(404, "model_not_found",
+                          "no route matched model " + sc.model, false);
+    }
+    if (!route->eligible) return fallback(req, body);
+    bool stream = sc.stream == 1 && req.path != "/v1/embeddings";
+
+    double retry_after = 0;
+    std::string rule_name;
+    if (!srv_->rl_check(&retry_after, &rule_name)) {
+      srv_->stats_.local_429++;
+      srv_->stats_.responses_4xx++;
+      std::string extra = "retry-after: " + std::to_string((int)retry_after + 1) + "\r\n";
+      return simple_reply(429, "rate_limit_exceeded", "token budget exhausted",
+                          false, extra);
+    }
+
+    int64_t gpu_tokens = 0;
+    if (srv_->gpu_ && req.path == "/v1/chat/completions" && !sc.text.empty()) {
+      gpu_tokens = srv_->gpu_->count_text(sc.text);
+      srv_->stats_.gpu_tokens += (uint64_t)gpu_tokens;
+    }
+
+    // attempt order: tiers by priority, weighted shuffle inside a tier
+    std::vector<const FastBackend*> order = attempt_order(*route);
+    int attempts_left = std::min((int)order.size(), route->retries + 1);
+    bool first = true;
+    for (const FastBackend* be : order) {
+      if (attempts_left-- <= 0) break;
+      if (!first) srv_->stats_.retries++;
+      first = false;
+      int outcome = try_backend(req, body, sc, *route, *be, stream, gpu_tokens, t0);
+      if (outcome == 0) return true;    // handled, keep-alive
+      if (outcome == 2) return false;   // handled, close
+      // outcome 1: retriable failure — next backend re-splices the
+      // ORIGINAL body (per-try translation, A.8)
+    }
+    srv_->stats_.responses_5xx++;
+    return simple_reply(503, "upstream_error", "no healthy upstream", false);
+  }
+
+  std::vector<const FastBackend*> attempt_order(const FastRoute& route) {
+    std::vector<const FastBackend*> order;
+    order.reserve(route.backends.size());
+    // backends are pre-sorted by priority; weighted order inside a tier
+    // (Efraimidis-Spirakis keys, matching aigw/extproc/router.py)
+    static thread_local std::mt19937_64 rng{std::random_device{}()};
+    std::uniform_real_distribution<double> uni(1e-12, 1.0);
+    size_t i = 0;
+    while (i < route.backends.size()) {
+      size_t j = i;
+      while (j < route.backends.size() &&
+             route.backends[j].priority == route.backends[i].priority)
+        ++j;
+      std::vector<std::pair<double, const FastBackend*>> keyed;
+      for (size_t k = i; k < j; ++k) {
+        double w = std::max(route.backends[k].weight, 1e-9);
+        keyed.push_back({-std::pow(uni(rng), 1.0 / w), &route.backends[k]});
+      }
+      std::sort(keyed.begin(), keyed.end(),
+                [](const auto& a, const auto& b) { return a.first < b.first; });
+      for (auto& kv : keyed) order.push_back(kv.second);
+      i = j;
+    }
+    return order;
+  }
+
+  // returns 0 done+keepalive, 1 retriable failure, 2 done+close
+  int try_backend(const HttpHead& req, const std::string& body,
+                  const aigw_core::Scan& sc, const FastRoute& route,
+                  const FastBackend& be, bool stream, int64_t gpu_tokens,
+                  int64_t t0) {
+    // per-try body: splice from ORIGINAL bytes each attempt
+    std::string out_body;
+    const std::string* send_body = &body;
+    if (!be.model_override.empty() && sc.model_ve > sc.model_vs) {
+      out_body = splice_model(body, sc.model_vs, sc.model_ve, be.model_override);
+      send_body = &out_body;
+    }
+    if (stream && route.has_costs) {
+      if (send_body == &body) out_body = body, send_body = &out_body;
+      splice_include_usage(&out_body);
+    }
+
+    std::string head;
+    head.reserve(512 + req.head_len);
+    head += "POST ";
+    head += req.path;
+    head += " HTTP/1.1\r\nhost: ";
+    head += be.host;
+    if (be.port != 80 && be.port != 443) head += ":" + std::to_string(be.port);
+    head += "\r\ncontent-length: " + std::to_string(send_body->size());
+    head += "\r\ncontent-type: application/json\r\n";
+    for (const auto& h : req.headers) {
+      if (banned_upstream_header(h.name)) continue;
+      head += h.name;
+      head += ": ";
+      head += h.value;
+      head += "\r\n";
+    }
+    if (!be.bearer.empty()) {
+      head += "authorization: Bearer ";
+      head += srv_->resolve_bearer(be);
+      head += "\r\n";
+    } else {
+      // propagate client Authorization for auth-less backends
+      const std::string* auth = req.get("authorization");
+      if (auth) head += "authorization: " + *auth + "\r\n";
+    }
+    head += "\r\n";
+
+    int ufd = srv_->pool_->acquire(be.host, be.port, be.timeout_s);
+    if (ufd < 0) return 1;
+    if (!write_all(ufd, head) || !write_all(ufd, *send_body)) {
+      ::close(ufd);
+      return 1;
+    }
+
+    // response head
+    std::string rbuf;
+    char tmp[65536];
+    size_t head_end;
+    for (;;) {
+      head_end = rbuf.find("\r\n\r\n");
+      if (head_end != std::string::npos) break;
+      if (rbuf.size() > kMaxHead) {
+        ::close(ufd);
+        return 1;
+      }
+      ssize_t r = read_some(ufd, tmp, sizeof(tmp));
+      if (r <= 0) {
+        ::close(ufd);
+        return 1;
+      }
+      rbuf.append(tmp, (size_t)r);
+    }
+    HttpHead resp;
+    if (!parse_head(rbuf, head_end, false, &resp)) {
+      ::close(ufd);
+      return 1;
+    }
+    rbuf.erase(0, resp.head_len);
+    bool up_close = false;
+    {
+      const std::string* c = resp.get("connection");
+      if (c && lower(*c).find("close") != std::string::npos) up_close = true;
+    }
+
+    static const int kRetriable[] = {429, 500, 502, 503, 504};
+    bool retriable = false;
+    for (int s : kRetriable) retriable |= (resp.status == s);
+
+    int64_t length = -1;
+    Framing fr = response_framing(resp, &length);
+
+    if (resp.status >= 400) {
+      std::string err_body;
+      if (!read_full_body(ufd, fr, length, rbuf, &err_body)) {
+        ::close(ufd);
+        return 1;
+      }
+      srv_->pool_->release(be.host, be.port, ufd, !up_close && fr != Framing::kClose);
+      if (retriable) return 1;
+      srv_->stats_.responses_4xx++;
+      record_latency(t0);
+      return forward_buffered(resp, err_body) ? 0 : 2;
+    }
+
+    if (stream) {
+      int rc = relay_stream(resp, fr, length, rbuf, ufd, be, up_close,
+                            gpu_tokens, t0);
+      return rc;
+    }
+
+    std::string resp_body;
+    if (!read_full_body(ufd, fr, length, rbuf, &resp_body)) {
+      ::close(ufd);
+      return 1;
+    }
+    srv_->pool_->release(be.host, be.port, ufd, !up_close && fr != Framing::kClose);
+    Usage u;
+    if (!resp.get("content-encoding"))
+      extract_usage(resp_body.data(), resp_body.size(), &u);
+    finish_usage(u, gpu_tokens);
+    srv_->stats_.responses_2xx++;
+    record_latency(t0);
+    return forward_buffered(resp, resp_body) ? 0 : 2;
+  }
+
+  // read a complete framed body from the upstream, decoded (de-chunked)
+  bool read_full_body(int fd, Framing fr, int64_t length, std::string& rbuf,
+                      std::string* out) {
+    char tmp[65536];
+    if (fr == Framing::kLength) {
+      out->assign(rbuf, 0, std::min((size_t)length, rbuf.size()));
+      rbuf.erase(0, out->size());
+      while ((int64_t)out->size() < length) {
+        ssize_t r = read_some(fd, tmp, sizeof(tmp));
+        if (r <= 0) return false;
+        out->append(tmp, (size_t)r);
+      }
+      if ((int64_t)out->size() > length) {
+        rbuf.insert(0, out->substr((size_t)length));
+        out->resize((size_t)length);
+      }
+      return true;
+    }
+    if (fr == Framing::kChunked) {
+      ChunkedParser cp;
+      auto tap = [&](const char* p, size_t n) { out->append(p, n); };
+      cp.feed(rbuf.data(), rbuf.size(), tap);
+      rbuf.clear();
+      while (!cp.done() && !cp.error()) {
+        ssize_t r = read_some(fd, tmp, sizeof(tmp));
+        if (r <= 0) return false;
+        cp.feed(tmp, (size_t)r, tap);
+      }
+      return cp.done();
+    }
+    // close-delimited
+    out->assign(rbuf);
+    rbuf.clear();
+    for (;;) {
+      ssize_t r = read_some(fd, tmp, sizeof(tmp));
+      if (r < 0) return false;
+      if (r == 0) return true;
+      out->append(tmp, (size_t)r);
+      if (out->size() > kMaxBody) return false;
+    }
+  }
+
+  // forward a buffered upstream response (status + filtered headers +
+  // explicit content-length + body); returns keep-alive viability
+  bool forward_buffered(const HttpHead& resp, const std::string& body) {
+    std::string head = "HTTP/1.1 " + std::to_string(resp.status) + " X\r\n";
+    for (const auto& h : resp.headers) {
+      if (h.name == "connection" || h.name == "keep-alive" ||
+          h.name == "transfer-encoding" || h.name == "content-length")
+        continue;
+      head += h.name;
+      head += ": ";
+      head += h.value;
+      head += "\r\n";
+    }
+    head += "content-length: " + std::to_string(body.size()) + "\r\n\r\n";
+    srv_->stats_.bytes_out += head.size() + body.size();
+    if (!write_all(fd_, head)) return false;
+    return write_all(fd_, body);
+  }
+
+  // relay a streamed response with its upstream framing preserved and a
+  // usage tap on the payload (SSE re-chunk parity: the bytes are already
+  // SSE; this path forwards them unmodified, per OpenAI passthrough)
+  int relay_stream(const HttpHead& resp, Framing fr, int64_t length,
+                   std::string& rbuf, int ufd, const FastBackend& be,
+                   bool up_close, int64_t gpu_tokens, int64_t t0) {
+    std::string head = "HTTP/1.1 " + std::to_string(resp.status) + " X\r\n";
+    for (const auto& h : resp.headers) {
+      if (h.name == "connection" || h.name == "keep-alive") continue;
+      // content-length dropped on streams (A.8) unless length-framed relay
+      if (h.name == "content-length" && fr != Framing::kLength) continue;
+      head += h.name;
+      head += ": ";
+      head += h.value;
+      head += "\r\n";
+    }
+    bool client_close = fr == Framing::kClose;
+    if (client_close) head += "connection: close\r\n";
+    head += "\r\n";
+    if (!write_all(fd_, head)) {
+      ::close(ufd);
+      return 2;
+    }
+    srv_->stats_.bytes_out += head.size();
+
+    Usage u;
+    bool scan_usage = !resp.get("content-encoding");
+    ChunkedParser cp;
+    char tmp[65536];
+    int64_t remaining = length;
+    bool done = false, up_err = false, client_err = false;
+    auto tap = [&](const char* p, size_t n) {
+      if (scan_usage && memmem(p, n, "usage", 5)) extract_usage(p, n, &u);
+    };
+    // first: whatever arrived with the head
+    auto pump = [&](const char* p, size_t n) -> bool {
+      if (n == 0) return true;
+      if (fr == Framing::kChunked) {
+        cp.feed(p, n, tap);
+        if (cp.error()) up_err = true;
+        if (cp.done()) done = true;
+      } else {
+        tap(p, n);
+        if (fr == Framing::kLength) {
+          remaining -= (int64_t)n;
+          if (remaining <= 0) done = true;
+        }
+      }
+      srv_->stats_.bytes_out += n;
+      if (!write_all(fd_, p, n)) {
+        client_err = true;
+        return false;
+      }
+      return true;
+    };
+    pump(rbuf.data(), rbuf.size());
+    rbuf.clear();
+    while (!done && !up_err && !client_err) {
+      ssize_t r = read_some(ufd, tmp, sizeof(tmp));
+      if (r < 0) {
+        up_err = true;
+        break;
+      }
+      if (r == 0) {
+        if (fr == Framing::kClose) done = true;
+        else up_err = true;
+        break;
+      }
+      if (!pump(tmp, (size_t)r)) break;
+    }
+    finish_usage(u, gpu_tokens);
+    record_latency(t0);
+    if (up_err || client_err || fr == Framing::kClose || up_close) {
+      ::close(ufd);
+    } else {
+      srv_->pool_->release(be.host, be.port, ufd, true);
+    }
+    if (client_err) return 2;
+    srv_->stats_.responses_2xx++;
+    // a mid-stream upstream failure after bytes went out cannot fall
+    // back; the client sees truncated framing (stream cut) — close
+    if (up_err) return 2;
+    return client_close ? 2 : 0;
+  }
+
+  // relay any non-hot request to the loopback Python fallback app
+  bool fallback(const HttpHead& req, const std::string& body) {
+    srv_->stats_.fallback++;
+    if (srv_->fallback_port_ == 0)
+      return simple_reply(404, "not_found", "not found", false);
+    int ufd = srv_->pool_->acquire(srv_->fallback_host_, srv_->fallback_port_, 300.0);
+    if (ufd < 0)
+      return simple_reply(502, "upstream_error", "fallback unavailable", false);
+    std::string head = req.method + " " + req.path + " HTTP/1.1\r\n";
+    const std::string* host = req.get("host");
+    head += "host: " + (host ? *host : "127.0.0.1") + "\r\n";
+    head += "content-length: " + std::to_string(body.size()) + "\r\n";
+    for (const auto& h : req.headers) {
+      if (h.name == "host" || h.name == "content-length" ||
+          h.name == "connection" || h.name == "transfer-encoding")
+        continue;
+      head += h.name + ": " + h.value + "\r\n";
+    }
+    head += "\r\n";
+    if (!write_all(ufd, head) || !write_all(ufd, body)) {
+      ::close(ufd);
+      return simple_reply(502, "upstream_error", "fallback write failed", false);
+    }
+    std::string rbuf;
+    char tmp[65536];
+    size_t head_end;
+    for (;;) {
+      head_end = rbuf.find("\r\n\r\n");
+      if (head_end != std::string::npos) break;
+      ssize_t r = read_some(ufd, tmp, sizeof(tmp));
+      if (r <= 0 || rbuf.size() > kMaxHead) {
+        ::close(ufd);
+        return simple_reply(502, "upstream_error", "fallback read failed", false);
+      }
+      rbuf.append(tmp, (size_t)r);
+    }
+    HttpHead resp;
+    if (!parse_head(rbuf, head_end, false, &resp)) {
+      ::close(ufd);
+      return simple_reply(502, "upstream_error", "fallback bad response", false);
+    }
+    rbuf.erase(0, resp.head_len);
+    int64_t length = -1;
+    Framing fr = response_framing(resp, &length);
+    bool up_close = false;
+    {
+      const std::string* c = resp.get("connection");
+      if (c && lower(*c).find("close") != std::string::npos) up_close = true;
+    }
+    // relay with framing preserved (chunked streams stay streams)
+    std::string chead = "HTTP/1.1 " + std::to_string(resp.status) + " X\r\n";
+    for (const auto& h : resp.headers) {
+      if (h.name == "connection" || h.name == "keep-alive") continue;
+      chead += h.name + ": " + h.value + "\r\n";
+    }
+    bool client_close = fr == Framing::kClose;
+    if (client_close) chead += "connection: close\r\n";
+    chead += "\r\n";
+    if (!write_all(fd_, chead)) {
+      ::close(ufd);
+      return false;
+    }
+    srv_->stats_.bytes_out += chead.size();
+    ChunkedParser cp;
+    int64_t remaining = length;
+    bool done = (fr == Framing::kLength && length == 0), up_err = false;
+    auto pump = [&](const char* p, size_t n) -> bool {
+      if (n == 0) return true;
+      if (fr == Framing::kChunked) {
+        cp.feed(p, n, [](const char*, size_t) {});
+        if (cp.error()) up_err = true;
+        if (cp.done()) done = true;
+      } else if (fr == Framing::kLength) {
+        remaining -= (int64_t)n;
+        if (remaining <= 0) done = true;
+      }
+      srv_->stats_.bytes_out += n;
+      return write_all(fd_, p, n);
+    };
+    if (!pump(rbuf.data(), rbuf.size())) {
+      ::close(ufd);
+      return false;
+    }
+    rbuf.clear();
+    while (!done && !up_err) {
+      ssize_t r = read_some(ufd, tmp, sizeof(tmp));
+      if (r < 0) {
+        up_err = true;
+        break;
+      }
+      if (r == 0) {
+        if (fr == Framing::kClose) done = true;
+        else up_err = true;
+        break;
+      }
+      if (!pump(tmp, (size_t)r)) {
+        ::close(ufd);
+        return false;
+      }
+    }
+    if (resp.status < 400) srv_->stats_.responses_2xx++;
+    else if (resp.status < 500) srv_->stats_.responses_4xx++;
+    else srv_->stats_.responses_5xx++;
+    if (up_err || fr == Framing::kClose || up_close)
+      ::close(ufd);
+    else
+      srv_->pool_->release(srv_->fallback_host_, srv_->fallback_port_, ufd, true);
+    return !client_close && !up_err;
+  }
+
+  void finish_usage(Usage& u, int64_t gpu_tokens) {
+    if (u.total == 0 && gpu_tokens > 0) {
+      // upstream reported no usage: gateway's own GPU-tokenized count
+      u.input = gpu_tokens;
+      u.total = gpu_tokens + u.output;
+    }
+    srv_->stats_.input_tokens += (uint64_t)u.input;
+    srv_->stats_.output_tokens += (uint64_t)u.output;
+    srv_->stats_.total_tokens += (uint64_t)u.total;
+    srv_->rl_charge(u);
+  }
+
+  void record_latency(int64_t t0) {
+    int64_t us = now_us() - t0;
+    if (us < 1) us = 1;
+    int b = 63 - __builtin_clzll((uint64_t)us);
+    if (b > 31) b = 31;
+    srv_->stats_.latency_us_log2[b]++;
+  }
+
+  bool raw_reply(int status, const char* ctype, const std::string& body) {
+    std::string head = "HTTP/1.1 " + std::to_string(status) + " X\r\ncontent-type: " +
+                       ctype + "\r\ncontent-length: " + std::to_string(body.size()) +
+                       "\r\n\r\n";
+    srv_->stats_.bytes_out += head.size() + body.size();
+    if (!write_all(fd_, head)) return false;
+    return write_all(fd_, body);
+  }
+
+  bool simple_reply(int status, const char* type, const std::string& msg,
+                    bool close_after, const std::string& extra_headers = "") {
+    std::string body = "{\"type\":\"error\",\"error\":{\"type\":\"";
+    body += type;
+    body += "\",\"code\":\"";
+    body += std::to_string(status);
+    body += "\",\"message\":\"";
+    for (char c : msg) {
+      if (c == '"' || c == '\\') body.push_back('\\');
+      if ((unsigned char)c >= 0x20) body.push_back(c);
+    }
+    body += "\"}}";
+    std::string head = "HTTP/1.1 " + std::to_string(status) +
+                       " X\r\ncontent-type: application/json\r\ncontent-length: " +
+                       std::to_string(body.size()) + "\r\n" + extra_headers;
+    if (close_after) head += "connection: close\r\n";
+    head += "\r\n";
+    srv_->stats_.bytes_out += head.size() + body.size();
+    bool ok = write_all(fd_, head) && write_all(fd_, body);
+    return ok && !close_after;
+  }
+
+  FastServer* srv_;
+  int fd_;
+
+  friend class FastServer;
+};
+
+// -------- FastServer --------------------------------------------------------
+
+FastServer::FastServer() : pool_(new UpstreamPool) {}
+
+FastServer::~FastServer() { stop(); }
+
+void FastServer::add_route(FastRoute r) {
+  std::stable_sort(r.backends.begin(), r.backends.end(),
+                   [](const FastBackend& a, const FastBackend& b) {
+                     return a.priority < b.priority;
+                   });
+  if (r.model_match.empty()) r.catch_all = true;
+  routes_.push_back(std::move(r));
+}
+
+void FastServer::add_rate_rule(const RateRule& r) {
+  auto rs = std::make_unique<RuleState>();
+  rs->rule = r;
+  rs->window_start_ms = now_ms();
+  rules_.push_back(std::move(rs));
+}
+
+void FastServer::set_fallback(const std::string& host, uint16_t port) {
+  fallback_host_ = host;
+  fallback_port_ = port;
+}
+
+void FastServer::enable_gpu(const std::string& socket_path, int window_us,
+                            int max_batch) {
+  gpu_ = std::make_unique<GpuAdmissionClient>();
+  if (!gpu_->start(socket_path, window_us, max_batch)) {
+    gpu_.reset();
+    throw std::runtime_error("cannot connect GPU admission service at " + socket_path);
+  }
+}
+
+int FastServer::start(const std::string& host, uint16_t port) {
+  listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+  if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
+  int one = 1;
+  setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  struct sockaddr_in addr;
+  memset(&addr, 0, sizeof(addr));
+  addr.sin_family = AF_INET;
+  addr.sin_port = htons(port);
+  if (inet_pton(AF_INET, host.c_str(), &addr.sin_addr) != 1)
+    addr.sin_addr.s_addr = INADDR_ANY;
+  if (::bind(listen_fd_, (struct sockaddr*)&addr, sizeof(addr)) != 0) {
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+    throw std::runtime_error("bind failed: " + std::string(strerror(errno)));
+  }
+  if (::listen(listen_fd_, 4096) != 0) {
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+    throw std::runtime_error("listen failed");
+  }
+  socklen_t alen = sizeof(addr);
+  getsockname(listen_fd_, (struct sockaddr*)&addr, &alen);
+  acceptor_ = std::thread([this] { accept_loop(); });
+  return ntohs(addr.sin_port);
+}
+
+void FastServer::accept_loop() {
+  for (;;) {
+    int fd = ::accept(listen_fd_, nullptr, nullptr);
+    if (fd < 0) {
+      if (errno == EINTR) continue;
+      return;  // listener closed (stop)
+    }
+    if (stopping_) {
+      ::close(fd);
+      return;
+    }
+    set_nodelay(fd);
+    set_timeout(fd, 300.0);
+    {
+      std::lock_guard<std::mutex> lk(conn_mu_);
+      if (conn_fds_.size() >= 8192) {
+        ::close(fd);
+        continue;
+      }
+      conn_fds_.insert(fd);
+    }
+    stats_.active_connections++;
+    std::thread([this, fd] {
+      ConnHandler(this, fd).run();
+      {
+        std::lock_guard<std::mutex> lk(conn_mu_);
+        conn_fds_.erase(fd);
+      }
+      ::close(fd);
+      stats_.active_connections--;
+    }).detach();
+  }
+}
+
+void FastServer::stop() {
+  if (stopping_.exchange(true)) return;
+  if (listen_fd_ >= 0) {
+    ::shutdown(listen_fd_, SHUT_RDWR);
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+  }
+  if (acceptor_.joinable()) acceptor_.join();
+  // nudge live connections to finish: shut down their sockets; the
+  // per-connection threads exit on the read error
+  {
+    std::lock_guard<std::mutex> lk(conn_mu_);
+    for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
+  }
+  for (int i = 0; i < 200 && stats_.active_connections.load() > 0; ++i)
+    std::this_thread::sleep_for(std::chrono::milliseconds(10));
+  if (gpu_) gpu_->stop();
+  pool_->close_all();
+}
+
+std::string FastServer::resolve_bearer(const FastBackend& be) {
+  if (be.api_key_file.empty()) return be.bearer;
+  // mtime-cached file credential (rotation without restart)
+  std::lock_guard<std::mutex> lk(cred_mu_);
+  auto& fc = file_creds_[be.api_key_file];
+  int64_t now = now_ms();
+  if (now - fc.checked_ms > 1000 || fc.value.empty()) {
+    fc.checked_ms = now;
+    FILE* f = fopen(be.api_key_file.c_str(), "rb");
+    if (f) {
+      char buf[4096];
+      size_t n = fread(buf, 1, sizeof(buf) - 1, f);
+      fclose(f);
+      while (n > 0 && (buf[n - 1] == '\n' || buf[n - 1] == '\r' ||
+                       buf[n - 1] == ' '))
+        --n;
+      fc.value.assign(buf, n);
+    }
+  }
+  return fc.value.empty() ? be.bearer : fc.value;
+}
+
+// -------- rate limiting (fixed-window, cross-shard syncable) ----------------
+
+void FastServer::rl_roll(RuleState& rs, int64_t now) {
+  int64_t ws = rs.window_start_ms.load(std::memory_order_relaxed);
+  if (now - ws >= (int64_t)(rs.rule.window_s * 1000.0)) {
+    if (rs.window_start_ms.compare_exchange_strong(ws, now)) {
+      rs.local_spent = 0;
+      rs.remote_spent = 0;
+    }
+  }
+}
+
+bool FastServer::rl_check(double* retry_after_s, std::string* rule_name) {
+  int64_t now = now_ms();
+  for (auto& rsp : rules_) {
+    auto& rs = *rsp;
+    rl_roll(rs, now);
+    int64_t total = rs.local_spent.load(std::memory_order_relaxed) +
+                    rs.remote_spent.load(std::memory_order_relaxed);
+    if (total >= rs.rule.limit) {
+      double elapsed = (double)(now - rs.window_start_ms.load()) / 1000.0;
+      *retry_after_s = std::max(rs.rule.window_s - elapsed, 0.0);
+      *rule_name = rs.rule.name;
+      return false;
+    }
+  }
+  return true;
+}
+
+void FastServer::rl_charge(const Usage& u) {
+  if (rules_.empty()) return;
+  int64_t now = now_ms();
+  for (auto& rsp : rules_) {
+    auto& rs = *rsp;
+    int64_t cost = rs.rule.metadata_key == 1   ? u.input
+                   : rs.rule.metadata_key == 2 ? u.output
+                                               : u.total;
+    if (!cost) continue;
+    rl_roll(rs, now);
+    rs.local_spent += cost;
+    rs.pending_delta += cost;
+  }
+}
+
+std::vector<int64_t> FastServer::rl_collect_deltas() {
+  std::vector<int64_t> out;
+  out.reserve(rules_.size());
+  for (auto& rsp : rules_) out.push_back(rsp->pending_delta.exchange(0));
+  return out;
+}
+
+void FastServer::rl_apply_remote(const std::vector<int64_t>& others_spend) {
+  for (size_t i = 0; i < rules_.size() && i < others_spend.size(); ++i)
+    rules_[i]->remote_spent += others_spend[i];
+}
+
+std::vector<int64_t> FastServer::rl_local_spent() const {
+  std::vector<int64_t> out;
+  for (const auto& rsp : rules_) out.push_back(rsp->local_spent.load());
+  return out;
+}
+
+}  // namespace aigw_fast

@@ -1,0 +1,152 @@
+// aigw fast path — native data-plane server.
+//
+// The reference splits its data plane between the Envoy C++ proxy (HTTP
+// codecs, routing, retries) and a Go extproc (cmd/aigw/run.go:224-231,
+// extensionserver/post_translate_modify.go:768); this gateway's round-1
+// equivalent ran the whole pipeline in CPython workers, which profiling
+// capped at ~117k req/s per node without GPU work. This module moves the
+// passthrough hot loop — frame → scan → route → rate-limit → auth →
+// upstream dispatch → usage extraction → relay — into native threads
+// behind the same RuntimeConfig semantics, with CPython only composing
+// configuration and serving cold paths via a loopback fallback.
+//
+// Concurrency model: one OS thread per client connection (the reference
+// runs one goroutine per stream, extproc/server.go:128; at gateway-scale
+// connection counts — hundreds per shard, not tens of thousands — the
+// thread stack cost is irrelevant and blocking I/O removes an entire
+// state-machine layer). An acceptor thread owns the listener; a batcher
+// thread coalesces GPU admission calls ACROSS all connections into one
+// micro-batch RPC stream to the per-shard GPU service
+// (aigw/gpu/service.py), which is strictly better coalescing than the
+// round-1 per-worker windows.
+#pragma once
+
+#include <atomic>
+#include <condition_variable>
+#include <cstdint>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <set>
+#include <string>
+#include <thread>
+#include <vector>
+
+namespace aigw_fast {
+
+struct FastBackend {
+  std::string name;
+  std::string host;
+  uint16_t port = 0;
+  std::string bearer;          // static API-key credential; empty = none
+  std::string api_key_file;    // rotated-credential file (mtime-cached)
+  std::string model_override;  // modelNameOverride, spliced into the body
+  double weight = 1.0;
+  int priority = 0;
+  double timeout_s = 60.0;
+};
+
+struct FastRoute {
+  std::string name;
+  std::string model_match;  // exact x-ai-eg-model value; "" = catch-all
+  bool catch_all = false;
+  int retries = 1;
+  bool has_costs = false;  // force include_usage on streams, charge tokens
+  bool eligible = true;    // false => relay matched requests to fallback
+  std::vector<FastBackend> backends;  // sorted by (priority)
+};
+
+struct RateRule {
+  std::string name;
+  int64_t limit = 0;
+  double window_s = 60.0;
+  int metadata_key = 0;  // 0=total, 1=input, 2=output tokens
+};
+
+struct Usage {
+  int64_t input = 0, output = 0, total = 0;
+};
+
+struct ServerStats {
+  std::atomic<uint64_t> requests{0};
+  std::atomic<uint64_t> responses_2xx{0};
+  std::atomic<uint64_t> responses_4xx{0};
+  std::atomic<uint64_t> responses_5xx{0};
+  std::atomic<uint64_t> local_429{0};
+  std::atomic<uint64_t> fallback{0};
+  std::atomic<uint64_t> retries{0};
+  std::atomic<uint64_t> gpu_tokens{0};
+  std::atomic<uint64_t> input_tokens{0};
+  std::atomic<uint64_t> output_tokens{0};
+  std::atomic<uint64_t> total_tokens{0};
+  std::atomic<uint64_t> bytes_in{0};
+  std::atomic<uint64_t> bytes_out{0};
+  std::atomic<uint64_t> active_connections{0};
+  // log2-microsecond latency histogram (bucket i: [2^i, 2^(i+1)) us)
+  std::atomic<uint64_t> latency_us_log2[32] = {};
+};
+
+class GpuAdmissionClient;
+class UpstreamPool;
+
+class FastServer {
+ public:
+  FastServer();
+  ~FastServer();
+
+  // configuration (before start)
+  void add_route(FastRoute r);
+  void add_rate_rule(const RateRule& r);
+  void set_fallback(const std::string& host, uint16_t port);
+  void enable_gpu(const std::string& socket_path, int window_us, int max_batch);
+
+  // lifecycle
+  int start(const std::string& host, uint16_t port);  // returns bound port
+  void stop();
+
+  const ServerStats& stats() const { return stats_; }
+  // cross-shard rate-limit sync hooks (aigw.parallel.StateSync bridge)
+  std::vector<int64_t> rl_collect_deltas();
+  void rl_apply_remote(const std::vector<int64_t>& others_spend);
+  std::vector<int64_t> rl_local_spent() const;
+
+  std::string resolve_bearer(const FastBackend& be);
+
+ private:
+  friend class ConnHandler;
+  struct FileCredState {
+    int64_t checked_ms = 0;
+    std::string value;
+  };
+  struct RuleState {
+    RateRule rule;
+    std::atomic<int64_t> window_start_ms{0};
+    std::atomic<int64_t> local_spent{0};
+    std::atomic<int64_t> remote_spent{0};
+    std::atomic<int64_t> pending_delta{0};
+  };
+
+  void accept_loop();
+  void handle_connection(int fd);
+  bool rl_check(double* retry_after_s, std::string* rule_name);
+  void rl_charge(const Usage& u);
+  void rl_roll(RuleState& rs, int64_t now_ms);
+
+  std::vector<FastRoute> routes_;
+  std::vector<std::unique_ptr<RuleState>> rules_;
+  std::string fallback_host_;
+  uint16_t fallback_port_ = 0;
+  std::unique_ptr<GpuAdmissionClient> gpu_;
+  std::unique_ptr<UpstreamPool> pool_;
+  ServerStats stats_;
+
+  int listen_fd_ = -1;
+  std::atomic<bool> stopping_{false};
+  std::thread acceptor_;
+  std::mutex conn_mu_;
+  std::set<int> conn_fds_;
+  std::mutex cred_mu_;
+  std::map<std::string, FileCredState> file_creds_;
+};
+
+}  // namespace aigw_fast

@@ -20,6 +20,8 @@ struct Scan {
   int stream = 0;
   std::string text;
   bool ok = true;
+  const char* base = nullptr;     // set to the body start to capture spans
+  size_t model_vs = 0, model_ve = 0;  // "model" value span incl. quotes
 
   void ws() {
     while (p < end && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
@@ -186,7 +188,18 @@ struct Scan {
       bool is_model = depth == 1 && key == "model";
       bool is_text = (in_msgs && (key == "content" || key == "text")) ||
                      (depth == 1 && key == "system");
-      if (is_model) return parse_string(&model);
+      if (is_model) {
+        // record the value byte span (incl. quotes) when the caller set
+        // `base`, so the fast path can splice a model override without
+        // re-parsing (sjson-style, translator model override)
+        const char* v0 = p;
+        bool r = parse_string(&model);
+        if (base != nullptr) {
+          model_vs = (size_t)(v0 - base);
+          model_ve = (size_t)(p - base);
+        }
+        return r;
+      }
       if (is_text) {
         bool r = parse_string(&text);
         text.push_back('\n');

@@ -34,6 +34,14 @@ setup(
             extra_compile_args=["-O3", "-std=c++17"],
             language="c++",
         ),
+        Extension(
+            name="aigw_fast",
+            sources=["csrc/aigw_fast_module.cpp", "csrc/fastpath.cpp"],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17"],
+            extra_link_args=["-lpthread"],
+            language="c++",
+        ),
     ],
     cmdclass={"build_ext": BuildExtension},
 )

@@ -1,0 +1,273 @@
+"""Native C++ fast front (csrc/fastpath.cpp): the passthrough hot loop in
+native threads must reproduce the Python pipeline's observable behavior —
+auth injection, model override splice, forced include_usage, retries on
+dead backends, rate limiting, desync hardening — and relay every cold
+path to the Python fallback app unchanged."""
+
+import asyncio
+import json
+
+import aiohttp
+import pytest
+import yaml
+
+from aigw.extproc.fast_front import FastFront, FastFrontUnsupported, build_fast_server
+from aigw.extproc.server import GatewayServer
+from aigw.filterapi import RuntimeConfig, load_config
+from aigw.testing.mockupstream import start_mock_upstream
+
+pytestmark = pytest.mark.timeout(120)
+
+
+def _cfg(up_port, *, dead_port=1, rate_limit=None):
+    cfg = {
+        "version": "v1",
+        "llmRequestCosts": [{"metadataKey": "llm_total_token", "type": "TotalToken"}],
+        "routes": [
+            {
+                "name": "fast",
+                "headers": [{"name": "x-ai-eg-model", "value": "fast-model"}],
+                "backends": [
+                    {"name": "mock", "schema": "OpenAI",
+                     "upstream": {"host": "127.0.0.1", "port": up_port},
+                     "modelNameOverride": "real-model",
+                     "auth": {"apiKey": "sk-fast"}},
+                ],
+            },
+            {
+                "name": "retry",
+                "headers": [{"name": "x-ai-eg-model", "value": "retry-model"}],
+                "retries": 2,
+                "backends": [
+                    {"name": "dead", "schema": "OpenAI", "priority": 0,
+                     "upstream": {"host": "127.0.0.1", "port": dead_port}},
+                    {"name": "live", "schema": "OpenAI", "priority": 1,
+                     "upstream": {"host": "127.0.0.1", "port": up_port},
+                     "auth": {"apiKey": "sk-live"}},
+                ],
+            },
+            {
+                "name": "cold",
+                "headers": [{"name": "x-ai-eg-model", "value": "cold-model"}],
+                "backends": [
+                    # header mutation makes this route ineligible for the
+                    # native path -> per-request fallback to Python
+                    {"name": "mut", "schema": "OpenAI",
+                     "upstream": {"host": "127.0.0.1", "port": up_port},
+                     "headerMutation": {"set": {"x-extra": "1"}},
+                     "auth": {"apiKey": "sk-cold"}},
+                ],
+            },
+        ],
+        "models": [{"name": "fast-model"}],
+    }
+    if rate_limit:
+        cfg["rateLimits"] = [rate_limit]
+    return load_config(cfg)
+
+
+async def _start(cfg):
+    server = GatewayServer(RuntimeConfig(cfg))
+    front = FastFront(server, server.runtime)
+    port = await front.start("127.0.0.1", 0)
+    return front, port
+
+
+def test_fast_front_full_matrix():
+    async def run():
+        mock, up_runner, up_port = await start_mock_upstream()
+        mock.record = True
+        front, port = await _start(_cfg(up_port))
+        base = f"http://127.0.0.1:{port}"
+        async with aiohttp.ClientSession() as c:
+            # unary passthrough: model override spliced, auth injected,
+            # spoofable/override headers stripped
+            async with c.post(
+                f"{base}/v1/chat/completions",
+                json={"model": "fast-model",
+                      "messages": [{"role": "user", "content": "hello world"}]},
+                headers={"x-ai-eg-model": "spoof", "x-aigw-aws-access-key-id": "AK",
+                         "x-client-note": "keep"},
+            ) as r:
+                body = await r.json()
+                assert r.status == 200, body
+                assert body["model"] == "real-model"
+                assert body["usage"]["total_tokens"] > 0
+            rec = mock.requests[-1]
+            assert rec["headers"]["Authorization"] == "Bearer sk-fast"
+            assert "x-ai-eg-model" not in {k.lower() for k in rec["headers"]}
+            assert "x-aigw-aws-access-key-id" not in {k.lower() for k in rec["headers"]}
+            assert {k.lower(): v for k, v in rec["headers"].items()}["x-client-note"] == "keep"
+            assert json.loads(rec["body"])["model"] == "real-model"
+
+            # streaming: SSE relayed, include_usage forced (route has costs)
+            async with c.post(
+                f"{base}/v1/chat/completions",
+                json={"model": "fast-model", "stream": True,
+                      "messages": [{"role": "user", "content": "hi"}]},
+            ) as r:
+                assert r.status == 200
+                text = (await r.read()).decode()
+                assert text.rstrip().endswith("data: [DONE]")
+                assert '"usage"' in text  # forced include_usage
+            assert json.loads(mock.requests[-1]["body"])["stream_options"][
+                "include_usage"] is True
+
+            # ineligible route (header mutation) -> Python fallback serves it
+            async with c.post(
+                f"{base}/v1/chat/completions",
+                json={"model": "cold-model",
+                      "messages": [{"role": "user", "content": "x"}]},
+            ) as r:
+                assert r.status == 200
+            rec = mock.requests[-1]
+            assert {k.lower(): v for k, v in rec["headers"].items()}["x-extra"] == "1"
+
+            # cold paths ride the fallback app
+            async with c.get(f"{base}/v1/models") as r:
+                data = await r.json()
+                assert r.status == 200
+                assert data["data"][0]["id"] == "fast-model"
+            async with c.get(f"{base}/health") as r:
+                assert (await r.json())["front"] == "fast"
+
+            # local replies
+            async with c.post(f"{base}/v1/chat/completions",
+                              data=b"{broken",
+                              headers={"content-type": "application/json"}) as r:
+                assert r.status == 400
+            async with c.post(
+                f"{base}/v1/chat/completions",
+                json={"model": "nope", "messages": []},
+            ) as r:
+                assert r.status == 404
+
+        st = front.stats()
+        assert st["responses_2xx"] >= 3
+        assert st["fallback"] >= 2
+        assert st["total_tokens"] > 0
+        metrics = front.render_metrics_extra()
+        assert "aigw_fast_requests_total" in metrics
+        await front.stop()
+        await up_runner.cleanup()
+
+    asyncio.run(run())
+
+
+def test_fast_front_retry_and_rate_limit():
+    async def run():
+        mock, up_runner, up_port = await start_mock_upstream()
+        mock.record = True
+        cfg = _cfg(up_port, rate_limit={
+            "name": "budget", "metadataKey": "llm_total_token",
+            "limit": 10, "windowS": 3600,
+        })
+        front, port = await _start(cfg)
+        base = f"http://127.0.0.1:{port}"
+        async with aiohttp.ClientSession() as c:
+            # first backend is a dead port: native retry moves to tier 1
+            async with c.post(
+                f"{base}/v1/chat/completions",
+                json={"model": "retry-model",
+                      "messages": [{"role": "user", "content": "fallback"}]},
+            ) as r:
+                assert r.status == 200
+            assert mock.requests[-1]["headers"]["Authorization"] == "Bearer sk-live"
+            assert front.stats()["retries"] >= 1
+
+            # the first response charged ~17 tokens against limit 10:
+            # next request must be denied locally with retry-after
+            async with c.post(
+                f"{base}/v1/chat/completions",
+                json={"model": "fast-model",
+                      "messages": [{"role": "user", "content": "x"}]},
+            ) as r:
+                assert r.status == 429
+                assert int(r.headers["retry-after"]) > 0
+        assert front.stats()["local_429"] == 1
+        await front.stop()
+        await up_runner.cleanup()
+
+    asyncio.run(run())
+
+
+def test_fast_front_desync_hardening():
+    async def run():
+        mock, up_runner, up_port = await start_mock_upstream()
+        front, port = await _start(_cfg(up_port))
+
+        async def send(payload: bytes) -> bytes:
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            writer.write(payload)
+            try:
+                await writer.drain()
+                return await asyncio.wait_for(reader.read(65536), timeout=6)
+            finally:
+                writer.close()
+
+        data = await send(
+            b"POST /v1/chat/completions HTTP/1.1\r\n"
+            b"transfer-encoding: chunked\r\n\r\n2\r\n{}\r\n0\r\n\r\n"
+            b"GET /health HTTP/1.1\r\n\r\n"
+        )
+        assert data.startswith(b"HTTP/1.1 501"), data[:60]
+        assert data.count(b"HTTP/1.1") == 1
+        data = await send(
+            b"POST /v1/chat/completions HTTP/1.1\r\n"
+            b"content-length: 2\r\ncontent-length: 9\r\n\r\n{}1234567"
+        )
+        assert data.startswith(b"HTTP/1.1 400"), data[:60]
+        await front.stop()
+        await up_runner.cleanup()
+
+    asyncio.run(run())
+
+
+def test_fast_front_concurrent_storm():
+    async def run():
+        mock, up_runner, up_port = await start_mock_upstream()
+        front, port = await _start(_cfg(up_port))
+        base = f"http://127.0.0.1:{port}"
+
+        async def one(c, i):
+            stream = i % 3 == 0
+            async with c.post(
+                f"{base}/v1/chat/completions",
+                json={"model": "fast-model", "stream": stream,
+                      "messages": [{"role": "user", "content": f"req {i}"}]},
+            ) as r:
+                await r.read()
+                assert r.status == 200
+
+        async with aiohttp.ClientSession(
+            connector=aiohttp.TCPConnector(limit=64)) as c:
+            await asyncio.gather(*(one(c, i) for i in range(200)))
+        st = front.stats()
+        assert st["responses_2xx"] == 200
+        await front.stop()
+        await up_runner.cleanup()
+
+    asyncio.run(run())
+
+
+def test_fast_front_unsupported_configs():
+    cfg = load_config(yaml.safe_load("""
+routes:
+  - name: r
+    headers: [{name: x-custom, value: v}]
+    backends:
+      - {name: b, schema: OpenAI, upstream: {host: h, port: 80}}
+"""))
+    with pytest.raises(FastFrontUnsupported):
+        build_fast_server(RuntimeConfig(cfg))
+
+    cfg = load_config({
+        "routes": [{"name": "r", "backends": [
+            {"name": "b", "schema": "OpenAI",
+             "upstream": {"host": "h", "port": 80}}]}],
+        "rateLimits": [{"name": "k", "metadataKey": "llm_total_token",
+                        "limit": 10, "windowS": 60,
+                        "keyHeaders": ["x-user"]}],
+    })
+    with pytest.raises(FastFrontUnsupported):
+        build_fast_server(RuntimeConfig(cfg))
