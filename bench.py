@@ -222,6 +222,159 @@ def worker_entry(args, local_rank, ready, go, out_q):
     asyncio.run(worker_main(args, local_rank, ready, go, out_q))
 
 
+# ---- fast-front mode: ONE native C++ gateway per rank ------------------------
+#
+# The Python-front mode above needs 12 worker processes per shard because
+# CPython caps one process near ~6k req/s; the C++ front (csrc/fastpath.cpp)
+# serves the whole shard from native threads, so the process group becomes:
+# rank primary = gateway (+ GPU admission host), M mock-upstream processes,
+# L load-generator processes.
+
+
+def upstream_entry(args, port_q):
+    from aigw.testing.fastmock import canned_chat_response, start_fast_mock
+
+    async def amain():
+        srv, port = await start_fast_mock(
+            response=canned_chat_response(prompt_tokens=args.tokens)
+        )
+        port_q.put(port)
+        await asyncio.Event().wait()
+
+    asyncio.run(amain())
+
+
+def loadgen_entry(args, gw_port, direct_port, ready, go, out_q):
+    from aigw.utils import tune_gc
+
+    tune_gc()
+
+    async def amain():
+        from aigw.extproc.upstream_client import LeanClient
+
+        client = LeanClient()
+        payloads = [json.dumps(build_payload(args.tokens)).encode()]
+        path = "/v1/chat/completions"
+        waves = max(args.waves, 1)
+        scratch: list[float] = []
+        for _ in range(max(args.warmup, 1) * min(waves, 10)):
+            await fire_step(client, direct_port, path, payloads, args.batch, scratch)
+        for _ in range(args.warmup * waves):
+            await fire_step(client, gw_port, path, payloads, args.batch, scratch)
+        direct_lat: list[float] = []
+        for _ in range(5):
+            await fire_step(client, direct_port, path, payloads, args.batch, direct_lat)
+
+        ready.set()
+        while not go.is_set():
+            await asyncio.sleep(0.001)
+
+        lat: list[float] = []
+        t0 = time.perf_counter()
+        for _ in range(args.steps * waves):
+            await fire_step(client, gw_port, path, payloads, args.batch, lat)
+        elapsed = time.perf_counter() - t0
+        out_q.put(
+            {
+                "elapsed": elapsed,
+                "requests": args.steps * waves * args.batch,
+                "p50": statistics.median(lat),
+                "p99": sorted(lat)[max(int(len(lat) * 0.99) - 1, 0)],
+                "p50_direct": statistics.median(direct_lat),
+            }
+        )
+        await client.close()
+
+    asyncio.run(amain())
+
+
+def _start_fast_front(args, upstream_ports, gpu_socket):
+    """Run the C++ gateway (and its Python fallback app) on a dedicated
+    thread/loop in the rank primary; returns (front, port)."""
+    import threading
+
+    from aigw.extproc.fast_front import FastFront
+    from aigw.extproc.server import GatewayServer
+    from aigw.filterapi import RuntimeConfig, load_config
+
+    cfg_dict = gateway_config(upstream_ports[0])
+    backends = [
+        {
+            "name": f"mock-openai-{i}",
+            "schema": "OpenAI",
+            "upstream": {"host": "127.0.0.1", "port": p},
+            "auth": {"apiKey": "sk-bench"},
+            "weight": 1,
+        }
+        for i, p in enumerate(upstream_ports)
+    ]
+    cfg_dict["routes"][0]["backends"] = backends
+    cfg = load_config(cfg_dict)
+    server = GatewayServer(RuntimeConfig(cfg))
+    front = FastFront(
+        server, server.runtime, gpu_socket=gpu_socket or "",
+        gpu_window_us=int(args.gpu_window * 1000.0), gpu_max_batch=1024,
+    )
+    done = threading.Event()
+    state = {}
+
+    def run():
+        async def amain():
+            state["port"] = await front.start("127.0.0.1", 0)
+            done.set()
+            await asyncio.Event().wait()
+
+        asyncio.run(amain())
+
+    t = threading.Thread(target=run, daemon=True, name="aigw-fast-front")
+    t.start()
+    if not done.wait(timeout=600):
+        raise RuntimeError("fast front failed to start")
+    return front, state["port"]
+
+
+def run_fast_mode(args, rank, world, local_rank, use_gpu):
+    ctx = mp.get_context("spawn")
+    cores = os.cpu_count() or 8
+    # the C++ gateway saturates well past what one loadgen generates:
+    # size the harness (not the gateway) to the machine — the MI355X node
+    # pairs 256 CUs with 128+ EPYC cores; CI containers are much smaller
+    n_up = args.upstreams if args.upstreams > 0 else max(1, min(8, cores // 12))
+    loadgens = args.workers if args.workers > 0 else max(2, min(24, cores // 5))
+
+    port_q = ctx.Queue()
+    up_procs = [ctx.Process(target=upstream_entry, args=(args, port_q))
+                for _ in range(n_up)]
+    for p in up_procs:
+        p.start()
+    ports = [port_q.get(timeout=600) for _ in range(n_up)]
+
+    gpu_socket = None
+    if use_gpu:
+        gpu_socket = f"/tmp/aigw-gpu-{rank}-{os.getpid()}.sock"
+        _start_gpu_host(gpu_socket, local_rank, args.gpu_window)
+    front, gw_port = _start_fast_front(args, ports, gpu_socket)
+
+    ready_evts = [ctx.Event() for _ in range(loadgens)]
+    go = ctx.Event()
+    out_q = ctx.Queue()
+    procs = [
+        ctx.Process(
+            target=loadgen_entry,
+            args=(args, gw_port, ports[i % n_up], ready_evts[i], go, out_q),
+        )
+        for i in range(loadgens)
+    ]
+    for p in procs:
+        p.start()
+    for ev in ready_evts:
+        if not ev.wait(timeout=900):
+            for p in procs + up_procs:
+                p.terminate()
+            raise RuntimeError("load generator failed to become ready")
+    return front, go, out_q, procs, up_procs, loadgens
+
+
 def _start_gpu_host(socket_path: str, local_rank: int, window_ms: float = 0.1):
     """Run the shard's GPU admission service on a dedicated thread/loop in
     the rank-primary process (the only process owning a GPU context)."""
@@ -272,6 +425,12 @@ def main():
                     help="semantic-cache mode: cycle this many distinct "
                          "payloads per worker (0 = cache off); steady-state "
                          "hit ratio approaches 1")
+    ap.add_argument("--front", choices=["lean", "fast"], default="fast",
+                    help="fast = native C++ gateway (csrc/fastpath.cpp), one "
+                         "per rank; lean = Python lean front across --workers "
+                         "processes (the round-1 configuration)")
+    ap.add_argument("--upstreams", type=int, default=0,
+                    help="fast mode: mock-upstream processes per rank (0 = 8)")
     ap.add_argument("--gpu-window", type=float, default=0.1,
                     help="GPU micro-batch window per worker, ms")
     ap.add_argument("--gpu-service", action="store_true",
@@ -293,37 +452,48 @@ def main():
             torch.cuda.set_device(local_rank)
         torch.distributed.init_process_group(backend)
 
-    workers = args.workers
-    if workers <= 0:
-        cores = os.cpu_count() or 8
-        # measured knee on the 256-core MI355X box: throughput peaks at
-        # ~12-16 workers/shard (14.1k req/s @16) and collapses by 32
-        # (profiles/r01 bench_wsweep); ~3 cores per worker keeps headroom
-        # when 8 ranks share the node
-        workers = max(1, min(12, cores // (3 * max(world, 1))))
+    fast_mode = args.front == "fast" and not args.cache_payloads
+    front = None
+    up_procs: list = []
+    if fast_mode:
+        try:
+            front, go, out_q, procs, up_procs, workers = run_fast_mode(
+                args, rank, world, local_rank, use_gpu
+            )
+        except ImportError:
+            fast_mode = False
+    if not fast_mode:
+        workers = args.workers
+        if workers <= 0:
+            cores = os.cpu_count() or 8
+            # measured knee on the 256-core MI355X box: throughput peaks at
+            # ~12-16 workers/shard (14.1k req/s @16) and collapses by 32
+            # (profiles/r01 bench_wsweep); ~3 cores per worker keeps headroom
+            # when 8 ranks share the node
+            workers = max(1, min(12, cores // (3 * max(world, 1))))
 
-    gpu_socket = None
-    gpu_host_state = None
-    if use_gpu and args.gpu_service:
-        gpu_socket = f"/tmp/aigw-gpu-{rank}-{os.getpid()}.sock"
-        gpu_host_state = _start_gpu_host(gpu_socket, local_rank, args.gpu_window)
-    args.gpu_socket = gpu_socket
+        gpu_socket = None
+        gpu_host_state = None
+        if use_gpu and args.gpu_service:
+            gpu_socket = f"/tmp/aigw-gpu-{rank}-{os.getpid()}.sock"
+            gpu_host_state = _start_gpu_host(gpu_socket, local_rank, args.gpu_window)
+        args.gpu_socket = gpu_socket
 
-    ctx = mp.get_context("spawn")
-    ready_evts = [ctx.Event() for _ in range(workers)]
-    go = ctx.Event()
-    out_q = ctx.Queue()
-    procs = [
-        ctx.Process(target=worker_entry, args=(args, local_rank, ready_evts[i], go, out_q))
-        for i in range(workers)
-    ]
-    for p in procs:
-        p.start()
-    for ev in ready_evts:
-        if not ev.wait(timeout=600):
-            for p in procs:
-                p.terminate()
-            raise RuntimeError("bench worker failed to become ready")
+        ctx = mp.get_context("spawn")
+        ready_evts = [ctx.Event() for _ in range(workers)]
+        go = ctx.Event()
+        out_q = ctx.Queue()
+        procs = [
+            ctx.Process(target=worker_entry, args=(args, local_rank, ready_evts[i], go, out_q))
+            for i in range(workers)
+        ]
+        for p in procs:
+            p.start()
+        for ev in ready_evts:
+            if not ev.wait(timeout=600):
+                for p in procs:
+                    p.terminate()
+                raise RuntimeError("bench worker failed to become ready")
 
     def barrier_sync():
         if world > 1:
@@ -344,6 +514,8 @@ def main():
         p.join(timeout=60)
         if p.is_alive():
             p.terminate()
+    for p in up_procs:
+        p.terminate()
 
     statesync_tick_us = None
     if world > 1:
@@ -401,6 +573,7 @@ def main():
                 "waves_per_step": args.waves,
                 "seq_len": args.tokens,
                 "parallelism": f"dp{world}",
+                "front": "fast" if fast_mode else "lean",
                 "workers_per_shard": workers,
                 "p50_ms": round(p50, 3),
                 "p99_ms": round(p99, 3),
