@@ -458,7 +458,7 @@ def main():
     if fast_mode:
         try:
             front, go, out_q, procs, up_procs, workers = run_fast_mode(
-                args, rank, world, local_rank, use_gpu
+                args, rank, world, local_rank, use_gpu and not args.no_gpu
             )
         except ImportError:
             fast_mode = False
