@@ -387,6 +387,75 @@ class GcpOidcRotator(Rotator):
         return expires_at
 
 
+def build_rotation_manager(cfg) -> Optional["RotationManager"]:
+    """Scan a filterapi Config for backend auths carrying ``rotation``
+    and build the manager driving them (the controller's
+    BackendSecurityPolicy → rotator wiring, controller.go:120-285
+    collapsed to one process). Returns None when nothing rotates."""
+    rotators: list[Rotator] = []
+    seen: set[str] = set()
+    for route in cfg.routes:
+        for b in route.backends:
+            auth = b.auth
+            if auth is None or auth.rotation is None:
+                continue
+            rot_cfg = auth.rotation
+            out = rot_cfg.out_file or auth.api_key_file or auth.aws_credentials_file
+            if not out:
+                raise RotationError(
+                    f"backend {b.name!r}: rotation needs an output file "
+                    "(set auth.apiKeyFile / auth.awsCredentialsFile or "
+                    "rotation.outFile)")
+            if out in seen:
+                continue  # one rotator per credential file
+            seen.add(out)
+            oidc = None
+            if rot_cfg.oidc is not None:
+                o = rot_cfg.oidc
+                oidc = OIDCTokenProvider(OIDCConfig(
+                    issuer=o.issuer, client_id=o.client_id,
+                    client_secret=o.client_secret,
+                    client_secret_file=o.client_secret_file,
+                    audience=o.audience,
+                    scopes=o.scopes or ["openid"],
+                    token_endpoint=o.token_endpoint,
+                ))
+            if rot_cfg.kind == "aws_oidc":
+                if oidc is None:
+                    raise RotationError(f"backend {b.name!r}: aws_oidc needs oidc")
+                rotators.append(AwsOidcRotator(
+                    oidc, role_arn=rot_cfg.aws_role_arn,
+                    region=rot_cfg.aws_region or auth.aws_region or "us-east-1",
+                    out_file=out, sts_endpoint=rot_cfg.sts_endpoint,
+                    pre_rotation_window_s=rot_cfg.pre_rotation_window_s))
+            elif rot_cfg.kind == "azure":
+                rotators.append(AzureTokenRotator(
+                    tenant_id=rot_cfg.azure_tenant_id,
+                    client_id=rot_cfg.azure_client_id,
+                    client_secret=rot_cfg.azure_client_secret,
+                    oidc=oidc, out_file=out,
+                    scope=rot_cfg.azure_scope or
+                    "https://cognitiveservices.azure.com/.default",
+                    authority=rot_cfg.azure_authority or
+                    "https://login.microsoftonline.com",
+                    pre_rotation_window_s=rot_cfg.pre_rotation_window_s))
+            elif rot_cfg.kind == "gcp_oidc":
+                if oidc is None:
+                    raise RotationError(f"backend {b.name!r}: gcp_oidc needs oidc")
+                rotators.append(GcpOidcRotator(
+                    oidc, project_number=rot_cfg.gcp_project_number,
+                    pool_name=rot_cfg.gcp_pool,
+                    provider_name=rot_cfg.gcp_provider,
+                    service_account=rot_cfg.gcp_service_account,
+                    out_file=out,
+                    sts_endpoint=rot_cfg.gcp_sts_endpoint or
+                    "https://sts.googleapis.com/v1/token",
+                    iam_endpoint=rot_cfg.gcp_iam_endpoint or
+                    "https://iamcredentials.googleapis.com",
+                    pre_rotation_window_s=rot_cfg.pre_rotation_window_s))
+    return RotationManager(rotators) if rotators else None
+
+
 class RotationManager:
     """Drives a set of rotators from one asyncio task: rotate whatever is
     (about to be) expired, then sleep until the earliest next deadline.

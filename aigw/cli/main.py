@@ -109,6 +109,16 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
         watcher = ConfigWatcher(watch_path, server.swap_runtime)
         await watcher.start()
 
+    # OIDC/token-exchange credential rotation (BSP rotator analogue):
+    # refreshes the credential files the auth handlers read
+    rotation_mgr = None
+    if not getattr(args, "_is_worker", False):
+        from aigw.backendauth.rotators import build_rotation_manager
+
+        rotation_mgr = build_rotation_manager(cfg)
+        if rotation_mgr is not None:
+            await rotation_mgr.start()
+
     port = args.port + rank
     reuse = getattr(args, "workers", 1) > 1
     if getattr(args, "front", "lean") == "aiohttp":
@@ -156,6 +166,8 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
     finally:
         if watcher:
             await watcher.stop()
+        if rotation_mgr is not None:
+            await rotation_mgr.stop()
         if sync:
             await sync.stop()
         if gpu_host is not None:

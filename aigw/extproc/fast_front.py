@@ -156,9 +156,11 @@ class FastFront:
         await site.start()
         fallback_port = self._fallback_runner.addresses[0][1]
         self.fast.set_fallback("127.0.0.1", fallback_port)
-        if self.gpu_socket:
-            self.fast.enable_gpu(self.gpu_socket, self.gpu_window_us,
-                                 self.gpu_max_batch)
+        sockets = ([self.gpu_socket] if isinstance(self.gpu_socket, str)
+                   else list(self.gpu_socket or []))
+        for s in sockets:
+            if s:
+                self.fast.enable_gpu(s, self.gpu_window_us, self.gpu_max_batch)
         self.port = self.fast.start(host, port)
         logger.info("fast front listening on %s:%d (fallback :%d)", host,
                     self.port, fallback_port)

@@ -91,6 +91,54 @@ class BodyMutation:
 
 
 @dataclass
+class OIDCSource:
+    """OIDC client-credentials source for credential rotation
+    (api/v1beta1 backendsecurity_policy.go BSPOIDC)."""
+
+    issuer: str = ""
+    client_id: str = ""
+    client_secret: str = ""
+    client_secret_file: str = ""
+    audience: str = ""
+    scopes: list[str] = field(default_factory=list)
+    token_endpoint: str = ""  # skip discovery (tests / fixed deployments)
+
+
+@dataclass
+class CredentialRotation:
+    """Background credential rotation for this backend's auth
+    (internal/controller/rotators/): the rotator refreshes the
+    credential FILE the auth handler reads (api_key_file /
+    aws_credentials_file), which is the reference's Secret-mount
+    contract collapsed to one process.
+
+    kind: aws_oidc | azure | gcp_oidc.
+    """
+
+    kind: str = ""
+    out_file: str = ""  # defaults to the auth's credential file
+    pre_rotation_window_s: float = 300.0
+    oidc: Optional[OIDCSource] = None
+    # aws_oidc (rotators/aws_oidc_rotator.go)
+    aws_role_arn: str = ""
+    aws_region: str = ""
+    sts_endpoint: str = ""  # test/override hook
+    # azure (rotators/azure_token_rotator.go)
+    azure_tenant_id: str = ""
+    azure_client_id: str = ""
+    azure_client_secret: str = ""
+    azure_scope: str = ""
+    azure_authority: str = ""
+    # gcp_oidc (rotators/gcp_oidc_token_rotator.go)
+    gcp_project_number: str = ""
+    gcp_pool: str = ""
+    gcp_provider: str = ""
+    gcp_service_account: str = ""
+    gcp_sts_endpoint: str = ""
+    gcp_iam_endpoint: str = ""
+
+
+@dataclass
 class CredentialOverride:
     """Per-request credential sourcing for a backend (filterconfig.go
     CredentialOverride :221-240). Override headers are only honored when
@@ -135,6 +183,8 @@ class BackendAuth:
     aws_credentials_file: str = ""
     # Per-request credential sourcing; None = overrides disabled.
     credential_override: Optional["CredentialOverride"] = None
+    # Background OIDC/token-exchange rotation of the credential file.
+    rotation: Optional["CredentialRotation"] = None
 
     @property
     def kind(self) -> str:
@@ -377,6 +427,15 @@ def _parse_backend_auth(d, ctx) -> BackendAuth:
             **_dc(CredentialOverride, kw["credential_override"],
                   f"{ctx}.credentialOverride")
         )
+    if kw.get("rotation") is not None:
+        rkw = _dc(CredentialRotation, kw["rotation"], f"{ctx}.rotation")
+        if rkw.get("oidc") is not None:
+            rkw["oidc"] = OIDCSource(**_dc(OIDCSource, rkw["oidc"],
+                                           f"{ctx}.rotation.oidc"))
+        rot = CredentialRotation(**rkw)
+        if rot.kind not in ("aws_oidc", "azure", "gcp_oidc"):
+            raise ConfigError(f"{ctx}.rotation: unknown kind {rot.kind!r}")
+        kw["rotation"] = rot
     return BackendAuth(**kw)
 
 

@@ -347,6 +347,27 @@ def _start_fast_front(args, upstream_ports, gpu_socket):
     return front, state["port"]
 
 
+def gpu_host_entry(socket_path, local_rank, window_ms, ready_evt):
+    """One GPU admission-host PROCESS (own GPU context + own GIL): the
+    native front shards count batches across several of these — one host
+    caps near ~900 MB/s of msgpack-decoded request text (measured:
+    36k req/s vs 82k without GPU), so admission decode must scale in
+    processes just like the round-1 HTTP workers did."""
+    torch.cuda.set_device(local_rank)
+    from aigw.gpu import GPUServices
+    from aigw.gpu.service import GPUServiceHost
+
+    async def amain():
+        gpu = GPUServices(device=f"cuda:{local_rank}", n_merges=32768,
+                          enable_cache=False, window_ms=window_ms, max_batch=1024)
+        host = GPUServiceHost(gpu, socket_path)
+        await host.start()
+        ready_evt.set()
+        await asyncio.Event().wait()
+
+    asyncio.run(amain())
+
+
 def run_fast_mode(args, rank, world, local_rank, use_gpu):
     ctx = mp.get_context("spawn")
     cores = os.cpu_count() or 8
@@ -363,11 +384,24 @@ def run_fast_mode(args, rank, world, local_rank, use_gpu):
         p.start()
     ports = [port_q.get(timeout=600) for _ in range(n_up)]
 
-    gpu_socket = None
+    gpu_sockets = []
+    gpu_host_procs = []
     if use_gpu:
-        gpu_socket = f"/tmp/aigw-gpu-{rank}-{os.getpid()}.sock"
-        _start_gpu_host(gpu_socket, local_rank, args.gpu_window)
-    front, gw_port = _start_fast_front(args, ports, gpu_socket)
+        n_hosts = max(args.gpu_hosts, 1)
+        host_ready = []
+        for h in range(n_hosts):
+            sock = f"/tmp/aigw-gpu-{rank}-{os.getpid()}-{h}.sock"
+            ev = ctx.Event()
+            p = ctx.Process(target=gpu_host_entry,
+                            args=(sock, local_rank, args.gpu_window, ev))
+            p.start()
+            gpu_sockets.append(sock)
+            gpu_host_procs.append(p)
+            host_ready.append(ev)
+        for ev in host_ready:
+            if not ev.wait(timeout=600):
+                raise RuntimeError("GPU admission host failed to start")
+    front, gw_port = _start_fast_front(args, ports, gpu_sockets)
 
     ready_evts = [ctx.Event() for _ in range(loadgens)]
     go = ctx.Event()
@@ -386,7 +420,7 @@ def run_fast_mode(args, rank, world, local_rank, use_gpu):
             for p in procs + up_procs:
                 p.terminate()
             raise RuntimeError("load generator failed to become ready")
-    return front, go, out_q, procs, up_procs, loadgens
+    return front, go, out_q, procs, up_procs + gpu_host_procs, loadgens
 
 
 def _start_gpu_host(socket_path: str, local_rank: int, window_ms: float = 0.1):
@@ -445,6 +479,8 @@ def main():
                          "processes (the round-1 configuration)")
     ap.add_argument("--upstreams", type=int, default=0,
                     help="fast mode: mock-upstream processes per rank (0 = 8)")
+    ap.add_argument("--gpu-hosts", type=int, default=4,
+                    help="fast mode: GPU admission host processes per rank")
     ap.add_argument("--gpu-window", type=float, default=0.1,
                     help="GPU micro-batch window per worker, ms")
     ap.add_argument("--gpu-service", action="store_true",

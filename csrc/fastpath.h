@@ -98,6 +98,8 @@ class FastServer {
   void add_route(FastRoute r);
   void add_rate_rule(const RateRule& r);
   void set_fallback(const std::string& host, uint16_t port);
+  // may be called multiple times: each socket is one admission-host
+  // process (one GPU context each); batches round-robin across them
   void enable_gpu(const std::string& socket_path, int window_us, int max_batch);
 
   // lifecycle
@@ -136,7 +138,9 @@ class FastServer {
   std::vector<std::unique_ptr<RuleState>> rules_;
   std::string fallback_host_;
   uint16_t fallback_port_ = 0;
-  std::unique_ptr<GpuAdmissionClient> gpu_;
+  std::vector<std::unique_ptr<GpuAdmissionClient>> gpu_;
+  std::atomic<uint64_t> gpu_rr_{0};
+  int64_t gpu_count(const std::string& text);
   std::unique_ptr<UpstreamPool> pool_;
   ServerStats stats_;
 
