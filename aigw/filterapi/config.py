@@ -326,6 +326,32 @@ class MCPOAuth:
 
 
 @dataclass
+class MCPAuthorizationRule:
+    """One authorization rule (mcpconfig.go MCPRouteAuthorizationRule
+    :126-170): action + optional CEL condition + optional tool target +
+    optional JWT source (scopes: ALL required; claims: each must exist
+    with one of the allowed values, dotted paths supported)."""
+
+    action: str = "Allow"  # Allow | Deny
+    cel: str = ""
+    tools: list[str] = field(default_factory=list)  # backend__tool names
+    jwt_scopes: list[str] = field(default_factory=list)
+    jwt_claims: list[dict] = field(default_factory=list)  # {name, values}
+
+
+@dataclass
+class MCPAuthorization:
+    """Route-level authorization (mcpconfig.go MCPRouteAuthorization
+    :101-115): first matching rule wins, else defaultAction; scope-only
+    denials answer 403 with an insufficient_scope WWW-Authenticate
+    challenge pointing at resourceMetadataURL."""
+
+    default_action: str = "Deny"  # Allow | Deny
+    rules: list[MCPAuthorizationRule] = field(default_factory=list)
+    resource_metadata_url: str = ""
+
+
+@dataclass
 class MCPRoute:
     name: str
     path: str = "/mcp"
@@ -337,6 +363,9 @@ class MCPRoute:
     # resource_metadata_url is set
     bearer_token: str = ""
     resource_metadata_url: str = ""
+    # fine-grained CEL/scope/claim rules evaluated per request after the
+    # authentication gate (authorization.go)
+    authorization: Optional[MCPAuthorization] = None
 
 
 @dataclass
@@ -526,6 +555,21 @@ def load_config(data: object) -> Config:
                 rkw["oauth"] = MCPOAuth(
                     **_dc(MCPOAuth, rkw["oauth"], f"mcp.routes[{i}].oauth")
                 )
+            if rkw.get("authorization") is not None:
+                akw = _dc(MCPAuthorization, rkw["authorization"],
+                          f"mcp.routes[{i}].authorization")
+                akw["rules"] = [
+                    MCPAuthorizationRule(**_dc(
+                        MCPAuthorizationRule, r,
+                        f"mcp.routes[{i}].authorization.rules[{j}]"))
+                    for j, r in enumerate(akw.get("rules", []))
+                ]
+                auth_cfg = MCPAuthorization(**akw)
+                if auth_cfg.default_action not in ("Allow", "Deny"):
+                    raise ConfigError(
+                        f"mcp.routes[{i}].authorization: defaultAction must "
+                        "be Allow or Deny")
+                rkw["authorization"] = auth_cfg
             routes.append(MCPRoute(**rkw))
         mkw["routes"] = routes
         kw["mcp"] = MCPConfig(**mkw)

@@ -71,6 +71,12 @@ class MCPProxy:
         self._tool_res: dict[str, list] = {}
         for b in route.backends:
             self._tool_res[b.name] = [re.compile(p) for p in b.tool_exclude + b.tool_include]
+        self._authz = None
+        if getattr(route, "authorization", None) is not None:
+            from aigw.mcp.authorization import CompiledAuthorization
+
+            # compiles CEL rules at startup: bad config fails loudly
+            self._authz = CompiledAuthorization(route.authorization)
 
     async def _client(self) -> aiohttp.ClientSession:
         if self._session is None:
@@ -189,6 +195,52 @@ class MCPProxy:
             )
         return resp
 
+    def _authorize_rules(self, request: web.Request, method: str, id_,
+                         payload: dict):
+        """Per-request CEL/scope/claim authorization after the
+        authentication gate (authorization.go:755-783): deny answers 403
+        with an insufficient_scope WWW-Authenticate challenge when the
+        denial was scope-only."""
+        from aigw.mcp.authorization import (
+            build_activation,
+            build_insufficient_scope_header,
+            parse_unverified_claims,
+        )
+
+        params = payload.get("params") or {}
+        backend_name = ""
+        tool = ""
+        if method in ("tools/call", "prompts/get"):
+            prefixed = str(params.get("name", ""))
+            be, stripped = self._backend_for(prefixed)
+            if be is not None:
+                backend_name, tool = be.name, stripped
+            else:
+                tool = prefixed
+        claims = parse_unverified_claims(request.headers.get("authorization", ""))
+        activation = build_activation(
+            http_method=request.method,
+            host=request.headers.get("host", ""),
+            path=request.path,
+            headers=dict(request.headers),
+            mcp_method=method,
+            backend=backend_name,
+            tool=tool,
+            params=params,
+            claims=claims,
+        )
+        decision = self._authz.authorize(
+            activation=activation, claims=claims, backend=backend_name, tool=tool
+        )
+        if decision.allowed:
+            return None
+        resp = _rpc_error(id_, -32001, "access denied", status=403)
+        if decision.required_scopes:
+            resp.headers["www-authenticate"] = build_insufficient_scope_header(
+                decision.required_scopes, self._authz.resource_metadata_url
+            )
+        return resp
+
     async def handle(self, request: web.Request) -> web.StreamResponse:
         import time as _time
 
@@ -235,6 +287,10 @@ class MCPProxy:
         method = payload.get("method", "")
         request[_METHOD_KEY] = method
         id_ = payload.get("id")
+        if self._authz is not None:
+            denied = self._authorize_rules(request, method, id_, payload)
+            if denied is not None:
+                return denied
         token = request.headers.get(internalapi.MCP_SESSION_ID_HEADER, "")
         sessions: dict[str, str] = {}
         if token:
