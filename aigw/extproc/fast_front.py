@@ -33,6 +33,7 @@ from aigw.filterapi.config import APISchemaName, LLMRequestCostType
 from aigw.filterapi.runtime import RuntimeConfig
 
 try:
+    import torch  # noqa: F401 - loads libc10/libamdhip64 for the extension
     import aigw_fast as _fast
 except ImportError:  # pragma: no cover - built by setup.py everywhere
     _fast = None
@@ -137,7 +138,9 @@ class FastFront:
     metrics bridge exposing the native counters on the Python /metrics."""
 
     def __init__(self, server, runtime: RuntimeConfig, *, gpu_socket: str = "",
-                 gpu_window_us: int = 100, gpu_max_batch: int = 256):
+                 gpu_window_us: int = 100, gpu_max_batch: int = 256,
+                 gpu_direct: bool = False, n_merges: int = 32768,
+                 tokenizer_seed: int = 1355):
         # `server` is the Python GatewayServer used for cold paths
         self.py_server = server
         self.runtime = runtime
@@ -145,6 +148,9 @@ class FastFront:
         self.gpu_socket = gpu_socket
         self.gpu_window_us = gpu_window_us
         self.gpu_max_batch = gpu_max_batch
+        self.gpu_direct = gpu_direct
+        self.n_merges = n_merges
+        self.tokenizer_seed = tokenizer_seed
         self._fallback_runner = None
         self.port = None
 
@@ -156,6 +162,17 @@ class FastFront:
         await site.start()
         fallback_port = self._fallback_runner.addresses[0][1]
         self.fast.set_fallback("127.0.0.1", fallback_port)
+        if self.gpu_direct:
+            # in-process HIP admission: the native server owns the BPE
+            # kernels on its own stream; the merge table comes from the
+            # same generator as the Python/CPU oracle, so counts are
+            # bit-identical (tests/test_bpe_ref.py)
+            from aigw.ops.bpe_ref import build_hash_table, make_merges
+
+            keys, ranks = build_hash_table(make_merges(self.n_merges,
+                                                       self.tokenizer_seed))
+            self.fast.enable_gpu_direct(keys, ranks,
+                                        max_batch=self.gpu_max_batch)
         sockets = ([self.gpu_socket] if isinstance(self.gpu_socket, str)
                    else list(self.gpu_socket or []))
         for s in sockets:

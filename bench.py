@@ -302,7 +302,7 @@ def loadgen_entry(args, gw_port, direct_port, ready, go, out_q):
     asyncio.run(amain())
 
 
-def _start_fast_front(args, upstream_ports, gpu_socket):
+def _start_fast_front(args, upstream_ports, gpu_socket, gpu_direct=False):
     """Run the C++ gateway (and its Python fallback app) on a dedicated
     thread/loop in the rank primary; returns (front, port)."""
     import threading
@@ -327,7 +327,8 @@ def _start_fast_front(args, upstream_ports, gpu_socket):
     server = GatewayServer(RuntimeConfig(cfg))
     front = FastFront(
         server, server.runtime, gpu_socket=gpu_socket or "",
-        gpu_window_us=int(args.gpu_window * 1000.0), gpu_max_batch=256,
+        gpu_window_us=int(args.gpu_window * 1000.0), gpu_max_batch=1024,
+        gpu_direct=gpu_direct,
     )
     done = threading.Event()
     state = {}
@@ -386,8 +387,16 @@ def run_fast_mode(args, rank, world, local_rank, use_gpu):
 
     gpu_sockets = []
     gpu_host_procs = []
-    if use_gpu:
-        n_hosts = max(args.gpu_hosts, 1)
+    gpu_direct = False
+    if use_gpu and args.gpu_hosts <= 0:
+        # default: in-process HIP admission — the native server launches
+        # the BPE kernels itself (no IPC, no extra GPU contexts; the UDS
+        # host topology measured 36k single-host / 10k 4-host vs 82k
+        # without GPU work)
+        gpu_direct = True
+        torch.cuda.set_device(local_rank)
+    elif use_gpu:
+        n_hosts = args.gpu_hosts
         host_ready = []
         for h in range(n_hosts):
             sock = f"/tmp/aigw-gpu-{rank}-{os.getpid()}-{h}.sock"
@@ -401,7 +410,7 @@ def run_fast_mode(args, rank, world, local_rank, use_gpu):
         for ev in host_ready:
             if not ev.wait(timeout=600):
                 raise RuntimeError("GPU admission host failed to start")
-    front, gw_port = _start_fast_front(args, ports, gpu_sockets)
+    front, gw_port = _start_fast_front(args, ports, gpu_sockets, gpu_direct)
 
     ready_evts = [ctx.Event() for _ in range(loadgens)]
     go = ctx.Event()
@@ -479,8 +488,9 @@ def main():
                          "processes (the round-1 configuration)")
     ap.add_argument("--upstreams", type=int, default=0,
                     help="fast mode: mock-upstream processes per rank (0 = 8)")
-    ap.add_argument("--gpu-hosts", type=int, default=4,
-                    help="fast mode: GPU admission host processes per rank")
+    ap.add_argument("--gpu-hosts", type=int, default=0,
+                    help="fast mode: 0 = in-process HIP admission (default); "
+                         "N>0 = N UDS admission-host processes per rank")
     ap.add_argument("--gpu-window", type=float, default=0.1,
                     help="GPU micro-batch window per worker, ms")
     ap.add_argument("--gpu-service", action="store_true",

@@ -590,7 +590,131 @@ class GpuAdmissionClient {
   std::thread batcher_, reader_;
 };
 
+
+// -------- in-process HIP admission batcher ----------------------------------
+
+// admission.hip entry points (same .so, separate HIP translation unit)
+GpuAdmissionDirect* admission_create(const long long* htab_keys,
+                                     const int32_t* htab_rank, int htab_n,
+                                     size_t max_bytes, int max_req);
+bool admission_count(GpuAdmissionDirect* a, const char* bytes, size_t n,
+                     const int64_t* offsets, int n_req, int32_t* counts_out);
+void admission_destroy(GpuAdmissionDirect* a);
+
+// Adaptive batching with NO timer window: one batcher thread drains
+// whatever accumulated while the previous GPU batch ran — the kernel
+// duration itself is the coalescing window, so light load gets ~zero
+// added latency and heavy load gets large batches automatically.
+class DirectGpuBatcher {
+ public:
+  bool start(const long long* htab_keys, const int32_t* htab_rank, int htab_n,
+             int max_batch, size_t max_bytes, int max_req) {
+    adm_ = admission_create(htab_keys, htab_rank, htab_n, max_bytes, max_req);
+    if (adm_ == nullptr) return false;
+    max_batch_ = std::min(max_batch, max_req);
+    max_bytes_ = max_bytes;
+    worker_ = std::thread([this] { loop(); });
+    return true;
+  }
+
+  void stop() {
+    stopping_ = true;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      cv_.notify_all();
+    }
+    if (worker_.joinable()) worker_.join();
+    if (adm_ != nullptr) admission_destroy(adm_), adm_ = nullptr;
+    fail_all();
+  }
+
+  ~DirectGpuBatcher() { stop(); }
+
+  int64_t count_text(const std::string& text) {
+    if (stopping_) return 0;
+    auto w = std::make_shared<GpuAdmissionClient::Waiter>();
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      q_texts_.push_back(text);
+      q_waiters_.push_back(w);
+      cv_.notify_one();
+    }
+    std::unique_lock<std::mutex> lk(w->m);
+    w->cv.wait_for(lk, std::chrono::seconds(30), [&] { return w->done; });
+    return w->count;
+  }
+
+ private:
+  void loop() {
+    std::string packed;
+    std::vector<int64_t> offs;
+    std::vector<int32_t> counts;
+    while (!stopping_) {
+      std::vector<std::string> texts;
+      std::vector<std::shared_ptr<GpuAdmissionClient::Waiter>> waiters;
+      {
+        std::unique_lock<std::mutex> lk(mu_);
+        cv_.wait(lk, [&] { return stopping_ || !q_texts_.empty(); });
+        if (stopping_) return;
+        size_t take = 0, bytes = 0;
+        while (take < q_texts_.size() && (int)take < max_batch_ &&
+               bytes + q_texts_[take].size() <= max_bytes_)
+          bytes += q_texts_[take++].size();
+        if (take == 0) take = 1;  // oversized single text: truncated below
+        texts.assign(q_texts_.begin(), q_texts_.begin() + take);
+        waiters.assign(q_waiters_.begin(), q_waiters_.begin() + take);
+        q_texts_.erase(q_texts_.begin(), q_texts_.begin() + take);
+        q_waiters_.erase(q_waiters_.begin(), q_waiters_.begin() + take);
+      }
+      packed.clear();
+      offs.clear();
+      for (auto& t : texts) {
+        offs.push_back((int64_t)packed.size());
+        size_t room = max_bytes_ - packed.size();
+        packed.append(t.data(), std::min(t.size(), room));
+      }
+      counts.assign(texts.size(), 0);
+      if (!packed.empty())
+        admission_count(adm_, packed.data(), packed.size(), offs.data(),
+                        (int)texts.size(), counts.data());
+      for (size_t i = 0; i < waiters.size(); ++i) {
+        auto& w = waiters[i];
+        std::lock_guard<std::mutex> lk(w->m);
+        w->count = counts[i];
+        w->done = true;
+        w->cv.notify_all();
+      }
+    }
+  }
+
+  void fail_all() {
+    std::vector<std::shared_ptr<GpuAdmissionClient::Waiter>> all;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      all = std::move(q_waiters_);
+      q_waiters_.clear();
+      q_texts_.clear();
+    }
+    for (auto& w : all) {
+      std::lock_guard<std::mutex> lk(w->m);
+      w->done = true;
+      w->cv.notify_all();
+    }
+  }
+
+  GpuAdmissionDirect* adm_ = nullptr;
+  int max_batch_ = 1024;
+  size_t max_bytes_ = 0;
+  std::atomic<bool> stopping_{false};
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::vector<std::string> q_texts_;
+  std::vector<std::shared_ptr<GpuAdmissionClient::Waiter>> q_waiters_;
+  std::thread worker_;
+};
+
 // -------- connection handler ------------------------------------------------
+
 
 namespace {
 
@@ -728,7 +852,7 @@ class ConnHandler {
     }
 
     int64_t gpu_tokens = 0;
-    if (!srv_->gpu_.empty() && req.path == "/v1/chat/completions" &&
+    if (srv_->gpu_enabled() && req.path == "/v1/chat/completions" &&
         !sc.text.empty()) {
       gpu_tokens = srv_->gpu_count(sc.text);
       srv_->stats_.gpu_tokens += (uint64_t)gpu_tokens;
@@ -1249,12 +1373,28 @@ void FastServer::enable_gpu(const std::string& socket_path, int window_us,
   gpu_.push_back(std::move(client));
 }
 
+bool FastServer::gpu_enabled() const {
+  return gpu_direct_ != nullptr || !gpu_.empty();
+}
+
 int64_t FastServer::gpu_count(const std::string& text) {
-  // round-robin across admission hosts: each is one Python process with
-  // its own GPU context; sharding removes the single-host msgpack-decode
-  // ceiling (~900 MB/s of request text per host, measured)
+  // in-process HIP admission wins when available; the UDS hosts remain
+  // for deployments where the gateway process must not own the device
+  if (gpu_direct_ != nullptr) return gpu_direct_->count_text(text);
   size_t i = (size_t)(gpu_rr_.fetch_add(1, std::memory_order_relaxed) % gpu_.size());
   return gpu_[i]->count_text(text);
+}
+
+void FastServer::enable_gpu_direct(const long long* htab_keys,
+                                   const int32_t* htab_rank, int htab_n,
+                                   int max_batch, size_t max_batch_bytes,
+                                   int max_req) {
+  auto b = std::make_unique<DirectGpuBatcher>();
+  if (!b->start(htab_keys, htab_rank, htab_n, max_batch, max_batch_bytes,
+                max_req))
+    throw std::runtime_error(
+        "direct GPU admission init failed (no device visible?)");
+  gpu_direct_ = std::move(b);
 }
 
 int FastServer::start(const std::string& host, uint16_t port) {
@@ -1334,6 +1474,7 @@ void FastServer::stop() {
   }
   for (int i = 0; i < 200 && stats_.active_connections.load() > 0; ++i)
     std::this_thread::sleep_for(std::chrono::milliseconds(10));
+  if (gpu_direct_) gpu_direct_->stop();
   for (auto& g : gpu_) g->stop();
   pool_->close_all();
 }

@@ -87,6 +87,8 @@ struct ServerStats {
 };
 
 class GpuAdmissionClient;
+class GpuAdmissionDirect;  // admission.hip — in-process HIP BPE counting
+class DirectGpuBatcher;
 class UpstreamPool;
 
 class FastServer {
@@ -101,6 +103,12 @@ class FastServer {
   // may be called multiple times: each socket is one admission-host
   // process (one GPU context each); batches round-robin across them
   void enable_gpu(const std::string& socket_path, int window_us, int max_batch);
+  // in-process HIP admission (preferred): the gateway owns one stream
+  // and launches the BPE kernels itself — no IPC, no extra GPU context.
+  // Throws when no GPU is visible or init fails.
+  void enable_gpu_direct(const long long* htab_keys, const int32_t* htab_rank,
+                         int htab_n, int max_batch, size_t max_batch_bytes,
+                         int max_req);
 
   // lifecycle
   int start(const std::string& host, uint16_t port);  // returns bound port
@@ -139,8 +147,10 @@ class FastServer {
   std::string fallback_host_;
   uint16_t fallback_port_ = 0;
   std::vector<std::unique_ptr<GpuAdmissionClient>> gpu_;
+  std::unique_ptr<DirectGpuBatcher> gpu_direct_;
   std::atomic<uint64_t> gpu_rr_{0};
   int64_t gpu_count(const std::string& text);
+  bool gpu_enabled() const;
   std::unique_ptr<UpstreamPool> pool_;
   ServerStats stats_;
 

@@ -34,13 +34,20 @@ setup(
             extra_compile_args=["-O3", "-std=c++17"],
             language="c++",
         ),
-        Extension(
+        # CUDAExtension so admission.hip (the in-process HIP BPE counting
+        # path) compiles with hipcc for gfx950 alongside the C++ server.
+        CUDAExtension(
             name="aigw_fast",
-            sources=["csrc/aigw_fast_module.cpp", "csrc/fastpath.cpp"],
-            include_dirs=[pybind11.get_include()],
-            extra_compile_args=["-O3", "-std=c++17"],
+            sources=[
+                "csrc/aigw_fast_module.cpp",
+                "csrc/fastpath.cpp",
+                "csrc/admission.hip",
+            ],
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
             extra_link_args=["-lpthread"],
-            language="c++",
         ),
     ],
     cmdclass={"build_ext": BuildExtension},
