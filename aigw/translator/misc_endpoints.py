@@ -199,6 +199,112 @@ class AnthropicTokenize(Translator):
         )
 
 
+def _tokenize_chat_body(body: dict, model: str) -> dict:
+    """Normalize a tokenize request to chat form: {messages} stays as-is,
+    a completion-style {prompt}/{text} becomes one user message
+    (tokenize_awsanthropic.go:84-101 ChatRequest synthesis)."""
+    if isinstance(body.get("messages"), list):
+        return {"model": model, "messages": body["messages"],
+                **({"tools": body["tools"]} if body.get("tools") else {})}
+    text = body.get("prompt") or body.get("text") or ""
+    return {"model": model, "messages": [{"role": "user", "content": text}]}
+
+
+def _aws_count_tokens_path(model: str) -> str:
+    """Bedrock CountTokens path; cross-region inference prefixes (e.g.
+    "us.anthropic.claude-...") are stripped because CountTokens rejects
+    CRIS ids (anthropic_helper.go awsAnthropicCountTokensPath:1385-1400)."""
+    bare = re.sub(r"^[a-z]{2}\.", "", model)
+    return f"/model/{bare}/count-tokens"
+
+
+@register("/tokenize", APISchemaName.AWS_BEDROCK)
+class AWSBedrockTokenize(Translator):
+    """/tokenize → Bedrock CountTokens in Converse form
+    (tokenize_awsbedrock.go: {"input":{"converse":{...}}} and the
+    inputTokens response field)."""
+
+    def __init__(self, api_version: str = "", gcp_project: str = "", gcp_region: str = ""):
+        self._model = ""
+
+    def request(self, body, *, model_override="", stream=False,
+                force_include_usage=False, raw=b""):
+        from aigw.translator.chat_bedrock import openai_to_converse_request
+
+        if model_override:
+            body["model"] = model_override
+        self._model = body.get("model", "")
+        chat = _tokenize_chat_body(body, self._model)
+        converse = openai_to_converse_request(chat)
+        inner = {"messages": converse.get("messages", [])}
+        if converse.get("system"):
+            inner["system"] = converse["system"]
+        if converse.get("toolConfig"):
+            inner["toolConfig"] = converse["toolConfig"]
+        return RequestTranslation(
+            path=_aws_count_tokens_path(self._model),
+            body=jdump({"input": {"converse": inner}}),
+        )
+
+    def response_body(self, status, body):
+        try:
+            n = json.loads(body).get("inputTokens", 0) or 0
+        except ValueError:
+            n = 0
+        return ResponseTranslation(
+            body=jdump({"count": n, "tokens": []}),
+            usage=Usage(input_tokens=n, total_tokens=n),
+            response_model=self._model,
+            end_of_stream=True,
+        )
+
+
+@register("/tokenize", APISchemaName.AWS_ANTHROPIC)
+class AWSAnthropicTokenize(Translator):
+    """/tokenize → Bedrock CountTokens in InvokeModel form: the Anthropic
+    count body rides base64-inside-JSON
+    (tokenize_awsanthropic.go: {"input":{"invokeModel":{"body":"<b64>"}}};
+    Bedrock validates the inner body as a real request, so it carries
+    anthropic_version and a default max_tokens)."""
+
+    def __init__(self, api_version: str = "", gcp_project: str = "", gcp_region: str = ""):
+        self.api_version = api_version or "bedrock-2023-05-31"
+        self._model = ""
+
+    def request(self, body, *, model_override="", stream=False,
+                force_include_usage=False, raw=b""):
+        import base64
+
+        from aigw.translator.anthropic_schema import openai_to_anthropic_request
+
+        if model_override:
+            body["model"] = model_override
+        self._model = body.get("model", "")
+        chat = _tokenize_chat_body(body, self._model)
+        areq = openai_to_anthropic_request(chat)
+        areq.pop("model", None)  # model travels in the URL path
+        areq["anthropic_version"] = self.api_version
+        areq.setdefault("max_tokens", 1)
+        b64 = base64.b64encode(jdump(areq)).decode()
+        return RequestTranslation(
+            path=_aws_count_tokens_path(self._model),
+            body=jdump({"input": {"invokeModel": {"body": b64}}}),
+        )
+
+    def response_body(self, status, body):
+        try:
+            doc = json.loads(body)
+            n = doc.get("inputTokens", doc.get("input_tokens", 0)) or 0
+        except ValueError:
+            n = 0
+        return ResponseTranslation(
+            body=jdump({"count": n, "tokens": []}),
+            usage=Usage(input_tokens=n, total_tokens=n),
+            response_model=self._model,
+            end_of_stream=True,
+        )
+
+
 # --- multipart audio endpoints -------------------------------------------------
 
 

@@ -66,7 +66,7 @@ llmRequestCosts:
                                      "content": t.decode("latin1")}]}
             # chat text extraction appends one newline per collected value
             chat_text = t + b"\n"
-            expected_total += len(ref.encode(chat_text))
+            expected_total += len(ref.encode_batch([chat_text])[0])
             reqs.append(json.dumps(payload).encode())
 
         async def one(body):
