@@ -375,7 +375,7 @@ def run_fast_mode(args, rank, world, local_rank, use_gpu):
     # the C++ gateway saturates well past what one loadgen generates:
     # size the harness (not the gateway) to the machine — the MI355X node
     # pairs 256 CUs with 128+ EPYC cores; CI containers are much smaller
-    n_up = args.upstreams if args.upstreams > 0 else max(1, min(8, cores // 12))
+    n_up = args.upstreams if args.upstreams > 0 else max(1, min(20, cores // 10))
     loadgens = args.workers if args.workers > 0 else max(2, min(24, cores // 5))
 
     port_q = ctx.Queue()
@@ -640,6 +640,11 @@ def main():
                 "p50_direct_ms": round(p50_direct, 3),
                 "p50_added_latency_ms": round(p50 - p50_direct, 3),
                 "gpu_token_accounting": use_gpu,
+                "gpu_admission": (
+                    dict(zip(("batches", "texts", "time_us", "max_us", "errors"),
+                             front.fast.gpu_direct_stats()))
+                    if fast_mode and front is not None else None
+                ),
                 "statesync_tick_us": round(statesync_tick_us, 1) if statesync_tick_us else None,
                 "semantic_cache_payload_pool": args.cache_payloads or None,
             },

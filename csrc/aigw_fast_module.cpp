@@ -84,6 +84,7 @@ PYBIND11_MODULE(aigw_fast, m) {
       .def("start", &FastServer::start, py::arg("host"), py::arg("port"),
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &FastServer::stop, py::call_guard<py::gil_scoped_release>())
+      .def("gpu_direct_stats", &FastServer::gpu_direct_stats)
       .def("rl_collect_deltas", &FastServer::rl_collect_deltas)
       .def("rl_apply_remote", &FastServer::rl_apply_remote)
       .def("rl_local_spent", &FastServer::rl_local_spent)

@@ -674,9 +674,18 @@ class DirectGpuBatcher {
         packed.append(t.data(), std::min(t.size(), room));
       }
       counts.assign(texts.size(), 0);
-      if (!packed.empty())
-        admission_count(adm_, packed.data(), packed.size(), offs.data(),
-                        (int)texts.size(), counts.data());
+      int64_t bt0 = now_us();
+      if (!packed.empty()) {
+        if (!admission_count(adm_, packed.data(), packed.size(), offs.data(),
+                             (int)texts.size(), counts.data()))
+          stats_errors++;
+      }
+      int64_t bt = now_us() - bt0;
+      stats_batches++;
+      stats_texts += (uint64_t)texts.size();
+      stats_time_us += (uint64_t)bt;
+      uint64_t prev = stats_max_us.load();
+      while ((uint64_t)bt > prev && !stats_max_us.compare_exchange_weak(prev, (uint64_t)bt)) {}
       for (size_t i = 0; i < waiters.size(); ++i) {
         auto& w = waiters[i];
         std::lock_guard<std::mutex> lk(w->m);
@@ -703,6 +712,15 @@ class DirectGpuBatcher {
   }
 
   GpuAdmissionDirect* adm_ = nullptr;
+
+ public:
+  std::atomic<uint64_t> stats_batches{0};
+  std::atomic<uint64_t> stats_texts{0};
+  std::atomic<uint64_t> stats_time_us{0};
+  std::atomic<uint64_t> stats_max_us{0};
+  std::atomic<uint64_t> stats_errors{0};
+
+ private:
   int max_batch_ = 1024;
   size_t max_bytes_ = 0;
   std::atomic<bool> stopping_{false};
@@ -1371,6 +1389,13 @@ void FastServer::enable_gpu(const std::string& socket_path, int window_us,
     throw std::runtime_error("cannot connect GPU admission service at " + socket_path);
   }
   gpu_.push_back(std::move(client));
+}
+
+std::vector<uint64_t> FastServer::gpu_direct_stats() const {
+  if (gpu_direct_ == nullptr) return {0, 0, 0, 0, 0};
+  return {gpu_direct_->stats_batches.load(), gpu_direct_->stats_texts.load(),
+          gpu_direct_->stats_time_us.load(), gpu_direct_->stats_max_us.load(),
+          gpu_direct_->stats_errors.load()};
 }
 
 bool FastServer::gpu_enabled() const {

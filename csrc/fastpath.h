@@ -115,6 +115,9 @@ class FastServer {
   void stop();
 
   const ServerStats& stats() const { return stats_; }
+  // direct-admission batcher counters: {batches, texts, time_us, max_us,
+  // errors}; zeros when direct admission is off
+  std::vector<uint64_t> gpu_direct_stats() const;
   // cross-shard rate-limit sync hooks (aigw.parallel.StateSync bridge)
   std::vector<int64_t> rl_collect_deltas();
   void rl_apply_remote(const std::vector<int64_t>& others_spend);
