@@ -44,6 +44,12 @@ class LeanResponse:
         enc = headers.get("content-encoding", "").lower()
         if enc in ("gzip", "deflate"):
             self._decomp = zlib.decompressobj(wbits=47 if enc == "gzip" else 15)
+        elif enc == "br":
+            # streamed brotli decode via the system libbrotlidec
+            # (extproc/util.go:57 parity)
+            from aigw.utils.brotli_dec import BrotliDecompressor
+
+            self._decomp = BrotliDecompressor()
         clen = headers.get("content-length")
         self._remaining = int(clen) if clen is not None else None
         self._chunked = headers.get("transfer-encoding", "").lower() == "chunked"

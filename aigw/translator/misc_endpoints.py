@@ -284,7 +284,7 @@ class AWSAnthropicTokenize(Translator):
         areq = openai_to_anthropic_request(chat)
         areq.pop("model", None)  # model travels in the URL path
         areq["anthropic_version"] = self.api_version
-        areq.setdefault("max_tokens", 1)
+        areq["max_tokens"] = 1  # unconditional, tokenize_awsanthropic.go:69-73
         b64 = base64.b64encode(jdump(areq)).decode()
         return RequestTranslation(
             path=_aws_count_tokens_path(self._model),

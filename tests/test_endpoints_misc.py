@@ -170,3 +170,57 @@ def test_misc_endpoints():
         await runner.cleanup()
 
     asyncio.run(main())
+
+
+def test_tokenize_aws_translators():
+    """/tokenize -> AWS Bedrock CountTokens (Converse form) and AWS
+    Anthropic CountTokens (base64 InvokeModel form) — parity:
+    tokenize_awsbedrock.go / tokenize_awsanthropic.go, incl. cross-region
+    model-id stripping and the inputTokens response mapping."""
+    import base64
+
+    from aigw.filterapi.config import APISchemaName
+    from aigw.translator import get_translator
+
+    # Bedrock Converse form, chat-shaped request
+    tr = get_translator("/tokenize", APISchemaName.AWS_BEDROCK)
+    res = tr.request({
+        "model": "us.amazon.nova-pro-v1",
+        "messages": [{"role": "system", "content": "be terse"},
+                     {"role": "user", "content": "hello"}],
+    }, model_override="", stream=False, force_include_usage=False)
+    assert res.path == "/model/amazon.nova-pro-v1/count-tokens"  # CRIS stripped
+    doc = json.loads(res.body)
+    conv = doc["input"]["converse"]
+    assert conv["messages"][0]["role"] == "user"
+    assert conv["messages"][0]["content"][0]["text"] == "hello"
+    assert conv["system"][0]["text"] == "be terse"
+    out = tr.response_body(200, b'{"inputTokens": 17}')
+    assert json.loads(out.body) == {"count": 17, "tokens": []}
+    assert out.usage.input_tokens == 17
+
+    # completion-shaped request becomes a single user message
+    tr = get_translator("/tokenize", APISchemaName.AWS_BEDROCK)
+    res = tr.request({"model": "m", "prompt": "count me"},
+                     model_override="", stream=False, force_include_usage=False)
+    conv = json.loads(res.body)["input"]["converse"]
+    assert conv["messages"] == [{"role": "user",
+                                 "content": [{"text": "count me"}]}]
+
+    # AWS Anthropic InvokeModel form: inner body is base64 Anthropic JSON
+    # with anthropic_version + max_tokens and NO model field
+    tr = get_translator("/tokenize", APISchemaName.AWS_ANTHROPIC)
+    res = tr.request({
+        "model": "us.anthropic.claude-sonnet-4",
+        "messages": [{"role": "user", "content": "hi"}],
+    }, model_override="", stream=False, force_include_usage=False)
+    assert res.path == "/model/anthropic.claude-sonnet-4/count-tokens"
+    inner = json.loads(base64.b64decode(
+        json.loads(res.body)["input"]["invokeModel"]["body"]))
+    assert inner["anthropic_version"] == "bedrock-2023-05-31"
+    assert inner["max_tokens"] == 1
+    assert "model" not in inner
+    assert inner["messages"][0]["content"] == "hi" or \
+        inner["messages"][0]["content"][0].get("text") == "hi"
+    out = tr.response_body(200, b'{"inputTokens": 9}')
+    assert json.loads(out.body) == {"count": 9, "tokens": []}

@@ -217,3 +217,75 @@ def test_bodyless_204_304_keepalive_not_close_delimited():
             await srv.stop()
 
     run(main())
+
+
+def _brotli_compress(data: bytes) -> bytes:
+    """One-shot brotli encode via the system libbrotlienc (test-only)."""
+    import ctypes
+    import ctypes.util
+
+    lib = ctypes.CDLL(ctypes.util.find_library("brotlienc") or "libbrotlienc.so.1")
+    lib.BrotliEncoderCompress.restype = ctypes.c_int
+    out_cap = ctypes.c_size_t(len(data) + 1024)
+    out_buf = (ctypes.c_uint8 * out_cap.value)()
+    ok = lib.BrotliEncoderCompress(
+        5, 22, 0, ctypes.c_size_t(len(data)),
+        (ctypes.c_uint8 * len(data)).from_buffer_copy(data),
+        ctypes.byref(out_cap), out_buf,
+    )
+    assert ok == 1
+    return bytes(out_buf[: out_cap.value])
+
+
+def test_brotli_response_decompression():
+    """content-encoding: br bodies decode incrementally like gzip
+    (extproc/util.go:57 parity), including chunked SSE split mid-stream."""
+    from aigw.utils.brotli_dec import BrotliDecompressor, available
+
+    assert available(), "libbrotlidec must ship with the image"
+    payload = b'{"usage": {"total_tokens": 42}, "pad": "' + b"x" * 5000 + b'"}'
+    br = _brotli_compress(payload)
+
+    # unit: incremental feed, byte at a time
+    d = BrotliDecompressor()
+    out = b"".join(d.decompress(br[i : i + 1]) for i in range(len(br)))
+    assert out + d.flush() == payload and d.eof
+
+    async def main():
+        resp = (b"HTTP/1.1 200 OK\r\ncontent-encoding: br\r\n"
+                b"content-length: %d\r\n\r\n" % len(br)) + br
+        srv = ScriptedServer([(resp, False)])
+        port = await srv.start()
+        c = LeanClient()
+        r = await c.post(host="127.0.0.1", port=port, tls=False, path="/x",
+                         headers={}, body=b"{}")
+        assert await r.read() == payload
+        r.release()
+        await c.close()
+        await srv.stop()
+
+        # chunked SSE stream, compressed, split across chunk boundaries
+        sse = b"".join(b"data: {\"n\": %d}\n\n" % i for i in range(50))
+        sbr = _brotli_compress(sse)
+        mid = len(sbr) // 3
+        chunks = [sbr[:mid], sbr[mid : 2 * mid], sbr[2 * mid :]]
+        blob = (b"HTTP/1.1 200 OK\r\ncontent-encoding: br\r\n"
+                b"content-type: text/event-stream\r\n"
+                b"transfer-encoding: chunked\r\n\r\n")
+        for ch in chunks:
+            blob += b"%x\r\n" % len(ch) + ch + b"\r\n"
+        blob += b"0\r\n\r\n"
+        srv = ScriptedServer([(blob, False)])
+        port = await srv.start()
+        c = LeanClient()
+        r = await c.post(host="127.0.0.1", port=port, tls=False, path="/x",
+                         headers={}, body=b"{}")
+        got = bytearray()
+        async for piece in r.iter_chunks():
+            got.extend(piece)
+        assert bytes(got) == sse
+        r.release()
+        await c.close()
+        await srv.stop()
+
+    run(main())
