@@ -1085,17 +1085,28 @@ class GatewayServer:
             )
             override = backend.model_name_override or route.route.model_name_override
             tr, model = translator.request_multipart(raw, content_type, model_override=override)
-            up_headers = dict(tr.headers)
+            # client headers forward exactly like the JSON path (fresh copy
+            # per try, hop-by-hop + internal stripped), then mutations,
+            # then translator headers (multipart content-type wins), then
+            # auth over the final body — A.8 ordering
+            up_headers = {k: v for k, v in headers.items() if k not in _HOP_BY_HOP}
             up_headers = apply_header_mutation(up_headers, route.route.header_mutation)
             up_headers = apply_header_mutation(up_headers, backend.header_mutation)
+            up_headers.update(tr.headers)
             auth = build_auth_handler(backend)
             if auth is not None:
                 try:
                     up_headers = auth(up_headers, tr.body, "POST", tr.path)
                 except CredentialMissingError as e:
                     return _json_error(401, str(e), "authentication_error")
+            else:
+                # propagate client Authorization for auth-less backends
+                if "authorization" in headers:
+                    up_headers.setdefault("authorization", headers["authorization"])
             for h in rt.override_strip_headers:
                 up_headers.pop(h, None)
+            if backend.upstream.hostname:
+                up_headers["host"] = backend.upstream.hostname
             try:
                 upstream = await self._session.post(
                     host=backend.upstream.host,

@@ -461,6 +461,58 @@ def _start_gpu_host(socket_path: str, local_rank: int, window_ms: float = 0.1):
     return t
 
 
+
+def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
+    """Single-process native harness: C++ mock upstream + C++ gateway
+    (with in-process HIP admission) + C++ closed-loop load generator —
+    the Python asyncio harness caps near ~2k req/s per loadgen process
+    for 25 KiB bodies and was measured to be the bottleneck, not the
+    gateway. Returns (results_row, elapsed_s, workers, front)."""
+    import aigw_fast
+
+    from aigw.testing.fastmock import canned_chat_response
+
+    canned = canned_chat_response(prompt_tokens=args.tokens)
+    mock = aigw_fast.FastMock()
+    up_port = mock.start("127.0.0.1", canned.decode("latin1"))
+    gpu_direct = use_gpu
+    if gpu_direct:
+        torch.cuda.set_device(local_rank)
+    front, gw_port = _start_fast_front(args, [up_port], gpu_direct)
+
+    payload = json.dumps(build_payload(args.tokens)).encode()
+    workers = args.workers if args.workers > 0 else 16
+    conns = args.batch * workers
+    waves = max(args.waves, 1)
+    path = "/v1/chat/completions"
+
+    # warmup (gateway) then warm direct baseline
+    aigw_fast.run_load("127.0.0.1", gw_port, path, payload, conns,
+                       max(args.warmup, 1) * waves)
+    direct = aigw_fast.run_load("127.0.0.1", up_port, path, payload, conns, 5)
+    if use_gpu:
+        torch.cuda.synchronize()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    res = aigw_fast.run_load("127.0.0.1", gw_port, path, payload, conns,
+                             args.steps * waves)
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    barrier_sync()
+    assert res["errors"] == 0, f"load errors: {res}"
+    row = {
+        "elapsed": elapsed,
+        "requests": res["completed"],
+        "p50": res["p50_ms"],
+        "p99": res["p99_ms"],
+        "p50_direct": direct["p50_ms"],
+    }
+    mock.stop()
+    return row, elapsed, workers, front
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -491,6 +543,11 @@ def main():
     ap.add_argument("--gpu-hosts", type=int, default=0,
                     help="fast mode: 0 = in-process HIP admission (default); "
                          "N>0 = N UDS admission-host processes per rank")
+    ap.add_argument("--harness", choices=["native", "python"], default="native",
+                    help="fast mode: native = C++ loadgen/mock in one "
+                         "process (default; the Python harness IS the "
+                         "bottleneck above ~50k req/s); python = asyncio "
+                         "loadgen + mock processes")
     ap.add_argument("--gpu-window", type=float, default=0.1,
                     help="GPU micro-batch window per worker, ms")
     ap.add_argument("--gpu-service", action="store_true",
@@ -513,9 +570,27 @@ def main():
         torch.distributed.init_process_group(backend)
 
     fast_mode = args.front == "fast" and not args.cache_payloads
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
     front = None
+    native = False
     up_procs: list = []
-    if fast_mode:
+    if fast_mode and args.harness == "native":
+        try:
+            row, elapsed, workers, front = run_native_bench(
+                args, rank, world, local_rank, use_gpu and not args.no_gpu,
+                barrier_sync,
+            )
+            results = [row]
+            native = True
+        except ImportError:
+            fast_mode = False
+    elif fast_mode:
         try:
             front, go, out_q, procs, up_procs, workers = run_fast_mode(
                 args, rank, world, local_rank, use_gpu and not args.no_gpu
@@ -555,27 +630,22 @@ def main():
                     p.terminate()
                 raise RuntimeError("bench worker failed to become ready")
 
-    def barrier_sync():
-        if world > 1:
-            torch.distributed.barrier()
+    if not native:
+        barrier_sync()
+        t0 = time.perf_counter()
+        go.set()
+        results = [out_q.get(timeout=600) for _ in range(workers)]
+        # closing bracket: all GPU work retired before the clock stops
         if use_gpu:
             torch.cuda.synchronize()
-
-    barrier_sync()
-    t0 = time.perf_counter()
-    go.set()
-    results = [out_q.get(timeout=600) for _ in range(workers)]
-    # closing bracket: all GPU work retired before the clock stops
-    if use_gpu:
-        torch.cuda.synchronize()
-    elapsed = time.perf_counter() - t0
-    barrier_sync()
-    for p in procs:
-        p.join(timeout=60)
-        if p.is_alive():
+        elapsed = time.perf_counter() - t0
+        barrier_sync()
+        for p in procs:
+            p.join(timeout=60)
+            if p.is_alive():
+                p.terminate()
+        for p in up_procs:
             p.terminate()
-    for p in up_procs:
-        p.terminate()
 
     statesync_tick_us = None
     if world > 1:

@@ -14,6 +14,34 @@ using namespace aigw_fast;
 PYBIND11_MODULE(aigw_fast, m) {
   m.doc() = "aigw native data-plane fast path";
 
+  py::class_<FastMock>(m, "FastMock")
+      .def(py::init<>())
+      .def("start", &FastMock::start, py::arg("host"), py::arg("response"))
+      .def("stop", &FastMock::stop)
+      .def("requests", &FastMock::requests);
+
+  m.def(
+      "run_load",
+      [](const std::string& host, uint16_t port, const std::string& path,
+         py::bytes body, int connections, int per_conn) {
+        LoadResult r;
+        {
+          std::string b = body;
+          py::gil_scoped_release release;
+          r = run_load(host, port, path, b, connections, per_conn);
+        }
+        py::dict d;
+        d["elapsed_s"] = r.elapsed_s;
+        d["completed"] = r.completed;
+        d["errors"] = r.errors;
+        d["p50_ms"] = r.p50_ms;
+        d["p99_ms"] = r.p99_ms;
+        return d;
+      },
+      py::arg("host"), py::arg("port"), py::arg("path"), py::arg("body"),
+      py::arg("connections"), py::arg("per_conn"),
+      "closed-loop native load generator (threads, keep-alive)");
+
   py::class_<FastServer>(m, "FastServer")
       .def(py::init<>())
       .def(

@@ -166,4 +166,32 @@ class FastServer {
   std::map<std::string, FileCredState> file_creds_;
 };
 
+// -------- native bench harness (see fastpath.cpp) ---------------------------
+
+class FastMock {
+ public:
+  int start(const std::string& host, const std::string& response);
+  void stop();
+  uint64_t requests() const;
+  ~FastMock() { stop(); }
+
+ private:
+  void serve(int fd);
+  std::string response_;
+  int listen_fd_ = -1;
+  std::thread acceptor_;
+  std::atomic<uint64_t> served_{0};
+};
+
+struct LoadResult {
+  double elapsed_s = 0;
+  uint64_t completed = 0;
+  uint64_t errors = 0;
+  double p50_ms = 0, p99_ms = 0;
+};
+
+LoadResult run_load(const std::string& host, uint16_t port,
+                    const std::string& path, const std::string& body,
+                    int connections, int per_conn);
+
 }  // namespace aigw_fast
