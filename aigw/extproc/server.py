@@ -289,6 +289,9 @@ class GatewayServer:
         # per-backend live stats for the KV-occupancy endpoint picker:
         # name -> [active_requests, estimated_active_tokens]
         self._ep_stats: dict[str, list] = {}
+        # replica telemetry poller (aigw.extproc.telemetry); started by
+        # the CLI / make_app when any backend configures telemetry
+        self.telemetry = None
 
     # ---- lifecycle -----------------------------------------------------------
 
@@ -297,11 +300,23 @@ class GatewayServer:
             from aigw.extproc.upstream_client import LeanClient
 
             self._session = LeanClient()
+        if self.telemetry is None and any(
+            b.telemetry is not None
+            for cr in self.runtime.routes for tier in cr.tiers for b in tier
+        ):
+            from aigw.extproc.telemetry import ReplicaTelemetry
+
+            self.telemetry = ReplicaTelemetry(self.runtime)
+            await self.telemetry.scrape_once()  # first rows before traffic
+            await self.telemetry.start()
 
     async def close(self) -> None:
         if self._session is not None:
             await self._session.close()
             self._session = None
+        if self.telemetry is not None:
+            await self.telemetry.stop()
+            self.telemetry = None
 
     def swap_runtime(self, rc: RuntimeConfig) -> None:
         """Hot reload: new requests see the new config; in-flight requests
@@ -651,8 +666,20 @@ class GatewayServer:
             tier0 = route.tiers[0]
             stats_rows = []
             for b in tier0:
-                st = self._ep_stats.setdefault(b.name, [0, 0.0])
-                stats_rows.append([st[1], 100000.0, float(st[0]), float(st[0])])
+                row = None
+                if self.telemetry is not None:
+                    # scraped replica state (vLLM /metrics): real KV
+                    # occupancy + queue depth; stale rows fall back to
+                    # the gateway-local in-flight estimate
+                    row = self.telemetry.fresh_row(
+                        b.name,
+                        max_age_s=(b.telemetry.interval_s * 3
+                                   if b.telemetry else 5.0),
+                    )
+                if row is None:
+                    st = self._ep_stats.setdefault(b.name, [0, 0.0])
+                    row = [st[1], 100000.0, float(st[0]), float(st[0])]
+                stats_rows.append(row)
             pick = await self.gpu.pick_endpoint(
                 stats_rows, float(gpu_input_tokens or 256)
             )

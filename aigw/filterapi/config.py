@@ -255,6 +255,25 @@ class Backend:
     # the Envoy cluster circuit breakers the reference relies on; 0 = off).
     # A saturated backend counts as a failed attempt and fallback proceeds.
     max_concurrency: int = 0
+    # Replica telemetry for the endpoint picker: poll this backend's
+    # metrics endpoint (vLLM Prometheus format) so KV-occupancy scoring
+    # runs on REAL replica load instead of gateway-local estimates
+    # (InferencePool EPP inputs, extensionserver/inferencepool.go:39-54).
+    telemetry: Optional["BackendTelemetry"] = None
+
+
+@dataclass
+class BackendTelemetry:
+    """Metrics scrape config for one replica backend."""
+
+    path: str = "/metrics"
+    interval_s: float = 1.0
+    # Prometheus gauge names (vLLM defaults); override for other servers
+    kv_usage_metric: str = "vllm:gpu_cache_usage_perc"  # 0..1 fraction
+    running_metric: str = "vllm:num_requests_running"
+    waiting_metric: str = "vllm:num_requests_waiting"
+    # KV block capacity used to convert the usage fraction into blocks
+    kv_total: float = 100000.0
 
 
 @dataclass
@@ -485,6 +504,10 @@ def _parse_backend(d, ctx) -> Backend:
     if kw.get("body_mutation") is not None:
         kw["body_mutation"] = BodyMutation(
             **_dc(BodyMutation, kw["body_mutation"], f"{ctx}.bodyMutation")
+        )
+    if kw.get("telemetry") is not None:
+        kw["telemetry"] = BackendTelemetry(
+            **_dc(BackendTelemetry, kw["telemetry"], f"{ctx}.telemetry")
         )
     return Backend(**kw)
 
