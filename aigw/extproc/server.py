@@ -785,6 +785,7 @@ class GatewayServer:
                     span, body,
                     provider=provider_from_schema(backend.schema.name.value, backend.name),
                     backend=backend.name,
+                    operation=JSON_ENDPOINTS.get(endpoint, "request"),
                 )
             st = self._ep_stats.setdefault(backend.name, [0, 0.0])
             if backend.max_concurrency and st[0] >= backend.max_concurrency:
@@ -1099,6 +1100,10 @@ class GatewayServer:
         if not decision.allowed:
             self.metrics.ratelimit_denials.labels(rule=decision.rule).inc()
             return _json_error(429, "token budget exhausted", "rate_limit_exceeded")
+        span = None
+        operation = MULTIPART_ENDPOINTS.get(endpoint, "audio")
+        if self.tracer is not None:
+            span = self.tracer.start_span(f"{operation} {model}", headers)
 
         assert self._session is not None
         attempts_left = max_attempts(route)
@@ -1134,6 +1139,13 @@ class GatewayServer:
                 up_headers.pop(h, None)
             if backend.upstream.hostname:
                 up_headers["host"] = backend.upstream.hostname
+            if span is not None:
+                up_headers["traceparent"] = span.traceparent()
+                self.span_recorder.record_request(
+                    span, {"model": model},
+                    provider=provider_from_schema(backend.schema.name.value, backend.name),
+                    backend=backend.name, operation=operation,
+                )
             try:
                 upstream = await self._session.post(
                     host=backend.upstream.host,
@@ -1156,19 +1168,26 @@ class GatewayServer:
                     upstream.release()
                     if upstream.status in RETRIABLE_STATUSES and attempts_left > 0:
                         continue
+                    if span is not None:
+                        self.tracer.end_span(span, error=f"status {upstream.status}")
                     return web.Response(
                         body=translator.response_error(upstream.status, err, {}),
                         status=upstream.status,
                         content_type="application/json",
                     )
-                return await self._unary_response(
+                resp = await self._unary_response(
                     endpoint, route, backend, translator, upstream, headers,
-                    model, start, 0, None, None,
+                    model, start, 0, None, None, span=span,
                 )
+                if span is not None:
+                    self.tracer.end_span(span)
+                return resp
             except (UpstreamError, OSError, asyncio.TimeoutError,
                     asyncio.IncompleteReadError) as e:
                 last_error = str(e)
                 continue
+        if span is not None:
+            self.tracer.end_span(span, error=last_error or "no healthy upstream")
         return _json_error(503, last_error or "no healthy upstream", "upstream_error")
 
 

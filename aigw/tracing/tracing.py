@@ -127,39 +127,84 @@ class Tracer:
 
 
 class ChatSpanRecorder:
-    """Fills span attributes for a chat request per the selected
-    convention (internal/tracing/openinference + otelgenai recorders)."""
+    """Fills span attributes per operation and the selected convention
+    (internal/tracing/openinference/{openai,anthropic,cohere} + otelgenai
+    recorder families). Despite the historical name this recorder is
+    operation-aware: embeddings get EMBEDDING span kind and
+    embedding.* attributes, rerank gets RERANKER + reranker.*, audio
+    operations carry their operation name — chat semantics never leak
+    onto non-chat spans."""
+
+    # operation -> OpenInference span kind (openinference semconv)
+    _OI_KIND = {
+        "chat": "LLM",
+        "text_completion": "LLM",
+        "responses": "LLM",
+        "embeddings": "EMBEDDING",
+        "rerank": "RERANKER",
+        "speech": "LLM",
+        "transcription": "LLM",
+        "translation": "LLM",
+        "image_generation": "LLM",
+        "count_tokens": "LLM",
+        "tokenize": "LLM",
+    }
 
     def __init__(self, tracer: Tracer, capture_content: bool = False):
         self.tracer = tracer
         self.capture_content = capture_content
 
-    def record_request(self, span: Span, body: dict, *, provider: str, backend: str) -> None:
+    def record_request(self, span: Span, body: dict, *, provider: str,
+                       backend: str, operation: str = "chat") -> None:
+        body = body or {}
         model = body.get("model", "")
         if self.tracer.semconv == "openinference":
-            span.set("openinference.span.kind", "LLM")
-            span.set("llm.model_name", model)
-            span.set("llm.provider", provider)
-            span.set("llm.system", provider)
-            params = {
-                k: body[k]
-                for k in ("temperature", "top_p", "max_tokens", "max_completion_tokens", "stream")
-                if k in body
-            }
-            span.set("llm.invocation_parameters", json.dumps(params) if params else None)
-            if self.capture_content:
-                for i, m in enumerate(body.get("messages") or []):
-                    span.set(f"llm.input_messages.{i}.message.role", m.get("role"))
-                    c = m.get("content")
-                    if isinstance(c, str):
-                        span.set(f"llm.input_messages.{i}.message.content", c)
+            span.set("openinference.span.kind",
+                     self._OI_KIND.get(operation, "LLM"))
+            if operation == "embeddings":
+                span.set("embedding.model_name", model)
+                inp = body.get("input")
+                n = len(inp) if isinstance(inp, list) else 1 if inp else 0
+                span.set("embedding.invocation_parameters", json.dumps(
+                    {k: body[k] for k in ("dimensions", "encoding_format")
+                     if k in body}) or None)
+                span.set("embedding.embeddings_count", n)
+            elif operation == "rerank":
+                span.set("reranker.model_name", model)
+                span.set("reranker.top_n", body.get("top_n"))
+                docs = body.get("documents")
+                span.set("reranker.documents_count",
+                         len(docs) if isinstance(docs, list) else None)
+                if self.capture_content:
+                    span.set("reranker.query", body.get("query"))
+            else:
+                span.set("llm.model_name", model)
+                span.set("llm.provider", provider)
+                span.set("llm.system", provider)
+                params = {
+                    k: body[k]
+                    for k in ("temperature", "top_p", "max_tokens",
+                              "max_completion_tokens", "stream", "voice",
+                              "response_format", "size", "language")
+                    if k in body
+                }
+                span.set("llm.invocation_parameters",
+                         json.dumps(params) if params else None)
+                if self.capture_content and operation == "chat":
+                    for i, m in enumerate(body.get("messages") or []):
+                        span.set(f"llm.input_messages.{i}.message.role", m.get("role"))
+                        c = m.get("content")
+                        if isinstance(c, str):
+                            span.set(f"llm.input_messages.{i}.message.content", c)
         else:
-            span.set("gen_ai.operation.name", "chat")
+            span.set("gen_ai.operation.name", operation)
             span.set("gen_ai.provider.name", provider)
             span.set("gen_ai.request.model", model)
-            span.set("gen_ai.request.temperature", body.get("temperature"))
-            span.set("gen_ai.request.max_tokens",
-                     body.get("max_completion_tokens") or body.get("max_tokens"))
+            if operation in ("chat", "text_completion", "responses"):
+                span.set("gen_ai.request.temperature", body.get("temperature"))
+                span.set("gen_ai.request.max_tokens",
+                         body.get("max_completion_tokens") or body.get("max_tokens"))
+        span.set("aigw.operation", operation)
         span.set("aigw.backend", backend)
 
     def record_chunk(self, span: Span) -> None:
