@@ -1,3 +1,3 @@
-from .state_sync import StateSync
+from .state_sync import FastLimiterBridge, StateSync
 
-__all__ = ["StateSync"]
+__all__ = ["FastLimiterBridge", "StateSync"]

@@ -175,3 +175,42 @@ class StateSync:
             except asyncio.CancelledError:
                 pass
             self._task = None
+
+
+class FastLimiterBridge:
+    """Adapts the native fast front's per-rule rate counters
+    (aigw_fast.FastServer rl_collect_deltas / rl_apply_remote) to the
+    limiter interface StateSync drives, so one code path covers both
+    fronts: the C++ server keeps its fixed-window atomics, this bridge
+    maps them onto the cross-shard slot table (rule name, "" descriptor)
+    and folds remote spend back in after each fused all-reduce."""
+
+    def __init__(self, fast_server, rule_names: list):
+        self.fast = fast_server
+        self.rule_names = list(rule_names)
+        self._slot_to_idx = {
+            bucket_slot(name, ""): i for i, name in enumerate(self.rule_names)
+        }
+
+    def collect_deltas(self) -> dict:
+        deltas = self.fast.rl_collect_deltas()
+        return {
+            (self.rule_names[i], ""): d
+            for i, d in enumerate(deltas)
+            if d
+        }
+
+    def _apply(self, idx: int, remote: int) -> None:
+        vec = [0] * len(self.rule_names)
+        vec[idx] = remote
+        self.fast.rl_apply_remote(vec)
+
+    def apply_remote(self, key, total: int, mine: int) -> None:
+        remote = total - mine
+        if remote > 0 and key[0] in self.rule_names:
+            self._apply(self.rule_names.index(key[0]), remote)
+
+    def apply_remote_slot(self, slot: int, remote: int) -> None:
+        idx = self._slot_to_idx.get(slot)
+        if idx is not None and remote > 0:
+            self._apply(idx, remote)

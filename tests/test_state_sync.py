@@ -172,3 +172,126 @@ def test_slot_remote_claimed_on_first_local_sight():
     limiter.apply_remote_slot(slot, 150)  # remote shard spent alice's budget
     d = limiter.check({"x-user-id": "alice"})
     assert not d.allowed
+
+
+def test_counter_sync_world8():
+    """The full node shape: 8 shards, same collective sequence the
+    round-end SCALE bench runs (gloo here, RCCL on hardware)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29781
+    procs = [ctx.Process(target=_world8_worker, args=(r, 8, port, q)) for r in range(8)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(8)]
+    for p in procs:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+            pytest.fail("worker hung")
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
+
+
+def _world8_worker(rank: int, world: int, port: int, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+
+        from aigw.parallel import StateSync
+
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        limiter = RateLimiter([
+            RateLimitRule(name="node", metadata_key="llm_total_token",
+                          limit=100, window_s=3600),
+        ])
+        sync = StateSync(limiter)
+        # every shard spends 14: below the limit locally, 112 > 100 globally
+        limiter.charge({}, {"llm_total_token": 14})
+        sync.tick_sync()
+        sync.tick_sync()
+        d = limiter.check({})
+        assert not d.allowed, f"rank {rank}: global budget not enforced: {d}"
+        dist.barrier()
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
+
+
+def _fastfront_worker(rank: int, world: int, port: int, q):
+    """One native fast-front shard per rank, real requests charging its
+    C++ rate counters, StateSync over the FastLimiterBridge."""
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import json as _json
+
+        import torch.distributed as dist
+
+        import aigw_fast
+        from aigw.parallel import FastLimiterBridge, StateSync
+
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        body = _json.dumps({
+            "id": "x", "object": "chat.completion", "model": "m",
+            "choices": [],
+            "usage": {"prompt_tokens": 5, "completion_tokens": 2,
+                      "total_tokens": 7},
+        }).encode()
+        resp = (b"HTTP/1.1 200 OK\r\ncontent-type: application/json\r\n"
+                b"content-length: %d\r\n\r\n" % len(body)) + body
+        mock = aigw_fast.FastMock()
+        up_port = mock.start("127.0.0.1", resp.decode("latin1"))
+        srv = aigw_fast.FastServer()
+        srv.add_route("r", "", 1, True, True,
+                      [{"name": "b", "host": "127.0.0.1", "port": up_port}])
+        srv.add_rate_rule("node", 50, 3600.0, "llm_total_token")
+        gw_port = srv.start("127.0.0.1", 0)
+        payload = _json.dumps({"model": "m", "messages": [
+            {"role": "user", "content": "hi"}]}).encode()
+        # each shard serves 1 request -> 7 tokens; 56 > 50 node-wide
+        r = aigw_fast.run_load("127.0.0.1", gw_port,
+                               "/v1/chat/completions", payload, 1, 1)
+        assert r["completed"] == 1, r
+        sync = StateSync(FastLimiterBridge(srv, ["node"]))
+        sync.tick_sync()
+        sync.tick_sync()
+        # every shard must now deny locally (global spend 56 >= 50)
+        r = aigw_fast.run_load("127.0.0.1", gw_port,
+                               "/v1/chat/completions", payload, 1, 1)
+        assert r["completed"] == 0 and r["errors"] == 1, r
+        st = srv.stats()
+        assert st["local_429"] == 1, st
+        dist.barrier()
+        dist.destroy_process_group()
+        srv.stop()
+        mock.stop()
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
+
+
+def test_fastfront_counter_sync_world8():
+    """8 native shards: real HTTP requests charge the C++ counters, one
+    fused gloo all-reduce folds the node-wide spend, every shard denies."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29783
+    procs = [ctx.Process(target=_fastfront_worker, args=(r, 8, port, q))
+             for r in range(8)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(8)]
+    for p in procs:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+            pytest.fail("worker hung")
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
