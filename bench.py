@@ -478,7 +478,7 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
     gpu_direct = use_gpu
     if gpu_direct:
         torch.cuda.set_device(local_rank)
-    front, gw_port = _start_fast_front(args, [up_port], gpu_direct)
+    front, gw_port = _start_fast_front(args, [up_port], None, gpu_direct)
 
     payload = json.dumps(build_payload(args.tokens)).encode()
     workers = args.workers if args.workers > 0 else 16
