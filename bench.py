@@ -462,6 +462,30 @@ def _start_gpu_host(socket_path: str, local_rank: int, window_ms: float = 0.1):
 
 
 
+
+def _cpu_quota() -> int:
+    """Effective CPU budget: the cgroup quota, NOT nproc — gpurun boxes
+    expose 256 cores but cap the container at 16 CPUs (cpu.max), and
+    oversubscribing the quota earns 100 ms CFS throttle stalls that
+    poison p99 (measured: 64 connections -> 78k req/s @ p99 1 ms;
+    512 connections -> 19k @ p99 102 ms on the same box)."""
+    for path in ("/sys/fs/cgroup/cpu.max",):
+        try:
+            parts = open(path).read().split()
+            if parts[0] != "max":
+                return max(1, int(int(parts[0]) / int(parts[1])))
+        except (OSError, ValueError, IndexError):
+            pass
+    try:
+        q = int(open("/sys/fs/cgroup/cpu/cpu.cfs_quota_us").read())
+        p = int(open("/sys/fs/cgroup/cpu/cpu.cfs_period_us").read())
+        if q > 0:
+            return max(1, q // p)
+    except (OSError, ValueError):
+        pass
+    return os.cpu_count() or 8
+
+
 def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
     """Single-process native harness: C++ mock upstream + C++ gateway
     (with in-process HIP admission) + C++ closed-loop load generator —
@@ -481,7 +505,7 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
     front, gw_port = _start_fast_front(args, [up_port], None, gpu_direct)
 
     payload = json.dumps(build_payload(args.tokens)).encode()
-    workers = args.workers if args.workers > 0 else 16
+    workers = args.workers if args.workers > 0 else max(1, _cpu_quota() // 8)
     conns = args.batch * workers
     waves = max(args.waves, 1)
     path = "/v1/chat/completions"
