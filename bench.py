@@ -543,10 +543,11 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--batch", type=int, default=32, help="concurrent requests per wave")
-    ap.add_argument("--waves", type=int, default=50,
-                    help="sequential waves of --batch requests per step "
-                         "(per worker); sizes the timed region so the "
-                         "driver's --steps 20 runs >=5 s at steady state")
+    ap.add_argument("--waves", type=int, default=250,
+                    help="sequential requests per connection per step; "
+                         "sizes the timed region so the driver's "
+                         "--steps 20 runs a multi-second region at the "
+                         "native harness's ~70-80k req/s")
     ap.add_argument("--tokens", type=int, default=4096)
     ap.add_argument("--workers", type=int, default=0,
                     help="HTTP worker processes per shard (0 = auto)")

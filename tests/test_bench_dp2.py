@@ -21,7 +21,7 @@ def test_torchrun_dp2_bench():
             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
             "--master-port", "29733", os.path.join(repo, "bench.py"),
             "--gpus", "2", "--steps", "2", "--warmup", "1", "--batch", "8",
-            "--workers", "1", "--waves", "2",
+            "--workers", "1", "--waves", "2", "--upstreams", "1",
         ],
         env=env,
         capture_output=True,
