@@ -91,7 +91,9 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
                 gpu_host = GPUServiceHost(gpu_services, gpu_socket)
                 await gpu_host.start()
     server = GatewayServer(
-        runtime, gpu_services=gpu_services, root_prefix=getattr(args, "root_prefix", "")
+        runtime, gpu_services=gpu_services,
+        root_prefix=getattr(args, "root_prefix", ""),
+        strict_schema=getattr(args, "strict_schema", False),
     )
 
     sync = None
@@ -121,7 +123,26 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
 
     port = args.port + rank
     reuse = getattr(args, "workers", 1) > 1
-    if getattr(args, "front", "lean") == "aiohttp":
+    front_choice = getattr(args, "front", "lean")
+    if front_choice == "fast":
+        # native C++ data plane (csrc/fastpath.cpp); cold paths ride its
+        # loopback fallback app. Configs the native server cannot express
+        # fail loudly here — pick --front lean for those.
+        import torch as _torch
+
+        from aigw.extproc.fast_front import FastFront
+
+        fast = FastFront(
+            server, runtime,
+            gpu_direct=bool(args.gpu and _torch.cuda.is_available()),
+        )
+        port = await fast.start(args.host, port)
+
+        async def _fast_cleanup():
+            await fast.stop()
+
+        front_cleanup = _fast_cleanup
+    elif front_choice == "aiohttp":
         runner = await run_server(server, host=args.host, port=port, reuse_port=reuse)
         front_cleanup = runner.cleanup
     else:
@@ -285,7 +306,10 @@ def main(argv=None) -> int:
     runp.add_argument("--shards", type=int, default=1, help="shards (one per GPU)")
     runp.add_argument("--root-prefix", default="", dest="root_prefix",
                       help="global path prefix for every endpoint")
-    runp.add_argument("--front", choices=["lean", "aiohttp"], default="lean",
+    runp.add_argument("--strict-schema", action="store_true", dest="strict_schema",
+                      help="typed-union request validation at ingress "
+                           "(aigw.apischema); wrong shapes answer 400")
+    runp.add_argument("--front", choices=["lean", "aiohttp", "fast"], default="lean",
                       help="HTTP front: lean raw-asyncio (benchmarked default; aiohttp loopback serves multipart/MCP) or pure aiohttp")
     runp.add_argument("--admin-port", type=int, default=0, dest="admin_port",
                       help="separate localhost admin server for /health and /metrics (0 = serve on the data port only)")

@@ -505,7 +505,7 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
     front, gw_port = _start_fast_front(args, [up_port], None, gpu_direct)
 
     payload = json.dumps(build_payload(args.tokens)).encode()
-    workers = args.workers if args.workers > 0 else max(1, _cpu_quota() // 8)
+    workers = args.workers if args.workers > 0 else max(1, _cpu_quota() // 4)
     conns = args.batch * workers
     waves = max(args.waves, 1)
     path = "/v1/chat/completions"

@@ -271,3 +271,76 @@ routes:
     })
     with pytest.raises(FastFrontUnsupported):
         build_fast_server(RuntimeConfig(cfg))
+
+
+def test_cli_run_fast_front():
+    """`aigw run --front fast` serves through the native server, with
+    cold paths on its fallback."""
+    import os
+    import subprocess
+    import sys
+    import tempfile
+    import time as _time
+    import urllib.request
+
+    import torch  # noqa: F401
+    import aigw_fast
+
+    body = json.dumps({"id": "x", "object": "chat.completion", "model": "m",
+                       "choices": [], "usage": {"total_tokens": 3}}).encode()
+    resp = (b"HTTP/1.1 200 OK\r\ncontent-type: application/json\r\n"
+            b"content-length: %d\r\n\r\n" % len(body)) + body
+    mock = aigw_fast.FastMock()
+    up_port = mock.start("127.0.0.1", resp.decode("latin1"))
+    cfg = {
+        "routes": [{"name": "r", "backends": [
+            {"name": "b", "schema": "OpenAI",
+             "upstream": {"host": "127.0.0.1", "port": up_port},
+             "auth": {"apiKey": "sk-cli"}}]}],
+        "models": [{"name": "cli-model"}],
+    }
+    with tempfile.NamedTemporaryFile("w", suffix=".yaml", delete=False) as f:
+        yaml.safe_dump(cfg, f)
+        cfg_path = f.name
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "aigw", "run", "--config", cfg_path,
+         "--front", "fast", "--host", "127.0.0.1", "--port", "0"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        port = None
+        for _ in range(300):
+            line = proc.stdout.readline()
+            if "aigw shard" in line and "listening on" in line:
+                port = int(line.rsplit(":", 1)[1].strip())
+                break
+        assert port, "gateway did not start"
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{port}/v1/chat/completions",
+            data=json.dumps({"model": "cli-model",
+                             "messages": [{"role": "user", "content": "x"}]}).encode(),
+            headers={"content-type": "application/json"})
+        for _ in range(50):
+            try:
+                with urllib.request.urlopen(req, timeout=5) as r:
+                    assert r.status == 200
+                break
+            except (ConnectionError, OSError):
+                _time.sleep(0.1)
+        else:
+            raise AssertionError("hot path never answered")
+        assert mock.requests() >= 1
+        with urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/v1/models", timeout=5) as r:
+            data = json.loads(r.read())
+            assert data["data"][0]["id"] == "cli-model"
+        with urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/health", timeout=5) as r:
+            assert json.loads(r.read())["front"] == "fast"
+    finally:
+        proc.terminate()
+        proc.wait(timeout=15)
+        mock.stop()
+        os.unlink(cfg_path)
