@@ -139,8 +139,8 @@ class FastFront:
 
     def __init__(self, server, runtime: RuntimeConfig, *, gpu_socket: str = "",
                  gpu_window_us: int = 100, gpu_max_batch: int = 256,
-                 gpu_direct: bool = False, n_merges: int = 32768,
-                 tokenizer_seed: int = 1355):
+                 gpu_direct: bool = False, gpu_device: int = 0,
+                 n_merges: int = 32768, tokenizer_seed: int = 1355):
         # `server` is the Python GatewayServer used for cold paths
         self.py_server = server
         self.runtime = runtime
@@ -149,6 +149,7 @@ class FastFront:
         self.gpu_window_us = gpu_window_us
         self.gpu_max_batch = gpu_max_batch
         self.gpu_direct = gpu_direct
+        self.gpu_device = gpu_device
         self.n_merges = n_merges
         self.tokenizer_seed = tokenizer_seed
         self._fallback_runner = None
@@ -172,7 +173,8 @@ class FastFront:
             keys, ranks = build_hash_table(make_merges(self.n_merges,
                                                        self.tokenizer_seed))
             self.fast.enable_gpu_direct(keys, ranks,
-                                        max_batch=self.gpu_max_batch)
+                                        max_batch=self.gpu_max_batch,
+                                        device=self.gpu_device)
         sockets = ([self.gpu_socket] if isinstance(self.gpu_socket, str)
                    else list(self.gpu_socket or []))
         for s in sockets:

@@ -77,9 +77,11 @@ __global__ void excl_scan_kernel(const int32_t* __restrict__ in, int n,
 class GpuAdmissionDirect {
  public:
   bool init(const long long* htab_keys, const int32_t* htab_rank, int htab_n,
-            size_t max_bytes, int max_req) {
+            size_t max_bytes, int max_req, int device) {
     int count = 0;
     if (hipGetDeviceCount(&count) != hipSuccess || count == 0) return false;
+    device_ = device >= 0 && device < count ? device : 0;
+    HIP_OK(hipSetDevice(device_));
     max_bytes_ = max_bytes;
     max_req_ = max_req;
     htab_mask_ = htab_n - 1;
@@ -125,6 +127,10 @@ class GpuAdmissionDirect {
              int32_t* counts_out) {
     if (!ready_ || n == 0 || n_req == 0 || n > max_bytes_ || n_req > max_req_)
       return false;
+    // launches follow the CALLER thread's current device; the batcher
+    // thread differs from the init thread, so pin it here (no-op when
+    // already current) — rank N of an 8-GPU node must stay on device N
+    HIP_OK(hipSetDevice(device_));
     memcpy(h_bytes_, bytes, n);
     memcpy(h_off_, offsets, sizeof(int64_t) * (size_t)n_req);
     HIP_OK(hipMemcpyAsync(d_bytes_, h_bytes_, n, hipMemcpyHostToDevice, stream_));
@@ -188,6 +194,7 @@ class GpuAdmissionDirect {
 
  private:
   bool ready_ = false;
+  int device_ = 0;
   size_t max_bytes_ = 0;
   int max_req_ = 0;
   int htab_mask_ = 0;
@@ -216,9 +223,9 @@ class GpuAdmissionDirect {
 // headers; admission.hip is the only HIP translation unit in aigw_fast).
 GpuAdmissionDirect* admission_create(const long long* htab_keys,
                                      const int32_t* htab_rank, int htab_n,
-                                     size_t max_bytes, int max_req) {
+                                     size_t max_bytes, int max_req, int device) {
   auto* a = new GpuAdmissionDirect();
-  if (!a->init(htab_keys, htab_rank, htab_n, max_bytes, max_req)) {
+  if (!a->init(htab_keys, htab_rank, htab_n, max_bytes, max_req, device)) {
     delete a;
     return nullptr;
   }

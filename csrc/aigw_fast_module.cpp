@@ -92,7 +92,7 @@ PYBIND11_MODULE(aigw_fast, m) {
       .def(
           "enable_gpu_direct",
           [](FastServer& s, py::buffer htab_keys, py::buffer htab_rank,
-             int max_batch, size_t max_batch_bytes, int max_req) {
+             int max_batch, size_t max_batch_bytes, int max_req, int device) {
             py::buffer_info ki = htab_keys.request();
             py::buffer_info ri = htab_rank.request();
             if (ki.itemsize != 8 || ri.itemsize != 4)
@@ -103,12 +103,12 @@ PYBIND11_MODULE(aigw_fast, m) {
             s.enable_gpu_direct(static_cast<const long long*>(ki.ptr),
                                 static_cast<const int32_t*>(ri.ptr),
                                 (int)ki.size, max_batch, max_batch_bytes,
-                                max_req);
+                                max_req, device);
           },
           py::arg("htab_keys"), py::arg("htab_rank"),
           py::arg("max_batch") = 1024,
           py::arg("max_batch_bytes") = (size_t)48 * 1024 * 1024,
-          py::arg("max_req") = 4096)
+          py::arg("max_req") = 4096, py::arg("device") = 0)
       .def("start", &FastServer::start, py::arg("host"), py::arg("port"),
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &FastServer::stop, py::call_guard<py::gil_scoped_release>())

@@ -596,7 +596,7 @@ class GpuAdmissionClient {
 // admission.hip entry points (same .so, separate HIP translation unit)
 GpuAdmissionDirect* admission_create(const long long* htab_keys,
                                      const int32_t* htab_rank, int htab_n,
-                                     size_t max_bytes, int max_req);
+                                     size_t max_bytes, int max_req, int device);
 bool admission_count(GpuAdmissionDirect* a, const char* bytes, size_t n,
                      const int64_t* offsets, int n_req, int32_t* counts_out);
 void admission_destroy(GpuAdmissionDirect* a);
@@ -608,8 +608,9 @@ void admission_destroy(GpuAdmissionDirect* a);
 class DirectGpuBatcher {
  public:
   bool start(const long long* htab_keys, const int32_t* htab_rank, int htab_n,
-             int max_batch, size_t max_bytes, int max_req) {
-    adm_ = admission_create(htab_keys, htab_rank, htab_n, max_bytes, max_req);
+             int max_batch, size_t max_bytes, int max_req, int device) {
+    adm_ = admission_create(htab_keys, htab_rank, htab_n, max_bytes, max_req,
+                            device);
     if (adm_ == nullptr) return false;
     max_batch_ = std::min(max_batch, max_req);
     max_bytes_ = max_bytes;
@@ -1571,10 +1572,10 @@ int64_t FastServer::gpu_count(const std::string& text) {
 void FastServer::enable_gpu_direct(const long long* htab_keys,
                                    const int32_t* htab_rank, int htab_n,
                                    int max_batch, size_t max_batch_bytes,
-                                   int max_req) {
+                                   int max_req, int device) {
   auto b = std::make_unique<DirectGpuBatcher>();
   if (!b->start(htab_keys, htab_rank, htab_n, max_batch, max_batch_bytes,
-                max_req))
+                max_req, device))
     throw std::runtime_error(
         "direct GPU admission init failed (no device visible?)");
   gpu_direct_ = std::move(b);

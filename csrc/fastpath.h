@@ -108,7 +108,7 @@ class FastServer {
   // Throws when no GPU is visible or init fails.
   void enable_gpu_direct(const long long* htab_keys, const int32_t* htab_rank,
                          int htab_n, int max_batch, size_t max_batch_bytes,
-                         int max_req);
+                         int max_req, int device);
 
   // lifecycle
   int start(const std::string& host, uint16_t port);  // returns bound port
