@@ -401,6 +401,15 @@ class GatewayServer:
                 self._mcp_proxies.append(proxy)
                 for method in ("POST", "GET", "DELETE"):
                     app.router.add_route(method, mr.path, proxy.handle)
+                if mr.oauth is not None:
+                    # RFC 9728/8414 metadata, outside the authn gate
+                    suffix = "" if mr.path == "/" else mr.path
+                    app.router.add_get(
+                        "/.well-known/oauth-protected-resource" + suffix,
+                        proxy.handle_protected_resource_metadata)
+                    app.router.add_get(
+                        "/.well-known/oauth-authorization-server" + suffix,
+                        proxy.handle_authorization_server_metadata)
 
             async def _close_mcp(_app):
                 for p in self._mcp_proxies:

@@ -342,6 +342,12 @@ class MCPOAuth:
     audiences: list[str] = field(default_factory=list)
     jwks: str = ""       # inline JWKS JSON (localJWKS analogue)
     jwks_file: str = ""
+    # OAuth protected-resource metadata (RFC 9728) served at
+    # /.well-known/oauth-protected-resource (mcp_route_security_policy.go
+    # buildOAuthProtectedResourceMetadataJSON :474-502)
+    resource: str = ""
+    resource_name: str = ""
+    scopes_supported: list[str] = field(default_factory=list)
 
 
 @dataclass

@@ -241,6 +241,56 @@ class MCPProxy:
             )
         return resp
 
+    @staticmethod
+    def _cors(resp: web.Response) -> web.Response:
+        # the MCP inspector runs in a browser on a different origin
+        # (mcp_route_security_policy.go ensureCORSHeaders)
+        resp.headers["Access-Control-Allow-Origin"] = "*"
+        resp.headers["Access-Control-Allow-Methods"] = "GET"
+        resp.headers["Access-Control-Allow-Headers"] = "mcp-protocol-version"
+        return resp
+
+    async def handle_protected_resource_metadata(self, request: web.Request) -> web.Response:
+        """RFC 9728 OAuth protected-resource metadata, served WITHOUT the
+        authn gate (buildOAuthProtectedResourceMetadataJSON :474-502)."""
+        oauth = self.route.oauth
+        if oauth is None:
+            return web.Response(status=404)
+        doc = {
+            "resource": oauth.resource or self.route.path,
+            "authorization_servers": [oauth.issuer] if oauth.issuer else [],
+            "bearer_methods_supported": ["header"],
+        }
+        if oauth.resource_name:
+            doc["resource_name"] = oauth.resource_name
+        if oauth.scopes_supported:
+            doc["scopes_supported"] = oauth.scopes_supported
+        return self._cors(web.json_response(doc))
+
+    async def handle_authorization_server_metadata(self, request: web.Request) -> web.Response:
+        """RFC 8414 authorization-server metadata. The reference proxies
+        the AS's own document with a hardcoded fallback
+        (buildOAuthAuthServerMetadataJSON :504-560); without egress this
+        gateway serves the fallback shape derived from the issuer."""
+        oauth = self.route.oauth
+        if oauth is None or not oauth.issuer:
+            return web.Response(status=404)
+        issuer = oauth.issuer.rstrip("/")
+        doc = {
+            "issuer": issuer,
+            "authorization_endpoint": f"{issuer}/authorize",
+            "token_endpoint": f"{issuer}/token",
+            "registration_endpoint": f"{issuer}/register",
+            "response_types_supported": ["code"],
+            "grant_types_supported": ["authorization_code", "refresh_token"],
+            "code_challenge_methods_supported": ["S256"],
+            "token_endpoint_auth_methods_supported": ["client_secret_post",
+                                                      "client_secret_basic", "none"],
+        }
+        if oauth.scopes_supported:
+            doc["scopes_supported"] = oauth.scopes_supported
+        return self._cors(web.json_response(doc))
+
     async def handle(self, request: web.Request) -> web.StreamResponse:
         import time as _time
 

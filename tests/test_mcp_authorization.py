@@ -323,3 +323,57 @@ def test_mcp_authz_scoped_jwt_allows():
         await runner.cleanup()
 
     asyncio.run(run())
+
+
+def test_wellknown_oauth_metadata_endpoints():
+    """RFC 9728 protected-resource + RFC 8414 authorization-server
+    metadata served outside the authn gate, with CORS for the browser
+    inspector (mcp_route_security_policy.go :340-560)."""
+    from aigw.extproc.server import GatewayServer
+    from aigw.filterapi import RuntimeConfig
+    from tests.test_mcp_jwt import _jwks
+
+    async def run():
+        cfg = load_config({
+            "routes": [],
+            "mcp": {"routes": [{
+                "name": "m", "path": "/mcp",
+                "bearerToken": "gate",
+                "backends": [{"name": "a", "upstream": {"host": "h", "port": 1}}],
+                "oauth": {"issuer": "https://as.example.com",
+                          "jwks": _jwks(),
+                          "resource": "https://gw.example.com/mcp",
+                          "resourceName": "aigw mcp",
+                          "scopesSupported": ["mcp:read", "mcp:call"]},
+            }]},
+        })
+        server = GatewayServer(RuntimeConfig(cfg))
+        app = server.make_app()
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.TCPSite(runner, "127.0.0.1", 0)
+        await site.start()
+        port = runner.addresses[0][1]
+        async with aiohttp.ClientSession() as c:
+            # no Authorization header needed
+            async with c.get(
+                f"http://127.0.0.1:{port}/.well-known/oauth-protected-resource/mcp"
+            ) as r:
+                assert r.status == 200
+                doc = await r.json()
+                assert doc["resource"] == "https://gw.example.com/mcp"
+                assert doc["authorization_servers"] == ["https://as.example.com"]
+                assert doc["bearer_methods_supported"] == ["header"]
+                assert doc["scopes_supported"] == ["mcp:read", "mcp:call"]
+                assert r.headers["Access-Control-Allow-Origin"] == "*"
+            async with c.get(
+                f"http://127.0.0.1:{port}/.well-known/oauth-authorization-server/mcp"
+            ) as r:
+                assert r.status == 200
+                doc = await r.json()
+                assert doc["issuer"] == "https://as.example.com"
+                assert doc["token_endpoint"] == "https://as.example.com/token"
+                assert "S256" in doc["code_challenge_methods_supported"]
+        await runner.cleanup()
+
+    asyncio.run(run())
