@@ -83,6 +83,42 @@ bool write_all(int fd, const char* p, size_t n) {
 
 bool write_all(int fd, const std::string& s) { return write_all(fd, s.data(), s.size()); }
 
+// vectored head+body send: one syscall for the common two-part write
+// (saves ~2 syscalls per hop at 80k req/s on a CPU-quota-bound box)
+bool write_two(int fd, const std::string& a, const std::string& b) {
+  struct iovec iov[2];
+  iov[0].iov_base = const_cast<char*>(a.data());
+  iov[0].iov_len = a.size();
+  iov[1].iov_base = const_cast<char*>(b.data());
+  iov[1].iov_len = b.size();
+  size_t total = a.size() + b.size();
+  size_t sent = 0;
+  struct msghdr msg;
+  memset(&msg, 0, sizeof(msg));
+  msg.msg_iov = iov;
+  msg.msg_iovlen = 2;
+  while (sent < total) {
+    ssize_t w = ::sendmsg(fd, &msg, MSG_NOSIGNAL);
+    if (w < 0) {
+      if (errno == EINTR) continue;
+      return false;
+    }
+    sent += (size_t)w;
+    // advance iovecs
+    size_t skip = (size_t)w;
+    while (msg.msg_iovlen > 0 && skip >= msg.msg_iov[0].iov_len) {
+      skip -= msg.msg_iov[0].iov_len;
+      ++msg.msg_iov;
+      --msg.msg_iovlen;
+    }
+    if (msg.msg_iovlen > 0) {
+      msg.msg_iov[0].iov_base = (char*)msg.msg_iov[0].iov_base + skip;
+      msg.msg_iov[0].iov_len -= skip;
+    }
+  }
+  return true;
+}
+
 // one recv; 0 = clean EOF, -1 = error/timeout
 ssize_t read_some(int fd, char* p, size_t cap) {
   for (;;) {
@@ -878,7 +914,12 @@ class ConnHandler {
     }
 
     // attempt order: tiers by priority, weighted shuffle inside a tier
-    std::vector<const FastBackend*> order = attempt_order(*route);
+    // (single-backend routes — the common case — skip the machinery)
+    std::vector<const FastBackend*> order;
+    if (route->backends.size() == 1)
+      order.push_back(&route->backends[0]);
+    else
+      order = attempt_order(*route);
     int attempts_left = std::min((int)order.size(), route->retries + 1);
     bool first = true;
     for (const FastBackend* be : order) {
@@ -967,7 +1008,7 @@ class ConnHandler {
 
     int ufd = srv_->pool_->acquire(be.host, be.port, be.timeout_s);
     if (ufd < 0) return 1;
-    if (!write_all(ufd, head) || !write_all(ufd, *send_body)) {
+    if (!write_two(ufd, head, *send_body)) {
       ::close(ufd);
       return 1;
     }
@@ -1100,8 +1141,7 @@ class ConnHandler {
     }
     head += "content-length: " + std::to_string(body.size()) + "\r\n\r\n";
     srv_->stats_.bytes_out += head.size() + body.size();
-    if (!write_all(fd_, head)) return false;
-    return write_all(fd_, body);
+    return write_two(fd_, head, body);
   }
 
   // relay a streamed response with its upstream framing preserved and a
@@ -1208,7 +1248,7 @@ class ConnHandler {
       head += h.name + ": " + h.value + "\r\n";
     }
     head += "\r\n";
-    if (!write_all(ufd, head) || !write_all(ufd, body)) {
+    if (!write_two(ufd, head, body)) {
       ::close(ufd);
       return simple_reply(502, "upstream_error", "fallback write failed", false);
     }
@@ -1324,8 +1364,7 @@ class ConnHandler {
                        ctype + "\r\ncontent-length: " + std::to_string(body.size()) +
                        "\r\n\r\n";
     srv_->stats_.bytes_out += head.size() + body.size();
-    if (!write_all(fd_, head)) return false;
-    return write_all(fd_, body);
+    return write_two(fd_, head, body);
   }
 
   bool simple_reply(int status, const char* type, const std::string& msg,
@@ -1346,7 +1385,7 @@ class ConnHandler {
     if (close_after) head += "connection: close\r\n";
     head += "\r\n";
     srv_->stats_.bytes_out += head.size() + body.size();
-    bool ok = write_all(fd_, head) && write_all(fd_, body);
+    bool ok = write_two(fd_, head, body);
     return ok && !close_after;
   }
 
