@@ -140,6 +140,8 @@ class FastFront:
     def __init__(self, server, runtime: RuntimeConfig, *, gpu_socket: str = "",
                  gpu_window_us: int = 100, gpu_max_batch: int = 256,
                  gpu_direct: bool = False, gpu_device: int = 0,
+                 gpu_cache: bool = False, cache_capacity: int = 65536,
+                 cache_threshold: float = 0.92, cache_seed: int = 7,
                  n_merges: int = 32768, tokenizer_seed: int = 1355):
         # `server` is the Python GatewayServer used for cold paths
         self.py_server = server
@@ -150,6 +152,10 @@ class FastFront:
         self.gpu_max_batch = gpu_max_batch
         self.gpu_direct = gpu_direct
         self.gpu_device = gpu_device
+        self.gpu_cache = gpu_cache
+        self.cache_capacity = cache_capacity
+        self.cache_threshold = cache_threshold
+        self.cache_seed = cache_seed
         self.n_merges = n_merges
         self.tokenizer_seed = tokenizer_seed
         self._fallback_runner = None
@@ -175,6 +181,21 @@ class FastFront:
             self.fast.enable_gpu_direct(keys, ranks,
                                         max_batch=self.gpu_max_batch,
                                         device=self.gpu_device)
+            if self.gpu_cache:
+                # same random-init weights as aigw.ops.semcache (seeded),
+                # bf16 passed as a uint16 view; the native server owns
+                # the index ring and serves hits without Python
+                import torch as _t
+
+                g = _t.Generator().manual_seed(self.cache_seed)
+                vocab = 256 + self.n_merges
+                dim = 384
+                emb = _t.randn(vocab, dim, generator=g).to(_t.bfloat16)
+                proj = (_t.randn(dim, dim, generator=g) / dim ** 0.5).to(_t.bfloat16)
+                self.fast.enable_gpu_direct_cache(
+                    emb.view(_t.uint16).numpy(), proj.view(_t.uint16).numpy(),
+                    dim=dim, capacity=self.cache_capacity,
+                    threshold=self.cache_threshold)
         sockets = ([self.gpu_socket] if isinstance(self.gpu_socket, str)
                    else list(self.gpu_socket or []))
         for s in sockets:
@@ -195,6 +216,7 @@ class FastFront:
         lines = []
         for k in ("requests", "responses_2xx", "responses_4xx", "responses_5xx",
                   "local_429", "fallback", "retries", "gpu_tokens",
+                  "cache_hits", "cache_misses",
                   "input_tokens", "output_tokens", "total_tokens",
                   "bytes_in", "bytes_out"):
             lines.append(f'aigw_fast_{k}_total {s[k]}')

@@ -23,6 +23,7 @@
 #include <vector>
 
 #include "bpe_kernels.cuh"
+#include "cache_kernels.cuh"
 
 namespace aigw_fast {
 
@@ -58,6 +59,12 @@ __global__ void excl_scan_kernel(const int32_t* __restrict__ in, int n,
     __syncthreads();
   }
   if (threadIdx.x == 0) *total = carry;
+}
+
+__global__ void f32_to_bf16_kernel(const float* __restrict__ in,
+                                   bf16* __restrict__ out, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = (bf16)in[i];
 }
 
 #define HIP_OK(expr)                                                   \
@@ -125,6 +132,19 @@ class GpuAdmissionDirect {
   // Returns per-request token counts. Single caller thread.
   bool count(const char* bytes, size_t n, const int64_t* offsets, int n_req,
              int32_t* counts_out) {
+    if (!count_submit(bytes, n, offsets, n_req)) return false;
+    HIP_OK(hipMemcpyAsync(h_counts_, d_counts_, sizeof(int32_t) * n_req,
+                          hipMemcpyDeviceToHost, stream_));
+    HIP_OK(hipEventRecord(event_, stream_));
+    HIP_OK(hipEventSynchronize(event_));
+    memcpy(counts_out, h_counts_, sizeof(int32_t) * (size_t)n_req);
+    return true;
+  }
+
+  // Launch the BPE pipeline (H2D + segmentation + merge) without the
+  // D2H/sync tail; shared by count() and count_lookup() below.
+  bool count_submit(const char* bytes, size_t n, const int64_t* offsets,
+                    int n_req) {
     if (!ready_ || n == 0 || n_req == 0 || n > max_bytes_ || n_req > max_req_)
       return false;
     // launches follow the CALLER thread's current device; the batcher
@@ -167,12 +187,149 @@ class GpuAdmissionDirect {
                        d_totals_, (int)n, d_ghead_, d_totals_ + 1,
                        d_htab_keys_, d_htab_rank_, htab_mask_, d_out_ids_,
                        d_counts_);
+    return true;
+  }
+
+  // ---- semantic cache (native path) ---------------------------------------
+  //
+  // Same pipeline as aigw/ops/semcache.py embed/lookup, launched on the
+  // admission stream right after the BPE batch: meanpool(out_ids) ->
+  // bf16 cast -> MFMA projection GEMM -> l2norm -> fused cosine/argmax
+  // over the HBM-resident index. Query vectors are parked in a pending
+  // pool so a later insert (after the upstream 200) can append them to
+  // the index without keeping the batch buffers alive.
+
+  bool init_cache(const uint16_t* emb, int vocab, const uint16_t* proj,
+                  int dim, long long capacity, float threshold,
+                  int pending_cap) {
+    if (!ready_ || dim != 384) return false;  // one tuned topk instantiation
+    HIP_OK(hipSetDevice(device_));
+    dim_ = dim;
+    cap_ = capacity;
+    threshold_ = threshold;
+    vocab_ = vocab;
+    pending_cap_ = pending_cap;
+    HIP_OK(hipMalloc(&d_emb_, sizeof(bf16) * (size_t)vocab * dim));
+    HIP_OK(hipMemcpy(d_emb_, emb, sizeof(bf16) * (size_t)vocab * dim,
+                     hipMemcpyHostToDevice));
+    HIP_OK(hipMalloc(&d_proj_, sizeof(bf16) * (size_t)dim * dim));
+    HIP_OK(hipMemcpy(d_proj_, proj, sizeof(bf16) * (size_t)dim * dim,
+                     hipMemcpyHostToDevice));
+    HIP_OK(hipMalloc(&d_index_, sizeof(bf16) * (size_t)capacity * dim));
+    HIP_OK(hipMemset(d_index_, 0, sizeof(bf16) * (size_t)capacity * dim));
+    HIP_OK(hipMalloc(&d_pool_, sizeof(float) * (size_t)max_req_ * dim));
+    HIP_OK(hipMalloc(&d_poolcnt_, sizeof(int32_t) * max_req_));
+    HIP_OK(hipMalloc(&d_poolbf_, sizeof(bf16) * (size_t)max_req_ * dim));
+    HIP_OK(hipMalloc(&d_gout_, sizeof(float) * (size_t)max_req_ * dim));
+    HIP_OK(hipMalloc(&d_q_, sizeof(bf16) * (size_t)max_req_ * dim));
+    HIP_OK(hipMalloc(&d_best_, sizeof(unsigned long long) * max_req_));
+    HIP_OK(hipHostMalloc(&h_best_, sizeof(unsigned long long) * max_req_,
+                         hipHostMallocDefault));
+    HIP_OK(hipMalloc(&d_pending_, sizeof(bf16) * (size_t)pending_cap * dim));
+    HIP_OK(hipStreamCreateWithFlags(&insert_stream_, hipStreamNonBlocking));
+    cache_on_ = true;
+    return true;
+  }
+
+  bool cache_on() const { return cache_on_; }
+
+  // Extended batch op: counts + cache lookup. pending_slots[i] >= 0 parks
+  // request i's query vector in that pending-pool slot (caller-managed
+  // free list). rows_out[i] = best index row (-1 below threshold or empty
+  // index); scores_out[i] = cosine.
+  bool count_lookup(const char* bytes, size_t n, const int64_t* offsets,
+                    int n_req, int32_t* counts_out,
+                    const int32_t* pending_slots, int32_t* rows_out,
+                    float* scores_out) {
+    if (!cache_on_) return false;
+    if (!count_submit(bytes, n, offsets, n_req)) return false;
+    int dim = dim_;
+    HIP_OK(hipMemsetAsync(d_pool_, 0, sizeof(float) * (size_t)n_req * dim,
+                          stream_));
+    HIP_OK(hipMemsetAsync(d_poolcnt_, 0, sizeof(int32_t) * n_req, stream_));
+    constexpr int P = 8;
+    hipLaunchKernelGGL(meanpool_accum_kernel, dim3(n_req, P), dim3(dim), 0,
+                       stream_, d_out_ids_, d_off_, n_req, (int)n, d_emb_,
+                       dim, P, d_pool_, d_poolcnt_);
+    hipLaunchKernelGGL(meanpool_div_kernel, dim3(n_req), dim3(dim), 0, stream_,
+                       d_pool_, d_poolcnt_, n_req, dim);
+    int total = n_req * dim;
+    hipLaunchKernelGGL(f32_to_bf16_kernel, dim3((total + 255) / 256), dim3(256),
+                       0, stream_, d_pool_, d_poolbf_, total);
+    dim3 ggrid((n_req + 15) / 16, (dim + 63) / 64);
+    hipLaunchKernelGGL(gemm_bf16_nt_kernel, ggrid, dim3(256), 0, stream_,
+                       d_poolbf_, d_proj_, d_gout_, n_req, dim, dim, nullptr, 0);
+    hipLaunchKernelGGL(l2norm_rows_kernel, dim3(n_req), dim3(dim), 0, stream_,
+                       d_gout_, d_q_, n_req, dim);
+    // park query vectors for possible insert
+    for (int i = 0; i < n_req; ++i) {
+      if (pending_slots[i] >= 0 && pending_slots[i] < pending_cap_)
+        HIP_OK(hipMemcpyAsync(d_pending_ + (size_t)pending_slots[i] * dim,
+                              d_q_ + (size_t)i * dim, sizeof(bf16) * dim,
+                              hipMemcpyDeviceToDevice, stream_));
+    }
+    long long rows_now = rows_visible_;
+    if (rows_now > 0) {
+      HIP_OK(hipMemsetAsync(d_best_, 0, sizeof(unsigned long long) * n_req,
+                            stream_));
+      constexpr int ROWTILES = 16;
+      long long blocks = (rows_now + 64 * ROWTILES - 1) / (64 * ROWTILES);
+      constexpr size_t L = 128 * (12 * 32 + 8) * sizeof(bf16) + 128 * 8;
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&cache_topk_lds_kernel_t<12, ROWTILES>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+      for (int q0 = 0; q0 < n_req; q0 += 128) {
+        int kq = (n_req - q0) < 128 ? (n_req - q0) : 128;
+        hipLaunchKernelGGL((cache_topk_lds_kernel_t<12, ROWTILES>),
+                           dim3((unsigned)blocks), dim3(256), L, stream_,
+                           (const bf16*)d_index_, rows_now,
+                           d_q_ + (size_t)q0 * dim, kq, dim, d_best_ + q0);
+      }
+      HIP_OK(hipMemcpyAsync(h_best_, d_best_,
+                            sizeof(unsigned long long) * n_req,
+                            hipMemcpyDeviceToHost, stream_));
+    }
     HIP_OK(hipMemcpyAsync(h_counts_, d_counts_, sizeof(int32_t) * n_req,
                           hipMemcpyDeviceToHost, stream_));
     HIP_OK(hipEventRecord(event_, stream_));
     HIP_OK(hipEventSynchronize(event_));
     memcpy(counts_out, h_counts_, sizeof(int32_t) * (size_t)n_req);
+    for (int i = 0; i < n_req; ++i) {
+      rows_out[i] = -1;
+      scores_out[i] = 0.f;
+      if (rows_now > 0) {
+        unsigned long long u = h_best_[i];
+        unsigned hi = (unsigned)(u >> 32);
+        unsigned bits = (hi >= 0x80000000u) ? (hi ^ 0x80000000u) : ~hi;
+        float score;
+        memcpy(&score, &bits, 4);
+        if (score >= threshold_) {
+          rows_out[i] = (int32_t)(u & 0xFFFFFFFFu);
+          scores_out[i] = score;
+        }
+      }
+    }
     return true;
+  }
+
+  // Append a parked query vector to the index ring; returns the row.
+  // Called from connection threads (post-response) on a dedicated stream;
+  // host-side ring state is guarded by the caller's mutex. A topk racing
+  // a half-written row can at worst misscore THAT row; the row→value
+  // mapping stays correct because the value is stored first.
+  long long cache_insert(int pending_slot) {
+    if (!cache_on_ || pending_slot < 0 || pending_slot >= pending_cap_)
+      return -1;
+    long long row = head_;
+    head_ = (head_ + 1) % cap_;
+    hipError_t e = hipMemcpyAsync(
+        (bf16*)d_index_ + (size_t)row * dim_,
+        d_pending_ + (size_t)pending_slot * dim_, sizeof(bf16) * dim_,
+        hipMemcpyDeviceToDevice, insert_stream_);
+    if (e != hipSuccess) return -1;
+    if (row + 1 > rows_visible_) rows_visible_ = row + 1;
+    if (rows_visible_ > cap_) rows_visible_ = cap_;
+    return row;
   }
 
   ~GpuAdmissionDirect() {
@@ -182,6 +339,15 @@ class GpuAdmissionDirect {
     (void)hipHostFree(h_bytes_);
     (void)hipHostFree(h_off_);
     (void)hipHostFree(h_counts_);
+    if (cache_on_) {
+      (void)hipStreamDestroy(insert_stream_);
+      (void)hipHostFree(h_best_);
+      for (void* p : {(void*)d_emb_, (void*)d_proj_, (void*)d_index_,
+                      (void*)d_pool_, (void*)d_poolcnt_, (void*)d_poolbf_,
+                      (void*)d_gout_, (void*)d_q_, (void*)d_best_,
+                      (void*)d_pending_})
+        (void)hipFree(p);
+    }
     for (void* p : {(void*)d_htab_keys_, (void*)d_htab_rank_, (void*)d_bytes_,
                     (void*)d_off_, (void*)d_flags_, (void*)d_gflags_,
                     (void*)d_blk_, (void*)d_excl_, (void*)d_totals_,
@@ -194,7 +360,27 @@ class GpuAdmissionDirect {
 
  private:
   bool ready_ = false;
+  bool cache_on_ = false;
   int device_ = 0;
+  int dim_ = 0;
+  int vocab_ = 0;
+  int pending_cap_ = 0;
+  long long cap_ = 0;
+  long long head_ = 0;
+  long long rows_visible_ = 0;
+  float threshold_ = 0.f;
+  bf16* d_emb_ = nullptr;
+  bf16* d_proj_ = nullptr;
+  void* d_index_ = nullptr;
+  float* d_pool_ = nullptr;
+  int32_t* d_poolcnt_ = nullptr;
+  bf16* d_poolbf_ = nullptr;
+  float* d_gout_ = nullptr;
+  bf16* d_q_ = nullptr;
+  unsigned long long* d_best_ = nullptr;
+  unsigned long long* h_best_ = nullptr;
+  bf16* d_pending_ = nullptr;
+  hipStream_t insert_stream_{};
   size_t max_bytes_ = 0;
   int max_req_ = 0;
   int htab_mask_ = 0;
@@ -238,5 +424,23 @@ bool admission_count(GpuAdmissionDirect* a, const char* bytes, size_t n,
 }
 
 void admission_destroy(GpuAdmissionDirect* a) { delete a; }
+
+bool admission_init_cache(GpuAdmissionDirect* a, const uint16_t* emb, int vocab,
+                          const uint16_t* proj, int dim, long long capacity,
+                          float threshold, int pending_cap) {
+  return a->init_cache(emb, vocab, proj, dim, capacity, threshold, pending_cap);
+}
+
+bool admission_count_lookup(GpuAdmissionDirect* a, const char* bytes, size_t n,
+                            const int64_t* offsets, int n_req,
+                            int32_t* counts_out, const int32_t* pending_slots,
+                            int32_t* rows_out, float* scores_out) {
+  return a->count_lookup(bytes, n, offsets, n_req, counts_out, pending_slots,
+                         rows_out, scores_out);
+}
+
+long long admission_cache_insert(GpuAdmissionDirect* a, int pending_slot) {
+  return a->cache_insert(pending_slot);
+}
 
 }  // namespace aigw_fast

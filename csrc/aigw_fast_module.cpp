@@ -109,6 +109,25 @@ PYBIND11_MODULE(aigw_fast, m) {
           py::arg("max_batch") = 1024,
           py::arg("max_batch_bytes") = (size_t)48 * 1024 * 1024,
           py::arg("max_req") = 4096, py::arg("device") = 0)
+      .def(
+          "enable_gpu_direct_cache",
+          [](FastServer& s, py::buffer emb, py::buffer proj, int dim,
+             long long capacity, float threshold) {
+            py::buffer_info ei = emb.request();
+            py::buffer_info pi = proj.request();
+            if (ei.itemsize != 2 || pi.itemsize != 2)
+              throw std::runtime_error("emb/proj must be bf16 (uint16 view)");
+            int vocab = (int)(ei.size / dim);
+            if ((long long)vocab * dim != ei.size || pi.size != (long long)dim * dim)
+              throw std::runtime_error("emb/proj shape mismatch");
+            if (!s.enable_gpu_direct_cache(
+                    static_cast<const uint16_t*>(ei.ptr), vocab,
+                    static_cast<const uint16_t*>(pi.ptr), dim, capacity,
+                    threshold))
+              throw std::runtime_error("native cache init failed");
+          },
+          py::arg("emb"), py::arg("proj"), py::arg("dim") = 384,
+          py::arg("capacity") = 65536, py::arg("threshold") = 0.92)
       .def("start", &FastServer::start, py::arg("host"), py::arg("port"),
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &FastServer::stop, py::call_guard<py::gil_scoped_release>())
@@ -127,6 +146,8 @@ PYBIND11_MODULE(aigw_fast, m) {
         d["fallback"] = st.fallback.load();
         d["retries"] = st.retries.load();
         d["gpu_tokens"] = st.gpu_tokens.load();
+        d["cache_hits"] = st.cache_hits.load();
+        d["cache_misses"] = st.cache_misses.load();
         d["input_tokens"] = st.input_tokens.load();
         d["output_tokens"] = st.output_tokens.load();
         d["total_tokens"] = st.total_tokens.load();

@@ -636,6 +636,14 @@ GpuAdmissionDirect* admission_create(const long long* htab_keys,
 bool admission_count(GpuAdmissionDirect* a, const char* bytes, size_t n,
                      const int64_t* offsets, int n_req, int32_t* counts_out);
 void admission_destroy(GpuAdmissionDirect* a);
+bool admission_init_cache(GpuAdmissionDirect* a, const uint16_t* emb, int vocab,
+                          const uint16_t* proj, int dim, long long capacity,
+                          float threshold, int pending_cap);
+bool admission_count_lookup(GpuAdmissionDirect* a, const char* bytes, size_t n,
+                            const int64_t* offsets, int n_req,
+                            int32_t* counts_out, const int32_t* pending_slots,
+                            int32_t* rows_out, float* scores_out);
+long long admission_cache_insert(GpuAdmissionDirect* a, int pending_slot);
 
 // Adaptive batching with NO timer window: one batcher thread drains
 // whatever accumulated while the previous GPU batch ran — the kernel
@@ -668,8 +676,23 @@ class DirectGpuBatcher {
   ~DirectGpuBatcher() { stop(); }
 
   int64_t count_text(const std::string& text) {
-    if (stopping_) return 0;
-    auto w = std::make_shared<GpuAdmissionClient::Waiter>();
+    return count_lookup_text(text).tokens;
+  }
+
+  struct CacheLookup {
+    int64_t tokens = 0;
+    int32_t row = -1;      // index row of a hit (score >= threshold)
+    float score = 0.f;
+    int32_t slot = -1;     // pending-pool slot holding the query vector
+  };
+
+  // count + (when the cache is enabled) lookup in ONE GPU batch. The
+  // returned slot (if >= 0) must be passed to cache_insert_slot() or
+  // cache_release_slot() by the caller exactly once.
+  CacheLookup count_lookup_text(const std::string& text) {
+    CacheLookup out;
+    if (stopping_) return out;
+    auto w = std::make_shared<Waiter2>();
     {
       std::lock_guard<std::mutex> lk(mu_);
       q_texts_.push_back(text);
@@ -678,17 +701,70 @@ class DirectGpuBatcher {
     }
     std::unique_lock<std::mutex> lk(w->m);
     w->cv.wait_for(lk, std::chrono::seconds(30), [&] { return w->done; });
-    return w->count;
+    out.tokens = w->count;
+    out.row = w->row;
+    out.score = w->score;
+    out.slot = w->slot;
+    return out;
   }
+
+  bool init_cache(const uint16_t* emb, int vocab, const uint16_t* proj,
+                  int dim, long long capacity, float threshold) {
+    if (adm_ == nullptr) return false;
+    const int pending_cap = 4096;
+    if (!admission_init_cache(adm_, emb, vocab, proj, dim, capacity, threshold,
+                              pending_cap))
+      return false;
+    {
+      std::lock_guard<std::mutex> lk(slot_mu_);
+      free_slots_.resize(pending_cap);
+      for (int i = 0; i < pending_cap; ++i) free_slots_[i] = pending_cap - 1 - i;
+    }
+    cache_on_ = true;
+    return true;
+  }
+
+  bool cache_on() const { return cache_on_; }
+
+  // append the parked query vector to the index; frees the slot. The
+  // index-ring state is guarded by insert_mu_ (callers also hold the
+  // server's value-store lock for the row->value write).
+  long long cache_insert_slot(int slot) {
+    long long row;
+    {
+      std::lock_guard<std::mutex> lk(insert_mu_);
+      row = admission_cache_insert(adm_, slot);
+    }
+    cache_release_slot(slot);
+    return row;
+  }
+
+  void cache_release_slot(int slot) {
+    if (slot < 0) return;
+    std::lock_guard<std::mutex> lk(slot_mu_);
+    free_slots_.push_back(slot);
+  }
+
+  struct Waiter2 {
+    std::mutex m;
+    std::condition_variable cv;
+    bool done = false;
+    int64_t count = 0;
+    int32_t row = -1;
+    float score = 0.f;
+    int32_t slot = -1;
+  };
 
  private:
   void loop() {
     std::string packed;
     std::vector<int64_t> offs;
     std::vector<int32_t> counts;
+    std::vector<int32_t> slots, rows;
+    std::vector<float> scores;
     while (!stopping_) {
       std::vector<std::string> texts;
-      std::vector<std::shared_ptr<GpuAdmissionClient::Waiter>> waiters;
+      std::vector<std::shared_ptr<Waiter2>> waiters;
       {
         std::unique_lock<std::mutex> lk(mu_);
         cv_.wait(lk, [&] { return stopping_ || !q_texts_.empty(); });
@@ -710,16 +786,32 @@ class DirectGpuBatcher {
         size_t room = max_bytes_ - packed.size();
         packed.append(t.data(), std::min(t.size(), room));
       }
-      counts.assign(texts.size(), 0);
+      size_t nt = texts.size();
+      counts.assign(nt, 0);
+      rows.assign(nt, -1);
+      scores.assign(nt, 0.f);
+      slots.assign(nt, -1);
+      if (cache_on_) {
+        std::lock_guard<std::mutex> lk(slot_mu_);
+        for (size_t i = 0; i < nt && !free_slots_.empty(); ++i) {
+          slots[i] = free_slots_.back();
+          free_slots_.pop_back();
+        }
+      }
       int64_t bt0 = now_us();
       if (!packed.empty()) {
-        if (!admission_count(adm_, packed.data(), packed.size(), offs.data(),
-                             (int)texts.size(), counts.data()))
-          stats_errors++;
+        bool ok = cache_on_
+                      ? admission_count_lookup(adm_, packed.data(), packed.size(),
+                                               offs.data(), (int)nt,
+                                               counts.data(), slots.data(),
+                                               rows.data(), scores.data())
+                      : admission_count(adm_, packed.data(), packed.size(),
+                                        offs.data(), (int)nt, counts.data());
+        if (!ok) stats_errors++;
       }
       int64_t bt = now_us() - bt0;
       stats_batches++;
-      stats_texts += (uint64_t)texts.size();
+      stats_texts += (uint64_t)nt;
       stats_time_us += (uint64_t)bt;
       uint64_t prev = stats_max_us.load();
       while ((uint64_t)bt > prev && !stats_max_us.compare_exchange_weak(prev, (uint64_t)bt)) {}
@@ -727,6 +819,9 @@ class DirectGpuBatcher {
         auto& w = waiters[i];
         std::lock_guard<std::mutex> lk(w->m);
         w->count = counts[i];
+        w->row = rows[i];
+        w->score = scores[i];
+        w->slot = slots[i];
         w->done = true;
         w->cv.notify_all();
       }
@@ -734,7 +829,7 @@ class DirectGpuBatcher {
   }
 
   void fail_all() {
-    std::vector<std::shared_ptr<GpuAdmissionClient::Waiter>> all;
+    std::vector<std::shared_ptr<Waiter2>> all;
     {
       std::lock_guard<std::mutex> lk(mu_);
       all = std::move(q_waiters_);
@@ -749,6 +844,10 @@ class DirectGpuBatcher {
   }
 
   GpuAdmissionDirect* adm_ = nullptr;
+  std::atomic<bool> cache_on_{false};
+  std::mutex slot_mu_;
+  std::vector<int32_t> free_slots_;
+  std::mutex insert_mu_;
 
  public:
   std::atomic<uint64_t> stats_batches{0};
@@ -764,7 +863,7 @@ class DirectGpuBatcher {
   std::mutex mu_;
   std::condition_variable cv_;
   std::vector<std::string> q_texts_;
-  std::vector<std::shared_ptr<GpuAdmissionClient::Waiter>> q_waiters_;
+  std::vector<std::shared_ptr<Waiter2>> q_waiters_;
   std::thread worker_;
 };
 
@@ -907,8 +1006,64 @@ class ConnHandler {
     }
 
     int64_t gpu_tokens = 0;
-    if (srv_->gpu_enabled() && req.path == "/v1/chat/completions" &&
-        !sc.text.empty()) {
+    int32_t cache_slot = -1;
+    uint64_t cache_fp = 0;
+    bool cache_eligible = srv_->gpu_direct_ != nullptr &&
+                          srv_->gpu_direct_->cache_on() && !stream &&
+                          req.path == "/v1/chat/completions" &&
+                          !sc.text.empty();
+    if (cache_eligible) {
+      // scope fingerprint: near-identical prompts may embed to the same
+      // vector, so the VALUE carries a hash of everything that must not
+      // be shared — model, route, client credential, and every body byte
+      // OUTSIDE the messages array (sampling params etc.); compared at
+      // hit time (same rule as the Python cache's tag)
+      cache_fp = 1469598103934665603ull;  // FNV-1a 64
+      auto fnv = [&](const char* p, size_t n) {
+        for (size_t i = 0; i < n; ++i) {
+          cache_fp ^= (unsigned char)p[i];
+          cache_fp *= 1099511628211ull;
+        }
+      };
+      fnv(route->name.data(), route->name.size());
+      fnv(sc.model.data(), sc.model.size());
+      const std::string* auth_h = req.get("authorization");
+      if (auth_h) fnv(auth_h->data(), auth_h->size());
+      if (sc.msgs_ve > sc.msgs_vs && sc.msgs_ve <= body.size()) {
+        fnv(body.data(), sc.msgs_vs);
+        fnv(body.data() + sc.msgs_ve, body.size() - sc.msgs_ve);
+      } else {
+        fnv(body.data(), body.size());
+      }
+      auto cl = srv_->gpu_direct_->count_lookup_text(sc.text);
+      gpu_tokens = cl.tokens;
+      srv_->stats_.gpu_tokens += (uint64_t)gpu_tokens;
+      cache_slot = cl.slot;
+      if (cl.row >= 0) {
+        std::string hit;
+        {
+          std::lock_guard<std::mutex> lk(srv_->cache_mu_);
+          if ((size_t)cl.row < srv_->cache_values_.size())
+            hit = srv_->cache_values_[cl.row];
+        }
+        if (hit.size() > 9 &&
+            memcmp(hit.data(), &cache_fp, 8) == 0 && hit[8] == 'U') {
+          srv_->gpu_direct_->cache_release_slot(cache_slot);
+          srv_->stats_.cache_hits++;
+          srv_->stats_.responses_2xx++;
+          record_latency(t0);
+          std::string h =
+              "HTTP/1.1 200 X\r\ncontent-type: application/json\r\n"
+              "x-aigw-cache: hit\r\ncontent-length: " +
+              std::to_string(hit.size() - 9) + "\r\n\r\n";
+          std::string bodyv = hit.substr(9);
+          srv_->stats_.bytes_out += h.size() + bodyv.size();
+          return write_two(fd_, h, bodyv);
+        }
+      }
+      srv_->stats_.cache_misses++;
+    } else if (srv_->gpu_enabled() && req.path == "/v1/chat/completions" &&
+               !sc.text.empty()) {
       gpu_tokens = srv_->gpu_count(sc.text);
       srv_->stats_.gpu_tokens += (uint64_t)gpu_tokens;
     }
@@ -926,12 +1081,14 @@ class ConnHandler {
       if (attempts_left-- <= 0) break;
       if (!first) srv_->stats_.retries++;
       first = false;
-      int outcome = try_backend(req, body, sc, *route, *be, stream, gpu_tokens, t0);
+      int outcome = try_backend(req, body, sc, *route, *be, stream, gpu_tokens,
+                                t0, cache_slot, cache_fp);
       if (outcome == 0) return true;    // handled, keep-alive
       if (outcome == 2) return false;   // handled, close
       // outcome 1: retriable failure — next backend re-splices the
       // ORIGINAL body (per-try translation, A.8)
     }
+    if (cache_slot >= 0) srv_->gpu_direct_->cache_release_slot(cache_slot);
     srv_->stats_.responses_5xx++;
     return simple_reply(503, "upstream_error", "no healthy upstream", false);
   }
@@ -966,7 +1123,7 @@ class ConnHandler {
   int try_backend(const HttpHead& req, const std::string& body,
                   const aigw_core::Scan& sc, const FastRoute& route,
                   const FastBackend& be, bool stream, int64_t gpu_tokens,
-                  int64_t t0) {
+                  int64_t t0, int32_t cache_slot = -1, uint64_t cache_fp = 0) {
     // per-try body: splice from ORIGINAL bytes each attempt
     std::string out_body;
     const std::string* send_body = &body;
@@ -1057,7 +1214,8 @@ class ConnHandler {
         return 1;
       }
       srv_->pool_->release(be.host, be.port, ufd, !up_close && fr != Framing::kClose);
-      if (retriable) return 1;
+      if (retriable) return 1;  // caller keeps the cache slot for the next try
+      if (cache_slot >= 0) srv_->gpu_direct_->cache_release_slot(cache_slot);
       srv_->stats_.responses_4xx++;
       record_latency(t0);
       return forward_buffered(resp, err_body) ? 0 : 2;
@@ -1079,6 +1237,24 @@ class ConnHandler {
     if (!resp.get("content-encoding"))
       extract_usage(resp_body.data(), resp_body.size(), &u);
     finish_usage(u, gpu_tokens);
+    if (cache_slot >= 0) {
+      if (resp_body.size() < (2u << 20)) {
+        // value first, then the index row: a lookup racing the insert
+        // can misscore the half-written vector but never map a row to a
+        // stale value (same cap as the Python cache)
+        std::string v;
+        v.reserve(9 + resp_body.size());
+        v.append(reinterpret_cast<const char*>(&cache_fp), 8);
+        v.push_back('U');
+        v += resp_body;
+        std::lock_guard<std::mutex> lk(srv_->cache_mu_);
+        long long row = srv_->gpu_direct_->cache_insert_slot(cache_slot);
+        if (row >= 0 && (size_t)row < srv_->cache_values_.size())
+          srv_->cache_values_[(size_t)row] = std::move(v);
+      } else {
+        srv_->gpu_direct_->cache_release_slot(cache_slot);
+      }
+    }
     srv_->stats_.responses_2xx++;
     record_latency(t0);
     return forward_buffered(resp, resp_body) ? 0 : 2;
@@ -1606,6 +1782,16 @@ int64_t FastServer::gpu_count(const std::string& text) {
   if (gpu_direct_ != nullptr) return gpu_direct_->count_text(text);
   size_t i = (size_t)(gpu_rr_.fetch_add(1, std::memory_order_relaxed) % gpu_.size());
   return gpu_[i]->count_text(text);
+}
+
+bool FastServer::enable_gpu_direct_cache(const uint16_t* emb, int vocab,
+                                         const uint16_t* proj, int dim,
+                                         long long capacity, float threshold) {
+  if (gpu_direct_ == nullptr) return false;
+  if (!gpu_direct_->init_cache(emb, vocab, proj, dim, capacity, threshold))
+    return false;
+  cache_values_.assign((size_t)capacity, std::string());
+  return true;
 }
 
 void FastServer::enable_gpu_direct(const long long* htab_keys,

@@ -76,6 +76,8 @@ struct ServerStats {
   std::atomic<uint64_t> fallback{0};
   std::atomic<uint64_t> retries{0};
   std::atomic<uint64_t> gpu_tokens{0};
+  std::atomic<uint64_t> cache_hits{0};
+  std::atomic<uint64_t> cache_misses{0};
   std::atomic<uint64_t> input_tokens{0};
   std::atomic<uint64_t> output_tokens{0};
   std::atomic<uint64_t> total_tokens{0};
@@ -109,6 +111,11 @@ class FastServer {
   void enable_gpu_direct(const long long* htab_keys, const int32_t* htab_rank,
                          int htab_n, int max_batch, size_t max_batch_bytes,
                          int max_req, int device);
+  // native semantic cache (requires enable_gpu_direct first): embedding
+  // table + projection weights, bf16 index ring of `capacity` rows.
+  bool enable_gpu_direct_cache(const uint16_t* emb, int vocab,
+                               const uint16_t* proj, int dim,
+                               long long capacity, float threshold);
 
   // lifecycle
   int start(const std::string& host, uint16_t port);  // returns bound port
@@ -164,6 +171,8 @@ class FastServer {
   std::set<int> conn_fds_;
   std::mutex cred_mu_;
   std::map<std::string, FileCredState> file_creds_;
+  std::mutex cache_mu_;
+  std::vector<std::string> cache_values_;  // row -> fp(8) + mode(1) + body
 };
 
 // -------- native bench harness (see fastpath.cpp) ---------------------------

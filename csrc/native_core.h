@@ -22,6 +22,7 @@ struct Scan {
   bool ok = true;
   const char* base = nullptr;     // set to the body start to capture spans
   size_t model_vs = 0, model_ve = 0;  // "model" value span incl. quotes
+  size_t msgs_vs = 0, msgs_ve = 0;    // top-level "messages" value span
 
   void ws() {
     while (p < end && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
@@ -148,8 +149,10 @@ struct Scan {
     // elsewhere in the body never leak into admission/cache-key text —
     // keeps this fast path byte-identical with the strict extractor
     // (aigw/gpu/services.py extract_chat_text)
-    in_msgs = in_msgs || (depth == 1 && key == "messages");
+    bool msgs_root = depth == 1 && key == "messages";
+    in_msgs = in_msgs || msgs_root;
     ws();
+    if (msgs_root && base != nullptr) msgs_vs = (size_t)(p - base);
     if (p >= end) return fail();
     char c = *p;
     if (c == '{') {
@@ -163,7 +166,9 @@ struct Scan {
         ws();
         if (p >= end || *p != ':') return fail();
         ++p;
+        bool k_is_msgs = depth == 0 && k == "messages";
         if (!parse_value(depth + 1, k, false, in_msgs)) return false;
+        if (k_is_msgs && base != nullptr) msgs_ve = (size_t)(p - base);
         ws();
         if (p < end && *p == ',') { ++p; continue; }
         if (p < end && *p == '}') { ++p; return true; }

@@ -93,3 +93,91 @@ llmRequestCosts:
         up_srv.close()
 
     asyncio.run(run())
+
+
+def test_fast_front_native_semantic_cache():
+    """Native cache path: MFMA embed + fused top-k on the admission
+    stream, value store + scope fingerprint in the C++ server. A repeat
+    request must hit without touching the upstream; model, sampling
+    params, and credential changes must miss (same isolation contract as
+    tests/test_cache_scope.py for the Python cache)."""
+    import aigw_fast
+    import yaml
+
+    from aigw.extproc.fast_front import FastFront
+    from aigw.extproc.server import GatewayServer
+    from aigw.extproc.upstream_client import LeanClient
+    from aigw.filterapi import RuntimeConfig, load_config
+
+    body = json.dumps({"id": "r1", "object": "chat.completion", "model": "m",
+                       "choices": [{"index": 0,
+                                    "message": {"role": "assistant",
+                                                "content": "cached answer"},
+                                    "finish_reason": "stop"}],
+                       "usage": {"prompt_tokens": 4, "completion_tokens": 2,
+                                 "total_tokens": 6}}).encode()
+    canned = (b"HTTP/1.1 200 OK\r\ncontent-type: application/json\r\n"
+              b"content-length: %d\r\n\r\n" % len(body)) + body
+
+    async def run():
+        mock = aigw_fast.FastMock()
+        up_port = mock.start("127.0.0.1", canned.decode("latin1"))
+        cfg = load_config(yaml.safe_load(f"""
+routes:
+  - name: r
+    backends:
+      - name: b
+        schema: OpenAI
+        upstream: {{host: 127.0.0.1, port: {up_port}}}
+"""))
+        server = GatewayServer(RuntimeConfig(cfg))
+        front = FastFront(server, server.runtime, gpu_direct=True,
+                          gpu_cache=True, n_merges=8192,
+                          cache_threshold=0.95)
+        port = await front.start("127.0.0.1", 0)
+        client = LeanClient()
+
+        async def chat(content, model="m", auth="Bearer user-a", **params):
+            payload = {"model": model,
+                       "messages": [{"role": "user", "content": content}]}
+            payload.update(params)
+            r = await client.post(
+                host="127.0.0.1", port=port, tls=False,
+                path="/v1/chat/completions",
+                headers={"content-type": "application/json",
+                         "authorization": auth},
+                body=json.dumps(payload).encode(), timeout_s=60.0)
+            data = await r.read()
+            hit = r.headers.get("x-aigw-cache") == "hit"
+            r.release()
+            assert r.status == 200, data[:200]
+            return hit, data
+
+        text = "what is the airspeed velocity of an unladen swallow " * 20
+        hit, _ = await chat(text)
+        assert not hit  # cold
+        hit, data = await chat(text)
+        assert hit, front.stats()
+        assert json.loads(data)["choices"][0]["message"]["content"] == "cached answer"
+        served_before = mock.requests()
+        hit, _ = await chat(text)
+        assert hit and mock.requests() == served_before  # upstream untouched
+
+        # isolation: model / sampling params / credential all miss
+        hit, _ = await chat(text, model="m2")
+        assert not hit
+        hit, _ = await chat(text, temperature=0.9)
+        assert not hit
+        hit, _ = await chat(text, auth="Bearer user-b")
+        assert not hit
+        # and a hit again for the original scope
+        hit, _ = await chat(text)
+        assert hit
+
+        st = front.stats()
+        assert st["cache_hits"] >= 3 and st["cache_misses"] >= 4, st
+        await client.close()
+        await front.stop()
+        mock.stop()
+
+    asyncio.run(run())
