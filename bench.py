@@ -302,7 +302,8 @@ def loadgen_entry(args, gw_port, direct_port, ready, go, out_q):
     asyncio.run(amain())
 
 
-def _start_fast_front(args, upstream_ports, gpu_socket, gpu_direct=False):
+def _start_fast_front(args, upstream_ports, gpu_socket, gpu_direct=False,
+                      gpu_cache=False):
     """Run the C++ gateway (and its Python fallback app) on a dedicated
     thread/loop in the rank primary; returns (front, port)."""
     import threading
@@ -328,7 +329,7 @@ def _start_fast_front(args, upstream_ports, gpu_socket, gpu_direct=False):
     front = FastFront(
         server, server.runtime, gpu_socket=gpu_socket or "",
         gpu_window_us=int(args.gpu_window * 1000.0), gpu_max_batch=1024,
-        gpu_direct=gpu_direct,
+        gpu_direct=gpu_direct, gpu_cache=gpu_cache,
         gpu_device=int(os.environ.get("LOCAL_RANK", 0)),
     )
     done = threading.Event()
@@ -503,25 +504,37 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
     gpu_direct = use_gpu
     if gpu_direct:
         torch.cuda.set_device(local_rank)
-    front, gw_port = _start_fast_front(args, [up_port], None, gpu_direct)
+    cache_mode = use_gpu and args.cache_payloads > 0
+    front, gw_port = _start_fast_front(args, [up_port], None, gpu_direct,
+                                       gpu_cache=cache_mode)
 
-    payload = json.dumps(build_payload(args.tokens)).encode()
+    if cache_mode:
+        # a fixed pool of distinct payloads: after one cold pass the
+        # native cache serves (pool exhausted) hits from the HBM index
+        payloads = []
+        for i in range(args.cache_payloads):
+            pl = build_payload(args.tokens)
+            pl["messages"][1]["content"] = f"variant {i}: " + pl["messages"][1]["content"]
+            payloads.append(json.dumps(pl).encode())
+    else:
+        payloads = [json.dumps(build_payload(args.tokens)).encode()]
     workers = args.workers if args.workers > 0 else max(1, _cpu_quota() // 4)
     conns = args.batch * workers
     waves = max(args.waves, 1)
     path = "/v1/chat/completions"
 
     # warmup (gateway) then warm direct baseline
-    aigw_fast.run_load("127.0.0.1", gw_port, path, payload, conns,
-                       max(args.warmup, 1) * waves)
-    direct = aigw_fast.run_load("127.0.0.1", up_port, path, payload, conns, 5)
+    aigw_fast.run_load_pool("127.0.0.1", gw_port, path, payloads, conns,
+                            max(args.warmup, 1) * waves)
+    direct = aigw_fast.run_load_pool("127.0.0.1", up_port, path, payloads,
+                                     conns, 5)
     if use_gpu:
         torch.cuda.synchronize()
 
     barrier_sync()
     t0 = time.perf_counter()
-    res = aigw_fast.run_load("127.0.0.1", gw_port, path, payload, conns,
-                             args.steps * waves)
+    res = aigw_fast.run_load_pool("127.0.0.1", gw_port, path, payloads, conns,
+                                  args.steps * waves)
     if use_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
@@ -595,7 +608,7 @@ def main():
             torch.cuda.set_device(local_rank)
         torch.distributed.init_process_group(backend)
 
-    fast_mode = args.front == "fast" and not args.cache_payloads
+    fast_mode = args.front == "fast"
 
     def barrier_sync():
         if world > 1:
@@ -743,6 +756,11 @@ def main():
                 ),
                 "statesync_tick_us": round(statesync_tick_us, 1) if statesync_tick_us else None,
                 "semantic_cache_payload_pool": args.cache_payloads or None,
+                "fast_cache_stats": (
+                    {k: front.stats()[k] for k in ("cache_hits", "cache_misses")}
+                    if fast_mode and front is not None and args.cache_payloads
+                    else None
+                ),
             },
         }
         print(json.dumps(out))

@@ -42,6 +42,29 @@ PYBIND11_MODULE(aigw_fast, m) {
       py::arg("connections"), py::arg("per_conn"),
       "closed-loop native load generator (threads, keep-alive)");
 
+  m.def(
+      "run_load_pool",
+      [](const std::string& host, uint16_t port, const std::string& path,
+         py::list bodies, int connections, int per_conn) {
+        std::vector<std::string> pool;
+        for (auto b : bodies) pool.push_back(b.cast<py::bytes>());
+        LoadResult r;
+        {
+          py::gil_scoped_release release;
+          r = run_load_pool(host, port, path, pool, connections, per_conn);
+        }
+        py::dict d;
+        d["elapsed_s"] = r.elapsed_s;
+        d["completed"] = r.completed;
+        d["errors"] = r.errors;
+        d["p50_ms"] = r.p50_ms;
+        d["p99_ms"] = r.p99_ms;
+        return d;
+      },
+      py::arg("host"), py::arg("port"), py::arg("path"), py::arg("bodies"),
+      py::arg("connections"), py::arg("per_conn"),
+      "run_load with a cycling body pool (semantic-cache hit mixes)");
+
   py::class_<FastServer>(m, "FastServer")
       .def(py::init<>())
       .def(

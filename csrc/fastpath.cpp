@@ -1651,13 +1651,18 @@ void FastMock::serve(int fd) {
 }
 
 // closed-loop load: `connections` threads, each one keep-alive connection
-// issuing `per_conn` sequential POSTs of `body` to host:port/path.
-LoadResult run_load(const std::string& host, uint16_t port,
-                    const std::string& path, const std::string& body,
-                    int connections, int per_conn) {
-  std::string req = "POST " + path + " HTTP/1.1\r\nhost: bench\r\n"
-                    "content-type: application/json\r\ncontent-length: " +
-                    std::to_string(body.size()) + "\r\n\r\n" + body;
+// issuing `per_conn` sequential POSTs to host:port/path, cycling the
+// body pool (one body = fixed payload; many = semantic-cache hit mix).
+LoadResult run_load_pool(const std::string& host, uint16_t port,
+                         const std::string& path,
+                         const std::vector<std::string>& bodies,
+                         int connections, int per_conn) {
+  std::vector<std::string> reqs;
+  reqs.reserve(bodies.size());
+  for (const auto& body : bodies)
+    reqs.push_back("POST " + path + " HTTP/1.1\r\nhost: bench\r\n"
+                   "content-type: application/json\r\ncontent-length: " +
+                   std::to_string(body.size()) + "\r\n\r\n" + body);
   std::vector<std::thread> threads;
   std::vector<std::vector<int64_t>> lat(connections);
   std::atomic<uint64_t> errors{0};
@@ -1670,6 +1675,7 @@ LoadResult run_load(const std::string& host, uint16_t port,
       std::string buf;
       char tmp[65536];
       for (int i = 0; i < per_conn; ++i) {
+        const std::string& req = reqs[((size_t)i * connections + c) % reqs.size()];
         int64_t r0 = now_us();
         if (fd < 0) {
           fd = tcp_connect(host, port, 120.0);
@@ -1727,6 +1733,12 @@ LoadResult run_load(const std::string& host, uint16_t port,
                                       (size_t)((double)all.size() * 0.99))] / 1e3;
   }
   return out;
+}
+
+LoadResult run_load(const std::string& host, uint16_t port,
+                    const std::string& path, const std::string& body,
+                    int connections, int per_conn) {
+  return run_load_pool(host, port, path, {body}, connections, per_conn);
 }
 
 // -------- FastServer --------------------------------------------------------

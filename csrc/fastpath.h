@@ -202,5 +202,9 @@ struct LoadResult {
 LoadResult run_load(const std::string& host, uint16_t port,
                     const std::string& path, const std::string& body,
                     int connections, int per_conn);
+LoadResult run_load_pool(const std::string& host, uint16_t port,
+                         const std::string& path,
+                         const std::vector<std::string>& bodies,
+                         int connections, int per_conn);
 
 }  // namespace aigw_fast
