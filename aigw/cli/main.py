@@ -135,6 +135,8 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
         fast = FastFront(
             server, runtime,
             gpu_direct=bool(args.gpu and _torch.cuda.is_available()),
+            gpu_cache=bool(args.gpu and args.semantic_cache
+                           and _torch.cuda.is_available()),
         )
         port = await fast.start(args.host, port)
 
