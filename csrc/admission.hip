@@ -16,6 +16,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <atomic>
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
@@ -268,7 +269,7 @@ class GpuAdmissionDirect {
                               d_q_ + (size_t)i * dim, sizeof(bf16) * dim,
                               hipMemcpyDeviceToDevice, stream_));
     }
-    long long rows_now = rows_visible_;
+    long long rows_now = rows_visible_.load(std::memory_order_acquire);
     if (rows_now > 0) {
       HIP_OK(hipMemsetAsync(d_best_, 0, sizeof(unsigned long long) * n_req,
                             stream_));
@@ -327,8 +328,10 @@ class GpuAdmissionDirect {
         d_pending_ + (size_t)pending_slot * dim_, sizeof(bf16) * dim_,
         hipMemcpyDeviceToDevice, insert_stream_);
     if (e != hipSuccess) return -1;
-    if (row + 1 > rows_visible_) rows_visible_ = row + 1;
-    if (rows_visible_ > cap_) rows_visible_ = cap_;
+    long long vis = rows_visible_.load(std::memory_order_relaxed);
+    long long want = row + 1 > vis ? row + 1 : vis;
+    if (want > cap_) want = cap_;
+    rows_visible_.store(want, std::memory_order_release);
     return row;
   }
 
@@ -366,8 +369,8 @@ class GpuAdmissionDirect {
   int vocab_ = 0;
   int pending_cap_ = 0;
   long long cap_ = 0;
-  long long head_ = 0;
-  long long rows_visible_ = 0;
+  long long head_ = 0;  // writers hold the caller's insert mutex
+  std::atomic<long long> rows_visible_{0};  // read lock-free by the batcher
   float threshold_ = 0.f;
   bf16* d_emb_ = nullptr;
   bf16* d_proj_ = nullptr;
