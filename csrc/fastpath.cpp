@@ -676,7 +676,12 @@ class DirectGpuBatcher {
   ~DirectGpuBatcher() { stop(); }
 
   int64_t count_text(const std::string& text) {
-    return count_lookup_text(text).tokens;
+    CacheLookup cl = count_lookup_text(text);
+    // the batcher parks a query vector whenever the cache is on; a
+    // count-only caller (streamed requests) must return the slot or the
+    // 4096-slot pending pool starves after enough streams
+    cache_release_slot(cl.slot);
+    return cl.tokens;
   }
 
   struct CacheLookup {
