@@ -267,6 +267,19 @@ def _parse_costs(costs: list) -> list[LLMRequestCost]:
 
 
 def translate_crds(docs: list[dict]) -> Config:
+    # typed validation first (the kube-apiserver's CRD/CEL gate analogue,
+    # aigw/controller/crds.py): malformed bundles fail with field-level
+    # errors before any compilation
+    from aigw.controller.crds import CRDValidationError, validate_bundle
+
+    try:
+        validate_bundle(docs)
+    except CRDValidationError as e:
+        raise ConfigError(str(e)) from e
+    return _translate_crds_validated(docs)
+
+
+def _translate_crds_validated(docs: list[dict]) -> Config:
     bundle = _Bundle(docs)
     routes: list[Route] = []
     models: list[Model] = []
