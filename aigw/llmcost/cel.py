@@ -36,15 +36,20 @@ _INT_VARS = (
 )
 _STR_VARS = ("model", "backend", "route_name")
 _ALLOWED_NAMES = frozenset(_INT_VARS + _STR_VARS)
-_ALLOWED_CALLS = frozenset({"uint", "int", "double", "min", "max"})
+_ALLOWED_CALLS = frozenset({"uint", "int", "double", "min", "max", "size"})
+# CEL string member functions, evaluated on the three string variables
+# (cel-go string extensions the reference's env exposes)
+_ALLOWED_METHODS = frozenset({"startsWith", "endsWith", "contains", "matches"})
 
 _ALLOWED_NODES = (
     ast.Expression, ast.BinOp, ast.UnaryOp, ast.BoolOp, ast.Compare,
     ast.IfExp, ast.Call, ast.Name, ast.Load, ast.Constant,
+    ast.Attribute, ast.List, ast.Tuple,
     ast.Add, ast.Sub, ast.Mult, ast.Div, ast.Mod,
     ast.USub, ast.UAdd, ast.Not,
     ast.And, ast.Or,
     ast.Eq, ast.NotEq, ast.Lt, ast.LtE, ast.Gt, ast.GtE,
+    ast.In, ast.NotIn,
 )
 
 
@@ -179,10 +184,16 @@ def _validate(tree: ast.AST, expr: str) -> None:
                 f"disallowed construct {type(node).__name__} in cost expression {expr!r}"
             )
         if isinstance(node, ast.Call):
-            if not isinstance(node.func, ast.Name) or node.func.id not in _ALLOWED_CALLS:
+            fn_ok = (isinstance(node.func, ast.Name) and node.func.id in _ALLOWED_CALLS) or (
+                isinstance(node.func, ast.Attribute)
+                and node.func.attr in _ALLOWED_METHODS)
+            if not fn_ok:
                 raise CostExpressionError(f"disallowed call in cost expression {expr!r}")
             if node.keywords:
                 raise CostExpressionError("keyword arguments not allowed")
+        if isinstance(node, ast.Attribute) and node.attr not in _ALLOWED_METHODS:
+            raise CostExpressionError(
+                f"disallowed attribute {node.attr!r} in cost expression {expr!r}")
         if isinstance(node, ast.Name) and node.id not in _ALLOWED_NAMES | _ALLOWED_CALLS:
             raise CostExpressionError(
                 f"unknown variable {node.id!r} in cost expression {expr!r}"
@@ -191,7 +202,26 @@ def _validate(tree: ast.AST, expr: str) -> None:
             raise CostExpressionError(f"disallowed literal {node.value!r}")
 
 
-_CALL_ENV = {"uint": int, "int": int, "double": float, "min": min, "max": max}
+_CALL_ENV = {"uint": int, "int": int, "double": float, "min": min, "max": max,
+             "size": len}
+
+
+class _CelStr(str):
+    """str + the CEL member functions (cel-go spelling) so expressions
+    like model.startsWith('gpt-4') compile without exposing Python's
+    attribute surface (the AST whitelist rejects every other attr)."""
+
+    def startsWith(self, a):  # noqa: N802 - CEL spelling
+        return self.startswith(a)
+
+    def endsWith(self, a):  # noqa: N802
+        return self.endswith(a)
+
+    def contains(self, a):
+        return a in self
+
+    def matches(self, a):
+        return re.search(a, self) is not None
 
 
 class CostProgram:
@@ -214,7 +244,8 @@ class CostProgram:
 
     def evaluate(self, v: CostVars) -> int:
         env = dict(_CALL_ENV)
-        env.update(v.as_dict())
+        for k, val in v.as_dict().items():
+            env[k] = _CelStr(val) if isinstance(val, str) else val
         try:
             out = eval(self._code, {"__builtins__": {}}, env)  # noqa: S307 - AST whitelisted
         except ZeroDivisionError:

@@ -74,3 +74,27 @@ def test_negative_cost_rejected():
 def test_string_literal_with_operators():
     p = CostProgram("model == 'a && b ? c : d' ? 1 : 2")
     assert p.evaluate(V) == 2
+
+
+def test_cel_string_functions_and_membership():
+    """cel-go string extensions the reference env exposes: member
+    functions on the string vars, list membership, size()."""
+    v = CostVars(model="gpt-4o-mini", backend="openai", route_name="prod",
+                 input_tokens=100, output_tokens=10, total_tokens=110)
+    from aigw.llmcost.cel import evaluate
+
+    cases = {
+        "model.startsWith('gpt-4') ? input_tokens * 3 : input_tokens": 300,
+        "model.endsWith('mini') ? 1 : 2": 1,
+        "model.contains('4o') ? output_tokens : 0": 10,
+        "model.matches('gpt-[0-9]+o') ? 7 : 8": 7,
+        "model in ['gpt-4o-mini', 'gpt-4o'] ? 5 : 6": 5,
+        "backend in ['azure'] ? 1 : 0": 0,
+        "size(model) + size(route_name)": 11 + 4,
+    }
+    for expr, want in cases.items():
+        assert evaluate(expr, v) == want, expr
+    # attribute surface stays closed
+    for bad in ("model.__class__", "model.upper()", "model.join(['x'])"):
+        with pytest.raises(CostExpressionError):
+            evaluate(bad, v)
