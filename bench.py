@@ -698,18 +698,25 @@ def main():
         # per rank, timed. Deterministic tick count keeps the collective
         # sequence identical across ranks (no timer-driven ticks racing
         # the final all-reduce).
-        from aigw.filterapi import RuntimeConfig as _RC
-        from aigw.filterapi import load_config as _lc
         from aigw.parallel import StateSync
-        from aigw.ratelimit import RateLimiter
 
-        lim = RateLimiter(_RC(_lc(gateway_config(9))).rate_limits)
-        lim.check({})  # touch the bucket so deltas exist
-        sync = StateSync(lim)
-        sync.tick_sync()  # warm
+        if native and front is not None:
+            # sync the NATIVE front's real rate counters — the serving run
+            # just charged them with every timed request's token usage
+            from aigw.parallel import FastLimiterBridge
+
+            sync = StateSync(FastLimiterBridge(front.fast, ["node-token-budget"]))
+        else:
+            from aigw.filterapi import RuntimeConfig as _RC
+            from aigw.filterapi import load_config as _lc
+            from aigw.ratelimit import RateLimiter
+
+            lim = RateLimiter(_RC(_lc(gateway_config(9))).rate_limits)
+            lim.check({})  # touch the bucket so deltas exist
+            sync = StateSync(lim)
+        sync.tick_sync()  # warm (drains the serving run's pending deltas)
         ts0 = time.perf_counter()
         for _ in range(20):
-            lim.charge({}, {"llm_total_token": 17})
             sync.tick_sync()
         statesync_tick_us = (time.perf_counter() - ts0) / 20 * 1e6
         t = torch.tensor([elapsed], dtype=torch.float64)
