@@ -10,10 +10,11 @@ front's fallback. CPython is out of the per-request path.
 
 Eligibility is decided per route at composition time (a matched but
 ineligible route is relayed to the fallback per request):
-- backend schema is OpenAI (byte-level passthrough; other providers need
-  the Python translators);
-- backend auth is None or a static/file API key (SigV4 and vendor auth
-  flows stay in Python);
+- backend schema is OpenAI (byte-level passthrough) or AzureOpenAI
+  (deployments-path rewrite + api-key header); other providers need the
+  Python translators;
+- backend auth is None, a static/file API key, or an Azure API key
+  (SigV4 and vendor token flows stay in Python);
 - plain HTTP upstream (TLS upstreams stay in Python);
 - no endpoint picker, no body mutations, no tracing.
 Whole-config blockers (any present -> FastFrontUnsupported, callers fall
@@ -46,7 +47,8 @@ class FastFrontUnsupported(RuntimeError):
 
 
 def _backend_eligible(b) -> tuple[bool, Optional[dict]]:
-    if b.schema.name is not APISchemaName.OPENAI:
+    azure = b.schema.name is APISchemaName.AZURE_OPENAI
+    if b.schema.name is not APISchemaName.OPENAI and not azure:
         return False, None
     if b.upstream.tls or b.upstream.path_prefix:
         return False, None
@@ -64,6 +66,8 @@ def _backend_eligible(b) -> tuple[bool, Optional[dict]]:
         if kind == "api_key":
             bearer = auth.api_key
             api_key_file = auth.api_key_file
+        elif azure and kind == "azure_api_key":
+            bearer = auth.azure_api_key
         elif kind == "":
             pass
         else:
@@ -75,6 +79,8 @@ def _backend_eligible(b) -> tuple[bool, Optional[dict]]:
         "bearer": bearer,
         "api_key_file": api_key_file,
         "model_override": b.model_name_override or "",
+        "azure": azure,
+        "azure_api_version": b.schema.version or "",
         "weight": float(b.weight),
         "priority": int(b.priority),
         "timeout_s": float(b.timeout_s),

@@ -91,6 +91,9 @@ PYBIND11_MODULE(aigw_fast, m) {
               if (d.contains("weight")) b.weight = d["weight"].cast<double>();
               if (d.contains("priority")) b.priority = d["priority"].cast<int>();
               if (d.contains("timeout_s")) b.timeout_s = d["timeout_s"].cast<double>();
+              if (d.contains("azure")) b.azure = d["azure"].cast<bool>();
+              if (d.contains("azure_api_version"))
+                b.azure_api_version = d["azure_api_version"].cast<std::string>();
               r.backends.push_back(std::move(b));
             }
             s.add_route(std::move(r));

@@ -1144,7 +1144,22 @@ class ConnHandler {
     std::string head;
     head.reserve(512 + req.head_len);
     head += "POST ";
-    head += req.path;
+    if (be.azure) {
+      // deployments-API rewrite; the deployment name is the effective
+      // model (override wins), suffix derived from the OpenAI path
+      const std::string& model_eff =
+          be.model_override.empty() ? sc.model : be.model_override;
+      const char* suffix = req.path == "/v1/embeddings" ? "embeddings"
+                           : req.path == "/v1/completions"
+                               ? "completions"
+                               : "chat/completions";
+      head += "/openai/deployments/" + model_eff + "/" + suffix +
+              "?api-version=" +
+              (be.azure_api_version.empty() ? "2025-01-01-preview"
+                                            : be.azure_api_version);
+    } else {
+      head += req.path;
+    }
     head += " HTTP/1.1\r\nhost: ";
     head += be.host;
     if (be.port != 80 && be.port != 443) head += ":" + std::to_string(be.port);
@@ -1157,7 +1172,11 @@ class ConnHandler {
       head += h.value;
       head += "\r\n";
     }
-    if (!be.bearer.empty()) {
+    if (be.azure && (!be.bearer.empty() || !be.api_key_file.empty())) {
+      head += "api-key: ";
+      head += srv_->resolve_bearer(be);
+      head += "\r\n";
+    } else if (!be.bearer.empty() || !be.api_key_file.empty()) {
       head += "authorization: Bearer ";
       head += srv_->resolve_bearer(be);
       head += "\r\n";

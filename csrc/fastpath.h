@@ -41,6 +41,11 @@ struct FastBackend {
   std::string bearer;          // static API-key credential; empty = none
   std::string api_key_file;    // rotated-credential file (mtime-cached)
   std::string model_override;  // modelNameOverride, spliced into the body
+  // AzureOpenAI deployments-API rewrite (openai_azureopenai.go): the
+  // request path becomes /openai/deployments/<model>/<suffix>?api-version=
+  // and the credential header is api-key instead of a bearer
+  bool azure = false;
+  std::string azure_api_version;
   double weight = 1.0;
   int priority = 0;
   double timeout_s = 60.0;

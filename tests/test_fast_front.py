@@ -344,3 +344,49 @@ def test_cli_run_fast_front():
         proc.wait(timeout=15)
         mock.stop()
         os.unlink(cfg_path)
+
+
+def test_fast_front_azure_deployments_rewrite():
+    """AzureOpenAI backends serve natively: deployments-path rewrite with
+    the effective model, api-version from the schema version, api-key
+    header (openai_azureopenai.go parity)."""
+
+    async def run():
+        mock, up_runner, up_port = await start_mock_upstream()
+        mock.record = True
+        cfg = load_config({
+            "routes": [{
+                "name": "az",
+                "headers": [{"name": "x-ai-eg-model", "value": "gpt-4o"}],
+                "backends": [{
+                    "name": "azure", "schema": {"name": "AzureOpenAI",
+                                                "version": "2024-10-21"},
+                    "upstream": {"host": "127.0.0.1", "port": up_port,
+                                 "pathPrefix": ""},
+                    "modelNameOverride": "my-deployment",
+                    "auth": {"azureApiKey": "az-key"},
+                }],
+            }],
+        })
+        server = GatewayServer(RuntimeConfig(cfg))
+        front = FastFront(server, server.runtime)
+        port = await front.start("127.0.0.1", 0)
+        async with aiohttp.ClientSession() as c:
+            async with c.post(
+                f"http://127.0.0.1:{port}/v1/chat/completions",
+                json={"model": "gpt-4o",
+                      "messages": [{"role": "user", "content": "hi"}]},
+            ) as r:
+                assert r.status == 200, await r.text()
+        rec = mock.requests[-1]
+        assert rec["path"] == ("/openai/deployments/my-deployment/"
+                               "chat/completions?api-version=2024-10-21"), rec["path"]
+        hdrs = {k.lower(): v for k, v in rec["headers"].items()}
+        assert hdrs["api-key"] == "az-key"
+        assert "authorization" not in hdrs
+        assert json.loads(rec["body"])["model"] == "my-deployment"
+        assert front.stats()["fallback"] == 0
+        await front.stop()
+        await up_runner.cleanup()
+
+    asyncio.run(run())
