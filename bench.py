@@ -522,7 +522,7 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
     # measured single-rank sweet spot (quota/4 workers) by the world size
     # so an 8-rank scale run does not earn CFS throttle stalls
     workers = args.workers if args.workers > 0 else max(
-        1, _cpu_quota() // (4 * max(world, 1)))
+        1, min(8, _cpu_quota() // (4 * max(world, 1))))
     conns = args.batch * workers
     waves = max(args.waves, 1)
     path = "/v1/chat/completions"
