@@ -148,6 +148,7 @@ class FastFront:
                  gpu_direct: bool = False, gpu_device: int = 0,
                  gpu_cache: bool = False, cache_capacity: int = 65536,
                  cache_threshold: float = 0.92, cache_seed: int = 7,
+                 cache_index_dtype: str = "bf16",
                  n_merges: int = 32768, tokenizer_seed: int = 1355):
         # `server` is the Python GatewayServer used for cold paths
         self.py_server = server
@@ -162,6 +163,7 @@ class FastFront:
         self.cache_capacity = cache_capacity
         self.cache_threshold = cache_threshold
         self.cache_seed = cache_seed
+        self.cache_index_dtype = cache_index_dtype
         self.n_merges = n_merges
         self.tokenizer_seed = tokenizer_seed
         self._fallback_runner = None
@@ -201,7 +203,8 @@ class FastFront:
                 self.fast.enable_gpu_direct_cache(
                     emb.view(_t.uint16).numpy(), proj.view(_t.uint16).numpy(),
                     dim=dim, capacity=self.cache_capacity,
-                    threshold=self.cache_threshold)
+                    threshold=self.cache_threshold,
+                    fp8=self.cache_index_dtype == "fp8")
         sockets = ([self.gpu_socket] if isinstance(self.gpu_socket, str)
                    else list(self.gpu_socket or []))
         for s in sockets:

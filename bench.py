@@ -330,6 +330,7 @@ def _start_fast_front(args, upstream_ports, gpu_socket, gpu_direct=False,
         server, server.runtime, gpu_socket=gpu_socket or "",
         gpu_window_us=int(args.gpu_window * 1000.0), gpu_max_batch=1024,
         gpu_direct=gpu_direct, gpu_cache=gpu_cache,
+        cache_index_dtype="fp8" if getattr(args, "cache_fp8", False) else "bf16",
         gpu_device=int(os.environ.get("LOCAL_RANK", 0)),
     )
     done = threading.Event()

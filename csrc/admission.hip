@@ -68,6 +68,20 @@ __global__ void f32_to_bf16_kernel(const float* __restrict__ in,
   if (i < n) out[i] = (bf16)in[i];
 }
 
+// bf16 -> OCP e4m3 via the gfx950 packed-convert instruction (the same
+// format torch.float8_e4m3fn uses on this architecture, so the fp8
+// index is interchangeable with the Python cache's)
+__global__ void bf16_to_fp8_kernel(const bf16* __restrict__ in,
+                                   uint8_t* __restrict__ out, int n) {
+  int i = (blockIdx.x * blockDim.x + threadIdx.x) * 2;
+  if (i >= n) return;
+  float a = (float)in[i];
+  float b = (i + 1 < n) ? (float)in[i + 1] : 0.f;
+  int packed = __builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0, false);
+  out[i] = (uint8_t)(packed & 0xff);
+  if (i + 1 < n) out[i + 1] = (uint8_t)((packed >> 8) & 0xff);
+}
+
 #define HIP_OK(expr)                                                   \
   do {                                                                 \
     hipError_t _e = (expr);                                            \
@@ -202,7 +216,7 @@ class GpuAdmissionDirect {
 
   bool init_cache(const uint16_t* emb, int vocab, const uint16_t* proj,
                   int dim, long long capacity, float threshold,
-                  int pending_cap) {
+                  int pending_cap, bool fp8) {
     if (!ready_ || dim != 384) return false;  // one tuned topk instantiation
     HIP_OK(hipSetDevice(device_));
     dim_ = dim;
@@ -210,19 +224,22 @@ class GpuAdmissionDirect {
     threshold_ = threshold;
     vocab_ = vocab;
     pending_cap_ = pending_cap;
+    fp8_ = fp8;
     HIP_OK(hipMalloc(&d_emb_, sizeof(bf16) * (size_t)vocab * dim));
     HIP_OK(hipMemcpy(d_emb_, emb, sizeof(bf16) * (size_t)vocab * dim,
                      hipMemcpyHostToDevice));
     HIP_OK(hipMalloc(&d_proj_, sizeof(bf16) * (size_t)dim * dim));
     HIP_OK(hipMemcpy(d_proj_, proj, sizeof(bf16) * (size_t)dim * dim,
                      hipMemcpyHostToDevice));
-    HIP_OK(hipMalloc(&d_index_, sizeof(bf16) * (size_t)capacity * dim));
-    HIP_OK(hipMemset(d_index_, 0, sizeof(bf16) * (size_t)capacity * dim));
+    size_t elt = fp8 ? 1 : sizeof(bf16);
+    HIP_OK(hipMalloc(&d_index_, elt * (size_t)capacity * dim));
+    HIP_OK(hipMemset(d_index_, 0, elt * (size_t)capacity * dim));
     HIP_OK(hipMalloc(&d_pool_, sizeof(float) * (size_t)max_req_ * dim));
     HIP_OK(hipMalloc(&d_poolcnt_, sizeof(int32_t) * max_req_));
     HIP_OK(hipMalloc(&d_poolbf_, sizeof(bf16) * (size_t)max_req_ * dim));
     HIP_OK(hipMalloc(&d_gout_, sizeof(float) * (size_t)max_req_ * dim));
     HIP_OK(hipMalloc(&d_q_, sizeof(bf16) * (size_t)max_req_ * dim));
+    if (fp8) HIP_OK(hipMalloc(&d_q8_, (size_t)max_req_ * dim));
     HIP_OK(hipMalloc(&d_best_, sizeof(unsigned long long) * max_req_));
     HIP_OK(hipHostMalloc(&h_best_, sizeof(unsigned long long) * max_req_,
                          hipHostMallocDefault));
@@ -275,16 +292,36 @@ class GpuAdmissionDirect {
                             stream_));
       constexpr int ROWTILES = 16;
       long long blocks = (rows_now + 64 * ROWTILES - 1) / (64 * ROWTILES);
-      constexpr size_t L = 128 * (12 * 32 + 8) * sizeof(bf16) + 128 * 8;
-      (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&cache_topk_lds_kernel_t<12, ROWTILES>),
-          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-      for (int q0 = 0; q0 < n_req; q0 += 128) {
-        int kq = (n_req - q0) < 128 ? (n_req - q0) : 128;
-        hipLaunchKernelGGL((cache_topk_lds_kernel_t<12, ROWTILES>),
-                           dim3((unsigned)blocks), dim3(256), L, stream_,
-                           (const bf16*)d_index_, rows_now,
-                           d_q_ + (size_t)q0 * dim, kq, dim, d_best_ + q0);
+      if (fp8_) {
+        // quantize the query block once; the fp8 index streams HBM at
+        // half the bytes per row (2x rows in the 288 GB budget)
+        hipLaunchKernelGGL(bf16_to_fp8_kernel,
+                           dim3((n_req * dim / 2 + 255) / 256), dim3(256), 0,
+                           stream_, d_q_, d_q8_, n_req * dim);
+        constexpr size_t L8 = 128 * (12 * 32 + 16) + 128 * 8;
+        (void)hipFuncSetAttribute(
+            reinterpret_cast<const void*>(
+                &cache_topk_fp8_lds_kernel_t<12, ROWTILES>),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        for (int q0 = 0; q0 < n_req; q0 += 128) {
+          int kq = (n_req - q0) < 128 ? (n_req - q0) : 128;
+          hipLaunchKernelGGL((cache_topk_fp8_lds_kernel_t<12, ROWTILES>),
+                             dim3((unsigned)blocks), dim3(256), L8, stream_,
+                             (const uint8_t*)d_index_, rows_now,
+                             d_q8_ + (size_t)q0 * dim, kq, dim, d_best_ + q0);
+        }
+      } else {
+        constexpr size_t L = 128 * (12 * 32 + 8) * sizeof(bf16) + 128 * 8;
+        (void)hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&cache_topk_lds_kernel_t<12, ROWTILES>),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        for (int q0 = 0; q0 < n_req; q0 += 128) {
+          int kq = (n_req - q0) < 128 ? (n_req - q0) : 128;
+          hipLaunchKernelGGL((cache_topk_lds_kernel_t<12, ROWTILES>),
+                             dim3((unsigned)blocks), dim3(256), L, stream_,
+                             (const bf16*)d_index_, rows_now,
+                             d_q_ + (size_t)q0 * dim, kq, dim, d_best_ + q0);
+        }
       }
       HIP_OK(hipMemcpyAsync(h_best_, d_best_,
                             sizeof(unsigned long long) * n_req,
@@ -323,11 +360,19 @@ class GpuAdmissionDirect {
       return -1;
     long long row = head_;
     head_ = (head_ + 1) % cap_;
-    hipError_t e = hipMemcpyAsync(
-        (bf16*)d_index_ + (size_t)row * dim_,
-        d_pending_ + (size_t)pending_slot * dim_, sizeof(bf16) * dim_,
-        hipMemcpyDeviceToDevice, insert_stream_);
-    if (e != hipSuccess) return -1;
+    if (fp8_) {
+      hipLaunchKernelGGL(bf16_to_fp8_kernel, dim3((dim_ / 2 + 255) / 256),
+                         dim3(256), 0, insert_stream_,
+                         d_pending_ + (size_t)pending_slot * dim_,
+                         (uint8_t*)d_index_ + (size_t)row * dim_, dim_);
+      if (hipGetLastError() != hipSuccess) return -1;
+    } else {
+      hipError_t e = hipMemcpyAsync(
+          (bf16*)d_index_ + (size_t)row * dim_,
+          d_pending_ + (size_t)pending_slot * dim_, sizeof(bf16) * dim_,
+          hipMemcpyDeviceToDevice, insert_stream_);
+      if (e != hipSuccess) return -1;
+    }
     long long vis = rows_visible_.load(std::memory_order_relaxed);
     long long want = row + 1 > vis ? row + 1 : vis;
     if (want > cap_) want = cap_;
@@ -347,8 +392,8 @@ class GpuAdmissionDirect {
       (void)hipHostFree(h_best_);
       for (void* p : {(void*)d_emb_, (void*)d_proj_, (void*)d_index_,
                       (void*)d_pool_, (void*)d_poolcnt_, (void*)d_poolbf_,
-                      (void*)d_gout_, (void*)d_q_, (void*)d_best_,
-                      (void*)d_pending_})
+                      (void*)d_gout_, (void*)d_q_, (void*)d_q8_,
+                      (void*)d_best_, (void*)d_pending_})
         (void)hipFree(p);
     }
     for (void* p : {(void*)d_htab_keys_, (void*)d_htab_rank_, (void*)d_bytes_,
@@ -380,6 +425,8 @@ class GpuAdmissionDirect {
   bf16* d_poolbf_ = nullptr;
   float* d_gout_ = nullptr;
   bf16* d_q_ = nullptr;
+  uint8_t* d_q8_ = nullptr;
+  bool fp8_ = false;
   unsigned long long* d_best_ = nullptr;
   unsigned long long* h_best_ = nullptr;
   bf16* d_pending_ = nullptr;
@@ -430,8 +477,9 @@ void admission_destroy(GpuAdmissionDirect* a) { delete a; }
 
 bool admission_init_cache(GpuAdmissionDirect* a, const uint16_t* emb, int vocab,
                           const uint16_t* proj, int dim, long long capacity,
-                          float threshold, int pending_cap) {
-  return a->init_cache(emb, vocab, proj, dim, capacity, threshold, pending_cap);
+                          float threshold, int pending_cap, bool fp8) {
+  return a->init_cache(emb, vocab, proj, dim, capacity, threshold, pending_cap,
+                       fp8);
 }
 
 bool admission_count_lookup(GpuAdmissionDirect* a, const char* bytes, size_t n,

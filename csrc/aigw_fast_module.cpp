@@ -138,7 +138,7 @@ PYBIND11_MODULE(aigw_fast, m) {
       .def(
           "enable_gpu_direct_cache",
           [](FastServer& s, py::buffer emb, py::buffer proj, int dim,
-             long long capacity, float threshold) {
+             long long capacity, float threshold, bool fp8) {
             py::buffer_info ei = emb.request();
             py::buffer_info pi = proj.request();
             if (ei.itemsize != 2 || pi.itemsize != 2)
@@ -149,11 +149,12 @@ PYBIND11_MODULE(aigw_fast, m) {
             if (!s.enable_gpu_direct_cache(
                     static_cast<const uint16_t*>(ei.ptr), vocab,
                     static_cast<const uint16_t*>(pi.ptr), dim, capacity,
-                    threshold))
+                    threshold, fp8))
               throw std::runtime_error("native cache init failed");
           },
           py::arg("emb"), py::arg("proj"), py::arg("dim") = 384,
-          py::arg("capacity") = 65536, py::arg("threshold") = 0.92)
+          py::arg("capacity") = 65536, py::arg("threshold") = 0.92,
+          py::arg("fp8") = false)
       .def("start", &FastServer::start, py::arg("host"), py::arg("port"),
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &FastServer::stop, py::call_guard<py::gil_scoped_release>())

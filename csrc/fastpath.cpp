@@ -638,7 +638,7 @@ bool admission_count(GpuAdmissionDirect* a, const char* bytes, size_t n,
 void admission_destroy(GpuAdmissionDirect* a);
 bool admission_init_cache(GpuAdmissionDirect* a, const uint16_t* emb, int vocab,
                           const uint16_t* proj, int dim, long long capacity,
-                          float threshold, int pending_cap);
+                          float threshold, int pending_cap, bool fp8);
 bool admission_count_lookup(GpuAdmissionDirect* a, const char* bytes, size_t n,
                             const int64_t* offsets, int n_req,
                             int32_t* counts_out, const int32_t* pending_slots,
@@ -714,11 +714,11 @@ class DirectGpuBatcher {
   }
 
   bool init_cache(const uint16_t* emb, int vocab, const uint16_t* proj,
-                  int dim, long long capacity, float threshold) {
+                  int dim, long long capacity, float threshold, bool fp8) {
     if (adm_ == nullptr) return false;
     const int pending_cap = 4096;
     if (!admission_init_cache(adm_, emb, vocab, proj, dim, capacity, threshold,
-                              pending_cap))
+                              pending_cap, fp8))
       return false;
     {
       std::lock_guard<std::mutex> lk(slot_mu_);
@@ -1822,9 +1822,10 @@ int64_t FastServer::gpu_count(const std::string& text) {
 
 bool FastServer::enable_gpu_direct_cache(const uint16_t* emb, int vocab,
                                          const uint16_t* proj, int dim,
-                                         long long capacity, float threshold) {
+                                         long long capacity, float threshold,
+                                         bool fp8) {
   if (gpu_direct_ == nullptr) return false;
-  if (!gpu_direct_->init_cache(emb, vocab, proj, dim, capacity, threshold))
+  if (!gpu_direct_->init_cache(emb, vocab, proj, dim, capacity, threshold, fp8))
     return false;
   cache_values_.assign((size_t)capacity, std::string());
   return true;

@@ -120,7 +120,8 @@ class FastServer {
   // table + projection weights, bf16 index ring of `capacity` rows.
   bool enable_gpu_direct_cache(const uint16_t* emb, int vocab,
                                const uint16_t* proj, int dim,
-                               long long capacity, float threshold);
+                               long long capacity, float threshold,
+                               bool fp8);
 
   // lifecycle
   int start(const std::string& host, uint16_t port);  // returns bound port

@@ -95,7 +95,8 @@ llmRequestCosts:
     asyncio.run(run())
 
 
-def test_fast_front_native_semantic_cache():
+@pytest.mark.parametrize("index_dtype", ["bf16", "fp8"])
+def test_fast_front_native_semantic_cache(index_dtype):
     """Native cache path: MFMA embed + fused top-k on the admission
     stream, value store + scope fingerprint in the C++ server. A repeat
     request must hit without touching the upstream; model, sampling
@@ -133,7 +134,8 @@ routes:
         server = GatewayServer(RuntimeConfig(cfg))
         front = FastFront(server, server.runtime, gpu_direct=True,
                           gpu_cache=True, n_merges=8192,
-                          cache_threshold=0.95)
+                          cache_threshold=0.95,
+                          cache_index_dtype=index_dtype)
         port = await front.start("127.0.0.1", 0)
         client = LeanClient()
 
