@@ -107,8 +107,14 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
             await sync.start()
 
     watcher = None
+    reload_cbs = [server.swap_runtime]
+
+    def _on_reload(rc):
+        for cb in reload_cbs:
+            cb(rc)
+
     if watch_path:
-        watcher = ConfigWatcher(watch_path, server.swap_runtime)
+        watcher = ConfigWatcher(watch_path, _on_reload)
         await watcher.start()
 
     # OIDC/token-exchange credential rotation (BSP rotator analogue):
@@ -139,9 +145,11 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
                            and _torch.cuda.is_available()),
         )
         port = await fast.start(args.host, port)
+        reload_cbs.append(fast.reload)
 
         async def _fast_cleanup():
-            await fast.stop()
+            await fast.stop(
+                drain_s=float(os.environ.get("AIGW_DRAIN_TIMEOUT", "30")))
 
         front_cleanup = _fast_cleanup
     elif front_choice == "aiohttp":

@@ -87,18 +87,11 @@ def _backend_eligible(b) -> tuple[bool, Optional[dict]]:
     }
 
 
-def build_fast_server(runtime: RuntimeConfig):
-    """Compose a FastServer from a RuntimeConfig; raises
-    FastFrontUnsupported when the config as a whole cannot be served."""
-    if _fast is None:
-        raise FastFrontUnsupported("aigw_fast extension not built")
-    cfg = runtime.config
-    for rule in cfg.rate_limits:
-        if rule.key_headers:
-            raise FastFrontUnsupported(
-                f"rate rule {rule.name!r} uses key headers (python fronts only)"
-            )
-    srv = _fast.FastServer()
+def route_specs(runtime: RuntimeConfig) -> list[dict]:
+    """Native route-table descriptors for a RuntimeConfig (shared by
+    initial composition and hot reload); raises FastFrontUnsupported on
+    whole-config blockers."""
+    specs: list[dict] = []
     for cr in runtime.routes:
         model_match = ""
         for m in cr.matches:
@@ -129,10 +122,30 @@ def build_fast_server(runtime: RuntimeConfig):
             for d in backends:
                 if not d["model_override"]:
                     d["model_override"] = cr.route.model_name_override
-        srv.add_route(
-            cr.route.name, model_match, int(cr.route.retries), has_costs,
-            eligible, backends if eligible else [],
-        )
+        specs.append({
+            "name": cr.route.name, "model_match": model_match,
+            "retries": int(cr.route.retries), "has_costs": has_costs,
+            "eligible": eligible,
+            "backends": backends if eligible else [],
+        })
+    return specs
+
+
+def build_fast_server(runtime: RuntimeConfig):
+    """Compose a FastServer from a RuntimeConfig; raises
+    FastFrontUnsupported when the config as a whole cannot be served."""
+    if _fast is None:
+        raise FastFrontUnsupported("aigw_fast extension not built")
+    cfg = runtime.config
+    for rule in cfg.rate_limits:
+        if rule.key_headers:
+            raise FastFrontUnsupported(
+                f"rate rule {rule.name!r} uses key headers (python fronts only)"
+            )
+    srv = _fast.FastServer()
+    for spec in route_specs(runtime):
+        srv.add_route(spec["name"], spec["model_match"], spec["retries"],
+                      spec["has_costs"], spec["eligible"], spec["backends"])
     for rule in cfg.rate_limits:
         srv.add_rate_rule(rule.name, int(rule.limit), float(rule.window_s),
                           rule.metadata_key)
@@ -215,6 +228,15 @@ class FastFront:
                     self.port, fallback_port)
         return self.port
 
+    def reload(self, runtime: RuntimeConfig) -> None:
+        """Hot reload: swap the native route table and the Python
+        fallback's runtime; in-flight requests keep the tables they
+        resolved (watcher.go swap semantics). Rate rules keep their
+        windows — changing rule SHAPES needs a restart (documented)."""
+        self.fast.swap_routes(route_specs(runtime))
+        self.py_server.swap_runtime(runtime)
+        self.runtime = runtime
+
     def stats(self) -> dict:
         return self.fast.stats()
 
@@ -238,7 +260,13 @@ class FastFront:
                 f'aigw_fast_latency_us_bucket{{le="{2 ** (i + 1)}"}} {cum}')
         return "\n".join(lines) + "\n"
 
-    async def stop(self) -> None:
+    async def stop(self, drain_s: float = 0.0) -> None:
+        if drain_s > 0:
+            import asyncio as _asyncio
+
+            left = await _asyncio.to_thread(self.fast.drain, drain_s)
+            if left:
+                logger.warning("fast front drain timeout with %d connections", left)
         self.fast.stop()
         if self._fallback_runner is not None:
             await self._fallback_runner.cleanup()

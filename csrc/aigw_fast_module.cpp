@@ -113,6 +113,44 @@ PYBIND11_MODULE(aigw_fast, m) {
              s.add_rate_rule(r);
            })
       .def("set_fallback", &FastServer::set_fallback)
+      .def(
+          "swap_routes",
+          [](FastServer& s, py::list route_specs) {
+            std::vector<FastRoute> routes;
+            for (auto item : route_specs) {
+              py::dict rd = item.cast<py::dict>();
+              FastRoute r;
+              r.name = rd["name"].cast<std::string>();
+              r.model_match = rd["model_match"].cast<std::string>();
+              r.retries = rd["retries"].cast<int>();
+              r.has_costs = rd["has_costs"].cast<bool>();
+              r.eligible = rd["eligible"].cast<bool>();
+              for (auto b : rd["backends"].cast<py::list>()) {
+                py::dict d = b.cast<py::dict>();
+                FastBackend be;
+                be.name = d["name"].cast<std::string>();
+                be.host = d["host"].cast<std::string>();
+                be.port = d["port"].cast<uint16_t>();
+                if (d.contains("bearer")) be.bearer = d["bearer"].cast<std::string>();
+                if (d.contains("api_key_file"))
+                  be.api_key_file = d["api_key_file"].cast<std::string>();
+                if (d.contains("model_override"))
+                  be.model_override = d["model_override"].cast<std::string>();
+                if (d.contains("weight")) be.weight = d["weight"].cast<double>();
+                if (d.contains("priority")) be.priority = d["priority"].cast<int>();
+                if (d.contains("timeout_s")) be.timeout_s = d["timeout_s"].cast<double>();
+                if (d.contains("azure")) be.azure = d["azure"].cast<bool>();
+                if (d.contains("azure_api_version"))
+                  be.azure_api_version = d["azure_api_version"].cast<std::string>();
+                r.backends.push_back(std::move(be));
+              }
+              routes.push_back(std::move(r));
+            }
+            s.swap_routes(std::move(routes));
+          },
+          "hot-swap the route table (in-flight requests keep the old one)")
+      .def("drain", &FastServer::drain, py::arg("drain_s"),
+           py::call_guard<py::gil_scoped_release>())
       .def("enable_gpu", &FastServer::enable_gpu, py::arg("socket_path"),
            py::arg("window_us") = 100, py::arg("max_batch") = 256)
       .def(

@@ -985,8 +985,13 @@ class ConnHandler {
       return simple_reply(400, "invalid_request_error",
                           "missing required field 'model'", false);
     }
+    // snapshot the route table: a concurrent swap_routes (hot reload)
+    // cannot free it under us, and this request finishes on the table
+    // it resolved (watcher.go:79-160 swap semantics)
+    std::shared_ptr<const std::vector<FastRoute>> table =
+        std::atomic_load(&srv_->routes_);
     const FastRoute* route = nullptr;
-    for (const auto& r : srv_->routes_) {
+    for (const auto& r : *table) {
       if (r.catch_all || r.model_match == sc.model) {
         route = &r;
         break;
@@ -1771,13 +1776,28 @@ FastServer::FastServer() : pool_(new UpstreamPool) {}
 
 FastServer::~FastServer() { stop(); }
 
-void FastServer::add_route(FastRoute r) {
+static void normalize_route(FastRoute& r) {
   std::stable_sort(r.backends.begin(), r.backends.end(),
                    [](const FastBackend& a, const FastBackend& b) {
                      return a.priority < b.priority;
                    });
   if (r.model_match.empty()) r.catch_all = true;
-  routes_.push_back(std::move(r));
+}
+
+void FastServer::add_route(FastRoute r) {
+  normalize_route(r);
+  auto next = std::make_shared<std::vector<FastRoute>>(*routes_);
+  next->push_back(std::move(r));
+  routes_ = std::move(next);
+}
+
+void FastServer::swap_routes(std::vector<FastRoute> routes) {
+  for (auto& r : routes) normalize_route(r);
+  // atomic shared_ptr store: handlers snapshot the table per request
+  std::atomic_store(&routes_,
+                    std::shared_ptr<const std::vector<FastRoute>>(
+                        std::make_shared<const std::vector<FastRoute>>(
+                            std::move(routes))));
 }
 
 void FastServer::add_rate_rule(const RateRule& r) {
@@ -1902,6 +1922,24 @@ void FastServer::accept_loop() {
       stats_.active_connections--;
     }).detach();
   }
+}
+
+int FastServer::drain(double drain_s) {
+  // fail new connections first (the LB health check sees the listener
+  // gone), let in-flight requests finish
+  if (listen_fd_ >= 0) {
+    ::shutdown(listen_fd_, SHUT_RDWR);
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+  }
+  if (acceptor_.joinable()) acceptor_.join();
+  int waited_ms = 0;
+  int limit_ms = (int)(drain_s * 1000.0);
+  while (stats_.active_connections.load() > 0 && waited_ms < limit_ms) {
+    std::this_thread::sleep_for(std::chrono::milliseconds(10));
+    waited_ms += 10;
+  }
+  return (int)stats_.active_connections.load();
 }
 
 void FastServer::stop() {

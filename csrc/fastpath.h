@@ -103,8 +103,11 @@ class FastServer {
   FastServer();
   ~FastServer();
 
-  // configuration (before start)
+  // configuration (add_route before start; swap_routes any time —
+  // in-flight requests keep the table they resolved, the Python
+  // config watcher's swap semantics)
   void add_route(FastRoute r);
+  void swap_routes(std::vector<FastRoute> routes);
   void add_rate_rule(const RateRule& r);
   void set_fallback(const std::string& host, uint16_t port);
   // may be called multiple times: each socket is one admission-host
@@ -125,6 +128,10 @@ class FastServer {
 
   // lifecycle
   int start(const std::string& host, uint16_t port);  // returns bound port
+  // graceful drain: stop accepting, wait up to drain_s for in-flight
+  // connections to finish, then force-close; returns connections still
+  // open at the deadline (0 = clean)
+  int drain(double drain_s);
   void stop();
 
   const ServerStats& stats() const { return stats_; }
@@ -158,7 +165,8 @@ class FastServer {
   void rl_charge(const Usage& u);
   void rl_roll(RuleState& rs, int64_t now_ms);
 
-  std::vector<FastRoute> routes_;
+  std::shared_ptr<const std::vector<FastRoute>> routes_ =
+      std::make_shared<const std::vector<FastRoute>>();
   std::vector<std::unique_ptr<RuleState>> rules_;
   std::string fallback_host_;
   uint16_t fallback_port_ = 0;

@@ -390,3 +390,70 @@ def test_fast_front_azure_deployments_rewrite():
         await up_runner.cleanup()
 
     asyncio.run(run())
+
+
+def test_fast_front_hot_reload_and_drain():
+    """swap_routes under live traffic: new requests see the new table,
+    nothing 5xxs during the swap; drain() stops accepting but lets
+    in-flight requests finish."""
+
+    async def run():
+        mock, up_runner, up_port = await start_mock_upstream()
+        mock2, up_runner2, up_port2 = await start_mock_upstream()
+        mock.record = mock2.record = True
+        front, port = await _start(_cfg(up_port))
+        base = f"http://127.0.0.1:{port}"
+
+        stop_flag = {"stop": False}
+        errors = []
+
+        async def pound(c):
+            while not stop_flag["stop"]:
+                try:
+                    async with c.post(
+                        f"{base}/v1/chat/completions",
+                        json={"model": "fast-model",
+                              "messages": [{"role": "user", "content": "x"}]},
+                    ) as r:
+                        if r.status >= 500:
+                            errors.append(r.status)
+                        await r.read()
+                except Exception as e:
+                    errors.append(repr(e))
+
+        async with aiohttp.ClientSession() as c:
+            pounders = [asyncio.ensure_future(pound(c)) for _ in range(8)]
+            await asyncio.sleep(0.2)
+            # hot-swap: fast-model now routes to mock2 with a new key
+            new_cfg = load_config({
+                "version": "v1",
+                "llmRequestCosts": [{"metadataKey": "llm_total_token",
+                                     "type": "TotalToken"}],
+                "routes": [{
+                    "name": "fast",
+                    "headers": [{"name": "x-ai-eg-model",
+                                 "value": "fast-model"}],
+                    "backends": [{"name": "mock2", "schema": "OpenAI",
+                                  "upstream": {"host": "127.0.0.1",
+                                               "port": up_port2},
+                                  "auth": {"apiKey": "sk-new"}}],
+                }],
+            })
+            from aigw.filterapi import RuntimeConfig as _RC
+
+            front.reload(_RC(new_cfg))
+            await asyncio.sleep(0.3)
+            stop_flag["stop"] = True
+            await asyncio.gather(*pounders)
+        assert not errors, errors[:5]
+        assert mock2.requests, "reloaded backend never hit"
+        assert mock2.requests[-1]["headers"]["Authorization"] == "Bearer sk-new"
+
+        # drain: no new connections, zero left after idle traffic stops
+        left = front.fast.drain(2.0)
+        assert left == 0, left
+        await front.stop()
+        await up_runner.cleanup()
+        await up_runner2.cleanup()
+
+    asyncio.run(run())
