@@ -223,6 +223,9 @@ class FastFront:
         for s in sockets:
             if s:
                 self.fast.enable_gpu(s, self.gpu_window_us, self.gpu_max_batch)
+        if self.render_metrics_extra not in self.py_server.metrics_extra:
+            # the fallback app's /metrics carries the native counters too
+            self.py_server.metrics_extra.append(self.render_metrics_extra)
         self.port = self.fast.start(host, port)
         logger.info("fast front listening on %s:%d (fallback :%d)", host,
                     self.port, fallback_port)

@@ -299,6 +299,8 @@ class GatewayServer:
         # replica telemetry poller (aigw.extproc.telemetry); started by
         # the CLI / make_app when any backend configures telemetry
         self.telemetry = None
+        # extra Prometheus providers appended to /metrics (native front)
+        self.metrics_extra: list = []
 
     # ---- lifecycle -----------------------------------------------------------
 
@@ -430,7 +432,15 @@ class GatewayServer:
         return web.json_response({"status": "ok", "uptime_s": time.time() - self._started_at})
 
     async def _handle_metrics(self, request: web.Request) -> web.Response:
-        return web.Response(body=self.metrics.render(), content_type="text/plain")
+        body = self.metrics.render()
+        for extra in self.metrics_extra:
+            # native-front counters etc. (providers return Prometheus text)
+            try:
+                chunk = extra()
+                body += chunk.encode() if isinstance(chunk, str) else chunk
+            except Exception:
+                logger.exception("extra metrics provider failed")
+        return web.Response(body=body, content_type="text/plain")
 
     async def _handle_debug_tasks(self, request: web.Request) -> web.Response:
         """Live asyncio task dump — the profiling surface the reference gets

@@ -457,3 +457,27 @@ def test_fast_front_hot_reload_and_drain():
         await up_runner2.cleanup()
 
     asyncio.run(run())
+
+
+def test_fast_front_metrics_exposed_on_fallback():
+    """/metrics (served by the fallback app) carries the native
+    counters alongside the Python registry."""
+
+    async def run():
+        mock, up_runner, up_port = await start_mock_upstream()
+        front, port = await _start(_cfg(up_port))
+        async with aiohttp.ClientSession() as c:
+            async with c.post(
+                f"http://127.0.0.1:{port}/v1/chat/completions",
+                json={"model": "fast-model",
+                      "messages": [{"role": "user", "content": "x"}]},
+            ) as r:
+                assert r.status == 200
+            async with c.get(f"http://127.0.0.1:{port}/metrics") as r:
+                text = await r.text()
+                assert "aigw_fast_requests_total" in text
+                assert "aigw_fast_responses_2xx_total 1" in text
+        await front.stop()
+        await up_runner.cleanup()
+
+    asyncio.run(run())
