@@ -94,3 +94,36 @@ async def start_fast_mock(host: str = "127.0.0.1", port: int = 0,
     server = await loop.create_server(lambda: _FastMockProtocol(resp), host, port)
     actual = server.sockets[0].getsockname()[1]
     return server, actual
+
+
+def canned_chat_sse(model: str = "bench-llm", prompt_tokens: int = 4096,
+                    n_chunks: int = 16) -> bytes:
+    """Chunked SSE chat stream (content deltas + usage chunk + [DONE]) as
+    one raw HTTP byte blob for the native mock (streamed-serving bench)."""
+    events = []
+    for i in range(n_chunks):
+        delta = {"role": "assistant", "content": ""} if i == 0 else {"content": f"tok{i} "}
+        events.append({"id": "chatcmpl-bench", "object": "chat.completion.chunk",
+                       "model": model,
+                       "choices": [{"index": 0, "delta": delta,
+                                    "finish_reason": None}]})
+    events.append({"id": "chatcmpl-bench", "object": "chat.completion.chunk",
+                   "model": model,
+                   "choices": [{"index": 0, "delta": {},
+                                "finish_reason": "stop"}]})
+    events.append({"id": "chatcmpl-bench", "object": "chat.completion.chunk",
+                   "model": model, "choices": [],
+                   "usage": {"prompt_tokens": prompt_tokens,
+                             "completion_tokens": n_chunks,
+                             "total_tokens": prompt_tokens + n_chunks}})
+    body = b""
+    for ev in events:
+        data = b"data: " + json.dumps(ev, separators=(",", ":")).encode() + b"\n\n"
+        body += b"%x\r\n" % len(data) + data + b"\r\n"
+    body += b"17\r\ndata: [DONE]\n\n\r\n"[:0]  # (terminal DONE below)
+    done = b"data: [DONE]\n\n"
+    body += b"%x\r\n" % len(done) + done + b"\r\n"
+    body += b"0\r\n\r\n"
+    head = (b"HTTP/1.1 200 OK\r\ncontent-type: text/event-stream\r\n"
+            b"transfer-encoding: chunked\r\n\r\n")
+    return head + body

@@ -499,7 +499,13 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
 
     from aigw.testing.fastmock import canned_chat_response
 
-    canned = canned_chat_response(prompt_tokens=args.tokens)
+    if args.stream:
+        from aigw.testing.fastmock import canned_chat_sse
+
+        canned = canned_chat_sse(prompt_tokens=args.tokens,
+                                 n_chunks=args.stream_chunks)
+    else:
+        canned = canned_chat_response(prompt_tokens=args.tokens)
     mock = aigw_fast.FastMock()
     up_port = mock.start("127.0.0.1", canned.decode("latin1"))
     gpu_direct = use_gpu
@@ -518,7 +524,11 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
             pl["messages"][1]["content"] = f"variant {i}: " + pl["messages"][1]["content"]
             payloads.append(json.dumps(pl).encode())
     else:
-        payloads = [json.dumps(build_payload(args.tokens)).encode()]
+        pl = build_payload(args.tokens)
+        if args.stream:
+            pl["stream"] = True
+            pl["stream_options"] = {"include_usage": True}
+        payloads = [json.dumps(pl).encode()]
     # the cgroup CPU quota is shared by ALL ranks on a node: divide the
     # measured single-rank sweet spot (quota/4 workers) by the world size
     # so an 8-rank scale run does not earn CFS throttle stalls
@@ -572,6 +582,12 @@ def main():
                     help="HTTP worker processes per shard (0 = auto)")
     ap.add_argument("--no-gpu", action="store_true",
                     help="disable GPU token accounting (contention diagnosis)")
+    ap.add_argument("--stream", action="store_true",
+                    help="streamed serving: clients send stream:true and "
+                         "the mock answers an SSE chunk sequence; measures "
+                         "the native relay + usage-tap path")
+    ap.add_argument("--stream-chunks", type=int, default=16,
+                    help="SSE content chunks per streamed response")
     ap.add_argument("--cache-fp8", action="store_true",
                     help="fp8 (e4m3) semantic-cache index instead of bf16")
     ap.add_argument("--cache-payloads", type=int, default=0,
@@ -755,6 +771,7 @@ def main():
                 "seq_len": args.tokens,
                 "parallelism": f"dp{world}",
                 "front": "fast" if fast_mode else "lean",
+                "streamed": bool(getattr(args, "stream", False)) or None,
                 "workers_per_shard": workers,
                 "p50_ms": round(p50, 3),
                 "p99_ms": round(p99, 3),

@@ -1726,21 +1726,41 @@ LoadResult run_load_pool(const std::string& host, uint16_t port,
           buf.append(tmp, (size_t)r);
         }
         if (fail) { ::close(fd); fd = -1; errors++; continue; }
+        bool chunked = false;
         int64_t clen = 0;
         {
           size_t p = buf.find("content-length:");
           if (p == std::string::npos) p = buf.find("Content-Length:");
           if (p != std::string::npos && p < head_end)
             clen = atoll(buf.c_str() + p + 15);
+          size_t t = buf.find("transfer-encoding: chunked");
+          if (t == std::string::npos)
+            t = buf.find("Transfer-Encoding: chunked");
+          chunked = t != std::string::npos && t < head_end;
         }
-        while (buf.size() < head_end + 4 + (size_t)clen) {
-          ssize_t r = read_some(fd, tmp, sizeof(tmp));
-          if (r <= 0) { fail = true; break; }
-          buf.append(tmp, (size_t)r);
-        }
-        if (fail) { ::close(fd); fd = -1; errors++; continue; }
         bool ok = buf.compare(9, 3, "200") == 0;
-        buf.erase(0, head_end + 4 + (size_t)clen);
+        if (chunked) {
+          // streamed response (SSE): consume chunked framing to the
+          // terminal chunk so the connection stays reusable
+          buf.erase(0, head_end + 4);
+          ChunkedParser cp;
+          cp.feed(buf.data(), buf.size(), [](const char*, size_t) {});
+          buf.clear();
+          while (!cp.done() && !cp.error()) {
+            ssize_t r = read_some(fd, tmp, sizeof(tmp));
+            if (r <= 0) { fail = true; break; }
+            cp.feed(tmp, (size_t)r, [](const char*, size_t) {});
+          }
+          if (fail || cp.error()) { ::close(fd); fd = -1; errors++; continue; }
+        } else {
+          while (buf.size() < head_end + 4 + (size_t)clen) {
+            ssize_t r = read_some(fd, tmp, sizeof(tmp));
+            if (r <= 0) { fail = true; break; }
+            buf.append(tmp, (size_t)r);
+          }
+          if (fail) { ::close(fd); fd = -1; errors++; continue; }
+          buf.erase(0, head_end + 4 + (size_t)clen);
+        }
         if (!ok) { errors++; continue; }
         completed++;
         lat[c].push_back(now_us() - r0);
