@@ -120,7 +120,6 @@ def canned_chat_sse(model: str = "bench-llm", prompt_tokens: int = 4096,
     for ev in events:
         data = b"data: " + json.dumps(ev, separators=(",", ":")).encode() + b"\n\n"
         body += b"%x\r\n" % len(data) + data + b"\r\n"
-    body += b"17\r\ndata: [DONE]\n\n\r\n"[:0]  # (terminal DONE below)
     done = b"data: [DONE]\n\n"
     body += b"%x\r\n" % len(done) + done + b"\r\n"
     body += b"0\r\n\r\n"
