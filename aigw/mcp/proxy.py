@@ -38,6 +38,8 @@ logger = logging.getLogger("aigw.mcp")
 PREFIX_SEP = "__"
 PROTOCOL_VERSION = "2025-06-18"
 _METHOD_KEY = web.RequestKey("aigw_mcp_method", str) if hasattr(web, "RequestKey") else "aigw_mcp_method"
+_BACKEND_KEY = web.RequestKey("aigw_mcp_backend", str) if hasattr(web, "RequestKey") else "aigw_mcp_backend"
+_TOOL_KEY = web.RequestKey("aigw_mcp_tool", str) if hasattr(web, "RequestKey") else "aigw_mcp_tool"
 
 
 def _rpc_error(id_, code: int, message: str, status: int = 200) -> web.Response:
@@ -315,9 +317,18 @@ class MCPProxy:
                     _time.monotonic() - t0
                 )
             if span is not None:
+                # per-RPC attributes (tracing/mcp*.go parity): method,
+                # route, resolved backend/tool, session presence, outcome
                 span.set("mcp.method", method)
                 span.set("mcp.route", self.route.name)
-                self.tracer.end_span(span)
+                span.set("mcp.backend", request.get(_BACKEND_KEY, "") or None)
+                span.set("mcp.tool", request.get(_TOOL_KEY, "") or None)
+                span.set("mcp.session",
+                         bool(request.headers.get(internalapi.MCP_SESSION_ID_HEADER)))
+                if isinstance(resp, web.Response) and resp.status >= 400:
+                    self.tracer.end_span(span, error=f"http_{resp.status}")
+                else:
+                    self.tracer.end_span(span)
 
     async def _handle_inner(self, request: web.Request) -> web.StreamResponse:
         denied = self._authorize(request)
@@ -336,6 +347,12 @@ class MCPProxy:
             return _rpc_error(None, -32600, "batch requests not supported", status=400)
         method = payload.get("method", "")
         request[_METHOD_KEY] = method
+        if method in ("tools/call", "prompts/get"):
+            prefixed = str((payload.get("params") or {}).get("name", ""))
+            be, bare = self._backend_for(prefixed)
+            if be is not None:
+                request[_BACKEND_KEY] = be.name
+                request[_TOOL_KEY] = bare
         id_ = payload.get("id")
         if self._authz is not None:
             denied = self._authorize_rules(request, method, id_, payload)
