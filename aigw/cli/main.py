@@ -134,6 +134,11 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
         # native C++ data plane (csrc/fastpath.cpp); cold paths ride its
         # loopback fallback app. Configs the native server cannot express
         # fail loudly here — pick --front lean for those.
+        if getattr(args, "strict_schema", False):
+            raise SystemExit(
+                "--strict-schema validates in the Python pipeline; the native "
+                "front's hot paths would bypass it — use --front lean (or "
+                "aiohttp) with --strict-schema")
         import torch as _torch
 
         from aigw.extproc.fast_front import FastFront
