@@ -159,3 +159,43 @@ def test_endpoint_picker_follows_scraped_occupancy():
         await runner_b.cleanup()
 
     asyncio.run(run())
+
+
+def test_telemetry_scrape_failure_degrades_gracefully():
+    """Dead metrics endpoint: rows go stale, the picker falls back to
+    local estimates, errors are recorded, traffic keeps flowing."""
+
+    async def run():
+        rep = FakeReplica("only")
+        runner, port = await rep.start()
+        cfg = load_config({
+            "routes": [{
+                "name": "pool", "endpointPicker": True,
+                "backends": [
+                    {"name": "only", "schema": "OpenAI",
+                     "upstream": {"host": "127.0.0.1", "port": port},
+                     "telemetry": {"path": "/metrics", "intervalS": 0.05}},
+                    {"name": "dead", "schema": "OpenAI",
+                     "upstream": {"host": "127.0.0.1", "port": 9},
+                     "telemetry": {"path": "/metrics", "intervalS": 0.05}},
+                ],
+            }],
+        })
+        server = GatewayServer(RuntimeConfig(cfg))
+        server.gpu = GreedyPickGPU()
+        await server.start()
+        t = server.telemetry
+        assert "only" in t.rows
+        assert "dead" in t.scrape_errors
+        assert t.fresh_row("dead", max_age_s=1.0) is None
+        assert t.fresh_row("only", max_age_s=1.0) is not None
+        # stale check: a row older than max_age is not served
+        import time as _time
+
+        ts, row = t.rows["only"]
+        t.rows["only"] = (ts - 100.0, row)
+        assert t.fresh_row("only", max_age_s=1.0) is None
+        await server.close()
+        await runner.cleanup()
+
+    asyncio.run(run())

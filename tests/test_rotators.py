@@ -256,3 +256,58 @@ def test_manager_expiry_driven_refresh_and_failure_keeps_old(cloud, tmp_path):
     # endpoint restored: next sweep succeeds again
     rot.sts_endpoint = f"{base}/sts"
     assert mgr.rotate_expired(now=time.time() + 4000) == 1
+
+
+def test_manager_async_lifecycle(cloud, tmp_path):
+    """The CLI-driven async loop: starts, performs the initial rotation,
+    and stops cleanly."""
+    import asyncio
+
+    srv, base = cloud
+    out = str(tmp_path / "credentials")
+    rot = AwsOidcRotator(_oidc(base), role_arn="arn:aws:iam::1:role/g",
+                        region="us-east-1", out_file=out,
+                        sts_endpoint=f"{base}/sts")
+    mgr = RotationManager([rot], check_interval_s=0.05)
+
+    async def run():
+        await mgr.start()
+        for _ in range(100):
+            if mgr.rotation_count:
+                break
+            await asyncio.sleep(0.02)
+        await mgr.stop()
+
+    asyncio.run(run())
+    assert mgr.rotation_count >= 1
+    assert "aws_access_key_id = ASIA0001" in open(out).read()
+
+
+def test_build_rotation_manager_from_config(cloud, tmp_path):
+    from aigw.backendauth.rotators import build_rotation_manager
+    from aigw.filterapi.config import load_config
+
+    srv, base = cloud
+    out = str(tmp_path / "rotated-key")
+    cfg = load_config({
+        "routes": [{"name": "r", "backends": [{
+            "name": "az", "schema": "AzureOpenAI",
+            "upstream": {"host": "h", "port": 443},
+            "auth": {"apiKeyFile": out,
+                     "rotation": {"kind": "azure",
+                                  "azureTenantId": "tenant-1",
+                                  "azureClientId": "app-1",
+                                  "azureClientSecret": "az-secret",
+                                  "azureAuthority": f"{base}/aad"}},
+        }]}],
+    })
+    mgr = build_rotation_manager(cfg)
+    assert mgr is not None and len(mgr.rotators) == 1
+    assert mgr.rotate_expired() == 1
+    assert open(out).read().startswith("aad-token-")
+    # config without rotation -> None
+    cfg2 = load_config({"routes": [{"name": "r", "backends": [{
+        "name": "b", "schema": "OpenAI",
+        "upstream": {"host": "h", "port": 80},
+        "auth": {"apiKey": "sk"}}]}]})
+    assert build_rotation_manager(cfg2) is None
