@@ -231,3 +231,107 @@ def test_slow_and_partial_requests():
         await cleanup()
 
     asyncio.run(run())
+
+
+async def _fast_gateway():
+    """Same surface as _gateway() but served by the native C++ front."""
+    from aigw.extproc.fast_front import FastFront
+
+    up_srv, up_port = await start_fast_mock("127.0.0.1", 0)
+    cfg = load_config(yaml.safe_load(f"""
+routes:
+  - name: r
+    backends:
+      - name: b
+        schema: OpenAI
+        upstream: {{host: 127.0.0.1, port: {up_port}}}
+"""))
+    server = GatewayServer(RuntimeConfig(cfg))
+    front = FastFront(server, server.runtime)
+    port = await front.start("127.0.0.1", 0)
+
+    async def cleanup():
+        await front.stop()
+        up_srv.close()
+
+    return server, port, cleanup
+
+
+@pytest.mark.timeout(120)
+def test_fast_front_malformed_json_bodies_get_4xx():
+    """The native server gets the same mutation corpus as the lean front:
+    clean 4xx or 200, never 5xx/hangs."""
+
+    async def run():
+        _, port, cleanup = await _fast_gateway()
+        client = LeanClient()
+        nxt = _rng(0xFA57)
+        valid = json.dumps({"model": "m", "messages": [{"role": "user", "content": "hi"}]})
+        alphabet = '{}[]",:null true false 0123456789.eE-\\u00'
+        cases = [b"", b"{", b"[1,", b'{"model":}', b'{"model":"m"',
+                 b"\xff\xfe\x00", b'{"model": "m", "messages": "not-a-list"}',
+                 b"[" * 100, json.dumps({"messages": []}).encode()]
+        for _ in range(150):
+            b = bytearray(valid.encode())
+            for _m in range(1 + nxt(4)):
+                b[nxt(len(b))] = ord(alphabet[nxt(len(alphabet))])
+            cases.append(bytes(b))
+        for body in cases:
+            r = await client.post(
+                host="127.0.0.1", port=port, tls=False,
+                path="/v1/chat/completions",
+                headers={"content-type": "application/json"}, body=body,
+            )
+            data = await r.read()
+            assert r.status < 500, (r.status, body[:120], data[:200])
+            r.release()
+        await client.close()
+        await cleanup()
+
+    asyncio.run(run())
+
+
+@pytest.mark.timeout(120)
+def test_fast_front_broken_framing_never_hangs():
+    async def run():
+        _, port, cleanup = await _fast_gateway()
+        frames = [
+            b"GARBAGE\r\n\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\ncontent-length: -5\r\n\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\ncontent-length: zzz\r\n\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\n" + b"x: y\r\n" * 5000 + b"\r\n",
+            b"POST /nope HTTP/1.1\r\ncontent-length: 2\r\n\r\n{}",
+            b"GET /health HTTP/1.1\r\n\r\n" * 3,  # pipelined
+            b"\r\n\r\n\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\n"
+            b"transfer-encoding: chunked\r\n\r\n2\r\n{}\r\n0\r\n\r\n",
+            b"POST /v1/chat/completions HTTP/1.1\r\n"
+            b"content-length: 2\r\ncontent-length: 9\r\n\r\n{}1234567",
+        ]
+        for payload in frames:
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            writer.write(payload)
+            try:
+                await writer.drain()
+                await asyncio.wait_for(reader.read(65536), timeout=6)
+            except (asyncio.TimeoutError,) as e:
+                raise AssertionError(f"hang on frame {payload[:60]!r}") from e
+            except (ConnectionResetError, BrokenPipeError):
+                pass
+            finally:
+                writer.close()
+        # healthy afterwards
+        client = LeanClient()
+        r = await client.post(
+            host="127.0.0.1", port=port, tls=False, path="/v1/chat/completions",
+            headers={"content-type": "application/json"},
+            body=json.dumps({"model": "m",
+                             "messages": [{"role": "user", "content": "ok"}]}).encode(),
+        )
+        assert r.status == 200
+        await r.read()
+        r.release()
+        await client.close()
+        await cleanup()
+
+    asyncio.run(run())
