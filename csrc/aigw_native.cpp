@@ -44,6 +44,28 @@ static py::tuple scan_chat_body(py::buffer buf) {
                         ok && sc.stream == 1, py::bytes(sc.text));
 }
 
+// (ok, model, stream, text, model_vs, model_ve, msgs_vs, msgs_ve) —
+// the byte spans the native fast path splices (model override) and
+// hashes around (cache scope fingerprint); exposed for property tests.
+static py::tuple scan_chat_body_spans(py::buffer buf) {
+  py::buffer_info info = buf.request();
+  Scan sc{static_cast<const char*>(info.ptr),
+          static_cast<const char*>(info.ptr) + info.size};
+  sc.base = static_cast<const char*>(info.ptr);
+  bool ok;
+  {
+    py::gil_scoped_release release;
+    ok = sc.parse_value(0, "", true) && sc.ok;
+    if (ok) {
+      sc.ws();
+      ok = sc.p == sc.end;
+    }
+  }
+  return py::make_tuple(ok, py::str(sc.model), ok && sc.stream == 1,
+                        py::bytes(sc.text), sc.model_vs, sc.model_ve,
+                        sc.msgs_vs, sc.msgs_ve);
+}
+
 static bool contains_usage(py::buffer buf) {
   py::buffer_info info = buf.request();
   const char* p = static_cast<const char*>(info.ptr);
@@ -81,6 +103,8 @@ PYBIND11_MODULE(aigw_native, m) {
   m.def("scan_chat_body", &scan_chat_body,
         "one-pass (ok, model, stream, text) extraction from a chat body");
   m.def("contains_usage", &contains_usage, "fast '\"usage\"' probe");
+  m.def("scan_chat_body_spans", &scan_chat_body_spans,
+        "scan + model/messages byte spans (fast-path splice points)");
   py::class_<SSEFeed>(m, "SSEFeed")
       .def(py::init<>())
       .def("feed", &SSEFeed::feed)

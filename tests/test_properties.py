@@ -204,3 +204,52 @@ def test_response_chunk_never_raises_unhandled(data, schema):
         t.response_flush()
     except (ValueError, TranslationError):
         pass
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    model=st.text(max_size=30),
+    stream=st.booleans(),
+    extra=st.dictionaries(st.text(max_size=8), json_values, max_size=4),
+    messages=st.lists(
+        st.fixed_dictionaries({"role": st.sampled_from(["user", "assistant"]),
+                               "content": st.text(max_size=60)}),
+        max_size=4,
+    ),
+    override=st.text(alphabet=st.characters(min_codepoint=33, max_codepoint=126,
+                                            blacklist_characters='"\\'),
+                     min_size=1, max_size=20),
+)
+def test_scanner_spans_splice_correctly(model, stream, extra, messages, override):
+    """The byte spans the native fast path uses must be EXACT: splicing a
+    model override at [model_vs, model_ve) yields JSON whose model is the
+    override and nothing else changed; slicing [msgs_vs, msgs_ve) yields
+    the messages array; hashing around the messages span keys on every
+    other byte (the cache scope fingerprint contract)."""
+    import aigw_native
+
+    body = dict(extra)
+    body["model"] = model
+    body["stream"] = stream
+    body["messages"] = messages
+    raw = json.dumps(body).encode()
+    ok, got_model, _, _, mvs, mve, svs, sve = aigw_native.scan_chat_body_spans(raw)
+    assert ok and got_model == model
+    # model span: value incl. quotes
+    assert raw[mvs:mve] == json.dumps(model).encode()
+    spliced = raw[:mvs] + json.dumps(override).encode() + raw[mve:]
+    doc = json.loads(spliced)
+    assert doc["model"] == override
+    doc["model"] = model
+    assert doc == body
+    # messages span: exactly the array value
+    assert sve > svs
+    assert json.loads(raw[svs:sve]) == messages
+    # bytes outside the messages span are identical for two bodies that
+    # differ only in message content
+    body2 = dict(body)
+    body2["messages"] = [{"role": "user", "content": "DIFFERENT"}]
+    raw2 = json.dumps(body2).encode()
+    ok2, _, _, _, _, _, svs2, sve2 = aigw_native.scan_chat_body_spans(raw2)
+    assert ok2
+    assert raw[:svs] + raw[sve:] == raw2[:svs2] + raw2[sve2:]
