@@ -71,6 +71,21 @@ __global__ void f32_to_bf16_kernel(const float* __restrict__ in,
 // bf16 -> OCP e4m3 via the gfx950 packed-convert instruction (the same
 // format torch.float8_e4m3fn uses on this architecture, so the fp8
 // index is interchangeable with the Python cache's)
+// scatter the batch's query vectors into their pending-pool slots in
+// ONE launch (a per-request hipMemcpyAsync costs ~10 us of enqueue CPU
+// each; at 100 misses/batch that dominated the whole cache batch)
+__global__ void park_pending_kernel(const bf16* __restrict__ q,
+                                    const int32_t* __restrict__ slots,
+                                    int n_req, int dim,
+                                    bf16* __restrict__ pending) {
+  int r = blockIdx.x;
+  int col = threadIdx.x;
+  if (r >= n_req || col >= dim) return;
+  int slot = slots[r];
+  if (slot < 0) return;
+  pending[(size_t)slot * dim + col] = q[(size_t)r * dim + col];
+}
+
 __global__ void bf16_to_fp8_kernel(const bf16* __restrict__ in,
                                    uint8_t* __restrict__ out, int n) {
   int i = (blockIdx.x * blockDim.x + threadIdx.x) * 2;
@@ -244,6 +259,9 @@ class GpuAdmissionDirect {
     HIP_OK(hipHostMalloc(&h_best_, sizeof(unsigned long long) * max_req_,
                          hipHostMallocDefault));
     HIP_OK(hipMalloc(&d_pending_, sizeof(bf16) * (size_t)pending_cap * dim));
+    HIP_OK(hipMalloc(&d_slots_, sizeof(int32_t) * max_req_));
+    HIP_OK(hipHostMalloc(&h_slots_, sizeof(int32_t) * max_req_,
+                         hipHostMallocDefault));
     HIP_OK(hipStreamCreateWithFlags(&insert_stream_, hipStreamNonBlocking));
     cache_on_ = true;
     return true;
@@ -279,13 +297,12 @@ class GpuAdmissionDirect {
                        d_poolbf_, d_proj_, d_gout_, n_req, dim, dim, nullptr, 0);
     hipLaunchKernelGGL(l2norm_rows_kernel, dim3(n_req), dim3(dim), 0, stream_,
                        d_gout_, d_q_, n_req, dim);
-    // park query vectors for possible insert
-    for (int i = 0; i < n_req; ++i) {
-      if (pending_slots[i] >= 0 && pending_slots[i] < pending_cap_)
-        HIP_OK(hipMemcpyAsync(d_pending_ + (size_t)pending_slots[i] * dim,
-                              d_q_ + (size_t)i * dim, sizeof(bf16) * dim,
-                              hipMemcpyDeviceToDevice, stream_));
-    }
+    // park query vectors for possible insert: one gather launch
+    memcpy(h_slots_, pending_slots, sizeof(int32_t) * (size_t)n_req);
+    HIP_OK(hipMemcpyAsync(d_slots_, h_slots_, sizeof(int32_t) * n_req,
+                          hipMemcpyHostToDevice, stream_));
+    hipLaunchKernelGGL(park_pending_kernel, dim3(n_req), dim3(dim), 0, stream_,
+                       d_q_, d_slots_, n_req, dim, d_pending_);
     long long rows_now = rows_visible_.load(std::memory_order_acquire);
     if (rows_now > 0) {
       HIP_OK(hipMemsetAsync(d_best_, 0, sizeof(unsigned long long) * n_req,
@@ -390,10 +407,11 @@ class GpuAdmissionDirect {
     if (cache_on_) {
       (void)hipStreamDestroy(insert_stream_);
       (void)hipHostFree(h_best_);
+      (void)hipHostFree(h_slots_);
       for (void* p : {(void*)d_emb_, (void*)d_proj_, (void*)d_index_,
                       (void*)d_pool_, (void*)d_poolcnt_, (void*)d_poolbf_,
                       (void*)d_gout_, (void*)d_q_, (void*)d_q8_,
-                      (void*)d_best_, (void*)d_pending_})
+                      (void*)d_best_, (void*)d_pending_, (void*)d_slots_})
         (void)hipFree(p);
     }
     for (void* p : {(void*)d_htab_keys_, (void*)d_htab_rank_, (void*)d_bytes_,
@@ -430,6 +448,8 @@ class GpuAdmissionDirect {
   unsigned long long* d_best_ = nullptr;
   unsigned long long* h_best_ = nullptr;
   bf16* d_pending_ = nullptr;
+  int32_t* d_slots_ = nullptr;
+  int32_t* h_slots_ = nullptr;
   hipStream_t insert_stream_{};
   size_t max_bytes_ = 0;
   int max_req_ = 0;
