@@ -109,10 +109,47 @@ __global__ void bf16_to_fp8_kernel(const bf16* __restrict__ in,
 
 }  // namespace
 
-// One instance per fast server; init once, count() called by the
-// admission batcher thread (single-threaded use of the stream).
+// One instance per fast server; init once. TWO batch sets on TWO
+// streams let the batcher keep one batch on the GPU while it packs,
+// fulfills, and submits the next (classic copy/compute overlap — the
+// per-batch pipeline is launch-bound, so overlapping whole batches
+// nearly doubles admission throughput).
 class GpuAdmissionDirect {
  public:
+  static constexpr int kSets = 2;
+
+  struct BatchSet {
+    hipStream_t stream{};
+    hipEvent_t event{};
+    char* h_bytes = nullptr;
+    int64_t* h_off = nullptr;
+    int32_t* h_counts = nullptr;
+    unsigned long long* h_best = nullptr;
+    int32_t* h_slots = nullptr;
+    uint8_t* d_bytes = nullptr;
+    int64_t* d_off = nullptr;
+    uint8_t* d_flags = nullptr;
+    uint8_t* d_gflags = nullptr;
+    int32_t* d_blk = nullptr;
+    int32_t* d_excl = nullptr;
+    int32_t* d_totals = nullptr;
+    int32_t* d_seg_start = nullptr;
+    int32_t* d_seg_req = nullptr;
+    int32_t* d_ghead = nullptr;
+    int32_t* d_out_ids = nullptr;
+    int32_t* d_counts = nullptr;
+    // cache-pipeline intermediates (allocated by init_cache)
+    float* d_pool = nullptr;
+    int32_t* d_poolcnt = nullptr;
+    bf16* d_poolbf = nullptr;
+    float* d_gout = nullptr;
+    bf16* d_q = nullptr;
+    uint8_t* d_q8 = nullptr;
+    unsigned long long* d_best = nullptr;
+    int32_t* d_slots = nullptr;
+    long long rows_in_flight = 0;  // index rows visible to this batch
+  };
+
   bool init(const long long* htab_keys, const int32_t* htab_rank, int htab_n,
             size_t max_bytes, int max_req, int device) {
     int count = 0;
@@ -122,101 +159,136 @@ class GpuAdmissionDirect {
     max_bytes_ = max_bytes;
     max_req_ = max_req;
     htab_mask_ = htab_n - 1;
-    HIP_OK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
-    // hipEventBlockingSync: the batcher thread YIELDS while the batch
-    // runs instead of busy-spinning — the serving container is CPU-quota
-    // bound (16 CPUs for the whole gateway), so a spinning core is ~6%
-    // of the entire budget
-    HIP_OK(hipEventCreateWithFlags(&event_,
-                                   hipEventDisableTiming | hipEventBlockingSync));
-    HIP_OK(hipHostMalloc(&h_bytes_, max_bytes, hipHostMallocDefault));
-    HIP_OK(hipHostMalloc(&h_off_, sizeof(int64_t) * (max_req + 1),
-                         hipHostMallocDefault));
-    HIP_OK(hipHostMalloc(&h_counts_, sizeof(int32_t) * max_req,
-                         hipHostMallocDefault));
     HIP_OK(hipMalloc(&d_htab_keys_, sizeof(long long) * htab_n));
     HIP_OK(hipMalloc(&d_htab_rank_, sizeof(int32_t) * htab_n));
     HIP_OK(hipMemcpy(d_htab_keys_, htab_keys, sizeof(long long) * htab_n,
                      hipMemcpyHostToDevice));
     HIP_OK(hipMemcpy(d_htab_rank_, htab_rank, sizeof(int32_t) * htab_n,
                      hipMemcpyHostToDevice));
-    HIP_OK(hipMalloc(&d_bytes_, max_bytes));
-    HIP_OK(hipMalloc(&d_off_, sizeof(int64_t) * (max_req + 1)));
-    HIP_OK(hipMalloc(&d_flags_, max_bytes));
-    HIP_OK(hipMalloc(&d_gflags_, max_bytes));
     int max_blocks = (int)((max_bytes + 255) / 256);
-    HIP_OK(hipMalloc(&d_blk_, sizeof(int32_t) * max_blocks));
-    HIP_OK(hipMalloc(&d_excl_, sizeof(int32_t) * max_blocks));
-    HIP_OK(hipMalloc(&d_totals_, sizeof(int32_t) * 2));  // [n_segs, n_groups]
-    HIP_OK(hipMalloc(&d_seg_start_, sizeof(int32_t) * max_bytes));
-    HIP_OK(hipMalloc(&d_seg_req_, sizeof(int32_t) * max_bytes));
-    HIP_OK(hipMalloc(&d_ghead_, sizeof(int32_t) * max_bytes));
-    HIP_OK(hipMalloc(&d_out_ids_, sizeof(int32_t) * max_bytes));
-    HIP_OK(hipMalloc(&d_counts_, sizeof(int32_t) * max_req));
+    for (int s = 0; s < kSets; ++s) {
+      BatchSet& b = sets_[s];
+      HIP_OK(hipStreamCreateWithFlags(&b.stream, hipStreamNonBlocking));
+      // hipEventBlockingSync: the batcher thread YIELDS while the batch
+      // runs instead of busy-spinning — the serving container is
+      // CPU-quota bound, so a spinning core is ~6% of the budget
+      HIP_OK(hipEventCreateWithFlags(
+          &b.event, hipEventDisableTiming | hipEventBlockingSync));
+      HIP_OK(hipHostMalloc(&b.h_bytes, max_bytes, hipHostMallocDefault));
+      HIP_OK(hipHostMalloc(&b.h_off, sizeof(int64_t) * (max_req + 1),
+                           hipHostMallocDefault));
+      HIP_OK(hipHostMalloc(&b.h_counts, sizeof(int32_t) * max_req,
+                           hipHostMallocDefault));
+      HIP_OK(hipMalloc(&b.d_bytes, max_bytes));
+      HIP_OK(hipMalloc(&b.d_off, sizeof(int64_t) * (max_req + 1)));
+      HIP_OK(hipMalloc(&b.d_flags, max_bytes));
+      HIP_OK(hipMalloc(&b.d_gflags, max_bytes));
+      HIP_OK(hipMalloc(&b.d_blk, sizeof(int32_t) * max_blocks));
+      HIP_OK(hipMalloc(&b.d_excl, sizeof(int32_t) * max_blocks));
+      HIP_OK(hipMalloc(&b.d_totals, sizeof(int32_t) * 2));
+      HIP_OK(hipMalloc(&b.d_seg_start, sizeof(int32_t) * max_bytes));
+      HIP_OK(hipMalloc(&b.d_seg_req, sizeof(int32_t) * max_bytes));
+      HIP_OK(hipMalloc(&b.d_ghead, sizeof(int32_t) * max_bytes));
+      HIP_OK(hipMalloc(&b.d_out_ids, sizeof(int32_t) * max_bytes));
+      HIP_OK(hipMalloc(&b.d_counts, sizeof(int32_t) * max_req));
+    }
     ready_ = true;
     return true;
   }
 
   // texts: concatenated; offsets[i] is text i's start (n_req entries,
   // matching the kernel contract in aigw/ops/tokenizer.py pack()).
-  // Returns per-request token counts. Single caller thread.
+  // Synchronous single-set wrapper (kept for the GPU-off fallbacks and
+  // simple callers); the pipelined batcher uses submit()/wait_set().
   bool count(const char* bytes, size_t n, const int64_t* offsets, int n_req,
              int32_t* counts_out) {
-    if (!count_submit(bytes, n, offsets, n_req)) return false;
-    HIP_OK(hipMemcpyAsync(h_counts_, d_counts_, sizeof(int32_t) * n_req,
-                          hipMemcpyDeviceToHost, stream_));
-    HIP_OK(hipEventRecord(event_, stream_));
-    HIP_OK(hipEventSynchronize(event_));
-    memcpy(counts_out, h_counts_, sizeof(int32_t) * (size_t)n_req);
-    return true;
+    if (!submit(0, bytes, n, offsets, n_req, nullptr)) return false;
+    return wait_set(0, n_req, counts_out, nullptr, nullptr);
   }
 
-  // Launch the BPE pipeline (H2D + segmentation + merge) without the
-  // D2H/sync tail; shared by count() and count_lookup() below.
-  bool count_submit(const char* bytes, size_t n, const int64_t* offsets,
-                    int n_req) {
+  // Launch one batch on set `s` (H2D + BPE + optional cache pipeline when
+  // pending_slots != nullptr and the cache is on), then record its event.
+  // The caller owns set rotation: a set must be wait_set()ed before reuse.
+  bool submit(int s, const char* bytes, size_t n, const int64_t* offsets,
+              int n_req, const int32_t* pending_slots) {
     if (!ready_ || n == 0 || n_req == 0 || n > max_bytes_ || n_req > max_req_)
       return false;
+    BatchSet& b = sets_[s];
     // launches follow the CALLER thread's current device; the batcher
     // thread differs from the init thread, so pin it here (no-op when
     // already current) — rank N of an 8-GPU node must stay on device N
     HIP_OK(hipSetDevice(device_));
-    memcpy(h_bytes_, bytes, n);
-    memcpy(h_off_, offsets, sizeof(int64_t) * (size_t)n_req);
-    HIP_OK(hipMemcpyAsync(d_bytes_, h_bytes_, n, hipMemcpyHostToDevice, stream_));
-    HIP_OK(hipMemcpyAsync(d_off_, h_off_, sizeof(int64_t) * (size_t)n_req,
-                          hipMemcpyHostToDevice, stream_));
+    memcpy(b.h_bytes, bytes, n);
+    memcpy(b.h_off, offsets, sizeof(int64_t) * (size_t)n_req);
+    hipStream_t st = b.stream;
+    HIP_OK(hipMemcpyAsync(b.d_bytes, b.h_bytes, n, hipMemcpyHostToDevice, st));
+    HIP_OK(hipMemcpyAsync(b.d_off, b.h_off, sizeof(int64_t) * (size_t)n_req,
+                          hipMemcpyHostToDevice, st));
     int blocks = (int)((n + 255) / 256);
-    hipLaunchKernelGGL(seg_flags_kernel, dim3(blocks), dim3(256), 0, stream_,
-                       d_bytes_, (int)n, d_flags_);
+    hipLaunchKernelGGL(seg_flags_kernel, dim3(blocks), dim3(256), 0, st,
+                       b.d_bytes, (int)n, b.d_flags);
     hipLaunchKernelGGL(seg_force_starts_kernel, dim3((n_req + 255) / 256),
-                       dim3(256), 0, stream_, d_off_, n_req, d_flags_);
+                       dim3(256), 0, st, b.d_off, n_req, b.d_flags);
     hipLaunchKernelGGL(seg_block_count_kernel, dim3(blocks), dim3(256), 0,
-                       stream_, d_flags_, (int)n, d_blk_);
-    hipLaunchKernelGGL(excl_scan_kernel, dim3(1), dim3(256), 0, stream_,
-                       d_blk_, blocks, d_excl_, d_totals_);
-    hipLaunchKernelGGL(seg_write_kernel, dim3(blocks), dim3(256), 0, stream_,
-                       d_flags_, (int)n, d_excl_, d_off_, n_req, d_seg_start_,
-                       d_seg_req_);
+                       st, b.d_flags, (int)n, b.d_blk);
+    hipLaunchKernelGGL(excl_scan_kernel, dim3(1), dim3(256), 0, st,
+                       b.d_blk, blocks, b.d_excl, b.d_totals);
+    hipLaunchKernelGGL(seg_write_kernel, dim3(blocks), dim3(256), 0, st,
+                       b.d_flags, (int)n, b.d_excl, b.d_off, n_req,
+                       b.d_seg_start, b.d_seg_req);
     // groups (device-bounded; gflags beyond n_segs cleared so counts stay 0)
-    HIP_OK(hipMemsetAsync(d_gflags_, 0, n, stream_));
+    HIP_OK(hipMemsetAsync(b.d_gflags, 0, n, st));
     hipLaunchKernelGGL(group_head_flags_dev_kernel, dim3(blocks), dim3(256), 0,
-                       stream_, d_seg_start_, d_seg_req_, d_totals_, d_off_,
-                       d_gflags_);
+                       st, b.d_seg_start, b.d_seg_req, b.d_totals, b.d_off,
+                       b.d_gflags);
     hipLaunchKernelGGL(seg_block_count_kernel, dim3(blocks), dim3(256), 0,
-                       stream_, d_gflags_, (int)n, d_blk_);
-    hipLaunchKernelGGL(excl_scan_kernel, dim3(1), dim3(256), 0, stream_,
-                       d_blk_, blocks, d_excl_, d_totals_ + 1);
+                       st, b.d_gflags, (int)n, b.d_blk);
+    hipLaunchKernelGGL(excl_scan_kernel, dim3(1), dim3(256), 0, st,
+                       b.d_blk, blocks, b.d_excl, b.d_totals + 1);
     hipLaunchKernelGGL(flag_compact_write_kernel, dim3(blocks), dim3(256), 0,
-                       stream_, d_gflags_, (int)n, d_excl_, d_ghead_);
-    HIP_OK(hipMemsetAsync(d_counts_, 0, sizeof(int32_t) * n_req, stream_));
+                       st, b.d_gflags, (int)n, b.d_excl, b.d_ghead);
+    HIP_OK(hipMemsetAsync(b.d_counts, 0, sizeof(int32_t) * n_req, st));
     long long group_bound = (long long)n / 32 + n_req + 1;
     int blocks2 = (int)((group_bound + 3) / 4);
     hipLaunchKernelGGL(bpe_encode_grouped_dev_kernel, dim3(blocks2), dim3(256),
-                       0, stream_, d_bytes_, d_seg_start_, d_seg_req_,
-                       d_totals_, (int)n, d_ghead_, d_totals_ + 1,
-                       d_htab_keys_, d_htab_rank_, htab_mask_, d_out_ids_,
-                       d_counts_);
+                       0, st, b.d_bytes, b.d_seg_start, b.d_seg_req,
+                       b.d_totals, (int)n, b.d_ghead, b.d_totals + 1,
+                       d_htab_keys_, d_htab_rank_, htab_mask_, b.d_out_ids,
+                       b.d_counts);
+    if (pending_slots != nullptr && cache_on_) {
+      if (!submit_cache(b, n, n_req, pending_slots)) return false;
+    }
+    HIP_OK(hipMemcpyAsync(b.h_counts, b.d_counts, sizeof(int32_t) * n_req,
+                          hipMemcpyDeviceToHost, st));
+    HIP_OK(hipEventRecord(b.event, st));
+    return true;
+  }
+
+  // Block until set `s` finishes; copy out counts (and lookup results
+  // when the batch ran the cache pipeline and rows/scores != nullptr).
+  bool wait_set(int s, int n_req, int32_t* counts_out, int32_t* rows_out,
+                float* scores_out) {
+    BatchSet& b = sets_[s];
+    HIP_OK(hipEventSynchronize(b.event));
+    memcpy(counts_out, b.h_counts, sizeof(int32_t) * (size_t)n_req);
+    if (rows_out != nullptr && scores_out != nullptr) {
+      long long rows_now = b.rows_in_flight;
+      for (int i = 0; i < n_req; ++i) {
+        rows_out[i] = -1;
+        scores_out[i] = 0.f;
+        if (rows_now > 0) {
+          unsigned long long u = b.h_best[i];
+          unsigned hi = (unsigned)(u >> 32);
+          unsigned bits = (hi >= 0x80000000u) ? (hi ^ 0x80000000u) : ~hi;
+          float score;
+          memcpy(&score, &bits, 4);
+          if (score >= threshold_) {
+            rows_out[i] = (int32_t)(u & 0xFFFFFFFFu);
+            scores_out[i] = score;
+          }
+        }
+      }
+    }
     return true;
   }
 
@@ -249,72 +321,71 @@ class GpuAdmissionDirect {
     size_t elt = fp8 ? 1 : sizeof(bf16);
     HIP_OK(hipMalloc(&d_index_, elt * (size_t)capacity * dim));
     HIP_OK(hipMemset(d_index_, 0, elt * (size_t)capacity * dim));
-    HIP_OK(hipMalloc(&d_pool_, sizeof(float) * (size_t)max_req_ * dim));
-    HIP_OK(hipMalloc(&d_poolcnt_, sizeof(int32_t) * max_req_));
-    HIP_OK(hipMalloc(&d_poolbf_, sizeof(bf16) * (size_t)max_req_ * dim));
-    HIP_OK(hipMalloc(&d_gout_, sizeof(float) * (size_t)max_req_ * dim));
-    HIP_OK(hipMalloc(&d_q_, sizeof(bf16) * (size_t)max_req_ * dim));
-    if (fp8) HIP_OK(hipMalloc(&d_q8_, (size_t)max_req_ * dim));
-    HIP_OK(hipMalloc(&d_best_, sizeof(unsigned long long) * max_req_));
-    HIP_OK(hipHostMalloc(&h_best_, sizeof(unsigned long long) * max_req_,
-                         hipHostMallocDefault));
     HIP_OK(hipMalloc(&d_pending_, sizeof(bf16) * (size_t)pending_cap * dim));
-    HIP_OK(hipMalloc(&d_slots_, sizeof(int32_t) * max_req_));
-    HIP_OK(hipHostMalloc(&h_slots_, sizeof(int32_t) * max_req_,
-                         hipHostMallocDefault));
     HIP_OK(hipStreamCreateWithFlags(&insert_stream_, hipStreamNonBlocking));
+    for (int s = 0; s < kSets; ++s) {
+      BatchSet& b = sets_[s];
+      HIP_OK(hipMalloc(&b.d_pool, sizeof(float) * (size_t)max_req_ * dim));
+      HIP_OK(hipMalloc(&b.d_poolcnt, sizeof(int32_t) * max_req_));
+      HIP_OK(hipMalloc(&b.d_poolbf, sizeof(bf16) * (size_t)max_req_ * dim));
+      HIP_OK(hipMalloc(&b.d_gout, sizeof(float) * (size_t)max_req_ * dim));
+      HIP_OK(hipMalloc(&b.d_q, sizeof(bf16) * (size_t)max_req_ * dim));
+      if (fp8) HIP_OK(hipMalloc(&b.d_q8, (size_t)max_req_ * dim));
+      HIP_OK(hipMalloc(&b.d_best, sizeof(unsigned long long) * max_req_));
+      HIP_OK(hipHostMalloc(&b.h_best, sizeof(unsigned long long) * max_req_,
+                           hipHostMallocDefault));
+      HIP_OK(hipMalloc(&b.d_slots, sizeof(int32_t) * max_req_));
+      HIP_OK(hipHostMalloc(&b.h_slots, sizeof(int32_t) * max_req_,
+                           hipHostMallocDefault));
+    }
     cache_on_ = true;
     return true;
   }
 
   bool cache_on() const { return cache_on_; }
 
-  // Extended batch op: counts + cache lookup. pending_slots[i] >= 0 parks
-  // request i's query vector in that pending-pool slot (caller-managed
-  // free list). rows_out[i] = best index row (-1 below threshold or empty
-  // index); scores_out[i] = cosine.
-  bool count_lookup(const char* bytes, size_t n, const int64_t* offsets,
-                    int n_req, int32_t* counts_out,
-                    const int32_t* pending_slots, int32_t* rows_out,
-                    float* scores_out) {
-    if (!cache_on_) return false;
-    if (!count_submit(bytes, n, offsets, n_req)) return false;
+  // The embed/lookup tail of one batch, launched on the set's stream
+  // right after the BPE pipeline (same pipeline as
+  // aigw/ops/semcache.py embed/lookup).
+  bool submit_cache(BatchSet& b, size_t n, int n_req,
+                    const int32_t* pending_slots) {
     int dim = dim_;
-    HIP_OK(hipMemsetAsync(d_pool_, 0, sizeof(float) * (size_t)n_req * dim,
-                          stream_));
-    HIP_OK(hipMemsetAsync(d_poolcnt_, 0, sizeof(int32_t) * n_req, stream_));
+    hipStream_t st = b.stream;
+    HIP_OK(hipMemsetAsync(b.d_pool, 0, sizeof(float) * (size_t)n_req * dim, st));
+    HIP_OK(hipMemsetAsync(b.d_poolcnt, 0, sizeof(int32_t) * n_req, st));
     constexpr int P = 8;
     hipLaunchKernelGGL(meanpool_accum_kernel, dim3(n_req, P), dim3(dim), 0,
-                       stream_, d_out_ids_, d_off_, n_req, (int)n, d_emb_,
-                       dim, P, d_pool_, d_poolcnt_);
-    hipLaunchKernelGGL(meanpool_div_kernel, dim3(n_req), dim3(dim), 0, stream_,
-                       d_pool_, d_poolcnt_, n_req, dim);
+                       st, b.d_out_ids, b.d_off, n_req, (int)n, d_emb_,
+                       dim, P, b.d_pool, b.d_poolcnt);
+    hipLaunchKernelGGL(meanpool_div_kernel, dim3(n_req), dim3(dim), 0, st,
+                       b.d_pool, b.d_poolcnt, n_req, dim);
     int total = n_req * dim;
     hipLaunchKernelGGL(f32_to_bf16_kernel, dim3((total + 255) / 256), dim3(256),
-                       0, stream_, d_pool_, d_poolbf_, total);
+                       0, st, b.d_pool, b.d_poolbf, total);
     dim3 ggrid((n_req + 15) / 16, (dim + 63) / 64);
-    hipLaunchKernelGGL(gemm_bf16_nt_kernel, ggrid, dim3(256), 0, stream_,
-                       d_poolbf_, d_proj_, d_gout_, n_req, dim, dim, nullptr, 0);
-    hipLaunchKernelGGL(l2norm_rows_kernel, dim3(n_req), dim3(dim), 0, stream_,
-                       d_gout_, d_q_, n_req, dim);
+    hipLaunchKernelGGL(gemm_bf16_nt_kernel, ggrid, dim3(256), 0, st,
+                       b.d_poolbf, d_proj_, b.d_gout, n_req, dim, dim,
+                       nullptr, 0);
+    hipLaunchKernelGGL(l2norm_rows_kernel, dim3(n_req), dim3(dim), 0, st,
+                       b.d_gout, b.d_q, n_req, dim);
     // park query vectors for possible insert: one gather launch
-    memcpy(h_slots_, pending_slots, sizeof(int32_t) * (size_t)n_req);
-    HIP_OK(hipMemcpyAsync(d_slots_, h_slots_, sizeof(int32_t) * n_req,
-                          hipMemcpyHostToDevice, stream_));
-    hipLaunchKernelGGL(park_pending_kernel, dim3(n_req), dim3(dim), 0, stream_,
-                       d_q_, d_slots_, n_req, dim, d_pending_);
+    memcpy(b.h_slots, pending_slots, sizeof(int32_t) * (size_t)n_req);
+    HIP_OK(hipMemcpyAsync(b.d_slots, b.h_slots, sizeof(int32_t) * n_req,
+                          hipMemcpyHostToDevice, st));
+    hipLaunchKernelGGL(park_pending_kernel, dim3(n_req), dim3(dim), 0, st,
+                       b.d_q, b.d_slots, n_req, dim, d_pending_);
     long long rows_now = rows_visible_.load(std::memory_order_acquire);
+    b.rows_in_flight = rows_now;
     if (rows_now > 0) {
-      HIP_OK(hipMemsetAsync(d_best_, 0, sizeof(unsigned long long) * n_req,
-                            stream_));
+      HIP_OK(hipMemsetAsync(b.d_best, 0, sizeof(unsigned long long) * n_req, st));
       constexpr int ROWTILES = 16;
       long long blocks = (rows_now + 64 * ROWTILES - 1) / (64 * ROWTILES);
       if (fp8_) {
-        // quantize the query block once; the fp8 index streams HBM at
-        // half the bytes per row (2x rows in the 288 GB budget)
+        // fp8 index streams HBM at half the bytes per row (2x rows in
+        // the 288 GB budget)
         hipLaunchKernelGGL(bf16_to_fp8_kernel,
                            dim3((n_req * dim / 2 + 255) / 256), dim3(256), 0,
-                           stream_, d_q_, d_q8_, n_req * dim);
+                           st, b.d_q, b.d_q8, n_req * dim);
         constexpr size_t L8 = 128 * (12 * 32 + 16) + 128 * 8;
         (void)hipFuncSetAttribute(
             reinterpret_cast<const void*>(
@@ -323,9 +394,9 @@ class GpuAdmissionDirect {
         for (int q0 = 0; q0 < n_req; q0 += 128) {
           int kq = (n_req - q0) < 128 ? (n_req - q0) : 128;
           hipLaunchKernelGGL((cache_topk_fp8_lds_kernel_t<12, ROWTILES>),
-                             dim3((unsigned)blocks), dim3(256), L8, stream_,
+                             dim3((unsigned)blocks), dim3(256), L8, st,
                              (const uint8_t*)d_index_, rows_now,
-                             d_q8_ + (size_t)q0 * dim, kq, dim, d_best_ + q0);
+                             b.d_q8 + (size_t)q0 * dim, kq, dim, b.d_best + q0);
         }
       } else {
         constexpr size_t L = 128 * (12 * 32 + 8) * sizeof(bf16) + 128 * 8;
@@ -335,36 +406,26 @@ class GpuAdmissionDirect {
         for (int q0 = 0; q0 < n_req; q0 += 128) {
           int kq = (n_req - q0) < 128 ? (n_req - q0) : 128;
           hipLaunchKernelGGL((cache_topk_lds_kernel_t<12, ROWTILES>),
-                             dim3((unsigned)blocks), dim3(256), L, stream_,
+                             dim3((unsigned)blocks), dim3(256), L, st,
                              (const bf16*)d_index_, rows_now,
-                             d_q_ + (size_t)q0 * dim, kq, dim, d_best_ + q0);
+                             b.d_q + (size_t)q0 * dim, kq, dim, b.d_best + q0);
         }
       }
-      HIP_OK(hipMemcpyAsync(h_best_, d_best_,
+      HIP_OK(hipMemcpyAsync(b.h_best, b.d_best,
                             sizeof(unsigned long long) * n_req,
-                            hipMemcpyDeviceToHost, stream_));
-    }
-    HIP_OK(hipMemcpyAsync(h_counts_, d_counts_, sizeof(int32_t) * n_req,
-                          hipMemcpyDeviceToHost, stream_));
-    HIP_OK(hipEventRecord(event_, stream_));
-    HIP_OK(hipEventSynchronize(event_));
-    memcpy(counts_out, h_counts_, sizeof(int32_t) * (size_t)n_req);
-    for (int i = 0; i < n_req; ++i) {
-      rows_out[i] = -1;
-      scores_out[i] = 0.f;
-      if (rows_now > 0) {
-        unsigned long long u = h_best_[i];
-        unsigned hi = (unsigned)(u >> 32);
-        unsigned bits = (hi >= 0x80000000u) ? (hi ^ 0x80000000u) : ~hi;
-        float score;
-        memcpy(&score, &bits, 4);
-        if (score >= threshold_) {
-          rows_out[i] = (int32_t)(u & 0xFFFFFFFFu);
-          scores_out[i] = score;
-        }
-      }
+                            hipMemcpyDeviceToHost, st));
     }
     return true;
+  }
+
+  // Synchronous single-set wrapper (simple callers / tests).
+  bool count_lookup(const char* bytes, size_t n, const int64_t* offsets,
+                    int n_req, int32_t* counts_out,
+                    const int32_t* pending_slots, int32_t* rows_out,
+                    float* scores_out) {
+    if (!cache_on_) return false;
+    if (!submit(0, bytes, n, offsets, n_req, pending_slots)) return false;
+    return wait_set(0, n_req, counts_out, rows_out, scores_out);
   }
 
   // Append a parked query vector to the index ring; returns the row.
@@ -399,26 +460,32 @@ class GpuAdmissionDirect {
 
   ~GpuAdmissionDirect() {
     if (!ready_) return;
-    (void)hipStreamDestroy(stream_);
-    (void)hipEventDestroy(event_);
-    (void)hipHostFree(h_bytes_);
-    (void)hipHostFree(h_off_);
-    (void)hipHostFree(h_counts_);
-    if (cache_on_) {
-      (void)hipStreamDestroy(insert_stream_);
-      (void)hipHostFree(h_best_);
-      (void)hipHostFree(h_slots_);
-      for (void* p : {(void*)d_emb_, (void*)d_proj_, (void*)d_index_,
-                      (void*)d_pool_, (void*)d_poolcnt_, (void*)d_poolbf_,
-                      (void*)d_gout_, (void*)d_q_, (void*)d_q8_,
-                      (void*)d_best_, (void*)d_pending_, (void*)d_slots_})
+    for (int s = 0; s < kSets; ++s) {
+      BatchSet& b = sets_[s];
+      (void)hipStreamDestroy(b.stream);
+      (void)hipEventDestroy(b.event);
+      (void)hipHostFree(b.h_bytes);
+      (void)hipHostFree(b.h_off);
+      (void)hipHostFree(b.h_counts);
+      (void)hipHostFree(b.h_best);
+      (void)hipHostFree(b.h_slots);
+      for (void* p : {(void*)b.d_bytes, (void*)b.d_off, (void*)b.d_flags,
+                      (void*)b.d_gflags, (void*)b.d_blk, (void*)b.d_excl,
+                      (void*)b.d_totals, (void*)b.d_seg_start,
+                      (void*)b.d_seg_req, (void*)b.d_ghead,
+                      (void*)b.d_out_ids, (void*)b.d_counts, (void*)b.d_pool,
+                      (void*)b.d_poolcnt, (void*)b.d_poolbf, (void*)b.d_gout,
+                      (void*)b.d_q, (void*)b.d_q8, (void*)b.d_best,
+                      (void*)b.d_slots})
         (void)hipFree(p);
     }
-    for (void* p : {(void*)d_htab_keys_, (void*)d_htab_rank_, (void*)d_bytes_,
-                    (void*)d_off_, (void*)d_flags_, (void*)d_gflags_,
-                    (void*)d_blk_, (void*)d_excl_, (void*)d_totals_,
-                    (void*)d_seg_start_, (void*)d_seg_req_, (void*)d_ghead_,
-                    (void*)d_out_ids_, (void*)d_counts_})
+    if (cache_on_) {
+      (void)hipStreamDestroy(insert_stream_);
+      for (void* p : {(void*)d_emb_, (void*)d_proj_, (void*)d_index_,
+                      (void*)d_pending_})
+        (void)hipFree(p);
+    }
+    for (void* p : {(void*)d_htab_keys_, (void*)d_htab_rank_})
       (void)hipFree(p);
   }
 
@@ -435,44 +502,18 @@ class GpuAdmissionDirect {
   long long head_ = 0;  // writers hold the caller's insert mutex
   std::atomic<long long> rows_visible_{0};  // read lock-free by the batcher
   float threshold_ = 0.f;
+  bool fp8_ = false;
   bf16* d_emb_ = nullptr;
   bf16* d_proj_ = nullptr;
   void* d_index_ = nullptr;
-  float* d_pool_ = nullptr;
-  int32_t* d_poolcnt_ = nullptr;
-  bf16* d_poolbf_ = nullptr;
-  float* d_gout_ = nullptr;
-  bf16* d_q_ = nullptr;
-  uint8_t* d_q8_ = nullptr;
-  bool fp8_ = false;
-  unsigned long long* d_best_ = nullptr;
-  unsigned long long* h_best_ = nullptr;
   bf16* d_pending_ = nullptr;
-  int32_t* d_slots_ = nullptr;
-  int32_t* h_slots_ = nullptr;
   hipStream_t insert_stream_{};
   size_t max_bytes_ = 0;
   int max_req_ = 0;
   int htab_mask_ = 0;
-  hipStream_t stream_{};
-  hipEvent_t event_{};
-  char* h_bytes_ = nullptr;
-  int64_t* h_off_ = nullptr;
-  int32_t* h_counts_ = nullptr;
   long long* d_htab_keys_ = nullptr;
   int32_t* d_htab_rank_ = nullptr;
-  uint8_t* d_bytes_ = nullptr;
-  int64_t* d_off_ = nullptr;
-  uint8_t* d_flags_ = nullptr;
-  uint8_t* d_gflags_ = nullptr;
-  int32_t* d_blk_ = nullptr;
-  int32_t* d_excl_ = nullptr;
-  int32_t* d_totals_ = nullptr;
-  int32_t* d_seg_start_ = nullptr;
-  int32_t* d_seg_req_ = nullptr;
-  int32_t* d_ghead_ = nullptr;
-  int32_t* d_out_ids_ = nullptr;
-  int32_t* d_counts_ = nullptr;
+  BatchSet sets_[kSets];
 };
 
 // C-style entry points used by fastpath.cpp (keeps fastpath free of HIP
@@ -512,6 +553,18 @@ bool admission_count_lookup(GpuAdmissionDirect* a, const char* bytes, size_t n,
 
 long long admission_cache_insert(GpuAdmissionDirect* a, int pending_slot) {
   return a->cache_insert(pending_slot);
+}
+
+bool admission_submit(GpuAdmissionDirect* a, int set, const char* bytes,
+                      size_t n, const int64_t* offsets, int n_req,
+                      const int32_t* pending_slots) {
+  return a->submit(set, bytes, n, offsets, n_req, pending_slots);
+}
+
+bool admission_wait(GpuAdmissionDirect* a, int set, int n_req,
+                    int32_t* counts_out, int32_t* rows_out,
+                    float* scores_out) {
+  return a->wait_set(set, n_req, counts_out, rows_out, scores_out);
 }
 
 }  // namespace aigw_fast

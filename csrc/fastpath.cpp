@@ -644,6 +644,11 @@ bool admission_count_lookup(GpuAdmissionDirect* a, const char* bytes, size_t n,
                             int32_t* counts_out, const int32_t* pending_slots,
                             int32_t* rows_out, float* scores_out);
 long long admission_cache_insert(GpuAdmissionDirect* a, int pending_slot);
+bool admission_submit(GpuAdmissionDirect* a, int set, const char* bytes,
+                      size_t n, const int64_t* offsets, int n_req,
+                      const int32_t* pending_slots);
+bool admission_wait(GpuAdmissionDirect* a, int set, int n_req,
+                    int32_t* counts_out, int32_t* rows_out, float* scores_out);
 
 // Adaptive batching with NO timer window: one batcher thread drains
 // whatever accumulated while the previous GPU batch ran — the kernel
@@ -761,19 +766,75 @@ class DirectGpuBatcher {
   };
 
  private:
+  // Two-deep software pipeline over the two admission BatchSets: while
+  // batch N runs on set s's stream, batch N+1 is collected, packed, and
+  // submitted on set 1-s. The GPU pipeline is launch/latency-bound per
+  // batch (~1.5 ms), so overlapping whole batches (plus the CPU-side
+  // pack/fulfill work) nearly doubles admission throughput under load.
+  struct InFlight {
+    std::vector<std::shared_ptr<Waiter2>> waiters;
+    std::vector<int32_t> slots;
+    int64_t t0 = 0;
+    bool valid = false;
+  };
+
+  // retire a submitted batch: block on its completion event, decode the
+  // D2H results, fulfill the waiters, record batch stats
+  void finish_set(int s) {
+    InFlight& fl = inflight_[s];
+    if (!fl.valid) return;
+    int n = (int)fl.waiters.size();
+    std::vector<int32_t> counts((size_t)n, 0);
+    std::vector<int32_t> rows((size_t)n, -1);
+    std::vector<float> scores((size_t)n, 0.f);
+    bool want_cache = cache_on_.load();
+    if (!admission_wait(adm_, s, n, counts.data(),
+                        want_cache ? rows.data() : nullptr,
+                        want_cache ? scores.data() : nullptr))
+      stats_errors++;
+    int64_t bt = now_us() - fl.t0;
+    stats_batches++;
+    stats_texts += (uint64_t)n;
+    stats_time_us += (uint64_t)bt;
+    uint64_t prev = stats_max_us.load();
+    while ((uint64_t)bt > prev &&
+           !stats_max_us.compare_exchange_weak(prev, (uint64_t)bt)) {}
+    for (int i = 0; i < n; ++i) {
+      auto& w = fl.waiters[i];
+      std::lock_guard<std::mutex> lk(w->m);
+      w->count = counts[i];
+      w->row = rows[i];
+      w->score = scores[i];
+      w->slot = fl.slots[i];
+      w->done = true;
+      w->cv.notify_all();
+    }
+    fl.waiters.clear();
+    fl.slots.clear();
+    fl.valid = false;
+  }
+
   void loop() {
     std::string packed;
     std::vector<int64_t> offs;
-    std::vector<int32_t> counts;
-    std::vector<int32_t> slots, rows;
-    std::vector<float> scores;
+    int next_set = 0;
     while (!stopping_) {
       std::vector<std::string> texts;
       std::vector<std::shared_ptr<Waiter2>> waiters;
       {
         std::unique_lock<std::mutex> lk(mu_);
-        cv_.wait(lk, [&] { return stopping_ || !q_texts_.empty(); });
-        if (stopping_) return;
+        if (q_texts_.empty()) {
+          int other = 1 - next_set;
+          if (inflight_[other].valid) {
+            // nothing queued yet: retire the running batch first (its
+            // kernel time is the coalescing window for the next batch)
+            lk.unlock();
+            finish_set(other);
+            continue;
+          }
+          cv_.wait(lk, [&] { return stopping_ || !q_texts_.empty(); });
+          if (stopping_) break;
+        }
         size_t take = 0, bytes = 0;
         while (take < q_texts_.size() && (int)take < max_batch_ &&
                bytes + q_texts_[take].size() <= max_bytes_)
@@ -792,10 +853,7 @@ class DirectGpuBatcher {
         packed.append(t.data(), std::min(t.size(), room));
       }
       size_t nt = texts.size();
-      counts.assign(nt, 0);
-      rows.assign(nt, -1);
-      scores.assign(nt, 0.f);
-      slots.assign(nt, -1);
+      std::vector<int32_t> slots((size_t)nt, -1);
       if (cache_on_) {
         std::lock_guard<std::mutex> lk(slot_mu_);
         for (size_t i = 0; i < nt && !free_slots_.empty(); ++i) {
@@ -803,34 +861,33 @@ class DirectGpuBatcher {
           free_slots_.pop_back();
         }
       }
-      int64_t bt0 = now_us();
-      if (!packed.empty()) {
-        bool ok = cache_on_
-                      ? admission_count_lookup(adm_, packed.data(), packed.size(),
-                                               offs.data(), (int)nt,
-                                               counts.data(), slots.data(),
-                                               rows.data(), scores.data())
-                      : admission_count(adm_, packed.data(), packed.size(),
-                                        offs.data(), (int)nt, counts.data());
-        if (!ok) stats_errors++;
+      finish_set(next_set);  // the set must be idle before reuse
+      bool ok = !packed.empty() &&
+                admission_submit(adm_, next_set, packed.data(), packed.size(),
+                                 offs.data(), (int)nt,
+                                 cache_on_ ? slots.data() : nullptr);
+      if (!ok) {
+        if (!packed.empty()) stats_errors++;
+        for (size_t i = 0; i < nt; ++i) {
+          auto& w = waiters[i];
+          std::lock_guard<std::mutex> lk(w->m);
+          w->slot = slots[i];
+          w->done = true;
+          w->cv.notify_all();
+        }
+        continue;
       }
-      int64_t bt = now_us() - bt0;
-      stats_batches++;
-      stats_texts += (uint64_t)nt;
-      stats_time_us += (uint64_t)bt;
-      uint64_t prev = stats_max_us.load();
-      while ((uint64_t)bt > prev && !stats_max_us.compare_exchange_weak(prev, (uint64_t)bt)) {}
-      for (size_t i = 0; i < waiters.size(); ++i) {
-        auto& w = waiters[i];
-        std::lock_guard<std::mutex> lk(w->m);
-        w->count = counts[i];
-        w->row = rows[i];
-        w->score = scores[i];
-        w->slot = slots[i];
-        w->done = true;
-        w->cv.notify_all();
-      }
+      InFlight& fl = inflight_[next_set];
+      fl.waiters = std::move(waiters);
+      fl.slots = std::move(slots);
+      fl.t0 = now_us();
+      fl.valid = true;
+      // overlap: retire the OLDER batch while this one runs on the GPU
+      finish_set(1 - next_set);
+      next_set ^= 1;
     }
+    finish_set(0);
+    finish_set(1);
   }
 
   void fail_all() {
@@ -869,6 +926,7 @@ class DirectGpuBatcher {
   std::condition_variable cv_;
   std::vector<std::string> q_texts_;
   std::vector<std::shared_ptr<Waiter2>> q_waiters_;
+  InFlight inflight_[2];
   std::thread worker_;
 };
 

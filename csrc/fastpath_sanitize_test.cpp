@@ -39,6 +39,13 @@ bool admission_count_lookup(GpuAdmissionDirect*, const char*, size_t,
   return false;
 }
 long long admission_cache_insert(GpuAdmissionDirect*, int) { return -1; }
+bool admission_submit(GpuAdmissionDirect*, int, const char*, size_t,
+                      const int64_t*, int, const int32_t*) {
+  return false;
+}
+bool admission_wait(GpuAdmissionDirect*, int, int, int32_t*, int32_t*, float*) {
+  return false;
+}
 void admission_destroy(GpuAdmissionDirect*) {}
 }  // namespace aigw_fast
 
