@@ -531,9 +531,14 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
         payloads = [json.dumps(pl).encode()]
     # the cgroup CPU quota is shared by ALL ranks on a node: divide the
     # measured single-rank sweet spot (quota/4 workers) by the world size
-    # so an 8-rank scale run does not earn CFS throttle stalls
+    # so an 8-rank scale run does not earn CFS throttle stalls. Without
+    # GPU admission the relay threads never sleep (no ~1ms batch wait
+    # parking them), so the CFS knee arrives at half the connection
+    # count — size nogpu mode to quota/8 (64 conns at the 16-CPU quota:
+    # 82k @ p99 0.9ms, vs 33k @ p99 88ms at 128 conns).
+    div = 4 if gpu_direct else 8
     workers = args.workers if args.workers > 0 else max(
-        1, min(8, _cpu_quota() // (4 * max(world, 1))))
+        1, min(8, _cpu_quota() // (div * max(world, 1))))
     conns = args.batch * workers
     waves = max(args.waves, 1)
     path = "/v1/chat/completions"
