@@ -853,6 +853,15 @@ class DirectGpuBatcher {
     std::vector<int64_t> offs;
     int next_set = 0;
     while (!stopping_) {
+      {
+        // wait for a free set FIRST: the time until the completer frees
+        // one is the coalescing window, so collecting the batch as late
+        // as possible maximizes batch size and means no request misses
+        // a batch it could have joined
+        std::unique_lock<std::mutex> lk(cmu_);
+        free_cv_.wait(lk, [&] { return stopping_ || !set_busy_[next_set]; });
+        if (stopping_) break;
+      }
       std::vector<std::string> texts;
       std::vector<std::shared_ptr<Waiter2>> waiters;
       {
@@ -883,22 +892,6 @@ class DirectGpuBatcher {
         for (size_t i = 0; i < nt && !free_slots_.empty(); ++i) {
           slots[i] = free_slots_.back();
           free_slots_.pop_back();
-        }
-      }
-      {
-        // the set must be idle (completed + fulfilled) before reuse
-        std::unique_lock<std::mutex> lk(cmu_);
-        free_cv_.wait(lk, [&] { return stopping_ || !set_busy_[next_set]; });
-        if (stopping_ && set_busy_[next_set]) {
-          lk.unlock();
-          for (size_t i = 0; i < nt; ++i) {
-            auto& w = waiters[i];
-            std::lock_guard<std::mutex> wl(w->m);
-            w->slot = slots[i];
-            w->done = true;
-            w->cv.notify_all();
-          }
-          break;
         }
       }
       bool ok = !packed.empty() &&
