@@ -664,7 +664,6 @@ class DirectGpuBatcher {
     max_batch_ = std::min(max_batch, max_req);
     max_bytes_ = max_bytes;
     worker_ = std::thread([this] { loop(); });
-    completer_ = std::thread([this] { completer_loop(); });
     return true;
   }
 
@@ -674,13 +673,7 @@ class DirectGpuBatcher {
       std::lock_guard<std::mutex> lk(mu_);
       cv_.notify_all();
     }
-    {
-      std::lock_guard<std::mutex> lk(cmu_);
-      submit_cv_.notify_all();
-      free_cv_.notify_all();
-    }
     if (worker_.joinable()) worker_.join();
-    if (completer_.joinable()) completer_.join();
     if (adm_ != nullptr) admission_destroy(adm_), adm_ = nullptr;
     fail_all();
   }
@@ -773,15 +766,11 @@ class DirectGpuBatcher {
   };
 
  private:
-  // Producer/consumer pipeline over the two admission BatchSets. The
-  // batcher thread collects, packs, and SUBMITS batches; a dedicated
-  // completer thread waits on each set's completion event, decodes the
-  // D2H results, and fulfills the waiters. Splitting submission from
-  // completion matters because fulfillment is real CPU work (~100
-  // waiter mutex+notify pairs per batch): single-threaded, the batch
-  // cycle was pack+submit+wait+fulfill serialized (~1.3 ms); split,
-  // batch N+1 is packed and submitted while N is fulfilled and N+1
-  // runs on the GPU.
+  // Two-deep software pipeline over the two admission BatchSets: while
+  // batch N runs on set s's stream, batch N+1 is collected, packed, and
+  // submitted on set 1-s. The GPU pipeline is launch/latency-bound per
+  // batch (~1.5 ms), so overlapping whole batches (plus the CPU-side
+  // pack/fulfill work) nearly doubles admission throughput under load.
   struct InFlight {
     std::vector<std::shared_ptr<Waiter2>> waiters;
     std::vector<int32_t> slots;
@@ -825,49 +814,27 @@ class DirectGpuBatcher {
     fl.valid = false;
   }
 
-  void completer_loop() {
-    for (;;) {
-      int s = -1;
-      {
-        std::unique_lock<std::mutex> lk(cmu_);
-        submit_cv_.wait(lk, [&] { return stopping_ || subq_n_ > 0; });
-        if (subq_n_ == 0) {
-          if (stopping_) return;  // drained: exit only with an empty queue
-          continue;
-        }
-        s = subq_[0];
-        subq_[0] = subq_[1];
-        subq_n_--;
-      }
-      finish_set(s);  // event wait + decode + fulfill, no lock held
-      {
-        std::lock_guard<std::mutex> lk(cmu_);
-        set_busy_[s] = false;
-      }
-      free_cv_.notify_one();
-    }
-  }
-
   void loop() {
     std::string packed;
     std::vector<int64_t> offs;
     int next_set = 0;
     while (!stopping_) {
-      {
-        // wait for a free set FIRST: the time until the completer frees
-        // one is the coalescing window, so collecting the batch as late
-        // as possible maximizes batch size and means no request misses
-        // a batch it could have joined
-        std::unique_lock<std::mutex> lk(cmu_);
-        free_cv_.wait(lk, [&] { return stopping_ || !set_busy_[next_set]; });
-        if (stopping_) break;
-      }
       std::vector<std::string> texts;
       std::vector<std::shared_ptr<Waiter2>> waiters;
       {
         std::unique_lock<std::mutex> lk(mu_);
-        cv_.wait(lk, [&] { return stopping_ || !q_texts_.empty(); });
-        if (stopping_) break;
+        if (q_texts_.empty()) {
+          int other = 1 - next_set;
+          if (inflight_[other].valid) {
+            // nothing queued yet: retire the running batch first (its
+            // kernel time is the coalescing window for the next batch)
+            lk.unlock();
+            finish_set(other);
+            continue;
+          }
+          cv_.wait(lk, [&] { return stopping_ || !q_texts_.empty(); });
+          if (stopping_) break;
+        }
         size_t take = 0, bytes = 0;
         while (take < q_texts_.size() && (int)take < max_batch_ &&
                bytes + q_texts_[take].size() <= max_bytes_)
@@ -894,6 +861,7 @@ class DirectGpuBatcher {
           free_slots_.pop_back();
         }
       }
+      finish_set(next_set);  // the set must be idle before reuse
       bool ok = !packed.empty() &&
                 admission_submit(adm_, next_set, packed.data(), packed.size(),
                                  offs.data(), (int)nt,
@@ -914,14 +882,12 @@ class DirectGpuBatcher {
       fl.slots = std::move(slots);
       fl.t0 = now_us();
       fl.valid = true;
-      {
-        std::lock_guard<std::mutex> lk(cmu_);
-        set_busy_[next_set] = true;
-        subq_[subq_n_++] = next_set;
-      }
-      submit_cv_.notify_one();
+      // overlap: retire the OLDER batch while this one runs on the GPU
+      finish_set(1 - next_set);
       next_set ^= 1;
     }
+    finish_set(0);
+    finish_set(1);
   }
 
   void fail_all() {
@@ -961,14 +927,7 @@ class DirectGpuBatcher {
   std::vector<std::string> q_texts_;
   std::vector<std::shared_ptr<Waiter2>> q_waiters_;
   InFlight inflight_[2];
-  std::mutex cmu_;  // guards subq_/set_busy_ (submission->completion handoff)
-  std::condition_variable submit_cv_;
-  std::condition_variable free_cv_;
-  int subq_[2] = {0, 0};
-  int subq_n_ = 0;
-  bool set_busy_[2] = {false, false};
   std::thread worker_;
-  std::thread completer_;
 };
 
 // -------- connection handler ------------------------------------------------
