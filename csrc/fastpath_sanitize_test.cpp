@@ -18,35 +18,66 @@
 #include "fastpath.h"
 
 namespace aigw_fast {
-// GPU admission stubs (admission.hip is hipcc-only; the sanitizer lane
-// runs the GPU-off code path)
-class GpuAdmissionDirect {};
+// Functional FAKE admission (admission.hip is hipcc-only): a 2-set
+// in-memory emulation so the sanitizer lane exercises the REAL
+// DirectGpuBatcher threading — queue handoff, two-set pipelining,
+// waiter fulfillment — with deterministic counts (bytes/3) and a
+// sleep standing in for the kernel time.
+class GpuAdmissionDirect {
+ public:
+  struct FakeSet {
+    std::vector<int32_t> counts;
+    int n_req = 0;
+  };
+  FakeSet sets[2];
+};
 GpuAdmissionDirect* admission_create(const long long*, const int32_t*, int,
                                      size_t, int, int) {
-  return nullptr;
+  return new GpuAdmissionDirect();
 }
-bool admission_count(GpuAdmissionDirect*, const char*, size_t, const int64_t*,
-                     int, int32_t*) {
-  return false;
+bool admission_submit(GpuAdmissionDirect* a, int set, const char*, size_t n,
+                      const int64_t* offsets, int n_req, const int32_t*) {
+  auto& s = a->sets[set];
+  s.counts.assign((size_t)n_req, 0);
+  s.n_req = n_req;
+  for (int i = 0; i < n_req; ++i) {
+    long long b = offsets[i];
+    long long e = (i + 1 < n_req) ? offsets[i + 1] : (long long)n;
+    s.counts[(size_t)i] = (int32_t)((e - b) / 3);
+  }
+  return true;
+}
+bool admission_wait(GpuAdmissionDirect* a, int set, int n_req,
+                    int32_t* counts_out, int32_t* rows_out,
+                    float* scores_out) {
+  std::this_thread::sleep_for(std::chrono::microseconds(200));
+  auto& s = a->sets[set];
+  if (s.n_req != n_req) return false;
+  for (int i = 0; i < n_req; ++i) {
+    counts_out[i] = s.counts[(size_t)i];
+    if (rows_out != nullptr) rows_out[i] = -1;
+    if (scores_out != nullptr) scores_out[i] = 0.f;
+  }
+  return true;
+}
+bool admission_count(GpuAdmissionDirect* a, const char* bytes, size_t n,
+                     const int64_t* offsets, int n_req, int32_t* counts_out) {
+  return admission_submit(a, 0, bytes, n, offsets, n_req, nullptr) &&
+         admission_wait(a, 0, n_req, counts_out, nullptr, nullptr);
 }
 bool admission_init_cache(GpuAdmissionDirect*, const uint16_t*, int,
                           const uint16_t*, int, long long, float, int, bool) {
   return false;
 }
-bool admission_count_lookup(GpuAdmissionDirect*, const char*, size_t,
-                            const int64_t*, int, int32_t*, const int32_t*,
-                            int32_t*, float*) {
-  return false;
+bool admission_count_lookup(GpuAdmissionDirect* a, const char* bytes, size_t n,
+                            const int64_t* offsets, int n_req,
+                            int32_t* counts_out, const int32_t*,
+                            int32_t* rows_out, float* scores_out) {
+  return admission_submit(a, 0, bytes, n, offsets, n_req, nullptr) &&
+         admission_wait(a, 0, n_req, counts_out, rows_out, scores_out);
 }
 long long admission_cache_insert(GpuAdmissionDirect*, int) { return -1; }
-bool admission_submit(GpuAdmissionDirect*, int, const char*, size_t,
-                      const int64_t*, int, const int32_t*) {
-  return false;
-}
-bool admission_wait(GpuAdmissionDirect*, int, int, int32_t*, int32_t*, float*) {
-  return false;
-}
-void admission_destroy(GpuAdmissionDirect*) {}
+void admission_destroy(GpuAdmissionDirect* a) { delete a; }
 }  // namespace aigw_fast
 
 using namespace aigw_fast;
@@ -140,6 +171,24 @@ int main() {
     assert(denied.completed == 0 && denied.errors == 1);
     assert(srv2.stats().local_429.load() == 1);
     srv2.stop();
+  }
+
+  // pipelined GPU admission batcher under concurrent load: the fake
+  // admission routes every request through the real two-set
+  // DirectGpuBatcher threading (queue -> pack -> submit/wait rotation
+  // -> waiter fulfillment), racing stop() against in-flight batches
+  {
+    FastServer srv3;
+    srv3.add_route(make_route("r", "model-a", up_port));
+    std::vector<long long> hk(16, -1);
+    std::vector<int32_t> hr(16, -1);
+    srv3.enable_gpu_direct(hk.data(), hr.data(), 16, 256, 1 << 20, 512, 0);
+    int p3 = srv3.start("127.0.0.1", 0);
+    LoadResult g = run_load("127.0.0.1", p3, "/v1/chat/completions", payload,
+                            8, 100);
+    assert(g.errors == 0 && g.completed == 800);
+    assert(srv3.stats().gpu_tokens.load() > 0);
+    srv3.stop();  // joins the batcher with sets possibly in flight
   }
 
   int left = srv.drain(2.0);
