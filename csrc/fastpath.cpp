@@ -689,6 +689,16 @@ class DirectGpuBatcher {
     return cl.tokens;
   }
 
+  struct Waiter2 {
+    std::mutex m;
+    std::condition_variable cv;
+    bool done = false;
+    int64_t count = 0;
+    int32_t row = -1;
+    float score = 0.f;
+    int32_t slot = -1;
+  };
+
   struct CacheLookup {
     int64_t tokens = 0;
     int32_t row = -1;      // index row of a hit (score >= threshold)
@@ -696,12 +706,14 @@ class DirectGpuBatcher {
     int32_t slot = -1;     // pending-pool slot holding the query vector
   };
 
-  // count + (when the cache is enabled) lookup in ONE GPU batch. The
-  // returned slot (if >= 0) must be passed to cache_insert_slot() or
-  // cache_release_slot() by the caller exactly once.
-  CacheLookup count_lookup_text(const std::string& text) {
-    CacheLookup out;
-    if (stopping_) return out;
+  // Async split: enqueue() submits a text for the next batch and
+  // returns a handle; collect() blocks until the batch completes. The
+  // serving path enqueues BEFORE dispatching upstream and collects at
+  // response time, hiding the admission batch behind the upstream
+  // round-trip (the reference also charges usage post-response:
+  // processor_impl.go buildDynamicMetadata runs on response complete).
+  std::shared_ptr<Waiter2> enqueue(const std::string& text) {
+    if (stopping_) return nullptr;
     auto w = std::make_shared<Waiter2>();
     {
       std::lock_guard<std::mutex> lk(mu_);
@@ -709,6 +721,12 @@ class DirectGpuBatcher {
       q_waiters_.push_back(w);
       cv_.notify_one();
     }
+    return w;
+  }
+
+  CacheLookup collect(const std::shared_ptr<Waiter2>& w) {
+    CacheLookup out;
+    if (w == nullptr) return out;
     std::unique_lock<std::mutex> lk(w->m);
     w->cv.wait_for(lk, std::chrono::seconds(30), [&] { return w->done; });
     out.tokens = w->count;
@@ -716,6 +734,13 @@ class DirectGpuBatcher {
     out.score = w->score;
     out.slot = w->slot;
     return out;
+  }
+
+  // count + (when the cache is enabled) lookup in ONE GPU batch. The
+  // returned slot (if >= 0) must be passed to cache_insert_slot() or
+  // cache_release_slot() by the caller exactly once.
+  CacheLookup count_lookup_text(const std::string& text) {
+    return collect(enqueue(text));
   }
 
   bool init_cache(const uint16_t* emb, int vocab, const uint16_t* proj,
@@ -755,15 +780,6 @@ class DirectGpuBatcher {
     free_slots_.push_back(slot);
   }
 
-  struct Waiter2 {
-    std::mutex m;
-    std::condition_variable cv;
-    bool done = false;
-    int64_t count = 0;
-    int32_t row = -1;
-    float score = 0.f;
-    int32_t slot = -1;
-  };
 
  private:
   // Two-deep software pipeline over the two admission BatchSets: while
@@ -1076,6 +1092,7 @@ class ConnHandler {
     int64_t gpu_tokens = 0;
     int32_t cache_slot = -1;
     uint64_t cache_fp = 0;
+    pending_count_.reset();  // per-request deferred admission handle
     bool cache_eligible = srv_->gpu_direct_ != nullptr &&
                           srv_->gpu_direct_->cache_on() && !stream &&
                           req.path == "/v1/chat/completions" &&
@@ -1132,8 +1149,15 @@ class ConnHandler {
       srv_->stats_.cache_misses++;
     } else if (srv_->gpu_enabled() && req.path == "/v1/chat/completions" &&
                !sc.text.empty()) {
-      gpu_tokens = srv_->gpu_count(sc.text);
-      srv_->stats_.gpu_tokens += (uint64_t)gpu_tokens;
+      if (srv_->gpu_direct_ != nullptr) {
+        // overlap the admission batch with the upstream round-trip:
+        // the count is only needed at response time (usage fallback +
+        // rate-limit charge), so enqueue now, collect in finish_usage
+        pending_count_ = srv_->gpu_direct_->enqueue(sc.text);
+      } else {
+        gpu_tokens = srv_->gpu_count(sc.text);
+        srv_->stats_.gpu_tokens += (uint64_t)gpu_tokens;
+      }
     }
 
     // attempt order: tiers by priority, weighted shuffle inside a tier
@@ -1157,6 +1181,7 @@ class ConnHandler {
       // ORIGINAL body (per-try translation, A.8)
     }
     if (cache_slot >= 0) srv_->gpu_direct_->cache_release_slot(cache_slot);
+    pending_count_.reset();  // failed request: nothing to charge
     srv_->stats_.responses_5xx++;
     return simple_reply(503, "upstream_error", "no healthy upstream", false);
   }
@@ -1602,7 +1627,23 @@ class ConnHandler {
     return !client_close && !up_err;
   }
 
+  // collect the deferred admission count (enqueued pre-dispatch): by
+  // now the batch has usually completed while the upstream round-trip
+  // was in flight, so this wait is ~zero
+  int64_t resolve_pending_count() {
+    if (pending_count_ == nullptr) return 0;
+    auto cl = srv_->gpu_direct_->collect(pending_count_);
+    pending_count_.reset();
+    srv_->gpu_direct_->cache_release_slot(cl.slot);
+    srv_->stats_.gpu_tokens += (uint64_t)cl.tokens;
+    return cl.tokens;
+  }
+
   void finish_usage(Usage& u, int64_t gpu_tokens) {
+    if (pending_count_ != nullptr) {
+      int64_t t = resolve_pending_count();
+      if (gpu_tokens == 0) gpu_tokens = t;
+    }
     if (u.total == 0 && gpu_tokens > 0) {
       // upstream reported no usage: gateway's own GPU-tokenized count
       u.input = gpu_tokens;
@@ -1654,6 +1695,9 @@ class ConnHandler {
 
   FastServer* srv_;
   int fd_;
+  // deferred admission count for the CURRENT request (enqueued before
+  // upstream dispatch, collected in finish_usage)
+  std::shared_ptr<DirectGpuBatcher::Waiter2> pending_count_;
 
   friend class FastServer;
 };
