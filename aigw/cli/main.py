@@ -117,6 +117,15 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
         watcher = ConfigWatcher(watch_path, _on_reload)
         await watcher.start()
 
+    # dynamic InferencePool membership (inferencepool.go analogue):
+    # DNS / members-file re-resolution swapping the runtime per change
+    pool_mgr = None
+    if any(r.pool is not None for r in cfg.routes):
+        from aigw.extproc.pool import PoolManager
+
+        pool_mgr = PoolManager(server)
+        await pool_mgr.start()
+
     # OIDC/token-exchange credential rotation (BSP rotator analogue):
     # refreshes the credential files the auth handlers read
     rotation_mgr = None
@@ -202,6 +211,8 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
     finally:
         if watcher:
             await watcher.stop()
+        if pool_mgr is not None:
+            await pool_mgr.stop()
         if rotation_mgr is not None:
             await rotation_mgr.stop()
         if sync:

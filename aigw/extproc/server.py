@@ -319,6 +319,20 @@ class GatewayServer:
             await self.telemetry.scrape_once()  # first rows before traffic
             await self.telemetry.start()
 
+    async def refresh_telemetry(self) -> None:
+        """Re-register the telemetry poller against the CURRENT runtime
+        (pool membership / hot reload changed the backend set)."""
+        if self.telemetry is not None:
+            await self.telemetry.stop()
+            self.telemetry = None
+        if any(b.telemetry is not None
+               for cr in self.runtime.routes for tier in cr.tiers for b in tier):
+            from aigw.extproc.telemetry import ReplicaTelemetry
+
+            self.telemetry = ReplicaTelemetry(self.runtime)
+            await self.telemetry.scrape_once()
+            await self.telemetry.start()
+
     async def close(self) -> None:
         if self._session is not None:
             await self._session.close()

@@ -289,6 +289,31 @@ class RateLimitRule:
 
 
 @dataclass
+class InferencePool:
+    """Dynamic pool membership (InferencePool analogue). The reference
+    resolves pool members behind a k8s selector and rewrites the Envoy
+    cluster per member (extensionserver/inferencepool.go:39-54,
+    post_cluster_modify.go:32); the single-node analogue re-resolves a
+    DNS name (headless-service style A records) or a members file on an
+    interval and swaps the runtime with the refreshed member set."""
+
+    # DNS name resolved to A records; every address becomes a member
+    service: str = ""
+    # newline-separated host:port list (airgapped / file-managed pools);
+    # takes precedence over `service` when both are set
+    members_file: str = ""
+    # member port when resolving via DNS (members_file carries its own)
+    port: int = 0
+    interval_s: float = 5.0
+    # template applied to every resolved member
+    schema: APISchema = field(default_factory=APISchema)
+    telemetry: Optional["BackendTelemetry"] = None
+    auth: Optional[BackendAuth] = None
+    timeout_s: float = 60.0
+    max_concurrency: int = 0
+
+
+@dataclass
 class Route:
     """One routing rule (filterconfig.go Config.Rules).
 
@@ -309,6 +334,9 @@ class Route:
     # (replaces the reference's external EPP service —
     # extensionserver/inferencepool.go:39-54).
     endpoint_picker: bool = False
+    # Dynamic member resolution; resolved members are APPENDED to the
+    # static `backends` list by aigw.extproc.pool.PoolManager
+    pool: Optional[InferencePool] = None
 
 
 @dataclass
@@ -536,6 +564,22 @@ def _parse_route(d, ctx) -> Route:
         kw["header_mutation"] = HeaderMutation(
             **_dc(HeaderMutation, kw["header_mutation"], f"{ctx}.headerMutation")
         )
+    if kw.get("pool") is not None:
+        pkw = _dc(InferencePool, kw["pool"], f"{ctx}.pool")
+        if "schema" in pkw:
+            pkw["schema"] = _parse_schema(pkw["schema"], f"{ctx}.pool.schema")
+        if pkw.get("auth") is not None:
+            pkw["auth"] = _parse_backend_auth(pkw["auth"], f"{ctx}.pool.auth")
+        if pkw.get("telemetry") is not None:
+            pkw["telemetry"] = BackendTelemetry(
+                **_dc(BackendTelemetry, pkw["telemetry"], f"{ctx}.pool.telemetry")
+            )
+        pool = InferencePool(**pkw)
+        if not pool.service and not pool.members_file:
+            raise ConfigError(f"{ctx}.pool: service or membersFile is required")
+        if pool.service and pool.port <= 0:
+            raise ConfigError(f"{ctx}.pool: port is required with service")
+        kw["pool"] = pool
     return Route(**kw)
 
 
@@ -610,7 +654,9 @@ def load_config(data: object) -> Config:
 def _validate(cfg: Config) -> None:
     cost_keys = {c.metadata_key for c in cfg.llm_request_costs}
     for r in cfg.routes:
-        if not r.backends:
+        if not r.backends and r.pool is None:
+            # a pool route may start empty: members arrive from the
+            # first PoolManager resolution sweep before traffic
             raise ConfigError(f"route {r.name!r} has no backends")
         for c in r.request_costs:
             cost_keys.add(c.metadata_key)

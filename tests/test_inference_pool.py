@@ -1,0 +1,184 @@
+"""Dynamic InferencePool member resolution (aigw/extproc/pool.py).
+
+Parity target: the reference resolves pool members behind a k8s selector
+and rewrites the cluster endpoints per member
+(extensionserver/inferencepool.go:39-54, post_cluster_modify.go:32).
+Here: a members file (and DNS) is re-resolved; membership changes swap
+the runtime, traffic follows the current member set, telemetry rows
+re-register, and resolution failures keep the last-known members.
+"""
+
+import asyncio
+import json
+
+import aiohttp
+from aiohttp import web
+
+from aigw.extproc.pool import PoolManager, resolve_members
+from aigw.extproc.server import GatewayServer
+from aigw.filterapi import RuntimeConfig, load_config
+from aigw.filterapi.config import ConfigError, InferencePool
+
+import pytest
+
+
+class FakeReplica:
+    def __init__(self, name):
+        self.name = name
+        self.served = 0
+        self.kv_usage = 0.1
+
+    async def chat(self, request):
+        self.served += 1
+        return web.json_response({
+            "id": "r", "object": "chat.completion", "model": self.name,
+            "choices": [{"index": 0,
+                         "message": {"role": "assistant", "content": self.name},
+                         "finish_reason": "stop"}],
+            "usage": {"prompt_tokens": 2, "completion_tokens": 1,
+                      "total_tokens": 3},
+        })
+
+    async def metrics(self, request):
+        return web.Response(
+            text=f'vllm:gpu_cache_usage_perc{{m="x"}} {self.kv_usage}\n',
+            content_type="text/plain")
+
+    async def start(self):
+        app = web.Application()
+        app.router.add_post("/v1/chat/completions", self.chat)
+        app.router.add_get("/metrics", self.metrics)
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.TCPSite(runner, "127.0.0.1", 0)
+        await site.start()
+        return runner, runner.addresses[0][1]
+
+
+def _pool_cfg(members_file, telemetry=False):
+    pool = {"membersFile": str(members_file), "intervalS": 0.05,
+            "schema": "OpenAI"}
+    if telemetry:
+        pool["telemetry"] = {"path": "/metrics", "intervalS": 0.05,
+                             "kvTotal": 1000}
+    return load_config({
+        "routes": [{"name": "pool-route", "pool": pool}],
+    })
+
+
+def test_pool_config_validation():
+    with pytest.raises(ConfigError):
+        load_config({"routes": [{"name": "r", "pool": {}}]})
+    with pytest.raises(ConfigError):  # DNS mode requires a port
+        load_config({"routes": [{"name": "r",
+                                 "pool": {"service": "svc.local"}}]})
+    cfg = load_config({"routes": [{
+        "name": "r", "pool": {"service": "svc.local", "port": 9000}}]})
+    assert cfg.routes[0].pool.service == "svc.local"
+
+
+def test_members_file_resolution(tmp_path):
+    f = tmp_path / "members"
+    f.write_text("# comment\n127.0.0.1:9001\n127.0.0.1:9002\n\n")
+    pool = InferencePool(members_file=str(f))
+    members = asyncio.run(resolve_members(pool))
+    assert members == [("127.0.0.1", 9001), ("127.0.0.1", 9002)]
+
+
+def test_dns_resolution():
+    pool = InferencePool(service="localhost", port=9000)
+    members = asyncio.run(resolve_members(pool))
+    assert ("127.0.0.1", 9000) in members
+
+
+def test_membership_changes_route_traffic(tmp_path):
+    async def run():
+        rep_a, rep_b = FakeReplica("a"), FakeReplica("b")
+        runner_a, port_a = await rep_a.start()
+        runner_b, port_b = await rep_b.start()
+        f = tmp_path / "members"
+        f.write_text(f"127.0.0.1:{port_a}\n")
+
+        server = GatewayServer(RuntimeConfig(_pool_cfg(f)))
+        await server.start()
+        mgr = PoolManager(server)
+        assert await mgr.resolve_once() is True
+
+        route = server.runtime.config.routes[0]
+        assert [b.upstream.port for b in route.backends] == [port_a]
+
+        from aigw.extproc.lean_front import serve_lean
+
+        _, port, cleanup = await serve_lean(server, "127.0.0.1", 0,
+                                            with_fallback=False)
+
+        async def burst(n=6):
+            async with aiohttp.ClientSession() as c:
+                for _ in range(n):
+                    async with c.post(
+                        f"http://127.0.0.1:{port}/v1/chat/completions",
+                        json={"model": "m",
+                              "messages": [{"role": "user", "content": "x"}]},
+                    ) as r:
+                        assert r.status == 200
+
+        await burst()
+        assert rep_a.served == 6 and rep_b.served == 0
+
+        # scale out: add replica B -> traffic spreads to the new member
+        f.write_text(f"127.0.0.1:{port_a}\n127.0.0.1:{port_b}\n")
+        assert await mgr.resolve_once() is True
+        ports = sorted(b.upstream.port
+                       for b in server.runtime.config.routes[0].backends)
+        assert ports == sorted([port_a, port_b])
+        rep_a.served = rep_b.served = 0
+        await burst(40)
+        assert rep_a.served > 0 and rep_b.served > 0
+        assert rep_a.served + rep_b.served == 40
+
+        # scale in: drop A -> only B serves
+        f.write_text(f"127.0.0.1:{port_b}\n")
+        assert await mgr.resolve_once() is True
+        rep_a.served = rep_b.served = 0
+        await burst()
+        assert rep_b.served == 6 and rep_a.served == 0
+
+        # no change -> no swap
+        assert await mgr.resolve_once() is False
+
+        # resolution failure keeps the last-known member set
+        f.unlink()
+        assert await mgr.resolve_once() is False
+        assert "pool-route" in mgr.resolve_errors
+        rep_b.served = 0
+        await burst()
+        assert rep_b.served == 6
+
+        await cleanup()
+        await server.close()
+        await runner_a.cleanup()
+        await runner_b.cleanup()
+
+    asyncio.run(run())
+
+
+def test_pool_members_register_telemetry(tmp_path):
+    async def run():
+        rep = FakeReplica("a")
+        runner, port_a = await rep.start()
+        f = tmp_path / "members"
+        f.write_text(f"127.0.0.1:{port_a}\n")
+        server = GatewayServer(RuntimeConfig(_pool_cfg(f, telemetry=True)))
+        await server.start()
+        assert server.telemetry is None  # no static telemetry backends
+        mgr = PoolManager(server)
+        await mgr.resolve_once()
+        assert server.telemetry is not None
+        name = f"pool:pool-route:127.0.0.1:{port_a}"
+        assert name in server.telemetry.rows
+        row = server.telemetry.fresh_row(name, max_age_s=5.0)
+        assert row is not None and row[1] == 1000  # kv_total from template
+        await server.close()
+        await runner.cleanup()
+
+    asyncio.run(run())
