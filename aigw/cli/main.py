@@ -37,14 +37,17 @@ def _is_crd_bundle(text: str) -> bool:
 
 
 def _load_any_config(path: str | None):
+    """Returns (config, watch_path, crd_mode). CRD bundles are watched
+    too: a change re-runs the controller translation (the single-node
+    reconcile loop; reference: controller/gateway.go watch -> push)."""
     if path is None:
-        return config_from_env(), None
+        return config_from_env(), None, False
     import os as _os
 
     if _os.path.isdir(path):  # sharded config bundle dir (filterapi.bundle)
         from aigw.filterapi import bundle as _bundle
 
-        return _bundle.load_bundle(path), path
+        return _bundle.load_bundle(path), path, False
     with open(path, "r", encoding="utf-8") as f:
         text = f.read()
     if _is_crd_bundle(text):
@@ -52,8 +55,8 @@ def _load_any_config(path: str | None):
 
         # ${ENV} expansion applies at load time on the run path (the
         # offline `aigw translate` keeps placeholders for GitOps)
-        return translate_yaml(_expand_env(text)), None
-    return load_config_file(path), path
+        return translate_yaml(_expand_env(text)), path, True
+    return load_config_file(path), path, False
 
 
 async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
@@ -63,7 +66,7 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
 
     tune_gc()
 
-    cfg, watch_path = _load_any_config(args.config)
+    cfg, watch_path, crd_mode = _load_any_config(args.config)
     runtime = RuntimeConfig(cfg)
 
     gpu_services = None
@@ -114,7 +117,7 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
             cb(rc)
 
     if watch_path:
-        watcher = ConfigWatcher(watch_path, _on_reload)
+        watcher = ConfigWatcher(watch_path, _on_reload, crd_mode=crd_mode)
         await watcher.start()
 
     # dynamic InferencePool membership (inferencepool.go analogue):

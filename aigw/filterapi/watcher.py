@@ -36,12 +36,26 @@ class ConfigWatcher:
         path: str,
         on_update: Callable[[RuntimeConfig], Optional[Awaitable[None]]],
         tick_s: float = 5.0,
+        crd_mode: bool = False,
     ):
         self.path = path
         self.on_update = on_update
         self.tick_s = tick_s
+        # crd_mode: the watched file is a CRD bundle (AIGatewayRoute /
+        # AIServiceBackend / ...); reload re-runs the controller
+        # translation — the single-node analogue of the reference's
+        # reconcile loop (controller/gateway.go watch -> filterapi push)
+        self.crd_mode = crd_mode
         self._last_key: str = ""
         self._task: Optional[asyncio.Task] = None
+
+    def _compile(self, raw: bytes):
+        if self.crd_mode:
+            from aigw.controller import translate_yaml
+            from aigw.filterapi.config import _expand_env
+
+            return translate_yaml(_expand_env(raw.decode("utf-8")))
+        return load_config(yaml.safe_load(raw))
 
     def load_once(self) -> RuntimeConfig:
         """Synchronous initial load; raises on invalid config."""
@@ -54,8 +68,11 @@ class ConfigWatcher:
             return RuntimeConfig(cfg)
         with open(self.path, "rb") as f:
             raw = f.read()
-        cfg = load_config(yaml.safe_load(raw))
-        self._last_key = cfg.uuid or hashlib.sha256(raw).hexdigest()
+        cfg = self._compile(raw)
+        # crd_mode keys on content hash only: translation may synthesize
+        # a fresh uuid per run, which would false-trigger the first tick
+        self._last_key = (hashlib.sha256(raw).hexdigest() if self.crd_mode
+                          else cfg.uuid or hashlib.sha256(raw).hexdigest())
         return RuntimeConfig(cfg)
 
     async def start(self) -> None:
@@ -95,6 +112,18 @@ class ConfigWatcher:
         else:
             with open(self.path, "rb") as f:
                 raw = f.read()
+            if self.crd_mode:
+                key = hashlib.sha256(raw).hexdigest()
+                if key == self._last_key:
+                    return
+                cfg = self._compile(raw)
+                rc = RuntimeConfig(cfg)
+                self._last_key = key
+                logger.info("CRD bundle re-translated (sha=%s)", key[:12])
+                res = self.on_update(rc)
+                if res is not None:
+                    await res
+                return
             data = yaml.safe_load(raw)
             uuid = (data or {}).get("uuid", "") if isinstance(data, dict) else ""
             key = uuid or hashlib.sha256(raw).hexdigest()

@@ -221,3 +221,61 @@ spec:
     assert m.name == "gpt-4o"
     assert m.owned_by == "my-org"
     assert m.created_at == 1714521600
+
+
+def test_crd_bundle_hot_reload(tmp_path):
+    """CRD-bundle configs are WATCHED: a change to the bundle file
+    re-runs the controller translation and swaps the runtime — the
+    single-node analogue of the reference's reconcile loop
+    (controller/gateway.go watch -> filterapi Secret push)."""
+    import asyncio
+
+    from aigw.filterapi import ConfigWatcher
+
+    bundle = """
+apiVersion: aigateway.envoyproxy.io/v1beta1
+kind: AIGatewayRoute
+metadata: {name: r, namespace: default}
+spec:
+  rules:
+    - matches:
+        - headers:
+            - {type: Exact, name: x-ai-eg-model, value: MODEL}
+      backendRefs: [{name: b}]
+---
+apiVersion: aigateway.envoyproxy.io/v1beta1
+kind: AIServiceBackend
+metadata: {name: b, namespace: default}
+spec:
+  schema: {name: OpenAI}
+  backendRef: {name: up, kind: Backend, group: gateway.envoyproxy.io}
+---
+apiVersion: gateway.envoyproxy.io/v1alpha1
+kind: Backend
+metadata: {name: up, namespace: default}
+spec:
+  endpoints: [{fqdn: {hostname: upstream.local, port: 8080}}]
+"""
+    path = tmp_path / "crds.yaml"
+    path.write_text(bundle.replace("MODEL", "gpt-4o"))
+    seen = []
+    w = ConfigWatcher(str(path), seen.append, crd_mode=True)
+    rc = w.load_once()
+    assert rc.routes[0].matches[0].value == "gpt-4o"
+
+    async def run():
+        await w._check()          # unchanged -> no callback
+        assert seen == []
+        path.write_text(bundle.replace("MODEL", "claude-3"))
+        await w._check()
+        assert len(seen) == 1
+        assert seen[0].routes[0].matches[0].value == "claude-3"
+        # a broken bundle must NOT clobber the last good runtime
+        path.write_text("apiVersion: aigateway.envoyproxy.io/v1beta1\nkind: Nope\n")
+        try:
+            await w._check()
+        except Exception:
+            pass  # watcher loop catches this in production (_loop)
+        assert len(seen) == 1
+
+    asyncio.run(run())

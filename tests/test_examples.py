@@ -52,10 +52,10 @@ def test_crd_bundle_example_runs_with_env_expansion(monkeypatch):
     monkeypatch.setenv("OPENAI_API_KEY", "sk-live-123")
     from aigw.cli.main import _load_any_config
 
-    cfg, watch = _load_any_config(
+    cfg, watch, crd_mode = _load_any_config(
         os.path.join(os.path.dirname(__file__), "..", "examples", "crd_bundle.yaml")
     )
-    assert watch is None  # CRD bundles don't hot-reload (one-shot translate)
+    assert watch is not None and crd_mode  # CRD bundles hot-reload via re-translation
     b = cfg.routes[0].backends[0]
     assert b.auth.api_key == "sk-live-123"
     assert b.stream_idle_timeout_s == 15.0
