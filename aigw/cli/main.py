@@ -126,7 +126,7 @@ async def _run_shard(args, rank: int = 0, world: int = 1) -> None:
     if any(r.pool is not None for r in cfg.routes):
         from aigw.extproc.pool import PoolManager
 
-        pool_mgr = PoolManager(server)
+        pool_mgr = PoolManager(server, swap_cb=_on_reload)
         await pool_mgr.start()
 
     # OIDC/token-exchange credential rotation (BSP rotator analogue):

@@ -81,8 +81,12 @@ class PoolManager:
     when membership actually changed.
     """
 
-    def __init__(self, server):
+    def __init__(self, server, swap_cb=None):
         self.server = server
+        # swap callback: the CLI passes its reload chain so EVERY front
+        # (including the native fast front's route table) follows
+        # membership changes, same as a file-watcher reload
+        self.swap_cb = swap_cb or server.swap_runtime
         self._task: Optional[asyncio.Task] = None
         self._last: dict[str, list[tuple[str, int]]] = {}
         self.resolve_errors: dict[str, str] = {}
@@ -125,7 +129,9 @@ class PoolManager:
             members = [build_member_backend(r.name, r.pool, host, port)
                        for host, port in desired.get(r.name, [])]
             r.backends = static + members
-        self.server.swap_runtime(RuntimeConfig(cfg))
+        res = self.swap_cb(RuntimeConfig(cfg))
+        if res is not None and hasattr(res, "__await__"):
+            await res
         await self.server.refresh_telemetry()
         self._last = desired
         logger.info("pool membership updated: %s",
