@@ -481,3 +481,65 @@ def test_fast_front_metrics_exposed_on_fallback():
         await up_runner.cleanup()
 
     asyncio.run(run())
+
+
+def test_fast_front_follows_pool_membership(tmp_path):
+    """Dynamic pool membership reaches the NATIVE front: PoolManager
+    swaps through the reload chain, so the C++ route table follows
+    members-file changes (scale-out lands new replicas in rotation)."""
+
+    async def run():
+        mock, up_runner, up_port = await start_mock_upstream()
+        mock2, up_runner2, up_port2 = await start_mock_upstream()
+        mock.record = mock2.record = True
+        f = tmp_path / "members"
+        f.write_text(f"127.0.0.1:{up_port}\n")
+        cfg = load_config({
+            "version": "v1",
+            "routes": [{
+                "name": "pool-route",
+                "headers": [{"name": "x-ai-eg-model", "value": "fast-model"}],
+                "pool": {"membersFile": str(f), "intervalS": 60,
+                         "schema": "OpenAI"},
+            }],
+        })
+        from aigw.extproc.pool import PoolManager
+        from aigw.extproc.server import GatewayServer
+        from aigw.filterapi import RuntimeConfig as _RC
+
+        server = GatewayServer(_RC(cfg))
+        await server.start()
+        mgr = PoolManager(server)
+        await mgr.resolve_once()  # inject member 1 BEFORE the front builds
+
+        from aigw.extproc.fast_front import FastFront
+
+        front = FastFront(server, server.runtime)
+        port = await front.start("127.0.0.1", 0)
+        mgr.swap_cb = lambda rc: (server.swap_runtime(rc), front.reload(rc))[0]
+
+        async def post(c):
+            async with c.post(
+                f"http://127.0.0.1:{port}/v1/chat/completions",
+                json={"model": "fast-model",
+                      "messages": [{"role": "user", "content": "x"}]},
+            ) as r:
+                assert r.status == 200
+                await r.read()
+
+        async with aiohttp.ClientSession() as c:
+            for _ in range(4):
+                await post(c)
+            assert len(mock.requests) == 4 and not mock2.requests
+            # scale out: second member appears in the native route table
+            f.write_text(f"127.0.0.1:{up_port}\n127.0.0.1:{up_port2}\n")
+            assert await mgr.resolve_once() is True
+            for _ in range(40):
+                await post(c)
+            assert mock2.requests, "new pool member never hit via fast front"
+        await front.stop()
+        await server.close()
+        await up_runner.cleanup()
+        await up_runner2.cleanup()
+
+    asyncio.run(run())
