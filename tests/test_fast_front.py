@@ -6,6 +6,8 @@ path to the Python fallback app unchanged."""
 
 import asyncio
 import json
+import socket
+import threading
 
 import aiohttp
 import pytest
@@ -541,5 +543,59 @@ def test_fast_front_follows_pool_membership(tmp_path):
         await server.close()
         await up_runner.cleanup()
         await up_runner2.cleanup()
+
+    asyncio.run(run())
+
+
+def test_fast_front_truncated_upstream_is_5xx_not_hang():
+    """An upstream that dies mid-body (content-length promised, half
+    delivered, connection closed) must produce a clean 503 — never a
+    hang, never a truncated 200 relayed to the client — and the
+    connection pool must recover for subsequent requests."""
+
+    async def run():
+        body = json.dumps({
+            "id": "x", "object": "chat.completion", "choices": [],
+            "usage": {"prompt_tokens": 5, "completion_tokens": 2,
+                      "total_tokens": 7}}).encode()
+        head = (f"HTTP/1.1 200 OK\r\ncontent-type: application/json\r\n"
+                f"content-length: {len(body)}\r\n\r\n").encode()
+        srv = socket.socket()
+        srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        srv.bind(("127.0.0.1", 0))
+        srv.listen(16)
+        port_up = srv.getsockname()[1]
+
+        def serve():
+            while True:
+                try:
+                    c, _ = srv.accept()
+                except OSError:
+                    return
+                try:
+                    c.recv(65536)
+                    c.sendall(head + body[: len(body) // 2])
+                    c.close()
+                except OSError:
+                    pass
+
+        t = threading.Thread(target=serve, daemon=True)
+        t.start()
+        front, port = await _start(_cfg(port_up))
+        payload = {"model": "fast-model",
+                   "messages": [{"role": "user", "content": "x"}]}
+        async with aiohttp.ClientSession() as c:
+            for _ in range(5):
+                async with c.post(
+                    f"http://127.0.0.1:{port}/v1/chat/completions",
+                    json=payload,
+                    timeout=aiohttp.ClientTimeout(total=5),
+                ) as r:
+                    assert r.status == 503
+                    await r.read()
+        st = front.fast.stats()
+        assert st["responses_5xx"] == 5 and st["responses_2xx"] == 0
+        await front.stop()
+        srv.close()
 
     asyncio.run(run())
