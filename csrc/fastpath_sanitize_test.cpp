@@ -184,10 +184,20 @@ int main() {
     std::vector<int32_t> hr(16, -1);
     srv3.enable_gpu_direct(hk.data(), hr.data(), 16, 256, 1 << 20, 512, 0);
     int p3 = srv3.start("127.0.0.1", 0);
+    std::thread swapper3([&] {
+      for (int i = 0; i < 30; ++i) {
+        std::vector<FastRoute> routes;
+        routes.push_back(make_route("r", "model-a", up_port));
+        srv3.swap_routes(std::move(routes));
+        std::this_thread::sleep_for(std::chrono::milliseconds(1));
+      }
+    });
     LoadResult g = run_load("127.0.0.1", p3, "/v1/chat/completions", payload,
                             8, 100);
+    swapper3.join();
     assert(g.errors == 0 && g.completed == 800);
     assert(srv3.stats().gpu_tokens.load() > 0);
+    assert(srv3.drain(2.0) == 0);
     srv3.stop();  // joins the batcher with sets possibly in flight
   }
 
