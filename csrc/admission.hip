@@ -80,21 +80,42 @@ __global__ void meanpool_tokens_kernel(
     const int32_t* __restrict__ req_excl, const int32_t* __restrict__ counts,
     int n_req, const bf16* __restrict__ emb, int dim, int P,
     float* __restrict__ out, int32_t* __restrict__ cnt) {
+  // Block (r, p) owns a CONTIGUOUS chunk of request r's tokens. The
+  // chunk's (ghead -> ids) indirections are staged through LDS by 64
+  // threads in parallel, so the embedding-row gathers in the inner loop
+  // have no dependent-load chain and the compiler can keep many in
+  // flight — the strided two-chain version measured 549 us/batch
+  // (344 GB/s, 4% of HBM peak) with this kernel as 54% of cache-mode
+  // GPU time.
+  constexpr int STAGE = 64;
+  __shared__ int32_t tok_ids[STAGE];
   int r = blockIdx.x;
   int p = blockIdx.y;
   if (r >= n_req) return;
   int col = threadIdx.x;
-  if (col >= dim) return;
   int s = req_excl[r], c = counts[r];
+  int chunk = (c + P - 1) / P;
+  int j0 = p * chunk;
+  int j1 = j0 + chunk;
+  if (j1 > c) j1 = c;
   float acc = 0.f;
   int n = 0;
-  for (int j = p; j < c; j += P) {
-    int tok = ids[ghead[s + j]];
-    if (tok < 0) continue;
-    acc += __bfloat162float(emb[(long long)tok * dim + col]);
-    ++n;
+  for (int base = j0; base < j1; base += STAGE) {
+    if (threadIdx.x < STAGE && base + (int)threadIdx.x < j1)
+      tok_ids[threadIdx.x] = ids[ghead[s + base + threadIdx.x]];
+    __syncthreads();
+    int here = j1 - base < STAGE ? j1 - base : STAGE;
+    if (col < dim) {
+      for (int q = 0; q < here; ++q) {
+        int tok = tok_ids[q];
+        if (tok < 0) continue;
+        acc += __bfloat162float(emb[(long long)tok * dim + col]);
+        ++n;
+      }
+    }
+    __syncthreads();
   }
-  if (n) {
+  if (col < dim && n) {
     atomicAdd(&out[(long long)r * dim + col], acc);
     if (col == 0) atomicAdd(&cnt[r], n);
   }
