@@ -87,7 +87,7 @@ __global__ void meanpool_tokens_kernel(
   // flight — the strided two-chain version measured 549 us/batch
   // (344 GB/s, 4% of HBM peak) with this kernel as 54% of cache-mode
   // GPU time.
-  constexpr int STAGE = 64;
+  constexpr int STAGE = 128;
   __shared__ int32_t tok_ids[STAGE];
   int r = blockIdx.x;
   int p = blockIdx.y;
@@ -409,7 +409,7 @@ class GpuAdmissionDirect {
     hipStream_t st = b.stream;
     HIP_OK(hipMemsetAsync(b.d_pool, 0, sizeof(float) * (size_t)n_req * dim, st));
     HIP_OK(hipMemsetAsync(b.d_poolcnt, 0, sizeof(int32_t) * n_req, st));
-    constexpr int P = 16;
+    constexpr int P = 32;
     hipLaunchKernelGGL(excl_scan_kernel, dim3(1), dim3(256), 0, st,
                        b.d_counts, n_req, b.d_req_excl,
                        b.d_req_excl + max_req_);
