@@ -599,3 +599,54 @@ def test_fast_front_truncated_upstream_is_5xx_not_hang():
         srv.close()
 
     asyncio.run(run())
+
+
+def test_fast_front_client_abort_mid_stream():
+    """A client that disconnects mid-SSE-relay must not wedge the relay
+    thread or poison the server: subsequent requests on fresh
+    connections keep succeeding."""
+
+    async def run():
+        from aigw.testing.fastmock import canned_chat_sse
+
+        import aigw_fast
+
+        mock = aigw_fast.FastMock()
+        port_up = mock.start("127.0.0.1",
+                             canned_chat_sse(prompt_tokens=64,
+                                             n_chunks=64).decode("latin1"))
+        front, port = await _start(_cfg(port_up))
+        payload = json.dumps({
+            "model": "fast-model", "stream": True,
+            "messages": [{"role": "user", "content": "x"}]}).encode()
+        req = (b"POST /v1/chat/completions HTTP/1.1\r\n"
+               b"host: x\r\ncontent-type: application/json\r\n"
+               b"content-length: " + str(len(payload)).encode() +
+               b"\r\n\r\n" + payload)
+        loop = asyncio.get_running_loop()
+
+        def abort_once():
+            s = socket.create_connection(("127.0.0.1", port), timeout=5)
+            s.sendall(req)
+            s.recv(256)   # first bytes of the relay
+            s.close()     # abort mid-stream
+
+        for _ in range(8):
+            await loop.run_in_executor(None, abort_once)
+        # server still healthy for well-behaved clients
+        async with aiohttp.ClientSession() as c:
+            for _ in range(4):
+                async with c.post(
+                    f"http://127.0.0.1:{port}/v1/chat/completions",
+                    json={"model": "fast-model", "stream": True,
+                          "messages": [{"role": "user", "content": "x"}]},
+                    timeout=aiohttp.ClientTimeout(total=5),
+                ) as r:
+                    assert r.status == 200
+                    body = await r.read()
+                    assert b"[DONE]" in body
+        assert front.fast.stats()["active_connections"] == 0
+        await front.stop()
+        mock.stop()
+
+    asyncio.run(run())
