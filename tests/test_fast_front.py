@@ -645,6 +645,12 @@ def test_fast_front_client_abort_mid_stream():
                     assert r.status == 200
                     body = await r.read()
                     assert b"[DONE]" in body
+        # aborted handlers notice the dead socket on their next write;
+        # give them a moment to drain
+        for _ in range(50):
+            if front.fast.stats()["active_connections"] == 0:
+                break
+            await asyncio.sleep(0.05)
         assert front.fast.stats()["active_connections"] == 0
         await front.stop()
         mock.stop()
