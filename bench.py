@@ -787,7 +787,9 @@ def main():
                 "p50_added_latency_ms": round(p50 - p50_direct, 3),
                 "gpu_token_accounting": use_gpu,
                 "gpu_admission": (
-                    dict(zip(("batches", "texts", "time_us", "max_us", "errors"),
+                    dict(zip(("batches", "texts", "time_us", "max_us",
+                              "errors", "pack_us", "submit_us", "wait_us",
+                              "fulfill_us"),
                              front.fast.gpu_direct_stats()))
                     if fast_mode and front is not None else None
                 ),

@@ -804,11 +804,14 @@ class DirectGpuBatcher {
     std::vector<int32_t> rows((size_t)n, -1);
     std::vector<float> scores((size_t)n, 0.f);
     bool want_cache = cache_on_.load();
+    int64_t w0 = now_us();
     if (!admission_wait(adm_, s, n, counts.data(),
                         want_cache ? rows.data() : nullptr,
                         want_cache ? scores.data() : nullptr))
       stats_errors++;
-    int64_t bt = now_us() - fl.t0;
+    int64_t w1 = now_us();
+    stats_wait_us += (uint64_t)(w1 - w0);
+    int64_t bt = w1 - fl.t0;
     stats_batches++;
     stats_texts += (uint64_t)n;
     stats_time_us += (uint64_t)bt;
@@ -825,6 +828,7 @@ class DirectGpuBatcher {
       w->done = true;
       w->cv.notify_all();
     }
+    stats_fulfill_us += (uint64_t)(now_us() - w1);
     fl.waiters.clear();
     fl.slots.clear();
     fl.valid = false;
@@ -861,6 +865,7 @@ class DirectGpuBatcher {
         q_texts_.erase(q_texts_.begin(), q_texts_.begin() + take);
         q_waiters_.erase(q_waiters_.begin(), q_waiters_.begin() + take);
       }
+      int64_t p0 = now_us();
       packed.clear();
       offs.clear();
       for (auto& t : texts) {
@@ -877,11 +882,14 @@ class DirectGpuBatcher {
           free_slots_.pop_back();
         }
       }
+      stats_pack_us += (uint64_t)(now_us() - p0);
       finish_set(next_set);  // the set must be idle before reuse
+      int64_t s0 = now_us();
       bool ok = !packed.empty() &&
                 admission_submit(adm_, next_set, packed.data(), packed.size(),
                                  offs.data(), (int)nt,
                                  cache_on_ ? slots.data() : nullptr);
+      stats_submit_us += (uint64_t)(now_us() - s0);
       if (!ok) {
         if (!packed.empty()) stats_errors++;
         for (size_t i = 0; i < nt; ++i) {
@@ -933,6 +941,11 @@ class DirectGpuBatcher {
   std::atomic<uint64_t> stats_time_us{0};
   std::atomic<uint64_t> stats_max_us{0};
   std::atomic<uint64_t> stats_errors{0};
+  // phase breakdown (batcher thread wall time per phase, summed)
+  std::atomic<uint64_t> stats_pack_us{0};
+  std::atomic<uint64_t> stats_submit_us{0};
+  std::atomic<uint64_t> stats_wait_us{0};
+  std::atomic<uint64_t> stats_fulfill_us{0};
 
  private:
   int max_batch_ = 1024;
@@ -1949,10 +1962,14 @@ void FastServer::enable_gpu(const std::string& socket_path, int window_us,
 }
 
 std::vector<uint64_t> FastServer::gpu_direct_stats() const {
-  if (gpu_direct_ == nullptr) return {0, 0, 0, 0, 0};
+  if (gpu_direct_ == nullptr) return {0, 0, 0, 0, 0, 0, 0, 0, 0};
   return {gpu_direct_->stats_batches.load(), gpu_direct_->stats_texts.load(),
           gpu_direct_->stats_time_us.load(), gpu_direct_->stats_max_us.load(),
-          gpu_direct_->stats_errors.load()};
+          gpu_direct_->stats_errors.load(),
+          gpu_direct_->stats_pack_us.load(),
+          gpu_direct_->stats_submit_us.load(),
+          gpu_direct_->stats_wait_us.load(),
+          gpu_direct_->stats_fulfill_us.load()};
 }
 
 bool FastServer::gpu_enabled() const {
