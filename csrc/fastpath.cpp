@@ -689,15 +689,10 @@ class DirectGpuBatcher {
     return cl.tokens;
   }
 
-  // Result holder. Waiters do NOT carry their own mutex/condvar:
-  // fulfillment is a sequence-number rendezvous — the batcher writes
-  // every result of a batch with plain stores, advances done_seq_
-  // under done_mu_, and issues ONE notify_all. The per-waiter
-  // mutex+notify design cost the batcher thread ~1.8 us x ~90 waiters
-  // per batch (163 us, measured via stats_fulfill_us) and the batcher
-  // thread is the pipeline's serial resource.
   struct Waiter2 {
-    uint64_t seq = 0;
+    std::mutex m;
+    std::condition_variable cv;
+    bool done = false;
     int64_t count = 0;
     int32_t row = -1;
     float score = 0.f;
@@ -722,7 +717,6 @@ class DirectGpuBatcher {
     auto w = std::make_shared<Waiter2>();
     {
       std::lock_guard<std::mutex> lk(mu_);
-      w->seq = next_seq_++;
       q_texts_.push_back(text);
       q_waiters_.push_back(w);
       cv_.notify_one();
@@ -733,13 +727,8 @@ class DirectGpuBatcher {
   CacheLookup collect(const std::shared_ptr<Waiter2>& w) {
     CacheLookup out;
     if (w == nullptr) return out;
-    {
-      std::unique_lock<std::mutex> lk(done_mu_);
-      done_cv_.wait_for(lk, std::chrono::seconds(30),
-                        [&] { return done_seq_ > w->seq; });
-      if (done_seq_ <= w->seq) return out;  // timed out: zeros
-    }
-    // done_mu_ orders the batcher's plain result stores before our reads
+    std::unique_lock<std::mutex> lk(w->m);
+    w->cv.wait_for(lk, std::chrono::seconds(30), [&] { return w->done; });
     out.tokens = w->count;
     out.row = w->row;
     out.score = w->score;
@@ -829,16 +818,16 @@ class DirectGpuBatcher {
     uint64_t prev = stats_max_us.load();
     while ((uint64_t)bt > prev &&
            !stats_max_us.compare_exchange_weak(prev, (uint64_t)bt)) {}
-    uint64_t last_seq = 0;
     for (int i = 0; i < n; ++i) {
       auto& w = fl.waiters[i];
+      std::lock_guard<std::mutex> lk(w->m);
       w->count = counts[i];
       w->row = rows[i];
       w->score = scores[i];
       w->slot = fl.slots[i];
-      last_seq = w->seq;
+      w->done = true;
+      w->cv.notify_all();
     }
-    if (n > 0) publish_done(last_seq + 1);
     stats_fulfill_us += (uint64_t)(now_us() - w1);
     fl.waiters.clear();
     fl.slots.clear();
@@ -903,12 +892,13 @@ class DirectGpuBatcher {
       stats_submit_us += (uint64_t)(now_us() - s0);
       if (!ok) {
         if (!packed.empty()) stats_errors++;
-        uint64_t last_seq = 0;
         for (size_t i = 0; i < nt; ++i) {
-          waiters[i]->slot = slots[i];
-          last_seq = waiters[i]->seq;
+          auto& w = waiters[i];
+          std::lock_guard<std::mutex> lk(w->m);
+          w->slot = slots[i];
+          w->done = true;
+          w->cv.notify_all();
         }
-        if (nt > 0) publish_done(last_seq + 1);
         continue;
       }
       InFlight& fl = inflight_[next_set];
@@ -925,23 +915,18 @@ class DirectGpuBatcher {
   }
 
   void fail_all() {
-    uint64_t up_to;
+    std::vector<std::shared_ptr<Waiter2>> all;
     {
       std::lock_guard<std::mutex> lk(mu_);
+      all = std::move(q_waiters_);
       q_waiters_.clear();
       q_texts_.clear();
-      up_to = next_seq_;
     }
-    publish_done(up_to);  // every enqueued waiter unblocks with zeros
-  }
-
-  // advance the completion horizon and wake every waiter once
-  void publish_done(uint64_t up_to) {
-    {
-      std::lock_guard<std::mutex> lk(done_mu_);
-      if (up_to > done_seq_) done_seq_ = up_to;
+    for (auto& w : all) {
+      std::lock_guard<std::mutex> lk(w->m);
+      w->done = true;
+      w->cv.notify_all();
     }
-    done_cv_.notify_all();
   }
 
   GpuAdmissionDirect* adm_ = nullptr;
@@ -970,10 +955,6 @@ class DirectGpuBatcher {
   std::condition_variable cv_;
   std::vector<std::string> q_texts_;
   std::vector<std::shared_ptr<Waiter2>> q_waiters_;
-  uint64_t next_seq_ = 1;  // guarded by mu_
-  std::mutex done_mu_;
-  std::condition_variable done_cv_;
-  uint64_t done_seq_ = 0;  // all seqs < done_seq_ have results published
   InFlight inflight_[2];
   std::thread worker_;
 };
