@@ -27,13 +27,20 @@ class GpuAdmissionDirect {
  public:
   struct FakeSet {
     std::vector<int32_t> counts;
+    std::vector<char> staging;
     int n_req = 0;
   };
   FakeSet sets[2];
 };
 GpuAdmissionDirect* admission_create(const long long*, const int32_t*, int,
-                                     size_t, int, int) {
-  return new GpuAdmissionDirect();
+                                     size_t max_bytes, int, int) {
+  auto* a = new GpuAdmissionDirect();
+  a->sets[0].staging.resize(max_bytes);
+  a->sets[1].staging.resize(max_bytes);
+  return a;
+}
+char* admission_staging(GpuAdmissionDirect* a, int set) {
+  return a->sets[set].staging.data();
 }
 bool admission_submit(GpuAdmissionDirect* a, int set, const char*, size_t n,
                       const int64_t* offsets, int n_req, const int32_t*) {
