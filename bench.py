@@ -633,8 +633,13 @@ def main():
     use_gpu = torch.cuda.is_available()
     if world > 1:
         backend = "nccl" if use_gpu else "gloo"
+        # AIGW_BENCH_BACKEND=gloo lets a multi-rank run share ONE GPU
+        # (RCCL refuses two ranks on a device) — used to rehearse the
+        # torchrun + fast-front + statesync + GPU-admission combination
+        # on a 1-GPU box before the driver's real 8-GPU scale run
+        backend = os.environ.get("AIGW_BENCH_BACKEND", backend)
         if use_gpu:
-            torch.cuda.set_device(local_rank)
+            torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
         torch.distributed.init_process_group(backend)
 
     fast_mode = args.front == "fast"
