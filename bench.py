@@ -42,6 +42,14 @@ WORDS = (
 ).split()
 
 
+def _gpu_device(local_rank: int) -> int:
+    """Clamp the rank's device (multi-rank rehearsal on fewer GPUs)."""
+    import torch as _t
+
+    n = max(_t.cuda.device_count(), 1)
+    return local_rank % n
+
+
 def build_payload(n_tokens: int) -> dict:
     """~n_tokens-token chat body (words ≈ tokens for the synthetic BPE)."""
     words = [WORDS[i % len(WORDS)] for i in range(n_tokens)]
@@ -130,7 +138,7 @@ async def worker_main(args, local_rank: int, ready, go, out_q):
 
     use_gpu = torch.cuda.is_available() and not getattr(args, "no_gpu", False)
     if use_gpu:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(_gpu_device(local_rank))
 
     up_server, up_port = await start_fast_mock(
         response=canned_chat_response(prompt_tokens=args.tokens)
@@ -331,7 +339,7 @@ def _start_fast_front(args, upstream_ports, gpu_socket, gpu_direct=False,
         gpu_window_us=int(args.gpu_window * 1000.0), gpu_max_batch=1024,
         gpu_direct=gpu_direct, gpu_cache=gpu_cache,
         cache_index_dtype="fp8" if getattr(args, "cache_fp8", False) else "bf16",
-        gpu_device=int(os.environ.get("LOCAL_RANK", 0)),
+        gpu_device=_gpu_device(int(os.environ.get("LOCAL_RANK", 0))),
     )
     done = threading.Event()
     state = {}
@@ -357,7 +365,7 @@ def gpu_host_entry(socket_path, local_rank, window_ms, ready_evt):
     caps near ~900 MB/s of msgpack-decoded request text (measured:
     36k req/s vs 82k without GPU), so admission decode must scale in
     processes just like the round-1 HTTP workers did."""
-    torch.cuda.set_device(local_rank)
+    torch.cuda.set_device(_gpu_device(local_rank))
     from aigw.gpu import GPUServices
     from aigw.gpu.service import GPUServiceHost
 
@@ -397,7 +405,7 @@ def run_fast_mode(args, rank, world, local_rank, use_gpu):
         # host topology measured 36k single-host / 10k 4-host vs 82k
         # without GPU work)
         gpu_direct = True
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(_gpu_device(local_rank))
     elif use_gpu:
         n_hosts = args.gpu_hosts
         host_ready = []
@@ -443,7 +451,7 @@ def _start_gpu_host(socket_path: str, local_rank: int, window_ms: float = 0.1):
     ready_evt = threading.Event()
 
     def run():
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(_gpu_device(local_rank))
         from aigw.gpu import GPUServices
         from aigw.gpu.service import GPUServiceHost
 
@@ -510,7 +518,7 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
     up_port = mock.start("127.0.0.1", canned.decode("latin1"))
     gpu_direct = use_gpu
     if gpu_direct:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(_gpu_device(local_rank))
     cache_mode = use_gpu and args.cache_payloads > 0
     front, gw_port = _start_fast_front(args, [up_port], None, gpu_direct,
                                        gpu_cache=cache_mode)
@@ -639,7 +647,7 @@ def main():
         # on a 1-GPU box before the driver's real 8-GPU scale run
         backend = os.environ.get("AIGW_BENCH_BACKEND", backend)
         if use_gpu:
-            torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+            torch.cuda.set_device(_gpu_device(local_rank))
         torch.distributed.init_process_group(backend)
 
     fast_mode = args.front == "fast"
