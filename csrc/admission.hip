@@ -87,37 +87,50 @@ __global__ void meanpool_tokens_kernel(
   // flight — the strided two-chain version measured 549 us/batch
   // (344 GB/s, 4% of HBM peak) with this kernel as 54% of cache-mode
   // GPU time.
+  // Launched with dim/2 threads: each thread owns a PAIR of columns
+  // loaded as one 4-byte bfloat16x2 (the scalar version was VALU-bound
+  // per the PMC counters — the embedding table is L2-resident — so
+  // halving the instruction stream is the lever, not bandwidth).
   constexpr int STAGE = 128;
   __shared__ int32_t tok_ids[STAGE];
   int r = blockIdx.x;
   int p = blockIdx.y;
   if (r >= n_req) return;
-  int col = threadIdx.x;
+  int dim2 = dim >> 1;
+  int col2 = threadIdx.x;
   int s = req_excl[r], c = counts[r];
   int chunk = (c + P - 1) / P;
   int j0 = p * chunk;
   int j1 = j0 + chunk;
   if (j1 > c) j1 = c;
-  float acc = 0.f;
+  const uint32_t* emb2 = reinterpret_cast<const uint32_t*>(emb);
+  float acc_lo = 0.f, acc_hi = 0.f;
   int n = 0;
   for (int base = j0; base < j1; base += STAGE) {
     if (threadIdx.x < STAGE && base + (int)threadIdx.x < j1)
       tok_ids[threadIdx.x] = ids[ghead[s + base + threadIdx.x]];
     __syncthreads();
     int here = j1 - base < STAGE ? j1 - base : STAGE;
-    if (col < dim) {
+    if (col2 < dim2) {
       for (int q = 0; q < here; ++q) {
         int tok = tok_ids[q];
         if (tok < 0) continue;
-        acc += __bfloat162float(emb[(long long)tok * dim + col]);
+        uint32_t v = emb2[(long long)tok * dim2 + col2];
+        bf16 lo, hi;
+        uint16_t vlo = (uint16_t)(v & 0xffffu), vhi = (uint16_t)(v >> 16);
+        memcpy(&lo, &vlo, 2);
+        memcpy(&hi, &vhi, 2);
+        acc_lo += __bfloat162float(lo);
+        acc_hi += __bfloat162float(hi);
         ++n;
       }
     }
     __syncthreads();
   }
-  if (col < dim && n) {
-    atomicAdd(&out[(long long)r * dim + col], acc);
-    if (col == 0) atomicAdd(&cnt[r], n);
+  if (col2 < dim2 && n) {
+    atomicAdd(&out[(long long)r * dim + 2 * col2], acc_lo);
+    atomicAdd(&out[(long long)r * dim + 2 * col2 + 1], acc_hi);
+    if (col2 == 0) atomicAdd(&cnt[r], n);
   }
 }
 
@@ -420,7 +433,8 @@ class GpuAdmissionDirect {
     hipLaunchKernelGGL(excl_scan_kernel, dim3(1), dim3(256), 0, st,
                        b.d_counts, n_req, b.d_req_excl,
                        b.d_req_excl + max_req_);
-    hipLaunchKernelGGL(meanpool_tokens_kernel, dim3(n_req, P), dim3(dim), 0,
+    hipLaunchKernelGGL(meanpool_tokens_kernel, dim3(n_req, P),
+                       dim3(dim / 2), 0,
                        st, b.d_out_ids, b.d_ghead, b.d_req_excl, b.d_counts,
                        n_req, d_emb_, dim, P, b.d_pool, b.d_poolcnt);
     hipLaunchKernelGGL(meanpool_div_kernel, dim3(n_req), dim3(dim), 0, st,
