@@ -545,11 +545,16 @@ def run_native_bench(args, rank, world, local_rank, use_gpu, barrier_sync):
     # count — size nogpu mode to quota/8 (64 conns at the 16-CPU quota:
     # 82k @ p99 0.9ms, vs 33k @ p99 88ms at 128 conns). Cache mode is the
     # other direction: every request sleeps a full embed+top-k batch
-    # cycle (~2 ms), so it takes quota/2 workers to keep the quota busy
-    # (measured 58.3k @ w4, 71.1k @ w6, 75.0k @ w8, 73.2k @ w10).
-    div = 2 if cache_mode else (4 if gpu_direct else 8)
-    workers = args.workers if args.workers > 0 else max(
-        1, min(8, _cpu_quota() // (div * max(world, 1))))
+    # cycle, so it takes ~3*quota/4 workers to keep the quota busy
+    # (vectorized-meanpool kernels: 104.1k @ w8, 112.0k @ w10,
+    # 115.3k @ w12 on the 16-CPU quota).
+    if cache_mode:
+        workers = args.workers if args.workers > 0 else max(
+            1, min(12, (3 * _cpu_quota() // 4) // max(world, 1)))
+    else:
+        div = 4 if gpu_direct else 8
+        workers = args.workers if args.workers > 0 else max(
+            1, min(8, _cpu_quota() // (div * max(world, 1))))
     conns = args.batch * workers
     waves = max(args.waves, 1)
     path = "/v1/chat/completions"
