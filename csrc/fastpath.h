@@ -136,7 +136,9 @@ class FastServer {
 
   const ServerStats& stats() const { return stats_; }
   // direct-admission batcher counters: {batches, texts, time_us, max_us,
-  // errors}; zeros when direct admission is off
+  // errors, pack_us, submit_us, wait_us, fulfill_us} — the last four are
+  // the batcher thread's per-phase wall-time sums (pipeline diagnosis);
+  // zeros when direct admission is off
   std::vector<uint64_t> gpu_direct_stats() const;
   // cross-shard rate-limit sync hooks (aigw.parallel.StateSync bridge)
   std::vector<int64_t> rl_collect_deltas();
