@@ -182,3 +182,29 @@ def test_pool_members_register_telemetry(tmp_path):
         await runner.cleanup()
 
     asyncio.run(run())
+
+
+def test_members_file_edge_cases(tmp_path):
+    f = tmp_path / "m"
+    # dedup, default port from pool.port, comments, whitespace
+    f.write_text("""
+# pool members
+10.0.0.1:8000
+10.0.0.1:8000
+10.0.0.2
+  10.0.0.3:9001
+""")
+    pool = InferencePool(members_file=str(f), port=8000)
+    members = asyncio.run(resolve_members(pool))
+    assert members == [("10.0.0.1", 8000), ("10.0.0.2", 8000),
+                       ("10.0.0.3", 9001)]
+
+    # malformed port -> resolution error (caller keeps last-known set)
+    f.write_text("10.0.0.1:not-a-port\n")
+    with pytest.raises(ValueError):
+        asyncio.run(resolve_members(pool))
+
+    # missing file -> RuntimeError
+    f.unlink()
+    with pytest.raises(RuntimeError):
+        asyncio.run(resolve_members(pool))
