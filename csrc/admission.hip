@@ -13,6 +13,13 @@
 //
 // The merge table is supplied by the caller (aigw.ops.tokenizer
 // make_merges) so counts stay bit-identical to the Python/CPU oracle.
+//
+// Reference parity anchor: the reference does NOT tokenize locally —
+// token accounting comes from provider usage JSON
+// (translator/openai_openai.go:185-223) and /tokenize passes through to
+// the backend (translator/tokenize.go:24-81). This module is the
+// MI355X-native replacement BASELINE.json demands: gateway-side GPU
+// token accounting feeding the same llm_*_token cost metadata.
 
 #include <hip/hip_runtime.h>
 
