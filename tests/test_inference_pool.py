@@ -208,3 +208,91 @@ def test_members_file_edge_cases(tmp_path):
     f.unlink()
     with pytest.raises(RuntimeError):
         asyncio.run(resolve_members(pool))
+
+
+class GreedyPickGPU:
+    """CPU stand-in for the KV scorer kernel (same greedy objective;
+    the kernel is oracle-tested on hardware in test_gpu_kernels)."""
+
+    cache_enabled = False
+
+    async def pick_endpoint(self, stats_rows, predicted):
+        def score(row):
+            used, total, waiting, running = row
+            return ((used + predicted) / max(total, 1.0)
+                    + 0.1 * waiting + 0.05 * running)
+
+        return min(range(len(stats_rows)), key=lambda i: score(stats_rows[i]))
+
+    async def count_text_tokens(self, text):
+        return 8
+
+
+def test_picker_over_dynamic_pool_members(tmp_path):
+    """The full integration: members resolved from a file, telemetry
+    scraped from each member, KV-occupancy picker routing to the least
+    loaded — and a freshly scaled-out member joining the SCORED set."""
+
+    async def run():
+        rep_a, rep_b = FakeReplica("a"), FakeReplica("b")
+        runner_a, port_a = await rep_a.start()
+        runner_b, port_b = await rep_b.start()
+        f = tmp_path / "members"
+        f.write_text(f"127.0.0.1:{port_a}\n")
+        cfg = load_config({
+            "routes": [{
+                "name": "pool-route", "endpointPicker": True,
+                "pool": {"membersFile": str(f), "intervalS": 60,
+                         "schema": "OpenAI",
+                         "telemetry": {"path": "/metrics",
+                                       "intervalS": 0.05,
+                                       "kvTotal": 1000}},
+            }],
+        })
+        server = GatewayServer(RuntimeConfig(cfg))
+        server.gpu = GreedyPickGPU()
+        await server.start()
+        mgr = PoolManager(server)
+        await mgr.resolve_once()
+
+        from aigw.extproc.lean_front import serve_lean
+
+        _, port, cleanup = await serve_lean(server, "127.0.0.1", 0,
+                                            with_fallback=False)
+
+        async def burst(n=8):
+            async with aiohttp.ClientSession() as c:
+                for _ in range(n):
+                    async with c.post(
+                        f"http://127.0.0.1:{port}/v1/chat/completions",
+                        json={"model": "m",
+                              "messages": [{"role": "user", "content": "x"}]},
+                    ) as r:
+                        assert r.status == 200
+
+        await burst(4)
+        assert rep_a.served == 4  # only member
+
+        # scale out; B reports LOW occupancy, A HIGH -> picker moves to B
+        rep_a.kv_usage, rep_b.kv_usage = 0.9, 0.05
+        f.write_text(f"127.0.0.1:{port_a}\n127.0.0.1:{port_b}\n")
+        await mgr.resolve_once()  # re-registers telemetry for both
+        await asyncio.sleep(0.15)  # let the poller scrape both members
+        rep_a.served = rep_b.served = 0
+        await burst()
+        assert rep_b.served == 8 and rep_a.served == 0, (
+            rep_a.served, rep_b.served)
+
+        # occupancy flips -> assignment flips on the next scrape
+        rep_a.kv_usage, rep_b.kv_usage = 0.05, 0.9
+        await asyncio.sleep(0.15)
+        rep_a.served = rep_b.served = 0
+        await burst()
+        assert rep_a.served == 8 and rep_b.served == 0
+
+        await cleanup()
+        await server.close()
+        await runner_a.cleanup()
+        await runner_b.cleanup()
+
+    asyncio.run(run())
