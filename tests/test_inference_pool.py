@@ -296,3 +296,53 @@ def test_picker_over_dynamic_pool_members(tmp_path):
         await runner_b.cleanup()
 
     asyncio.run(run())
+
+
+def test_fast_front_relays_picker_pool_route_to_fallback(tmp_path):
+    """--front fast with an endpointPicker pool route: the route is
+    ineligible for the native hot path (the picker runs in Python), so
+    the C++ server must relay it per-request to the fallback app — and
+    the picker must still see scraped occupancy."""
+
+    async def run():
+        rep = FakeReplica("a")
+        runner, port_a = await rep.start()
+        f = tmp_path / "members"
+        f.write_text(f"127.0.0.1:{port_a}\n")
+        cfg = load_config({
+            "routes": [{
+                "name": "pool-route", "endpointPicker": True,
+                "pool": {"membersFile": str(f), "intervalS": 60,
+                         "schema": "OpenAI",
+                         "telemetry": {"path": "/metrics",
+                                       "intervalS": 0.05,
+                                       "kvTotal": 1000}},
+            }],
+        })
+        server = GatewayServer(RuntimeConfig(cfg))
+        server.gpu = GreedyPickGPU()
+        await server.start()
+        mgr = PoolManager(server)
+        await mgr.resolve_once()
+
+        from aigw.extproc.fast_front import FastFront
+
+        front = FastFront(server, server.runtime)
+        port = await front.start("127.0.0.1", 0)
+        async with aiohttp.ClientSession() as c:
+            for _ in range(4):
+                async with c.post(
+                    f"http://127.0.0.1:{port}/v1/chat/completions",
+                    json={"model": "m",
+                          "messages": [{"role": "user", "content": "x"}]},
+                ) as r:
+                    assert r.status == 200
+                    body = await r.json()
+                    assert body["model"] == "a"
+        assert rep.served == 4
+        assert front.fast.stats()["fallback"] >= 4
+        await front.stop()
+        await server.close()
+        await runner.cleanup()
+
+    asyncio.run(run())
