@@ -40,6 +40,10 @@ def parse_prometheus_gauge(text: str, name: str) -> Optional[float]:
             v = float(value_part)
         except ValueError:
             continue
+        # a NaN/Inf gauge from a misbehaving replica would poison the
+        # scorer's min() comparison; treat it as absent
+        if v != v or v in (float("inf"), float("-inf")):
+            continue
         total = v if total is None else total + v
     return total
 

@@ -199,3 +199,34 @@ def test_telemetry_scrape_failure_degrades_gracefully():
         await runner.cleanup()
 
     asyncio.run(run())
+
+
+def test_prometheus_parser_never_crashes():
+    """The scraper feeds arbitrary replica output into the parser; any
+    bytes must parse to a float-or-None, never raise (a misbehaving
+    replica must not take down the poller)."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.text(max_size=300), st.text(min_size=1, max_size=40))
+    def check(text, name):
+        out = parse_prometheus_gauge(text, name)
+        assert out is None or isinstance(out, float)
+
+    check()
+
+
+def test_prometheus_parser_hostile_lines():
+    hostile = (
+        "m NaN\nm +Inf\nm -Inf\n"          # non-finite -> ignored
+        "m{label=\"}\"} 1.5\n"              # brace inside label value
+        "m\n"                                # no value at all
+        "m 1 2 3\n"                          # extra fields (last wins per split)
+        "\x00\x01\x02 m 7\n"                 # binary garbage
+    )
+    out = parse_prometheus_gauge(hostile, "m")
+    # the finite samples sum; NaN/Inf are dropped so the scorer's min()
+    # comparison can never be poisoned
+    assert out is not None and out == out and abs(out) != float("inf")
+    assert parse_prometheus_gauge("m NaN\n", "m") is None
