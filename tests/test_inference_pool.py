@@ -346,3 +346,46 @@ def test_fast_front_relays_picker_pool_route_to_fallback(tmp_path):
         await runner.cleanup()
 
     asyncio.run(run())
+
+
+def test_empty_members_file_is_scale_to_zero(tmp_path):
+    """An empty members file empties the dynamic set: with no static
+    backends the route 503s; repopulating the file restores service."""
+
+    async def run():
+        rep = FakeReplica("a")
+        runner, port_a = await rep.start()
+        f = tmp_path / "members"
+        f.write_text(f"127.0.0.1:{port_a}\n")
+        server = GatewayServer(RuntimeConfig(_pool_cfg(f)))
+        await server.start()
+        mgr = PoolManager(server)
+        await mgr.resolve_once()
+
+        from aigw.extproc.lean_front import serve_lean
+
+        _, port, cleanup = await serve_lean(server, "127.0.0.1", 0,
+                                            with_fallback=False)
+
+        async def status():
+            async with aiohttp.ClientSession() as c:
+                async with c.post(
+                    f"http://127.0.0.1:{port}/v1/chat/completions",
+                    json={"model": "m",
+                          "messages": [{"role": "user", "content": "x"}]},
+                ) as r:
+                    return r.status
+
+        assert await status() == 200
+        f.write_text("")  # scale to zero
+        assert await mgr.resolve_once() is True
+        assert await status() >= 500
+        f.write_text(f"127.0.0.1:{port_a}\n")
+        assert await mgr.resolve_once() is True
+        assert await status() == 200
+
+        await cleanup()
+        await server.close()
+        await runner.cleanup()
+
+    asyncio.run(run())
