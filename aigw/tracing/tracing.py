@@ -207,9 +207,6 @@ class ChatSpanRecorder:
         span.set("aigw.operation", operation)
         span.set("aigw.backend", backend)
 
-    def record_chunk(self, span: Span) -> None:
-        span.add_event("gen_ai.chunk")
-
     def record_response(self, span: Span, usage, response_model: str,
                         finish_reason: str = "") -> None:
         if self.tracer.semconv == "openinference":

@@ -53,15 +53,6 @@ def usage_from_openai(u: dict) -> Usage:
     )
 
 
-def usage_to_openai(u: Usage) -> dict:
-    return {
-        "prompt_tokens": u.input_tokens,
-        "completion_tokens": u.output_tokens,
-        "total_tokens": u.total_tokens or (u.input_tokens + u.output_tokens),
-        "prompt_tokens_details": {"cached_tokens": u.cached_input_tokens},
-        "completion_tokens_details": {"reasoning_tokens": u.reasoning_tokens},
-    }
-
 
 @dataclass
 class RequestTranslation:

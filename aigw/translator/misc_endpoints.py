@@ -9,7 +9,6 @@ from __future__ import annotations
 
 import json
 import re
-import secrets
 
 from aigw.filterapi.config import APISchemaName
 from aigw.translator.base import (
@@ -407,6 +406,3 @@ class OpenAITranscription(_MultipartAudio):
 class OpenAITranslation(_MultipartAudio):
     PATH = "/v1/audio/translations"
 
-
-def make_boundary() -> str:
-    return "aigw" + secrets.token_hex(16)
