@@ -183,3 +183,78 @@ routes:
         mock.stop()
 
     asyncio.run(run())
+
+
+def test_fast_front_stream_deferred_counts_match_oracle():
+    """Streamed requests ENQUEUE their admission count before upstream
+    dispatch and collect it at stream end (deferred path); the deferred
+    count must still be bit-identical to the CPU oracle and feed the
+    usage accounting when the stream carries no usage chunk."""
+    import yaml
+
+    import aigw_fast
+    from aigw.extproc.fast_front import FastFront
+    from aigw.extproc.server import GatewayServer
+    from aigw.extproc.upstream_client import LeanClient
+    from aigw.ops.bpe_ref import BPERef, make_merges
+
+    # SSE stream WITHOUT a usage chunk -> gateway's deferred GPU count
+    # is the only token source
+    frames = (b"data: {\"id\":\"c\",\"object\":\"chat.completion.chunk\","
+              b"\"choices\":[{\"index\":0,\"delta\":{\"content\":\"hi\"},"
+              b"\"finish_reason\":null}]}\n\n"
+              b"data: [DONE]\n\n")
+    head = (b"HTTP/1.1 200 OK\r\ncontent-type: text/event-stream\r\n"
+            b"transfer-encoding: chunked\r\n\r\n")
+    chunked = (b"%x\r\n" % len(frames)) + frames + b"\r\n0\r\n\r\n"
+    canned = head + chunked
+
+    ref = BPERef(make_merges(8192, 1355))
+    text = b"stream accounting check with some longer body text " * 30
+
+    async def run():
+        mock = aigw_fast.FastMock()
+        up_port = mock.start("127.0.0.1", canned.decode("latin1"))
+        cfg = load_config_yaml = yaml.safe_load(f"""
+routes:
+  - name: r
+    backends:
+      - name: b
+        schema: OpenAI
+        upstream: {{host: 127.0.0.1, port: {up_port}}}
+llmRequestCosts:
+  - metadataKey: llm_total_token
+    type: TotalToken
+""")
+        from aigw.filterapi import RuntimeConfig, load_config
+
+        server = GatewayServer(RuntimeConfig(load_config(cfg)))
+        front = FastFront(server, server.runtime, gpu_direct=True,
+                          n_merges=8192)
+        port = await front.start("127.0.0.1", 0)
+        client = LeanClient()
+        payload = json.dumps({
+            "model": "m", "stream": True,
+            "messages": [{"role": "user",
+                          "content": text.decode("latin1")}]}).encode()
+        expected = len(ref.encode_batch([text + b"\n"])[0])
+
+        async def one():
+            r = await client.post(host="127.0.0.1", port=port, tls=False,
+                                  path="/v1/chat/completions",
+                                  headers={"content-type": "application/json"},
+                                  body=payload, timeout_s=60.0)
+            data = await r.read()
+            r.release()
+            assert r.status == 200 and b"[DONE]" in data
+
+        n = 24
+        await asyncio.gather(*(one() for _ in range(n)))
+        st = front.stats()
+        assert st["gpu_tokens"] == expected * n, (st["gpu_tokens"], expected, n)
+        assert st["input_tokens"] == expected * n
+        await client.close()
+        await front.stop()
+        mock.stop()
+
+    asyncio.run(run())
