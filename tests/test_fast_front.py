@@ -656,3 +656,13 @@ def test_fast_front_client_abort_mid_stream():
         mock.stop()
 
     asyncio.run(run())
+
+
+def test_gpu_direct_stats_shape_contract():
+    """bench.py zips gpu_direct_stats() with 9 field names; the shape is
+    a cross-component contract (csrc/fastpath.h), so lock it."""
+    import aigw_fast
+
+    srv = aigw_fast.FastServer()
+    stats = srv.gpu_direct_stats()
+    assert list(stats) == [0] * 9
