@@ -250,8 +250,14 @@ def tracing_from_env(env: Optional[dict] = None) -> Optional[Tracer]:
         exporter = ConsoleExporter()
     else:
         return None
+    # span header->attribute mapping. Unset defaults to session-id
+    # tracking; an EXPLICITLY EMPTY value clears the default
+    # (requestheaderattrs/resolve.go:15-21 semantics)
+    raw = env.get("AIGW_SPAN_REQUEST_HEADER_ATTRIBUTES")
+    if raw is None:
+        raw = "agent-session-id:session.id"
     header_attrs = {}
-    for pair in env.get("AIGW_SPAN_REQUEST_HEADER_ATTRIBUTES", "").split(","):
+    for pair in raw.split(","):
         if ":" in pair:
             h, a = pair.split(":", 1)
             header_attrs[h.strip()] = a.strip()

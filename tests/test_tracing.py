@@ -88,3 +88,23 @@ def test_gateway_emits_spans_and_injects_traceparent():
         await up_runner.cleanup()
 
     asyncio.run(main())
+
+
+def test_session_id_header_attribute_default():
+    """Unset span header-attrs default to agent-session-id -> session.id;
+    an explicitly EMPTY value clears the default
+    (requestheaderattrs/resolve.go semantics)."""
+    from aigw.tracing.tracing import tracing_from_env as build_tracer
+
+    env = {"OTEL_EXPORTER_OTLP_ENDPOINT": "http://127.0.0.1:9",
+           "OTEL_TRACES_EXPORTER": "otlp"}
+    t = build_tracer(env)
+    assert t.header_attributes == {"agent-session-id": "session.id"}
+
+    t = build_tracer({**env, "AIGW_SPAN_REQUEST_HEADER_ATTRIBUTES": ""})
+    assert t.header_attributes == {}
+
+    t = build_tracer({**env,
+                      "AIGW_SPAN_REQUEST_HEADER_ATTRIBUTES":
+                          "x-team:team.id, x-env:deploy.env"})
+    assert t.header_attributes == {"x-team": "team.id", "x-env": "deploy.env"}
