@@ -301,6 +301,19 @@ class GatewayServer:
         self.telemetry = None
         # extra Prometheus providers appended to /metrics (native front)
         self.metrics_extra: list = []
+        # access-log header->attribute mapping; unset defaults to
+        # session tracking, explicitly empty clears it
+        # (requestheaderattrs/resolve.go:15-21)
+        import os as _osm
+
+        raw = _osm.environ.get("AIGW_LOG_REQUEST_HEADER_ATTRIBUTES")
+        if raw is None:
+            raw = "agent-session-id:session.id"
+        self.log_header_attrs: dict[str, str] = {}
+        for pair in raw.split(","):
+            if ":" in pair:
+                h, a = pair.split(":", 1)
+                self.log_header_attrs[h.strip().lower()] = a.strip()
 
     # ---- lifecycle -----------------------------------------------------------
 
@@ -945,6 +958,7 @@ class GatewayServer:
     def _finish_metrics(
         self, endpoint, route, backend, model, response_model, usage, start,
         *, status: int, error_type: str = "", ttft: float = -1.0,
+        headers=None,
     ) -> None:
         provider = (
             provider_from_schema(backend.schema.name.value, backend.name) if backend else ""
@@ -985,6 +999,11 @@ class GatewayServer:
                         "total_tokens": usage.total_tokens,
                         "ttft_ms": round(ttft * 1000.0, 2) if ttft >= 0 else None,
                         "error_type": error_type or None,
+                        # header->attribute mapping (requestheaderattrs
+                        # defaults: agent-session-id -> session.id)
+                        **{attr: headers.get(h)
+                           for h, attr in self.log_header_attrs.items()
+                           if headers is not None and headers.get(h)},
                     },
                     separators=(",", ":"),
                 )
@@ -1010,7 +1029,7 @@ class GatewayServer:
             self.span_recorder.record_response(span, usage, rtl.response_model)
         self._finish_metrics(
             endpoint, route, backend, model, rtl.response_model, usage, start,
-            status=upstream.status,
+            status=upstream.status, headers=headers,
         )
         hdrs = translator.response_headers(upstream.status, upstream.headers)
         content_type = hdrs.get("content-type") or upstream.headers.get(
