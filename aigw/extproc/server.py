@@ -1131,6 +1131,7 @@ class GatewayServer:
             self._finish_metrics(
                 endpoint, route, backend, model, response_model, usage, start,
                 status=upstream.status, ttft=ttft if ttft >= 0 else 0.0,
+                headers=headers,
             )
         if cut:
             abort = getattr(writer, "abort", None)
